@@ -1,0 +1,884 @@
+// kernels.hip — MI355X (gfx950/CDNA4) kernels + C-ABI host layer of
+// libballista_gpu.so, the GPU stage executor behind Ballista's
+// ExecutionEngine seam (see include/ballista_gpu.h for the ABI contract and
+// per-entry-point reference citations).
+//
+// Design notes (MI355X-first; /opt/skills/guides/cdna_hip_programming.md):
+//  - Everything here is HBM-bound integer/byte work (scan, compare, hash,
+//    split, gather, 128-bit accumulate) — no dense contraction, no MFMA.
+//  - wave64 is the unit: selection masks are built with 64-bit __ballot
+//    (Arrow LSB bitmap order == ballot lane order), stable compaction and
+//    the stable multi-split use wave-contiguous row chunks so row order is
+//    preserved without cross-wave coordination.
+//  - Loads are coalesced full-width: Decimal128 as ulong2 (16 B/lane),
+//    Date32 as i32 (256 B/wave).  Grid-stride loops capped at ~2048 blocks
+//    (guide §6 G11).
+//  - Exact Decimal128 arithmetic: two's-complement i128 adds/muls are
+//    wrap-exact and associative, so block/wave reduction order never changes
+//    the result; global accumulation uses the u64 carry-propagating
+//    atomicAdd pair (deterministic, order-independent).
+//  - Grouped aggregation clusters the wave's rows by group with ballot
+//    leader-loops (few distinct groups per wave in the TPC-H shapes), does
+//    register wave-reductions, and lands ONE LDS atomic set per
+//    (group, wave-iteration) instead of per row.
+
+#include <hip/hip_runtime.h>
+
+#include <cstdio>
+#include <cstring>
+
+#include "../../include/ballista_gpu.h"
+#include "bg_ahash.h"
+
+#define BG_BLOCK 256
+#define BG_MAX_BLOCKS 2048
+#define BG_WAVE 64
+#define BG_MAX_PREDS 8
+#define BG_MAX_KEYS 4
+#define BG_MAX_PAYLOAD 16
+
+static inline int64_t bg_imin64(int64_t a, int64_t b) { return a < b ? a : b; }
+
+// ---------------------------------------------------------------------------
+// error plumbing
+// ---------------------------------------------------------------------------
+static thread_local char g_err[512] = "";
+static thread_local bool g_inited = false;
+
+extern "C" const char* bg_last_error(void) { return g_err; }
+extern "C" int bg_version(void) { return 10; }
+
+static int set_err(int code, const char* msg) {
+  snprintf(g_err, sizeof(g_err), "%s", msg);
+  return code;
+}
+
+static int set_hip_err(hipError_t e, const char* what) {
+  snprintf(g_err, sizeof(g_err), "%s: %s", what, hipGetErrorString(e));
+  return BG_ERR_HIP;
+}
+
+#define HIP_TRY(call)                                   \
+  do {                                                  \
+    hipError_t _e = (call);                             \
+    if (_e != hipSuccess) return set_hip_err(_e, #call); \
+  } while (0)
+
+#define REQUIRE_INIT()                                                       \
+  do {                                                                       \
+    if (!g_inited)                                                           \
+      return set_err(BG_ERR_NO_GPU,                                          \
+                     "ballista_gpu: bg_init() not called or no gfx950 GPU "  \
+                     "present — the GPU stage executor has NO CPU fallback"); \
+  } while (0)
+
+// ---------------------------------------------------------------------------
+// session
+// ---------------------------------------------------------------------------
+extern "C" int bg_init(int device_ordinal) {
+  int n = 0;
+  hipError_t e = hipGetDeviceCount(&n);
+  if (e != hipSuccess || n == 0)
+    return set_err(BG_ERR_NO_GPU, "ballista_gpu: no HIP device visible");
+  HIP_TRY(hipSetDevice(device_ordinal));
+  // touch the device so a broken runtime fails here, loudly
+  HIP_TRY(hipFree(nullptr));
+  g_inited = true;
+  return BG_OK;
+}
+
+extern "C" int bg_device_count(int* out) {
+  int n = 0;
+  hipError_t e = hipGetDeviceCount(&n);
+  if (e != hipSuccess) n = 0;
+  *out = n;
+  return BG_OK;
+}
+
+extern "C" int bg_synchronize(void) {
+  REQUIRE_INIT();
+  HIP_TRY(hipDeviceSynchronize());
+  return BG_OK;
+}
+
+extern "C" int bg_malloc(uint64_t bytes, void** d_ptr) {
+  REQUIRE_INIT();
+  HIP_TRY(hipMalloc(d_ptr, bytes));
+  return BG_OK;
+}
+extern "C" int bg_free(void* d_ptr) {
+  REQUIRE_INIT();
+  HIP_TRY(hipFree(d_ptr));
+  return BG_OK;
+}
+extern "C" int bg_memset(void* d_ptr, int value, uint64_t bytes) {
+  REQUIRE_INIT();
+  HIP_TRY(hipMemset(d_ptr, value, bytes));
+  return BG_OK;
+}
+extern "C" int bg_memcpy_h2d(void* d_dst, const void* h_src, uint64_t bytes) {
+  REQUIRE_INIT();
+  HIP_TRY(hipMemcpy(d_dst, h_src, bytes, hipMemcpyHostToDevice));
+  return BG_OK;
+}
+extern "C" int bg_memcpy_d2h(void* h_dst, const void* d_src, uint64_t bytes) {
+  REQUIRE_INIT();
+  HIP_TRY(hipMemcpy(h_dst, d_src, bytes, hipMemcpyDeviceToHost));
+  return BG_OK;
+}
+
+// ---------------------------------------------------------------------------
+// device helpers
+// ---------------------------------------------------------------------------
+using u64 = unsigned long long;
+using i64 = long long;
+using u128 = unsigned __int128;
+using i128 = __int128;
+
+__device__ __forceinline__ int lane_id() { return threadIdx.x & (BG_WAVE - 1); }
+
+__device__ __forceinline__ bool bit_valid(const uint8_t* bm, int64_t i) {
+  return bm == nullptr || ((bm[i >> 3] >> (i & 7)) & 1);
+}
+
+__device__ __forceinline__ i128 make_i128(u64 lo, i64 hi) {
+  return ((i128)hi << 64) | (i128)(u128)lo;
+}
+
+__device__ __forceinline__ i128 load_dec128(const void* base, int64_t row) {
+  const ulong2 v = reinterpret_cast<const ulong2*>(base)[row];
+  return make_i128(v.x, (i64)v.y);
+}
+
+// order-independent exact i128 accumulate into a global (lo, hi) pair
+__device__ __forceinline__ void atomic_add_i128(u64* lo, u64* hi, i128 v) {
+  u64 vlo = (u64)(u128)v;
+  u64 vhi = (u64)((u128)v >> 64);
+  u64 old = atomicAdd(lo, vlo);
+  u64 carry = (old + vlo) < vlo ? 1ull : 0ull;
+  if (vhi + carry) atomicAdd(hi, vhi + carry);
+}
+
+__device__ __forceinline__ void atomic_add_i128_lds(u64* lo, u64* hi, i128 v) {
+  u64 vlo = (u64)(u128)v;
+  u64 vhi = (u64)((u128)v >> 64);
+  u64 old = atomicAdd(lo, vlo);
+  u64 carry = (old + vlo) < vlo ? 1ull : 0ull;
+  if (vhi + carry) atomicAdd(hi, vhi + carry);
+}
+
+// full-wave i128 sum (all 64 lanes get the total)
+__device__ __forceinline__ i128 wave_reduce_i128(i128 v) {
+  u64 lo = (u64)(u128)v;
+  u64 hi = (u64)((u128)v >> 64);
+#pragma unroll
+  for (int s = 32; s > 0; s >>= 1) {
+    u64 olo = (u64)__shfl_xor((long long)lo, s, BG_WAVE);
+    u64 ohi = (u64)__shfl_xor((long long)hi, s, BG_WAVE);
+    u64 nlo = lo + olo;
+    hi = hi + ohi + (nlo < olo ? 1 : 0);
+    lo = nlo;
+  }
+  return make_i128(lo, (i64)hi);
+}
+
+// ---------------------------------------------------------------------------
+// predicate evaluation -> Arrow LSB bitmask
+// ---------------------------------------------------------------------------
+struct PredDev {
+  const void* data;
+  const uint8_t* valid;
+  int dtype;
+  int op;
+  u64 lo_lo; i64 lo_hi;
+  u64 hi_lo; i64 hi_hi;
+};
+
+struct PredArgs {
+  int npreds;
+  PredDev p[BG_MAX_PREDS];
+};
+
+__device__ __forceinline__ bool pred_eval(i128 x, int op, i128 lo, i128 hi) {
+  switch (op) {
+    case BG_PRED_GE_LT: return x >= lo && x < hi;
+    case BG_PRED_BETWEEN: return x >= lo && x <= hi;
+    case BG_PRED_LT: return x < hi;
+    case BG_PRED_EQ: return x == lo;
+    case BG_PRED_GT: return x > lo;
+    default: return false;
+  }
+}
+
+// one wave evaluates 64 consecutive rows per step; ballot bit order ==
+// Arrow LSB bitmap order, lane 0 stores the u64 word.
+__global__ void k_eval_predicates(PredArgs args, int64_t n, u64* mask_words,
+                                  int64_t nwords) {
+  const int64_t wave_global =
+      ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) / BG_WAVE;
+  const int64_t nwaves = ((int64_t)gridDim.x * blockDim.x) / BG_WAVE;
+  for (int64_t w = wave_global; w < nwords; w += nwaves) {
+    const int64_t row = w * BG_WAVE + lane_id();
+    bool keep = row < n;
+    if (keep) {
+      for (int pi = 0; pi < args.npreds; ++pi) {
+        const PredDev& pr = args.p[pi];
+        if (!bit_valid(pr.valid, row)) { keep = false; break; }
+        i128 x;
+        switch (pr.dtype) {
+          case BG_DT_INT32:
+          case BG_DT_DATE32:
+            x = (i128) reinterpret_cast<const int32_t*>(pr.data)[row];
+            break;
+          case BG_DT_INT64:
+            x = (i128) reinterpret_cast<const int64_t*>(pr.data)[row];
+            break;
+          case BG_DT_DECIMAL128:
+            x = load_dec128(pr.data, row);
+            break;
+          case BG_DT_DICT8:
+            x = (i128) reinterpret_cast<const uint8_t*>(pr.data)[row];
+            break;
+          default:
+            x = 0;
+        }
+        if (!pred_eval(x, pr.op, make_i128(pr.lo_lo, pr.lo_hi),
+                       make_i128(pr.hi_lo, pr.hi_hi))) {
+          keep = false;
+          break;
+        }
+      }
+    }
+    u64 m = __ballot(keep);
+    if (lane_id() == 0) mask_words[w] = m;
+  }
+}
+
+extern "C" int bg_eval_predicates(const bg_column* cols, int32_t ncols,
+                                  const bg_pred* preds, int32_t npreds,
+                                  int64_t n, uint8_t* d_mask) {
+  REQUIRE_INIT();
+  if (npreds <= 0 || npreds > BG_MAX_PREDS)
+    return set_err(BG_ERR_INVALID, "npreds out of range [1,8]");
+  PredArgs a{};
+  a.npreds = npreds;
+  for (int i = 0; i < npreds; ++i) {
+    int c = preds[i].column;
+    if (c < 0 || c >= ncols) return set_err(BG_ERR_INVALID, "pred column oob");
+    a.p[i].data = cols[c].d_data;
+    a.p[i].valid = cols[c].d_validity;
+    a.p[i].dtype = cols[c].dtype;
+    a.p[i].op = preds[i].op;
+    a.p[i].lo_lo = (u64)preds[i].lo_lo;
+    a.p[i].lo_hi = preds[i].lo_hi;
+    a.p[i].hi_lo = (u64)preds[i].hi_lo;
+    a.p[i].hi_hi = preds[i].hi_hi;
+  }
+  int64_t nwords = (n + 63) / 64;
+  int blocks = (int)bg_imin64((nwords * BG_WAVE + BG_BLOCK - 1) / BG_BLOCK,
+                                 BG_MAX_BLOCKS);
+  if (blocks == 0) blocks = 1;
+  hipLaunchKernelGGL(k_eval_predicates, dim3(blocks), dim3(BG_BLOCK), 0, 0, a,
+                     n, reinterpret_cast<u64*>(d_mask), nwords);
+  HIP_TRY(hipGetLastError());
+  return BG_OK;
+}
+
+// ---------------------------------------------------------------------------
+// generic exclusive scan (single block; m up to a few million)
+// used for compaction offsets and the multi-split histogram
+// ---------------------------------------------------------------------------
+__global__ void k_exclusive_scan_i64(const u64* in, int64_t m, i64* out,
+                                     i64* total) {
+  // chunked serial scan: thread t owns chunk [t*ch, (t+1)*ch)
+  __shared__ i64 sums[BG_BLOCK];
+  __shared__ i64 carry;
+  const int t = threadIdx.x;
+  const int64_t ch = (m + blockDim.x - 1) / blockDim.x;
+  const int64_t lo = (int64_t)t * ch, hi = min(lo + ch, m);
+  i64 s = 0;
+  for (int64_t i = lo; i < hi; ++i) s += (i64)in[i];
+  sums[t] = s;
+  __syncthreads();
+  if (t == 0) {
+    i64 acc = 0;
+    for (int i = 0; i < (int)blockDim.x; ++i) {
+      i64 v = sums[i];
+      sums[i] = acc;
+      acc += v;
+    }
+    carry = acc;
+  }
+  __syncthreads();
+  i64 acc = sums[t];
+  for (int64_t i = lo; i < hi; ++i) {
+    i64 v = (i64)in[i];
+    out[i] = acc;
+    acc += v;
+  }
+  if (t == 0 && total) *total = carry;
+}
+
+// ---------------------------------------------------------------------------
+// stable mask compaction
+// ---------------------------------------------------------------------------
+#define MC_WORDS_PER_WAVE 1024  // 65536 rows per wave chunk
+
+__global__ void k_mask_count(const u64* words, int64_t nwords, int64_t n,
+                             u64* wave_counts, int64_t nchunks) {
+  const int64_t wave_global =
+      ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) / BG_WAVE;
+  const int64_t nwaves = ((int64_t)gridDim.x * blockDim.x) / BG_WAVE;
+  for (int64_t c = wave_global; c < nchunks; c += nwaves) {
+    const int64_t w0 = c * MC_WORDS_PER_WAVE;
+    const int64_t w1 = min(w0 + (int64_t)MC_WORDS_PER_WAVE, nwords);
+    u64 cnt = 0;
+    for (int64_t w = w0 + lane_id(); w < w1; w += BG_WAVE) {
+      u64 m = words[w];
+      // trim tail bits beyond n in the last word
+      if (w == nwords - 1 && (n & 63)) m &= (1ull << (n & 63)) - 1;
+      cnt += __popcll(m);
+    }
+    // wave sum
+#pragma unroll
+    for (int s = 32; s > 0; s >>= 1) cnt += (u64)__shfl_xor((long long)cnt, s, BG_WAVE);
+    if (lane_id() == 0) wave_counts[c] = cnt;
+  }
+}
+
+__global__ void k_mask_scatter(const u64* words, int64_t nwords, int64_t n,
+                               const i64* chunk_offsets, int64_t nchunks,
+                               uint32_t* out) {
+  const int64_t wave_global =
+      ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) / BG_WAVE;
+  const int64_t nwaves = ((int64_t)gridDim.x * blockDim.x) / BG_WAVE;
+  for (int64_t c = wave_global; c < nchunks; c += nwaves) {
+    const int64_t w0 = c * MC_WORDS_PER_WAVE;
+    const int64_t w1 = min(w0 + (int64_t)MC_WORDS_PER_WAVE, nwords);
+    i64 base = chunk_offsets[c];
+    for (int64_t w = w0; w < w1; ++w) {  // in order => stability
+      u64 m = words[w];
+      if (w == nwords - 1 && (n & 63)) m &= (1ull << (n & 63)) - 1;
+      const int l = lane_id();
+      if ((m >> l) & 1) {
+        int pos = __popcll(m & ((1ull << l) - 1));
+        out[base + pos] = (uint32_t)(w * BG_WAVE + l);
+      }
+      base += __popcll(m);
+    }
+  }
+}
+
+extern "C" int bg_mask_to_indices(const uint8_t* d_mask, int64_t n,
+                                  uint32_t* d_indices, int64_t* out_count) {
+  REQUIRE_INIT();
+  const int64_t nwords = (n + 63) / 64;
+  const int64_t nchunks = (nwords + MC_WORDS_PER_WAVE - 1) / MC_WORDS_PER_WAVE;
+  u64* d_counts;
+  i64* d_offs;
+  i64* d_total;
+  HIP_TRY(hipMalloc(&d_counts, sizeof(u64) * (nchunks ? nchunks : 1)));
+  HIP_TRY(hipMalloc(&d_offs, sizeof(i64) * (nchunks ? nchunks : 1)));
+  HIP_TRY(hipMalloc(&d_total, sizeof(i64)));
+  int blocks = (int)bg_imin64(nchunks, BG_MAX_BLOCKS);
+  if (blocks == 0) blocks = 1;
+  hipLaunchKernelGGL(k_mask_count, dim3(blocks), dim3(BG_BLOCK), 0, 0,
+                     reinterpret_cast<const u64*>(d_mask), nwords, n, d_counts,
+                     nchunks);
+  hipLaunchKernelGGL(k_exclusive_scan_i64, dim3(1), dim3(BG_BLOCK), 0, 0,
+                     d_counts, nchunks, d_offs, d_total);
+  hipLaunchKernelGGL(k_mask_scatter, dim3(blocks), dim3(BG_BLOCK), 0, 0,
+                     reinterpret_cast<const u64*>(d_mask), nwords, n, d_offs,
+                     nchunks, d_indices);
+  i64 total = 0;
+  HIP_TRY(hipMemcpy(&total, d_total, sizeof(i64), hipMemcpyDeviceToHost));
+  HIP_TRY(hipFree(d_counts));
+  HIP_TRY(hipFree(d_offs));
+  HIP_TRY(hipFree(d_total));
+  *out_count = total;
+  return BG_OK;
+}
+
+// ---------------------------------------------------------------------------
+// gather (take)
+// ---------------------------------------------------------------------------
+template <typename T>
+__global__ void k_gather(const T* src, const uint32_t* idx, int64_t m, T* dst) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < m;
+       i += (int64_t)gridDim.x * blockDim.x)
+    dst[i] = src[idx[i]];
+}
+
+extern "C" int bg_gather(const void* d_src, int64_t elem_size,
+                         const uint32_t* d_idx, int64_t m, void* d_dst) {
+  REQUIRE_INIT();
+  int blocks = (int)bg_imin64((m + BG_BLOCK - 1) / BG_BLOCK, BG_MAX_BLOCKS);
+  if (blocks == 0) blocks = 1;
+  switch (elem_size) {
+    case 1:
+      hipLaunchKernelGGL(k_gather<uint8_t>, dim3(blocks), dim3(BG_BLOCK), 0, 0,
+                         (const uint8_t*)d_src, d_idx, m, (uint8_t*)d_dst);
+      break;
+    case 2:
+      hipLaunchKernelGGL(k_gather<uint16_t>, dim3(blocks), dim3(BG_BLOCK), 0, 0,
+                         (const uint16_t*)d_src, d_idx, m, (uint16_t*)d_dst);
+      break;
+    case 4:
+      hipLaunchKernelGGL(k_gather<uint32_t>, dim3(blocks), dim3(BG_BLOCK), 0, 0,
+                         (const uint32_t*)d_src, d_idx, m, (uint32_t*)d_dst);
+      break;
+    case 8:
+      hipLaunchKernelGGL(k_gather<u64>, dim3(blocks), dim3(BG_BLOCK), 0, 0,
+                         (const u64*)d_src, d_idx, m, (u64*)d_dst);
+      break;
+    case 16:
+      hipLaunchKernelGGL(k_gather<ulong2>, dim3(blocks), dim3(BG_BLOCK), 0, 0,
+                         (const ulong2*)d_src, d_idx, m, (ulong2*)d_dst);
+      break;
+    default:
+      return set_err(BG_ERR_INVALID, "elem_size must be 1/2/4/8/16");
+  }
+  HIP_TRY(hipGetLastError());
+  return BG_OK;
+}
+
+// ---------------------------------------------------------------------------
+// create_hashes (bg_ahash.h restatement) + partition ids
+// ---------------------------------------------------------------------------
+struct KeyArgs {
+  int nkeys;
+  struct {
+    const void* data;
+    const uint8_t* valid;
+    int dtype;
+  } k[BG_MAX_KEYS];
+};
+
+__global__ void k_hash_columns(KeyArgs keys, int64_t n, u64* hashes) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    u64 h = 0;
+    for (int c = 0; c < keys.nkeys; ++c) {
+      if (!bit_valid(keys.k[c].valid, i)) continue;  // null: leave running hash
+      u64 hc;
+      switch (keys.k[c].dtype) {
+        case BG_DT_INT64:
+          hc = bg_hash_u64((u64) reinterpret_cast<const int64_t*>(keys.k[c].data)[i]);
+          break;
+        case BG_DT_INT32:
+        case BG_DT_DATE32:
+          hc = bg_hash_u32((uint32_t) reinterpret_cast<const int32_t*>(keys.k[c].data)[i]);
+          break;
+        case BG_DT_DECIMAL128: {
+          const ulong2 v = reinterpret_cast<const ulong2*>(keys.k[c].data)[i];
+          hc = bg_hash_u128(v.x, v.y);
+          break;
+        }
+        case BG_DT_DICT8:
+          // dictionary codes hash as their u8 value zero-extended (the host
+          // maps codes back to dictionary values before hashing when exact
+          // Utf8 hash parity is required; DICT8 here is for synthetic keys)
+          hc = bg_hash_u64((u64) reinterpret_cast<const uint8_t*>(keys.k[c].data)[i]);
+          break;
+        default:
+          hc = 0;
+      }
+      h = (c == 0) ? hc : bg_combine_hashes(hc, h);
+    }
+    hashes[i] = h;
+  }
+}
+
+extern "C" int bg_hash_columns(const bg_column* key_cols, int32_t nkeys,
+                               int64_t n, uint64_t* d_hashes) {
+  REQUIRE_INIT();
+  if (nkeys <= 0 || nkeys > BG_MAX_KEYS)
+    return set_err(BG_ERR_INVALID, "nkeys out of range [1,4]");
+  KeyArgs a{};
+  a.nkeys = nkeys;
+  for (int i = 0; i < nkeys; ++i) {
+    a.k[i].data = key_cols[i].d_data;
+    a.k[i].valid = key_cols[i].d_validity;
+    a.k[i].dtype = key_cols[i].dtype;
+  }
+  int blocks = (int)bg_imin64((n + BG_BLOCK - 1) / BG_BLOCK, BG_MAX_BLOCKS);
+  if (blocks == 0) blocks = 1;
+  hipLaunchKernelGGL(k_hash_columns, dim3(blocks), dim3(BG_BLOCK), 0, 0, a, n,
+                     reinterpret_cast<u64*>(d_hashes));
+  HIP_TRY(hipGetLastError());
+  return BG_OK;
+}
+
+__global__ void k_partition_ids(const u64* hashes, int64_t n, uint32_t k,
+                                uint32_t* pids) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x)
+    pids[i] = (uint32_t)(hashes[i] % (u64)k);
+}
+
+extern "C" int bg_partition_ids(const uint64_t* d_hashes, int64_t n, uint32_t k,
+                                uint32_t* d_pids) {
+  REQUIRE_INIT();
+  if (k == 0) return set_err(BG_ERR_INVALID, "k == 0");
+  int blocks = (int)bg_imin64((n + BG_BLOCK - 1) / BG_BLOCK, BG_MAX_BLOCKS);
+  if (blocks == 0) blocks = 1;
+  hipLaunchKernelGGL(k_partition_ids, dim3(blocks), dim3(BG_BLOCK), 0, 0,
+                     reinterpret_cast<const u64*>(d_hashes), n, k, d_pids);
+  HIP_TRY(hipGetLastError());
+  return BG_OK;
+}
+
+// ---------------------------------------------------------------------------
+// stable multi-split (compute_partition_indices device equivalent)
+// layout: hist[p * nchunks + c] -> p-major so the exclusive scan directly
+// yields each (partition, chunk) start and offsets[p] = scan[p * nchunks].
+// ---------------------------------------------------------------------------
+#define PS_ROWS_PER_WAVE 16384
+
+__global__ void k_part_hist(const uint32_t* pids, int64_t n, uint32_t k,
+                            u64* hist, int64_t nchunks) {
+  extern __shared__ __attribute__((aligned(16))) char smem_raw[];
+  // per-wave counters: waves_per_block * k u32
+  uint32_t* cnt = reinterpret_cast<uint32_t*>(smem_raw);
+  const int wave_in_block = threadIdx.x / BG_WAVE;
+  uint32_t* my = cnt + (size_t)wave_in_block * k;
+  const int64_t wave_global =
+      ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) / BG_WAVE;
+  const int64_t nwaves = ((int64_t)gridDim.x * blockDim.x) / BG_WAVE;
+  for (int64_t c = wave_global; c < nchunks; c += nwaves) {
+    for (uint32_t p = lane_id(); p < k; p += BG_WAVE) my[p] = 0;
+    __builtin_amdgcn_wave_barrier();
+    const int64_t r0 = c * PS_ROWS_PER_WAVE;
+    const int64_t r1 = min(r0 + (int64_t)PS_ROWS_PER_WAVE, n);
+    for (int64_t r = r0 + lane_id(); r < r1; r += BG_WAVE)
+      atomicAdd(&my[pids[r]], 1u);
+    __builtin_amdgcn_wave_barrier();
+    for (uint32_t p = lane_id(); p < k; p += BG_WAVE)
+      hist[(int64_t)p * nchunks + c] = my[p];
+  }
+}
+
+__global__ void k_part_scatter(const uint32_t* pids, int64_t n, uint32_t k,
+                               const i64* start, int64_t nchunks,
+                               uint32_t* out) {
+  extern __shared__ __attribute__((aligned(16))) char smem_raw[];
+  i64* cur = reinterpret_cast<i64*>(smem_raw);  // waves_per_block * k
+  const int wave_in_block = threadIdx.x / BG_WAVE;
+  i64* my = cur + (size_t)wave_in_block * k;
+  const int64_t wave_global =
+      ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) / BG_WAVE;
+  const int64_t nwaves = ((int64_t)gridDim.x * blockDim.x) / BG_WAVE;
+  for (int64_t c = wave_global; c < nchunks; c += nwaves) {
+    for (uint32_t p = lane_id(); p < k; p += BG_WAVE)
+      my[p] = start[(int64_t)p * nchunks + c];
+    __builtin_amdgcn_wave_barrier();
+    const int64_t r0 = c * PS_ROWS_PER_WAVE;
+    const int64_t r1 = min(r0 + (int64_t)PS_ROWS_PER_WAVE, n);
+    for (int64_t rb = r0; rb < r1; rb += BG_WAVE) {
+      const int64_t r = rb + lane_id();
+      const bool active = r < r1;
+      uint32_t pid = active ? pids[r] : 0xffffffffu;
+      u64 remaining = __ballot(active);
+      while (remaining) {
+        const int leader = __ffsll((i64)remaining) - 1;
+        const uint32_t pg = (uint32_t)__shfl((int)pid, leader, BG_WAVE);
+        const u64 m = __ballot(active && pid == pg);
+        const i64 base = my[pg];
+        if (active && pid == pg) {
+          const int pos = __popcll(m & ((1ull << lane_id()) - 1));
+          out[base + pos] = (uint32_t)r;
+        }
+        __builtin_amdgcn_wave_barrier();
+        if (lane_id() == leader) my[pg] = base + __popcll(m);
+        __builtin_amdgcn_wave_barrier();
+        remaining &= ~m;
+      }
+    }
+  }
+}
+
+__global__ void k_extract_offsets(const i64* start, int64_t nchunks, uint32_t k,
+                                  int64_t n, int64_t* offsets) {
+  for (uint32_t p = blockIdx.x * blockDim.x + threadIdx.x; p <= k;
+       p += gridDim.x * blockDim.x)
+    offsets[p] = (p == k) ? n : start[(int64_t)p * nchunks];
+}
+
+extern "C" int bg_partition_indices(const uint32_t* d_pids, int64_t n,
+                                    uint32_t k, uint32_t* d_indices,
+                                    int64_t* d_offsets) {
+  REQUIRE_INIT();
+  if (k == 0 || k > 4096) return set_err(BG_ERR_INVALID, "k out of range [1,4096]");
+  const int64_t nchunks = (n + PS_ROWS_PER_WAVE - 1) / PS_ROWS_PER_WAVE;
+  const int waves_per_block = BG_BLOCK / BG_WAVE;
+  const size_t lds_hist = (size_t)waves_per_block * k * sizeof(uint32_t);
+  const size_t lds_scat = (size_t)waves_per_block * k * sizeof(i64);
+  u64* d_hist;
+  i64* d_start;
+  const int64_t hist_len = (int64_t)k * (nchunks ? nchunks : 1);
+  HIP_TRY(hipMalloc(&d_hist, sizeof(u64) * hist_len));
+  HIP_TRY(hipMalloc(&d_start, sizeof(i64) * hist_len));
+  int blocks = (int)bg_imin64(nchunks, BG_MAX_BLOCKS);
+  if (blocks == 0) blocks = 1;
+  hipLaunchKernelGGL(k_part_hist, dim3(blocks), dim3(BG_BLOCK), lds_hist, 0,
+                     d_pids, n, k, d_hist, nchunks);
+  hipLaunchKernelGGL(k_exclusive_scan_i64, dim3(1), dim3(BG_BLOCK), 0, 0,
+                     d_hist, hist_len, d_start, nullptr);
+  hipLaunchKernelGGL(k_extract_offsets, dim3(1), dim3(BG_BLOCK), 0, 0, d_start,
+                     nchunks, k, n, d_offsets);
+  hipLaunchKernelGGL(k_part_scatter, dim3(blocks), dim3(BG_BLOCK), lds_scat, 0,
+                     d_pids, n, k, d_start, nchunks, d_indices);
+  HIP_TRY(hipGetLastError());
+  HIP_TRY(hipDeviceSynchronize());
+  HIP_TRY(hipFree(d_hist));
+  HIP_TRY(hipFree(d_start));
+  return BG_OK;
+}
+
+static int64_t dtype_size(int dtype) {
+  switch (dtype) {
+    case BG_DT_INT32:
+    case BG_DT_DATE32: return 4;
+    case BG_DT_INT64: return 8;
+    case BG_DT_DECIMAL128: return 16;
+    case BG_DT_DICT8: return 1;
+    default: return 0;
+  }
+}
+
+extern "C" int bg_hash_repartition(const bg_column* key_cols, int32_t nkeys,
+                                   const bg_column* payload_cols, int32_t ncols,
+                                   int64_t n, uint32_t k, uint32_t* d_indices,
+                                   int64_t* d_offsets, void** d_out) {
+  REQUIRE_INIT();
+  if (ncols > BG_MAX_PAYLOAD) return set_err(BG_ERR_INVALID, "too many payload cols");
+  uint64_t* d_hashes;
+  uint32_t* d_pids;
+  HIP_TRY(hipMalloc(&d_hashes, sizeof(u64) * (n ? n : 1)));
+  HIP_TRY(hipMalloc(&d_pids, sizeof(uint32_t) * (n ? n : 1)));
+  int rc = bg_hash_columns(key_cols, nkeys, n, d_hashes);
+  if (rc == BG_OK) rc = bg_partition_ids(d_hashes, n, k, d_pids);
+  if (rc == BG_OK) rc = bg_partition_indices(d_pids, n, k, d_indices, d_offsets);
+  if (rc == BG_OK) {
+    for (int c = 0; c < ncols && rc == BG_OK; ++c) {
+      int64_t esz = dtype_size(payload_cols[c].dtype);
+      if (esz == 0) { rc = set_err(BG_ERR_UNSUPPORTED, "payload dtype"); break; }
+      rc = bg_gather(payload_cols[c].d_data, esz, d_indices, n, d_out[c]);
+    }
+  }
+  hipError_t e1 = hipFree(d_hashes);
+  hipError_t e2 = hipFree(d_pids);
+  if (rc != BG_OK) return rc;
+  if (e1 != hipSuccess) return set_hip_err(e1, "hipFree");
+  if (e2 != hipSuccess) return set_hip_err(e2, "hipFree");
+  return BG_OK;
+}
+
+// ---------------------------------------------------------------------------
+// q6: fused scan + filter + SUM(l_extendedprice * l_discount)
+// 52 B/row algorithmic (Date32 4 + 3 x Decimal128 16) — HBM-bound.
+// ---------------------------------------------------------------------------
+__global__ void k_q6_agg(const int32_t* shipdate, const ulong2* discount,
+                         const ulong2* quantity, const ulong2* extendedprice,
+                         int64_t n, int32_t date_lo, int32_t date_hi,
+                         i64 disc_lo, i64 disc_hi, i64 qty_lt, u64* out_lo,
+                         u64* out_hi, u64* out_count) {
+  i128 acc = 0;
+  u64 cnt = 0;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    const int32_t d = shipdate[i];
+    const ulong2 dv = discount[i];
+    const ulong2 qv = quantity[i];
+    const ulong2 pv = extendedprice[i];
+    const i128 disc = make_i128(dv.x, (i64)dv.y);
+    const i128 qty = make_i128(qv.x, (i64)qv.y);
+    const bool keep = (d >= date_lo) & (d < date_hi) & (disc >= disc_lo) &
+                      (disc <= disc_hi) & (qty < qty_lt);
+    if (keep) {
+      const i128 price = make_i128(pv.x, (i64)pv.y);
+      acc += price * disc;
+      cnt++;
+    }
+  }
+  // wave -> block -> global
+  acc = wave_reduce_i128(acc);
+#pragma unroll
+  for (int s = 32; s > 0; s >>= 1) cnt += (u64)__shfl_xor((long long)cnt, s, BG_WAVE);
+  __shared__ u64 slo[BG_BLOCK / BG_WAVE], shi[BG_BLOCK / BG_WAVE],
+      scnt[BG_BLOCK / BG_WAVE];
+  const int w = threadIdx.x / BG_WAVE;
+  if (lane_id() == 0) {
+    slo[w] = (u64)(u128)acc;
+    shi[w] = (u64)((u128)acc >> 64);
+    scnt[w] = cnt;
+  }
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    i128 bacc = 0;
+    u64 bcnt = 0;
+    for (int i = 0; i < BG_BLOCK / BG_WAVE; ++i) {
+      bacc += make_i128(slo[i], (i64)shi[i]);
+      bcnt += scnt[i];
+    }
+    atomic_add_i128(out_lo, out_hi, bacc);
+    atomicAdd(out_count, bcnt);
+  }
+}
+
+extern "C" int bg_q6_agg(const bg_column* shipdate, const bg_column* discount,
+                         const bg_column* quantity,
+                         const bg_column* extendedprice, int32_t date_lo,
+                         int32_t date_hi, int64_t disc_lo, int64_t disc_hi,
+                         int64_t qty_lt, uint64_t* out_sum_lo,
+                         int64_t* out_sum_hi, int64_t* out_count) {
+  REQUIRE_INIT();
+  if (shipdate->dtype != BG_DT_DATE32 && shipdate->dtype != BG_DT_INT32)
+    return set_err(BG_ERR_INVALID, "shipdate must be DATE32");
+  const int64_t n = shipdate->len;
+  u64* d_acc;  // [lo, hi, count]
+  HIP_TRY(hipMalloc(&d_acc, 3 * sizeof(u64)));
+  HIP_TRY(hipMemset(d_acc, 0, 3 * sizeof(u64)));
+  int blocks = (int)bg_imin64((n + BG_BLOCK - 1) / BG_BLOCK, BG_MAX_BLOCKS);
+  if (blocks == 0) blocks = 1;
+  hipLaunchKernelGGL(k_q6_agg, dim3(blocks), dim3(BG_BLOCK), 0, 0,
+                     (const int32_t*)shipdate->d_data,
+                     (const ulong2*)discount->d_data,
+                     (const ulong2*)quantity->d_data,
+                     (const ulong2*)extendedprice->d_data, n, date_lo, date_hi,
+                     disc_lo, disc_hi, qty_lt, d_acc, d_acc + 1, d_acc + 2);
+  HIP_TRY(hipGetLastError());
+  u64 h[3];
+  HIP_TRY(hipMemcpy(h, d_acc, 3 * sizeof(u64), hipMemcpyDeviceToHost));
+  HIP_TRY(hipFree(d_acc));
+  *out_sum_lo = h[0];
+  *out_sum_hi = (int64_t)h[1];
+  *out_count = (int64_t)h[2];
+  return BG_OK;
+}
+
+// ---------------------------------------------------------------------------
+// q1: fused filter + grouped partial aggregate.
+// Wave-clustered: per 64 rows, loop over the distinct groups present
+// (ballot leader loop — 4 in TPC-H q1), full-wave register reductions, one
+// LDS atomic set per (group, iteration); block LDS table flushed once.
+// 70 B/row algorithmic (2 x u8 + 4 x Decimal128 + Date32).
+// ---------------------------------------------------------------------------
+#define Q1_GROUPS 256
+#define Q1_ACCS 5  // qty, price, disc_price, charge, disc
+
+__global__ void k_q1_agg(const uint8_t* rf, const uint8_t* ls,
+                         const ulong2* quantity, const ulong2* extendedprice,
+                         const ulong2* discount, const ulong2* tax,
+                         const int32_t* shipdate, int64_t n, int32_t date_le,
+                         u64* g_sums /*Q1_GROUPS*Q1_ACCS*2*/,
+                         u64* g_counts /*Q1_GROUPS*/) {
+  __shared__ u64 s_sums[Q1_GROUPS * Q1_ACCS * 2];
+  __shared__ u64 s_counts[Q1_GROUPS];
+  for (int i = threadIdx.x; i < Q1_GROUPS * Q1_ACCS * 2; i += blockDim.x)
+    s_sums[i] = 0;
+  for (int i = threadIdx.x; i < Q1_GROUPS; i += blockDim.x) s_counts[i] = 0;
+  __syncthreads();
+
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t base = (int64_t)blockIdx.x * blockDim.x + threadIdx.x -
+                      lane_id();
+       base < n; base += stride) {
+    const int64_t i = base + lane_id();
+    const bool in = i < n;
+    const bool keep = in && shipdate[i] <= date_le;
+    i128 vals[Q1_ACCS];
+    uint32_t g = 0xffffffffu;
+    if (keep) {
+      g = ((uint32_t)rf[i] << 4) | (uint32_t)ls[i];
+      const ulong2 qv = quantity[i];
+      const ulong2 pv = extendedprice[i];
+      const ulong2 dv = discount[i];
+      const ulong2 tv = tax[i];
+      const i128 qty = make_i128(qv.x, (i64)qv.y);
+      const i128 price = make_i128(pv.x, (i64)pv.y);
+      const i128 disc = make_i128(dv.x, (i64)dv.y);
+      const i128 tx = make_i128(tv.x, (i64)tv.y);
+      const i128 disc_price = price * (100 - disc);
+      vals[0] = qty;
+      vals[1] = price;
+      vals[2] = disc_price;
+      vals[3] = disc_price * (100 + tx);
+      vals[4] = disc;
+    } else {
+#pragma unroll
+      for (int a = 0; a < Q1_ACCS; ++a) vals[a] = 0;
+    }
+    u64 remaining = __ballot(keep);
+    while (remaining) {
+      const int leader = __ffsll((i64)remaining) - 1;
+      const uint32_t pg = (uint32_t)__shfl((int)g, leader, BG_WAVE);
+      const u64 m = __ballot(keep && g == pg);
+      const bool member = keep && g == pg;
+#pragma unroll
+      for (int a = 0; a < Q1_ACCS; ++a) {
+        const i128 contrib = member ? vals[a] : (i128)0;
+        const i128 tot = wave_reduce_i128(contrib);
+        if (lane_id() == leader)
+          atomic_add_i128_lds(&s_sums[(pg * Q1_ACCS + a) * 2],
+                              &s_sums[(pg * Q1_ACCS + a) * 2 + 1], tot);
+      }
+      if (lane_id() == leader) atomicAdd(&s_counts[pg], (u64)__popcll(m));
+      remaining &= ~m;
+    }
+  }
+  __syncthreads();
+  for (int gidx = threadIdx.x; gidx < Q1_GROUPS; gidx += blockDim.x) {
+    if (s_counts[gidx]) atomicAdd(&g_counts[gidx], s_counts[gidx]);
+    for (int a = 0; a < Q1_ACCS; ++a) {
+      const u64 lo = s_sums[(gidx * Q1_ACCS + a) * 2];
+      const u64 hi = s_sums[(gidx * Q1_ACCS + a) * 2 + 1];
+      if (lo | hi)
+        atomic_add_i128(&g_sums[(gidx * Q1_ACCS + a) * 2],
+                        &g_sums[(gidx * Q1_ACCS + a) * 2 + 1],
+                        make_i128(lo, (i64)hi));
+    }
+  }
+}
+
+extern "C" int bg_q1_agg(const bg_column* rf, const bg_column* ls,
+                         const bg_column* quantity,
+                         const bg_column* extendedprice,
+                         const bg_column* discount, const bg_column* tax,
+                         const bg_column* shipdate, int32_t date_le,
+                         int64_t* h_counts, uint8_t* h_sums) {
+  REQUIRE_INIT();
+  const int64_t n = shipdate->len;
+  u64* d_sums;
+  u64* d_counts;
+  HIP_TRY(hipMalloc(&d_sums, Q1_GROUPS * Q1_ACCS * 2 * sizeof(u64)));
+  HIP_TRY(hipMalloc(&d_counts, Q1_GROUPS * sizeof(u64)));
+  HIP_TRY(hipMemset(d_sums, 0, Q1_GROUPS * Q1_ACCS * 2 * sizeof(u64)));
+  HIP_TRY(hipMemset(d_counts, 0, Q1_GROUPS * sizeof(u64)));
+  int blocks = (int)bg_imin64((n + BG_BLOCK - 1) / BG_BLOCK, BG_MAX_BLOCKS);
+  if (blocks == 0) blocks = 1;
+  hipLaunchKernelGGL(k_q1_agg, dim3(blocks), dim3(BG_BLOCK), 0, 0,
+                     (const uint8_t*)rf->d_data, (const uint8_t*)ls->d_data,
+                     (const ulong2*)quantity->d_data,
+                     (const ulong2*)extendedprice->d_data,
+                     (const ulong2*)discount->d_data,
+                     (const ulong2*)tax->d_data,
+                     (const int32_t*)shipdate->d_data, n, date_le, d_sums,
+                     d_counts);
+  HIP_TRY(hipGetLastError());
+  u64 sums[Q1_GROUPS * Q1_ACCS * 2];
+  u64 counts[Q1_GROUPS];
+  HIP_TRY(hipMemcpy(sums, d_sums, sizeof(sums), hipMemcpyDeviceToHost));
+  HIP_TRY(hipMemcpy(counts, d_counts, sizeof(counts), hipMemcpyDeviceToHost));
+  HIP_TRY(hipFree(d_sums));
+  HIP_TRY(hipFree(d_counts));
+  for (int gidx = 0; gidx < Q1_GROUPS; ++gidx) {
+    h_counts[gidx] = (int64_t)counts[gidx];
+    for (int a = 0; a < Q1_ACCS; ++a) {
+      memcpy(h_sums + (gidx * Q1_ACCS + a) * 16, &sums[(gidx * Q1_ACCS + a) * 2], 8);
+      memcpy(h_sums + (gidx * Q1_ACCS + a) * 16 + 8,
+             &sums[(gidx * Q1_ACCS + a) * 2 + 1], 8);
+    }
+  }
+  return BG_OK;
+}

@@ -1,0 +1,155 @@
+"""GPU query-stage executor — host mirror of Ballista's QueryStageExecutor
+for the hash-repartition (sort-shuffle) stage.
+
+Mirrors (reference citations, /root/reference):
+  - ExecutionEngine::create_query_stage_exec /
+    QueryStageExecutor::execute_query_stage —
+    ballista/executor/src/execution_engine.rs:53-103 (the plugin seam the
+    Rust binding in INTEGRATION.md drops this engine into).
+  - SortShuffleWriterExec::execute_shuffle_write (writer.rs:564-753):
+    per-input bucket rows by hash partition -> materialise per-partition
+    batches -> encode IPC+LZ4 -> consolidate partition-major, one file per
+    task + index; summaries returned as ShuffleWritePartition
+    (proto ballista.proto:779-791: partition_id, path, num_batches,
+    num_rows, num_bytes).
+
+Device work (hash, partition split, gather) runs through libballista_gpu.so;
+IPC encode + file write stay host-side.
+"""
+
+from dataclasses import dataclass
+
+import numpy as np
+import pyarrow as pa
+
+from . import gpu
+from . import shuffle
+
+
+@dataclass
+class ShuffleWritePartition:
+    """ShuffleWritePartition proto restatement (ballista.proto:779-791)."""
+    partition_id: int
+    path: str
+    num_batches: int
+    num_rows: int
+    num_bytes: int
+
+
+_PA_TO_BG = {
+    pa.int32(): gpu.BG_DT_INT32,
+    pa.int64(): gpu.BG_DT_INT64,
+    pa.date32(): gpu.BG_DT_DATE32,
+    pa.uint8(): gpu.BG_DT_DICT8,
+}
+
+
+def _bg_dtype(t: pa.DataType) -> int:
+    if pa.types.is_decimal128(t):
+        return gpu.BG_DT_DECIMAL128
+    try:
+        return _PA_TO_BG[t]
+    except KeyError:
+        raise RuntimeError(f"dtype {t} not yet accelerated (BG_ERR_UNSUPPORTED)")
+
+
+def _np_for(t: pa.DataType):
+    if pa.types.is_decimal128(t):
+        return np.uint8  # 16 B/elem raw
+    return {pa.int32(): np.int32, pa.int64(): np.int64,
+            pa.date32(): np.int32, pa.uint8(): np.uint8}[t]
+
+
+def _col_raw(arr: pa.Array) -> np.ndarray:
+    """Arrow array -> contiguous primitive numpy view (no nulls yet)."""
+    if arr.null_count:
+        raise RuntimeError("null-carrying payload columns: next round")
+    t = arr.type
+    if pa.types.is_decimal128(t):
+        buf = arr.buffers()[1]
+        off = arr.offset * 16
+        return np.frombuffer(buf, dtype=np.uint8,
+                             count=len(arr) * 16, offset=off)
+    np_t = _np_for(t)
+    buf = arr.buffers()[1]
+    return np.frombuffer(buf, dtype=np_t, count=len(arr), offset=arr.offset *
+                         np.dtype(np_t).itemsize)
+
+
+def _array_from_raw(t: pa.DataType, raw: np.ndarray, n: int) -> pa.Array:
+    if pa.types.is_decimal128(t):
+        return pa.Array.from_buffers(t, n, [None, pa.py_buffer(raw.tobytes())])
+    return pa.Array.from_buffers(t, n, [None, pa.py_buffer(raw.tobytes())])
+
+
+class GpuQueryStageExecutor:
+    """Executes one sort-shuffle task: hash-repartition `batch`'s rows into
+    k partitions on the GPU, write the consolidated shuffle file, return
+    ShuffleWritePartition summaries — the GPU implementation of
+    execute_query_stage for a SortShuffleWriterExec-rooted stage."""
+
+    def __init__(self, ctx: "gpu.GpuStageContext", job_id: str, stage_id: int,
+                 work_dir: str, key_columns, num_partitions: int,
+                 batch_size: int = shuffle.DEFAULT_BATCH_SIZE):
+        self.ctx = ctx
+        self.job_id = job_id
+        self.stage_id = stage_id
+        self.work_dir = work_dir
+        self.key_columns = key_columns  # column indices (already-evaluated keys)
+        self.k = num_partitions
+        self.batch_size = batch_size
+
+    def execute_query_stage(self, task_id: int, table: pa.Table):
+        ctx = self.ctx
+        table = table.combine_chunks()
+        n = table.num_rows
+        schema = table.schema
+
+        cols = []
+        for i in range(table.num_columns):
+            arr = table.column(i).combine_chunks()
+            if isinstance(arr, pa.ChunkedArray):
+                arr = arr.chunk(0) if arr.num_chunks else pa.array(
+                    [], type=arr.type)
+            raw = _col_raw(arr)
+            bgdt = _bg_dtype(arr.type)
+            buf = ctx.upload(raw)
+            nelem = n
+            cols.append(ctx.column(bgdt, buf, nelem))
+
+        key_cols = [cols[i] for i in self.key_columns]
+        idx_buf, offs_buf, out_bufs = ctx.hash_repartition(
+            key_cols, cols, n, self.k)
+        ctx.synchronize()
+        offsets = offs_buf.download(np.int64, self.k + 1)
+
+        # download partition-major buffers and slice per partition
+        col_raws = []
+        for c, b in zip(cols, out_bufs):
+            esz = gpu._DT_SIZE[c.dtype]
+            col_raws.append(b.download(np.uint8, max(n, 1) * esz))
+
+        partition_streams = []
+        for p in range(self.k):
+            lo, hi = int(offsets[p]), int(offsets[p + 1])
+            m = hi - lo
+            if m == 0:
+                partition_streams.append([b""])
+                continue
+            arrays = []
+            for ci in range(table.num_columns):
+                t = schema.types[ci]
+                esz = 16 if pa.types.is_decimal128(t) else \
+                    np.dtype(_np_for(t)).itemsize
+                raw = col_raws[ci][lo * esz: hi * esz]
+                arrays.append(_array_from_raw(t, raw, m))
+            part_table = pa.Table.from_arrays(arrays, schema=schema)
+            batches = shuffle.rechunk(part_table, self.batch_size)
+            partition_streams.append(
+                [shuffle.encode_partition_stream(batches, schema)])
+
+        data_path, index_path, stats = shuffle.write_task_consolidated(
+            self.work_dir, self.job_id, self.stage_id, task_id, schema,
+            partition_streams)
+        return [ShuffleWritePartition(p, data_path, nb, nr, nbytes)
+                for (p, nb, nr, nbytes) in stats]

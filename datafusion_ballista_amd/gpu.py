@@ -1,0 +1,274 @@
+"""ctypes host bindings for libballista_gpu.so (include/ballista_gpu.h).
+
+The GpuStageContext owns device memory for one task's columns and exposes
+the hot-path ops.  It fails loudly (RuntimeError) when the HIP extension or
+the GPU is missing — no CPU fallback exists on the product path.
+"""
+
+import ctypes
+import os
+
+import numpy as np
+
+_DIR = os.path.dirname(os.path.abspath(__file__))
+
+BG_DT_INT32 = 1
+BG_DT_INT64 = 2
+BG_DT_DATE32 = 3
+BG_DT_DECIMAL128 = 4
+BG_DT_DICT8 = 5
+
+BG_PRED_GE_LT = 0
+BG_PRED_BETWEEN = 1
+BG_PRED_LT = 2
+BG_PRED_EQ = 3
+BG_PRED_GT = 4
+
+_DT_SIZE = {BG_DT_INT32: 4, BG_DT_DATE32: 4, BG_DT_INT64: 8,
+            BG_DT_DECIMAL128: 16, BG_DT_DICT8: 1}
+
+_NP_TO_DT = {
+    np.dtype(np.int32): BG_DT_INT32,
+    np.dtype(np.int64): BG_DT_INT64,
+    np.dtype(np.uint8): BG_DT_DICT8,
+}
+
+
+class BgColumn(ctypes.Structure):
+    _fields_ = [
+        ("dtype", ctypes.c_int32),
+        ("precision", ctypes.c_int32),
+        ("scale", ctypes.c_int32),
+        ("_pad", ctypes.c_int32),
+        ("d_data", ctypes.c_void_p),
+        ("d_validity", ctypes.c_void_p),
+        ("len", ctypes.c_int64),
+    ]
+
+
+class BgPred(ctypes.Structure):
+    _fields_ = [
+        ("column", ctypes.c_int32),
+        ("op", ctypes.c_int32),
+        ("lo_lo", ctypes.c_int64),
+        ("lo_hi", ctypes.c_int64),
+        ("hi_lo", ctypes.c_int64),
+        ("hi_hi", ctypes.c_int64),
+    ]
+
+
+def lib_path() -> str:
+    return os.path.join(_DIR, "libballista_gpu.so")
+
+
+_lib = None
+
+
+def load_library() -> ctypes.CDLL:
+    """dlopen the HIP extension; raises if it is absent or unloadable."""
+    global _lib
+    if _lib is None:
+        p = lib_path()
+        if not os.path.exists(p):
+            raise RuntimeError(
+                f"libballista_gpu.so not found at {p}: the MI355X stage "
+                "executor requires its HIP extension (run "
+                "__graft_entry__.build()); there is no CPU fallback")
+        _lib = ctypes.CDLL(p)
+        _lib.bg_last_error.restype = ctypes.c_char_p
+    return _lib
+
+
+def _check(rc: int, what: str):
+    if rc != 0:
+        err = load_library().bg_last_error().decode()
+        raise RuntimeError(f"{what} failed (rc={rc}): {err}")
+
+
+def _split_i128(v: int):
+    v = int(v)
+    return (ctypes.c_int64(v & 0xFFFFFFFFFFFFFFFF).value,
+            ctypes.c_int64((v >> 64) & 0xFFFFFFFFFFFFFFFF).value)
+
+
+class DeviceBuffer:
+    """Caller-owned device allocation."""
+
+    def __init__(self, ctx, nbytes: int):
+        self._ctx = ctx
+        self.nbytes = nbytes
+        ptr = ctypes.c_void_p()
+        _check(ctx.L.bg_malloc(ctypes.c_uint64(nbytes), ctypes.byref(ptr)),
+               "bg_malloc")
+        self.ptr = ptr
+
+    def free(self):
+        if self.ptr and self.ptr.value:
+            self._ctx.L.bg_free(self.ptr)
+            self.ptr = ctypes.c_void_p()
+
+    def upload(self, arr: np.ndarray):
+        a = np.ascontiguousarray(arr)
+        assert a.nbytes <= self.nbytes
+        _check(self._ctx.L.bg_memcpy_h2d(
+            self.ptr, a.ctypes.data_as(ctypes.c_void_p),
+            ctypes.c_uint64(a.nbytes)), "bg_memcpy_h2d")
+        return self
+
+    def download(self, dtype, count) -> np.ndarray:
+        out = np.empty(count, dtype=dtype)
+        _check(self._ctx.L.bg_memcpy_d2h(
+            out.ctypes.data_as(ctypes.c_void_p), self.ptr,
+            ctypes.c_uint64(out.nbytes)), "bg_memcpy_d2h")
+        return out
+
+
+class GpuStageContext:
+    """One GPU-backed task context (one process per GPU; the device ordinal
+    is normally fixed by HIP_VISIBLE_DEVICES, SURVEY.md §2 executor row)."""
+
+    def __init__(self, device: int = 0):
+        self.L = load_library()
+        _check(self.L.bg_init(device), "bg_init")
+        self._bufs = []
+
+    # ---- memory ----
+    def alloc(self, nbytes: int) -> DeviceBuffer:
+        b = DeviceBuffer(self, nbytes)
+        self._bufs.append(b)
+        return b
+
+    def upload(self, arr: np.ndarray) -> DeviceBuffer:
+        return self.alloc(arr.nbytes).upload(arr)
+
+    def close(self):
+        for b in self._bufs:
+            b.free()
+        self._bufs = []
+
+    def synchronize(self):
+        _check(self.L.bg_synchronize(), "bg_synchronize")
+
+    # ---- columns ----
+    def column(self, dtype: int, buf: DeviceBuffer, n: int,
+               validity: DeviceBuffer = None, precision=0, scale=0) -> BgColumn:
+        return BgColumn(dtype, precision, scale, 0, buf.ptr,
+                        validity.ptr if validity else None, n)
+
+    def upload_column(self, arr: np.ndarray, dtype: int = None,
+                      validity: np.ndarray = None):
+        if dtype is None:
+            dtype = _NP_TO_DT[arr.dtype]
+        buf = self.upload(arr)
+        vbuf = self.upload(validity) if validity is not None else None
+        n = len(arr) if dtype != BG_DT_DECIMAL128 else arr.nbytes // 16
+        return self.column(dtype, buf, n, vbuf), buf
+
+    # ---- ops ----
+    def eval_predicates(self, cols, preds, n: int) -> DeviceBuffer:
+        """cols: list[BgColumn]; preds: list[(col, op, lo, hi)] with python
+        int bounds. Returns the device mask (ceil(n/64)*8 bytes)."""
+        nwords = (n + 63) // 64
+        mask = self.alloc(max(nwords * 8, 8))
+        carr = (BgColumn * len(cols))(*cols)
+        parr = (BgPred * len(preds))()
+        for i, (c, op, lo, hi) in enumerate(preds):
+            lo_lo, lo_hi = _split_i128(lo)
+            hi_lo, hi_hi = _split_i128(hi)
+            parr[i] = BgPred(c, op, lo_lo, lo_hi, hi_lo, hi_hi)
+        _check(self.L.bg_eval_predicates(carr, len(cols), parr, len(preds),
+                                         ctypes.c_int64(n), mask.ptr),
+               "bg_eval_predicates")
+        return mask
+
+    def mask_to_indices(self, mask: DeviceBuffer, n: int):
+        idx = self.alloc(max(4 * n, 4))
+        count = ctypes.c_int64()
+        _check(self.L.bg_mask_to_indices(mask.ptr, ctypes.c_int64(n), idx.ptr,
+                                         ctypes.byref(count)),
+               "bg_mask_to_indices")
+        return idx, count.value
+
+    def gather(self, src: DeviceBuffer, elem_size: int, idx: DeviceBuffer,
+               m: int) -> DeviceBuffer:
+        dst = self.alloc(max(elem_size * m, elem_size))
+        _check(self.L.bg_gather(src.ptr, ctypes.c_int64(elem_size), idx.ptr,
+                                ctypes.c_int64(m), dst.ptr), "bg_gather")
+        return dst
+
+    def hash_columns(self, key_cols, n: int) -> DeviceBuffer:
+        hashes = self.alloc(max(8 * n, 8))
+        karr = (BgColumn * len(key_cols))(*key_cols)
+        _check(self.L.bg_hash_columns(karr, len(key_cols), ctypes.c_int64(n),
+                                      hashes.ptr), "bg_hash_columns")
+        return hashes
+
+    def partition_ids(self, hashes: DeviceBuffer, n: int, k: int) -> DeviceBuffer:
+        pids = self.alloc(max(4 * n, 4))
+        _check(self.L.bg_partition_ids(hashes.ptr, ctypes.c_int64(n), k,
+                                       pids.ptr), "bg_partition_ids")
+        return pids
+
+    def partition_indices(self, pids: DeviceBuffer, n: int, k: int):
+        idx = self.alloc(max(4 * n, 4))
+        offs = self.alloc(8 * (k + 1))
+        _check(self.L.bg_partition_indices(pids.ptr, ctypes.c_int64(n), k,
+                                           idx.ptr, offs.ptr),
+               "bg_partition_indices")
+        return idx, offs
+
+    def hash_repartition(self, key_cols, payload_cols, n: int, k: int):
+        """-> (indices buf, offsets buf, [out bufs partition-major])."""
+        idx = self.alloc(max(4 * n, 4))
+        offs = self.alloc(8 * (k + 1))
+        outs = []
+        optrs = (ctypes.c_void_p * len(payload_cols))()
+        for i, c in enumerate(payload_cols):
+            esz = _DT_SIZE[c.dtype]
+            b = self.alloc(max(esz * n, esz))
+            outs.append(b)
+            optrs[i] = b.ptr.value
+        karr = (BgColumn * len(key_cols))(*key_cols)
+        parr = (BgColumn * len(payload_cols))(*payload_cols)
+        _check(self.L.bg_hash_repartition(karr, len(key_cols), parr,
+                                          len(payload_cols),
+                                          ctypes.c_int64(n), k, idx.ptr,
+                                          offs.ptr, optrs),
+               "bg_hash_repartition")
+        return idx, offs, outs
+
+    def q6_agg(self, shipdate: BgColumn, discount: BgColumn,
+               quantity: BgColumn, price: BgColumn, date_lo: int, date_hi: int,
+               disc_lo: int, disc_hi: int, qty_lt: int):
+        """-> (count, exact i128 sum as python int)."""
+        s_lo = ctypes.c_uint64()
+        s_hi = ctypes.c_int64()
+        cnt = ctypes.c_int64()
+        _check(self.L.bg_q6_agg(ctypes.byref(shipdate), ctypes.byref(discount),
+                                ctypes.byref(quantity), ctypes.byref(price),
+                                date_lo, date_hi,
+                                ctypes.c_int64(disc_lo), ctypes.c_int64(disc_hi),
+                                ctypes.c_int64(qty_lt), ctypes.byref(s_lo),
+                                ctypes.byref(s_hi), ctypes.byref(cnt)),
+               "bg_q6_agg")
+        return cnt.value, (s_hi.value << 64) + s_lo.value
+
+    def q1_agg(self, rf, ls, qty, price, disc, tax, shipdate, date_le: int):
+        """-> dict group -> (count, [5 exact i128 sums])."""
+        counts = np.zeros(256, dtype=np.int64)
+        sums = np.zeros(256 * 5 * 16, dtype=np.uint8)
+        _check(self.L.bg_q1_agg(
+            ctypes.byref(rf), ctypes.byref(ls), ctypes.byref(qty),
+            ctypes.byref(price), ctypes.byref(disc), ctypes.byref(tax),
+            ctypes.byref(shipdate), date_le,
+            counts.ctypes.data_as(ctypes.POINTER(ctypes.c_int64)),
+            sums.ctypes.data_as(ctypes.POINTER(ctypes.c_uint8))), "bg_q1_agg")
+        out = {}
+        raw = sums.reshape(256, 5, 16)
+        for g in range(256):
+            if counts[g] == 0:
+                continue
+            vals = [int.from_bytes(bytes(raw[g, a]), "little", signed=True)
+                    for a in range(5)]
+            out[g] = (int(counts[g]), vals)
+        return out

@@ -1,0 +1,154 @@
+"""Sort-shuffle file format, byte-compatible with the reference.
+
+Restates (reference citations, /root/reference):
+  - consolidated data file layout — write_task_consolidated,
+    ballista/core/src/execution_plans/sort_shuffle/writer.rs:794-895:
+    `{work_dir}/{job_id}/{stage_id}/{file_id}/data.arrow` =
+    leading schema-only Arrow IPC stream (schema message + EOS, no batches),
+    then K partition-major byte ranges, each a concatenation of zero or more
+    COMPLETE Arrow IPC streams (batches LZ4_FRAME-compressed; codec default
+    `lz4`, core/src/config.rs:413-415).
+  - index file — sort_shuffle/index.rs:21-33:
+    `data.arrow.index` = (K+1) little-endian i64 absolute offsets; offset_i =
+    absolute byte where partition i starts; last entry = total file length.
+  - readers cross sub-stream boundaries transparently —
+    sort_shuffle/multi_stream_reader.rs:17-34; a reader fetches partition p
+    with one ranged read [offset_p, offset_{p+1}) (shuffle_reader.rs:1120-1168).
+  - batches re-chunked to `batch_size` (default 8192,
+    sort_shuffle/config.rs:35-43).
+
+File IO + IPC framing stay host-side by design ("file
+write stays CPU/IO"); the GPU produces the partition-major column buffers.
+"""
+
+import io
+import os
+import struct
+
+import pyarrow as pa
+
+DEFAULT_BATCH_SIZE = 8192  # sort_shuffle/config.rs:35-43
+
+
+def ipc_write_options():
+    # LZ4_FRAME body-buffer compression = Ballista's default shuffle codec
+    # (config.rs:413-415, codec map :729-742).
+    return pa.ipc.IpcWriteOptions(compression="lz4")
+
+
+def encode_partition_stream(batches, schema) -> bytes:
+    """One complete IPC stream (schema .. batches .. EOS) for one partition's
+    in-memory rows — encode_buffered_partitions (writer.rs:121-164). Empty
+    partitions encode to b'' (the reference skips the writer entirely)."""
+    if not batches or all(b.num_rows == 0 for b in batches):
+        return b""
+    sink = io.BytesIO()
+    with pa.ipc.new_stream(sink, schema, options=ipc_write_options()) as w:
+        for b in batches:
+            if b.num_rows:
+                w.write_batch(b)
+    return sink.getvalue()
+
+
+def rechunk(table_like, batch_size=DEFAULT_BATCH_SIZE):
+    """Slice a pyarrow Table/RecordBatch into <=batch_size record batches
+    (PartitionedBatchIterator materialises interleaved rows in batch_size
+    chunks — partitioned_batch_iterator.rs)."""
+    if isinstance(table_like, pa.RecordBatch):
+        table_like = pa.Table.from_batches([table_like])
+    out = []
+    n = table_like.num_rows
+    for s in range(0, n, batch_size):
+        chunk = table_like.slice(s, min(batch_size, n - s))
+        out.extend(chunk.combine_chunks().to_batches())
+    return out
+
+
+def task_output_dir(work_dir, job_id, stage_id, file_id) -> str:
+    # write_task_consolidated path layout (writer.rs:821-828)
+    return os.path.join(work_dir, str(job_id), str(stage_id), str(file_id))
+
+
+def write_task_consolidated(work_dir, job_id, stage_id, file_id, schema,
+                            partition_streams):
+    """partition_streams: list (len K) of lists of encoded IPC stream bytes
+    (one per input partition, concatenated verbatim — writer.rs:861-884).
+
+    Returns (data_path, index_path, [(partition_id, num_batches, num_rows,
+    num_bytes)]) with stats counted like EncodedStats (writer.rs:778-844);
+    num_rows/num_batches must be supplied via the streams' metadata, so this
+    function re-parses each stream cheaply for counts (host-side, test/IO
+    path only)."""
+    out_dir = task_output_dir(work_dir, job_id, stage_id, file_id)
+    os.makedirs(out_dir, exist_ok=True)
+    data_path = os.path.join(out_dir, "data.arrow")
+    index_path = os.path.join(out_dir, "data.arrow.index")
+
+    k = len(partition_streams)
+    offsets = [0] * (k + 1)
+    stats = []
+    with open(data_path, "wb") as f:
+        # leading schema-only stream so a reader of an empty partition can
+        # still recover the schema (writer.rs:838-846)
+        sink = io.BytesIO()
+        with pa.ipc.new_stream(sink, schema, options=ipc_write_options()):
+            pass
+        f.write(sink.getvalue())
+        for p in range(k):
+            offsets[p] = f.tell()
+            num_batches = num_rows = num_bytes = 0
+            for stream in partition_streams[p]:
+                if not stream:
+                    continue
+                f.write(stream)
+                rd = pa.ipc.open_stream(pa.BufferReader(stream))
+                for b in rd:
+                    num_batches += 1
+                    num_rows += b.num_rows
+                    num_bytes += b.get_total_buffer_size()
+            stats.append((p, num_batches, num_rows, num_bytes))
+        offsets[k] = f.tell()
+
+    write_index(index_path, offsets)
+    return data_path, index_path, stats
+
+
+def write_index(index_path, offsets):
+    # (K+1) little-endian i64 (index.rs:21-33)
+    with open(index_path, "wb") as f:
+        f.write(struct.pack(f"<{len(offsets)}q", *offsets))
+
+
+def read_index(index_path):
+    raw = open(index_path, "rb").read()
+    n = len(raw) // 8
+    return list(struct.unpack(f"<{n}q", raw))
+
+
+def read_partition(data_path, index_path, partition_id):
+    """Ranged read of one partition; crosses concatenated IPC sub-stream
+    boundaries transparently (multi_stream_reader.rs:17-34).
+    Returns a list of RecordBatches (possibly empty)."""
+    offsets = read_index(index_path)
+    lo = offsets[partition_id]
+    hi = offsets[partition_id + 1]
+    with open(data_path, "rb") as f:
+        f.seek(lo)
+        raw = f.read(hi - lo)
+    batches = []
+    buf = pa.BufferReader(raw)
+    while buf.tell() < len(raw):
+        sub = pa.BufferReader(raw[buf.tell():])
+        rd = pa.ipc.open_stream(sub)
+        for b in rd:
+            batches.append(b)
+        buf.seek(buf.tell() + sub.tell())
+    return batches
+
+
+def read_schema(data_path):
+    """Recover the schema from the leading schema-only stream."""
+    with open(data_path, "rb") as f:
+        raw = f.read(1 << 16)
+    rd = pa.ipc.open_stream(pa.BufferReader(raw))
+    return rd.schema

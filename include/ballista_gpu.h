@@ -1,0 +1,185 @@
+/* ballista_gpu.h — C ABI of the MI355X-native Ballista stage-executor
+ * library (libballista_gpu.so).
+ *
+ * This is the device boundary of SURVEY.md §8(b), seam 3: the thin
+ * `extern "C"` FFI a Rust `GpuExecutionEngine` (implementing Ballista's
+ * `ExecutionEngine` trait, ballista/executor/src/execution_engine.rs:53-103,
+ * injected via `Executor::new` arg 8 / `ExecutorProcessConfig.
+ * override_execution_engine`, executor_process.rs:319) binds to run the
+ * executor-side physical-operator hot path on a gfx950 GPU.  The Rust-side
+ * binding stub a maintainer would add is shown in INTEGRATION.md.
+ *
+ * Conventions:
+ *  - All functions return BG_OK (0) or a negative bg_status code;
+ *    bg_last_error() returns a thread-local message for the last failure.
+ *  - Column buffers are Arrow C-Data-Interface-style: raw data pointer +
+ *    optional validity bitmap (LSB bit order) + length.  Pointers named d_*
+ *    are DEVICE pointers owned by the caller (allocated via bg_malloc);
+ *    the library never frees caller memory.
+ *  - There is no CPU fallback anywhere behind this ABI: every compute entry
+ *    point requires an initialised HIP device and fails loudly otherwise.
+ *
+ * Reference interfaces each entry point replaces (file:line under
+ * /root/reference):
+ *  - bg_hash_columns / bg_partition_ids / bg_partition_indices:
+ *      compute_partition_indices,
+ *      ballista/core/src/execution_plans/sort_shuffle/writer.rs:1259-1279
+ *      (evaluate key exprs -> create_hashes(REPARTITION_RANDOM_STATE) ->
+ *       h % K -> per-partition row-index lists).
+ *  - bg_eval_predicates / bg_mask_to_indices / bg_gather:
+ *      DataFusion 55 FilterExec predicate eval + filter compaction and the
+ *      sort-shuffle writer's interleave-gather
+ *      (sort_shuffle/partitioned_batch_iterator.rs; SURVEY.md §8a rows 1,5).
+ *  - bg_hash_repartition: the device half of
+ *      SortShuffleWriterExec::execute_shuffle_write (writer.rs:564-753) —
+ *      bucket rows by partition and materialise partition-major column
+ *      buffers; IPC encode + file write stay host-side
+ *      (write_task_consolidated writer.rs:810-895 + index.rs:21-33).
+ *  - bg_q6_agg: the fused Filter+Partial-Aggregate stage of TPC-H q6
+ *      (plan golden scheduler/tests/tpch_plan_stability/approved/q6.txt;
+ *       SUM(Decimal128) exact i128 per DataFusion AggregateExec).
+ *  - bg_q1_agg: the fused Filter+Partial-Aggregate stage of TPC-H q1
+ *      (approved/q1.txt; grouped sums/counts, groups dictionary-encoded).
+ */
+
+#ifndef BALLISTA_GPU_H
+#define BALLISTA_GPU_H
+
+#include <stdint.h>
+
+#ifdef __cplusplus
+extern "C" {
+#endif
+
+/* ---- status ---- */
+typedef enum {
+  BG_OK = 0,
+  BG_ERR_HIP = -1,        /* HIP runtime failure (see bg_last_error) */
+  BG_ERR_NO_GPU = -2,     /* no usable gfx950 device / not initialised */
+  BG_ERR_INVALID = -3,    /* bad argument */
+  BG_ERR_UNSUPPORTED = -4 /* dtype/op not yet accelerated */
+} bg_status;
+
+const char* bg_last_error(void);
+int bg_version(void);
+
+/* ---- device/session ---- */
+int bg_init(int device_ordinal); /* fails loudly when no GPU is present */
+int bg_device_count(int* out);
+int bg_synchronize(void);
+
+/* ---- device memory (caller-owned) ---- */
+int bg_malloc(uint64_t bytes, void** d_ptr);
+int bg_free(void* d_ptr);
+int bg_memset(void* d_ptr, int value, uint64_t bytes);
+int bg_memcpy_h2d(void* d_dst, const void* h_src, uint64_t bytes);
+int bg_memcpy_d2h(void* h_dst, const void* d_src, uint64_t bytes);
+
+/* ---- columns ---- */
+typedef enum {
+  BG_DT_INT32 = 1,
+  BG_DT_INT64 = 2,
+  BG_DT_DATE32 = 3,      /* Arrow Date32: i32 days since epoch */
+  BG_DT_DECIMAL128 = 4,  /* Arrow Decimal128: 16-byte LE two's complement */
+  BG_DT_DICT8 = 5,       /* dictionary codes as u8 (small Utf8 dictionaries) */
+} bg_dtype;
+
+typedef struct {
+  int32_t dtype;          /* bg_dtype */
+  int32_t precision;      /* decimals only */
+  int32_t scale;          /* decimals only */
+  int32_t _pad;
+  const void* d_data;     /* device pointer, element-contiguous */
+  const uint8_t* d_validity; /* device pointer or NULL (all valid); LSB bits */
+  int64_t len;
+} bg_column;
+
+/* ---- predicates (FilterExec subset; ops shared with the oracle) ---- */
+typedef enum {
+  BG_PRED_GE_LT = 0,   /* lo <= x <  hi */
+  BG_PRED_BETWEEN = 1, /* lo <= x <= hi */
+  BG_PRED_LT = 2,      /* x <  hi */
+  BG_PRED_EQ = 3,      /* x == lo */
+  BG_PRED_GT = 4,      /* x >  lo */
+} bg_pred_op;
+
+typedef struct {
+  int32_t column; /* index into the cols array */
+  int32_t op;     /* bg_pred_op */
+  /* bounds as i128 split into (lo64, hi64); narrower dtypes use the low part */
+  int64_t lo_lo; int64_t lo_hi;
+  int64_t hi_lo; int64_t hi_hi;
+} bg_pred;
+
+/* AND-fold `npreds` predicates over `cols` into an Arrow LSB bitmask
+ * d_mask (ceil(n/64)*8 bytes, 8-byte aligned). Null input => false. */
+int bg_eval_predicates(const bg_column* cols, int32_t ncols,
+                       const bg_pred* preds, int32_t npreds,
+                       int64_t n, uint8_t* d_mask);
+
+/* Stable compaction: selection bitmask -> ascending row indices.
+ * d_indices capacity n; *out_count receives the selected count. */
+int bg_mask_to_indices(const uint8_t* d_mask, int64_t n, uint32_t* d_indices,
+                       int64_t* out_count);
+
+/* Row gather (take): d_dst[i] = d_src[d_idx[i]], elem_size in {1,2,4,8,16}. */
+int bg_gather(const void* d_src, int64_t elem_size, const uint32_t* d_idx,
+              int64_t m, void* d_dst);
+
+/* ---- hash repartition (SortShuffleWriterExec device half) ---- */
+
+/* create_hashes restatement over the key columns (bg_ahash.h; parity
+ * unpinned for the ahash seed constants — SURVEY.md §8c). */
+int bg_hash_columns(const bg_column* key_cols, int32_t nkeys, int64_t n,
+                    uint64_t* d_hashes);
+
+/* pid[i] = hash[i] % k  (writer.rs:1274-1276) */
+int bg_partition_ids(const uint64_t* d_hashes, int64_t n, uint32_t k,
+                     uint32_t* d_pids);
+
+/* Stable multi-split: partition-major row-index lists, ascending row order
+ * inside each partition — the device equivalent of the Vec<Vec<u32>> that
+ * compute_partition_indices returns.  d_indices: n entries; d_offsets: k+1
+ * exclusive prefix (offsets[k] == n).  k <= 4096. */
+int bg_partition_indices(const uint32_t* d_pids, int64_t n, uint32_t k,
+                         uint32_t* d_indices, int64_t* d_offsets);
+
+/* Fused convenience: hash + pids + stable split + gather every payload
+ * column partition-major.  d_out[c] must hold len*elem_size(c) bytes; rows
+ * of partition p for column c live at [offsets[p]*esz, offsets[p+1]*esz).
+ * offsets (k+1, device) and the permutation d_indices are also returned so
+ * the host can slice buffers for IPC encoding. */
+int bg_hash_repartition(const bg_column* key_cols, int32_t nkeys,
+                        const bg_column* payload_cols, int32_t ncols,
+                        int64_t n, uint32_t k,
+                        uint32_t* d_indices, int64_t* d_offsets,
+                        void** d_out /* ncols device pointers */);
+
+/* ---- fused filter+aggregate stages ---- */
+
+/* TPC-H q6 stage 1 (scan+filter+aggregate, approved/q6.txt):
+ *   sum(l_extendedprice * l_discount), count of qualifying rows.
+ * shipdate: DATE32; discount/quantity/extendedprice: DECIMAL128.
+ * Result is the exact i128 sum (scale 4). */
+int bg_q6_agg(const bg_column* shipdate, const bg_column* discount,
+              const bg_column* quantity, const bg_column* extendedprice,
+              int32_t date_lo, int32_t date_hi, int64_t disc_lo,
+              int64_t disc_hi, int64_t qty_lt,
+              uint64_t* out_sum_lo, int64_t* out_sum_hi, int64_t* out_count);
+
+/* TPC-H q1 stage 1 (filter + grouped partial aggregate, approved/q1.txt):
+ * group = (rf_code<<4)|ls_code over u8 dictionary codes (<16 each);
+ * per group: count + exact i128 sums of {qty, price, price*(100-disc),
+ * price*(100-disc)*(100+tax), disc} — scales {2,2,4,6,2}.
+ * h_counts: host i64[256]; h_sums: host bytes 256*5*16 (LE i128 each). */
+int bg_q1_agg(const bg_column* rf, const bg_column* ls,
+              const bg_column* quantity, const bg_column* extendedprice,
+              const bg_column* discount, const bg_column* tax,
+              const bg_column* shipdate, int32_t date_le,
+              int64_t* h_counts, uint8_t* h_sums);
+
+#ifdef __cplusplus
+} /* extern "C" */
+#endif
+
+#endif /* BALLISTA_GPU_H */

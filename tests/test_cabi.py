@@ -1,0 +1,60 @@
+"""C-ABI surface checks (CPU-safe: loads the library, no compute calls)."""
+import ctypes
+import os
+import re
+
+import pytest
+
+HERE = os.path.dirname(os.path.abspath(__file__))
+ROOT = os.path.dirname(HERE)
+HEADER = os.path.join(ROOT, "include", "ballista_gpu.h")
+
+
+def header_symbols():
+    """Every function declared in include/ballista_gpu.h."""
+    src = open(HEADER).read()
+    # strip comments
+    src = re.sub(r"/\*.*?\*/", "", src, flags=re.S)
+    syms = re.findall(r"^\s*(?:const\s+char\s*\*|int)\s+(bg_\w+)\s*\(", src,
+                      flags=re.M)
+    assert len(syms) >= 15
+    return syms
+
+
+def test_library_exports_every_header_symbol():
+    from datafusion_ballista_amd import gpu
+    L = gpu.load_library()
+    for sym in header_symbols():
+        assert hasattr(L, sym), f"libballista_gpu.so missing symbol {sym}"
+
+
+def test_version():
+    from datafusion_ballista_amd import gpu
+    L = gpu.load_library()
+    assert L.bg_version() >= 10
+
+
+def test_init_fails_loudly_without_gpu():
+    """On a GPU-less host bg_init must refuse (no silent CPU fallback)."""
+    import torch
+    if torch.cuda.is_available():
+        pytest.skip("GPU present; covered by gpu-marked tests")
+    from datafusion_ballista_amd import gpu
+    L = gpu.load_library()
+    rc = L.bg_init(0)
+    assert rc == -2  # BG_ERR_NO_GPU
+    msg = L.bg_last_error().decode()
+    assert "no HIP device" in msg or "no gfx950" in msg
+
+
+def test_compute_requires_init():
+    """Compute entry points refuse before bg_init (fail-loud contract)."""
+    import torch
+    if torch.cuda.is_available():
+        pytest.skip("GPU present")
+    from datafusion_ballista_amd import gpu
+    L = gpu.load_library()
+    out = ctypes.c_int64()
+    rc = L.bg_mask_to_indices(None, ctypes.c_int64(0), None, ctypes.byref(out))
+    assert rc == -2
+    assert b"fallback" in L.bg_last_error()
