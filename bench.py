@@ -1,0 +1,229 @@
+#!/usr/bin/env python3
+"""Flagship bench: TPC-H q6 SF100 hot path (scan+filter+SUM(Decimal128)) on
+MI355X, inputs resident in HBM — BASELINE.json configs[1], the largest
+single-GPU configuration ("TPC-H q6 SF100 on 1 MI355X, Parquet resident in
+HBM").  A "step" = one full q6 pass over the SF100-shaped lineitem columns
+(600,037,902 rows pinned by tpch_plan_stability/fixtures.rs:50; synthetic —
+no network for real dbgen data).
+
+Contract: python bench.py --gpus N --steps K --warmup W
+  - W untimed warmup steps, then exactly K timed steps bracketed by a
+    barrier + torch.cuda.synchronize() on both sides; MAX over ranks;
+    rank 0 prints ONE JSON line.
+  - value = whole-job rows/s over all N GPUs; scaling "weak" (each rank owns
+    its own SF100-shaped shard; the only exchange is the final-aggregate
+    merge of one Decimal128 partial per rank — q6 has no data-path
+    collective).
+  - roofline: dominant kernel = k_q6_agg; achieved = algorithmic bytes
+    (52 B/row: Date32 4 + 3 x Decimal128 16) / kernel ms
+    measured with hipEvents on the launch stream inside libballista_gpu.so
+    (bg_last_kernel_ms); peak 8 TB/s HBM3E (MI355X_MICROARCH.md);
+    traffic from profiles/pmc_q6.json when a PMC calibration exists.
+  - cpu_baseline: the oracle's q6 C path (kind "port", test infrastructure
+    used here only as the reported baseline) timed on this box's host cores
+    over a bounded sample.
+"""
+
+import argparse
+import json
+import os
+import sys
+import time
+
+ROOT = os.path.dirname(os.path.abspath(__file__))
+sys.path.insert(0, ROOT)
+
+ROWS_SF100 = 600_037_902  # lineitem row count at SF100 (fixtures.rs:50)
+BYTES_PER_ROW = 52        # Date32(4) + 3 x Decimal128(16) — SURVEY.md §8d
+HBM_PEAK_GBS = 8000.0     # 8 TB/s HBM3E spec peak (MI355X_MICROARCH.md)
+
+
+def parse_args():
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=10)
+    p.add_argument("--warmup", type=int, default=3)
+    p.add_argument("--rows", type=int, default=ROWS_SF100,
+                   help="rows per GPU (default SF100 lineitem)")
+    p.add_argument("--cpu-sample-rows", type=int, default=30_000_000)
+    p.add_argument("--skip-cpu-baseline", action="store_true")
+    return p.parse_args()
+
+
+def setup_dist(args):
+    import torch
+    rank = int(os.environ.get("RANK", "0"))
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
+    if world > 1:
+        import torch.distributed as dist
+        os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+        dist.init_process_group("nccl")
+        torch.cuda.set_device(local_rank)
+        return dist, rank, world, local_rank
+    torch.cuda.set_device(local_rank)
+    return None, rank, world, local_rank
+
+
+def i128_to_limbs(v: int):
+    """exact i128 -> 4 x 32-bit limbs (for exact allreduce in int64)."""
+    u = v & ((1 << 128) - 1)
+    return [(u >> (32 * i)) & 0xFFFFFFFF for i in range(4)]
+
+
+def limbs_to_i128(limbs):
+    u = 0
+    for i, l in enumerate(limbs):
+        u += int(l) << (32 * i)
+    u &= (1 << 128) - 1
+    if u >= 1 << 127:
+        u -= 1 << 128
+    return u
+
+
+def cpu_baseline_leg(sample_rows: int):
+    """Oracle q6 C path on host cores — reported baseline, not the target."""
+    import numpy as np
+    import oracle
+    from datafusion_ballista_amd import tpch_synth
+    li = tpch_synth.lineitem_numpy(sample_rows, seed=123)
+    d16 = tpch_synth.dec128_pairs_np(li["l_discount"]).view(np.uint8).reshape(-1)
+    q16 = tpch_synth.dec128_pairs_np(li["l_quantity"]).view(np.uint8).reshape(-1)
+    p16 = tpch_synth.dec128_pairs_np(li["l_extendedprice"]).view(np.uint8).reshape(-1)
+    t0 = time.perf_counter()
+    oracle.q6(li["l_shipdate"], d16, q16, p16, tpch_synth.Q6_DATE_LO,
+              tpch_synth.Q6_DATE_HI, tpch_synth.Q6_DISC_LO,
+              tpch_synth.Q6_DISC_HI, tpch_synth.Q6_QTY_LT)
+    dt = time.perf_counter() - t0
+    return {
+        "value": sample_rows / dt,
+        "unit": "rows/s",
+        "cores": 1,
+        "kind": "port",
+        "sample": f"{sample_rows} synthetic q6-shaped rows, "
+                  f"{dt:.2f}s single-thread C oracle",
+    }
+
+
+def main():
+    args = parse_args()
+    import torch
+    if not torch.cuda.is_available():
+        print(json.dumps({"error": "no GPU visible; bench requires MI355X"}))
+        sys.exit(1)
+    dist, rank, world, local_rank = setup_dist(args)
+    device = torch.device(f"cuda:{local_rank}")
+
+    from datafusion_ballista_amd import gpu, tpch_synth
+    ctx = gpu.GpuStageContext(local_rank)
+
+    n = args.rows
+    # generate the shard directly in HBM (no host staging)
+    cols = tpch_synth.lineitem_torch(n, device, seed=1000 + rank)
+    torch.cuda.synchronize()
+
+    def col_of(t, dtype):
+        return gpu.BgColumn(dtype, 15, 2, 0, t.data_ptr(), None,
+                            t.shape[0])
+
+    sd = col_of(cols["l_shipdate"], gpu.BG_DT_DATE32)
+    cd = col_of(cols["l_discount"], gpu.BG_DT_DECIMAL128)
+    cq = col_of(cols["l_quantity"], gpu.BG_DT_DECIMAL128)
+    cp = col_of(cols["l_extendedprice"], gpu.BG_DT_DECIMAL128)
+
+    def step():
+        cnt, total = ctx.q6_agg(sd, cd, cq, cp, tpch_synth.Q6_DATE_LO,
+                                tpch_synth.Q6_DATE_HI, tpch_synth.Q6_DISC_LO,
+                                tpch_synth.Q6_DISC_HI, tpch_synth.Q6_QTY_LT)
+        # final-aggregate merge across ranks (exact: 32-bit limbs in int64)
+        if dist is not None:
+            t = torch.tensor(i128_to_limbs(total) + [cnt], dtype=torch.int64,
+                             device=device)
+            dist.all_reduce(t)
+            merged = t.cpu().tolist()
+            total = limbs_to_i128(merged[:4])
+            cnt = merged[4]
+        return cnt, total
+
+    def barrier_sync():
+        if dist is not None:
+            dist.barrier()
+        torch.cuda.synchronize()
+
+    kernel_ms_samples = []
+    for _ in range(args.warmup):
+        step()
+    barrier_sync()
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        cnt, total = step()
+        kernel_ms_samples.append(ctx.L.bg_last_kernel_ms())
+    barrier_sync()
+    elapsed = time.perf_counter() - t0
+
+    # MAX over ranks
+    if dist is not None:
+        t = torch.tensor([elapsed], dtype=torch.float64, device=device)
+        dist.all_reduce(t, op=torch.distributed.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    if rank == 0:
+        ms_per_step = elapsed / args.steps * 1000.0
+        total_rows = n * world * args.steps
+        value = total_rows / elapsed
+
+        kms = sorted(kernel_ms_samples)[len(kernel_ms_samples) // 2]
+        algo_bytes = float(n) * BYTES_PER_ROW
+        achieved_gbs = algo_bytes / (kms * 1e-3) / 1e9
+
+        traffic = None
+        pmc_path = os.path.join(ROOT, "profiles", "pmc_q6.json")
+        if os.path.exists(pmc_path):
+            try:
+                cal = json.load(open(pmc_path))
+                traffic = cal["hbm_bytes_per_row"] * n
+            except Exception:
+                traffic = None
+
+        out = {
+            "metric": "tpch_q6_sf100_rows_per_s",
+            "value": value,
+            "unit": "rows/s",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": ms_per_step,
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "int128",
+            "data": "synthetic",
+            "config": {
+                "workload": "tpch_q6_sf100_hbm_resident",
+                "rows_per_gpu": n,
+                "bytes_per_row": BYTES_PER_ROW,
+                "k_filter_selectivity": float(cnt) / n,
+                "parallelism": f"dp{world}",
+            },
+            "roofline": {
+                "bound": "hbm",
+                "achieved": achieved_gbs,
+                "peak": HBM_PEAK_GBS,
+                "unit": "GB/s",
+                "frac": achieved_gbs / HBM_PEAK_GBS,
+                "traffic": traffic,
+                "kernel": "k_q6_agg",
+                "kernel_ms_median": kms,
+            },
+            "cpu_baseline": None,
+        }
+        if not args.skip_cpu_baseline and world == 1:
+            out["cpu_baseline"] = cpu_baseline_leg(args.cpu_sample_rows)
+        print(json.dumps(out), flush=True)
+
+    if dist is not None:
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

@@ -45,8 +45,15 @@ static inline int64_t bg_imin64(int64_t a, int64_t b) { return a < b ? a : b; }
 static thread_local char g_err[512] = "";
 static thread_local bool g_inited = false;
 
+static thread_local double g_last_kernel_ms = 0.0;
+
 extern "C" const char* bg_last_error(void) { return g_err; }
 extern "C" int bg_version(void) { return 10; }
+
+// Duration of the most recent timed hot kernel (bg_q6_agg / bg_q1_agg),
+// measured with hipEvents on the launch stream — feeds bench.py's
+// roofline.achieved (algorithmic bytes / kernel time).
+extern "C" double bg_last_kernel_ms(void) { return g_last_kernel_ms; }
 
 static int set_err(int code, const char* msg) {
   snprintf(g_err, sizeof(g_err), "%s", msg);
@@ -741,6 +748,10 @@ extern "C" int bg_q6_agg(const bg_column* shipdate, const bg_column* discount,
   HIP_TRY(hipMemset(d_acc, 0, 3 * sizeof(u64)));
   int blocks = (int)bg_imin64((n + BG_BLOCK - 1) / BG_BLOCK, BG_MAX_BLOCKS);
   if (blocks == 0) blocks = 1;
+  hipEvent_t ev0, ev1;
+  HIP_TRY(hipEventCreate(&ev0));
+  HIP_TRY(hipEventCreate(&ev1));
+  HIP_TRY(hipEventRecord(ev0, 0));
   hipLaunchKernelGGL(k_q6_agg, dim3(blocks), dim3(BG_BLOCK), 0, 0,
                      (const int32_t*)shipdate->d_data,
                      (const ulong2*)discount->d_data,
@@ -748,6 +759,13 @@ extern "C" int bg_q6_agg(const bg_column* shipdate, const bg_column* discount,
                      (const ulong2*)extendedprice->d_data, n, date_lo, date_hi,
                      disc_lo, disc_hi, qty_lt, d_acc, d_acc + 1, d_acc + 2);
   HIP_TRY(hipGetLastError());
+  HIP_TRY(hipEventRecord(ev1, 0));
+  HIP_TRY(hipEventSynchronize(ev1));
+  float ms = 0.f;
+  HIP_TRY(hipEventElapsedTime(&ms, ev0, ev1));
+  g_last_kernel_ms = (double)ms;
+  HIP_TRY(hipEventDestroy(ev0));
+  HIP_TRY(hipEventDestroy(ev1));
   u64 h[3];
   HIP_TRY(hipMemcpy(h, d_acc, 3 * sizeof(u64), hipMemcpyDeviceToHost));
   HIP_TRY(hipFree(d_acc));
@@ -857,6 +875,10 @@ extern "C" int bg_q1_agg(const bg_column* rf, const bg_column* ls,
   HIP_TRY(hipMemset(d_counts, 0, Q1_GROUPS * sizeof(u64)));
   int blocks = (int)bg_imin64((n + BG_BLOCK - 1) / BG_BLOCK, BG_MAX_BLOCKS);
   if (blocks == 0) blocks = 1;
+  hipEvent_t ev0, ev1;
+  HIP_TRY(hipEventCreate(&ev0));
+  HIP_TRY(hipEventCreate(&ev1));
+  HIP_TRY(hipEventRecord(ev0, 0));
   hipLaunchKernelGGL(k_q1_agg, dim3(blocks), dim3(BG_BLOCK), 0, 0,
                      (const uint8_t*)rf->d_data, (const uint8_t*)ls->d_data,
                      (const ulong2*)quantity->d_data,
@@ -866,6 +888,13 @@ extern "C" int bg_q1_agg(const bg_column* rf, const bg_column* ls,
                      (const int32_t*)shipdate->d_data, n, date_le, d_sums,
                      d_counts);
   HIP_TRY(hipGetLastError());
+  HIP_TRY(hipEventRecord(ev1, 0));
+  HIP_TRY(hipEventSynchronize(ev1));
+  float ms = 0.f;
+  HIP_TRY(hipEventElapsedTime(&ms, ev0, ev1));
+  g_last_kernel_ms = (double)ms;
+  HIP_TRY(hipEventDestroy(ev0));
+  HIP_TRY(hipEventDestroy(ev1));
   u64 sums[Q1_GROUPS * Q1_ACCS * 2];
   u64 counts[Q1_GROUPS];
   HIP_TRY(hipMemcpy(sums, d_sums, sizeof(sums), hipMemcpyDeviceToHost));

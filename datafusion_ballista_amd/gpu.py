@@ -76,6 +76,7 @@ def load_library() -> ctypes.CDLL:
                 "__graft_entry__.build()); there is no CPU fallback")
         _lib = ctypes.CDLL(p)
         _lib.bg_last_error.restype = ctypes.c_char_p
+        _lib.bg_last_kernel_ms.restype = ctypes.c_double
     return _lib
 
 

@@ -62,6 +62,9 @@ typedef enum {
 
 const char* bg_last_error(void);
 int bg_version(void);
+/* duration (ms) of the most recent fused-aggregate kernel launch, measured
+ * with hipEvents on the launch stream — roofline evidence for bench.py */
+double bg_last_kernel_ms(void);
 
 /* ---- device/session ---- */
 int bg_init(int device_ordinal); /* fails loudly when no GPU is present */

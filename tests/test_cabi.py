@@ -15,7 +15,7 @@ def header_symbols():
     src = open(HEADER).read()
     # strip comments
     src = re.sub(r"/\*.*?\*/", "", src, flags=re.S)
-    syms = re.findall(r"^\s*(?:const\s+char\s*\*|int)\s+(bg_\w+)\s*\(", src,
+    syms = re.findall(r"^\s*(?:const\s+char\s*\*|int|double)\s+(bg_\w+)\s*\(", src,
                       flags=re.M)
     assert len(syms) >= 15
     return syms
