@@ -911,3 +911,160 @@ extern "C" int bg_q1_agg(const bg_column* rf, const bg_column* ls,
   }
   return BG_OK;
 }
+
+// ---------------------------------------------------------------------------
+// HashJoinExec build/probe (DataFusion 55 HashJoinExec
+// partitioned-mode semantics for INNER equi-joins on integer keys).
+// Chained hash table: head[nb] + next[n_build]; insertion via atomicExch is
+// order-free (chain order is not part of the contract — SQL pins the result
+// MULTISET; parity tests compare sorted pairs).  Probe is two-phase
+// (count -> exclusive scan -> fill) so the output is probe-major with exact
+// offsets and no atomic append nondeterminism in sizes.
+// ---------------------------------------------------------------------------
+struct BgJoinTable {
+  u64* keys;      // build keys (copied for probe-time equality check)
+  int* head;      // bucket heads (-1 empty)
+  int* next;      // chain links
+  int64_t n_build;
+  u64 mask;       // nb - 1
+  i64* probe_offsets = nullptr;  // per-probe-row output offsets (count phase)
+  int64_t probe_n = 0;
+};
+
+__global__ void k_join_build(const int64_t* keys, int64_t n, int* head,
+                             int* next, u64* key_copy, u64 mask) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    const u64 k = (u64)keys[i];
+    key_copy[i] = k;
+    const u64 b = bg_hash_u64(k) & mask;
+    next[i] = atomicExch(&head[b], (int)i);
+  }
+}
+
+__global__ void k_join_count(const int64_t* probe_keys, int64_t n_probe,
+                             const int* head, const int* next,
+                             const u64* build_keys, u64 mask, u64* counts) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n_probe;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    const u64 k = (u64)probe_keys[i];
+    const u64 b = bg_hash_u64(k) & mask;
+    u64 c = 0;
+    for (int j = head[b]; j >= 0; j = next[j])
+      if (build_keys[j] == k) c++;
+    counts[i] = c;
+  }
+}
+
+__global__ void k_join_fill(const int64_t* probe_keys, int64_t n_probe,
+                            const int* head, const int* next,
+                            const u64* build_keys, u64 mask,
+                            const i64* offsets, uint32_t* out_probe,
+                            uint32_t* out_build) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n_probe;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    const u64 k = (u64)probe_keys[i];
+    const u64 b = bg_hash_u64(k) & mask;
+    i64 w = offsets[i];
+    for (int j = head[b]; j >= 0; j = next[j]) {
+      if (build_keys[j] == k) {
+        out_probe[w] = (uint32_t)i;
+        out_build[w] = (uint32_t)j;
+        ++w;
+      }
+    }
+  }
+}
+
+static u64 next_pow2_u64(u64 x) {
+  u64 p = 1;
+  while (p < x) p <<= 1;
+  return p;
+}
+
+extern "C" int bg_hashjoin_build(const bg_column* build_keys, int64_t n,
+                                 void** out_handle) {
+  REQUIRE_INIT();
+  if (build_keys->dtype != BG_DT_INT64)
+    return set_err(BG_ERR_UNSUPPORTED, "join keys must be INT64 (round 1)");
+  if (n > 0x7fffffffLL) return set_err(BG_ERR_INVALID, "build side > 2^31 rows");
+  BgJoinTable t{};
+  t.n_build = n;
+  const u64 nb = next_pow2_u64((u64)(n > 4 ? n * 2 : 8));
+  t.mask = nb - 1;
+  HIP_TRY(hipMalloc(&t.keys, sizeof(u64) * (n ? n : 1)));
+  HIP_TRY(hipMalloc(&t.head, sizeof(int) * nb));
+  HIP_TRY(hipMalloc(&t.next, sizeof(int) * (n ? n : 1)));
+  HIP_TRY(hipMemset(t.head, 0xff, sizeof(int) * nb));  // -1
+  int blocks = (int)bg_imin64((n + BG_BLOCK - 1) / BG_BLOCK, BG_MAX_BLOCKS);
+  if (blocks == 0) blocks = 1;
+  hipLaunchKernelGGL(k_join_build, dim3(blocks), dim3(BG_BLOCK), 0, 0,
+                     (const int64_t*)build_keys->d_data, n, t.head, t.next,
+                     t.keys, t.mask);
+  HIP_TRY(hipGetLastError());
+  BgJoinTable* h = new BgJoinTable(t);
+  *out_handle = h;
+  return BG_OK;
+}
+
+extern "C" int bg_hashjoin_probe_count(void* handle,
+                                       const bg_column* probe_keys, int64_t n,
+                                       int64_t* out_matches) {
+  REQUIRE_INIT();
+  BgJoinTable* t = (BgJoinTable*)handle;
+  if (probe_keys->dtype != BG_DT_INT64)
+    return set_err(BG_ERR_UNSUPPORTED, "join keys must be INT64 (round 1)");
+  u64* d_counts;
+  i64* d_offs;
+  i64* d_total;
+  HIP_TRY(hipMalloc(&d_counts, sizeof(u64) * (n ? n : 1)));
+  HIP_TRY(hipMalloc(&d_offs, sizeof(i64) * (n ? n : 1)));
+  HIP_TRY(hipMalloc(&d_total, sizeof(i64)));
+  int blocks = (int)bg_imin64((n + BG_BLOCK - 1) / BG_BLOCK, BG_MAX_BLOCKS);
+  if (blocks == 0) blocks = 1;
+  hipLaunchKernelGGL(k_join_count, dim3(blocks), dim3(BG_BLOCK), 0, 0,
+                     (const int64_t*)probe_keys->d_data, n, t->head, t->next,
+                     t->keys, t->mask, d_counts);
+  hipLaunchKernelGGL(k_exclusive_scan_i64, dim3(1), dim3(BG_BLOCK), 0, 0,
+                     d_counts, n, d_offs, d_total);
+  i64 total = 0;
+  HIP_TRY(hipMemcpy(&total, d_total, sizeof(i64), hipMemcpyDeviceToHost));
+  HIP_TRY(hipFree(d_counts));
+  // stash offsets on the handle for the fill call
+  if (t->probe_offsets) (void)hipFree(t->probe_offsets);
+  t->probe_offsets = d_offs;
+  t->probe_n = n;
+  HIP_TRY(hipFree(d_total));
+  *out_matches = total;
+  return BG_OK;
+}
+
+extern "C" int bg_hashjoin_probe_fill(void* handle,
+                                      const bg_column* probe_keys, int64_t n,
+                                      uint32_t* d_out_probe,
+                                      uint32_t* d_out_build) {
+  REQUIRE_INIT();
+  BgJoinTable* t = (BgJoinTable*)handle;
+  if (!t->probe_offsets || t->probe_n != n)
+    return set_err(BG_ERR_INVALID, "call bg_hashjoin_probe_count first");
+  int blocks = (int)bg_imin64((n + BG_BLOCK - 1) / BG_BLOCK, BG_MAX_BLOCKS);
+  if (blocks == 0) blocks = 1;
+  hipLaunchKernelGGL(k_join_fill, dim3(blocks), dim3(BG_BLOCK), 0, 0,
+                     (const int64_t*)probe_keys->d_data, n, t->head, t->next,
+                     t->keys, t->mask, t->probe_offsets, d_out_probe,
+                     d_out_build);
+  HIP_TRY(hipGetLastError());
+  return BG_OK;
+}
+
+extern "C" int bg_hashjoin_free(void* handle) {
+  REQUIRE_INIT();
+  BgJoinTable* t = (BgJoinTable*)handle;
+  if (!t) return BG_OK;
+  (void)hipFree(t->keys);
+  (void)hipFree(t->head);
+  (void)hipFree(t->next);
+  if (t->probe_offsets) (void)hipFree(t->probe_offsets);
+  delete t;
+  return BG_OK;
+}

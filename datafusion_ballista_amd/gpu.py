@@ -273,3 +273,40 @@ class GpuStageContext:
                     for a in range(5)]
             out[g] = (int(counts[g]), vals)
         return out
+
+
+class GpuHashJoin:
+    """INNER equi-join (Int64 keys) — bg_hashjoin_* wrapper.
+    Build once, probe many (HashJoinExec build/probe semantics)."""
+
+    def __init__(self, ctx: GpuStageContext, build_keys: BgColumn, n_build: int):
+        self.ctx = ctx
+        self._handle = ctypes.c_void_p()
+        _check(ctx.L.bg_hashjoin_build(ctypes.byref(build_keys),
+                                       ctypes.c_int64(n_build),
+                                       ctypes.byref(self._handle)),
+               "bg_hashjoin_build")
+
+    def probe(self, probe_keys: BgColumn, n_probe: int):
+        """-> (probe_idx DeviceBuffer, build_idx DeviceBuffer, n_matches)."""
+        ctx = self.ctx
+        matches = ctypes.c_int64()
+        _check(ctx.L.bg_hashjoin_probe_count(self._handle,
+                                             ctypes.byref(probe_keys),
+                                             ctypes.c_int64(n_probe),
+                                             ctypes.byref(matches)),
+               "bg_hashjoin_probe_count")
+        m = matches.value
+        pbuf = ctx.alloc(max(4 * m, 4))
+        bbuf = ctx.alloc(max(4 * m, 4))
+        _check(ctx.L.bg_hashjoin_probe_fill(self._handle,
+                                            ctypes.byref(probe_keys),
+                                            ctypes.c_int64(n_probe),
+                                            pbuf.ptr, bbuf.ptr),
+               "bg_hashjoin_probe_fill")
+        return pbuf, bbuf, m
+
+    def free(self):
+        if self._handle and self._handle.value:
+            self.ctx.L.bg_hashjoin_free(self._handle)
+            self._handle = ctypes.c_void_p()

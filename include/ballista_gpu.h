@@ -158,6 +158,24 @@ int bg_hash_repartition(const bg_column* key_cols, int32_t nkeys,
                         uint32_t* d_indices, int64_t* d_offsets,
                         void** d_out /* ncols device pointers */);
 
+/* ---- HashJoinExec build/probe (INNER equi-join, Int64 keys) ----
+ * Replaces DataFusion 55 HashJoinExec's build/probe arithmetic for
+ * partitioned hash joins (planned when
+ * `prefer_hash_join=true`, core/src/extension.rs:850-858 + the opt-in test
+ * client/tests/context_checks.rs:1034-1063).  Contract: the emitted
+ * (probe_idx, build_idx) pair MULTISET equals the reference's inner-join
+ * result set; pairs are probe-major (ascending probe_idx, exact per-row
+ * offsets from the count phase); chain order within one probe row is
+ * unspecified, as in the reference (SQL does not pin intra-row order). */
+int bg_hashjoin_build(const bg_column* build_keys, int64_t n_build,
+                      void** out_handle);
+int bg_hashjoin_probe_count(void* handle, const bg_column* probe_keys,
+                            int64_t n_probe, int64_t* out_matches);
+int bg_hashjoin_probe_fill(void* handle, const bg_column* probe_keys,
+                           int64_t n_probe, uint32_t* d_out_probe,
+                           uint32_t* d_out_build);
+int bg_hashjoin_free(void* handle);
+
 /* ---- fused filter+aggregate stages ---- */
 
 /* TPC-H q6 stage 1 (scan+filter+aggregate, approved/q6.txt):

@@ -247,3 +247,29 @@ def dec128_from_ints(vals) -> np.ndarray:
         out[i * 16:(i + 1) * 16] = np.frombuffer(
             int(v).to_bytes(16, "little", signed=True), dtype=np.uint8)
     return out
+
+
+def hashjoin_pairs(build: np.ndarray, probe: np.ndarray):
+    """-> (probe_idx u32[], build_idx u32[]) inner-join pair set (O(n*m))."""
+    L = lib()
+    L.oracle_hashjoin_count.restype = ctypes.c_int64
+    L.oracle_hashjoin_count.argtypes = [ctypes.POINTER(ctypes.c_int64),
+                                        ctypes.c_int64,
+                                        ctypes.POINTER(ctypes.c_int64),
+                                        ctypes.c_int64]
+    L.oracle_hashjoin_pairs.argtypes = [ctypes.POINTER(ctypes.c_int64),
+                                        ctypes.c_int64,
+                                        ctypes.POINTER(ctypes.c_int64),
+                                        ctypes.c_int64,
+                                        ctypes.POINTER(ctypes.c_uint32),
+                                        ctypes.POINTER(ctypes.c_uint32)]
+    b = np.ascontiguousarray(build, dtype=np.int64)
+    p = np.ascontiguousarray(probe, dtype=np.int64)
+    total = L.oracle_hashjoin_count(_p(b, ctypes.c_int64), len(b),
+                                    _p(p, ctypes.c_int64), len(p))
+    op = np.empty(total, dtype=np.uint32)
+    ob = np.empty(total, dtype=np.uint32)
+    L.oracle_hashjoin_pairs(_p(b, ctypes.c_int64), len(b),
+                            _p(p, ctypes.c_int64), len(p),
+                            _p(op, ctypes.c_uint32), _p(ob, ctypes.c_uint32))
+    return op, ob

@@ -310,3 +310,68 @@ EXPORT void oracle_shuffle_index_encode(const int64_t* offsets, int64_t k1,
     memcpy(out + 8 * i, &v, 8); /* little-endian host assumed (x86-64) */
   }
 }
+
+/* ------------------------------------------------------------------ */
+/* HashJoinExec restatement (INNER equi-join, Int64 keys): serial chained
+ * hash build + probe emitting probe-major (probe_idx, build_idx) pairs —
+ * the same build/probe algorithm class as DataFusion's JoinHashMap.
+ * The pair multiset is the contract (DataFusion inner-join result set);
+ * within a probe row this oracle emits build indices ASCENDING (it inserts
+ * build rows in reverse so chain traversal is ascending) — parity tests
+ * compare sorted pairs, so any intra-row order on the GPU side is fine.
+ * Two-phase like the GPU: count then fill.                               */
+#include <stdlib.h>
+
+typedef struct {
+  int32_t* head;
+  int32_t* next;
+  uint64_t mask;
+} join_table;
+
+static join_table join_build(const int64_t* build, int64_t nb) {
+  uint64_t cap = 8;
+  while (cap < (uint64_t)(nb * 2)) cap <<= 1;
+  join_table t;
+  t.mask = cap - 1;
+  t.head = malloc(sizeof(int32_t) * cap);
+  t.next = malloc(sizeof(int32_t) * (nb ? nb : 1));
+  for (uint64_t b = 0; b < cap; ++b) t.head[b] = -1;
+  for (int64_t j = nb - 1; j >= 0; --j) { /* reverse => ascending chains */
+    uint64_t b = bg_hash_u64((uint64_t)build[j]) & t.mask;
+    t.next[j] = t.head[b];
+    t.head[b] = (int32_t)j;
+  }
+  return t;
+}
+
+EXPORT int64_t oracle_hashjoin_count(const int64_t* build, int64_t nb,
+                                     const int64_t* probe, int64_t np) {
+  join_table t = join_build(build, nb);
+  int64_t total = 0;
+  for (int64_t i = 0; i < np; ++i) {
+    uint64_t b = bg_hash_u64((uint64_t)probe[i]) & t.mask;
+    for (int32_t j = t.head[b]; j >= 0; j = t.next[j])
+      if (build[j] == probe[i]) total++;
+  }
+  free(t.head);
+  free(t.next);
+  return total;
+}
+
+EXPORT void oracle_hashjoin_pairs(const int64_t* build, int64_t nb,
+                                  const int64_t* probe, int64_t np,
+                                  uint32_t* out_probe, uint32_t* out_build) {
+  join_table t = join_build(build, nb);
+  int64_t w = 0;
+  for (int64_t i = 0; i < np; ++i) {
+    uint64_t b = bg_hash_u64((uint64_t)probe[i]) & t.mask;
+    for (int32_t j = t.head[b]; j >= 0; j = t.next[j])
+      if (build[j] == probe[i]) {
+        out_probe[w] = (uint32_t)i;
+        out_build[w] = (uint32_t)j;
+        w++;
+      }
+  }
+  free(t.head);
+  free(t.next);
+}

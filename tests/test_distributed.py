@@ -1,0 +1,71 @@
+"""Multi-process CPU coverage of the N>1 path (gloo, world_size=2).
+
+bench.py's distributed pieces — the exact Decimal128 limb all-reduce that
+merges per-rank q6 partial aggregates, and the MAX-over-ranks elapsed-time
+reduction — must be correct by construction before the driver ever runs the
+8-GPU scaling bench (RCCL path, same torch.distributed calls)."""
+
+import os
+
+import numpy as np
+import pytest
+import torch
+import torch.multiprocessing as mp
+
+from bench import i128_to_limbs, limbs_to_i128
+
+
+def test_limb_codec_roundtrip():
+    rng = np.random.default_rng(0)
+    vals = [0, 1, -1, 2**64, -(2**64), 2**126, -(2**126) + 5]
+    vals += [int(rng.integers(-2**62, 2**62)) * int(rng.integers(1, 2**40))
+             for _ in range(50)]
+    for v in vals:
+        assert limbs_to_i128(i128_to_limbs(v)) == v
+
+
+def _worker(rank, world, port, results):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    torch.distributed.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        # each rank holds an exact i128 partial (as a q6 shard would)
+        partials = [123456789012345678901234567890 * 3,
+                    -987654321098765432109876543210]
+        cnt = [17, 25]
+        t = torch.tensor(i128_to_limbs(partials[rank]) + [cnt[rank]],
+                         dtype=torch.int64)
+        torch.distributed.all_reduce(t)
+        merged = t.tolist()
+        total = limbs_to_i128(merged[:4])
+        count = merged[4]
+
+        # MAX-over-ranks elapsed
+        elapsed = [1.5, 2.5][rank]
+        e = torch.tensor([elapsed], dtype=torch.float64)
+        torch.distributed.all_reduce(e, op=torch.distributed.ReduceOp.MAX)
+
+        results[rank] = (total, count, float(e.item()))
+    finally:
+        torch.distributed.destroy_process_group()
+
+
+def test_exact_i128_allreduce_world2():
+    ctx = mp.get_context("spawn")
+    with ctx.Manager() as mgr:
+        results = mgr.dict()
+        port = 29631
+        ps = [ctx.Process(target=_worker, args=(r, 2, port, results))
+              for r in range(2)]
+        for p in ps:
+            p.start()
+        for p in ps:
+            p.join(120)
+            assert p.exitcode == 0
+        want_total = (123456789012345678901234567890 * 3
+                      - 987654321098765432109876543210)
+        for r in range(2):
+            total, count, emax = results[r]
+            assert total == want_total
+            assert count == 42
+            assert emax == pytest.approx(2.5)
