@@ -1068,3 +1068,275 @@ extern "C" int bg_hashjoin_free(void* handle) {
   delete t;
   return BG_OK;
 }
+
+// ---------------------------------------------------------------------------
+// General hash group-by (AggregateExec Partial/Single for arbitrary group
+// cardinality — SURVEY.md §8a row 2; q3-class: ~11.6M groups over 100M+
+// rows).  Open-addressing table with linear probing; a slot stores the
+// FIRST row index carrying its key (claimed with one atomicCAS) and the
+// SLOT INDEX is the (sparse) group id — no group-counter handshake, no
+// cross-workgroup spin on a second word (placement-independent by
+// construction: claims and accumulations are device-scope atomics).
+// Accumulators are slot-indexed; the host compacts occupied slots to dense
+// groups with the existing mask->indices + gather machinery.
+// ---------------------------------------------------------------------------
+#define BG_MAX_AGGS 8
+#define BG_AGG_SUM_DEC128 0
+#define BG_AGG_SUM_I64 1
+
+struct AggArgs {
+  int naggs;
+  struct {
+    const void* data;
+    int op;
+  } a[BG_MAX_AGGS];
+};
+
+__device__ __forceinline__ bool keys_equal_rows(const KeyArgs& keys,
+                                                int64_t r1, int64_t r2) {
+  for (int c = 0; c < keys.nkeys; ++c) {
+    switch (keys.k[c].dtype) {
+      case BG_DT_INT64:
+        if (reinterpret_cast<const int64_t*>(keys.k[c].data)[r1] !=
+            reinterpret_cast<const int64_t*>(keys.k[c].data)[r2])
+          return false;
+        break;
+      case BG_DT_INT32:
+      case BG_DT_DATE32:
+        if (reinterpret_cast<const int32_t*>(keys.k[c].data)[r1] !=
+            reinterpret_cast<const int32_t*>(keys.k[c].data)[r2])
+          return false;
+        break;
+      case BG_DT_DECIMAL128: {
+        const ulong2 a = reinterpret_cast<const ulong2*>(keys.k[c].data)[r1];
+        const ulong2 b = reinterpret_cast<const ulong2*>(keys.k[c].data)[r2];
+        if (a.x != b.x || a.y != b.y) return false;
+        break;
+      }
+      case BG_DT_DICT8:
+        if (reinterpret_cast<const uint8_t*>(keys.k[c].data)[r1] !=
+            reinterpret_cast<const uint8_t*>(keys.k[c].data)[r2])
+          return false;
+        break;
+      default:
+        return false;
+    }
+  }
+  return true;
+}
+
+__device__ __forceinline__ u64 hash_keys_row(const KeyArgs& keys, int64_t i) {
+  u64 h = 0;
+  for (int c = 0; c < keys.nkeys; ++c) {
+    u64 hc;
+    switch (keys.k[c].dtype) {
+      case BG_DT_INT64:
+        hc = bg_hash_u64((u64) reinterpret_cast<const int64_t*>(keys.k[c].data)[i]);
+        break;
+      case BG_DT_INT32:
+      case BG_DT_DATE32:
+        hc = bg_hash_u32((uint32_t) reinterpret_cast<const int32_t*>(keys.k[c].data)[i]);
+        break;
+      case BG_DT_DECIMAL128: {
+        const ulong2 v = reinterpret_cast<const ulong2*>(keys.k[c].data)[i];
+        hc = bg_hash_u128(v.x, v.y);
+        break;
+      }
+      case BG_DT_DICT8:
+        hc = bg_hash_u64((u64) reinterpret_cast<const uint8_t*>(keys.k[c].data)[i]);
+        break;
+      default:
+        hc = 0;
+    }
+    h = (c == 0) ? hc : bg_combine_hashes(hc, h);
+  }
+  return h;
+}
+
+// slot_row: i32[cap], -1 empty; acc: u64[cap * naggs * 2]; counts: u64[cap]
+__global__ void k_hashagg(KeyArgs keys, AggArgs aggs, const u64* mask_words,
+                          int64_t n, int* slot_row, u64 cap_mask,
+                          u64* acc, u64* counts, int* err_flag) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    if (mask_words && !((mask_words[i >> 6] >> (i & 63)) & 1)) continue;
+    const u64 h = hash_keys_row(keys, i);
+    u64 slot = h & cap_mask;
+    int64_t probes = 0;
+    int64_t my_slot = -1;
+    while (true) {
+      int cur = __hip_atomic_load(&slot_row[slot], __ATOMIC_RELAXED,
+                                  __HIP_MEMORY_SCOPE_AGENT);
+      if (cur == -1) {
+        int old = atomicCAS(&slot_row[slot], -1, (int)i);
+        if (old == -1) { my_slot = (int64_t)slot; break; }
+        cur = old;
+      }
+      if (cur >= 0 && keys_equal_rows(keys, (int64_t)cur, i)) {
+        my_slot = (int64_t)slot;
+        break;
+      }
+      slot = (slot + 1) & cap_mask;
+      if (++probes > (int64_t)cap_mask) {  // table full
+        atomicExch(err_flag, 1);
+        return;
+      }
+    }
+    atomicAdd(&counts[my_slot], 1ull);
+    for (int a = 0; a < aggs.naggs; ++a) {
+      u64* base = acc + ((u64)my_slot * aggs.naggs + a) * 2;
+      switch (aggs.a[a].op) {
+        case BG_AGG_SUM_DEC128: {
+          const ulong2 v =
+              reinterpret_cast<const ulong2*>(aggs.a[a].data)[i];
+          atomic_add_i128(base, base + 1, make_i128(v.x, (i64)v.y));
+          break;
+        }
+        case BG_AGG_SUM_I64: {
+          const i64 v = reinterpret_cast<const int64_t*>(aggs.a[a].data)[i];
+          // exact i64 sum in two's complement (wrap == i64 semantics);
+          // keep the i128 carry so large sums stay exact at scale
+          atomic_add_i128(base, base + 1, (i128)v);
+          break;
+        }
+        default:
+          break;
+      }
+    }
+  }
+}
+
+// occupancy bitmask over slots (feeds the stable compaction)
+__global__ void k_slot_occupancy(const int* slot_row, int64_t cap,
+                                 u64* mask_words) {
+  const int64_t wave_global =
+      ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) / BG_WAVE;
+  const int64_t nwaves = ((int64_t)gridDim.x * blockDim.x) / BG_WAVE;
+  const int64_t nwords = (cap + 63) / 64;
+  for (int64_t w = wave_global; w < nwords; w += nwaves) {
+    const int64_t s = w * BG_WAVE + lane_id();
+    const bool occ = s < cap && slot_row[s] >= 0;
+    u64 m = __ballot(occ);
+    if (lane_id() == 0) mask_words[w] = m;
+  }
+}
+
+// gather group outputs: for dense group g = 0..ngroups-1 with slot index
+// sidx[g]: first_row[g], counts_out[g], acc_out[g*naggs*2 ..]
+__global__ void k_hashagg_gather(const uint32_t* sidx, int64_t ngroups,
+                                 const int* slot_row, const u64* acc,
+                                 const u64* counts, int naggs,
+                                 uint32_t* first_row, u64* counts_out,
+                                 u64* acc_out) {
+  for (int64_t g = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; g < ngroups;
+       g += (int64_t)gridDim.x * blockDim.x) {
+    const int64_t s = sidx[g];
+    first_row[g] = (uint32_t)slot_row[s];
+    counts_out[g] = counts[s];
+    for (int a = 0; a < naggs; ++a) {
+      acc_out[((u64)g * naggs + a) * 2] = acc[((u64)s * naggs + a) * 2];
+      acc_out[((u64)g * naggs + a) * 2 + 1] = acc[((u64)s * naggs + a) * 2 + 1];
+    }
+  }
+}
+
+extern "C" int bg_hashagg(const bg_column* key_cols, int32_t nkeys,
+                          const bg_column* agg_cols, const int32_t* agg_ops,
+                          int32_t naggs, const uint8_t* d_mask, int64_t n,
+                          int64_t max_groups, uint32_t* d_first_row,
+                          uint8_t* d_acc_out, int64_t* d_counts_out,
+                          int64_t* out_ngroups) {
+  REQUIRE_INIT();
+  if (nkeys <= 0 || nkeys > BG_MAX_KEYS)
+    return set_err(BG_ERR_INVALID, "nkeys out of range [1,4]");
+  if (naggs < 0 || naggs > BG_MAX_AGGS)
+    return set_err(BG_ERR_INVALID, "naggs out of range [0,8]");
+  KeyArgs keys{};
+  keys.nkeys = nkeys;
+  for (int i = 0; i < nkeys; ++i) {
+    keys.k[i].data = key_cols[i].d_data;
+    keys.k[i].valid = key_cols[i].d_validity;
+    keys.k[i].dtype = key_cols[i].dtype;
+  }
+  AggArgs aggs{};
+  aggs.naggs = naggs;
+  for (int i = 0; i < naggs; ++i) {
+    aggs.a[i].data = agg_cols[i].d_data;
+    aggs.a[i].op = agg_ops[i];
+  }
+  u64 cap = 8;
+  while (cap < (u64)(max_groups * 2)) cap <<= 1;
+  int* slot_row;
+  u64* acc;
+  u64* counts;
+  int* err_flag;
+  HIP_TRY(hipMalloc(&slot_row, sizeof(int) * cap));
+  HIP_TRY(hipMalloc(&acc, sizeof(u64) * cap * (naggs ? naggs : 1) * 2));
+  HIP_TRY(hipMalloc(&counts, sizeof(u64) * cap));
+  HIP_TRY(hipMalloc(&err_flag, sizeof(int)));
+  HIP_TRY(hipMemset(slot_row, 0xff, sizeof(int) * cap));
+  HIP_TRY(hipMemset(acc, 0, sizeof(u64) * cap * (naggs ? naggs : 1) * 2));
+  HIP_TRY(hipMemset(counts, 0, sizeof(u64) * cap));
+  HIP_TRY(hipMemset(err_flag, 0, sizeof(int)));
+
+  int blocks = (int)bg_imin64((n + BG_BLOCK - 1) / BG_BLOCK, BG_MAX_BLOCKS);
+  if (blocks == 0) blocks = 1;
+  hipEvent_t ev0, ev1;
+  HIP_TRY(hipEventCreate(&ev0));
+  HIP_TRY(hipEventCreate(&ev1));
+  HIP_TRY(hipEventRecord(ev0, 0));
+  hipLaunchKernelGGL(k_hashagg, dim3(blocks), dim3(BG_BLOCK), 0, 0, keys, aggs,
+                     reinterpret_cast<const u64*>(d_mask), n, slot_row,
+                     cap - 1, acc, counts, err_flag);
+  HIP_TRY(hipGetLastError());
+  HIP_TRY(hipEventRecord(ev1, 0));
+  HIP_TRY(hipEventSynchronize(ev1));
+  float ms = 0.f;
+  HIP_TRY(hipEventElapsedTime(&ms, ev0, ev1));
+  g_last_kernel_ms = (double)ms;
+  HIP_TRY(hipEventDestroy(ev0));
+  HIP_TRY(hipEventDestroy(ev1));
+
+  int err = 0;
+  HIP_TRY(hipMemcpy(&err, err_flag, sizeof(int), hipMemcpyDeviceToHost));
+  if (err) {
+    (void)hipFree(slot_row); (void)hipFree(acc); (void)hipFree(counts);
+    (void)hipFree(err_flag);
+    return set_err(BG_ERR_INVALID, "bg_hashagg: table full (raise max_groups)");
+  }
+
+  // compact occupied slots (ascending slot order — deterministic for a
+  // fixed capacity) using the occupancy mask + stable compaction
+  const int64_t nwords = ((int64_t)cap + 63) / 64;
+  u64* occ_mask;
+  uint32_t* sidx;
+  HIP_TRY(hipMalloc(&occ_mask, sizeof(u64) * nwords));
+  HIP_TRY(hipMalloc(&sidx, sizeof(uint32_t) * cap));
+  int oblocks = (int)bg_imin64((nwords * BG_WAVE + BG_BLOCK - 1) / BG_BLOCK,
+                               BG_MAX_BLOCKS);
+  if (oblocks == 0) oblocks = 1;
+  hipLaunchKernelGGL(k_slot_occupancy, dim3(oblocks), dim3(BG_BLOCK), 0, 0,
+                     slot_row, (int64_t)cap, occ_mask);
+  int64_t ngroups = 0;
+  int rc = bg_mask_to_indices(reinterpret_cast<const uint8_t*>(occ_mask),
+                              (int64_t)cap, sidx, &ngroups);
+  if (rc != BG_OK) return rc;
+
+  int gblocks = (int)bg_imin64((ngroups + BG_BLOCK - 1) / BG_BLOCK,
+                               BG_MAX_BLOCKS);
+  if (gblocks == 0) gblocks = 1;
+  hipLaunchKernelGGL(k_hashagg_gather, dim3(gblocks), dim3(BG_BLOCK), 0, 0,
+                     sidx, ngroups, slot_row, acc, counts, naggs, d_first_row,
+                     reinterpret_cast<u64*>(d_counts_out),
+                     reinterpret_cast<u64*>(d_acc_out));
+  HIP_TRY(hipGetLastError());
+  HIP_TRY(hipDeviceSynchronize());
+  HIP_TRY(hipFree(slot_row));
+  HIP_TRY(hipFree(acc));
+  HIP_TRY(hipFree(counts));
+  HIP_TRY(hipFree(err_flag));
+  HIP_TRY(hipFree(occ_mask));
+  HIP_TRY(hipFree(sidx));
+  *out_ngroups = ngroups;
+  return BG_OK;
+}

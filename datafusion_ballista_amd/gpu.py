@@ -310,3 +310,33 @@ class GpuHashJoin:
         if self._handle and self._handle.value:
             self.ctx.L.bg_hashjoin_free(self._handle)
             self._handle = ctypes.c_void_p()
+
+
+BG_AGG_OP_SUM_DEC128 = 0
+BG_AGG_OP_SUM_I64 = 1
+
+
+def _ctx_hashagg(self, key_cols, agg_cols, agg_ops, n, max_groups,
+                 mask: "DeviceBuffer" = None):
+    """General hash group-by. -> (first_row u32[g], acc i128 bytes
+    [g, naggs], counts i64[g]) as numpy arrays (downloaded)."""
+    naggs = len(agg_cols)
+    first = self.alloc(max(4 * max_groups, 4))
+    acc = self.alloc(max(16 * max_groups * max(naggs, 1), 16))
+    counts = self.alloc(max(8 * max_groups, 8))
+    karr = (BgColumn * len(key_cols))(*key_cols)
+    aarr = (BgColumn * max(naggs, 1))(*(agg_cols or [BgColumn()]))
+    oarr = (ctypes.c_int32 * max(naggs, 1))(*(agg_ops or [0]))
+    ng = ctypes.c_int64()
+    _check(self.L.bg_hashagg(karr, len(key_cols), aarr, oarr, naggs,
+                             mask.ptr if mask else None, ctypes.c_int64(n),
+                             ctypes.c_int64(max_groups), first.ptr, acc.ptr,
+                             counts.ptr, ctypes.byref(ng)), "bg_hashagg")
+    g = ng.value
+    return (first.download(np.uint32, g),
+            acc.download(np.uint8, 16 * g * naggs).reshape(g, naggs, 16)
+            if naggs else np.zeros((g, 0, 16), dtype=np.uint8),
+            counts.download(np.int64, g))
+
+
+GpuStageContext.hashagg = _ctx_hashagg

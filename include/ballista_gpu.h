@@ -176,6 +176,25 @@ int bg_hashjoin_probe_fill(void* handle, const bg_column* probe_keys,
                            uint32_t* d_out_build);
 int bg_hashjoin_free(void* handle);
 
+/* ---- general hash group-by (AggregateExec Partial/Single) ----
+ * Arbitrary group cardinality (q3-class: millions of groups).  Aggregate
+ * ops: 0 = SUM over Decimal128 (exact i128), 1 = SUM over Int64 (exact,
+ * accumulated in i128).  COUNT(*) is always produced.  d_mask (optional,
+ * Arrow LSB bitmask words) pre-filters rows (fused Filter+Aggregate).
+ * Outputs are dense groups in ascending hash-slot order (deterministic for
+ * a fixed max_groups): d_first_row[g] = first input row carrying the
+ * group's key (gather key values from it), d_acc_out = i128 LE per
+ * (group, agg), d_counts_out = i64 per group. */
+#define BG_AGG_OP_SUM_DEC128 0
+#define BG_AGG_OP_SUM_I64 1
+int bg_hashagg(const bg_column* key_cols, int32_t nkeys,
+               const bg_column* agg_cols, const int32_t* agg_ops,
+               int32_t naggs, const uint8_t* d_mask, int64_t n,
+               int64_t max_groups, uint32_t* d_first_row,
+               uint8_t* d_acc_out /* max_groups*naggs*16 */,
+               int64_t* d_counts_out /* max_groups */,
+               int64_t* out_ngroups);
+
 /* ---- fused filter+aggregate stages ---- */
 
 /* TPC-H q6 stage 1 (scan+filter+aggregate, approved/q6.txt):

@@ -273,3 +273,22 @@ def hashjoin_pairs(build: np.ndarray, probe: np.ndarray):
                             _p(p, ctypes.c_int64), len(p),
                             _p(op, ctypes.c_uint32), _p(ob, ctypes.c_uint32))
     return op, ob
+
+
+def hashagg(keys, aggs, n, mask=None):
+    """Serial group-by restatement (exact python-int arithmetic).
+
+    keys: list of numpy arrays (any comparable dtype incl. i64 views of
+    dec128 pairs); aggs: list of ("sum", int-array-or-python-int-list);
+    mask: optional boolean numpy array.
+    -> dict key_tuple -> (count, [exact sums])."""
+    out = {}
+    for i in range(n):
+        if mask is not None and not mask[i]:
+            continue
+        kt = tuple(int(k[i]) if hasattr(k[i], "item") else k[i] for k in keys)
+        e = out.setdefault(kt, [0] + [0] * len(aggs))
+        e[0] += 1
+        for a, (_, vals) in enumerate(aggs):
+            e[a + 1] += int(vals[i])
+    return {k: (v[0], v[1:]) for k, v in out.items()}

@@ -375,3 +375,66 @@ def test_hashjoin_golden_alltypes(ctx):
     # every id unique -> m == n; project t0.id, order desc, limit 5
     joined_ids = sorted((int(ids[i]) for i in got_p), reverse=True)[:5]
     assert joined_ids == e["join_ids_desc5"]
+
+
+# ---------------------------------------------------------------------------
+# general hash group-by
+# ---------------------------------------------------------------------------
+@pytest.mark.parametrize("n,ngroups_hint", [(10_000, 50), (500_000, 100_000),
+                                            (64, 64)])
+def test_hashagg_parity(ctx, n, ngroups_hint):
+    """q3-class group-by: composite (i64, date32) keys, SUM(Decimal128) +
+    SUM(Int64) + COUNT — exact vs the oracle as a key->values map."""
+    rng = np.random.default_rng(n)
+    k1 = rng.integers(0, max(ngroups_hint // 2, 2), size=n, dtype=np.int64)
+    k2 = rng.integers(0, 4, size=n, dtype=np.int32)
+    dec = rng.integers(-10**10, 10**10, size=n, dtype=np.int64)
+    dec16 = dec_bytes(dec)
+    v64 = rng.integers(-10**6, 10**6, size=n, dtype=np.int64)
+
+    c1, _ = ctx.upload_column(k1, gpu.BG_DT_INT64)
+    c2, _ = ctx.upload_column(k2, gpu.BG_DT_INT32)
+    ca = ctx.column(gpu.BG_DT_DECIMAL128, ctx.upload(dec16), n)
+    cb, _ = ctx.upload_column(v64, gpu.BG_DT_INT64)
+    first, acc, counts = ctx.hashagg(
+        [c1, c2], [ca, cb],
+        [gpu.BG_AGG_OP_SUM_DEC128, gpu.BG_AGG_OP_SUM_I64], n,
+        max_groups=max(ngroups_hint * 4, 64))
+
+    got = {}
+    for g in range(len(first)):
+        r = int(first[g])
+        key = (int(k1[r]), int(k2[r]))
+        sums = [int.from_bytes(bytes(acc[g, a]), "little", signed=True)
+                for a in range(2)]
+        assert key not in got, "duplicate group emitted"
+        got[key] = (int(counts[g]), sums)
+
+    want = oracle.hashagg([k1, k2], [("sum", dec), ("sum", v64)], n)
+    assert got == want
+
+
+def test_hashagg_with_mask_fused_filter(ctx):
+    """Fused Filter+Aggregate: the mask from bg_eval_predicates feeds
+    bg_hashagg (the q1/q3 stage-1 shape)."""
+    n = 200_000
+    li = tpch_synth.lineitem_numpy(n, seed=31)
+    sd, _ = ctx.upload_column(li["l_shipdate"], gpu.BG_DT_DATE32)
+    mask = ctx.eval_predicates(
+        [sd], [(0, gpu.BG_PRED_LT, 0, tpch_synth.Q1_DATE_LE + 1)], n)
+    rf, _ = ctx.upload_column(li["l_returnflag"], gpu.BG_DT_DICT8)
+    qty16 = dec_bytes(li["l_quantity"])
+    cq = ctx.column(gpu.BG_DT_DECIMAL128, ctx.upload(qty16), n)
+    first, acc, counts = ctx.hashagg([rf], [cq], [gpu.BG_AGG_OP_SUM_DEC128],
+                                     n, max_groups=64, mask=mask)
+    got = {}
+    for g in range(len(first)):
+        r = int(first[g])
+        got[int(li["l_returnflag"][r])] = (
+            int(counts[g]),
+            [int.from_bytes(bytes(acc[g, 0]), "little", signed=True)])
+    np_mask = li["l_shipdate"] <= tpch_synth.Q1_DATE_LE
+    want = oracle.hashagg([li["l_returnflag"]], [("sum", li["l_quantity"])],
+                          n, mask=np_mask)
+    want = {k[0]: v for k, v in want.items()}
+    assert got == want
