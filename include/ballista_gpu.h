@@ -147,6 +147,19 @@ int bg_partition_ids(const uint64_t* d_hashes, int64_t n, uint32_t k,
 int bg_partition_indices(const uint32_t* d_pids, int64_t n, uint32_t k,
                          uint32_t* d_indices, int64_t* d_offsets);
 
+/* As bg_partition_indices, also emitting the inverse permutation
+ * d_rank[row] = output position (coalesced write) — feeds
+ * bg_scatter_rows so payload materialisation reads inputs sequentially. */
+int bg_partition_indices_ex(const uint32_t* d_pids, int64_t n, uint32_t k,
+                            uint32_t* d_indices, int64_t* d_offsets,
+                            uint32_t* d_rank);
+
+/* Scatter-materialise: d_dst[d_rank[i]] = d_src[i] (sequential reads,
+ * partition-major writes — the write-side dual of bg_gather, avoiding the
+ * gather's k-fold read amplification on permuted input). */
+int bg_scatter_rows(const void* d_src, int64_t elem_size,
+                    const uint32_t* d_rank, int64_t n, void* d_dst);
+
 /* Fused convenience: hash + pids + stable split + gather every payload
  * column partition-major.  d_out[c] must hold len*elem_size(c) bytes; rows
  * of partition p for column c live at [offsets[p]*esz, offsets[p+1]*esz).

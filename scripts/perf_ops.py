@@ -99,15 +99,30 @@ def main():
     sec = timeit(ctx, do_gather)
     report("gather_4cols_24B_row", sec, gather_bytes,
            {"note": "4B idx + read+write per col elem; random reads"})
+    # scatter-based materialisation (sequential reads, k write streams)
+    rankbuf = ctx.alloc(4 * n)
+    gpu._check(L.bg_partition_indices_ex(pbuf.ptr, ctypes.c_int64(n), k,
+                                         ibuf.ptr, obuf.ptr, rankbuf.ptr),
+               "split_ex")
+    def do_scatter():
+        for c, o in zip(payload, outs):
+            esz = gpu._DT_SIZE[c.dtype]
+            gpu._check(L.bg_scatter_rows(ctypes.c_void_p(c.d_data),
+                                         ctypes.c_int64(esz), rankbuf.ptr,
+                                         ctypes.c_int64(n), o.ptr), "scat")
+    sec = timeit(ctx, do_scatter)
+    report("scatter_4cols_24B_row", sec, gather_bytes,
+           {"note": "sequential reads via inverse perm, k=16 write streams"})
     whole_bytes = (8 + 8) * n + (8 + 4) * n + (4 * 3) * n + gather_bytes
     sec = timeit(ctx, lambda: (
         gpu._check(L.bg_hash_columns((gpu.BgColumn * 1)(kc), 1,
                                      ctypes.c_int64(n), hbuf.ptr), "h"),
         gpu._check(L.bg_partition_ids(hbuf.ptr, ctypes.c_int64(n), k,
                                       pbuf.ptr), "p"),
-        gpu._check(L.bg_partition_indices(pbuf.ptr, ctypes.c_int64(n), k,
-                                          ibuf.ptr, obuf.ptr), "s"),
-        do_gather()))
+        gpu._check(L.bg_partition_indices_ex(pbuf.ptr, ctypes.c_int64(n), k,
+                                             ibuf.ptr, obuf.ptr,
+                                             rankbuf.ptr), "s"),
+        do_scatter()))
     report("repartition_pipeline_total", sec, whole_bytes,
            {"rows_per_s": n / sec})
 
