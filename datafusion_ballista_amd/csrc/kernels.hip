@@ -71,6 +71,77 @@ static int set_hip_err(hipError_t e, const char* what) {
     if (_e != hipSuccess) return set_hip_err(_e, #call); \
   } while (0)
 
+// ---------------------------------------------------------------------------
+// pooled device allocator — hipMalloc/hipFree cost ~100 ms on GB-scale
+// temporaries (measured: join-probe pair buffers), so every internal
+// allocation (and the public bg_malloc/bg_free) goes through a size-bucketed
+// free list (power-of-two rounding, per-process, mutex-guarded).  Sized for
+// 288 GB HBM: memory stays pooled until bg_pool_trim().
+// ---------------------------------------------------------------------------
+#include <map>
+#include <mutex>
+#include <unordered_map>
+#include <vector>
+
+static std::mutex g_pool_mu;
+static std::unordered_map<uint64_t, std::vector<void*>> g_pool_free;
+static std::unordered_map<void*, uint64_t> g_pool_sizes;
+
+static uint64_t pool_round(uint64_t bytes) {
+  uint64_t p = 256;
+  while (p < bytes) p <<= 1;
+  return p;
+}
+
+static hipError_t pool_malloc(void** out, uint64_t bytes) {
+  const uint64_t sz = pool_round(bytes ? bytes : 1);
+  {
+    std::lock_guard<std::mutex> g(g_pool_mu);
+    auto it = g_pool_free.find(sz);
+    if (it != g_pool_free.end() && !it->second.empty()) {
+      *out = it->second.back();
+      it->second.pop_back();
+      return hipSuccess;
+    }
+  }
+  hipError_t e = hipMalloc(out, sz);
+  if (e != hipSuccess) {  // pressure: drop the cache and retry once
+    std::lock_guard<std::mutex> g(g_pool_mu);
+    for (auto& kv : g_pool_free)
+      for (void* p : kv.second) {
+        g_pool_sizes.erase(p);
+        (void)hipFree(p);
+      }
+    g_pool_free.clear();
+    e = hipMalloc(out, sz);
+  }
+  if (e == hipSuccess) {
+    std::lock_guard<std::mutex> g(g_pool_mu);
+    g_pool_sizes[*out] = sz;
+  }
+  return e;
+}
+
+static hipError_t pool_release(void* p) {
+  if (!p) return hipSuccess;
+  std::lock_guard<std::mutex> g(g_pool_mu);
+  auto it = g_pool_sizes.find(p);
+  if (it == g_pool_sizes.end()) return hipFree(p);  // not pooled
+  g_pool_free[it->second].push_back(p);
+  return hipSuccess;
+}
+
+extern "C" int bg_pool_trim(void) {
+  std::lock_guard<std::mutex> g(g_pool_mu);
+  for (auto& kv : g_pool_free)
+    for (void* p : kv.second) {
+      g_pool_sizes.erase(p);
+      (void)hipFree(p);
+    }
+  g_pool_free.clear();
+  return BG_OK;
+}
+
 #define REQUIRE_INIT()                                                       \
   do {                                                                       \
     if (!g_inited)                                                           \
@@ -110,12 +181,12 @@ extern "C" int bg_synchronize(void) {
 
 extern "C" int bg_malloc(uint64_t bytes, void** d_ptr) {
   REQUIRE_INIT();
-  HIP_TRY(hipMalloc(d_ptr, bytes));
+  HIP_TRY(pool_malloc((void**)d_ptr, bytes));
   return BG_OK;
 }
 extern "C" int bg_free(void* d_ptr) {
   REQUIRE_INIT();
-  HIP_TRY(hipFree(d_ptr));
+  HIP_TRY(pool_release(d_ptr));
   return BG_OK;
 }
 extern "C" int bg_memset(void* d_ptr, int value, uint64_t bytes) {
@@ -370,9 +441,9 @@ static int scan_exclusive_i64(const u64* d_in, int64_t m, i64* d_out,
   u64* d_segsums;
   i64* d_segbases;
   hipError_t e;
-  e = hipMalloc(&d_segsums, sizeof(u64) * nseg);
+  e = pool_malloc((void**)&d_segsums, sizeof(u64) * nseg);
   if (e != hipSuccess) return set_hip_err(e, "scan malloc");
-  e = hipMalloc(&d_segbases, sizeof(i64) * nseg);
+  e = pool_malloc((void**)&d_segbases, sizeof(i64) * nseg);
   if (e != hipSuccess) return set_hip_err(e, "scan malloc");
   int blocks = (int)bg_imin64(nseg, BG_MAX_BLOCKS);
   hipLaunchKernelGGL(k_segment_sums_i64, dim3(blocks), dim3(BG_BLOCK), 0, 0,
@@ -383,8 +454,8 @@ static int scan_exclusive_i64(const u64* d_in, int64_t m, i64* d_out,
                      dim3(BG_BLOCK), 0, 0, d_in, m, d_out, nullptr, seg,
                      d_segbases);
   e = hipGetLastError();
-  (void)hipFree(d_segsums);
-  (void)hipFree(d_segbases);
+  (void)pool_release(d_segsums);
+  (void)pool_release(d_segbases);
   if (e != hipSuccess) return set_hip_err(e, "scan large");
   return BG_OK;
 }
@@ -447,9 +518,9 @@ extern "C" int bg_mask_to_indices(const uint8_t* d_mask, int64_t n,
   u64* d_counts;
   i64* d_offs;
   i64* d_total;
-  HIP_TRY(hipMalloc(&d_counts, sizeof(u64) * (nchunks ? nchunks : 1)));
-  HIP_TRY(hipMalloc(&d_offs, sizeof(i64) * (nchunks ? nchunks : 1)));
-  HIP_TRY(hipMalloc(&d_total, sizeof(i64)));
+  HIP_TRY(pool_malloc((void**)&d_counts, sizeof(u64) * (nchunks ? nchunks : 1)));
+  HIP_TRY(pool_malloc((void**)&d_offs, sizeof(i64) * (nchunks ? nchunks : 1)));
+  HIP_TRY(pool_malloc((void**)&d_total, sizeof(i64)));
   int blocks = (int)bg_imin64(nchunks, BG_MAX_BLOCKS);
   if (blocks == 0) blocks = 1;
   hipLaunchKernelGGL(k_mask_count, dim3(blocks), dim3(BG_BLOCK), 0, 0,
@@ -464,9 +535,9 @@ extern "C" int bg_mask_to_indices(const uint8_t* d_mask, int64_t n,
                      nchunks, d_indices);
   i64 total = 0;
   HIP_TRY(hipMemcpy(&total, d_total, sizeof(i64), hipMemcpyDeviceToHost));
-  HIP_TRY(hipFree(d_counts));
-  HIP_TRY(hipFree(d_offs));
-  HIP_TRY(hipFree(d_total));
+  HIP_TRY(pool_release(d_counts));
+  HIP_TRY(pool_release(d_offs));
+  HIP_TRY(pool_release(d_total));
   *out_count = total;
   return BG_OK;
 }
@@ -706,8 +777,8 @@ extern "C" int bg_partition_indices_ex(const uint32_t* d_pids, int64_t n,
   u64* d_hist;
   i64* d_start;
   const int64_t hist_len = (int64_t)k * (nchunks ? nchunks : 1);
-  HIP_TRY(hipMalloc(&d_hist, sizeof(u64) * hist_len));
-  HIP_TRY(hipMalloc(&d_start, sizeof(i64) * hist_len));
+  HIP_TRY(pool_malloc((void**)&d_hist, sizeof(u64) * hist_len));
+  HIP_TRY(pool_malloc((void**)&d_start, sizeof(i64) * hist_len));
   int blocks = (int)bg_imin64(nchunks, BG_MAX_BLOCKS);
   if (blocks == 0) blocks = 1;
   hipLaunchKernelGGL(k_part_hist, dim3(blocks), dim3(BG_BLOCK), lds_hist, 0,
@@ -722,8 +793,8 @@ extern "C" int bg_partition_indices_ex(const uint32_t* d_pids, int64_t n,
                      d_pids, n, k, d_start, nchunks, d_indices, d_rank);
   HIP_TRY(hipGetLastError());
   HIP_TRY(hipDeviceSynchronize());
-  HIP_TRY(hipFree(d_hist));
-  HIP_TRY(hipFree(d_start));
+  HIP_TRY(pool_release(d_hist));
+  HIP_TRY(pool_release(d_start));
   return BG_OK;
 }
 
@@ -794,9 +865,9 @@ extern "C" int bg_hash_repartition(const bg_column* key_cols, int32_t nkeys,
   uint64_t* d_hashes;
   uint32_t* d_pids;
   uint32_t* d_rank;
-  HIP_TRY(hipMalloc(&d_hashes, sizeof(u64) * (n ? n : 1)));
-  HIP_TRY(hipMalloc(&d_pids, sizeof(uint32_t) * (n ? n : 1)));
-  HIP_TRY(hipMalloc(&d_rank, sizeof(uint32_t) * (n ? n : 1)));
+  HIP_TRY(pool_malloc((void**)&d_hashes, sizeof(u64) * (n ? n : 1)));
+  HIP_TRY(pool_malloc((void**)&d_pids, sizeof(uint32_t) * (n ? n : 1)));
+  HIP_TRY(pool_malloc((void**)&d_rank, sizeof(uint32_t) * (n ? n : 1)));
   int rc = bg_hash_columns(key_cols, nkeys, n, d_hashes);
   if (rc == BG_OK) rc = bg_partition_ids(d_hashes, n, k, d_pids);
   if (rc == BG_OK)
@@ -809,9 +880,9 @@ extern "C" int bg_hash_repartition(const bg_column* key_cols, int32_t nkeys,
       rc = bg_scatter_rows(payload_cols[c].d_data, esz, d_rank, n, d_out[c]);
     }
   }
-  hipError_t e1 = hipFree(d_hashes);
-  hipError_t e2 = hipFree(d_pids);
-  hipError_t e3 = hipFree(d_rank);
+  hipError_t e1 = pool_release(d_hashes);
+  hipError_t e2 = pool_release(d_pids);
+  hipError_t e3 = pool_release(d_rank);
   if (rc != BG_OK) return rc;
   if (e1 != hipSuccess) return set_hip_err(e1, "hipFree");
   if (e2 != hipSuccess) return set_hip_err(e2, "hipFree");
@@ -882,7 +953,7 @@ extern "C" int bg_q6_agg(const bg_column* shipdate, const bg_column* discount,
     return set_err(BG_ERR_INVALID, "shipdate must be DATE32");
   const int64_t n = shipdate->len;
   u64* d_acc;  // [lo, hi, count]
-  HIP_TRY(hipMalloc(&d_acc, 3 * sizeof(u64)));
+  HIP_TRY(pool_malloc((void**)&d_acc, 3 * sizeof(u64)));
   HIP_TRY(hipMemset(d_acc, 0, 3 * sizeof(u64)));
   int blocks = (int)bg_imin64((n + BG_BLOCK - 1) / BG_BLOCK, BG_MAX_BLOCKS);
   if (blocks == 0) blocks = 1;
@@ -906,7 +977,7 @@ extern "C" int bg_q6_agg(const bg_column* shipdate, const bg_column* discount,
   HIP_TRY(hipEventDestroy(ev1));
   u64 h[3];
   HIP_TRY(hipMemcpy(h, d_acc, 3 * sizeof(u64), hipMemcpyDeviceToHost));
-  HIP_TRY(hipFree(d_acc));
+  HIP_TRY(pool_release(d_acc));
   *out_sum_lo = h[0];
   *out_sum_hi = (int64_t)h[1];
   *out_count = (int64_t)h[2];
@@ -981,8 +1052,8 @@ extern "C" int bg_q1_agg(const bg_column* rf, const bg_column* ls,
   const int64_t n = shipdate->len;
   u64* d_sums;
   u64* d_counts;
-  HIP_TRY(hipMalloc(&d_sums, Q1_GROUPS * Q1_ACCS * 2 * sizeof(u64)));
-  HIP_TRY(hipMalloc(&d_counts, Q1_GROUPS * sizeof(u64)));
+  HIP_TRY(pool_malloc((void**)&d_sums, Q1_GROUPS * Q1_ACCS * 2 * sizeof(u64)));
+  HIP_TRY(pool_malloc((void**)&d_counts, Q1_GROUPS * sizeof(u64)));
   HIP_TRY(hipMemset(d_sums, 0, Q1_GROUPS * Q1_ACCS * 2 * sizeof(u64)));
   HIP_TRY(hipMemset(d_counts, 0, Q1_GROUPS * sizeof(u64)));
   int blocks = (int)bg_imin64((n + BG_BLOCK - 1) / BG_BLOCK, BG_MAX_BLOCKS);
@@ -1011,8 +1082,8 @@ extern "C" int bg_q1_agg(const bg_column* rf, const bg_column* ls,
   u64 counts[Q1_GROUPS];
   HIP_TRY(hipMemcpy(sums, d_sums, sizeof(sums), hipMemcpyDeviceToHost));
   HIP_TRY(hipMemcpy(counts, d_counts, sizeof(counts), hipMemcpyDeviceToHost));
-  HIP_TRY(hipFree(d_sums));
-  HIP_TRY(hipFree(d_counts));
+  HIP_TRY(pool_release(d_sums));
+  HIP_TRY(pool_release(d_counts));
   for (int gidx = 0; gidx < Q1_GROUPS; ++gidx) {
     h_counts[gidx] = (int64_t)counts[gidx];
     for (int a = 0; a < Q1_ACCS; ++a) {
@@ -1104,9 +1175,9 @@ extern "C" int bg_hashjoin_build(const bg_column* build_keys, int64_t n,
   t.n_build = n;
   const u64 nb = next_pow2_u64((u64)(n > 4 ? n * 2 : 8));
   t.mask = nb - 1;
-  HIP_TRY(hipMalloc(&t.keys, sizeof(u64) * (n ? n : 1)));
-  HIP_TRY(hipMalloc(&t.head, sizeof(int) * nb));
-  HIP_TRY(hipMalloc(&t.next, sizeof(int) * (n ? n : 1)));
+  HIP_TRY(pool_malloc((void**)&t.keys, sizeof(u64) * (n ? n : 1)));
+  HIP_TRY(pool_malloc((void**)&t.head, sizeof(int) * nb));
+  HIP_TRY(pool_malloc((void**)&t.next, sizeof(int) * (n ? n : 1)));
   HIP_TRY(hipMemset(t.head, 0xff, sizeof(int) * nb));  // -1
   int blocks = (int)bg_imin64((n + BG_BLOCK - 1) / BG_BLOCK, BG_MAX_BLOCKS);
   if (blocks == 0) blocks = 1;
@@ -1129,9 +1200,9 @@ extern "C" int bg_hashjoin_probe_count(void* handle,
   u64* d_counts;
   i64* d_offs;
   i64* d_total;
-  HIP_TRY(hipMalloc(&d_counts, sizeof(u64) * (n ? n : 1)));
-  HIP_TRY(hipMalloc(&d_offs, sizeof(i64) * (n ? n : 1)));
-  HIP_TRY(hipMalloc(&d_total, sizeof(i64)));
+  HIP_TRY(pool_malloc((void**)&d_counts, sizeof(u64) * (n ? n : 1)));
+  HIP_TRY(pool_malloc((void**)&d_offs, sizeof(i64) * (n ? n : 1)));
+  HIP_TRY(pool_malloc((void**)&d_total, sizeof(i64)));
   int blocks = (int)bg_imin64((n + BG_BLOCK - 1) / BG_BLOCK, BG_MAX_BLOCKS);
   if (blocks == 0) blocks = 1;
   hipLaunchKernelGGL(k_join_count, dim3(blocks), dim3(BG_BLOCK), 0, 0,
@@ -1143,12 +1214,12 @@ extern "C" int bg_hashjoin_probe_count(void* handle,
   }
   i64 total = 0;
   HIP_TRY(hipMemcpy(&total, d_total, sizeof(i64), hipMemcpyDeviceToHost));
-  HIP_TRY(hipFree(d_counts));
+  HIP_TRY(pool_release(d_counts));
   // stash offsets on the handle for the fill call
-  if (t->probe_offsets) (void)hipFree(t->probe_offsets);
+  if (t->probe_offsets) (void)pool_release(t->probe_offsets);
   t->probe_offsets = d_offs;
   t->probe_n = n;
-  HIP_TRY(hipFree(d_total));
+  HIP_TRY(pool_release(d_total));
   *out_matches = total;
   return BG_OK;
 }
@@ -1175,10 +1246,10 @@ extern "C" int bg_hashjoin_free(void* handle) {
   REQUIRE_INIT();
   BgJoinTable* t = (BgJoinTable*)handle;
   if (!t) return BG_OK;
-  (void)hipFree(t->keys);
-  (void)hipFree(t->head);
-  (void)hipFree(t->next);
-  if (t->probe_offsets) (void)hipFree(t->probe_offsets);
+  (void)pool_release(t->keys);
+  (void)pool_release(t->head);
+  (void)pool_release(t->next);
+  if (t->probe_offsets) (void)pool_release(t->probe_offsets);
   delete t;
   return BG_OK;
 }
@@ -1384,10 +1455,10 @@ extern "C" int bg_hashagg(const bg_column* key_cols, int32_t nkeys,
   u64* acc;
   u64* counts;
   int* err_flag;
-  HIP_TRY(hipMalloc(&slot_row, sizeof(int) * cap));
-  HIP_TRY(hipMalloc(&acc, sizeof(u64) * cap * (naggs ? naggs : 1) * 2));
-  HIP_TRY(hipMalloc(&counts, sizeof(u64) * cap));
-  HIP_TRY(hipMalloc(&err_flag, sizeof(int)));
+  HIP_TRY(pool_malloc((void**)&slot_row, sizeof(int) * cap));
+  HIP_TRY(pool_malloc((void**)&acc, sizeof(u64) * cap * (naggs ? naggs : 1) * 2));
+  HIP_TRY(pool_malloc((void**)&counts, sizeof(u64) * cap));
+  HIP_TRY(pool_malloc((void**)&err_flag, sizeof(int)));
   HIP_TRY(hipMemset(slot_row, 0xff, sizeof(int) * cap));
   HIP_TRY(hipMemset(acc, 0, sizeof(u64) * cap * (naggs ? naggs : 1) * 2));
   HIP_TRY(hipMemset(counts, 0, sizeof(u64) * cap));
@@ -1414,8 +1485,8 @@ extern "C" int bg_hashagg(const bg_column* key_cols, int32_t nkeys,
   int err = 0;
   HIP_TRY(hipMemcpy(&err, err_flag, sizeof(int), hipMemcpyDeviceToHost));
   if (err) {
-    (void)hipFree(slot_row); (void)hipFree(acc); (void)hipFree(counts);
-    (void)hipFree(err_flag);
+    (void)pool_release(slot_row); (void)pool_release(acc); (void)pool_release(counts);
+    (void)pool_release(err_flag);
     return set_err(BG_ERR_INVALID, "bg_hashagg: table full (raise max_groups)");
   }
 
@@ -1424,8 +1495,8 @@ extern "C" int bg_hashagg(const bg_column* key_cols, int32_t nkeys,
   const int64_t nwords = ((int64_t)cap + 63) / 64;
   u64* occ_mask;
   uint32_t* sidx;
-  HIP_TRY(hipMalloc(&occ_mask, sizeof(u64) * nwords));
-  HIP_TRY(hipMalloc(&sidx, sizeof(uint32_t) * cap));
+  HIP_TRY(pool_malloc((void**)&occ_mask, sizeof(u64) * nwords));
+  HIP_TRY(pool_malloc((void**)&sidx, sizeof(uint32_t) * cap));
   int oblocks = (int)bg_imin64((nwords * BG_WAVE + BG_BLOCK - 1) / BG_BLOCK,
                                BG_MAX_BLOCKS);
   if (oblocks == 0) oblocks = 1;
@@ -1445,12 +1516,12 @@ extern "C" int bg_hashagg(const bg_column* key_cols, int32_t nkeys,
                      reinterpret_cast<u64*>(d_acc_out));
   HIP_TRY(hipGetLastError());
   HIP_TRY(hipDeviceSynchronize());
-  HIP_TRY(hipFree(slot_row));
-  HIP_TRY(hipFree(acc));
-  HIP_TRY(hipFree(counts));
-  HIP_TRY(hipFree(err_flag));
-  HIP_TRY(hipFree(occ_mask));
-  HIP_TRY(hipFree(sidx));
+  HIP_TRY(pool_release(slot_row));
+  HIP_TRY(pool_release(acc));
+  HIP_TRY(pool_release(counts));
+  HIP_TRY(pool_release(err_flag));
+  HIP_TRY(pool_release(occ_mask));
+  HIP_TRY(pool_release(sidx));
   *out_ngroups = ngroups;
   return BG_OK;
 }
