@@ -730,9 +730,7 @@ __global__ void k_part_scatter(const uint32_t* pids, int64_t n, uint32_t k,
           const i64 base = my[p];
           if (active && pid == p)
             my_pos = base + __popcll(m & ((1ull << lane_id()) - 1));
-          __builtin_amdgcn_wave_barrier();
           if (lane_id() == (__ffsll((i64)m) - 1)) my[p] = base + __popcll(m);
-          __builtin_amdgcn_wave_barrier();
         }
       } else {
         // leader loop: iterations = distinct partitions present (<= 64)
