@@ -1,0 +1,51 @@
+"""Multi-GPU shuffle exchange — RCCL all-to-all over xGMI.
+
+Replaces the reference's byte movement between co-located executors
+(ShuffleReaderExec remote fetch over Flight/raw-block transport,
+shuffle_reader.rs:704-716, 1049-1118) with `torch.distributed`
+all_to_all_single on the nccl backend (= RCCL on ROCm), one rank per
+GPU-backed executor.  The file/Flight path stays the cross-node/parity
+fallback (RCCL replaces only the byte movement between
+co-located executors).
+
+Partitioning contract: the writer hash-partitions into K = k_local x world
+global partitions (`hash % K`, writer.rs:1274-1276); partition p is owned
+by rank p // k_local — so the all-to-all send split for rank r is the byte
+range [offsets[r*k_local], offsets[(r+1)*k_local]) of the partition-major
+buffer that bg_hash_repartition materialises.  Backend-agnostic (gloo for
+the CPU tests, nccl/RCCL on the 8-GPU node).
+"""
+
+import torch
+import torch.distributed as dist
+
+
+def owner_of(partition: int, k_local: int) -> int:
+    return partition // k_local
+
+
+def send_splits(offsets, world: int, elem_size: int = 1):
+    """offsets: (K+1,) int64 row offsets of the partition-major buffer
+    (K = world * k_local).  Returns per-rank element counts (len `world`)."""
+    k = len(offsets) - 1
+    assert k % world == 0
+    k_local = k // world
+    return [int(offsets[(r + 1) * k_local] - offsets[r * k_local]) * elem_size
+            for r in range(world)]
+
+
+def all_to_all_rows(t: torch.Tensor, offsets, world: int):
+    """Exchange a partition-major row buffer: rank r keeps the rows of its
+    own k_local partitions from every peer.  `t` is 1-D with one element
+    per row unit (any dtype); returns (received tensor, recv_splits)."""
+    in_splits = send_splits(offsets, world)
+    counts = torch.tensor(in_splits, dtype=torch.int64, device=t.device)
+    all_counts = torch.empty(world * world, dtype=torch.int64,
+                             device=t.device)
+    dist.all_gather_into_tensor(all_counts, counts)
+    # all_counts[s * world + r] = what rank s sends to rank r
+    me = dist.get_rank()
+    out_splits = [int(all_counts[s * world + me]) for s in range(world)]
+    out = torch.empty(sum(out_splits), dtype=t.dtype, device=t.device)
+    dist.all_to_all_single(out, t.contiguous(), out_splits, in_splits)
+    return out, out_splits

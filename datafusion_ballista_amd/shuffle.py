@@ -152,3 +152,71 @@ def read_schema(data_path):
         raw = f.read(1 << 16)
     rd = pa.ipc.open_stream(pa.BufferReader(raw))
     return rd.schema
+
+
+# ---------------------------------------------------------------------------
+# Passthrough shuffle writer (ShuffleWriterExec, no repartition):
+# K child partitions -> K files `{work_dir}/{job}/{stage}/{global_partition}/
+# data-{task_id}.arrow`, each one complete LZ4_FRAME IPC stream
+# (shuffle_writer.rs:528-622; path layout core/src/execution_plans/
+# mod.rs:95-127), with the local->global partition-id mapping of
+# walk_child_partition_mapping (shuffle_writer.rs:76-165).
+# ---------------------------------------------------------------------------
+
+class GlobalPartitionMap:
+    """Local output index -> global partition id (shuffle_writer.rs:76-108):
+    COLLAPSED: every local -> 0 (SortPreservingMerge fan-in)
+    KSPACE: local == global (fresh K-space: hash/round-robin/range repart)
+    PASSTHROUGH: local i -> slice[i] (scheduler-stamped ids; identity when
+    the slice is empty)."""
+    COLLAPSED = "collapsed"
+    KSPACE = "kspace"
+    PASSTHROUGH = "passthrough"
+
+    def __init__(self, kind, slice_ids=None):
+        self.kind = kind
+        self.slice_ids = slice_ids or []
+
+    def resolve(self, local: int) -> int:
+        if self.kind == self.COLLAPSED:
+            return 0
+        if self.kind == self.KSPACE:
+            return local
+        if local < len(self.slice_ids):
+            return int(self.slice_ids[local])
+        return local
+
+
+def passthrough_partition_path(work_dir, job_id, stage_id, global_partition,
+                               task_id) -> str:
+    return os.path.join(work_dir, str(job_id), str(stage_id),
+                        str(global_partition), f"data-{task_id}.arrow")
+
+
+def write_passthrough_partition(work_dir, job_id, stage_id, global_partition,
+                                task_id, schema, batches):
+    """One local output partition -> one LZ4_FRAME IPC file.  Returns
+    (path, num_batches, num_rows, num_bytes) — the ShuffleWritePartition
+    fields (proto :779-791)."""
+    path = passthrough_partition_path(work_dir, job_id, stage_id,
+                                      global_partition, task_id)
+    os.makedirs(os.path.dirname(path), exist_ok=True)
+    num_batches = num_rows = num_bytes = 0
+    with open(path, "wb") as f:
+        with pa.ipc.new_stream(f, schema, options=ipc_write_options()) as w:
+            for b in batches:
+                if b.num_rows == 0:
+                    continue
+                w.write_batch(b)
+                num_batches += 1
+                num_rows += b.num_rows
+                num_bytes += b.get_total_buffer_size()
+    return path, num_batches, num_rows, num_bytes
+
+
+def read_passthrough_partition(path):
+    """ShuffleReaderExec local read of a passthrough file
+    (shuffle_reader.rs:1120-1168, plain single-stream case)."""
+    with open(path, "rb") as f:
+        raw = f.read()
+    return list(pa.ipc.open_stream(pa.BufferReader(raw)))

@@ -152,3 +152,31 @@ def test_lz4_frame_batches(tmp_work):
     raw_sink_len = sum(
         b.get_total_buffer_size() for b in shuffle.rechunk(table, 8192))
     assert len(enc) < raw_sink_len + 1000
+
+
+def test_passthrough_writer_roundtrip(tmp_work):
+    """Passthrough ShuffleWriterExec format: one LZ4 IPC file per global
+    partition at {work}/{job}/{stage}/{global}/data-{task}.arrow
+    (shuffle_writer.rs:528-622)."""
+    table = make_table(3000, seed=9)
+    batches = shuffle.rechunk(table, 512)
+    path, nb, nr, nbytes = shuffle.write_passthrough_partition(
+        tmp_work, "job-p", 4, 7, 3, table.schema, batches)
+    assert path == os.path.join(tmp_work, "job-p", "4", "7", "data-3.arrow")
+    assert nr == 3000 and nb == len(batches)
+    got = pa.Table.from_batches(shuffle.read_passthrough_partition(path),
+                                schema=table.schema)
+    assert got.equals(table)
+
+
+def test_global_partition_map():
+    """Local->global mapping rules (shuffle_writer.rs:76-108):
+    Collapsed -> 0; KSpace -> identity; PassThrough -> slice (identity
+    fallback past the slice)."""
+    m = shuffle.GlobalPartitionMap(shuffle.GlobalPartitionMap.COLLAPSED)
+    assert [m.resolve(i) for i in range(3)] == [0, 0, 0]
+    m = shuffle.GlobalPartitionMap(shuffle.GlobalPartitionMap.KSPACE)
+    assert [m.resolve(i) for i in range(3)] == [0, 1, 2]
+    m = shuffle.GlobalPartitionMap(shuffle.GlobalPartitionMap.PASSTHROUGH,
+                                   [5, 9])
+    assert [m.resolve(i) for i in range(3)] == [5, 9, 2]
