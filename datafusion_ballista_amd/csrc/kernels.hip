@@ -1523,3 +1523,49 @@ extern "C" int bg_hashagg(const bg_column* key_cols, int32_t nkeys,
   *out_ngroups = ngroups;
   return BG_OK;
 }
+
+// ---------------------------------------------------------------------------
+// ProjectionExec expression subset: Decimal128 binary arithmetic
+// (DataFusion 55 expression eval on the hot path — SURVEY.md §1 "expression
+// eval"; exact i128, Arrow decimal scale handling is the planner's concern).
+// ops: 0 a*b, 1 a+b, 2 a-b, 3 lit-a, 4 a*lit, 5 a+lit
+// ---------------------------------------------------------------------------
+__global__ void k_project_dec128(int op, const ulong2* a, const ulong2* b,
+                                 i128 lit, int64_t n, ulong2* out) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    const ulong2 av = a[i];
+    const i128 x = make_i128(av.x, (i64)av.y);
+    i128 r;
+    switch (op) {
+      case 0: { const ulong2 bv = b[i]; r = x * make_i128(bv.x, (i64)bv.y); break; }
+      case 1: { const ulong2 bv = b[i]; r = x + make_i128(bv.x, (i64)bv.y); break; }
+      case 2: { const ulong2 bv = b[i]; r = x - make_i128(bv.x, (i64)bv.y); break; }
+      case 3: r = lit - x; break;
+      case 4: r = x * lit; break;
+      case 5: r = x + lit; break;
+      default: r = 0;
+    }
+    ulong2 o;
+    o.x = (u64)(u128)r;
+    o.y = (u64)((u128)r >> 64);
+    out[i] = o;
+  }
+}
+
+extern "C" int bg_project_dec128(int32_t op, const bg_column* a,
+                                 const bg_column* b, int64_t lit_lo,
+                                 int64_t lit_hi, int64_t n, void* d_out) {
+  REQUIRE_INIT();
+  if (op <= 2 && b == nullptr)
+    return set_err(BG_ERR_INVALID, "binary op needs column b");
+  int blocks = (int)bg_imin64((n + BG_BLOCK - 1) / BG_BLOCK, BG_MAX_BLOCKS);
+  if (blocks == 0) blocks = 1;
+  const i128 lit = ((i128)lit_hi << 64) | (i128)(u128)(u64)lit_lo;
+  hipLaunchKernelGGL(k_project_dec128, dim3(blocks), dim3(BG_BLOCK), 0, 0, op,
+                     (const ulong2*)a->d_data,
+                     b ? (const ulong2*)b->d_data : nullptr, lit, n,
+                     (ulong2*)d_out);
+  HIP_TRY(hipGetLastError());
+  return BG_OK;
+}

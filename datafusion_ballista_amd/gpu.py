@@ -340,3 +340,25 @@ def _ctx_hashagg(self, key_cols, agg_cols, agg_ops, n, max_groups,
 
 
 GpuStageContext.hashagg = _ctx_hashagg
+
+
+BG_PROJ_MUL = 0
+BG_PROJ_ADD = 1
+BG_PROJ_SUB = 2
+BG_PROJ_RSUB_LIT = 3
+BG_PROJ_MUL_LIT = 4
+BG_PROJ_ADD_LIT = 5
+
+
+def _ctx_project_dec128(self, op, a: BgColumn, b: BgColumn, lit, n):
+    out = self.alloc(max(16 * n, 16))
+    lo, hi = _split_i128(lit)
+    _check(self.L.bg_project_dec128(op, ctypes.byref(a),
+                                    ctypes.byref(b) if b else None,
+                                    ctypes.c_int64(lo), ctypes.c_int64(hi),
+                                    ctypes.c_int64(n), out.ptr),
+           "bg_project_dec128")
+    return out
+
+
+GpuStageContext.project_dec128 = _ctx_project_dec128

@@ -210,6 +210,12 @@ int bg_hashagg(const bg_column* key_cols, int32_t nkeys,
                int64_t* d_counts_out /* max_groups */,
                int64_t* out_ngroups);
 
+/* ---- ProjectionExec expression subset: Decimal128 arithmetic ----
+ * ops: 0 a*b, 1 a+b, 2 a-b, 3 lit-a, 4 a*lit, 5 a+lit (exact i128;
+ * lit as (lo,hi) i128 halves).  d_out: n x 16 B. */
+int bg_project_dec128(int32_t op, const bg_column* a, const bg_column* b,
+                      int64_t lit_lo, int64_t lit_hi, int64_t n, void* d_out);
+
 /* ---- fused filter+aggregate stages ---- */
 
 /* TPC-H q6 stage 1 (scan+filter+aggregate, approved/q6.txt):
