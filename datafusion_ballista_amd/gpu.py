@@ -362,3 +362,17 @@ def _ctx_project_dec128(self, op, a: BgColumn, b: BgColumn, lit, n):
 
 
 GpuStageContext.project_dec128 = _ctx_project_dec128
+
+
+def _ctx_sort_rows(self, key_cols, descending, n):
+    """Stable multi-column ORDER BY -> row permutation DeviceBuffer."""
+    perm = self.alloc(max(4 * n, 4))
+    karr = (BgColumn * len(key_cols))(*key_cols)
+    darr = (ctypes.c_int32 * len(key_cols))(*[1 if d else 0
+                                              for d in descending])
+    _check(self.L.bg_sort_rows(karr, darr, len(key_cols),
+                               ctypes.c_int64(n), perm.ptr), "bg_sort_rows")
+    return perm
+
+
+GpuStageContext.sort_rows = _ctx_sort_rows

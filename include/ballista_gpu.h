@@ -216,6 +216,13 @@ int bg_hashagg(const bg_column* key_cols, int32_t nkeys,
 int bg_project_dec128(int32_t op, const bg_column* a, const bg_column* b,
                       int64_t lit_lo, int64_t lit_hi, int64_t n, void* d_out);
 
+/* ---- SortExec (stable multi-column ORDER BY) ----
+ * d_perm (u32[n]) receives the stable row permutation realising ORDER BY
+ * key_cols[0] [DESC], key_cols[1] [DESC], ...; Top-K = first K entries.
+ * Keys: INT64/INT32/DATE32 (round 1). */
+int bg_sort_rows(const bg_column* key_cols, const int32_t* descending,
+                 int32_t nkeys, int64_t n, uint32_t* d_perm);
+
 /* ---- fused filter+aggregate stages ---- */
 
 /* TPC-H q6 stage 1 (scan+filter+aggregate, approved/q6.txt):

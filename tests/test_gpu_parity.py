@@ -438,3 +438,39 @@ def test_hashagg_with_mask_fused_filter(ctx):
                           n, mask=np_mask)
     want = {k[0]: v for k, v in want.items()}
     assert got == want
+
+
+# ---------------------------------------------------------------------------
+# SortExec / Top-K
+# ---------------------------------------------------------------------------
+@pytest.mark.parametrize("n", [1, 64, 1000, 1_000_000])
+def test_sort_rows_i64_parity(ctx, n):
+    """Stable sort permutation == numpy stable argsort (both stable =>
+    unique answer), asc and desc."""
+    rng = np.random.default_rng(n)
+    keys = rng.integers(-2**62, 2**62, size=n, dtype=np.int64)
+    # inject duplicates to exercise stability
+    if n >= 1000:
+        keys[::7] = 42
+    col, _ = ctx.upload_column(keys, gpu.BG_DT_INT64)
+    # order-preserving u64 transform; stable desc = stable argsort of ~u
+    u = keys.view(np.uint64) ^ np.uint64(1 << 63)
+    for desc in (False, True):
+        perm = ctx.sort_rows([col], [desc], n).download(np.uint32, n)
+        want = np.argsort(~u if desc else u, kind="stable")
+        assert np.array_equal(perm.astype(np.int64), want)
+
+
+def test_sort_multi_column_topk(ctx):
+    """q3 final-stage shape: ORDER BY revenue DESC, o_orderdate ASC,
+    LIMIT 10 (SortExec: TopK(fetch=10), approved/q3.txt)."""
+    n = 100_000
+    rng = np.random.default_rng(8)
+    revenue = rng.integers(0, 5000, size=n, dtype=np.int64)  # many ties
+    odate = rng.integers(8000, 11000, size=n, dtype=np.int32)
+    rc, _ = ctx.upload_column(revenue, gpu.BG_DT_INT64)
+    dc, _ = ctx.upload_column(odate, gpu.BG_DT_DATE32)
+    perm = ctx.sort_rows([rc, dc], [True, False], n).download(np.uint32, n)
+    order = np.lexsort((np.arange(n), odate, -revenue))
+    assert np.array_equal(perm[:10].astype(np.int64), order[:10])
+    assert np.array_equal(perm.astype(np.int64), order)
