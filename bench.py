@@ -123,7 +123,7 @@ def main():
     torch.cuda.synchronize()
 
     def col_of(t, dtype):
-        return gpu.BgColumn(dtype, 15, 2, 0, t.data_ptr(), None,
+        return gpu.BgColumn(dtype, 15, 2, 0, t.data_ptr(), None, None,
                             t.shape[0])
 
     sd = col_of(cols["l_shipdate"], gpu.BG_DT_DATE32)

@@ -593,6 +593,7 @@ struct KeyArgs {
   struct {
     const void* data;
     const uint8_t* valid;
+    const int32_t* offsets;  // UTF8 only
     int dtype;
   } k[BG_MAX_KEYS];
 };
@@ -623,6 +624,14 @@ __global__ void k_hash_columns(KeyArgs keys, int64_t n, u64* hashes) {
           // Utf8 hash parity is required; DICT8 here is for synthetic keys)
           hc = bg_hash_u64((u64) reinterpret_cast<const uint8_t*>(keys.k[c].data)[i]);
           break;
+        case BG_DT_UTF8: {
+          const int32_t lo = keys.k[c].offsets[i];
+          const int32_t hi_off = keys.k[c].offsets[i + 1];
+          hc = bg_hash_str(
+              reinterpret_cast<const uint8_t*>(keys.k[c].data) + lo,
+              (uint64_t)(hi_off - lo));
+          break;
+        }
         default:
           hc = 0;
       }
@@ -642,6 +651,7 @@ extern "C" int bg_hash_columns(const bg_column* key_cols, int32_t nkeys,
   for (int i = 0; i < nkeys; ++i) {
     a.k[i].data = key_cols[i].d_data;
     a.k[i].valid = key_cols[i].d_validity;
+    a.k[i].offsets = key_cols[i].d_offsets;
     a.k[i].dtype = key_cols[i].dtype;
   }
   int blocks = (int)bg_imin64((n + BG_BLOCK - 1) / BG_BLOCK, BG_MAX_BLOCKS);
@@ -1301,6 +1311,16 @@ __device__ __forceinline__ bool keys_equal_rows(const KeyArgs& keys,
             reinterpret_cast<const uint8_t*>(keys.k[c].data)[r2])
           return false;
         break;
+      case BG_DT_UTF8: {
+        const int32_t* offs = keys.k[c].offsets;
+        const int32_t l1 = offs[r1], h1 = offs[r1 + 1];
+        const int32_t l2 = offs[r2], h2 = offs[r2 + 1];
+        if (h1 - l1 != h2 - l2) return false;
+        const uint8_t* d = reinterpret_cast<const uint8_t*>(keys.k[c].data);
+        for (int32_t j = 0; j < h1 - l1; ++j)
+          if (d[l1 + j] != d[l2 + j]) return false;
+        break;
+      }
       default:
         return false;
     }
@@ -1445,6 +1465,7 @@ extern "C" int bg_hashagg(const bg_column* key_cols, int32_t nkeys,
   for (int i = 0; i < nkeys; ++i) {
     keys.k[i].data = key_cols[i].d_data;
     keys.k[i].valid = key_cols[i].d_validity;
+    keys.k[i].offsets = key_cols[i].d_offsets;
     keys.k[i].dtype = key_cols[i].dtype;
   }
   AggArgs aggs{};

@@ -17,6 +17,7 @@ BG_DT_INT64 = 2
 BG_DT_DATE32 = 3
 BG_DT_DECIMAL128 = 4
 BG_DT_DICT8 = 5
+BG_DT_UTF8 = 6
 
 BG_PRED_GE_LT = 0
 BG_PRED_BETWEEN = 1
@@ -42,6 +43,7 @@ class BgColumn(ctypes.Structure):
         ("_pad", ctypes.c_int32),
         ("d_data", ctypes.c_void_p),
         ("d_validity", ctypes.c_void_p),
+        ("d_offsets", ctypes.c_void_p),
         ("len", ctypes.c_int64),
     ]
 
@@ -152,9 +154,22 @@ class GpuStageContext:
 
     # ---- columns ----
     def column(self, dtype: int, buf: DeviceBuffer, n: int,
-               validity: DeviceBuffer = None, precision=0, scale=0) -> BgColumn:
+               validity: DeviceBuffer = None, precision=0, scale=0,
+               offsets: DeviceBuffer = None) -> BgColumn:
         return BgColumn(dtype, precision, scale, 0, buf.ptr,
-                        validity.ptr if validity else None, n)
+                        validity.ptr if validity else None,
+                        offsets.ptr if offsets else None, n)
+
+    def upload_utf8_column(self, strings):
+        """list[bytes] -> BG_DT_UTF8 column (Arrow i32 offsets + data)."""
+        data = b"".join(strings)
+        offs = np.zeros(len(strings) + 1, dtype=np.int32)
+        for i, b in enumerate(strings):
+            offs[i + 1] = offs[i] + len(b)
+        dbuf = self.upload(np.frombuffer(data, dtype=np.uint8)
+                           if data else np.zeros(1, dtype=np.uint8))
+        obuf = self.upload(offs)
+        return self.column(BG_DT_UTF8, dbuf, len(strings), offsets=obuf)
 
     def upload_column(self, arr: np.ndarray, dtype: int = None,
                       validity: np.ndarray = None):

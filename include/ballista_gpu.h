@@ -87,6 +87,7 @@ typedef enum {
   BG_DT_DATE32 = 3,      /* Arrow Date32: i32 days since epoch */
   BG_DT_DECIMAL128 = 4,  /* Arrow Decimal128: 16-byte LE two's complement */
   BG_DT_DICT8 = 5,       /* dictionary codes as u8 (small Utf8 dictionaries) */
+  BG_DT_UTF8 = 6,        /* Arrow Utf8: i32 offsets (d_offsets) + byte data */
 } bg_dtype;
 
 typedef struct {
@@ -94,8 +95,9 @@ typedef struct {
   int32_t precision;      /* decimals only */
   int32_t scale;          /* decimals only */
   int32_t _pad;
-  const void* d_data;     /* device pointer, element-contiguous */
+  const void* d_data;     /* device pointer; UTF8: byte data buffer */
   const uint8_t* d_validity; /* device pointer or NULL (all valid); LSB bits */
+  const int32_t* d_offsets;  /* UTF8 only: n+1 Arrow offsets; else NULL */
   int64_t len;
 } bg_column;
 

@@ -54,7 +54,7 @@ def main():
         print(json.dumps(rec), flush=True)
 
     def col_of(t, dtype):
-        return gpu.BgColumn(dtype, 15, 2, 0, t.data_ptr(), None, t.shape[0])
+        return gpu.BgColumn(dtype, 15, 2, 0, t.data_ptr(), None, None, t.shape[0])
 
     # ---- hash repartition (150M rows, k=16) ----
     n = 150_000_000

@@ -31,7 +31,7 @@ def main():
     torch.cuda.synchronize()
 
     def col_of(t, dtype):
-        return gpu.BgColumn(dtype, 15, 2, 0, t.data_ptr(), None, t.shape[0])
+        return gpu.BgColumn(dtype, 15, 2, 0, t.data_ptr(), None, None, t.shape[0])
 
     sd = col_of(cols["l_shipdate"], gpu.BG_DT_DATE32)
     cd = col_of(cols["l_discount"], gpu.BG_DT_DECIMAL128)

@@ -474,3 +474,46 @@ def test_sort_multi_column_topk(ctx):
     order = np.lexsort((np.arange(n), odate, -revenue))
     assert np.array_equal(perm[:10].astype(np.int64), order[:10])
     assert np.array_equal(perm.astype(np.int64), order)
+
+
+def test_hash_utf8_parity(ctx):
+    """Utf8 key hashing (q1's repartition keys are 1-char Utf8) — bit-exact
+    vs the oracle's bg_hash_str restatement, incl. multi-column combine."""
+    strings = [b"A", b"N", b"R", b"", b"hello world", b"x" * 40,
+               b"A", b"NFNF", b"\x00\x01\x02"] * 300
+    n = len(strings)
+    col = ctx.upload_utf8_column(strings)
+    h = ctx.hash_columns([col], n).download(np.uint64, n)
+    data = np.frombuffer(b"".join(strings), dtype=np.uint8)
+    offs = np.zeros(n + 1, dtype=np.int32)
+    for i, b in enumerate(strings):
+        offs[i + 1] = offs[i] + len(b)
+    want = oracle.hash_columns([("utf8", data, offs)], n)
+    assert np.array_equal(h, want)
+    # multi-column: (utf8, i64) combine
+    ids = np.arange(n, dtype=np.int64) % 7
+    icol, _ = ctx.upload_column(ids, gpu.BG_DT_INT64)
+    h2 = ctx.hash_columns([col, icol], n).download(np.uint64, n)
+    want2 = oracle.hash_columns([("utf8", data, offs), ("i64", ids)], n)
+    assert np.array_equal(h2, want2)
+
+
+def test_hashagg_utf8_keys(ctx):
+    """Group-by on a Utf8 key column (q1's shape pre-dictionary)."""
+    strings = ([b"A|F", b"N|F", b"N|O", b"R|F"] * 2500)
+    n = len(strings)
+    col = ctx.upload_utf8_column(strings)
+    vals = np.arange(n, dtype=np.int64)
+    vcol, _ = ctx.upload_column(vals, gpu.BG_DT_INT64)
+    first, acc, counts = ctx.hashagg([col], [vcol], [gpu.BG_AGG_OP_SUM_I64],
+                                     n, max_groups=64)
+    got = {}
+    for g in range(len(first)):
+        key = strings[first[g]]
+        got[key] = (int(counts[g]),
+                    int.from_bytes(bytes(acc[g, 0]), "little", signed=True))
+    want = {}
+    for i, s_ in enumerate(strings):
+        c, t = want.get(s_, (0, 0))
+        want[s_] = (c + 1, t + i)
+    assert got == want
