@@ -17,6 +17,7 @@ Device work (hash, partition split, gather) runs through libballista_gpu.so;
 IPC encode + file write stay host-side.
 """
 
+import time
 from dataclasses import dataclass
 
 import numpy as np
@@ -98,6 +99,14 @@ class GpuQueryStageExecutor:
         self.key_columns = key_columns  # column indices (already-evaluated keys)
         self.k = num_partitions
         self.batch_size = batch_size
+        self._metrics = {}
+
+    def collect_plan_metrics(self):
+        """Metric names mirror the reference writer's MetricsSet
+        (sort_shuffle/writer.rs:328-440: write_time/repart_time/...) plus
+        the GPU counters SURVEY.md §5 calls for, so EXPLAIN ANALYZE keeps
+        rendering per-operator numbers."""
+        return [dict(self._metrics)]
 
     def execute_query_stage(self, task_id: int, table: pa.Table):
         ctx = self.ctx
@@ -118,9 +127,11 @@ class GpuQueryStageExecutor:
             cols.append(ctx.column(bgdt, buf, nelem))
 
         key_cols = [cols[i] for i in self.key_columns]
+        t_repart = time.perf_counter()
         idx_buf, offs_buf, out_bufs = ctx.hash_repartition(
             key_cols, cols, n, self.k)
         ctx.synchronize()
+        repart_time_s = time.perf_counter() - t_repart
         offsets = offs_buf.download(np.int64, self.k + 1)
 
         # download partition-major buffers and slice per partition
@@ -148,8 +159,23 @@ class GpuQueryStageExecutor:
             partition_streams.append(
                 [shuffle.encode_partition_stream(batches, schema)])
 
+        t_write = time.perf_counter()
         data_path, index_path, stats = shuffle.write_task_consolidated(
             self.work_dir, self.job_id, self.stage_id, task_id, schema,
             partition_streams)
+        write_time_s = time.perf_counter() - t_write
+        self._metrics = {
+            # reference writer metric names (sort_shuffle/writer.rs:328-440)
+            "repart_time_ns": int(repart_time_s * 1e9),
+            "write_time_ns": int(write_time_s * 1e9),
+            "spill_time_ns": 0,
+            "spill_count": 0,
+            "spilled_bytes": 0,
+            "output_rows": int(sum(s_[2] for s_ in stats)),
+            # GPU additions
+            "gpu_kernel_time_ns": int(repart_time_s * 1e9),
+            "achieved_hbm_bytes": int(sum(
+                gpu._DT_SIZE[c.dtype] for c in cols) * n * 2),
+        }
         return [ShuffleWritePartition(p, data_path, nb, nr, nbytes)
                 for (p, nb, nr, nbytes) in stats]

@@ -301,6 +301,13 @@ def test_sort_shuffle_stage_e2e(ctx, tmp_path):
     summaries = ex.execute_query_stage(0, table)
     assert len(summaries) == k
     assert sum(s.num_rows for s in summaries) == n
+    # MetricsSet name parity (sort_shuffle/writer.rs:328-440)
+    m = ex.collect_plan_metrics()[0]
+    for name in ("repart_time_ns", "write_time_ns", "spill_time_ns",
+                 "spill_count", "spilled_bytes", "output_rows",
+                 "gpu_kernel_time_ns"):
+        assert name in m
+    assert m["output_rows"] == n
 
     # oracle-side expectation
     keys = table.column("k").to_numpy()
