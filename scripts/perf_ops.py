@@ -178,6 +178,18 @@ def main():
            {"ngroups": ng.value, "rows_per_s": n / sec,
             "kernel_ms": L.bg_last_kernel_ms()})
 
+    # ---- sort: 150M i64 keys (8 radix passes) ----
+    permbuf = ctx.alloc(4 * n)
+    kcarr = (gpu.BgColumn * 1)(kc)
+    darr = (ctypes.c_int32 * 1)(0)
+    def do_sort():
+        gpu._check(L.bg_sort_rows(kcarr, darr, 1, ctypes.c_int64(n),
+                                  permbuf.ptr), "sort")
+    sec = timeit(ctx, do_sort, iters=2)
+    report("sort_rows_i64_150M", sec, n * 8 * 16,
+           {"note": "8 stable radix passes; bytes = 16B/row/pass model",
+            "rows_per_s": n / sec})
+
     # free big tensors before q1
     del keys, dates, prio, cust, bkeys, gk, dec
     torch.cuda.empty_cache()
