@@ -141,6 +141,7 @@ def main():
            {"note": "read keys, write key copy + head CAS + next"})
     pb, bb, m = join2.probe(pcol, n)
     ctx.synchronize()
+    pb.free(); bb.free()  # return pair buffers to the pool (steady state)
     t0 = time.perf_counter()
     pb2, bb2, m2 = join2.probe(pcol, n)
     ctx.synchronize()
