@@ -1603,11 +1603,14 @@ extern "C" int bg_project_dec128(int32_t op, const bg_column* a,
 // first, then each earlier key (classic LSD over columns), and makes the
 // permutation bit-exact against a stable argsort oracle.
 // ---------------------------------------------------------------------------
-__global__ void k_sort_digit(const u64* keyu, const uint32_t* perm, int64_t n,
-                             int shift, uint32_t* pids) {
+__global__ void k_sort_digit_seq(const u64* keyu_cur, int64_t n, int shift,
+                                 uint32_t* pids) {
+  // keys are kept in CURRENT permutation order, so the digit read is a
+  // sequential stream (the gather-by-perm variant random-reads the key
+  // array and pays k-fold line amplification)
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
        i += (int64_t)gridDim.x * blockDim.x)
-    pids[i] = (uint32_t)((keyu[perm[i]] >> shift) & 0xffu);
+    pids[i] = (uint32_t)((keyu_cur[i] >> shift) & 0xffu);
 }
 
 __global__ void k_key_transform_i64(const int64_t* keys, int64_t n,
@@ -1628,35 +1631,76 @@ __global__ void k_key_transform_i32(const int32_t* keys, int64_t n,
   }
 }
 
+// Decimal128 -> order-preserving (hi, lo) u64 pair (sign-flip the high word)
+__global__ void k_key_transform_dec128(const ulong2* keys, int64_t n,
+                                       int descending, u64* key_lo,
+                                       u64* key_hi) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    const ulong2 v = keys[i];
+    u64 lo = v.x;
+    u64 hi = v.y ^ 0x8000000000000000ull;
+    if (descending) { lo = ~lo; hi = ~hi; }
+    key_lo[i] = lo;
+    key_hi[i] = hi;
+  }
+}
+
+// apply the initial permutation (or identity) to a key stream
+__global__ void k_permute_u64(const u64* src, const uint32_t* perm, int64_t n,
+                              u64* dst) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x)
+    dst[i] = src[perm[i]];
+}
+
 __global__ void k_iota_u32(uint32_t* p, int64_t n) {
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
        i += (int64_t)gridDim.x * blockDim.x)
     p[i] = (uint32_t)i;
 }
 
-// Stable sort by one already-transformed u64 key array, refining an
-// existing permutation (identity for a fresh sort).  npasses: 4 for
-// 32-bit-range keys, 8 for full u64.
-static int radix_sort_passes(const u64* d_keyu, int64_t n, uint32_t* d_perm,
-                             int npasses) {
+// Stable radix passes over one or two u64 key words held in CURRENT
+// permutation order: each pass extracts a digit sequentially, stable-splits,
+// then SCATTERS perm and the remaining key words through the inverse
+// permutation (sequential reads, 256 write streams).
+static int radix_sort_passes(u64* d_key_words[2], int nwords, int64_t n,
+                             uint32_t* d_perm, int npasses_per_word) {
   uint32_t* d_pids;
   uint32_t* d_idx;
   int64_t* d_offs;
+  uint32_t* d_rank;
   uint32_t* d_newperm;
+  u64* d_newkey[2] = {nullptr, nullptr};
   HIP_TRY(pool_malloc((void**)&d_pids, sizeof(uint32_t) * (n ? n : 1)));
   HIP_TRY(pool_malloc((void**)&d_idx, sizeof(uint32_t) * (n ? n : 1)));
-  HIP_TRY(pool_malloc((void**)&d_offs, sizeof(i64) * 257));
+  HIP_TRY(pool_malloc((void**)&d_offs, sizeof(int64_t) * 257));
+  HIP_TRY(pool_malloc((void**)&d_rank, sizeof(uint32_t) * (n ? n : 1)));
   HIP_TRY(pool_malloc((void**)&d_newperm, sizeof(uint32_t) * (n ? n : 1)));
+  for (int w = 0; w < nwords; ++w)
+    HIP_TRY(pool_malloc((void**)&d_newkey[w], sizeof(u64) * (n ? n : 1)));
   int blocks = (int)bg_imin64((n + BG_BLOCK - 1) / BG_BLOCK, BG_MAX_BLOCKS);
   if (blocks == 0) blocks = 1;
   int rc = BG_OK;
-  for (int p = 0; p < npasses && rc == BG_OK; ++p) {
-    hipLaunchKernelGGL(k_sort_digit, dim3(blocks), dim3(BG_BLOCK), 0, 0,
-                       d_keyu, d_perm, n, 8 * p, d_pids);
-    rc = bg_partition_indices(d_pids, n, 256, d_idx, d_offs);
-    if (rc == BG_OK)
-      rc = bg_gather(d_perm, 4, d_idx, n, d_newperm);
+  const int total_passes = npasses_per_word * nwords;
+  for (int p = 0; p < total_passes && rc == BG_OK; ++p) {
+    const int word = p / npasses_per_word;   // lo word first (LSD)
+    const int shift = 8 * (p % npasses_per_word);
+    hipLaunchKernelGGL(k_sort_digit_seq, dim3(blocks), dim3(BG_BLOCK), 0, 0,
+                       d_key_words[word], n, shift, d_pids);
+    rc = bg_partition_indices_ex(d_pids, n, 256, d_idx, d_offs, d_rank);
+    if (rc == BG_OK) rc = bg_scatter_rows(d_perm, 4, d_rank, n, d_newperm);
+    for (int w = word; w < nwords && rc == BG_OK; ++w) {
+      rc = bg_scatter_rows(d_key_words[w], 8, d_rank, n, d_newkey[w]);
+      if (rc == BG_OK) {
+        u64* t = d_key_words[w];
+        d_key_words[w] = d_newkey[w];
+        d_newkey[w] = t;
+      }
+    }
     if (rc == BG_OK) {
+      uint32_t* t = d_perm ? nullptr : nullptr;
+      (void)t;
       hipError_t e = hipMemcpyAsync(d_perm, d_newperm, sizeof(uint32_t) * n,
                                     hipMemcpyDeviceToDevice, 0);
       if (e != hipSuccess) rc = set_hip_err(e, "perm copy");
@@ -1665,7 +1709,9 @@ static int radix_sort_passes(const u64* d_keyu, int64_t n, uint32_t* d_perm,
   (void)pool_release(d_pids);
   (void)pool_release(d_idx);
   (void)pool_release(d_offs);
+  (void)pool_release(d_rank);
   (void)pool_release(d_newperm);
+  for (int w = 0; w < nwords; ++w) (void)pool_release(d_newkey[w]);
   return rc;
 }
 
@@ -1681,33 +1727,57 @@ extern "C" int bg_sort_rows(const bg_column* key_cols,
   int blocks = (int)bg_imin64((n + BG_BLOCK - 1) / BG_BLOCK, BG_MAX_BLOCKS);
   if (blocks == 0) blocks = 1;
   hipLaunchKernelGGL(k_iota_u32, dim3(blocks), dim3(BG_BLOCK), 0, 0, d_perm, n);
-  u64* d_keyu;
-  HIP_TRY(pool_malloc((void**)&d_keyu, sizeof(u64) * (n ? n : 1)));
+  u64* d_keyu[2];
+  u64* d_keyperm[2];
+  HIP_TRY(pool_malloc((void**)&d_keyu[0], sizeof(u64) * (n ? n : 1)));
+  HIP_TRY(pool_malloc((void**)&d_keyu[1], sizeof(u64) * (n ? n : 1)));
+  HIP_TRY(pool_malloc((void**)&d_keyperm[0], sizeof(u64) * (n ? n : 1)));
+  HIP_TRY(pool_malloc((void**)&d_keyperm[1], sizeof(u64) * (n ? n : 1)));
   int rc = BG_OK;
   for (int c = nkeys - 1; c >= 0 && rc == BG_OK; --c) {
-    int npasses;
+    int npasses, nwords;
     switch (key_cols[c].dtype) {
       case BG_DT_INT64:
         hipLaunchKernelGGL(k_key_transform_i64, dim3(blocks), dim3(BG_BLOCK),
                            0, 0, (const int64_t*)key_cols[c].d_data, n,
-                           descending[c], d_keyu);
-        npasses = 8;
+                           descending[c], d_keyu[0]);
+        npasses = 8; nwords = 1;
         break;
       case BG_DT_INT32:
       case BG_DT_DATE32:
         hipLaunchKernelGGL(k_key_transform_i32, dim3(blocks), dim3(BG_BLOCK),
                            0, 0, (const int32_t*)key_cols[c].d_data, n,
-                           descending[c], d_keyu);
-        npasses = 4;
+                           descending[c], d_keyu[0]);
+        npasses = 4; nwords = 1;
+        break;
+      case BG_DT_DECIMAL128:
+        hipLaunchKernelGGL(k_key_transform_dec128, dim3(blocks),
+                           dim3(BG_BLOCK), 0, 0,
+                           (const ulong2*)key_cols[c].d_data, n,
+                           descending[c], d_keyu[0], d_keyu[1]);
+        npasses = 8; nwords = 2;
         break;
       default:
         rc = set_err(BG_ERR_UNSUPPORTED,
-                     "sort keys: INT64/INT32/DATE32 (round 1)");
+                     "sort keys: INT64/INT32/DATE32/DECIMAL128 (round 1)");
         continue;
     }
-    rc = radix_sort_passes(d_keyu, n, d_perm, npasses);
+    // bring the key stream into the CURRENT permutation order once
+    u64* words[2] = {d_keyperm[0], d_keyperm[1]};
+    for (int w = 0; w < nwords; ++w)
+      hipLaunchKernelGGL(k_permute_u64, dim3(blocks), dim3(BG_BLOCK), 0, 0,
+                         d_keyu[w], d_perm, n, d_keyperm[w]);
+    rc = radix_sort_passes(words, nwords, n, d_perm, npasses);
+    // radix passes ping-pong the key buffers: re-adopt whatever pointers
+    // ended up live so the release below frees each buffer exactly once
+    d_keyperm[0] = words[0];
+    if (nwords > 1) d_keyperm[1] = words[1];
   }
-  (void)pool_release(d_keyu);
+  (void)pool_release(d_keyu[0]);
+  (void)pool_release(d_keyu[1]);
+  (void)pool_release(d_keyperm[0]);
+  (void)pool_release(d_keyperm[1]);
   if (rc == BG_OK) HIP_TRY(hipGetLastError());
   return rc;
 }
+

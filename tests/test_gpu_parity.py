@@ -524,3 +524,22 @@ def test_hashagg_utf8_keys(ctx):
         c, t = want.get(s_, (0, 0))
         want[s_] = (c + 1, t + i)
     assert got == want
+
+
+def test_sort_dec128_keys(ctx):
+    """Decimal128 ORDER BY (q3's revenue DESC is Decimal(38,4)): 16-pass
+    radix over the (hi, lo) order-preserving pair."""
+    n = 50_000
+    rng = np.random.default_rng(12)
+    lo_part = rng.integers(-10**12, 10**12, size=n, dtype=np.int64)
+    vals = [int(v) * (10**6) + int(w) for v, w in
+            zip(lo_part, rng.integers(0, 10**6, size=n))]
+    dec16 = np.zeros(16 * n, dtype=np.uint8)
+    for i, v in enumerate(vals):
+        dec16[16*i:16*(i+1)] = np.frombuffer(
+            int(v).to_bytes(16, "little", signed=True), dtype=np.uint8)
+    col = ctx.column(gpu.BG_DT_DECIMAL128, ctx.upload(dec16), n)
+    for desc in (False, True):
+        perm = ctx.sort_rows([col], [desc], n).download(np.uint32, n)
+        order = sorted(range(n), key=lambda i: (-vals[i] if desc else vals[i], i))
+        assert perm.astype(np.int64).tolist() == order
