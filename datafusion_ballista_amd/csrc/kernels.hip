@@ -1205,6 +1205,13 @@ extern "C" int bg_hashjoin_probe_count(void* handle,
   BgJoinTable* t = (BgJoinTable*)handle;
   if (probe_keys->dtype != BG_DT_INT64)
     return set_err(BG_ERR_UNSUPPORTED, "join keys must be INT64 (round 1)");
+  // release the previous probe's offsets FIRST so the pool can recycle the
+  // buffer for this call (allocating before releasing forces a fresh
+  // multi-GB hipMalloc inside the hot path)
+  if (t->probe_offsets) {
+    (void)pool_release(t->probe_offsets);
+    t->probe_offsets = nullptr;
+  }
   u64* d_counts;
   i64* d_offs;
   i64* d_total;
@@ -1224,7 +1231,6 @@ extern "C" int bg_hashjoin_probe_count(void* handle,
   HIP_TRY(hipMemcpy(&total, d_total, sizeof(i64), hipMemcpyDeviceToHost));
   HIP_TRY(pool_release(d_counts));
   // stash offsets on the handle for the fill call
-  if (t->probe_offsets) (void)pool_release(t->probe_offsets);
   t->probe_offsets = d_offs;
   t->probe_n = n;
   HIP_TRY(pool_release(d_total));
