@@ -90,17 +90,29 @@ def cpu_baseline_leg(sample_rows: int):
     d16 = tpch_synth.dec128_pairs_np(li["l_discount"]).view(np.uint8).reshape(-1)
     q16 = tpch_synth.dec128_pairs_np(li["l_quantity"]).view(np.uint8).reshape(-1)
     p16 = tpch_synth.dec128_pairs_np(li["l_extendedprice"]).view(np.uint8).reshape(-1)
+    # bounded sample, repeated until >= ~10 s of CPU work (the contract's
+    # 10-30 s window) — the rate is per-row so repetition is neutral
+    def one_pass():
+        oracle.q6(li["l_shipdate"], d16, q16, p16, tpch_synth.Q6_DATE_LO,
+                  tpch_synth.Q6_DATE_HI, tpch_synth.Q6_DISC_LO,
+                  tpch_synth.Q6_DISC_HI, tpch_synth.Q6_QTY_LT)
     t0 = time.perf_counter()
-    oracle.q6(li["l_shipdate"], d16, q16, p16, tpch_synth.Q6_DATE_LO,
-              tpch_synth.Q6_DATE_HI, tpch_synth.Q6_DISC_LO,
-              tpch_synth.Q6_DISC_HI, tpch_synth.Q6_QTY_LT)
+    one_pass()
     dt = time.perf_counter() - t0
+    passes = 1
+    target = 10.0
+    while dt < target and passes < 64:
+        more = min(63, max(1, int((target - dt) / max(dt / passes, 1e-3))))
+        for _ in range(more):
+            one_pass()
+        passes += more
+        dt = time.perf_counter() - t0
     return {
-        "value": sample_rows / dt,
+        "value": sample_rows * passes / dt,
         "unit": "rows/s",
         "cores": 1,
         "kind": "port",
-        "sample": f"{sample_rows} synthetic q6-shaped rows, "
+        "sample": f"{sample_rows} synthetic q6-shaped rows x {passes} passes, "
                   f"{dt:.2f}s single-thread C oracle",
     }
 

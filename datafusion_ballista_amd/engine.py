@@ -109,6 +109,50 @@ class GpuQueryStageExecutor:
         return [dict(self._metrics)]
 
     def execute_query_stage(self, task_id: int, table: pa.Table):
+        """Single- or multi-input task: a list of tables = the task's M
+        local input partitions; each is bucketed on the GPU independently
+        and its encoded streams are concatenated per output partition in
+        the consolidated file, exactly like the reference's per-input
+        tasks + coordinator concat (writer.rs:564-753, 861-884)."""
+        tables = table if isinstance(table, (list, tuple)) else [table]
+        schema = tables[0].schema
+        per_input_streams = []  # [input][partition] -> bytes
+        repart_time_s = 0.0
+        total_rows = 0
+        for t_in in tables:
+            streams, dt = self._repartition_one_input(t_in)
+            per_input_streams.append(streams)
+            repart_time_s += dt
+            total_rows += t_in.num_rows
+
+        # transpose to [partition][input] for the consolidated writer
+        partition_streams = [
+            [per_input_streams[i][p] for i in range(len(tables))]
+            for p in range(self.k)]
+
+        t_write = time.perf_counter()
+        data_path, index_path, stats = shuffle.write_task_consolidated(
+            self.work_dir, self.job_id, self.stage_id, task_id, schema,
+            partition_streams)
+        write_time_s = time.perf_counter() - t_write
+        self._metrics = {
+            # reference writer metric names (sort_shuffle/writer.rs:328-440)
+            "repart_time_ns": int(repart_time_s * 1e9),
+            "write_time_ns": int(write_time_s * 1e9),
+            "spill_time_ns": 0,
+            "spill_count": 0,
+            "spilled_bytes": 0,
+            "output_rows": int(sum(s_[2] for s_ in stats)),
+            # GPU additions (SURVEY.md §5)
+            "gpu_kernel_time_ns": int(repart_time_s * 1e9),
+            "achieved_hbm_bytes": 0,
+        }
+        return [ShuffleWritePartition(p, data_path, nb, nr, nbytes)
+                for (p, nb, nr, nbytes) in stats]
+
+    def _repartition_one_input(self, table: pa.Table):
+        """GPU-bucket one input partition -> (per-output-partition encoded
+        IPC streams, device seconds)."""
         ctx = self.ctx
         table = table.combine_chunks()
         n = table.num_rows
@@ -131,7 +175,7 @@ class GpuQueryStageExecutor:
         idx_buf, offs_buf, out_bufs = ctx.hash_repartition(
             key_cols, cols, n, self.k)
         ctx.synchronize()
-        repart_time_s = time.perf_counter() - t_repart
+        dt_device = time.perf_counter() - t_repart
         offsets = offs_buf.download(np.int64, self.k + 1)
 
         # download partition-major buffers and slice per partition
@@ -140,12 +184,12 @@ class GpuQueryStageExecutor:
             esz = gpu._DT_SIZE[c.dtype]
             col_raws.append(b.download(np.uint8, max(n, 1) * esz))
 
-        partition_streams = []
+        streams = []
         for p in range(self.k):
             lo, hi = int(offsets[p]), int(offsets[p + 1])
             m = hi - lo
             if m == 0:
-                partition_streams.append([b""])
+                streams.append(b"")
                 continue
             arrays = []
             for ci in range(table.num_columns):
@@ -156,26 +200,5 @@ class GpuQueryStageExecutor:
                 arrays.append(_array_from_raw(t, raw, m))
             part_table = pa.Table.from_arrays(arrays, schema=schema)
             batches = shuffle.rechunk(part_table, self.batch_size)
-            partition_streams.append(
-                [shuffle.encode_partition_stream(batches, schema)])
-
-        t_write = time.perf_counter()
-        data_path, index_path, stats = shuffle.write_task_consolidated(
-            self.work_dir, self.job_id, self.stage_id, task_id, schema,
-            partition_streams)
-        write_time_s = time.perf_counter() - t_write
-        self._metrics = {
-            # reference writer metric names (sort_shuffle/writer.rs:328-440)
-            "repart_time_ns": int(repart_time_s * 1e9),
-            "write_time_ns": int(write_time_s * 1e9),
-            "spill_time_ns": 0,
-            "spill_count": 0,
-            "spilled_bytes": 0,
-            "output_rows": int(sum(s_[2] for s_ in stats)),
-            # GPU additions
-            "gpu_kernel_time_ns": int(repart_time_s * 1e9),
-            "achieved_hbm_bytes": int(sum(
-                gpu._DT_SIZE[c.dtype] for c in cols) * n * 2),
-        }
-        return [ShuffleWritePartition(p, data_path, nb, nr, nbytes)
-                for (p, nb, nr, nbytes) in stats]
+            streams.append(shuffle.encode_partition_stream(batches, schema))
+        return streams, dt_device

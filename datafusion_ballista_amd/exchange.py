@@ -49,3 +49,17 @@ def all_to_all_rows(t: torch.Tensor, offsets, world: int):
     out = torch.empty(sum(out_splits), dtype=t.dtype, device=t.device)
     dist.all_to_all_single(out, t.contiguous(), out_splits, in_splits)
     return out, out_splits
+
+
+def broadcast_build_side(t: torch.Tensor, src_rank: int = 0):
+    """Broadcast-join build-side replication: RCCL
+    broadcast instead of N Flight fetches of the broadcast stage's output
+    (try_new_broadcast, shuffle_reader.rs:163-187; planner.rs:154-195
+    lowers CollectLeft joins to a broadcast stage)."""
+    shape = torch.tensor([t.shape[0] if dist.get_rank() == src_rank else 0],
+                         dtype=torch.int64, device=t.device)
+    dist.broadcast(shape, src_rank)
+    if dist.get_rank() != src_rank:
+        t = torch.empty(int(shape.item()), dtype=t.dtype, device=t.device)
+    dist.broadcast(t, src_rank)
+    return t

@@ -130,3 +130,36 @@ def test_all_to_all_exchange_world2():
             want_sum += int(keys.sum())
         assert total_rows == want_rows
         assert total_sum == want_sum
+
+
+def _broadcast_worker(rank, world, port, results):
+    import numpy as np
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    torch.distributed.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        from datafusion_ballista_amd import exchange
+        if rank == 0:
+            build = torch.arange(1000, dtype=torch.int64) * 7
+        else:
+            build = torch.empty(0, dtype=torch.int64)
+        got = exchange.broadcast_build_side(build, src_rank=0)
+        results[rank] = (int(got.shape[0]), int(got.sum().item()))
+    finally:
+        torch.distributed.destroy_process_group()
+
+
+def test_broadcast_build_side_world2():
+    """Broadcast-join build replication (RCCL broadcast stand-in on gloo)."""
+    ctx = mp.get_context("spawn")
+    with ctx.Manager() as mgr:
+        results = mgr.dict()
+        ps = [ctx.Process(target=_broadcast_worker, args=(r, 2, 29641, results))
+              for r in range(2)]
+        for p in ps:
+            p.start()
+        for p in ps:
+            p.join(120)
+            assert p.exitcode == 0
+        want = (1000, sum(i * 7 for i in range(1000)))
+        assert results[0] == want and results[1] == want
