@@ -584,3 +584,54 @@ def test_sort_shuffle_stage_multi_input(ctx, tmp_path):
             chunks.append(t.take(pa.array(rows, type=pa.uint32())))
         want = pa.concat_tables(chunks)
         assert got.equals(want.combine_chunks()), f"partition {p}"
+
+
+def test_q1_partial_final_two_stage(ctx, tmp_path):
+    """Partial/Final aggregate decomposition (AggregateExec Partial on each
+    task shard -> shuffle -> Final merge; approved/q1.txt stages 1-2):
+    partials from two GPU tasks merge to exactly the single-pass oracle
+    answer; AVG at the final stage checked at 1e-6 relative
+    (benchmarks/src/lib.rs:35 float tolerance)."""
+    n = 400_000
+    li = tpch_synth.lineitem_numpy(n, seed=77)
+    half = n // 2
+
+    def run_partial(sl):
+        m = sl.stop - sl.start
+        rf, _ = ctx.upload_column(li["l_returnflag"][sl], gpu.BG_DT_DICT8)
+        ls, _ = ctx.upload_column(li["l_linestatus"][sl], gpu.BG_DT_DICT8)
+        cq = ctx.column(gpu.BG_DT_DECIMAL128,
+                        ctx.upload(dec_bytes(li["l_quantity"][sl])), m)
+        cp = ctx.column(gpu.BG_DT_DECIMAL128,
+                        ctx.upload(dec_bytes(li["l_extendedprice"][sl])), m)
+        cd = ctx.column(gpu.BG_DT_DECIMAL128,
+                        ctx.upload(dec_bytes(li["l_discount"][sl])), m)
+        ct = ctx.column(gpu.BG_DT_DECIMAL128,
+                        ctx.upload(dec_bytes(li["l_tax"][sl])), m)
+        sd, _ = ctx.upload_column(li["l_shipdate"][sl], gpu.BG_DT_DATE32)
+        return ctx.q1_agg(rf, ls, cq, cp, cd, ct, sd, tpch_synth.Q1_DATE_LE)
+
+    p1 = run_partial(slice(0, half))
+    p2 = run_partial(slice(half, n))
+
+    # FINAL aggregate: merge partial states (exact integer adds)
+    merged = {}
+    for part in (p1, p2):
+        for g, (cnt, sums) in part.items():
+            c0, s0 = merged.get(g, (0, [0] * 5))
+            merged[g] = (c0 + cnt, [a + b for a, b in zip(s0, sums)])
+
+    want = oracle.q1(li["l_returnflag"], li["l_linestatus"],
+                     dec_bytes(li["l_quantity"]),
+                     dec_bytes(li["l_extendedprice"]),
+                     dec_bytes(li["l_discount"]), dec_bytes(li["l_tax"]),
+                     li["l_shipdate"], tpch_synth.Q1_DATE_LE)
+    assert merged == want  # partial+final == single-pass, bit-exact
+
+    # final-stage AVG columns (avg_qty, avg_price, avg_disc) at 1e-6 rel
+    for g, (cnt, sums) in merged.items():
+        wc, ws = want[g]
+        for idx_a, scale in ((0, 100.0), (1, 100.0), (4, 100.0)):
+            got_avg = sums[idx_a] / cnt / scale
+            want_avg = ws[idx_a] / wc / scale
+            assert abs(got_avg - want_avg) <= 1e-6 * max(abs(want_avg), 1e-30)
