@@ -725,6 +725,15 @@ __global__ void k_part_scatter(const uint32_t* pids, int64_t n, uint32_t k,
     for (uint32_t p = lane_id(); p < k; p += BG_WAVE)
       my[p] = start[(int64_t)p * nchunks + c];
     __builtin_amdgcn_wave_barrier();
+    // register cursors for the k<=64 fast path (lane p owns partition p)
+    int cur_lo = 0, cur_hi = 0;
+    if (k <= BG_WAVE) {
+      const i64 seed = (lane_id() < (int)k)
+                           ? start[(int64_t)lane_id() * nchunks + c]
+                           : 0;
+      cur_lo = (int)(uint32_t)((u64)seed & 0xffffffff);
+      cur_hi = (int)(uint32_t)((u64)seed >> 32);
+    }
     const int64_t r0 = c * PS_ROWS_PER_WAVE;
     const int64_t r1 = min(r0 + (int64_t)PS_ROWS_PER_WAVE, n);
     for (int64_t rb = r0; rb < r1; rb += BG_WAVE) {
@@ -733,14 +742,25 @@ __global__ void k_part_scatter(const uint32_t* pids, int64_t n, uint32_t k,
       const uint32_t pid = active ? pids[r] : 0xffffffffu;
       i64 my_pos = -1;
       if (k <= BG_WAVE) {
-        // dense partition loop: no cross-lane shuffles, one ballot per p
+        // dense partition loop with REGISTER cursors: lane p's element of
+        // `cur_lane` holds partition p's cursor; readlane/writelane avoid
+        // the ~50-cycle LDS round trip per partition iteration that
+        // serialises the whole wave (cursors were seeded from LDS `my`
+        // before the row loop)
         for (uint32_t p = 0; p < k; ++p) {
           const u64 m = __ballot(active && pid == p);
           if (!m) continue;
-          const i64 base = my[p];
+          const i64 base =
+              ((i64)(uint32_t)__builtin_amdgcn_readlane(cur_hi, p) << 32) |
+              (i64)(uint32_t)__builtin_amdgcn_readlane(cur_lo, p);
           if (active && pid == p)
             my_pos = base + __popcll(m & ((1ull << lane_id()) - 1));
-          if (lane_id() == (__ffsll((i64)m) - 1)) my[p] = base + __popcll(m);
+          const i64 nxt = base + __popcll(m);
+          // per-lane select: lane p adopts the new cursor (nxt is wave-
+          // uniform; one v_cndmask per word, no LDS round trip)
+          const bool mine = (uint32_t)lane_id() == p;
+          cur_lo = mine ? (int)(uint32_t)((u64)nxt & 0xffffffff) : cur_lo;
+          cur_hi = mine ? (int)(uint32_t)((u64)nxt >> 32) : cur_hi;
         }
       } else {
         // leader loop: iterations = distinct partitions present (<= 64)
@@ -1113,9 +1133,8 @@ extern "C" int bg_q1_agg(const bg_column* rf, const bg_column* ls,
 // offsets and no atomic append nondeterminism in sizes.
 // ---------------------------------------------------------------------------
 struct BgJoinTable {
-  u64* keys;      // build keys (copied for probe-time equality check)
+  ulong2* nodes;  // packed chain node: {key, next} — ONE line per hop
   int* head;      // bucket heads (-1 empty)
-  int* next;      // chain links
   int64_t n_build;
   u64 mask;       // nb - 1
   i64* probe_offsets = nullptr;  // per-probe-row output offsets (count phase)
@@ -1123,33 +1142,38 @@ struct BgJoinTable {
 };
 
 __global__ void k_join_build(const int64_t* keys, int64_t n, int* head,
-                             int* next, u64* key_copy, u64 mask) {
+                             ulong2* nodes, u64 mask) {
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
        i += (int64_t)gridDim.x * blockDim.x) {
     const u64 k = (u64)keys[i];
-    key_copy[i] = k;
     const u64 b = bg_hash_u64(k) & mask;
-    next[i] = atomicExch(&head[b], (int)i);
+    const int prev = atomicExch(&head[b], (int)i);
+    ulong2 node;
+    node.x = k;
+    node.y = (u64)(int64_t)prev;  // sign-extended: -1 terminates
+    nodes[i] = node;
   }
 }
 
 __global__ void k_join_count(const int64_t* probe_keys, int64_t n_probe,
-                             const int* head, const int* next,
-                             const u64* build_keys, u64 mask, u64* counts) {
+                             const int* head, const ulong2* nodes, u64 mask,
+                             u64* counts) {
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n_probe;
        i += (int64_t)gridDim.x * blockDim.x) {
     const u64 k = (u64)probe_keys[i];
     const u64 b = bg_hash_u64(k) & mask;
     u64 c = 0;
-    for (int j = head[b]; j >= 0; j = next[j])
-      if (build_keys[j] == k) c++;
+    for (int64_t j = head[b]; j >= 0;) {
+      const ulong2 node = nodes[j];
+      if (node.x == k) c++;
+      j = (int64_t)node.y;
+    }
     counts[i] = c;
   }
 }
 
 __global__ void k_join_fill(const int64_t* probe_keys, int64_t n_probe,
-                            const int* head, const int* next,
-                            const u64* build_keys, u64 mask,
+                            const int* head, const ulong2* nodes, u64 mask,
                             const i64* offsets, uint32_t* out_probe,
                             uint32_t* out_build) {
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n_probe;
@@ -1157,12 +1181,14 @@ __global__ void k_join_fill(const int64_t* probe_keys, int64_t n_probe,
     const u64 k = (u64)probe_keys[i];
     const u64 b = bg_hash_u64(k) & mask;
     i64 w = offsets[i];
-    for (int j = head[b]; j >= 0; j = next[j]) {
-      if (build_keys[j] == k) {
+    for (int64_t j = head[b]; j >= 0;) {
+      const ulong2 node = nodes[j];
+      if (node.x == k) {
         out_probe[w] = (uint32_t)i;
         out_build[w] = (uint32_t)j;
         ++w;
       }
+      j = (int64_t)node.y;
     }
   }
 }
@@ -1183,15 +1209,14 @@ extern "C" int bg_hashjoin_build(const bg_column* build_keys, int64_t n,
   t.n_build = n;
   const u64 nb = next_pow2_u64((u64)(n > 4 ? n * 2 : 8));
   t.mask = nb - 1;
-  HIP_TRY(pool_malloc((void**)&t.keys, sizeof(u64) * (n ? n : 1)));
+  HIP_TRY(pool_malloc((void**)&t.nodes, sizeof(ulong2) * (n ? n : 1)));
   HIP_TRY(pool_malloc((void**)&t.head, sizeof(int) * nb));
-  HIP_TRY(pool_malloc((void**)&t.next, sizeof(int) * (n ? n : 1)));
   HIP_TRY(hipMemset(t.head, 0xff, sizeof(int) * nb));  // -1
   int blocks = (int)bg_imin64((n + BG_BLOCK - 1) / BG_BLOCK, BG_MAX_BLOCKS);
   if (blocks == 0) blocks = 1;
   hipLaunchKernelGGL(k_join_build, dim3(blocks), dim3(BG_BLOCK), 0, 0,
-                     (const int64_t*)build_keys->d_data, n, t.head, t.next,
-                     t.keys, t.mask);
+                     (const int64_t*)build_keys->d_data, n, t.head, t.nodes,
+                     t.mask);
   HIP_TRY(hipGetLastError());
   BgJoinTable* h = new BgJoinTable(t);
   *out_handle = h;
@@ -1221,8 +1246,8 @@ extern "C" int bg_hashjoin_probe_count(void* handle,
   int blocks = (int)bg_imin64((n + BG_BLOCK - 1) / BG_BLOCK, BG_MAX_BLOCKS);
   if (blocks == 0) blocks = 1;
   hipLaunchKernelGGL(k_join_count, dim3(blocks), dim3(BG_BLOCK), 0, 0,
-                     (const int64_t*)probe_keys->d_data, n, t->head, t->next,
-                     t->keys, t->mask, d_counts);
+                     (const int64_t*)probe_keys->d_data, n, t->head, t->nodes,
+                     t->mask, d_counts);
   {
     int rc = scan_exclusive_i64(d_counts, n, d_offs, d_total);
     if (rc != BG_OK) return rc;
@@ -1249,8 +1274,8 @@ extern "C" int bg_hashjoin_probe_fill(void* handle,
   int blocks = (int)bg_imin64((n + BG_BLOCK - 1) / BG_BLOCK, BG_MAX_BLOCKS);
   if (blocks == 0) blocks = 1;
   hipLaunchKernelGGL(k_join_fill, dim3(blocks), dim3(BG_BLOCK), 0, 0,
-                     (const int64_t*)probe_keys->d_data, n, t->head, t->next,
-                     t->keys, t->mask, t->probe_offsets, d_out_probe,
+                     (const int64_t*)probe_keys->d_data, n, t->head, t->nodes,
+                     t->mask, t->probe_offsets, d_out_probe,
                      d_out_build);
   HIP_TRY(hipGetLastError());
   return BG_OK;
@@ -1260,9 +1285,8 @@ extern "C" int bg_hashjoin_free(void* handle) {
   REQUIRE_INIT();
   BgJoinTable* t = (BgJoinTable*)handle;
   if (!t) return BG_OK;
-  (void)pool_release(t->keys);
+  (void)pool_release(t->nodes);
   (void)pool_release(t->head);
-  (void)pool_release(t->next);
   if (t->probe_offsets) (void)pool_release(t->probe_offsets);
   delete t;
   return BG_OK;
@@ -1362,31 +1386,33 @@ __device__ __forceinline__ u64 hash_keys_row(const KeyArgs& keys, int64_t i) {
   return h;
 }
 
-// slot_row: i32[cap], -1 empty (probe path touches only this array);
-// slot_data: u64[cap * (1 + 2*naggs)] interleaved [count, acc0_lo, acc0_hi,
-// ...] so one row's accumulation lands in ONE cache line (vs 3 random
-// lines with split count/acc arrays — measured 2x on the 12M-group sweep)
+// slot_data: u64[cap * (2 + 2*naggs)] fully interleaved records
+// [claim = first_row+1 (0 empty), count, acc0_lo, acc0_hi, ...] — the
+// probe's claim word and the accumulators share ONE cache line per slot,
+// so a row costs ~1 random line total (claim CAS/read + adds hit the same
+// line); claims are device-scope u64 atomics (placement-independent)
 __global__ void k_hashagg(KeyArgs keys, AggArgs aggs, const u64* mask_words,
-                          int64_t n, int* slot_row, u64 cap_mask,
-                          u64* slot_data, int* err_flag) {
-  const int rec = 1 + 2 * aggs.naggs;
+                          int64_t n, u64 cap_mask, u64* slot_data,
+                          int* err_flag) {
+  const int rec = 2 + 2 * aggs.naggs;
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
        i += (int64_t)gridDim.x * blockDim.x) {
     if (mask_words && !((mask_words[i >> 6] >> (i & 63)) & 1)) continue;
     const u64 h = hash_keys_row(keys, i);
     u64 slot = h & cap_mask;
     int64_t probes = 0;
-    int64_t my_slot = -1;
+    u64* srec = nullptr;
     while (true) {
-      int cur = __hip_atomic_load(&slot_row[slot], __ATOMIC_RELAXED,
+      u64* cand = slot_data + slot * rec;
+      u64 cur = __hip_atomic_load(cand, __ATOMIC_RELAXED,
                                   __HIP_MEMORY_SCOPE_AGENT);
-      if (cur == -1) {
-        int old = atomicCAS(&slot_row[slot], -1, (int)i);
-        if (old == -1) { my_slot = (int64_t)slot; break; }
+      if (cur == 0) {
+        u64 old = atomicCAS(cand, 0ull, (u64)i + 1);
+        if (old == 0) { srec = cand; break; }
         cur = old;
       }
-      if (cur >= 0 && keys_equal_rows(keys, (int64_t)cur, i)) {
-        my_slot = (int64_t)slot;
+      if (cur != 0 && keys_equal_rows(keys, (int64_t)cur - 1, i)) {
+        srec = cand;
         break;
       }
       slot = (slot + 1) & cap_mask;
@@ -1395,10 +1421,9 @@ __global__ void k_hashagg(KeyArgs keys, AggArgs aggs, const u64* mask_words,
         return;
       }
     }
-    u64* srec = slot_data + (u64)my_slot * rec;
-    atomicAdd(&srec[0], 1ull);
+    atomicAdd(&srec[1], 1ull);
     for (int a = 0; a < aggs.naggs; ++a) {
-      u64* base = srec + 1 + 2 * a;
+      u64* base = srec + 2 + 2 * a;
       switch (aggs.a[a].op) {
         case BG_AGG_SUM_DEC128: {
           const ulong2 v =
@@ -1421,7 +1446,7 @@ __global__ void k_hashagg(KeyArgs keys, AggArgs aggs, const u64* mask_words,
 }
 
 // occupancy bitmask over slots (feeds the stable compaction)
-__global__ void k_slot_occupancy(const int* slot_row, int64_t cap,
+__global__ void k_slot_occupancy(const u64* slot_data, int rec, int64_t cap,
                                  u64* mask_words) {
   const int64_t wave_global =
       ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) / BG_WAVE;
@@ -1429,7 +1454,7 @@ __global__ void k_slot_occupancy(const int* slot_row, int64_t cap,
   const int64_t nwords = (cap + 63) / 64;
   for (int64_t w = wave_global; w < nwords; w += nwaves) {
     const int64_t s = w * BG_WAVE + lane_id();
-    const bool occ = s < cap && slot_row[s] >= 0;
+    const bool occ = s < cap && slot_data[(u64)s * rec] != 0;
     u64 m = __ballot(occ);
     if (lane_id() == 0) mask_words[w] = m;
   }
@@ -1438,19 +1463,19 @@ __global__ void k_slot_occupancy(const int* slot_row, int64_t cap,
 // gather group outputs: for dense group g = 0..ngroups-1 with slot index
 // sidx[g]: first_row[g], counts_out[g], acc_out[g*naggs*2 ..]
 __global__ void k_hashagg_gather(const uint32_t* sidx, int64_t ngroups,
-                                 const int* slot_row, const u64* slot_data,
-                                 int naggs, uint32_t* first_row,
-                                 u64* counts_out, u64* acc_out) {
-  const int rec = 1 + 2 * naggs;
+                                 const u64* slot_data, int naggs,
+                                 uint32_t* first_row, u64* counts_out,
+                                 u64* acc_out) {
+  const int rec = 2 + 2 * naggs;
   for (int64_t g = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; g < ngroups;
        g += (int64_t)gridDim.x * blockDim.x) {
     const int64_t s = sidx[g];
     const u64* srec = slot_data + (u64)s * rec;
-    first_row[g] = (uint32_t)slot_row[s];
-    counts_out[g] = srec[0];
+    first_row[g] = (uint32_t)(srec[0] - 1);
+    counts_out[g] = srec[1];
     for (int a = 0; a < naggs; ++a) {
-      acc_out[((u64)g * naggs + a) * 2] = srec[1 + 2 * a];
-      acc_out[((u64)g * naggs + a) * 2 + 1] = srec[2 + 2 * a];
+      acc_out[((u64)g * naggs + a) * 2] = srec[2 + 2 * a];
+      acc_out[((u64)g * naggs + a) * 2 + 1] = srec[3 + 2 * a];
     }
   }
 }
@@ -1482,14 +1507,11 @@ extern "C" int bg_hashagg(const bg_column* key_cols, int32_t nkeys,
   }
   u64 cap = 8;
   while (cap < (u64)(max_groups * 2)) cap <<= 1;
-  const int rec = 1 + 2 * naggs;
-  int* slot_row;
+  const int rec = 2 + 2 * naggs;
   u64* slot_data;
   int* err_flag;
-  HIP_TRY(pool_malloc((void**)&slot_row, sizeof(int) * cap));
   HIP_TRY(pool_malloc((void**)&slot_data, sizeof(u64) * cap * rec));
   HIP_TRY(pool_malloc((void**)&err_flag, sizeof(int)));
-  HIP_TRY(hipMemset(slot_row, 0xff, sizeof(int) * cap));
   HIP_TRY(hipMemset(slot_data, 0, sizeof(u64) * cap * rec));
   HIP_TRY(hipMemset(err_flag, 0, sizeof(int)));
 
@@ -1500,8 +1522,8 @@ extern "C" int bg_hashagg(const bg_column* key_cols, int32_t nkeys,
   HIP_TRY(hipEventCreate(&ev1));
   HIP_TRY(hipEventRecord(ev0, 0));
   hipLaunchKernelGGL(k_hashagg, dim3(blocks), dim3(BG_BLOCK), 0, 0, keys, aggs,
-                     reinterpret_cast<const u64*>(d_mask), n, slot_row,
-                     cap - 1, slot_data, err_flag);
+                     reinterpret_cast<const u64*>(d_mask), n, cap - 1,
+                     slot_data, err_flag);
   HIP_TRY(hipGetLastError());
   HIP_TRY(hipEventRecord(ev1, 0));
   HIP_TRY(hipEventSynchronize(ev1));
@@ -1514,7 +1536,7 @@ extern "C" int bg_hashagg(const bg_column* key_cols, int32_t nkeys,
   int err = 0;
   HIP_TRY(hipMemcpy(&err, err_flag, sizeof(int), hipMemcpyDeviceToHost));
   if (err) {
-    (void)pool_release(slot_row); (void)pool_release(slot_data);
+    (void)pool_release(slot_data);
     (void)pool_release(err_flag);
     return set_err(BG_ERR_INVALID, "bg_hashagg: table full (raise max_groups)");
   }
@@ -1530,7 +1552,7 @@ extern "C" int bg_hashagg(const bg_column* key_cols, int32_t nkeys,
                                BG_MAX_BLOCKS);
   if (oblocks == 0) oblocks = 1;
   hipLaunchKernelGGL(k_slot_occupancy, dim3(oblocks), dim3(BG_BLOCK), 0, 0,
-                     slot_row, (int64_t)cap, occ_mask);
+                     slot_data, rec, (int64_t)cap, occ_mask);
   int64_t ngroups = 0;
   int rc = bg_mask_to_indices(reinterpret_cast<const uint8_t*>(occ_mask),
                               (int64_t)cap, sidx, &ngroups);
@@ -1540,12 +1562,11 @@ extern "C" int bg_hashagg(const bg_column* key_cols, int32_t nkeys,
                                BG_MAX_BLOCKS);
   if (gblocks == 0) gblocks = 1;
   hipLaunchKernelGGL(k_hashagg_gather, dim3(gblocks), dim3(BG_BLOCK), 0, 0,
-                     sidx, ngroups, slot_row, slot_data, naggs, d_first_row,
+                     sidx, ngroups, slot_data, naggs, d_first_row,
                      reinterpret_cast<u64*>(d_counts_out),
                      reinterpret_cast<u64*>(d_acc_out));
   HIP_TRY(hipGetLastError());
   HIP_TRY(hipDeviceSynchronize());
-  HIP_TRY(pool_release(slot_row));
   HIP_TRY(pool_release(slot_data));
   HIP_TRY(pool_release(err_flag));
   HIP_TRY(pool_release(occ_mask));
