@@ -1808,3 +1808,97 @@ extern "C" int bg_sort_rows(const bg_column* key_cols,
   return rc;
 }
 
+
+// ---------------------------------------------------------------------------
+// SortMergeJoinExec (INNER, Int64 keys; SURVEY.md §8f row 2 second half —
+// the reference's DEFAULT partitioned join, prefer_hash_join=false,
+// core/src/extension.rs:850-858, every approved/q*.txt join stage).
+// Both inputs sorted on the join key (bg_sort_rows upstream); each probe
+// row binary-searches its build-side equal range -> exact per-row counts ->
+// scan -> fill.  Output is FULLY ordered: probe-major, build ascending —
+// stronger determinism than the hash join's chain order.
+// ---------------------------------------------------------------------------
+__device__ __forceinline__ int64_t lower_bound_i64(const int64_t* a,
+                                                   int64_t n, int64_t key) {
+  int64_t lo = 0, hi = n;
+  while (lo < hi) {
+    const int64_t mid = (lo + hi) >> 1;
+    if (a[mid] < key) lo = mid + 1;
+    else hi = mid;
+  }
+  return lo;
+}
+
+__global__ void k_merge_count(const int64_t* build_sorted, int64_t nb,
+                              const int64_t* probe_sorted, int64_t np,
+                              u64* counts, int64_t* lbs) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < np;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    const int64_t k = probe_sorted[i];
+    const int64_t lb = lower_bound_i64(build_sorted, nb, k);
+    int64_t ub = lb;
+    while (ub < nb && build_sorted[ub] == k) ++ub;  // runs are short in
+    counts[i] = (u64)(ub - lb);                      // equi-join practice
+    lbs[i] = lb;
+  }
+}
+
+__global__ void k_merge_fill(const int64_t* build_sorted, int64_t nb,
+                             int64_t np, const i64* offsets,
+                             const int64_t* lbs, const u64* counts,
+                             uint32_t* out_probe, uint32_t* out_build) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < np;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    i64 w = offsets[i];
+    const int64_t lb = lbs[i];
+    const int64_t c = (int64_t)counts[i];
+    for (int64_t j = 0; j < c; ++j) {
+      out_probe[w + j] = (uint32_t)i;
+      out_build[w + j] = (uint32_t)(lb + j);
+    }
+  }
+}
+
+/* Inner merge join of two KEY-SORTED Int64 arrays.  Emits positions INTO
+ * THE SORTED ORDERS (compose with the sort permutations to recover
+ * original row ids).  Two-phase: call with d_out_* = NULL to get the match
+ * count, then with buffers sized accordingly (state cached per call pair
+ * is avoided by recomputing the cheap count pass). */
+extern "C" int bg_merge_join(const int64_t* d_build_sorted, int64_t nb,
+                             const int64_t* d_probe_sorted, int64_t np,
+                             int64_t* out_matches, uint32_t* d_out_probe,
+                             uint32_t* d_out_build) {
+  REQUIRE_INIT();
+  u64* d_counts;
+  int64_t* d_lbs;
+  i64* d_offs;
+  i64* d_total;
+  HIP_TRY(pool_malloc((void**)&d_counts, sizeof(u64) * (np ? np : 1)));
+  HIP_TRY(pool_malloc((void**)&d_lbs, sizeof(int64_t) * (np ? np : 1)));
+  HIP_TRY(pool_malloc((void**)&d_offs, sizeof(i64) * (np ? np : 1)));
+  HIP_TRY(pool_malloc((void**)&d_total, sizeof(i64)));
+  int blocks = (int)bg_imin64((np + BG_BLOCK - 1) / BG_BLOCK, BG_MAX_BLOCKS);
+  if (blocks == 0) blocks = 1;
+  hipLaunchKernelGGL(k_merge_count, dim3(blocks), dim3(BG_BLOCK), 0, 0,
+                     d_build_sorted, nb, d_probe_sorted, np, d_counts, d_lbs);
+  int rc = scan_exclusive_i64(d_counts, np, d_offs, d_total);
+  i64 total = 0;
+  if (rc == BG_OK) {
+    hipError_t e = hipMemcpy(&total, d_total, sizeof(i64),
+                             hipMemcpyDeviceToHost);
+    if (e != hipSuccess) rc = set_hip_err(e, "total copy");
+  }
+  if (rc == BG_OK && d_out_probe && d_out_build) {
+    hipLaunchKernelGGL(k_merge_fill, dim3(blocks), dim3(BG_BLOCK), 0, 0,
+                       d_build_sorted, nb, np, d_offs, d_lbs, d_counts,
+                       d_out_probe, d_out_build);
+    hipError_t e = hipGetLastError();
+    if (e != hipSuccess) rc = set_hip_err(e, "merge fill");
+  }
+  (void)pool_release(d_counts);
+  (void)pool_release(d_lbs);
+  (void)pool_release(d_offs);
+  (void)pool_release(d_total);
+  if (rc == BG_OK) *out_matches = total;
+  return rc;
+}

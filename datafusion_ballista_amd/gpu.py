@@ -391,3 +391,25 @@ def _ctx_sort_rows(self, key_cols, descending, n):
 
 
 GpuStageContext.sort_rows = _ctx_sort_rows
+
+
+def _ctx_merge_join(self, build_sorted: "DeviceBuffer", nb: int,
+                    probe_sorted: "DeviceBuffer", np_: int):
+    """Inner merge join of key-sorted i64 buffers -> (probe_pos, build_pos
+    DeviceBuffers into the sorted orders, n_matches)."""
+    matches = ctypes.c_int64()
+    _check(self.L.bg_merge_join(build_sorted.ptr, ctypes.c_int64(nb),
+                                probe_sorted.ptr, ctypes.c_int64(np_),
+                                ctypes.byref(matches), None, None),
+           "bg_merge_join(count)")
+    m = matches.value
+    pbuf = self.alloc(max(4 * m, 4))
+    bbuf = self.alloc(max(4 * m, 4))
+    _check(self.L.bg_merge_join(build_sorted.ptr, ctypes.c_int64(nb),
+                                probe_sorted.ptr, ctypes.c_int64(np_),
+                                ctypes.byref(matches), pbuf.ptr, bbuf.ptr),
+           "bg_merge_join(fill)")
+    return pbuf, bbuf, m
+
+
+GpuStageContext.merge_join = _ctx_merge_join

@@ -225,6 +225,17 @@ int bg_project_dec128(int32_t op, const bg_column* a, const bg_column* b,
 int bg_sort_rows(const bg_column* key_cols, const int32_t* descending,
                  int32_t nkeys, int64_t n, uint32_t* d_perm);
 
+/* ---- SortMergeJoinExec (INNER, key-sorted Int64 inputs) ----
+ * The reference's DEFAULT partitioned join (prefer_hash_join=false,
+ * extension.rs:850-858; every approved/q*.txt join stage is SMJ).  Inputs
+ * are key-sorted (bg_sort_rows); emits positions into the sorted orders,
+ * probe-major with build ascending (fully deterministic).  Call once with
+ * d_out_* NULL to size the output, then again with buffers. */
+int bg_merge_join(const int64_t* d_build_sorted, int64_t nb,
+                  const int64_t* d_probe_sorted, int64_t np,
+                  int64_t* out_matches, uint32_t* d_out_probe,
+                  uint32_t* d_out_build);
+
 /* ---- fused filter+aggregate stages ---- */
 
 /* TPC-H q6 stage 1 (scan+filter+aggregate, approved/q6.txt):

@@ -635,3 +635,42 @@ def test_q1_partial_final_two_stage(ctx, tmp_path):
             got_avg = sums[idx_a] / cnt / scale
             want_avg = ws[idx_a] / wc / scale
             assert abs(got_avg - want_avg) <= 1e-6 * max(abs(want_avg), 1e-30)
+
+
+# ---------------------------------------------------------------------------
+# sort-merge join (reference-default partitioned join shape)
+# ---------------------------------------------------------------------------
+@pytest.mark.parametrize("nb,np_,kspace", [(1000, 5000, 500),
+                                           (100_000, 300_000, 50_000)])
+def test_merge_join_parity(ctx, nb, np_, kspace):
+    """SMJ(sorted build, sorted probe) emits exactly the hash-join pair
+    multiset, mapped through the sort permutations; output fully ordered
+    (probe-major, build ascending)."""
+    rng = np.random.default_rng(nb * 3 + np_)
+    build = rng.integers(0, kspace, size=nb, dtype=np.int64)
+    probe = rng.integers(0, kspace, size=np_, dtype=np.int64)
+
+    bcol, bbuf_raw = ctx.upload_column(build, gpu.BG_DT_INT64)
+    pcol, pbuf_raw = ctx.upload_column(probe, gpu.BG_DT_INT64)
+    bperm = ctx.sort_rows([bcol], [False], nb)
+    pperm = ctx.sort_rows([pcol], [False], np_)
+    bsorted = ctx.gather(bbuf_raw, 8, bperm, nb)
+    psorted = ctx.gather(pbuf_raw, 8, pperm, np_)
+    ppos_buf, bpos_buf, m = ctx.merge_join(bsorted, nb, psorted, np_)
+    ppos = ppos_buf.download(np.uint32, m)
+    bpos = bpos_buf.download(np.uint32, m)
+    # map sorted positions back to original row ids
+    bperm_h = bperm.download(np.uint32, nb)
+    pperm_h = pperm.download(np.uint32, np_)
+    got_pairs = np.sort(pperm_h[ppos].astype(np.uint64) << np.uint64(32)
+                        | bperm_h[bpos])
+
+    want_p, want_b = oracle.hashjoin_pairs(build, probe)
+    want_pairs = np.sort(want_p.astype(np.uint64) << np.uint64(32) | want_b)
+    assert m == len(want_p)
+    assert np.array_equal(got_pairs, want_pairs)
+    # ordered-output contract: probe positions non-decreasing, build
+    # positions ascending within a probe row
+    assert np.all(np.diff(ppos.astype(np.int64)) >= 0)
+    same = np.diff(ppos.astype(np.int64)) == 0
+    assert np.all(np.diff(bpos.astype(np.int64))[same] > 0)
