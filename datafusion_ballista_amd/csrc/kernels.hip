@@ -1902,3 +1902,115 @@ extern "C" int bg_merge_join(const int64_t* d_build_sorted, int64_t nb,
   if (rc == BG_OK) *out_matches = total;
   return rc;
 }
+
+// ---------------------------------------------------------------------------
+// GPU Parquet decode, stage 1: Snappy page
+// decompression.  Parquet compresses each page independently, so the
+// parallelism unit is the PAGE: one thread decodes one page serially
+// (the format is a strict byte-serial LZ77 variant), thousands of pages
+// decode concurrently — throughput scales with resident pages, latency
+// hidden by oversubscription.  Format: varint uncompressed length, then
+// tagged elements: tag&3==0 literal (len up to 4-byte extension),
+// 1 copy-1B (len 4-11, 11-bit offset), 2 copy-2B, 3 copy-4B.
+// ---------------------------------------------------------------------------
+struct SnappyPage {
+  const uint8_t* src;
+  uint8_t* dst;
+  int64_t src_len;
+  int64_t dst_cap;
+};
+
+__global__ void k_snappy_decompress(const SnappyPage* pages, int64_t npages,
+                                    int64_t* out_lens /* -1 on error */) {
+  for (int64_t p = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; p < npages;
+       p += (int64_t)gridDim.x * blockDim.x) {
+    const uint8_t* s = pages[p].src;
+    const uint8_t* send = s + pages[p].src_len;
+    uint8_t* d = pages[p].dst;
+    // varint32 uncompressed length
+    u64 ulen = 0;
+    int shift = 0;
+    bool ok = true;
+    while (s < send) {
+      const uint8_t b = *s++;
+      ulen |= (u64)(b & 0x7f) << shift;
+      if (!(b & 0x80)) break;
+      shift += 7;
+      if (shift > 32) { ok = false; break; }
+    }
+    if (!ok || (int64_t)ulen > pages[p].dst_cap) {
+      out_lens[p] = -1;
+      continue;
+    }
+    uint8_t* dend = d + ulen;
+    while (ok && s < send && d < dend) {
+      const uint8_t tag = *s++;
+      if ((tag & 3) == 0) {  // literal
+        int64_t len = (tag >> 2) + 1;
+        if (len > 60) {
+          const int nb = (int)len - 60;  // 1..4 length bytes
+          if (s + nb > send) { ok = false; break; }
+          len = 0;
+          for (int i = 0; i < nb; ++i) len |= (int64_t)s[i] << (8 * i);
+          len += 1;
+          s += nb;
+        }
+        if (s + len > send || d + len > dend) { ok = false; break; }
+        for (int64_t i = 0; i < len; ++i) d[i] = s[i];
+        d += len;
+        s += len;
+      } else {
+        int64_t len, off;
+        if ((tag & 3) == 1) {  // copy with 1-byte offset
+          if (s >= send) { ok = false; break; }
+          len = ((tag >> 2) & 7) + 4;
+          off = ((int64_t)(tag >> 5) << 8) | *s++;
+        } else if ((tag & 3) == 2) {  // 2-byte offset
+          if (s + 2 > send) { ok = false; break; }
+          len = (tag >> 2) + 1;
+          off = (int64_t)s[0] | ((int64_t)s[1] << 8);
+          s += 2;
+        } else {  // 4-byte offset
+          if (s + 4 > send) { ok = false; break; }
+          len = (tag >> 2) + 1;
+          off = (int64_t)s[0] | ((int64_t)s[1] << 8) |
+                ((int64_t)s[2] << 16) | ((int64_t)s[3] << 24);
+          s += 4;
+        }
+        if (off == 0 || off > d - pages[p].dst || d + len > dend) {
+          ok = false;
+          break;
+        }
+        const uint8_t* cs = d - off;  // may overlap forward (RLE-style)
+        for (int64_t i = 0; i < len; ++i) d[i] = cs[i];
+        d += len;
+      }
+    }
+    out_lens[p] = (ok && d == dend) ? (int64_t)ulen : -1;
+  }
+}
+
+/* Decompress npages independent Snappy blocks.  pages: HOST array copied
+ * internally; each entry's src/dst are DEVICE pointers.  out_lens (host,
+ * npages): decompressed length or -1 on malformed input. */
+extern "C" int bg_snappy_decompress(const void* h_pages, int64_t npages,
+                                    int64_t* h_out_lens) {
+  REQUIRE_INIT();
+  SnappyPage* d_pages;
+  int64_t* d_lens;
+  HIP_TRY(pool_malloc((void**)&d_pages, sizeof(SnappyPage) * (npages ? npages : 1)));
+  HIP_TRY(pool_malloc((void**)&d_lens, sizeof(int64_t) * (npages ? npages : 1)));
+  HIP_TRY(hipMemcpy(d_pages, h_pages, sizeof(SnappyPage) * npages,
+                    hipMemcpyHostToDevice));
+  int blocks = (int)bg_imin64((npages + BG_BLOCK - 1) / BG_BLOCK,
+                              BG_MAX_BLOCKS);
+  if (blocks == 0) blocks = 1;
+  hipLaunchKernelGGL(k_snappy_decompress, dim3(blocks), dim3(BG_BLOCK), 0, 0,
+                     d_pages, npages, d_lens);
+  HIP_TRY(hipGetLastError());
+  HIP_TRY(hipMemcpy(h_out_lens, d_lens, sizeof(int64_t) * npages,
+                    hipMemcpyDeviceToHost));
+  (void)pool_release(d_pages);
+  (void)pool_release(d_lens);
+  return BG_OK;
+}

@@ -413,3 +413,26 @@ def _ctx_merge_join(self, build_sorted: "DeviceBuffer", nb: int,
 
 
 GpuStageContext.merge_join = _ctx_merge_join
+
+
+class BgSnappyPage(ctypes.Structure):
+    _fields_ = [("d_src", ctypes.c_void_p), ("d_dst", ctypes.c_void_p),
+                ("src_len", ctypes.c_int64), ("dst_cap", ctypes.c_int64)]
+
+
+def _ctx_snappy_decompress(self, pages):
+    """pages: list of (src DeviceBuffer, src_len, dst DeviceBuffer,
+    dst_cap). -> list of decompressed lengths (-1 = malformed)."""
+    n = len(pages)
+    arr = (BgSnappyPage * n)()
+    for i, (src, slen, dst, dcap) in enumerate(pages):
+        arr[i] = BgSnappyPage(src.ptr, dst.ptr, slen, dcap)
+    lens = np.zeros(n, dtype=np.int64)
+    _check(self.L.bg_snappy_decompress(
+        arr, ctypes.c_int64(n),
+        lens.ctypes.data_as(ctypes.POINTER(ctypes.c_int64))),
+        "bg_snappy_decompress")
+    return lens.tolist()
+
+
+GpuStageContext.snappy_decompress = _ctx_snappy_decompress

@@ -236,6 +236,20 @@ int bg_merge_join(const int64_t* d_build_sorted, int64_t nb,
                   int64_t* out_matches, uint32_t* d_out_probe,
                   uint32_t* d_out_build);
 
+/* ---- GPU Parquet decode, stage 1: Snappy page decompression ----
+ * (SURVEY.md §8f row 1).  One device thread decodes one independently-
+ * compressed page; thousands of pages decode concurrently.  h_pages is a
+ * host array of {src, dst, src_len, dst_cap} with DEVICE src/dst pointers;
+ * h_out_lens receives decompressed lengths (-1 = malformed page). */
+typedef struct {
+  const void* d_src;
+  void* d_dst;
+  int64_t src_len;
+  int64_t dst_cap;
+} bg_snappy_page;
+int bg_snappy_decompress(const void* h_pages /* bg_snappy_page[npages] */,
+                         int64_t npages, int64_t* h_out_lens);
+
 /* ---- fused filter+aggregate stages ---- */
 
 /* TPC-H q6 stage 1 (scan+filter+aggregate, approved/q6.txt):
