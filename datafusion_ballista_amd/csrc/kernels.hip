@@ -2023,9 +2023,9 @@ extern "C" int bg_snappy_decompress(const void* h_pages, int64_t npages,
 // value bytes to the column buffer; FLBA(16) decimals (parquet big-endian)
 // are byte-reversed to Arrow little-endian in flight.
 // ---------------------------------------------------------------------------
-__global__ void k_page_extract(const uint8_t* page, uint8_t* out,
-                               int64_t nvals, int64_t esz, int has_def,
-                               int flba_reverse, int* err) {
+__global__ void k_page_extract(const uint8_t* page, int64_t page_len,
+                               uint8_t* out, int64_t nvals, int64_t src_esz,
+                               int has_def, int flba_reverse, int* err) {
   __shared__ int64_t s_voff;
   if (threadIdx.x == 0) {
     int64_t voff = 0;
@@ -2068,40 +2068,46 @@ __global__ void k_page_extract(const uint8_t* page, uint8_t* out,
       if (seen < nvals) ok = 0;
       voff = 4 + dlen;
     }
+    if (voff + nvals * src_esz > page_len) ok = 0;  // geometry bound
     if (!ok) atomicExch(err, 2);  // nulls / malformed: unsupported this round
-    s_voff = voff;
+    s_voff = ok ? voff : -1;
   }
   __syncthreads();
   const int64_t voff = s_voff;
-  const int64_t nbytes = nvals * esz;
+  if (voff < 0) return;
   if (!flba_reverse) {
+    const int64_t nbytes = nvals * src_esz;
     for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
          i < nbytes; i += (int64_t)gridDim.x * blockDim.x)
       out[i] = page[voff + i];
   } else {
-    // reverse each 16-byte big-endian decimal to little-endian
+    // FLBA(src_esz) big-endian two's complement -> 16-byte LE decimal128
+    // with sign extension (parquet stores decimals at the minimal width
+    // for the precision, e.g. 7 bytes for Decimal(15,2))
     for (int64_t v = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
          v < nvals; v += (int64_t)gridDim.x * blockDim.x) {
-      const uint8_t* src = page + voff + v * 16;
+      const uint8_t* src = page + voff + v * src_esz;
       uint8_t* dst = out + v * 16;
-      for (int b = 0; b < 16; ++b) dst[b] = src[15 - b];
+      const uint8_t sign = (src[0] & 0x80) ? 0xff : 0x00;
+      for (int64_t b = 0; b < 16; ++b)
+        dst[b] = (b < src_esz) ? src[src_esz - 1 - b] : sign;
     }
   }
 }
 
-extern "C" int bg_page_extract(const void* d_page, void* d_out,
-                               int64_t dst_byte_off, int64_t nvals,
-                               int64_t esz, int32_t has_def,
-                               int32_t flba_reverse) {
+extern "C" int bg_page_extract(const void* d_page, int64_t page_len,
+                               void* d_out, int64_t dst_byte_off,
+                               int64_t nvals, int64_t src_esz,
+                               int32_t has_def, int32_t flba_reverse) {
   REQUIRE_INIT();
   int* d_err;
   HIP_TRY(pool_malloc((void**)&d_err, sizeof(int)));
   HIP_TRY(hipMemset(d_err, 0, sizeof(int)));
-  int blocks = (int)bg_imin64((nvals * esz + BG_BLOCK - 1) / BG_BLOCK, 512);
+  int blocks = (int)bg_imin64((nvals * src_esz + BG_BLOCK - 1) / BG_BLOCK, 512);
   if (blocks == 0) blocks = 1;
   hipLaunchKernelGGL(k_page_extract, dim3(blocks), dim3(BG_BLOCK), 0, 0,
-                     (const uint8_t*)d_page,
-                     (uint8_t*)d_out + dst_byte_off, nvals, esz, has_def,
+                     (const uint8_t*)d_page, page_len,
+                     (uint8_t*)d_out + dst_byte_off, nvals, src_esz, has_def,
                      flba_reverse, d_err);
   HIP_TRY(hipGetLastError());
   int err = 0;

@@ -159,10 +159,14 @@ class GpuParquetColumnReader:
             raise RuntimeError(f"codec {codec} not GPU-decodable yet")
         phys = meta.physical_type
         if phys in _PHYS_NP:
-            esz = _PHYS_NP[phys][1]
+            src_esz = dst_esz = _PHYS_NP[phys][1]
             flba = False
         elif phys == "FIXED_LEN_BYTE_ARRAY":
-            esz = 16  # decimal128, stored big-endian; device flips to LE
+            # parquet stores decimals at the minimal width for the
+            # precision (FLBA(7) for Decimal(15,2)), big-endian; the device
+            # sign-extends to 16-byte LE decimal128
+            src_esz = self.pf.schema.column(col).length
+            dst_esz = 16
             flba = True
         else:
             raise RuntimeError(f"physical type {phys} not GPU-decodable yet")
@@ -177,7 +181,7 @@ class GpuParquetColumnReader:
         total_values = meta.num_values
 
         # column output buffer + per-page scratch
-        out = ctx.alloc(max(total_values * esz, esz))
+        out = ctx.alloc(max(total_values * dst_esz, dst_esz))
         pages = []           # (src buf, src_len, scratch buf, ulen)
         extracts = []        # (scratch, dst_off, num_values, def_prefixed)
         got_values = 0
@@ -203,7 +207,7 @@ class GpuParquetColumnReader:
                     pages.append((src, csz, scratch, usz))
                 else:
                     scratch.upload(np.frombuffer(payload, dtype=np.uint8))
-                extracts.append((scratch, got_values, nvals))
+                extracts.append((scratch, usz, got_values, nvals))
                 got_values += nvals
             pos = data_pos + csz
 
@@ -220,10 +224,11 @@ class GpuParquetColumnReader:
         # required column => values at offset 0
         max_def = self.pf.schema.column(col).max_definition_level
         import ctypes
-        for scratch, dst_off, nvals in extracts:
+        for scratch, usz, dst_off, nvals in extracts:
             rc = ctx.L.bg_page_extract(
-                scratch.ptr, out.ptr, ctypes.c_int64(dst_off * esz),
-                ctypes.c_int64(nvals), ctypes.c_int64(esz),
+                scratch.ptr, ctypes.c_int64(usz), out.ptr,
+                ctypes.c_int64(dst_off * dst_esz),
+                ctypes.c_int64(nvals), ctypes.c_int64(src_esz),
                 1 if max_def > 0 else 0, 1 if flba else 0)
             gpu._check(rc, "bg_page_extract")
         ctx.synchronize()

@@ -254,9 +254,9 @@ int bg_snappy_decompress(const void* h_pages /* bg_snappy_page[npages] */,
  * block (optional columns) on device, copies the value bytes into the
  * column buffer at dst_byte_off; flba_reverse flips 16-B big-endian
  * decimals to Arrow LE.  Fails loudly on nulls/V2/dictionary pages. */
-int bg_page_extract(const void* d_page, void* d_out, int64_t dst_byte_off,
-                    int64_t nvals, int64_t esz, int32_t has_def,
-                    int32_t flba_reverse);
+int bg_page_extract(const void* d_page, int64_t page_len, void* d_out,
+                    int64_t dst_byte_off, int64_t nvals, int64_t src_esz,
+                    int32_t has_def, int32_t flba_reverse);
 
 /* ---- fused filter+aggregate stages ---- */
 
