@@ -2119,3 +2119,121 @@ extern "C" int bg_page_extract(const void* d_page, int64_t page_len,
                    "levels (null decode: later round)");
   return BG_OK;
 }
+
+// ---------------------------------------------------------------------------
+// Parquet dictionary-encoded data pages (PLAIN_DICTIONARY/RLE_DICTIONARY —
+// the default layout modern writers produce): after the def-level block the
+// page holds [u8 bit_width][RLE/bit-packed hybrid dictionary indices].
+// One thread expands one page's indices serially (strictly byte-serial
+// format; pages decode concurrently); the column is then materialised with
+// the existing bg_gather from the (PLAIN-decoded) dictionary.
+// ---------------------------------------------------------------------------
+__global__ void k_dict_indices(const uint8_t* page, int64_t page_len,
+                               int64_t nvals, int has_def, uint32_t* out_idx,
+                               int* err) {
+  if (blockIdx.x != 0 || threadIdx.x != 0) return;
+  const uint8_t* d = page;
+  const uint8_t* pend = page + page_len;
+  if (has_def) {
+    if (d + 4 > pend) { atomicExch(err, 2); return; }
+    const uint32_t dlen = (uint32_t)d[0] | ((uint32_t)d[1] << 8) |
+                          ((uint32_t)d[2] << 16) | ((uint32_t)d[3] << 24);
+    // validate all-1 def levels exactly like k_page_extract
+    const uint8_t* dl = d + 4;
+    const uint8_t* dlend = dl + dlen;
+    if (dlend > pend) { atomicExch(err, 2); return; }
+    int64_t seen = 0;
+    while (dl < dlend && seen < nvals) {
+      u64 header = 0;
+      int shift = 0;
+      while (dl < dlend) {
+        const uint8_t b = *dl++;
+        header |= (u64)(b & 0x7f) << shift;
+        if (!(b & 0x80)) break;
+        shift += 7;
+      }
+      if (header & 1) {
+        const int64_t groups = (int64_t)(header >> 1);
+        for (int64_t g = 0; g < groups; ++g) {
+          if (dl >= dlend) { atomicExch(err, 2); return; }
+          const uint8_t byte = *dl++;
+          const int64_t take = nvals - seen >= 8 ? 8 : nvals - seen;
+          for (int64_t t = 0; t < take; ++t)
+            if (!((byte >> t) & 1)) { atomicExch(err, 2); return; }
+          seen += take;
+        }
+      } else {
+        const int64_t run = (int64_t)(header >> 1);
+        if (dl >= dlend) { atomicExch(err, 2); return; }
+        if (*dl++ != 1) { atomicExch(err, 2); return; }
+        seen += run;
+      }
+    }
+    if (seen < nvals) { atomicExch(err, 2); return; }
+    d += 4 + dlen;
+  }
+  if (d >= pend) { atomicExch(err, 3); return; }
+  const int bw = *d++;  // index bit width (0..32)
+  if (bw < 0 || bw > 32) { atomicExch(err, 3); return; }
+  int64_t outp = 0;
+  if (bw == 0) {  // every index is 0
+    for (; outp < nvals; ++outp) out_idx[outp] = 0;
+    return;
+  }
+  const u64 vmask = (bw == 32) ? 0xffffffffull : ((1ull << bw) - 1);
+  while (d < pend && outp < nvals) {
+    u64 header = 0;
+    int shift = 0;
+    while (d < pend) {
+      const uint8_t b = *d++;
+      header |= (u64)(b & 0x7f) << shift;
+      if (!(b & 0x80)) break;
+      shift += 7;
+    }
+    if (header & 1) {  // bit-packed: (ngroups<<1)|1, groups of 8 values
+      const int64_t groups = (int64_t)(header >> 1);
+      for (int64_t g = 0; g < groups && outp < nvals; ++g) {
+        if (d + bw > pend) { atomicExch(err, 3); return; }
+        u64 bits = 0;
+        for (int b2 = 0; b2 < bw; ++b2) bits |= (u64)d[b2] << (8 * b2);
+        d += bw;
+        const int64_t take = nvals - outp >= 8 ? 8 : nvals - outp;
+        for (int64_t t = 0; t < take; ++t)
+          out_idx[outp + t] = (uint32_t)((bits >> (t * bw)) & vmask);
+        outp += take;
+      }
+    } else {  // RLE run: value in ceil(bw/8) LE bytes
+      const int64_t run = (int64_t)(header >> 1);
+      const int nb = (bw + 7) / 8;
+      if (d + nb > pend) { atomicExch(err, 3); return; }
+      u64 v = 0;
+      for (int b2 = 0; b2 < nb; ++b2) v |= (u64)d[b2] << (8 * b2);
+      d += nb;
+      const int64_t take = run <= nvals - outp ? run : nvals - outp;
+      for (int64_t t = 0; t < take; ++t) out_idx[outp + t] = (uint32_t)v;
+      outp += take;
+    }
+  }
+  if (outp < nvals) atomicExch(err, 3);
+}
+
+extern "C" int bg_dict_indices(const void* d_page, int64_t page_len,
+                               int64_t nvals, int32_t has_def,
+                               uint32_t* d_out_idx) {
+  REQUIRE_INIT();
+  int* d_err;
+  HIP_TRY(pool_malloc((void**)&d_err, sizeof(int)));
+  HIP_TRY(hipMemset(d_err, 0, sizeof(int)));
+  hipLaunchKernelGGL(k_dict_indices, dim3(1), dim3(64), 0, 0,
+                     (const uint8_t*)d_page, page_len, nvals, has_def,
+                     d_out_idx, d_err);
+  HIP_TRY(hipGetLastError());
+  int err = 0;
+  HIP_TRY(hipMemcpy(&err, d_err, sizeof(int), hipMemcpyDeviceToHost));
+  (void)pool_release(d_err);
+  if (err == 2)
+    return set_err(BG_ERR_UNSUPPORTED, "bg_dict_indices: nulls (later round)");
+  if (err)
+    return set_err(BG_ERR_INVALID, "bg_dict_indices: malformed index block");
+  return BG_OK;
+}

@@ -183,7 +183,9 @@ class GpuParquetColumnReader:
         # column output buffer + per-page scratch
         out = ctx.alloc(max(total_values * dst_esz, dst_esz))
         pages = []           # (src buf, src_len, scratch buf, ulen)
-        extracts = []        # (scratch, dst_off, num_values, def_prefixed)
+        extracts = []        # PLAIN pages: (scratch, usz, dst_off, nvals)
+        dict_pages = []      # DICT-coded pages: (scratch, usz, dst_off, nvals)
+        dict_info = None     # (scratch, usz, num_dict_values)
         got_values = 0
         while pos < end:
             h, data_pos = parse_page_header(self.raw, pos)
@@ -191,23 +193,36 @@ class GpuParquetColumnReader:
             usz = h[2]
             csz = h[3]
             payload = self.raw[data_pos:data_pos + csz]
-            if ptype == 2:
-                raise RuntimeError("dictionary-encoded chunk: not GPU-decodable yet")
             if ptype == 3:
                 raise RuntimeError("DATA_PAGE_V2: not GPU-decodable yet")
-            if ptype == 0:
-                dph = h.get(5, {})
-                nvals = dph.get(1, 0)
-                enc = dph.get(2, 0)
-                if enc != 0:
-                    raise RuntimeError(f"encoding {enc}: not GPU-decodable yet")
+            if ptype == 2:
+                dsh = h.get(7, {})
+                ndict = dsh.get(1, 0)
                 scratch = ctx.alloc(max(usz, 1))
                 if codec == "SNAPPY":
                     src = ctx.upload(np.frombuffer(payload, dtype=np.uint8))
                     pages.append((src, csz, scratch, usz))
                 else:
                     scratch.upload(np.frombuffer(payload, dtype=np.uint8))
-                extracts.append((scratch, usz, got_values, nvals))
+                dict_info = (scratch, usz, ndict)
+            elif ptype == 0:
+                dph = h.get(5, {})
+                nvals = dph.get(1, 0)
+                enc = dph.get(2, 0)
+                scratch = ctx.alloc(max(usz, 1))
+                if codec == "SNAPPY":
+                    src = ctx.upload(np.frombuffer(payload, dtype=np.uint8))
+                    pages.append((src, csz, scratch, usz))
+                else:
+                    scratch.upload(np.frombuffer(payload, dtype=np.uint8))
+                if enc == 0:  # PLAIN
+                    extracts.append((scratch, usz, got_values, nvals))
+                elif enc in (2, 8):  # PLAIN_DICTIONARY / RLE_DICTIONARY
+                    if dict_info is None:
+                        raise RuntimeError("dict-coded page without dict page")
+                    dict_pages.append((scratch, usz, got_values, nvals))
+                else:
+                    raise RuntimeError(f"encoding {enc}: not GPU-decodable yet")
                 got_values += nvals
             pos = data_pos + csz
 
@@ -231,5 +246,25 @@ class GpuParquetColumnReader:
                 ctypes.c_int64(nvals), ctypes.c_int64(src_esz),
                 1 if max_def > 0 else 0, 1 if flba else 0)
             gpu._check(rc, "bg_page_extract")
+
+        if dict_pages:
+            dscratch, dusz, ndict = dict_info
+            # PLAIN-decode (and endianness-convert) the dictionary itself
+            dict_buf = ctx.alloc(max(ndict * dst_esz, dst_esz))
+            rc = ctx.L.bg_page_extract(
+                dscratch.ptr, ctypes.c_int64(dusz), dict_buf.ptr,
+                ctypes.c_int64(0), ctypes.c_int64(ndict),
+                ctypes.c_int64(src_esz), 0, 1 if flba else 0)
+            gpu._check(rc, "bg_page_extract(dict)")
+            for scratch, usz, dst_off, nvals in dict_pages:
+                idx = ctx.alloc(max(4 * nvals, 4))
+                rc = ctx.L.bg_dict_indices(
+                    scratch.ptr, ctypes.c_int64(usz), ctypes.c_int64(nvals),
+                    1 if max_def > 0 else 0, idx.ptr)
+                gpu._check(rc, "bg_dict_indices")
+                dst_ptr = ctypes.c_void_p(out.ptr.value + dst_off * dst_esz)
+                rc = ctx.L.bg_gather(dict_buf.ptr, ctypes.c_int64(dst_esz),
+                                     idx.ptr, ctypes.c_int64(nvals), dst_ptr)
+                gpu._check(rc, "bg_gather(dict)")
         ctx.synchronize()
         return out, total_values, phys if not flba else "DECIMAL128"

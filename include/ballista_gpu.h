@@ -258,6 +258,13 @@ int bg_page_extract(const void* d_page, int64_t page_len, void* d_out,
                     int64_t dst_byte_off, int64_t nvals, int64_t src_esz,
                     int32_t has_def, int32_t flba_reverse);
 
+/* Parquet dictionary-index expansion (PLAIN_DICTIONARY/RLE_DICTIONARY
+ * data pages): one device thread expands one page's RLE/bit-packed index
+ * block to u32 indices; materialise with bg_gather from the PLAIN-decoded
+ * dictionary. */
+int bg_dict_indices(const void* d_page, int64_t page_len, int64_t nvals,
+                    int32_t has_def, uint32_t* d_out_idx);
+
 /* ---- fused filter+aggregate stages ---- */
 
 /* TPC-H q6 stage 1 (scan+filter+aggregate, approved/q6.txt):

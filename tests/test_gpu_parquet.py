@@ -23,7 +23,7 @@ def ctx():
     c.close()
 
 
-def make_file(tmp_path, compression, n=200_000):
+def make_file(tmp_path, compression, n=200_000, use_dictionary=False):
     rng = np.random.default_rng(3)
     dec_vals = [decimal.Decimal(int(v)) / 100 for v in
                 rng.integers(-10**9, 10**9, size=n)]
@@ -32,17 +32,23 @@ def make_file(tmp_path, compression, n=200_000):
         "d32": pa.array(rng.integers(-10**9, 10**9, size=n, dtype=np.int32)),
         "f64": pa.array(rng.standard_normal(n)),
         "dec": pa.array(dec_vals, type=pa.decimal128(15, 2)),
+        "lowcard": pa.array(rng.integers(0, 50, size=n, dtype=np.int64)),
     })
-    path = str(tmp_path / f"t_{compression}.parquet")
+    path = str(tmp_path / f"t_{compression}_{use_dictionary}.parquet")
     pq.write_table(table, path, compression=compression,
-                   use_dictionary=False, data_page_size=64 * 1024,
-                   write_statistics=False)
+                   use_dictionary=use_dictionary,
+                   data_page_size=64 * 1024, write_statistics=False)
     return path, table
 
 
 @pytest.mark.parametrize("compression", ["snappy", "none"])
-def test_parquet_column_decode(ctx, tmp_path, compression, ):
-    path, table = make_file(tmp_path, compression)
+@pytest.mark.parametrize("use_dictionary", [False, True])
+def test_parquet_column_decode(ctx, tmp_path, compression, use_dictionary):
+    """PLAIN and dictionary-encoded chunks (pyarrow's default is
+    dictionary; low-cardinality columns stay dict-coded, high-cardinality
+    ones fall back to PLAIN mid-chunk — both paths on device)."""
+    path, table = make_file(tmp_path, compression,
+                            use_dictionary=use_dictionary)
     rd = GpuParquetColumnReader(ctx, path)
     n = table.num_rows
 
@@ -65,6 +71,10 @@ def test_parquet_column_decode(ctx, tmp_path, compression, ):
     want = table.column("dec").combine_chunks().buffers()[1]
     want_np = np.frombuffer(want, dtype=np.uint8, count=16 * n)
     assert np.array_equal(got, want_np)
+
+    buf, nv, phys = rd.read_column(0, 4)   # lowcard: dict-coded when enabled
+    assert np.array_equal(buf.download(np.int64, n),
+                          table.column("lowcard").to_numpy())
 
 
 def test_parquet_decode_feeds_q6_kernel(ctx, tmp_path):
