@@ -2191,15 +2191,24 @@ __global__ void k_dict_indices(const uint8_t* page, int64_t page_len,
       shift += 7;
     }
     if (header & 1) {  // bit-packed: (ngroups<<1)|1, groups of 8 values
+      // one group = 8*bw BITS = bw bytes; a value can span up to 5 bytes
+      // (bw<=32), so extract per value from a bounded 8-byte window
       const int64_t groups = (int64_t)(header >> 1);
       for (int64_t g = 0; g < groups && outp < nvals; ++g) {
         if (d + bw > pend) { atomicExch(err, 3); return; }
-        u64 bits = 0;
-        for (int b2 = 0; b2 < bw; ++b2) bits |= (u64)d[b2] << (8 * b2);
-        d += bw;
         const int64_t take = nvals - outp >= 8 ? 8 : nvals - outp;
-        for (int64_t t = 0; t < take; ++t)
-          out_idx[outp + t] = (uint32_t)((bits >> (t * bw)) & vmask);
+        for (int64_t t = 0; t < take; ++t) {
+          const int bitpos = (int)(t * bw);
+          const int bytepos = bitpos >> 3;
+          const int shift2 = bitpos & 7;
+          u64 w = 0;
+          const int avail = bw - bytepos;  // bytes left in this group
+          const int nload = avail < 8 ? avail : 8;
+          for (int b2 = 0; b2 < nload; ++b2)
+            w |= (u64)d[bytepos + b2] << (8 * b2);
+          out_idx[outp + t] = (uint32_t)((w >> shift2) & vmask);
+        }
+        d += bw;
         outp += take;
       }
     } else {  // RLE run: value in ceil(bw/8) LE bytes
