@@ -204,6 +204,11 @@ extern "C" int bg_memcpy_d2h(void* h_dst, const void* d_src, uint64_t bytes) {
   HIP_TRY(hipMemcpy(h_dst, d_src, bytes, hipMemcpyDeviceToHost));
   return BG_OK;
 }
+extern "C" int bg_memcpy_dtod(void* d_dst, const void* d_src, uint64_t bytes) {
+  REQUIRE_INIT();
+  HIP_TRY(hipMemcpy(d_dst, d_src, bytes, hipMemcpyDeviceToDevice));
+  return BG_OK;
+}
 
 // ---------------------------------------------------------------------------
 // device helpers

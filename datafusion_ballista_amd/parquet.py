@@ -171,6 +171,7 @@ class GpuParquetColumnReader:
         else:
             raise RuntimeError(f"physical type {phys} not GPU-decodable yet")
 
+        import ctypes
         start = meta.data_page_offset
         if meta.has_dictionary_page and \
                 meta.dictionary_page_offset is not None and \
@@ -180,91 +181,113 @@ class GpuParquetColumnReader:
         pos = start
         total_values = meta.num_values
 
-        # column output buffer + per-page scratch
-        out = ctx.alloc(max(total_values * dst_esz, dst_esz))
-        pages = []           # (src buf, src_len, scratch buf, ulen)
-        extracts = []        # PLAIN pages: (scratch, usz, dst_off, nvals)
-        dict_pages = []      # DICT-coded pages: (scratch, usz, dst_off, nvals)
-        dict_info = None     # (scratch, usz, num_dict_values)
-        got_values = 0
+        # ONE H2D upload of the whole chunk + ONE decompression scratch;
+        # pages are addressed by offset (no per-page copies)
+        chunk = ctx.upload(np.frombuffer(self.raw[start:end], dtype=np.uint8))
+        headers = []  # (ptype, page_off_in_chunk, csz, usz, nvals, enc, ndict)
+        scratch_total = 0
         while pos < end:
             h, data_pos = parse_page_header(self.raw, pos)
             ptype = h.get(1, 0)
             usz = h[2]
             csz = h[3]
-            payload = self.raw[data_pos:data_pos + csz]
             if ptype == 3:
                 raise RuntimeError("DATA_PAGE_V2: not GPU-decodable yet")
+            nvals = enc = ndict = 0
             if ptype == 2:
-                dsh = h.get(7, {})
-                ndict = dsh.get(1, 0)
-                scratch = ctx.alloc(max(usz, 1))
-                if codec == "SNAPPY":
-                    src = ctx.upload(np.frombuffer(payload, dtype=np.uint8))
-                    pages.append((src, csz, scratch, usz))
-                else:
-                    scratch.upload(np.frombuffer(payload, dtype=np.uint8))
-                dict_info = (scratch, usz, ndict)
+                ndict = h.get(7, {}).get(1, 0)
             elif ptype == 0:
                 dph = h.get(5, {})
                 nvals = dph.get(1, 0)
                 enc = dph.get(2, 0)
-                scratch = ctx.alloc(max(usz, 1))
-                if codec == "SNAPPY":
-                    src = ctx.upload(np.frombuffer(payload, dtype=np.uint8))
-                    pages.append((src, csz, scratch, usz))
-                else:
-                    scratch.upload(np.frombuffer(payload, dtype=np.uint8))
-                if enc == 0:  # PLAIN
-                    extracts.append((scratch, usz, got_values, nvals))
-                elif enc in (2, 8):  # PLAIN_DICTIONARY / RLE_DICTIONARY
-                    if dict_info is None:
-                        raise RuntimeError("dict-coded page without dict page")
-                    dict_pages.append((scratch, usz, got_values, nvals))
-                else:
+                if enc not in (0, 2, 8):
                     raise RuntimeError(f"encoding {enc}: not GPU-decodable yet")
-                got_values += nvals
+            headers.append((ptype, data_pos - start, csz, usz, nvals, enc,
+                            ndict, scratch_total))
+            scratch_total += (usz + 255) & ~255
             pos = data_pos + csz
 
+        scratch = ctx.alloc(max(scratch_total, 256))
+
+        def page_ptr(base, off):
+            return ctypes.c_void_p(base.ptr.value + off)
+
+        out = ctx.alloc(max(total_values * dst_esz, dst_esz))
+        extracts = []        # PLAIN: (scratch_off, usz, dst_off, nvals)
+        dict_pages = []      # DICT-coded: (scratch_off, usz, dst_off, nvals)
+        dict_info = None     # (scratch_off, usz, ndict)
+        got_values = 0
+        snappy_jobs = []
+        for (ptype, poff, csz, usz, nvals, enc, ndict, soff) in headers:
+            if codec == "SNAPPY":
+                snappy_jobs.append((poff, csz, soff, usz))
+            if ptype == 2:
+                dict_info = (soff, usz, ndict)
+            elif ptype == 0:
+                if enc == 0:
+                    extracts.append((soff, usz, got_values, nvals))
+                else:
+                    if dict_info is None:
+                        raise RuntimeError("dict-coded page without dict page")
+                    dict_pages.append((soff, usz, got_values, nvals))
+                got_values += nvals
         if got_values != total_values:
             raise RuntimeError(f"decoded {got_values} != {total_values} values")
-        if pages:
-            lens = ctx.snappy_decompress(
-                [(s, sl, d, dc) for (s, sl, d, dc) in pages])
-            for i, ln in enumerate(lens):
-                if ln != pages[i][3]:
-                    raise RuntimeError(f"snappy page {i} failed ({ln})")
+
+        if codec == "SNAPPY":
+            arr = (gpu.BgSnappyPage * len(snappy_jobs))()
+            for i, (poff, csz, soff, usz) in enumerate(snappy_jobs):
+                arr[i] = gpu.BgSnappyPage(
+                    ctypes.c_void_p(chunk.ptr.value + poff),
+                    ctypes.c_void_p(scratch.ptr.value + soff), csz, usz)
+            lens = np.zeros(len(snappy_jobs), dtype=np.int64)
+            gpu._check(ctx.L.bg_snappy_decompress(
+                arr, ctypes.c_int64(len(snappy_jobs)),
+                lens.ctypes.data_as(ctypes.POINTER(ctypes.c_int64))),
+                "bg_snappy_decompress")
+            for i, (_, _, _, usz) in enumerate(snappy_jobs):
+                if lens[i] != usz:
+                    raise RuntimeError(f"snappy page {i} failed ({lens[i]})")
+        else:
+            # uncompressed: device-to-device copy page payloads into the
+            # aligned scratch slots
+            for (ptype, poff, csz, usz, nvals, enc, ndict, soff) in headers:
+                gpu._check(ctx.L.bg_memcpy_dtod(
+                    ctypes.c_void_p(scratch.ptr.value + soff),
+                    ctypes.c_void_p(chunk.ptr.value + poff),
+                    ctypes.c_uint64(csz)), "bg_memcpy_dtod")
 
         # optional column => 4-byte def-level length prefix + RLE run;
         # required column => values at offset 0
         max_def = self.pf.schema.column(col).max_definition_level
-        import ctypes
-        for scratch, usz, dst_off, nvals in extracts:
+        for soff, usz, dst_off, nvals in extracts:
             rc = ctx.L.bg_page_extract(
-                scratch.ptr, ctypes.c_int64(usz), out.ptr,
+                page_ptr(scratch, soff), ctypes.c_int64(usz), out.ptr,
                 ctypes.c_int64(dst_off * dst_esz),
                 ctypes.c_int64(nvals), ctypes.c_int64(src_esz),
                 1 if max_def > 0 else 0, 1 if flba else 0)
             gpu._check(rc, "bg_page_extract")
 
         if dict_pages:
-            dscratch, dusz, ndict = dict_info
+            dsoff, dusz, ndict = dict_info
             # PLAIN-decode (and endianness-convert) the dictionary itself
             dict_buf = ctx.alloc(max(ndict * dst_esz, dst_esz))
             rc = ctx.L.bg_page_extract(
-                dscratch.ptr, ctypes.c_int64(dusz), dict_buf.ptr,
+                page_ptr(scratch, dsoff), ctypes.c_int64(dusz), dict_buf.ptr,
                 ctypes.c_int64(0), ctypes.c_int64(ndict),
                 ctypes.c_int64(src_esz), 0, 1 if flba else 0)
             gpu._check(rc, "bg_page_extract(dict)")
-            for scratch, usz, dst_off, nvals in dict_pages:
+            for soff, usz, dst_off, nvals in dict_pages:
                 idx = ctx.alloc(max(4 * nvals, 4))
                 rc = ctx.L.bg_dict_indices(
-                    scratch.ptr, ctypes.c_int64(usz), ctypes.c_int64(nvals),
+                    page_ptr(scratch, soff), ctypes.c_int64(usz),
+                    ctypes.c_int64(nvals),
                     1 if max_def > 0 else 0, idx.ptr)
                 gpu._check(rc, "bg_dict_indices")
                 dst_ptr = ctypes.c_void_p(out.ptr.value + dst_off * dst_esz)
                 rc = ctx.L.bg_gather(dict_buf.ptr, ctypes.c_int64(dst_esz),
                                      idx.ptr, ctypes.c_int64(nvals), dst_ptr)
                 gpu._check(rc, "bg_gather(dict)")
+            ctx.synchronize()
         ctx.synchronize()
         return out, total_values, phys if not flba else "DECIMAL128"

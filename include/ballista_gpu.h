@@ -77,6 +77,7 @@ int bg_free(void* d_ptr);
 int bg_memset(void* d_ptr, int value, uint64_t bytes);
 int bg_memcpy_h2d(void* d_dst, const void* h_src, uint64_t bytes);
 int bg_memcpy_d2h(void* h_dst, const void* d_src, uint64_t bytes);
+int bg_memcpy_dtod(void* d_dst, const void* d_src, uint64_t bytes);
 /* release the allocator pool's cached device memory back to HIP */
 int bg_pool_trim(void);
 
