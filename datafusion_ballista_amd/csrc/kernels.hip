@@ -1925,73 +1925,118 @@ struct SnappyPage {
   int64_t dst_cap;
 };
 
+// Wave-cooperative decode: lane 0 walks the (strictly serial) tag stream
+// and broadcasts each element; all 64 lanes execute the copy.  Overlapping
+// LZ77 copies (offset < length) replicate a pattern, which parallelises as
+// dst[i] = window[i % offset].  One wave per page.
 __global__ void k_snappy_decompress(const SnappyPage* pages, int64_t npages,
                                     int64_t* out_lens /* -1 on error */) {
-  for (int64_t p = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; p < npages;
-       p += (int64_t)gridDim.x * blockDim.x) {
-    const uint8_t* s = pages[p].src;
-    const uint8_t* send = s + pages[p].src_len;
-    uint8_t* d = pages[p].dst;
-    // varint32 uncompressed length
-    u64 ulen = 0;
-    int shift = 0;
-    bool ok = true;
-    while (s < send) {
-      const uint8_t b = *s++;
-      ulen |= (u64)(b & 0x7f) << shift;
-      if (!(b & 0x80)) break;
-      shift += 7;
-      if (shift > 32) { ok = false; break; }
-    }
-    if (!ok || (int64_t)ulen > pages[p].dst_cap) {
-      out_lens[p] = -1;
-      continue;
-    }
-    uint8_t* dend = d + ulen;
-    while (ok && s < send && d < dend) {
-      const uint8_t tag = *s++;
-      if ((tag & 3) == 0) {  // literal
-        int64_t len = (tag >> 2) + 1;
-        if (len > 60) {
-          const int nb = (int)len - 60;  // 1..4 length bytes
-          if (s + nb > send) { ok = false; break; }
-          len = 0;
-          for (int i = 0; i < nb; ++i) len |= (int64_t)s[i] << (8 * i);
-          len += 1;
-          s += nb;
-        }
-        if (s + len > send || d + len > dend) { ok = false; break; }
-        for (int64_t i = 0; i < len; ++i) d[i] = s[i];
-        d += len;
-        s += len;
-      } else {
-        int64_t len, off;
-        if ((tag & 3) == 1) {  // copy with 1-byte offset
-          if (s >= send) { ok = false; break; }
-          len = ((tag >> 2) & 7) + 4;
-          off = ((int64_t)(tag >> 5) << 8) | *s++;
-        } else if ((tag & 3) == 2) {  // 2-byte offset
-          if (s + 2 > send) { ok = false; break; }
-          len = (tag >> 2) + 1;
-          off = (int64_t)s[0] | ((int64_t)s[1] << 8);
-          s += 2;
-        } else {  // 4-byte offset
-          if (s + 4 > send) { ok = false; break; }
-          len = (tag >> 2) + 1;
-          off = (int64_t)s[0] | ((int64_t)s[1] << 8) |
-                ((int64_t)s[2] << 16) | ((int64_t)s[3] << 24);
-          s += 4;
-        }
-        if (off == 0 || off > d - pages[p].dst || d + len > dend) {
-          ok = false;
-          break;
-        }
-        const uint8_t* cs = d - off;  // may overlap forward (RLE-style)
-        for (int64_t i = 0; i < len; ++i) d[i] = cs[i];
-        d += len;
+  const int64_t wave_global =
+      ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) / BG_WAVE;
+  const int64_t nwaves = ((int64_t)gridDim.x * blockDim.x) / BG_WAVE;
+  const int lane = lane_id();
+  for (int64_t p = wave_global; p < npages; p += nwaves) {
+    const uint8_t* base_s = pages[p].src;
+    uint8_t* base_d = pages[p].dst;
+    const int64_t src_len = pages[p].src_len;
+    int64_t si = 0, di = 0;
+    int64_t ulen = 0;
+    int ok = 1;
+    if (lane == 0) {
+      int shift = 0;
+      while (si < src_len) {
+        const uint8_t b = base_s[si++];
+        ulen |= (int64_t)(b & 0x7f) << shift;
+        if (!(b & 0x80)) break;
+        shift += 7;
+        if (shift > 32) { ok = 0; break; }
       }
+      if (ulen > pages[p].dst_cap) ok = 0;
     }
-    out_lens[p] = (ok && d == dend) ? (int64_t)ulen : -1;
+    ok = (int)__shfl(ok, 0, BG_WAVE);
+    ulen = __shfl((long long)ulen, 0, BG_WAVE);
+    si = __shfl((long long)si, 0, BG_WAVE);
+    while (ok) {
+      int64_t lit_len = 0, cp_len = 0, cp_off = 0, nsi = si;
+      if (lane == 0 && si < src_len && di < ulen) {
+        const uint8_t tag = base_s[nsi++];
+        if ((tag & 3) == 0) {
+          int64_t len = (tag >> 2) + 1;
+          if (len > 60) {
+            const int nb = (int)len - 60;
+            if (nsi + nb > src_len) { ok = 0; }
+            else {
+              len = 0;
+              for (int i = 0; i < nb; ++i)
+                len |= (int64_t)base_s[nsi + i] << (8 * i);
+              len += 1;
+              nsi += nb;
+            }
+          }
+          if (ok) {
+            if (nsi + len > src_len || di + len > ulen) ok = 0;
+            else lit_len = len;
+          }
+        } else {
+          int64_t len = 0, off = 0;
+          if ((tag & 3) == 1) {
+            if (nsi >= src_len) ok = 0;
+            else {
+              len = ((tag >> 2) & 7) + 4;
+              off = ((int64_t)(tag >> 5) << 8) | base_s[nsi];
+              nsi += 1;
+            }
+          } else if ((tag & 3) == 2) {
+            if (nsi + 2 > src_len) ok = 0;
+            else {
+              len = (tag >> 2) + 1;
+              off = (int64_t)base_s[nsi] | ((int64_t)base_s[nsi + 1] << 8);
+              nsi += 2;
+            }
+          } else {
+            if (nsi + 4 > src_len) ok = 0;
+            else {
+              len = (tag >> 2) + 1;
+              off = (int64_t)base_s[nsi] | ((int64_t)base_s[nsi + 1] << 8) |
+                    ((int64_t)base_s[nsi + 2] << 16) |
+                    ((int64_t)base_s[nsi + 3] << 24);
+              nsi += 4;
+            }
+          }
+          if (ok) {
+            if (off == 0 || off > di || di + len > ulen) ok = 0;
+            else { cp_len = len; cp_off = off; }
+          }
+        }
+      }
+      ok = (int)__shfl(ok, 0, BG_WAVE);
+      if (!ok) break;
+      lit_len = __shfl((long long)lit_len, 0, BG_WAVE);
+      cp_len = __shfl((long long)cp_len, 0, BG_WAVE);
+      cp_off = __shfl((long long)cp_off, 0, BG_WAVE);
+      nsi = __shfl((long long)nsi, 0, BG_WAVE);
+      if (lit_len == 0 && cp_len == 0) break;  // end of stream
+      if (lit_len) {
+        const uint8_t* src = base_s + nsi;
+        uint8_t* dst = base_d + di;
+        for (int64_t i = lane; i < lit_len; i += BG_WAVE) dst[i] = src[i];
+        si = nsi + lit_len;
+        di += lit_len;
+      } else {
+        const uint8_t* win = base_d + di - cp_off;
+        uint8_t* dst = base_d + di;
+        if (cp_off >= cp_len) {
+          for (int64_t i = lane; i < cp_len; i += BG_WAVE) dst[i] = win[i];
+        } else {
+          for (int64_t i = lane; i < cp_len; i += BG_WAVE)
+            dst[i] = win[i % cp_off];
+        }
+        si = nsi;
+        di += cp_len;
+      }
+      if (si >= src_len || di >= ulen) break;
+    }
+    if (lane == 0) out_lens[p] = (ok && di == ulen) ? ulen : -1;
   }
 }
 
@@ -2007,7 +2052,9 @@ extern "C" int bg_snappy_decompress(const void* h_pages, int64_t npages,
   HIP_TRY(pool_malloc((void**)&d_lens, sizeof(int64_t) * (npages ? npages : 1)));
   HIP_TRY(hipMemcpy(d_pages, h_pages, sizeof(SnappyPage) * npages,
                     hipMemcpyHostToDevice));
-  int blocks = (int)bg_imin64((npages + BG_BLOCK - 1) / BG_BLOCK,
+  // one WAVE per page
+  const int waves_per_block = BG_BLOCK / BG_WAVE;
+  int blocks = (int)bg_imin64((npages + waves_per_block - 1) / waves_per_block,
                               BG_MAX_BLOCKS);
   if (blocks == 0) blocks = 1;
   hipLaunchKernelGGL(k_snappy_decompress, dim3(blocks), dim3(BG_BLOCK), 0, 0,
