@@ -2180,102 +2180,151 @@ extern "C" int bg_page_extract(const void* d_page, int64_t page_len,
 // format; pages decode concurrently); the column is then materialised with
 // the existing bg_gather from the (PLAIN-decoded) dictionary.
 // ---------------------------------------------------------------------------
+// Wave-cooperative index expansion: lane 0 parses the serial headers
+// (def levels + hybrid run headers) and broadcasts; all 64 lanes expand
+// bit-packed groups (one group of 8 per lane) and RLE fills in parallel.
 __global__ void k_dict_indices(const uint8_t* page, int64_t page_len,
                                int64_t nvals, int has_def, uint32_t* out_idx,
                                int* err) {
-  if (blockIdx.x != 0 || threadIdx.x != 0) return;
-  const uint8_t* d = page;
-  const uint8_t* pend = page + page_len;
-  if (has_def) {
-    if (d + 4 > pend) { atomicExch(err, 2); return; }
-    const uint32_t dlen = (uint32_t)d[0] | ((uint32_t)d[1] << 8) |
-                          ((uint32_t)d[2] << 16) | ((uint32_t)d[3] << 24);
-    // validate all-1 def levels exactly like k_page_extract
-    const uint8_t* dl = d + 4;
-    const uint8_t* dlend = dl + dlen;
-    if (dlend > pend) { atomicExch(err, 2); return; }
-    int64_t seen = 0;
-    while (dl < dlend && seen < nvals) {
-      u64 header = 0;
+  if (blockIdx.x != 0 || threadIdx.x >= BG_WAVE) return;
+  const int lane = (int)threadIdx.x;
+  int64_t doff = 0;
+  int ok = 1;
+  int bw = 0;
+  if (lane == 0) {
+    const uint8_t* d = page;
+    const uint8_t* pend = page + page_len;
+    if (has_def) {
+      if (d + 4 > pend) ok = 0;
+      else {
+        const uint32_t dlen = (uint32_t)d[0] | ((uint32_t)d[1] << 8) |
+                              ((uint32_t)d[2] << 16) | ((uint32_t)d[3] << 24);
+        const uint8_t* dl = d + 4;
+        const uint8_t* dlend = dl + dlen;
+        if (dlend > pend) ok = 0;
+        int64_t seen = 0;
+        while (ok && dl < dlend && seen < nvals) {
+          u64 header = 0;
+          int shift = 0;
+          while (dl < dlend) {
+            const uint8_t b = *dl++;
+            header |= (u64)(b & 0x7f) << shift;
+            if (!(b & 0x80)) break;
+            shift += 7;
+          }
+          if (header & 1) {
+            const int64_t groups = (int64_t)(header >> 1);
+            for (int64_t g = 0; g < groups && ok; ++g) {
+              if (dl >= dlend) { ok = 0; break; }
+              const uint8_t byte = *dl++;
+              const int64_t take = nvals - seen >= 8 ? 8 : nvals - seen;
+              for (int64_t t = 0; t < take; ++t)
+                if (!((byte >> t) & 1)) ok = 0;
+              seen += take;
+            }
+          } else {
+            const int64_t run = (int64_t)(header >> 1);
+            if (dl >= dlend || *dl++ != 1) ok = 0;
+            seen += run;
+          }
+        }
+        if (seen < nvals) ok = 0;
+        doff = 4 + dlen;
+      }
+    }
+    if (ok) {
+      if (doff >= page_len) ok = 0;
+      else {
+        bw = page[doff];
+        doff += 1;
+        if (bw < 0 || bw > 32) ok = 0;
+      }
+    }
+  }
+  ok = (int)__shfl(ok, 0, BG_WAVE);
+  bw = (int)__shfl(bw, 0, BG_WAVE);
+  doff = __shfl((long long)doff, 0, BG_WAVE);
+  if (!ok) {
+    if (lane == 0) atomicExch(err, 2);
+    return;
+  }
+  if (bw == 0) {
+    for (int64_t t = lane; t < nvals; t += BG_WAVE) out_idx[t] = 0;
+    return;
+  }
+  const u64 vmask = (bw == 32) ? 0xffffffffull : ((1ull << bw) - 1);
+  int64_t outp = 0;
+  int64_t si = doff;
+  while (outp < nvals) {
+    // lane 0 parses one run header
+    u64 header = 0;
+    int hok = 1;
+    int64_t nsi = si;
+    u64 rle_val = 0;
+    if (lane == 0) {
+      if (nsi >= page_len) hok = 0;
       int shift = 0;
-      while (dl < dlend) {
-        const uint8_t b = *dl++;
+      while (hok && nsi < page_len) {
+        const uint8_t b = page[nsi++];
         header |= (u64)(b & 0x7f) << shift;
         if (!(b & 0x80)) break;
         shift += 7;
       }
-      if (header & 1) {
-        const int64_t groups = (int64_t)(header >> 1);
-        for (int64_t g = 0; g < groups; ++g) {
-          if (dl >= dlend) { atomicExch(err, 2); return; }
-          const uint8_t byte = *dl++;
-          const int64_t take = nvals - seen >= 8 ? 8 : nvals - seen;
-          for (int64_t t = 0; t < take; ++t)
-            if (!((byte >> t) & 1)) { atomicExch(err, 2); return; }
-          seen += take;
+      if (hok && !(header & 1)) {
+        const int nb = (bw + 7) / 8;
+        if (nsi + nb > page_len) hok = 0;
+        else {
+          for (int b2 = 0; b2 < nb; ++b2)
+            rle_val |= (u64)page[nsi + b2] << (8 * b2);
+          nsi += nb;
         }
-      } else {
-        const int64_t run = (int64_t)(header >> 1);
-        if (dl >= dlend) { atomicExch(err, 2); return; }
-        if (*dl++ != 1) { atomicExch(err, 2); return; }
-        seen += run;
       }
     }
-    if (seen < nvals) { atomicExch(err, 2); return; }
-    d += 4 + dlen;
-  }
-  if (d >= pend) { atomicExch(err, 3); return; }
-  const int bw = *d++;  // index bit width (0..32)
-  if (bw < 0 || bw > 32) { atomicExch(err, 3); return; }
-  int64_t outp = 0;
-  if (bw == 0) {  // every index is 0
-    for (; outp < nvals; ++outp) out_idx[outp] = 0;
-    return;
-  }
-  const u64 vmask = (bw == 32) ? 0xffffffffull : ((1ull << bw) - 1);
-  while (d < pend && outp < nvals) {
-    u64 header = 0;
-    int shift = 0;
-    while (d < pend) {
-      const uint8_t b = *d++;
-      header |= (u64)(b & 0x7f) << shift;
-      if (!(b & 0x80)) break;
-      shift += 7;
-    }
-    if (header & 1) {  // bit-packed: (ngroups<<1)|1, groups of 8 values
-      // one group = 8*bw BITS = bw bytes; a value can span up to 5 bytes
-      // (bw<=32), so extract per value from a bounded 8-byte window
+    hok = (int)__shfl(hok, 0, BG_WAVE);
+    if (!hok) { if (lane == 0) atomicExch(err, 3); return; }
+    header = (u64)__shfl((long long)header, 0, BG_WAVE);
+    nsi = __shfl((long long)nsi, 0, BG_WAVE);
+    rle_val = (u64)__shfl((long long)rle_val, 0, BG_WAVE);
+    if (header & 1) {  // bit-packed: ngroups groups of 8, bw bytes each
       const int64_t groups = (int64_t)(header >> 1);
-      for (int64_t g = 0; g < groups && outp < nvals; ++g) {
-        if (d + bw > pend) { atomicExch(err, 3); return; }
-        const int64_t take = nvals - outp >= 8 ? 8 : nvals - outp;
-        for (int64_t t = 0; t < take; ++t) {
-          const int bitpos = (int)(t * bw);
-          const int bytepos = bitpos >> 3;
-          const int shift2 = bitpos & 7;
-          u64 w = 0;
-          const int avail = bw - bytepos;  // bytes left in this group
-          const int nload = avail < 8 ? avail : 8;
-          for (int b2 = 0; b2 < nload; ++b2)
-            w |= (u64)d[bytepos + b2] << (8 * b2);
-          out_idx[outp + t] = (uint32_t)((w >> shift2) & vmask);
-        }
-        d += bw;
-        outp += take;
+      if (nsi + groups * bw > page_len) {
+        if (lane == 0) atomicExch(err, 3);
+        return;
       }
-    } else {  // RLE run: value in ceil(bw/8) LE bytes
+      // lane L expands group (gb + L)
+      for (int64_t gb = 0; gb < groups; gb += BG_WAVE) {
+        const int64_t g = gb + lane;
+        if (g < groups) {
+          const uint8_t* gp = page + nsi + g * bw;
+          const int64_t base_out = outp + g * 8;
+          const int64_t take = nvals - base_out >= 8
+                                   ? 8
+                                   : (nvals > base_out ? nvals - base_out : 0);
+          for (int64_t t = 0; t < take; ++t) {
+            const int bitpos = (int)(t * bw);
+            const int bytepos = bitpos >> 3;
+            const int shift2 = bitpos & 7;
+            u64 w = 0;
+            const int avail = bw - bytepos;
+            const int nload = avail < 8 ? avail : 8;
+            for (int b2 = 0; b2 < nload; ++b2)
+              w |= (u64)gp[bytepos + b2] << (8 * b2);
+            out_idx[base_out + t] = (uint32_t)((w >> shift2) & vmask);
+          }
+        }
+      }
+      const int64_t produced = groups * 8;
+      outp += produced < nvals - outp ? produced : nvals - outp;
+      si = nsi + groups * bw;
+    } else {  // RLE run: parallel fill
       const int64_t run = (int64_t)(header >> 1);
-      const int nb = (bw + 7) / 8;
-      if (d + nb > pend) { atomicExch(err, 3); return; }
-      u64 v = 0;
-      for (int b2 = 0; b2 < nb; ++b2) v |= (u64)d[b2] << (8 * b2);
-      d += nb;
       const int64_t take = run <= nvals - outp ? run : nvals - outp;
-      for (int64_t t = 0; t < take; ++t) out_idx[outp + t] = (uint32_t)v;
+      for (int64_t t = lane; t < take; t += BG_WAVE)
+        out_idx[outp + t] = (uint32_t)rle_val;
       outp += take;
+      si = nsi;
     }
   }
-  if (outp < nvals) atomicExch(err, 3);
 }
 
 extern "C" int bg_dict_indices(const void* d_page, int64_t page_len,
