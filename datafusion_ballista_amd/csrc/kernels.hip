@@ -2075,10 +2075,12 @@ extern "C" int bg_snappy_decompress(const void* h_pages, int64_t npages,
 // value bytes to the column buffer; FLBA(16) decimals (parquet big-endian)
 // are byte-reversed to Arrow little-endian in flight.
 // ---------------------------------------------------------------------------
-__global__ void k_page_extract(const uint8_t* page, int64_t page_len,
-                               uint8_t* out, int64_t nvals, int64_t src_esz,
-                               int has_def, int flba_reverse, int* err) {
+__device__ void k_page_extract_body(const uint8_t* page, int64_t page_len,
+                                    uint8_t* out, int64_t nvals,
+                                    int64_t src_esz, int has_def,
+                                    int flba_reverse, int* err) {
   __shared__ int64_t s_voff;
+  __syncthreads();
   if (threadIdx.x == 0) {
     int64_t voff = 0;
     int ok = 1;
@@ -2129,15 +2131,13 @@ __global__ void k_page_extract(const uint8_t* page, int64_t page_len,
   if (voff < 0) return;
   if (!flba_reverse) {
     const int64_t nbytes = nvals * src_esz;
-    for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-         i < nbytes; i += (int64_t)gridDim.x * blockDim.x)
+    for (int64_t i = threadIdx.x; i < nbytes; i += blockDim.x)
       out[i] = page[voff + i];
   } else {
     // FLBA(src_esz) big-endian two's complement -> 16-byte LE decimal128
     // with sign extension (parquet stores decimals at the minimal width
     // for the precision, e.g. 7 bytes for Decimal(15,2))
-    for (int64_t v = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-         v < nvals; v += (int64_t)gridDim.x * blockDim.x) {
+    for (int64_t v = threadIdx.x; v < nvals; v += blockDim.x) {
       const uint8_t* src = page + voff + v * src_esz;
       uint8_t* dst = out + v * 16;
       const uint8_t sign = (src[0] & 0x80) ? 0xff : 0x00;
@@ -2145,6 +2145,14 @@ __global__ void k_page_extract(const uint8_t* page, int64_t page_len,
         dst[b] = (b < src_esz) ? src[src_esz - 1 - b] : sign;
     }
   }
+}
+
+__global__ void k_page_extract(const uint8_t* page, int64_t page_len,
+                               uint8_t* out, int64_t nvals, int64_t src_esz,
+                               int has_def, int flba_reverse, int* err) {
+  if (blockIdx.x != 0) return;
+  k_page_extract_body(page, page_len, out, nvals, src_esz, has_def,
+                      flba_reverse, err);
 }
 
 extern "C" int bg_page_extract(const void* d_page, int64_t page_len,
@@ -2183,11 +2191,10 @@ extern "C" int bg_page_extract(const void* d_page, int64_t page_len,
 // Wave-cooperative index expansion: lane 0 parses the serial headers
 // (def levels + hybrid run headers) and broadcasts; all 64 lanes expand
 // bit-packed groups (one group of 8 per lane) and RLE fills in parallel.
-__global__ void k_dict_indices(const uint8_t* page, int64_t page_len,
-                               int64_t nvals, int has_def, uint32_t* out_idx,
-                               int* err) {
-  if (blockIdx.x != 0 || threadIdx.x >= BG_WAVE) return;
-  const int lane = (int)threadIdx.x;
+__device__ void k_dict_indices_body(const uint8_t* page, int64_t page_len,
+                                    int64_t nvals, int has_def,
+                                    uint32_t* out_idx, int* err) {
+  const int lane = lane_id();
   int64_t doff = 0;
   int ok = 1;
   int bw = 0;
@@ -2327,6 +2334,13 @@ __global__ void k_dict_indices(const uint8_t* page, int64_t page_len,
   }
 }
 
+__global__ void k_dict_indices(const uint8_t* page, int64_t page_len,
+                               int64_t nvals, int has_def, uint32_t* out_idx,
+                               int* err) {
+  if (blockIdx.x != 0 || threadIdx.x >= BG_WAVE) return;
+  k_dict_indices_body(page, page_len, nvals, has_def, out_idx, err);
+}
+
 extern "C" int bg_dict_indices(const void* d_page, int64_t page_len,
                                int64_t nvals, int32_t has_def,
                                uint32_t* d_out_idx) {
@@ -2345,5 +2359,101 @@ extern "C" int bg_dict_indices(const void* d_page, int64_t page_len,
     return set_err(BG_ERR_UNSUPPORTED, "bg_dict_indices: nulls (later round)");
   if (err)
     return set_err(BG_ERR_INVALID, "bg_dict_indices: malformed index block");
+  return BG_OK;
+}
+
+// ---------------------------------------------------------------------------
+// Batched page ops: the per-page host loop (launch + sync error check per
+// page) starves the GPU when chunks hold few pages — these take the whole
+// column's page list in one call.
+// ---------------------------------------------------------------------------
+struct PageExtractJob {
+  const uint8_t* page;
+  uint8_t* out;
+  int64_t page_len;
+  int64_t nvals;
+  int64_t src_esz;
+  int32_t has_def;
+  int32_t flba_reverse;
+};
+
+__global__ void k_page_extract_batch(const PageExtractJob* jobs, int64_t njobs,
+                                     int* err) {
+  for (int64_t j = blockIdx.x; j < njobs; j += gridDim.x) {
+    const PageExtractJob job = jobs[j];
+    k_page_extract_body(job.page, job.page_len, job.out, job.nvals,
+                        job.src_esz, job.has_def, job.flba_reverse, err);
+  }
+}
+
+extern "C" int bg_page_extract_batch(const void* h_jobs, int64_t njobs) {
+  REQUIRE_INIT();
+  PageExtractJob* d_jobs;
+  int* d_err;
+  HIP_TRY(pool_malloc((void**)&d_jobs, sizeof(PageExtractJob) * (njobs ? njobs : 1)));
+  HIP_TRY(pool_malloc((void**)&d_err, sizeof(int)));
+  HIP_TRY(hipMemset(d_err, 0, sizeof(int)));
+  HIP_TRY(hipMemcpy(d_jobs, h_jobs, sizeof(PageExtractJob) * njobs,
+                    hipMemcpyHostToDevice));
+  int blocks = (int)bg_imin64(njobs, BG_MAX_BLOCKS);
+  if (blocks == 0) blocks = 1;
+  hipLaunchKernelGGL(k_page_extract_batch, dim3(blocks), dim3(BG_BLOCK), 0, 0,
+                     d_jobs, njobs, d_err);
+  HIP_TRY(hipGetLastError());
+  int err = 0;
+  HIP_TRY(hipMemcpy(&err, d_err, sizeof(int), hipMemcpyDeviceToHost));
+  (void)pool_release(d_jobs);
+  (void)pool_release(d_err);
+  if (err)
+    return set_err(BG_ERR_UNSUPPORTED,
+                   "bg_page_extract_batch: nulls or malformed page");
+  return BG_OK;
+}
+
+struct DictIndicesJob {
+  const uint8_t* page;
+  uint32_t* out_idx;
+  int64_t page_len;
+  int64_t nvals;
+  int32_t has_def;
+  int32_t _pad;
+};
+
+__global__ void k_dict_indices_batch(const DictIndicesJob* jobs, int64_t njobs,
+                                     int* err) {
+  const int64_t wave_global =
+      ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) / BG_WAVE;
+  const int64_t nwaves = ((int64_t)gridDim.x * blockDim.x) / BG_WAVE;
+  for (int64_t j = wave_global; j < njobs; j += nwaves) {
+    const DictIndicesJob job = jobs[j];
+    k_dict_indices_body(job.page, job.page_len, job.nvals, job.has_def,
+                        job.out_idx, err);
+  }
+}
+
+extern "C" int bg_dict_indices_batch(const void* h_jobs, int64_t njobs) {
+  REQUIRE_INIT();
+  DictIndicesJob* d_jobs;
+  int* d_err;
+  HIP_TRY(pool_malloc((void**)&d_jobs, sizeof(DictIndicesJob) * (njobs ? njobs : 1)));
+  HIP_TRY(pool_malloc((void**)&d_err, sizeof(int)));
+  HIP_TRY(hipMemset(d_err, 0, sizeof(int)));
+  HIP_TRY(hipMemcpy(d_jobs, h_jobs, sizeof(DictIndicesJob) * njobs,
+                    hipMemcpyHostToDevice));
+  const int waves_per_block = BG_BLOCK / BG_WAVE;
+  int blocks = (int)bg_imin64((njobs + waves_per_block - 1) / waves_per_block,
+                              BG_MAX_BLOCKS);
+  if (blocks == 0) blocks = 1;
+  hipLaunchKernelGGL(k_dict_indices_batch, dim3(blocks), dim3(BG_BLOCK), 0, 0,
+                     d_jobs, njobs, d_err);
+  HIP_TRY(hipGetLastError());
+  int err = 0;
+  HIP_TRY(hipMemcpy(&err, d_err, sizeof(int), hipMemcpyDeviceToHost));
+  (void)pool_release(d_jobs);
+  (void)pool_release(d_err);
+  if (err == 2)
+    return set_err(BG_ERR_UNSUPPORTED, "bg_dict_indices_batch: nulls");
+  if (err)
+    return set_err(BG_ERR_INVALID, "bg_dict_indices_batch: malformed block");
   return BG_OK;
 }

@@ -436,3 +436,16 @@ def _ctx_snappy_decompress(self, pages):
 
 
 GpuStageContext.snappy_decompress = _ctx_snappy_decompress
+
+
+class BgPageExtractJob(ctypes.Structure):
+    _fields_ = [("d_page", ctypes.c_void_p), ("d_out", ctypes.c_void_p),
+                ("page_len", ctypes.c_int64), ("nvals", ctypes.c_int64),
+                ("src_esz", ctypes.c_int64), ("has_def", ctypes.c_int32),
+                ("flba_reverse", ctypes.c_int32)]
+
+
+class BgDictIndicesJob(ctypes.Structure):
+    _fields_ = [("d_page", ctypes.c_void_p), ("d_out_idx", ctypes.c_void_p),
+                ("page_len", ctypes.c_int64), ("nvals", ctypes.c_int64),
+                ("has_def", ctypes.c_int32), ("_pad", ctypes.c_int32)]

@@ -148,6 +148,8 @@ class GpuParquetColumnReader:
         self.path = path
         self.pf = pq.ParquetFile(path)
         self.raw = open(path, "rb").read()
+        # whole file resident once; every page is addressed by file offset
+        self._file_buf = ctx.upload(np.frombuffer(self.raw, dtype=np.uint8))
 
     def read_column(self, rg: int, col: int):
         """-> (DeviceBuffer of raw values, num_values, physical_type).
@@ -181,9 +183,9 @@ class GpuParquetColumnReader:
         pos = start
         total_values = meta.num_values
 
-        # ONE H2D upload of the whole chunk + ONE decompression scratch;
-        # pages are addressed by offset (no per-page copies)
-        chunk = ctx.upload(np.frombuffer(self.raw[start:end], dtype=np.uint8))
+        # pages live in the resident file buffer; ONE decompression scratch
+        chunk = self._file_buf
+        chunk_base = 0  # offsets below are absolute file offsets
         headers = []  # (ptype, page_off_in_chunk, csz, usz, nvals, enc, ndict)
         scratch_total = 0
         while pos < end:
@@ -202,7 +204,7 @@ class GpuParquetColumnReader:
                 enc = dph.get(2, 0)
                 if enc not in (0, 2, 8):
                     raise RuntimeError(f"encoding {enc}: not GPU-decodable yet")
-            headers.append((ptype, data_pos - start, csz, usz, nvals, enc,
+            headers.append((ptype, data_pos, csz, usz, nvals, enc,
                             ndict, scratch_total))
             scratch_total += (usz + 255) & ~255
             pos = data_pos + csz
@@ -258,15 +260,18 @@ class GpuParquetColumnReader:
                     ctypes.c_uint64(csz)), "bg_memcpy_dtod")
 
         # optional column => 4-byte def-level length prefix + RLE run;
-        # required column => values at offset 0
+        # required column => values at offset 0  (ONE batched launch)
         max_def = self.pf.schema.column(col).max_definition_level
-        for soff, usz, dst_off, nvals in extracts:
-            rc = ctx.L.bg_page_extract(
-                page_ptr(scratch, soff), ctypes.c_int64(usz), out.ptr,
-                ctypes.c_int64(dst_off * dst_esz),
-                ctypes.c_int64(nvals), ctypes.c_int64(src_esz),
-                1 if max_def > 0 else 0, 1 if flba else 0)
-            gpu._check(rc, "bg_page_extract")
+        if extracts:
+            jobs = (gpu.BgPageExtractJob * len(extracts))()
+            for i, (soff, usz, dst_off, nvals) in enumerate(extracts):
+                jobs[i] = gpu.BgPageExtractJob(
+                    page_ptr(scratch, soff).value,
+                    ctypes.c_void_p(out.ptr.value + dst_off * dst_esz).value,
+                    usz, nvals, src_esz,
+                    1 if max_def > 0 else 0, 1 if flba else 0)
+            gpu._check(ctx.L.bg_page_extract_batch(
+                jobs, ctypes.c_int64(len(extracts))), "bg_page_extract_batch")
 
         if dict_pages:
             dsoff, dusz, ndict = dict_info
@@ -277,17 +282,33 @@ class GpuParquetColumnReader:
                 ctypes.c_int64(0), ctypes.c_int64(ndict),
                 ctypes.c_int64(src_esz), 0, 1 if flba else 0)
             gpu._check(rc, "bg_page_extract(dict)")
-            for soff, usz, dst_off, nvals in dict_pages:
-                idx = ctx.alloc(max(4 * nvals, 4))
-                rc = ctx.L.bg_dict_indices(
-                    page_ptr(scratch, soff), ctypes.c_int64(usz),
-                    ctypes.c_int64(nvals),
-                    1 if max_def > 0 else 0, idx.ptr)
-                gpu._check(rc, "bg_dict_indices")
-                dst_ptr = ctypes.c_void_p(out.ptr.value + dst_off * dst_esz)
-                rc = ctx.L.bg_gather(dict_buf.ptr, ctypes.c_int64(dst_esz),
-                                     idx.ptr, ctypes.c_int64(nvals), dst_ptr)
-                gpu._check(rc, "bg_gather(dict)")
-            ctx.synchronize()
+            # batched index expansion into ONE contiguous index buffer,
+            # then ONE gather for all dict-coded rows of this chunk
+            nidx_total = sum(nv for (_, _, _, nv) in dict_pages)
+            idx = ctx.alloc(max(4 * nidx_total, 4))
+            jobs = (gpu.BgDictIndicesJob * len(dict_pages))()
+            run = 0
+            first_dst = dict_pages[0][2]
+            # the one-gather fast path needs the dict-coded pages to cover a
+            # contiguous dst range (they do: the PLAIN fallback only ever
+            # follows the dict-coded prefix)
+            expect = first_dst
+            for (_, _, dst_off, nvals) in dict_pages:
+                if dst_off != expect:
+                    raise RuntimeError("non-contiguous dict-coded pages")
+                expect = dst_off + nvals
+            for i, (soff, usz, dst_off, nvals) in enumerate(dict_pages):
+                jobs[i] = gpu.BgDictIndicesJob(
+                    page_ptr(scratch, soff).value,
+                    ctypes.c_void_p(idx.ptr.value + 4 * run).value,
+                    usz, nvals, 1 if max_def > 0 else 0, 0)
+                run += nvals
+            gpu._check(ctx.L.bg_dict_indices_batch(
+                jobs, ctypes.c_int64(len(dict_pages))),
+                "bg_dict_indices_batch")
+            dst_ptr = ctypes.c_void_p(out.ptr.value + first_dst * dst_esz)
+            gpu._check(ctx.L.bg_gather(
+                dict_buf.ptr, ctypes.c_int64(dst_esz), idx.ptr,
+                ctypes.c_int64(nidx_total), dst_ptr), "bg_gather(dict)")
         ctx.synchronize()
         return out, total_values, phys if not flba else "DECIMAL128"

@@ -266,6 +266,29 @@ int bg_page_extract(const void* d_page, int64_t page_len, void* d_out,
 int bg_dict_indices(const void* d_page, int64_t page_len, int64_t nvals,
                     int32_t has_def, uint32_t* d_out_idx);
 
+/* Batched forms: the whole column's page list in one launch (chunks hold
+ * few pages; per-page host loops starve the GPU). */
+typedef struct {
+  const void* d_page;
+  void* d_out;
+  int64_t page_len;
+  int64_t nvals;
+  int64_t src_esz;
+  int32_t has_def;
+  int32_t flba_reverse;
+} bg_page_extract_job;
+int bg_page_extract_batch(const void* h_jobs, int64_t njobs);
+
+typedef struct {
+  const void* d_page;
+  uint32_t* d_out_idx;
+  int64_t page_len;
+  int64_t nvals;
+  int32_t has_def;
+  int32_t _pad;
+} bg_dict_indices_job;
+int bg_dict_indices_batch(const void* h_jobs, int64_t njobs);
+
 /* ---- fused filter+aggregate stages ---- */
 
 /* TPC-H q6 stage 1 (scan+filter+aggregate, approved/q6.txt):
