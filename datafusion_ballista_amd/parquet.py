@@ -152,10 +152,18 @@ class GpuParquetColumnReader:
         self._file_buf = ctx.upload(np.frombuffer(self.raw, dtype=np.uint8))
 
     def read_column(self, rg: int, col: int):
-        """-> (DeviceBuffer of raw values, num_values, physical_type).
-        FLBA(16) decimals are byte-reversed on device to Arrow LE."""
+        return self.read_column_all(col, rgs=[rg])
+
+    def read_column_all(self, col: int, rgs=None):
+        """Decode a column across row groups in ONE batched pass (one
+        snappy launch over every page of every chunk — chunks alone hold
+        too few pages to fill 256 CUs).
+        -> (DeviceBuffer of raw values, num_values, physical_type)."""
         ctx = self.ctx
-        meta = self.pf.metadata.row_group(rg).column(col)
+        if rgs is None:
+            rgs = range(self.pf.metadata.num_row_groups)
+        rgs = list(rgs)
+        meta = self.pf.metadata.row_group(rgs[0]).column(col)
         codec = meta.compression  # 'SNAPPY' | 'UNCOMPRESSED' | ...
         if codec not in ("SNAPPY", "UNCOMPRESSED"):
             raise RuntimeError(f"codec {codec} not GPU-decodable yet")
@@ -174,40 +182,41 @@ class GpuParquetColumnReader:
             raise RuntimeError(f"physical type {phys} not GPU-decodable yet")
 
         import ctypes
-        start = meta.data_page_offset
-        if meta.has_dictionary_page and \
-                meta.dictionary_page_offset is not None and \
-                meta.dictionary_page_offset < start:
-            start = meta.dictionary_page_offset
-        end = start + meta.total_compressed_size
-        pos = start
-        total_values = meta.num_values
-
-        # pages live in the resident file buffer; ONE decompression scratch
-        chunk = self._file_buf
-        chunk_base = 0  # offsets below are absolute file offsets
-        headers = []  # (ptype, page_off_in_chunk, csz, usz, nvals, enc, ndict)
+        chunk = self._file_buf  # offsets below are absolute file offsets
+        headers = []  # (rg, ptype, file_off, csz, usz, nvals, enc, ndict, soff)
         scratch_total = 0
-        while pos < end:
-            h, data_pos = parse_page_header(self.raw, pos)
-            ptype = h.get(1, 0)
-            usz = h[2]
-            csz = h[3]
-            if ptype == 3:
-                raise RuntimeError("DATA_PAGE_V2: not GPU-decodable yet")
-            nvals = enc = ndict = 0
-            if ptype == 2:
-                ndict = h.get(7, {}).get(1, 0)
-            elif ptype == 0:
-                dph = h.get(5, {})
-                nvals = dph.get(1, 0)
-                enc = dph.get(2, 0)
-                if enc not in (0, 2, 8):
-                    raise RuntimeError(f"encoding {enc}: not GPU-decodable yet")
-            headers.append((ptype, data_pos, csz, usz, nvals, enc,
-                            ndict, scratch_total))
-            scratch_total += (usz + 255) & ~255
-            pos = data_pos + csz
+        total_values = 0
+        for rg in rgs:
+            m = self.pf.metadata.row_group(rg).column(col)
+            start = m.data_page_offset
+            if m.has_dictionary_page and \
+                    m.dictionary_page_offset is not None and \
+                    m.dictionary_page_offset < start:
+                start = m.dictionary_page_offset
+            end = start + m.total_compressed_size
+            pos = start
+            total_values += m.num_values
+            while pos < end:
+                h, data_pos = parse_page_header(self.raw, pos)
+                ptype = h.get(1, 0)
+                usz = h[2]
+                csz = h[3]
+                if ptype == 3:
+                    raise RuntimeError("DATA_PAGE_V2: not GPU-decodable yet")
+                nvals = enc = ndict = 0
+                if ptype == 2:
+                    ndict = h.get(7, {}).get(1, 0)
+                elif ptype == 0:
+                    dph = h.get(5, {})
+                    nvals = dph.get(1, 0)
+                    enc = dph.get(2, 0)
+                    if enc not in (0, 2, 8):
+                        raise RuntimeError(
+                            f"encoding {enc}: not GPU-decodable yet")
+                headers.append((rg, ptype, data_pos, csz, usz, nvals, enc,
+                                ndict, scratch_total))
+                scratch_total += (usz + 255) & ~255
+                pos = data_pos + csz
 
         scratch = ctx.alloc(max(scratch_total, 256))
 
@@ -216,22 +225,23 @@ class GpuParquetColumnReader:
 
         out = ctx.alloc(max(total_values * dst_esz, dst_esz))
         extracts = []        # PLAIN: (scratch_off, usz, dst_off, nvals)
-        dict_pages = []      # DICT-coded: (scratch_off, usz, dst_off, nvals)
-        dict_info = None     # (scratch_off, usz, ndict)
+        dict_runs = {}       # rg -> {"dict": (soff, usz, ndict), "pages": []}
         got_values = 0
         snappy_jobs = []
-        for (ptype, poff, csz, usz, nvals, enc, ndict, soff) in headers:
+        for (rg, ptype, poff, csz, usz, nvals, enc, ndict, soff) in headers:
             if codec == "SNAPPY":
                 snappy_jobs.append((poff, csz, soff, usz))
             if ptype == 2:
-                dict_info = (soff, usz, ndict)
+                dict_runs.setdefault(rg, {"dict": None, "pages": []})
+                dict_runs[rg]["dict"] = (soff, usz, ndict)
             elif ptype == 0:
                 if enc == 0:
                     extracts.append((soff, usz, got_values, nvals))
                 else:
-                    if dict_info is None:
+                    if rg not in dict_runs or dict_runs[rg]["dict"] is None:
                         raise RuntimeError("dict-coded page without dict page")
-                    dict_pages.append((soff, usz, got_values, nvals))
+                    dict_runs[rg]["pages"].append(
+                        (soff, usz, got_values, nvals))
                 got_values += nvals
         if got_values != total_values:
             raise RuntimeError(f"decoded {got_values} != {total_values} values")
@@ -253,7 +263,7 @@ class GpuParquetColumnReader:
         else:
             # uncompressed: device-to-device copy page payloads into the
             # aligned scratch slots
-            for (ptype, poff, csz, usz, nvals, enc, ndict, soff) in headers:
+            for (rg, ptype, poff, csz, usz, nvals, enc, ndict, soff) in headers:
                 gpu._check(ctx.L.bg_memcpy_dtod(
                     ctypes.c_void_p(scratch.ptr.value + soff),
                     ctypes.c_void_p(chunk.ptr.value + poff),
@@ -273,42 +283,55 @@ class GpuParquetColumnReader:
             gpu._check(ctx.L.bg_page_extract_batch(
                 jobs, ctypes.c_int64(len(extracts))), "bg_page_extract_batch")
 
-        if dict_pages:
-            dsoff, dusz, ndict = dict_info
-            # PLAIN-decode (and endianness-convert) the dictionary itself
-            dict_buf = ctx.alloc(max(ndict * dst_esz, dst_esz))
-            rc = ctx.L.bg_page_extract(
-                page_ptr(scratch, dsoff), ctypes.c_int64(dusz), dict_buf.ptr,
-                ctypes.c_int64(0), ctypes.c_int64(ndict),
-                ctypes.c_int64(src_esz), 0, 1 if flba else 0)
-            gpu._check(rc, "bg_page_extract(dict)")
-            # batched index expansion into ONE contiguous index buffer,
-            # then ONE gather for all dict-coded rows of this chunk
-            nidx_total = sum(nv for (_, _, _, nv) in dict_pages)
+        all_dict_pages = [(rg, pg) for rg, d in dict_runs.items()
+                          for pg in d["pages"]]
+        if all_dict_pages:
+            # PLAIN-decode every row group's dictionary (batched)
+            dict_bufs = {}
+            djobs = []
+            for rg, d in dict_runs.items():
+                if not d["pages"]:
+                    continue
+                dsoff, dusz, ndict = d["dict"]
+                dbuf = ctx.alloc(max(ndict * dst_esz, dst_esz))
+                dict_bufs[rg] = dbuf
+                djobs.append(gpu.BgPageExtractJob(
+                    page_ptr(scratch, dsoff).value, dbuf.ptr.value, dusz,
+                    ndict, src_esz, 0, 1 if flba else 0))
+            jarr = (gpu.BgPageExtractJob * len(djobs))(*djobs)
+            gpu._check(ctx.L.bg_page_extract_batch(
+                jarr, ctypes.c_int64(len(djobs))), "bg_page_extract(dicts)")
+            # ONE batched index expansion over every dict-coded page
+            nidx_total = sum(pg[3] for (_, pg) in all_dict_pages)
             idx = ctx.alloc(max(4 * nidx_total, 4))
-            jobs = (gpu.BgDictIndicesJob * len(dict_pages))()
+            jobs = (gpu.BgDictIndicesJob * len(all_dict_pages))()
             run = 0
-            first_dst = dict_pages[0][2]
-            # the one-gather fast path needs the dict-coded pages to cover a
-            # contiguous dst range (they do: the PLAIN fallback only ever
-            # follows the dict-coded prefix)
-            expect = first_dst
-            for (_, _, dst_off, nvals) in dict_pages:
-                if dst_off != expect:
-                    raise RuntimeError("non-contiguous dict-coded pages")
-                expect = dst_off + nvals
-            for i, (soff, usz, dst_off, nvals) in enumerate(dict_pages):
+            gathers = []  # (rg, idx_off, dst_off, nvals) merged per rg below
+            for i, (rg, (soff, usz, dst_off, nvals)) in \
+                    enumerate(all_dict_pages):
                 jobs[i] = gpu.BgDictIndicesJob(
                     page_ptr(scratch, soff).value,
                     ctypes.c_void_p(idx.ptr.value + 4 * run).value,
                     usz, nvals, 1 if max_def > 0 else 0, 0)
+                gathers.append((rg, run, dst_off, nvals))
                 run += nvals
             gpu._check(ctx.L.bg_dict_indices_batch(
-                jobs, ctypes.c_int64(len(dict_pages))),
+                jobs, ctypes.c_int64(len(all_dict_pages))),
                 "bg_dict_indices_batch")
-            dst_ptr = ctypes.c_void_p(out.ptr.value + first_dst * dst_esz)
-            gpu._check(ctx.L.bg_gather(
-                dict_buf.ptr, ctypes.c_int64(dst_esz), idx.ptr,
-                ctypes.c_int64(nidx_total), dst_ptr), "bg_gather(dict)")
+            # merge adjacent pages of one rg into single gathers
+            merged = []
+            for rg, ioff, dst_off, nvals in gathers:
+                if merged and merged[-1][0] == rg and \
+                        merged[-1][2] + merged[-1][3] == dst_off:
+                    merged[-1][3] += nvals
+                else:
+                    merged.append([rg, ioff, dst_off, nvals])
+            for rg, ioff, dst_off, nvals in merged:
+                gpu._check(ctx.L.bg_gather(
+                    dict_bufs[rg].ptr, ctypes.c_int64(dst_esz),
+                    ctypes.c_void_p(idx.ptr.value + 4 * ioff),
+                    ctypes.c_int64(nvals),
+                    ctypes.c_void_p(out.ptr.value + dst_off * dst_esz)),
+                    "bg_gather(dict)")
         ctx.synchronize()
         return out, total_values, phys if not flba else "DECIMAL128"
