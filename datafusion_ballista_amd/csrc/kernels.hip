@@ -2468,3 +2468,88 @@ extern "C" int bg_dict_indices_batch(const void* h_jobs, int64_t njobs) {
     return set_err(BG_ERR_INVALID, "bg_dict_indices_batch: malformed block");
   return BG_OK;
 }
+
+// ---------------------------------------------------------------------------
+// Variable-length (Utf8/Binary) row gather — the take/interleave the
+// sort-shuffle writer needs for string payload columns
+// (PartitionedBatchIterator handles every Arrow type).
+// Two-phase: lengths -> exclusive scan -> wave-per-row byte copy.
+// ---------------------------------------------------------------------------
+__global__ void k_varlen_lens(const int32_t* src_offsets, const uint32_t* idx,
+                              int64_t m, u64* lens) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < m;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    const uint32_t r = idx[i];
+    lens[i] = (u64)(src_offsets[r + 1] - src_offsets[r]);
+  }
+}
+
+__global__ void k_varlen_copy(const uint8_t* src_data,
+                              const int32_t* src_offsets, const uint32_t* idx,
+                              int64_t m, const i64* out_offs,
+                              uint8_t* out_data, int32_t* out_offsets32) {
+  const int64_t wave_global =
+      ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) / BG_WAVE;
+  const int64_t nwaves = ((int64_t)gridDim.x * blockDim.x) / BG_WAVE;
+  const int lane = lane_id();
+  for (int64_t i = wave_global; i < m; i += nwaves) {
+    const uint32_t r = idx[i];
+    const int32_t lo = src_offsets[r];
+    const int32_t len = src_offsets[r + 1] - lo;
+    const i64 dst = out_offs[i];
+    for (int32_t b = lane; b < len; b += BG_WAVE)
+      out_data[dst + b] = src_data[lo + b];
+    if (lane == 0) out_offsets32[i] = (int32_t)dst;
+  }
+}
+
+__global__ void k_varlen_tail(int64_t m, const i64* total,
+                              int32_t* out_offsets32) {
+  if (blockIdx.x == 0 && threadIdx.x == 0) out_offsets32[m] = (int32_t)*total;
+}
+
+/* Gather m variable-length rows: writes Arrow i32 offsets (m+1) and packed
+ * bytes; *out_total_bytes receives the data length (must fit i32 per Arrow
+ * Utf8). */
+extern "C" int bg_gather_varlen(const void* d_src_data,
+                                const int32_t* d_src_offsets,
+                                const uint32_t* d_idx, int64_t m,
+                                int32_t* d_out_offsets /* m+1 */,
+                                void* d_out_data, int64_t out_data_cap,
+                                int64_t* out_total_bytes) {
+  REQUIRE_INIT();
+  u64* d_lens;
+  i64* d_offs;
+  i64* d_total;
+  HIP_TRY(pool_malloc((void**)&d_lens, sizeof(u64) * (m ? m : 1)));
+  HIP_TRY(pool_malloc((void**)&d_offs, sizeof(i64) * (m ? m : 1)));
+  HIP_TRY(pool_malloc((void**)&d_total, sizeof(i64)));
+  int blocks = (int)bg_imin64((m + BG_BLOCK - 1) / BG_BLOCK, BG_MAX_BLOCKS);
+  if (blocks == 0) blocks = 1;
+  hipLaunchKernelGGL(k_varlen_lens, dim3(blocks), dim3(BG_BLOCK), 0, 0,
+                     d_src_offsets, d_idx, m, d_lens);
+  int rc = scan_exclusive_i64(d_lens, m, d_offs, d_total);
+  if (rc != BG_OK) return rc;
+  i64 total = 0;
+  HIP_TRY(hipMemcpy(&total, d_total, sizeof(i64), hipMemcpyDeviceToHost));
+  if (total > out_data_cap)
+    return set_err(BG_ERR_INVALID, "bg_gather_varlen: out_data too small");
+  if (total > 0x7fffffffLL)
+    return set_err(BG_ERR_INVALID,
+                   "bg_gather_varlen: >2GiB Utf8 data (LargeUtf8: later)");
+  const int waves_per_block = BG_BLOCK / BG_WAVE;
+  int cblocks = (int)bg_imin64((m + waves_per_block - 1) / waves_per_block,
+                               BG_MAX_BLOCKS);
+  if (cblocks == 0) cblocks = 1;
+  hipLaunchKernelGGL(k_varlen_copy, dim3(cblocks), dim3(BG_BLOCK), 0, 0,
+                     (const uint8_t*)d_src_data, d_src_offsets, d_idx, m,
+                     d_offs, (uint8_t*)d_out_data, d_out_offsets);
+  hipLaunchKernelGGL(k_varlen_tail, dim3(1), dim3(1), 0, 0, m, d_total,
+                     d_out_offsets);
+  HIP_TRY(hipGetLastError());
+  (void)pool_release(d_lens);
+  (void)pool_release(d_offs);
+  (void)pool_release(d_total);
+  *out_total_bytes = total;
+  return BG_OK;
+}

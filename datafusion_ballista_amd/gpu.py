@@ -449,3 +449,19 @@ class BgDictIndicesJob(ctypes.Structure):
     _fields_ = [("d_page", ctypes.c_void_p), ("d_out_idx", ctypes.c_void_p),
                 ("page_len", ctypes.c_int64), ("nvals", ctypes.c_int64),
                 ("has_def", ctypes.c_int32), ("_pad", ctypes.c_int32)]
+
+
+def _ctx_gather_varlen(self, src_data: "DeviceBuffer", src_offsets: "DeviceBuffer",
+                       idx: "DeviceBuffer", m: int, max_bytes: int):
+    """Variable-length take -> (offsets buf i32[m+1], data buf, total)."""
+    out_offs = self.alloc(max(4 * (m + 1), 8))
+    out_data = self.alloc(max(max_bytes, 1))
+    total = ctypes.c_int64()
+    _check(self.L.bg_gather_varlen(src_data.ptr, src_offsets.ptr, idx.ptr,
+                                   ctypes.c_int64(m), out_offs.ptr,
+                                   out_data.ptr, ctypes.c_int64(max_bytes),
+                                   ctypes.byref(total)), "bg_gather_varlen")
+    return out_offs, out_data, total.value
+
+
+GpuStageContext.gather_varlen = _ctx_gather_varlen

@@ -134,6 +134,14 @@ int bg_mask_to_indices(const uint8_t* d_mask, int64_t n, uint32_t* d_indices,
 int bg_gather(const void* d_src, int64_t elem_size, const uint32_t* d_idx,
               int64_t m, void* d_dst);
 
+/* Variable-length (Utf8/Binary) gather: m rows by index -> Arrow i32
+ * offsets (m+1) + packed bytes (the writer's take/interleave for string
+ * payload columns). */
+int bg_gather_varlen(const void* d_src_data, const int32_t* d_src_offsets,
+                     const uint32_t* d_idx, int64_t m,
+                     int32_t* d_out_offsets, void* d_out_data,
+                     int64_t out_data_cap, int64_t* out_total_bytes);
+
 /* ---- hash repartition (SortShuffleWriterExec device half) ---- */
 
 /* create_hashes restatement over the key columns (bg_ahash.h; parity

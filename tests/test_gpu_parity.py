@@ -718,3 +718,31 @@ def test_snappy_malformed_is_rejected(ctx):
         pages.append((src, len(c), dst, 5000))
     lens = ctx.snappy_decompress(pages)
     assert lens == [-1, -1]
+
+
+def test_gather_varlen_parity(ctx):
+    """Utf8 take (string shuffle payload): offsets + bytes vs pyarrow."""
+    rng = np.random.default_rng(17)
+    n, m = 50_000, 20_000
+    strings = [bytes(rng.integers(65, 91, size=int(L), dtype=np.uint8))
+               for L in rng.integers(0, 40, size=n)]
+    data = np.frombuffer(b"".join(strings), dtype=np.uint8)
+    offs = np.zeros(n + 1, dtype=np.int32)
+    for i, s_ in enumerate(strings):
+        offs[i + 1] = offs[i] + len(s_)
+    idx = rng.integers(0, n, size=m, dtype=np.uint32)
+
+    dbuf = ctx.upload(data if len(data) else np.zeros(1, dtype=np.uint8))
+    obuf = ctx.upload(offs)
+    ibuf = ctx.upload(idx)
+    out_offs, out_data, total = ctx.gather_varlen(dbuf, obuf, ibuf, m,
+                                                  max_bytes=int(data.nbytes) + 1)
+    got_offs = out_offs.download(np.int32, m + 1)
+    got_data = out_data.download(np.uint8, max(total, 1))[:total].tobytes()
+
+    arr = pa.array([s_.decode() for s_ in strings], type=pa.utf8())
+    want = arr.take(pa.array(idx, type=pa.uint32()))
+    want_strings = [v.as_py().encode() for v in want]
+    assert got_offs[0] == 0 and got_offs[-1] == total
+    got_strings = [got_data[got_offs[i]:got_offs[i + 1]] for i in range(m)]
+    assert got_strings == want_strings
