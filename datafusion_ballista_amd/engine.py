@@ -48,6 +48,8 @@ _PA_TO_BG = {
 def _bg_dtype(t: pa.DataType) -> int:
     if pa.types.is_decimal128(t):
         return gpu.BG_DT_DECIMAL128
+    if pa.types.is_string(t):
+        return gpu.BG_DT_UTF8
     try:
         return _PA_TO_BG[t]
     except KeyError:
@@ -158,31 +160,60 @@ class GpuQueryStageExecutor:
         n = table.num_rows
         schema = table.schema
 
-        cols = []
+        cols = []           # BgColumn per table column (fixed OR utf8)
+        utf8_src = {}       # col index -> (data buf, offsets buf, total)
         for i in range(table.num_columns):
             arr = table.column(i).combine_chunks()
             if isinstance(arr, pa.ChunkedArray):
                 arr = arr.chunk(0) if arr.num_chunks else pa.array(
                     [], type=arr.type)
-            raw = _col_raw(arr)
-            bgdt = _bg_dtype(arr.type)
-            buf = ctx.upload(raw)
-            nelem = n
-            cols.append(ctx.column(bgdt, buf, nelem))
+            t = arr.type
+            if pa.types.is_string(t):
+                if arr.null_count:
+                    raise RuntimeError("null strings: next round")
+                offs = np.frombuffer(arr.buffers()[1], dtype=np.int32,
+                                     count=n + 1 + arr.offset)[arr.offset:]
+                data_buf = arr.buffers()[2]
+                nbytes = int(offs[-1]) - int(offs[0])
+                base = int(offs[0])
+                data_np = np.frombuffer(data_buf, dtype=np.uint8,
+                                        count=nbytes, offset=base)                     if nbytes else np.zeros(1, dtype=np.uint8)
+                offs_rb = (offs.astype(np.int32) - base).astype(np.int32)
+                dbuf = ctx.upload(data_np)
+                obuf = ctx.upload(offs_rb)
+                utf8_src[i] = (dbuf, obuf, nbytes)
+                cols.append(ctx.column(gpu.BG_DT_UTF8, dbuf, n, offsets=obuf))
+            else:
+                raw = _col_raw(arr)
+                bgdt = _bg_dtype(t)
+                buf = ctx.upload(raw)
+                cols.append(ctx.column(bgdt, buf, n))
 
         key_cols = [cols[i] for i in self.key_columns]
+        fixed_ids = [i for i in range(table.num_columns) if i not in utf8_src]
+        fixed_cols = [cols[i] for i in fixed_ids]
         t_repart = time.perf_counter()
         idx_buf, offs_buf, out_bufs = ctx.hash_repartition(
-            key_cols, cols, n, self.k)
+            key_cols, fixed_cols, n, self.k)
+        # variable-length payload: gather through the same permutation
+        utf8_out = {}
+        for i, (dbuf, obuf, nbytes) in utf8_src.items():
+            oo, od, tot = ctx.gather_varlen(dbuf, obuf, idx_buf, n,
+                                            max_bytes=max(nbytes, 1))
+            utf8_out[i] = (oo, od, tot)
         ctx.synchronize()
         dt_device = time.perf_counter() - t_repart
         offsets = offs_buf.download(np.int64, self.k + 1)
 
         # download partition-major buffers and slice per partition
-        col_raws = []
-        for c, b in zip(cols, out_bufs):
-            esz = gpu._DT_SIZE[c.dtype]
-            col_raws.append(b.download(np.uint8, max(n, 1) * esz))
+        col_raws = {}
+        for ci, b in zip(fixed_ids, out_bufs):
+            esz = gpu._DT_SIZE[cols[ci].dtype]
+            col_raws[ci] = b.download(np.uint8, max(n, 1) * esz)
+        utf8_raws = {}
+        for i, (oo, od, tot) in utf8_out.items():
+            utf8_raws[i] = (oo.download(np.int32, n + 1),
+                            od.download(np.uint8, max(tot, 1))[:tot])
 
         streams = []
         for p in range(self.k):
@@ -194,6 +225,16 @@ class GpuQueryStageExecutor:
             arrays = []
             for ci in range(table.num_columns):
                 t = schema.types[ci]
+                if ci in utf8_raws:
+                    offs_all, data_all = utf8_raws[ci]
+                    o_lo, o_hi = int(offs_all[lo]), int(offs_all[hi])
+                    sub_offs = (offs_all[lo:hi + 1] - o_lo).astype(np.int32)
+                    sub_data = data_all[o_lo:o_hi]
+                    arrays.append(pa.Array.from_buffers(
+                        pa.utf8(), m,
+                        [None, pa.py_buffer(sub_offs.tobytes()),
+                         pa.py_buffer(sub_data.tobytes())]))
+                    continue
                 esz = 16 if pa.types.is_decimal128(t) else \
                     np.dtype(_np_for(t)).itemsize
                 raw = col_raws[ci][lo * esz: hi * esz]

@@ -746,3 +746,42 @@ def test_gather_varlen_parity(ctx):
     assert got_offs[0] == 0 and got_offs[-1] == total
     got_strings = [got_data[got_offs[i]:got_offs[i + 1]] for i in range(m)]
     assert got_strings == want_strings
+
+
+def test_sort_shuffle_stage_with_utf8_payload(ctx, tmp_path):
+    """Shuffle stage carrying a Utf8 payload column AND a Utf8 key column
+    (q1's repartition keys are 1-char Utf8 — SURVEY.md §8a row 4)."""
+    from datafusion_ballista_amd import engine, shuffle
+    n, k = 30_000, 8
+    rng = np.random.default_rng(71)
+    flags = ["A", "N", "R"]
+    strs = [flags[i] for i in rng.integers(0, 3, size=n)]
+    comments = ["".join(chr(65 + c) for c in rng.integers(0, 26, size=int(L)))
+                for L in rng.integers(0, 30, size=n)]
+    table = pa.table({
+        "flag": pa.array(strs, type=pa.utf8()),
+        "v": pa.array(rng.integers(-10**9, 10**9, size=n, dtype=np.int64)),
+        "comment": pa.array(comments, type=pa.utf8()),
+    })
+    ex = engine.GpuQueryStageExecutor(ctx, "job-utf8", 6, str(tmp_path),
+                                      key_columns=[0], num_partitions=k)
+    summaries = ex.execute_query_stage(0, table)
+    assert sum(s.num_rows for s in summaries) == n
+
+    # oracle expectation: hash the utf8 key column
+    data = np.frombuffer("".join(strs).encode(), dtype=np.uint8)
+    offs = np.zeros(n + 1, dtype=np.int32)
+    for i, s_ in enumerate(strs):
+        offs[i + 1] = offs[i] + len(s_)
+    h = oracle.hash_columns([("utf8", data, offs)], n)
+    pids = oracle.partition_ids(h, k)
+    idx, poffs = oracle.partition_indices(pids, k)
+
+    data_path = summaries[0].path
+    for p in range(k):
+        batches = shuffle.read_partition(data_path, data_path + ".index", p)
+        got = pa.Table.from_batches(batches, schema=table.schema) if batches \
+            else table.schema.empty_table()
+        rows = idx[poffs[p]:poffs[p + 1]]
+        want = table.take(pa.array(rows, type=pa.uint32()))
+        assert got.equals(want), f"partition {p}"
