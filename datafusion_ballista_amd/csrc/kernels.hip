@@ -1311,6 +1311,13 @@ extern "C" int bg_hashjoin_free(void* handle) {
 #define BG_MAX_AGGS 8
 #define BG_AGG_SUM_DEC128 0
 #define BG_AGG_SUM_I64 1
+// MIN/MAX store an order-preserving u64 encoding with atomicMax so the
+// zero-initialised slot is the identity (no per-slot init race):
+//   MAX_I64: enc = v ^ (1<<63);          decode: enc ^ (1<<63)
+//   MIN_I64: enc = ~(v ^ (1<<63));       decode: ~enc ^ (1<<63)
+// (groups always hold >= 1 row, so enc 0 is never read back as a value)
+#define BG_AGG_MIN_I64 2
+#define BG_AGG_MAX_I64 3
 
 struct AggArgs {
   int naggs;
@@ -1441,6 +1448,18 @@ __global__ void k_hashagg(KeyArgs keys, AggArgs aggs, const u64* mask_words,
           // exact i64 sum in two's complement (wrap == i64 semantics);
           // keep the i128 carry so large sums stay exact at scale
           atomic_add_i128(base, base + 1, (i128)v);
+          break;
+        }
+        case BG_AGG_MAX_I64: {
+          const u64 v = (u64) reinterpret_cast<const int64_t*>(
+                            aggs.a[a].data)[i] ^ 0x8000000000000000ull;
+          atomicMax(base, v);
+          break;
+        }
+        case BG_AGG_MIN_I64: {
+          const u64 v = ~((u64) reinterpret_cast<const int64_t*>(
+                              aggs.a[a].data)[i] ^ 0x8000000000000000ull);
+          atomicMax(base, v);
           break;
         }
         default:

@@ -329,6 +329,20 @@ class GpuHashJoin:
 
 BG_AGG_OP_SUM_DEC128 = 0
 BG_AGG_OP_SUM_I64 = 1
+BG_AGG_OP_MIN_I64 = 2
+BG_AGG_OP_MAX_I64 = 3
+
+
+def decode_agg_value(op, raw16: bytes) -> int:
+    """Accumulator bytes -> python int per aggregate op."""
+    if op in (BG_AGG_OP_SUM_DEC128, BG_AGG_OP_SUM_I64):
+        return int.from_bytes(raw16, "little", signed=True)
+    enc = int.from_bytes(raw16[:8], "little")
+    if op == BG_AGG_OP_MAX_I64:
+        v = enc ^ (1 << 63)
+    else:  # MIN
+        v = (~enc & 0xFFFFFFFFFFFFFFFF) ^ (1 << 63)
+    return v - (1 << 64) if v >= 1 << 63 else v
 
 
 def _ctx_hashagg(self, key_cols, agg_cols, agg_ops, n, max_groups,

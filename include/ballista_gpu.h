@@ -213,6 +213,8 @@ int bg_hashjoin_free(void* handle);
  * (group, agg), d_counts_out = i64 per group. */
 #define BG_AGG_OP_SUM_DEC128 0
 #define BG_AGG_OP_SUM_I64 1
+#define BG_AGG_OP_MIN_I64 2 /* acc = order-preserving u64; host decodes */
+#define BG_AGG_OP_MAX_I64 3
 int bg_hashagg(const bg_column* key_cols, int32_t nkeys,
                const bg_column* agg_cols, const int32_t* agg_ops,
                int32_t naggs, const uint8_t* d_mask, int64_t n,

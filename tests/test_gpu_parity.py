@@ -785,3 +785,30 @@ def test_sort_shuffle_stage_with_utf8_payload(ctx, tmp_path):
         rows = idx[poffs[p]:poffs[p + 1]]
         want = table.take(pa.array(rows, type=pa.uint32()))
         assert got.equals(want), f"partition {p}"
+
+
+def test_hashagg_min_max(ctx):
+    """MIN/MAX aggregate ops (AggregateExec breadth) vs a numpy oracle."""
+    n = 300_000
+    rng = np.random.default_rng(23)
+    keys = rng.integers(0, 1000, size=n, dtype=np.int64)
+    vals = rng.integers(-2**62, 2**62, size=n, dtype=np.int64)
+    kc, _ = ctx.upload_column(keys, gpu.BG_DT_INT64)
+    vc, _ = ctx.upload_column(vals, gpu.BG_DT_INT64)
+    first, acc, counts = ctx.hashagg(
+        [kc], [vc, vc, vc],
+        [gpu.BG_AGG_OP_MIN_I64, gpu.BG_AGG_OP_MAX_I64, gpu.BG_AGG_OP_SUM_I64],
+        n, max_groups=4096)
+    got = {}
+    for g in range(len(first)):
+        key = int(keys[first[g]])
+        got[key] = (gpu.decode_agg_value(gpu.BG_AGG_OP_MIN_I64, bytes(acc[g, 0])),
+                    gpu.decode_agg_value(gpu.BG_AGG_OP_MAX_I64, bytes(acc[g, 1])),
+                    gpu.decode_agg_value(gpu.BG_AGG_OP_SUM_I64, bytes(acc[g, 2])),
+                    int(counts[g]))
+    want = {}
+    for k_ in np.unique(keys):
+        sel = vals[keys == k_]
+        want[int(k_)] = (int(sel.min()), int(sel.max()),
+                         int(sel.astype(object).sum()), int(len(sel)))
+    assert got == want
