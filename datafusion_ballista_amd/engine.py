@@ -243,3 +243,35 @@ class GpuQueryStageExecutor:
             batches = shuffle.rechunk(part_table, self.batch_size)
             streams.append(shuffle.encode_partition_stream(batches, schema))
         return streams, dt_device
+
+
+class GpuResidentShuffleStage(GpuQueryStageExecutor):
+    """`ballista.gpu.resident_shuffle` mode: the task's
+    partition-major column buffers STAY in HBM for an RCCL all-to-all to
+    co-located executors (exchange.all_to_all_rows) instead of being
+    IPC-encoded to disk; the file path remains the durability/parity
+    fallback (documented trade-off of shuffle.md:66-68 — resident shuffle
+    gives up recompute-from-files recovery for the exchange's speed)."""
+
+    def execute_query_stage_resident(self, task_id: int, table: pa.Table):
+        """-> (offsets np.int64[k+1], {col_idx: DeviceBuffer partition-major},
+        schema).  Fixed-width columns only in resident mode (strings fall
+        back to the file path)."""
+        ctx = self.ctx
+        table = table.combine_chunks()
+        n = table.num_rows
+        cols = []
+        for i in range(table.num_columns):
+            arr = table.column(i).combine_chunks()
+            if isinstance(arr, pa.ChunkedArray):
+                arr = arr.chunk(0)
+            if pa.types.is_string(arr.type):
+                raise RuntimeError("resident shuffle: fixed-width only (r1)")
+            raw = _col_raw(arr)
+            cols.append(ctx.column(_bg_dtype(arr.type), ctx.upload(raw), n))
+        key_cols = [cols[i] for i in self.key_columns]
+        idx_buf, offs_buf, out_bufs = ctx.hash_repartition(
+            key_cols, cols, n, self.k)
+        ctx.synchronize()
+        offsets = offs_buf.download(np.int64, self.k + 1)
+        return offsets, {i: b for i, b in enumerate(out_bufs)}, table.schema

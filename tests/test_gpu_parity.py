@@ -812,3 +812,33 @@ def test_hashagg_min_max(ctx):
         want[int(k_)] = (int(sel.min()), int(sel.max()),
                          int(sel.astype(object).sum()), int(len(sel)))
     assert got == want
+
+
+def test_resident_shuffle_matches_file_path(ctx, tmp_path):
+    """ballista.gpu.resident_shuffle: the HBM-resident partition buffers
+    hold exactly the rows the file path writes (the RCCL exchange then
+    moves these buffers instead of Flight moving file bytes)."""
+    from datafusion_ballista_amd import engine, shuffle
+    n, k = 60_000, 8
+    rng = np.random.default_rng(99)
+    table = pa.table({
+        "k": pa.array(rng.integers(0, 5_000, size=n, dtype=np.int64)),
+        "v": pa.array(rng.integers(-10**9, 10**9, size=n, dtype=np.int64)),
+    })
+    ex = engine.GpuResidentShuffleStage(ctx, "job-res", 9, str(tmp_path),
+                                        key_columns=[0], num_partitions=k)
+    offsets, bufs, schema = ex.execute_query_stage_resident(0, table)
+    assert offsets[0] == 0 and offsets[-1] == n
+    keys_pm = bufs[0].download(np.int64, n)
+    vals_pm = bufs[1].download(np.int64, n)
+
+    # file path on the same input
+    summaries = ex.execute_query_stage(1, table)
+    data_path = summaries[0].path
+    for p in range(k):
+        lo, hi = int(offsets[p]), int(offsets[p + 1])
+        batches = shuffle.read_partition(data_path, data_path + ".index", p)
+        want = pa.Table.from_batches(batches, schema=schema) if batches \
+            else schema.empty_table()
+        assert np.array_equal(keys_pm[lo:hi], want.column("k").to_numpy())
+        assert np.array_equal(vals_pm[lo:hi], want.column("v").to_numpy())
