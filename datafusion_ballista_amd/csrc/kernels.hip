@@ -2572,3 +2572,40 @@ extern "C" int bg_gather_varlen(const void* d_src_data,
   *out_total_bytes = total;
   return BG_OK;
 }
+
+// ---------------------------------------------------------------------------
+// Validity-bitmap gather: out bit i = valid[idx[i]] — the take of a
+// null-carrying column's validity buffer (Arrow LSB order; ballot word ==
+// bitmap word).  Groundwork for nullable payload materialisation.
+// ---------------------------------------------------------------------------
+__global__ void k_gather_bits(const uint8_t* valid, const uint32_t* idx,
+                              int64_t m, u64* out_words) {
+  const int64_t nwords = (m + 63) / 64;
+  const int64_t wave_global =
+      ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) / BG_WAVE;
+  const int64_t nwaves = ((int64_t)gridDim.x * blockDim.x) / BG_WAVE;
+  for (int64_t w = wave_global; w < nwords; w += nwaves) {
+    const int64_t i = w * BG_WAVE + lane_id();
+    bool bit = false;
+    if (i < m) {
+      const uint32_t r = idx[i];
+      bit = (valid[r >> 3] >> (r & 7)) & 1;
+    }
+    const u64 word = __ballot(bit);
+    if (lane_id() == 0) out_words[w] = word;
+  }
+}
+
+extern "C" int bg_gather_bits(const uint8_t* d_valid, const uint32_t* d_idx,
+                              int64_t m, uint8_t* d_out_bits) {
+  REQUIRE_INIT();
+  const int64_t nwords = (m + 63) / 64;
+  const int waves_per_block = BG_BLOCK / BG_WAVE;
+  int blocks = (int)bg_imin64((nwords + waves_per_block - 1) / waves_per_block,
+                              BG_MAX_BLOCKS);
+  if (blocks == 0) blocks = 1;
+  hipLaunchKernelGGL(k_gather_bits, dim3(blocks), dim3(BG_BLOCK), 0, 0,
+                     d_valid, d_idx, m, reinterpret_cast<u64*>(d_out_bits));
+  HIP_TRY(hipGetLastError());
+  return BG_OK;
+}

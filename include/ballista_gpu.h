@@ -142,6 +142,11 @@ int bg_gather_varlen(const void* d_src_data, const int32_t* d_src_offsets,
                      int32_t* d_out_offsets, void* d_out_data,
                      int64_t out_data_cap, int64_t* out_total_bytes);
 
+/* Validity-bitmap gather: out bit i = valid[idx[i]] (Arrow LSB order;
+ * d_out_bits holds ceil(m/64)*8 bytes) — take for null-carrying columns. */
+int bg_gather_bits(const uint8_t* d_valid, const uint32_t* d_idx, int64_t m,
+                   uint8_t* d_out_bits);
+
 /* ---- hash repartition (SortShuffleWriterExec device half) ---- */
 
 /* create_hashes restatement over the key columns (bg_ahash.h; parity

@@ -842,3 +842,24 @@ def test_resident_shuffle_matches_file_path(ctx, tmp_path):
             else schema.empty_table()
         assert np.array_equal(keys_pm[lo:hi], want.column("k").to_numpy())
         assert np.array_equal(vals_pm[lo:hi], want.column("v").to_numpy())
+
+
+def test_gather_bits_parity(ctx):
+    """Validity-bitmap take vs a numpy bit oracle (null columns through the
+    shuffle/filter materialisation)."""
+    rng = np.random.default_rng(41)
+    n, m = 100_000, 40_000
+    valid = rng.integers(0, 256, size=(n + 7) // 8, dtype=np.uint8)
+    idx = rng.integers(0, n, size=m, dtype=np.uint32)
+    vbuf = ctx.upload(valid)
+    ibuf = ctx.upload(idx)
+    nwords = (m + 63) // 64
+    out = ctx.alloc(nwords * 8)
+    import ctypes
+    gpu._check(ctx.L.bg_gather_bits(vbuf.ptr, ibuf.ptr, ctypes.c_int64(m),
+                                    out.ptr), "bg_gather_bits")
+    got_bits = np.unpackbits(out.download(np.uint8, nwords * 8),
+                             bitorder="little")[:m]
+    src_bits = np.unpackbits(valid, bitorder="little")[:n]
+    want = src_bits[idx]
+    assert np.array_equal(got_bits, want)
