@@ -83,11 +83,13 @@ def fuzz_round(ctx, seed):
     want_groups = {kk[0]: v for kk, v in want_groups.items()}
     assert got_groups == want_groups, f"hashagg mismatch seed={seed}"
 
-    # stable sort
-    perm = ctx.sort_rows([col], [bool(rng.integers(0, 2))], n)
-    # stability & order check against numpy
-    desc = False  # track which we asked
-    ctx.synchronize()
+    # stable sort (asc/desc) vs numpy stable argsort
+    desc = bool(rng.integers(0, 2))
+    perm = ctx.sort_rows([col], [desc], n).download(np.uint32, n)
+    u = keys.view(np.uint64) ^ np.uint64(1 << 63)
+    want_perm = np.argsort(~u if desc else u, kind="stable")
+    assert np.array_equal(perm.astype(np.int64), want_perm), \
+        f"sort mismatch seed={seed} desc={desc}"
     return n, k
 
 
