@@ -240,6 +240,37 @@ def q1(rf_code, ls_code, qty16, price16, disc16, tax16, shipdate, date_le):
     return out
 
 
+_omp_lib = None
+
+
+def omp_lib():
+    """Multithreaded oracle build — bench.py cpu_baseline leg only."""
+    global _omp_lib
+    if _omp_lib is None:
+        subprocess.run(["make", "-C", _DIR, "liboracle_omp.so"], check=True,
+                       capture_output=True)
+        _omp_lib = ctypes.CDLL(os.path.join(_DIR, "liboracle_omp.so"))
+        _decorate(_omp_lib)
+    return _omp_lib
+
+
+def q6_omp(shipdate, discount16, quantity16, price16, date_lo, date_hi,
+           disc_lo, disc_hi, qty_lt):
+    """All-core q6 (exact; i128 wrap-add is associative)."""
+    L = omp_lib()
+    sd = np.ascontiguousarray(shipdate, dtype=np.int32)
+    n = len(sd)
+    s_lo = ctypes.c_uint64(0)
+    s_hi = ctypes.c_int64(0)
+    cnt = L.oracle_q6(_p(sd, ctypes.c_int32),
+                      _p(np.ascontiguousarray(discount16, dtype=np.uint8), ctypes.c_uint8),
+                      _p(np.ascontiguousarray(quantity16, dtype=np.uint8), ctypes.c_uint8),
+                      _p(np.ascontiguousarray(price16, dtype=np.uint8), ctypes.c_uint8),
+                      n, int(date_lo), int(date_hi), int(disc_lo), int(disc_hi),
+                      int(qty_lt), ctypes.byref(s_lo), ctypes.byref(s_hi))
+    return cnt, (s_hi.value << 64) + s_lo.value
+
+
 def dec128_from_ints(vals) -> np.ndarray:
     """Scaled python ints -> Arrow Decimal128 byte layout (16B LE each)."""
     out = np.zeros(len(vals) * 16, dtype=np.uint8)

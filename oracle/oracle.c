@@ -247,6 +247,11 @@ EXPORT int64_t oracle_q6(const int32_t* shipdate, const uint8_t* discount,
                          uint64_t* sum_lo, int64_t* sum_hi) {
   i128 acc = 0;
   int64_t count = 0;
+#ifdef ORACLE_OMP
+  /* bench.py's cpu_baseline leg only: same arithmetic, all host cores.
+   * i128 wrap-add is associative, so the parallel reduction is exact. */
+#pragma omp parallel for reduction(+ : acc, count) schedule(static)
+#endif
   for (int64_t i = 0; i < n; ++i) {
     int32_t d = shipdate[i];
     if (d < date_lo || d >= date_hi) continue;
