@@ -45,7 +45,7 @@ def parse_args():
     p.add_argument("--warmup", type=int, default=3)
     p.add_argument("--rows", type=int, default=ROWS_SF100,
                    help="rows per GPU (default SF100 lineitem)")
-    p.add_argument("--cpu-sample-rows", type=int, default=30_000_000)
+    p.add_argument("--cpu-sample-rows", type=int, default=100_000_000)
     p.add_argument("--skip-cpu-baseline", action="store_true")
     return p.parse_args()
 
@@ -104,8 +104,8 @@ def cpu_baseline_leg(sample_rows: int):
     dt = time.perf_counter() - t0
     passes = 1
     target = 10.0
-    while dt < target and passes < 64:
-        more = min(63, max(1, int((target - dt) / max(dt / passes, 1e-3))))
+    while dt < target and passes < 1024:
+        more = min(1023, max(1, int((target - dt) / max(dt / passes, 1e-3))))
         for _ in range(more):
             one_pass()
         passes += more
