@@ -80,8 +80,6 @@ def _col_raw(arr: pa.Array) -> np.ndarray:
 
 
 def _array_from_raw(t: pa.DataType, raw: np.ndarray, n: int) -> pa.Array:
-    if pa.types.is_decimal128(t):
-        return pa.Array.from_buffers(t, n, [None, pa.py_buffer(raw.tobytes())])
     return pa.Array.from_buffers(t, n, [None, pa.py_buffer(raw.tobytes())])
 
 

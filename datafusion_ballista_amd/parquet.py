@@ -182,7 +182,7 @@ class GpuParquetColumnReader:
             raise RuntimeError(f"physical type {phys} not GPU-decodable yet")
 
         import ctypes
-        chunk = self._file_buf  # offsets below are absolute file offsets
+        chunk = self._file_buf  # page offsets below are absolute file offsets
         headers = []  # (rg, ptype, file_off, csz, usz, nvals, enc, ndict, soff)
         scratch_total = 0
         total_values = 0
