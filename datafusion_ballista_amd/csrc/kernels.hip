@@ -2609,3 +2609,132 @@ extern "C" int bg_gather_bits(const uint8_t* d_valid, const uint32_t* d_idx,
   HIP_TRY(hipGetLastError());
   return BG_OK;
 }
+
+// ---------------------------------------------------------------------------
+// Device LZ4 block compression (SURVEY.md §8f row 3 — the GPU shuffle
+// codec's compress half; decompression lives in k_snappy/... for parquet,
+// and LZ4-frame DECODE is the CPU reader's job).  64 KiB blocks, one wave
+// per block: lane 0 runs the greedy matcher (strictly serial emission)
+// with a 4096-entry LDS hash table; blocks that don't shrink are reported
+// as stored (negative size) and the host frames them uncompressed.
+// Frame assembly (constant 7-byte header 04224d18/40/40/c0 + [u32 size]
+// blocks + end mark) is host-side glue over the returned sizes.
+// ---------------------------------------------------------------------------
+#define LZ4_BLOCK 65536
+#define LZ4_HASH_LOG 12
+#define LZ4_SLOT_STRIDE (LZ4_BLOCK + 8)
+
+__global__ void k_lz4_compress(const uint8_t* src, int64_t len,
+                               uint8_t* out_slots, int64_t slot_stride,
+                               int64_t* block_sizes, int64_t nblocks) {
+  extern __shared__ __attribute__((aligned(16))) char smem_raw[];
+  uint16_t* tab_all = reinterpret_cast<uint16_t*>(smem_raw);
+  const int wave_in_block = threadIdx.x / BG_WAVE;
+  uint16_t* tab = tab_all + (size_t)wave_in_block * (1 << LZ4_HASH_LOG);
+  const int64_t wave_global =
+      ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) / BG_WAVE;
+  const int64_t nwaves = ((int64_t)gridDim.x * blockDim.x) / BG_WAVE;
+  const int lane = lane_id();
+  for (int64_t b = wave_global; b < nblocks; b += nwaves) {
+    const int64_t boff = b * LZ4_BLOCK;
+    const int32_t blen = (int32_t)(len - boff < LZ4_BLOCK ? len - boff
+                                                          : LZ4_BLOCK);
+    const uint8_t* s = src + boff;
+    uint8_t* d = out_slots + b * slot_stride;
+    // parallel table clear (0xffff = empty)
+    for (int i = lane; i < (1 << LZ4_HASH_LOG); i += BG_WAVE) tab[i] = 0xffff;
+    __builtin_amdgcn_wave_barrier();
+    if (lane == 0) {
+      int32_t pos = 0, anchor = 0, w = 0;
+      const int32_t mflimit = blen - 12;  // no match may START past here
+      const int32_t matchlimit = blen - 5;  // last 5 bytes stay literals
+      bool overflow = false;
+      auto emit_seq = [&](int32_t lit_len, int32_t mlen, int32_t moff) {
+        // worst case bytes: 1 + lit_len/255+1 + lit_len + 2 + mlen/255+1
+        if (w + lit_len + (lit_len / 255) + (mlen / 255) + 12 > blen) {
+          overflow = true;
+          return;
+        }
+        const int32_t ml_token = mlen >= 0 ? (mlen - 4) : 0;
+        uint8_t token = (uint8_t)((lit_len < 15 ? lit_len : 15) << 4);
+        if (mlen >= 0) token |= (uint8_t)(ml_token < 15 ? ml_token : 15);
+        d[w++] = token;
+        if (lit_len >= 15) {
+          int32_t r = lit_len - 15;
+          while (r >= 255) { d[w++] = 255; r -= 255; }
+          d[w++] = (uint8_t)r;
+        }
+        for (int32_t i = 0; i < lit_len; ++i) d[w++] = s[anchor + i];
+        if (mlen >= 0) {
+          d[w++] = (uint8_t)(moff & 0xff);
+          d[w++] = (uint8_t)(moff >> 8);
+          if (ml_token >= 15) {
+            int32_t r = ml_token - 15;
+            while (r >= 255) { d[w++] = 255; r -= 255; }
+            d[w++] = (uint8_t)r;
+          }
+        }
+      };
+      if (blen >= 13) {
+        while (pos <= mflimit && !overflow) {
+          uint32_t v;
+          __builtin_memcpy(&v, s + pos, 4);
+          const uint32_t h = (v * 2654435761u) >> (32 - LZ4_HASH_LOG);
+          const int32_t cand = tab[h] == 0xffff ? -1 : (int32_t)tab[h];
+          tab[h] = (uint16_t)pos;
+          uint32_t cv = 0;
+          if (cand >= 0) __builtin_memcpy(&cv, s + cand, 4);
+          if (cand >= 0 && cv == v) {
+            int32_t mlen = 4;
+            while (pos + mlen < matchlimit && s[cand + mlen] == s[pos + mlen])
+              ++mlen;
+            emit_seq(pos - anchor, mlen, pos - cand);
+            pos += mlen;
+            anchor = pos;
+          } else {
+            ++pos;
+          }
+        }
+      }
+      // trailing literals-only sequence
+      if (!overflow) emit_seq(blen - anchor, -1, 0);
+      if (overflow || w >= blen) {
+        // stored block: raw copy, negative size signals "uncompressed"
+        block_sizes[b] = -(int64_t)blen;
+      } else {
+        block_sizes[b] = (int64_t)w;
+      }
+    }
+    __builtin_amdgcn_wave_barrier();
+    // stored blocks: the whole wave copies raw bytes into the slot
+    const int64_t bs = block_sizes[b];
+    __builtin_amdgcn_wave_barrier();
+    if (bs < 0) {
+      for (int32_t i = lane; i < blen; i += BG_WAVE) d[i] = s[i];
+    }
+  }
+}
+
+extern "C" int bg_lz4_compress(const void* d_src, int64_t len,
+                               void* d_out_slots, int64_t* h_block_sizes,
+                               int64_t* out_nblocks) {
+  REQUIRE_INIT();
+  const int64_t nblocks = (len + LZ4_BLOCK - 1) / LZ4_BLOCK;
+  int64_t* d_sizes;
+  HIP_TRY(pool_malloc((void**)&d_sizes, sizeof(int64_t) * (nblocks ? nblocks : 1)));
+  const int waves_per_block = BG_BLOCK / BG_WAVE;
+  const size_t lds = (size_t)waves_per_block * (1 << LZ4_HASH_LOG) *
+                     sizeof(uint16_t);
+  int blocks = (int)bg_imin64((nblocks + waves_per_block - 1) / waves_per_block,
+                              BG_MAX_BLOCKS);
+  if (blocks == 0) blocks = 1;
+  hipLaunchKernelGGL(k_lz4_compress, dim3(blocks), dim3(BG_BLOCK), lds, 0,
+                     (const uint8_t*)d_src, len, (uint8_t*)d_out_slots,
+                     (int64_t)LZ4_SLOT_STRIDE, d_sizes, nblocks);
+  HIP_TRY(hipGetLastError());
+  HIP_TRY(hipMemcpy(h_block_sizes, d_sizes, sizeof(int64_t) * nblocks,
+                    hipMemcpyDeviceToHost));
+  (void)pool_release(d_sizes);
+  *out_nblocks = nblocks;
+  return BG_OK;
+}

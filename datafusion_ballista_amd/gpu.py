@@ -479,3 +479,40 @@ def _ctx_gather_varlen(self, src_data: "DeviceBuffer", src_offsets: "DeviceBuffe
 
 
 GpuStageContext.gather_varlen = _ctx_gather_varlen
+
+
+LZ4_FRAME_HEADER = bytes.fromhex("04224d184040c0")  # magic+FLG+BD+HC
+
+
+def _ctx_lz4_compress(self, src: "DeviceBuffer", length: int):
+    """Device LZ4 -> (list of (size, is_stored), slots DeviceBuffer)."""
+    nblocks = (length + 65536 - 1) // 65536 if length else 0
+    slots = self.alloc(max(nblocks * 65544, 8))
+    sizes = np.zeros(max(nblocks, 1), dtype=np.int64)
+    nb = ctypes.c_int64()
+    _check(self.L.bg_lz4_compress(src.ptr, ctypes.c_int64(length), slots.ptr,
+                                  sizes.ctypes.data_as(
+                                      ctypes.POINTER(ctypes.c_int64)),
+                                  ctypes.byref(nb)), "bg_lz4_compress")
+    return sizes[:nblocks], slots
+
+
+def lz4_frame_assemble(sizes, slot_bytes: bytes, length: int) -> bytes:
+    """Host glue: device block slots -> one LZ4 frame (constant header,
+    [u32 size] blocks — high bit = stored — end mark)."""
+    import struct
+    out = [LZ4_FRAME_HEADER]
+    for i, sz in enumerate(sizes):
+        base = i * 65544
+        blen = min(65536, length - i * 65536)
+        if sz < 0:  # stored raw
+            out.append(struct.pack("<I", blen | 0x80000000))
+            out.append(slot_bytes[base:base + blen])
+        else:
+            out.append(struct.pack("<I", int(sz)))
+            out.append(slot_bytes[base:base + int(sz)])
+    out.append(b"\x00\x00\x00\x00")
+    return b"".join(out)
+
+
+GpuStageContext.lz4_compress = _ctx_lz4_compress

@@ -304,6 +304,17 @@ typedef struct {
 } bg_dict_indices_job;
 int bg_dict_indices_batch(const void* h_jobs, int64_t njobs);
 
+/* Device LZ4 block compression (the GPU shuffle codec's compress half,
+ * SURVEY.md §8f row 3): 64 KiB blocks, one wave per block.  d_out_slots
+ * holds nblocks slots of 65544 B; h_block_sizes[i] = compressed size, or
+ * negative (-usize) when the block is stored raw.  Host glue frames the
+ * blocks (constant header 04224d18 40 40 c0 + [u32 size] blocks + end
+ * mark) into the LZ4_FRAME streams Arrow IPC carries. */
+#define BG_LZ4_BLOCK 65536
+#define BG_LZ4_SLOT_STRIDE 65544
+int bg_lz4_compress(const void* d_src, int64_t len, void* d_out_slots,
+                    int64_t* h_block_sizes, int64_t* out_nblocks);
+
 /* ---- fused filter+aggregate stages ---- */
 
 /* TPC-H q6 stage 1 (scan+filter+aggregate, approved/q6.txt):

@@ -863,3 +863,32 @@ def test_gather_bits_parity(ctx):
     src_bits = np.unpackbits(valid, bitorder="little")[:n]
     want = src_bits[idx]
     assert np.array_equal(got_bits, want)
+
+
+# ---------------------------------------------------------------------------
+# device LZ4 (shuffle codec compress half)
+# ---------------------------------------------------------------------------
+@pytest.mark.parametrize("case", ["zeros", "random", "text", "ints",
+                                  "multiblock", "tiny"])
+def test_lz4_compress_roundtrip(ctx, case):
+    """Device-compressed LZ4 frames must decode bit-exact with Arrow's own
+    (reference) LZ4_FRAME codec — the same decoder ShuffleReaderExec uses
+    for shuffle batches (codec default lz4, config.rs:413-415)."""
+    rng = np.random.default_rng(len(case))
+    payloads = {
+        "zeros": b"\x00" * 200_000,
+        "random": bytes(rng.integers(0, 256, 150_000, dtype=np.uint8)),
+        "text": (b"the quick brown fox jumps over the lazy dog. " * 5000),
+        "ints": np.arange(50_000, dtype=np.int64).tobytes(),
+        "multiblock": bytes(rng.integers(0, 64, 300_000, dtype=np.uint8)),
+        "tiny": b"abc",
+    }
+    data = payloads[case]
+    src = ctx.upload(np.frombuffer(data, dtype=np.uint8))
+    sizes, slots = ctx.lz4_compress(src, len(data))
+    slot_bytes = slots.download(np.uint8, max(len(sizes) * 65544, 1)).tobytes()
+    frame = gpu.lz4_frame_assemble(sizes, slot_bytes, len(data))
+    got = pa.decompress(frame, len(data), codec="lz4", asbytes=True)
+    assert got == data
+    if case in ("zeros", "text", "ints"):
+        assert len(frame) < len(data) // 2, "compressible data must shrink"
