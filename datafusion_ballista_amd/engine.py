@@ -24,6 +24,7 @@ import numpy as np
 import pyarrow as pa
 
 from . import gpu
+from . import ipc as bgipc
 from . import shuffle
 
 
@@ -91,7 +92,8 @@ class GpuQueryStageExecutor:
 
     def __init__(self, ctx: "gpu.GpuStageContext", job_id: str, stage_id: int,
                  work_dir: str, key_columns, num_partitions: int,
-                 batch_size: int = shuffle.DEFAULT_BATCH_SIZE):
+                 batch_size: int = shuffle.DEFAULT_BATCH_SIZE,
+                 gpu_codec: bool = False):
         self.ctx = ctx
         self.job_id = job_id
         self.stage_id = stage_id
@@ -99,6 +101,11 @@ class GpuQueryStageExecutor:
         self.key_columns = key_columns  # column indices (already-evaluated keys)
         self.k = num_partitions
         self.batch_size = batch_size
+        # gpu_codec: LZ4-compress the shuffle batch bodies ON DEVICE
+        # (bg_lz4_compress) and emit IPC metadata with the handwritten
+        # writer (ipc.py) — removes the host LZ4+encode cost (SURVEY.md
+        # §8f row 3).  Fixed-width columns; one batch per partition.
+        self.gpu_codec = gpu_codec
         self._metrics = {}
 
     def collect_plan_metrics(self):
@@ -213,6 +220,12 @@ class GpuQueryStageExecutor:
             utf8_raws[i] = (oo.download(np.int32, n + 1),
                             od.download(np.uint8, max(tot, 1))[:tot])
 
+        if self.gpu_codec:
+            if utf8_src:
+                raise RuntimeError("gpu_codec: fixed-width columns only (r1)")
+            return self._encode_partitions_gpu(
+                schema, offsets, cols, out_bufs), dt_device
+
         streams = []
         for p in range(self.k):
             lo, hi = int(offsets[p]), int(offsets[p + 1])
@@ -273,3 +286,47 @@ class GpuResidentShuffleStage(GpuQueryStageExecutor):
         ctx.synchronize()
         offsets = offs_buf.download(np.int64, self.k + 1)
         return offsets, {i: b for i, b in enumerate(out_bufs)}, table.schema
+
+
+    def _encode_partitions_gpu(self, schema, offsets, cols, out_bufs):
+        """Device-LZ4 the partition-major column slices and frame them with
+        the handwritten IPC writer; the host only concatenates compressed
+        bytes."""
+        import ctypes
+        import struct
+        ctx = self.ctx
+        streams = []
+        for p in range(self.k):
+            lo, hi = int(offsets[p]), int(offsets[p + 1])
+            m = hi - lo
+            if m == 0:
+                streams.append(b"")
+                continue
+            parts = []
+            nodes = []
+            for ci in range(len(cols)):
+                esz = gpu._DT_SIZE[cols[ci].dtype]
+                length = m * esz
+                src_ptr = ctypes.c_void_p(out_bufs[ci].ptr.value + lo * esz)
+                nblocks = (length + 65536 - 1) // 65536
+                slots = ctx.alloc(max(nblocks * 65544, 8))
+                sizes = np.zeros(max(nblocks, 1), dtype=np.int64)
+                nb = ctypes.c_int64()
+                gpu._check(ctx.L.bg_lz4_compress(
+                    src_ptr, ctypes.c_int64(length), slots.ptr,
+                    sizes.ctypes.data_as(ctypes.POINTER(ctypes.c_int64)),
+                    ctypes.byref(nb)), "bg_lz4_compress")
+                slot_bytes = slots.download(
+                    np.uint8, max(nblocks * 65544, 1)).tobytes()
+                frame = gpu.lz4_frame_assemble(sizes[:nblocks], slot_bytes,
+                                               length)
+                parts.append((None, struct.pack("<q", length) + frame))
+                nodes.append((m, 0))
+                slots.free()
+            buffer_parts = []
+            for validity, data_part in parts:
+                buffer_parts.append(validity)
+                buffer_parts.append(data_part)
+            streams.append(bgipc.stream_from_compressed_batches(
+                schema, [(m, nodes, buffer_parts)]))
+        return streams

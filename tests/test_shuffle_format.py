@@ -180,3 +180,31 @@ def test_global_partition_map():
     m = shuffle.GlobalPartitionMap(shuffle.GlobalPartitionMap.PASSTHROUGH,
                                    [5, 9])
     assert [m.resolve(i) for i in range(3)] == [5, 9, 2]
+
+
+def test_handwritten_ipc_writer_roundtrip():
+    """The minimal flatbuffers IPC writer (datafusion_ballista_amd.ipc) must
+    be readable by Arrow's own reader with CPU-compressed bodies (the GPU
+    path swaps in device-compressed frames byte-for-byte)."""
+    import struct
+    from datafusion_ballista_amd import ipc as bgipc
+
+    def cpu_part(data: bytes) -> bytes:
+        comp = pa.compress(data, codec="lz4", asbytes=True)
+        return struct.pack("<q", len(data)) + comp
+
+    rng = np.random.default_rng(13)
+    n = 10_000
+    a = rng.integers(-2**60, 2**60, size=n, dtype=np.int64)
+    b = rng.integers(-10**9, 10**9, size=n, dtype=np.int32)
+    schema = pa.schema([("a", pa.int64()), ("b", pa.int32())])
+    stream = bgipc.stream_from_compressed_batches(schema, [
+        (n, [(n, 0), (n, 0)],
+         [None, cpu_part(a.tobytes()), None, cpu_part(b.tobytes())]),
+        (n, [(n, 0), (n, 0)],
+         [None, cpu_part(a.tobytes()), None, cpu_part(b.tobytes())]),
+    ])
+    got = pa.ipc.open_stream(pa.BufferReader(stream)).read_all()
+    one = pa.table({"a": a, "b": b})
+    want = pa.concat_tables([one, one])
+    assert got.equals(want)
