@@ -288,45 +288,48 @@ class GpuResidentShuffleStage(GpuQueryStageExecutor):
         return offsets, {i: b for i, b in enumerate(out_bufs)}, table.schema
 
 
-    def _encode_partitions_gpu(self, schema, offsets, cols, out_bufs):
-        """Device-LZ4 the partition-major column slices and frame them with
-        the handwritten IPC writer; the host only concatenates compressed
-        bytes."""
-        import ctypes
-        import struct
-        ctx = self.ctx
-        streams = []
-        for p in range(self.k):
-            lo, hi = int(offsets[p]), int(offsets[p + 1])
-            m = hi - lo
-            if m == 0:
-                streams.append(b"")
-                continue
-            parts = []
-            nodes = []
-            for ci in range(len(cols)):
-                esz = gpu._DT_SIZE[cols[ci].dtype]
-                length = m * esz
-                src_ptr = ctypes.c_void_p(out_bufs[ci].ptr.value + lo * esz)
-                nblocks = (length + 65536 - 1) // 65536
-                slots = ctx.alloc(max(nblocks * 65544, 8))
-                sizes = np.zeros(max(nblocks, 1), dtype=np.int64)
-                nb = ctypes.c_int64()
-                gpu._check(ctx.L.bg_lz4_compress(
-                    src_ptr, ctypes.c_int64(length), slots.ptr,
-                    sizes.ctypes.data_as(ctypes.POINTER(ctypes.c_int64)),
-                    ctypes.byref(nb)), "bg_lz4_compress")
-                slot_bytes = slots.download(
-                    np.uint8, max(nblocks * 65544, 1)).tobytes()
-                frame = gpu.lz4_frame_assemble(sizes[:nblocks], slot_bytes,
-                                               length)
-                parts.append((None, struct.pack("<q", length) + frame))
-                nodes.append((m, 0))
-                slots.free()
-            buffer_parts = []
-            for validity, data_part in parts:
-                buffer_parts.append(validity)
-                buffer_parts.append(data_part)
-            streams.append(bgipc.stream_from_compressed_batches(
-                schema, [(m, nodes, buffer_parts)]))
-        return streams
+def _encode_partitions_gpu(self, schema, offsets, cols, out_bufs):
+    """Device-LZ4 the partition-major column slices and frame them with
+    the handwritten IPC writer; the host only concatenates compressed
+    bytes."""
+    import ctypes
+    import struct
+    ctx = self.ctx
+    streams = []
+    for p in range(self.k):
+        lo, hi = int(offsets[p]), int(offsets[p + 1])
+        m = hi - lo
+        if m == 0:
+            streams.append(b"")
+            continue
+        parts = []
+        nodes = []
+        for ci in range(len(cols)):
+            esz = gpu._DT_SIZE[cols[ci].dtype]
+            length = m * esz
+            src_ptr = ctypes.c_void_p(out_bufs[ci].ptr.value + lo * esz)
+            nblocks = (length + 65536 - 1) // 65536
+            slots = ctx.alloc(max(nblocks * 65544, 8))
+            sizes = np.zeros(max(nblocks, 1), dtype=np.int64)
+            nb = ctypes.c_int64()
+            gpu._check(ctx.L.bg_lz4_compress(
+                src_ptr, ctypes.c_int64(length), slots.ptr,
+                sizes.ctypes.data_as(ctypes.POINTER(ctypes.c_int64)),
+                ctypes.byref(nb)), "bg_lz4_compress")
+            slot_bytes = slots.download(
+                np.uint8, max(nblocks * 65544, 1)).tobytes()
+            frame = gpu.lz4_frame_assemble(sizes[:nblocks], slot_bytes,
+                                           length)
+            parts.append((None, struct.pack("<q", length) + frame))
+            nodes.append((m, 0))
+            slots.free()
+        buffer_parts = []
+        for validity, data_part in parts:
+            buffer_parts.append(validity)
+            buffer_parts.append(data_part)
+        streams.append(bgipc.stream_from_compressed_batches(
+            schema, [(m, nodes, buffer_parts)]))
+    return streams
+
+
+GpuQueryStageExecutor._encode_partitions_gpu = _encode_partitions_gpu
