@@ -2738,3 +2738,47 @@ extern "C" int bg_lz4_compress(const void* d_src, int64_t len,
   *out_nblocks = nblocks;
   return BG_OK;
 }
+
+// ---------------------------------------------------------------------------
+// Batched block pack: assemble LZ4 frame bodies on device — each job writes
+// its [u32 size-word][block bytes] at the precomputed frame offset, so the
+// host downloads ONE contiguous compressed stream per buffer (no
+// slot-strided D2H, no host-side slicing).
+// ---------------------------------------------------------------------------
+struct PackJob {
+  const uint8_t* src;
+  uint8_t* dst;        // points AT the 4-byte size word
+  int64_t nbytes;      // block payload bytes
+  uint32_t size_word;  // little-endian u32 (high bit = stored)
+  uint32_t _pad;
+};
+
+__global__ void k_pack_blocks(const PackJob* jobs, int64_t njobs) {
+  for (int64_t j = blockIdx.x; j < njobs; j += gridDim.x) {
+    const PackJob job = jobs[j];
+    if (threadIdx.x == 0) {
+      job.dst[0] = (uint8_t)(job.size_word & 0xff);
+      job.dst[1] = (uint8_t)((job.size_word >> 8) & 0xff);
+      job.dst[2] = (uint8_t)((job.size_word >> 16) & 0xff);
+      job.dst[3] = (uint8_t)((job.size_word >> 24) & 0xff);
+    }
+    uint8_t* d = job.dst + 4;
+    for (int64_t i = threadIdx.x; i < job.nbytes; i += blockDim.x)
+      d[i] = job.src[i];
+  }
+}
+
+extern "C" int bg_pack_blocks(const void* h_jobs, int64_t njobs) {
+  REQUIRE_INIT();
+  PackJob* d_jobs;
+  HIP_TRY(pool_malloc((void**)&d_jobs, sizeof(PackJob) * (njobs ? njobs : 1)));
+  HIP_TRY(hipMemcpy(d_jobs, h_jobs, sizeof(PackJob) * njobs,
+                    hipMemcpyHostToDevice));
+  int blocks = (int)bg_imin64(njobs, BG_MAX_BLOCKS);
+  if (blocks == 0) blocks = 1;
+  hipLaunchKernelGGL(k_pack_blocks, dim3(blocks), dim3(BG_BLOCK), 0, 0, d_jobs,
+                     njobs);
+  HIP_TRY(hipGetLastError());
+  (void)pool_release(d_jobs);
+  return BG_OK;
+}

@@ -289,9 +289,9 @@ class GpuResidentShuffleStage(GpuQueryStageExecutor):
 
 
 def _encode_partitions_gpu(self, schema, offsets, cols, out_bufs):
-    """Device-LZ4 the partition-major column slices and frame them with
-    the handwritten IPC writer; the host only concatenates compressed
-    bytes."""
+    """Device-LZ4 the partition-major column slices, PACK the frame bodies
+    on device, and download only the compressed bytes (one D2H per
+    buffer); the handwritten IPC writer supplies the metadata."""
     import ctypes
     import struct
     ctx = self.ctx
@@ -302,8 +302,8 @@ def _encode_partitions_gpu(self, schema, offsets, cols, out_bufs):
         if m == 0:
             streams.append(b"")
             continue
-        parts = []
         nodes = []
+        buffer_parts = []
         for ci in range(len(cols)):
             esz = gpu._DT_SIZE[cols[ci].dtype]
             length = m * esz
@@ -316,17 +316,32 @@ def _encode_partitions_gpu(self, schema, offsets, cols, out_bufs):
                 src_ptr, ctypes.c_int64(length), slots.ptr,
                 sizes.ctypes.data_as(ctypes.POINTER(ctypes.c_int64)),
                 ctypes.byref(nb)), "bg_lz4_compress")
-            slot_bytes = slots.download(
-                np.uint8, max(nblocks * 65544, 1)).tobytes()
-            frame = gpu.lz4_frame_assemble(sizes[:nblocks], slot_bytes,
-                                           length)
-            parts.append((None, struct.pack("<q", length) + frame))
+            # frame-body layout: [u32 size][block]... packed on device
+            jobs = (gpu.BgPackJob * nblocks)()
+            dst_off = 0
+            for i in range(nblocks):
+                blen = min(65536, length - i * 65536)
+                sz = int(sizes[i])
+                payload = blen if sz < 0 else sz
+                word = (blen | 0x80000000) if sz < 0 else sz
+                jobs[i] = gpu.BgPackJob(
+                    ctypes.c_void_p(slots.ptr.value + i * 65544).value,
+                    None, payload, word, 0)
+                dst_off += 4 + payload
+            packed = ctx.alloc(max(dst_off, 8))
+            off = 0
+            for i in range(nblocks):
+                jobs[i].d_dst = packed.ptr.value + off
+                off += 4 + jobs[i].nbytes
+            gpu._check(ctx.L.bg_pack_blocks(jobs, ctypes.c_int64(nblocks)),
+                       "bg_pack_blocks")
+            body = packed.download(np.uint8, dst_off).tobytes()
+            frame = gpu.LZ4_FRAME_HEADER + body + b"\x00\x00\x00\x00"
             nodes.append((m, 0))
+            buffer_parts.append(None)
+            buffer_parts.append(struct.pack("<q", length) + frame)
             slots.free()
-        buffer_parts = []
-        for validity, data_part in parts:
-            buffer_parts.append(validity)
-            buffer_parts.append(data_part)
+            packed.free()
         streams.append(bgipc.stream_from_compressed_batches(
             schema, [(m, nodes, buffer_parts)]))
     return streams

@@ -516,3 +516,9 @@ def lz4_frame_assemble(sizes, slot_bytes: bytes, length: int) -> bytes:
 
 
 GpuStageContext.lz4_compress = _ctx_lz4_compress
+
+
+class BgPackJob(ctypes.Structure):
+    _fields_ = [("d_src", ctypes.c_void_p), ("d_dst", ctypes.c_void_p),
+                ("nbytes", ctypes.c_int64), ("size_word", ctypes.c_uint32),
+                ("_pad", ctypes.c_uint32)]

@@ -314,6 +314,15 @@ int bg_dict_indices_batch(const void* h_jobs, int64_t njobs);
 #define BG_LZ4_SLOT_STRIDE 65544
 int bg_lz4_compress(const void* d_src, int64_t len, void* d_out_slots,
                     int64_t* h_block_sizes, int64_t* out_nblocks);
+/* assemble [u32 size][block] sequences on device at precomputed offsets */
+typedef struct {
+  const void* d_src;
+  void* d_dst;        /* at the 4-byte size word */
+  int64_t nbytes;
+  uint32_t size_word; /* high bit = stored block */
+  uint32_t _pad;
+} bg_pack_job;
+int bg_pack_blocks(const void* h_jobs, int64_t njobs);
 
 /* ---- fused filter+aggregate stages ---- */
 
