@@ -2624,23 +2624,12 @@ extern "C" int bg_gather_bits(const uint8_t* d_valid, const uint32_t* d_idx,
 #define LZ4_HASH_LOG 12
 #define LZ4_SLOT_STRIDE (LZ4_BLOCK + 8)
 
-__global__ void k_lz4_compress(const uint8_t* src, int64_t len,
-                               uint8_t* out_slots, int64_t slot_stride,
-                               int64_t* block_sizes, int64_t nblocks) {
-  extern __shared__ __attribute__((aligned(16))) char smem_raw[];
-  uint16_t* tab_all = reinterpret_cast<uint16_t*>(smem_raw);
-  const int wave_in_block = threadIdx.x / BG_WAVE;
-  uint16_t* tab = tab_all + (size_t)wave_in_block * (1 << LZ4_HASH_LOG);
-  const int64_t wave_global =
-      ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) / BG_WAVE;
-  const int64_t nwaves = ((int64_t)gridDim.x * blockDim.x) / BG_WAVE;
+// wave-cooperative single-block compress; returns csize or -blen (stored)
+__device__ int64_t lz4_compress_one(const uint8_t* s, int32_t blen,
+                                    uint8_t* d, uint16_t* tab) {
   const int lane = lane_id();
-  for (int64_t b = wave_global; b < nblocks; b += nwaves) {
-    const int64_t boff = b * LZ4_BLOCK;
-    const int32_t blen = (int32_t)(len - boff < LZ4_BLOCK ? len - boff
-                                                          : LZ4_BLOCK);
-    const uint8_t* s = src + boff;
-    uint8_t* d = out_slots + b * slot_stride;
+  int64_t result = 0;
+  {
     // parallel table clear (0xffff = empty)
     for (int i = lane; i < (1 << LZ4_HASH_LOG); i += BG_WAVE) tab[i] = 0xffff;
     __builtin_amdgcn_wave_barrier();
@@ -2698,21 +2687,83 @@ __global__ void k_lz4_compress(const uint8_t* src, int64_t len,
       }
       // trailing literals-only sequence
       if (!overflow) emit_seq(blen - anchor, -1, 0);
-      if (overflow || w >= blen) {
-        // stored block: raw copy, negative size signals "uncompressed"
-        block_sizes[b] = -(int64_t)blen;
-      } else {
-        block_sizes[b] = (int64_t)w;
-      }
+      result = (overflow || w >= blen) ? -(int64_t)blen : (int64_t)w;
     }
-    __builtin_amdgcn_wave_barrier();
+    result = bcast64(result);
     // stored blocks: the whole wave copies raw bytes into the slot
-    const int64_t bs = block_sizes[b];
-    __builtin_amdgcn_wave_barrier();
-    if (bs < 0) {
+    if (result < 0)
       for (int32_t i = lane; i < blen; i += BG_WAVE) d[i] = s[i];
-    }
   }
+  return result;
+}
+
+__global__ void k_lz4_compress(const uint8_t* src, int64_t len,
+                               uint8_t* out_slots, int64_t slot_stride,
+                               int64_t* block_sizes, int64_t nblocks) {
+  extern __shared__ __attribute__((aligned(16))) char smem_raw[];
+  uint16_t* tab_all = reinterpret_cast<uint16_t*>(smem_raw);
+  const int wave_in_block = threadIdx.x / BG_WAVE;
+  uint16_t* tab = tab_all + (size_t)wave_in_block * (1 << LZ4_HASH_LOG);
+  const int64_t wave_global =
+      ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) / BG_WAVE;
+  const int64_t nwaves = ((int64_t)gridDim.x * blockDim.x) / BG_WAVE;
+  for (int64_t b = wave_global; b < nblocks; b += nwaves) {
+    const int64_t boff = b * LZ4_BLOCK;
+    const int32_t blen = (int32_t)(len - boff < LZ4_BLOCK ? len - boff
+                                                          : LZ4_BLOCK);
+    const int64_t r = lz4_compress_one(src + boff, blen,
+                                       out_slots + b * slot_stride, tab);
+    if (lane_id() == 0) block_sizes[b] = r;
+  }
+}
+
+// flat batched form: one wave per (buffer, block) pair across many buffers
+struct Lz4BlockJob {
+  const uint8_t* src;
+  uint8_t* dst_slot;
+  int32_t blen;
+  int32_t _pad;
+};
+
+__global__ void k_lz4_compress_flat(const Lz4BlockJob* jobs, int64_t njobs,
+                                    int64_t* block_sizes) {
+  extern __shared__ __attribute__((aligned(16))) char smem_raw[];
+  uint16_t* tab_all = reinterpret_cast<uint16_t*>(smem_raw);
+  const int wave_in_block = threadIdx.x / BG_WAVE;
+  uint16_t* tab = tab_all + (size_t)wave_in_block * (1 << LZ4_HASH_LOG);
+  const int64_t wave_global =
+      ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) / BG_WAVE;
+  const int64_t nwaves = ((int64_t)gridDim.x * blockDim.x) / BG_WAVE;
+  for (int64_t j = wave_global; j < njobs; j += nwaves) {
+    const Lz4BlockJob job = jobs[j];
+    const int64_t r = lz4_compress_one(job.src, job.blen, job.dst_slot, tab);
+    if (lane_id() == 0) block_sizes[j] = r;
+  }
+}
+
+extern "C" int bg_lz4_compress_flat(const void* h_jobs, int64_t njobs,
+                                    int64_t* h_block_sizes) {
+  REQUIRE_INIT();
+  Lz4BlockJob* d_jobs;
+  int64_t* d_sizes;
+  HIP_TRY(pool_malloc((void**)&d_jobs, sizeof(Lz4BlockJob) * (njobs ? njobs : 1)));
+  HIP_TRY(pool_malloc((void**)&d_sizes, sizeof(int64_t) * (njobs ? njobs : 1)));
+  HIP_TRY(hipMemcpy(d_jobs, h_jobs, sizeof(Lz4BlockJob) * njobs,
+                    hipMemcpyHostToDevice));
+  const int waves_per_block = BG_BLOCK / BG_WAVE;
+  const size_t lds = (size_t)waves_per_block * (1 << LZ4_HASH_LOG) *
+                     sizeof(uint16_t);
+  int blocks = (int)bg_imin64((njobs + waves_per_block - 1) / waves_per_block,
+                              BG_MAX_BLOCKS);
+  if (blocks == 0) blocks = 1;
+  hipLaunchKernelGGL(k_lz4_compress_flat, dim3(blocks), dim3(BG_BLOCK), lds, 0,
+                     d_jobs, njobs, d_sizes);
+  HIP_TRY(hipGetLastError());
+  HIP_TRY(hipMemcpy(h_block_sizes, d_sizes, sizeof(int64_t) * njobs,
+                    hipMemcpyDeviceToHost));
+  (void)pool_release(d_jobs);
+  (void)pool_release(d_sizes);
+  return BG_OK;
 }
 
 extern "C" int bg_lz4_compress(const void* d_src, int64_t len,

@@ -289,12 +289,75 @@ class GpuResidentShuffleStage(GpuQueryStageExecutor):
 
 
 def _encode_partitions_gpu(self, schema, offsets, cols, out_bufs):
-    """Device-LZ4 the partition-major column slices, PACK the frame bodies
-    on device, and download only the compressed bytes (one D2H per
-    buffer); the handwritten IPC writer supplies the metadata."""
+    """Device-LZ4 every partition's column slices in ONE flat launch, pack
+    the frame bodies on device in ONE launch, then download only the
+    compressed bytes (one D2H per (partition, column) buffer); the
+    handwritten IPC writer supplies the metadata."""
     import ctypes
     import struct
     ctx = self.ctx
+    # phase 1: enumerate every (partition, column, 64KiB block)
+    buf_meta = []   # per (p, ci): (length, nblocks, first_job)
+    jobs = []
+    slot_cursor = 0
+    for p in range(self.k):
+        lo, hi = int(offsets[p]), int(offsets[p + 1])
+        m = hi - lo
+        if m == 0:
+            continue
+        for ci in range(len(cols)):
+            esz = gpu._DT_SIZE[cols[ci].dtype]
+            length = m * esz
+            nblocks = (length + 65536 - 1) // 65536
+            buf_meta.append((p, ci, m, length, nblocks, len(jobs),
+                             slot_cursor))
+            for i in range(nblocks):
+                blen = min(65536, length - i * 65536)
+                jobs.append((out_bufs[ci].ptr.value + lo * esz + i * 65536,
+                             slot_cursor, blen))
+                slot_cursor += 65544
+    if not jobs:
+        return [b""] * self.k
+    slots = ctx.alloc(max(slot_cursor, 8))
+    jarr = (gpu.BgLz4BlockJob * len(jobs))()
+    for i, (src, soff, blen) in enumerate(jobs):
+        jarr[i] = gpu.BgLz4BlockJob(src, slots.ptr.value + soff, blen, 0)
+    sizes = np.zeros(len(jobs), dtype=np.int64)
+    gpu._check(ctx.L.bg_lz4_compress_flat(
+        jarr, ctypes.c_int64(len(jobs)),
+        sizes.ctypes.data_as(ctypes.POINTER(ctypes.c_int64))),
+        "bg_lz4_compress_flat")
+
+    # phase 2: pack each buffer's [u32 size][block] sequence contiguously
+    pack_jobs = []
+    packed_layout = {}  # (p, ci) -> (packed_off, body_len, length, m)
+    packed_cursor = 0
+    for (p, ci, m, length, nblocks, job0, _sc) in buf_meta:
+        body = 0
+        for i in range(nblocks):
+            sz = int(sizes[job0 + i])
+            payload = -sz if sz < 0 else sz
+            body += 4 + payload
+        packed_layout[(p, ci)] = (packed_cursor, body, length, m)
+        off = packed_cursor
+        for i in range(nblocks):
+            sz = int(sizes[job0 + i])
+            blen = min(65536, length - i * 65536)
+            payload = blen if sz < 0 else sz
+            word = (blen | 0x80000000) if sz < 0 else sz
+            pack_jobs.append((jobs[job0 + i][1], off, payload, word))
+            off += 4 + payload
+        packed_cursor += body
+    packed = ctx.alloc(max(packed_cursor, 8))
+    parr = (gpu.BgPackJob * len(pack_jobs))()
+    for i, (soff, doff, payload, word) in enumerate(pack_jobs):
+        parr[i] = gpu.BgPackJob(slots.ptr.value + soff,
+                                packed.ptr.value + doff, payload, word, 0)
+    gpu._check(ctx.L.bg_pack_blocks(parr, ctypes.c_int64(len(pack_jobs))),
+               "bg_pack_blocks")
+
+    # phase 3: one D2H of the whole packed region, then assemble streams
+    packed_host = packed.download(np.uint8, packed_cursor).tobytes()
     streams = []
     for p in range(self.k):
         lo, hi = int(offsets[p]), int(offsets[p + 1])
@@ -305,45 +368,17 @@ def _encode_partitions_gpu(self, schema, offsets, cols, out_bufs):
         nodes = []
         buffer_parts = []
         for ci in range(len(cols)):
-            esz = gpu._DT_SIZE[cols[ci].dtype]
-            length = m * esz
-            src_ptr = ctypes.c_void_p(out_bufs[ci].ptr.value + lo * esz)
-            nblocks = (length + 65536 - 1) // 65536
-            slots = ctx.alloc(max(nblocks * 65544, 8))
-            sizes = np.zeros(max(nblocks, 1), dtype=np.int64)
-            nb = ctypes.c_int64()
-            gpu._check(ctx.L.bg_lz4_compress(
-                src_ptr, ctypes.c_int64(length), slots.ptr,
-                sizes.ctypes.data_as(ctypes.POINTER(ctypes.c_int64)),
-                ctypes.byref(nb)), "bg_lz4_compress")
-            # frame-body layout: [u32 size][block]... packed on device
-            jobs = (gpu.BgPackJob * nblocks)()
-            dst_off = 0
-            for i in range(nblocks):
-                blen = min(65536, length - i * 65536)
-                sz = int(sizes[i])
-                payload = blen if sz < 0 else sz
-                word = (blen | 0x80000000) if sz < 0 else sz
-                jobs[i] = gpu.BgPackJob(
-                    ctypes.c_void_p(slots.ptr.value + i * 65544).value,
-                    None, payload, word, 0)
-                dst_off += 4 + payload
-            packed = ctx.alloc(max(dst_off, 8))
-            off = 0
-            for i in range(nblocks):
-                jobs[i].d_dst = packed.ptr.value + off
-                off += 4 + jobs[i].nbytes
-            gpu._check(ctx.L.bg_pack_blocks(jobs, ctypes.c_int64(nblocks)),
-                       "bg_pack_blocks")
-            body = packed.download(np.uint8, dst_off).tobytes()
-            frame = gpu.LZ4_FRAME_HEADER + body + b"\x00\x00\x00\x00"
+            poff, body_len, length, m2 = packed_layout[(p, ci)]
+            frame = (gpu.LZ4_FRAME_HEADER +
+                     packed_host[poff:poff + body_len] +
+                     b"\x00\x00\x00\x00")
             nodes.append((m, 0))
             buffer_parts.append(None)
             buffer_parts.append(struct.pack("<q", length) + frame)
-            slots.free()
-            packed.free()
         streams.append(bgipc.stream_from_compressed_batches(
             schema, [(m, nodes, buffer_parts)]))
+    slots.free()
+    packed.free()
     return streams
 
 

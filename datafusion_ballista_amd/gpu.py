@@ -522,3 +522,8 @@ class BgPackJob(ctypes.Structure):
     _fields_ = [("d_src", ctypes.c_void_p), ("d_dst", ctypes.c_void_p),
                 ("nbytes", ctypes.c_int64), ("size_word", ctypes.c_uint32),
                 ("_pad", ctypes.c_uint32)]
+
+
+class BgLz4BlockJob(ctypes.Structure):
+    _fields_ = [("d_src", ctypes.c_void_p), ("d_dst_slot", ctypes.c_void_p),
+                ("blen", ctypes.c_int32), ("_pad", ctypes.c_int32)]

@@ -314,6 +314,15 @@ int bg_dict_indices_batch(const void* h_jobs, int64_t njobs);
 #define BG_LZ4_SLOT_STRIDE 65544
 int bg_lz4_compress(const void* d_src, int64_t len, void* d_out_slots,
                     int64_t* h_block_sizes, int64_t* out_nblocks);
+/* flat batched compression over many buffers' blocks in one launch */
+typedef struct {
+  const void* d_src;
+  void* d_dst_slot;
+  int32_t blen;
+  int32_t _pad;
+} bg_lz4_block_job;
+int bg_lz4_compress_flat(const void* h_jobs, int64_t njobs,
+                         int64_t* h_block_sizes);
 /* assemble [u32 size][block] sequences on device at precomputed offsets */
 typedef struct {
   const void* d_src;
