@@ -2665,9 +2665,10 @@ __device__ int64_t lz4_compress_one(const uint8_t* s, int32_t blen,
         }
       };
       if (blen >= 13) {
-        while (pos <= mflimit && !overflow) {
-          uint32_t v;
-          __builtin_memcpy(&v, s + pos, 4);
+        int32_t misses = 0;  // LZ4-style skip acceleration: after a run of
+        while (pos <= mflimit && !overflow) {  // misses, stride grows so
+          uint32_t v;                          // incompressible data is
+          __builtin_memcpy(&v, s + pos, 4);    // skimmed, not crawled
           const uint32_t h = (v * 2654435761u) >> (32 - LZ4_HASH_LOG);
           const int32_t cand = tab[h] == 0xffff ? -1 : (int32_t)tab[h];
           tab[h] = (uint16_t)pos;
@@ -2680,8 +2681,10 @@ __device__ int64_t lz4_compress_one(const uint8_t* s, int32_t blen,
             emit_seq(pos - anchor, mlen, pos - cand);
             pos += mlen;
             anchor = pos;
+            misses = 0;
           } else {
-            ++pos;
+            pos += 1 + (misses >> 6);
+            ++misses;
           }
         }
       }
