@@ -2836,3 +2836,302 @@ extern "C" int bg_pack_blocks(const void* h_jobs, int64_t njobs) {
   (void)pool_release(d_jobs);
   return BG_OK;
 }
+
+// ---------------------------------------------------------------------------
+// Fused hash-repartition materialiser (k <= 64, <= 4 fixed-width payload
+// columns): ONE pass recomputes the key hash (cheaper than writing/reading
+// a pids array), ranks rows per partition with ballots, stages rows in
+// per-partition LDS tiles and flushes 64-row bursts per column — write
+// combining turns the k scattered payload streams into contiguous
+// wave-wide stores.  Start offsets come from the same hist+scan prework
+// as the stable split (chunk-contiguous => stable order preserved).
+// ---------------------------------------------------------------------------
+#define FM_TILE 64
+#define FM_WAVES 2  // waves per 128-thread block (LDS budget)
+
+struct FmArgs {
+  KeyArgs keys;
+  int ncols;
+  struct {
+    const void* data;
+    void* out;
+    int esz;  // 1/4/8/16
+  } c[BG_MAX_KEYS];
+};
+
+__global__ void __launch_bounds__(FM_WAVES * BG_WAVE)
+k_fused_materialize(FmArgs args, int64_t n, uint32_t k, const i64* start,
+                    int64_t nchunks, uint32_t* out_idx, uint32_t* rank_out) {
+  // LDS: per wave: [k][FM_TILE] u32 row-index tile + per-col value tiles
+  extern __shared__ __attribute__((aligned(16))) char smem_raw[];
+  const int wave_in_block = threadIdx.x / BG_WAVE;
+  const int lane = lane_id();
+  // layout per wave: idx tile (k*FM_TILE*4) + cols (k*FM_TILE*esz each)
+  size_t wave_bytes = (size_t)k * FM_TILE * 4;
+  for (int c = 0; c < args.ncols; ++c)
+    wave_bytes += (size_t)k * FM_TILE * args.c[c].esz;
+  char* base = smem_raw + (size_t)wave_in_block * wave_bytes;
+  uint32_t* t_idx = reinterpret_cast<uint32_t*>(base);
+  char* t_cols[BG_MAX_KEYS];
+  {
+    char* p = base + (size_t)k * FM_TILE * 4;
+    for (int c = 0; c < args.ncols; ++c) {
+      t_cols[c] = p;
+      p += (size_t)k * FM_TILE * args.c[c].esz;
+    }
+  }
+  const int64_t wave_global =
+      ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) / BG_WAVE;
+  const int64_t nwaves = ((int64_t)gridDim.x * blockDim.x) / BG_WAVE;
+  for (int64_t ch = wave_global; ch < nchunks; ch += nwaves) {
+    // register cursors: lane p holds partition p's next global slot and
+    // the count of rows THIS WAVE has staged-but-not-flushed (cursors are
+    // not tile-aligned: tile slot 0 sits at global position cursor - fill)
+    i64 my_cursor = (lane < (int)k)
+                        ? start[(int64_t)lane * nchunks + ch]
+                        : 0;
+    int my_fill = 0;
+    const int64_t r0 = ch * PS_ROWS_PER_WAVE;
+    const int64_t r1 = min(r0 + (int64_t)PS_ROWS_PER_WAVE, n);
+    for (int64_t rb = r0; rb < r1; rb += BG_WAVE) {
+      const int64_t r = rb + lane;
+      const bool active = r < r1;
+      uint32_t pid = 0xffffffffu;
+      if (active) {
+        const u64 h = hash_keys_row(args.keys, r);
+        pid = (uint32_t)(h % (u64)k);
+      }
+      for (uint32_t p = 0; p < k; ++p) {
+        const u64 m = __ballot(active && pid == p);
+        if (!m) continue;
+        const i64 cur =
+            ((i64)(uint32_t)__builtin_amdgcn_readlane(
+                 (int)(uint32_t)((u64)my_cursor >> 32), p)
+             << 32) |
+            (i64)(uint32_t)__builtin_amdgcn_readlane(
+                (int)(uint32_t)((u64)my_cursor & 0xffffffff), p);
+        const int fill = __builtin_amdgcn_readlane(my_fill, p);
+        const int cnt = __popcll(m);
+        const int rank = __popcll(m & ((1ull << lane) - 1));
+        const int slot = fill + rank;
+        const bool mine = active && pid == p;
+        if (mine && slot < FM_TILE) {
+          t_idx[p * FM_TILE + slot] = (uint32_t)r;
+          for (int c = 0; c < args.ncols; ++c) {
+            switch (args.c[c].esz) {
+              case 1:
+                t_cols[c][p * FM_TILE + slot] =
+                    reinterpret_cast<const uint8_t*>(args.c[c].data)[r];
+                break;
+              case 4:
+                reinterpret_cast<uint32_t*>(t_cols[c])[p * FM_TILE + slot] =
+                    reinterpret_cast<const uint32_t*>(args.c[c].data)[r];
+                break;
+              case 8:
+                reinterpret_cast<u64*>(t_cols[c])[p * FM_TILE + slot] =
+                    reinterpret_cast<const u64*>(args.c[c].data)[r];
+                break;
+              case 16:
+                reinterpret_cast<ulong2*>(t_cols[c])[p * FM_TILE + slot] =
+                    reinterpret_cast<const ulong2*>(args.c[c].data)[r];
+                break;
+            }
+          }
+        }
+        if (mine && rank_out) rank_out[r] = (uint32_t)(cur + rank);
+        __builtin_amdgcn_wave_barrier();
+        // tile crossed a 64 boundary? flush the completed tile
+        const int new_fill = fill + cnt;
+        if (new_fill >= FM_TILE) {
+          const i64 flush_base = cur - fill;  // global position of slot 0
+          for (int i = lane; i < FM_TILE; i += BG_WAVE)
+            out_idx[flush_base + i] = t_idx[p * FM_TILE + i];
+          for (int c = 0; c < args.ncols; ++c) {
+            switch (args.c[c].esz) {
+              case 1:
+                for (int i = lane; i < FM_TILE; i += BG_WAVE)
+                  reinterpret_cast<uint8_t*>(args.c[c].out)[flush_base + i] =
+                      t_cols[c][p * FM_TILE + i];
+                break;
+              case 4:
+                for (int i = lane; i < FM_TILE; i += BG_WAVE)
+                  reinterpret_cast<uint32_t*>(args.c[c].out)[flush_base + i] =
+                      reinterpret_cast<uint32_t*>(t_cols[c])[p * FM_TILE + i];
+                break;
+              case 8:
+                for (int i = lane; i < FM_TILE; i += BG_WAVE)
+                  reinterpret_cast<u64*>(args.c[c].out)[flush_base + i] =
+                      reinterpret_cast<u64*>(t_cols[c])[p * FM_TILE + i];
+                break;
+              case 16:
+                for (int i = lane; i < FM_TILE; i += BG_WAVE)
+                  reinterpret_cast<ulong2*>(args.c[c].out)[flush_base + i] =
+                      reinterpret_cast<ulong2*>(t_cols[c])[p * FM_TILE + i];
+                break;
+            }
+          }
+          __builtin_amdgcn_wave_barrier();
+          // restage this group's overflow rows into the emptied tile
+          if (mine && slot >= FM_TILE) {
+            const int slot2 = slot - FM_TILE;
+            t_idx[p * FM_TILE + slot2] = (uint32_t)r;
+            for (int c = 0; c < args.ncols; ++c) {
+              switch (args.c[c].esz) {
+                case 1:
+                  t_cols[c][p * FM_TILE + slot2] =
+                      reinterpret_cast<const uint8_t*>(args.c[c].data)[r];
+                  break;
+                case 4:
+                  reinterpret_cast<uint32_t*>(t_cols[c])[p * FM_TILE + slot2] =
+                      reinterpret_cast<const uint32_t*>(args.c[c].data)[r];
+                  break;
+                case 8:
+                  reinterpret_cast<u64*>(t_cols[c])[p * FM_TILE + slot2] =
+                      reinterpret_cast<const u64*>(args.c[c].data)[r];
+                  break;
+                case 16:
+                  reinterpret_cast<ulong2*>(t_cols[c])[p * FM_TILE + slot2] =
+                      reinterpret_cast<const ulong2*>(args.c[c].data)[r];
+                  break;
+              }
+            }
+          }
+        }
+        // cursor/fill advance on owning lane
+        if (lane == (int)p) {
+          my_cursor += cnt;
+          my_fill = new_fill >= FM_TILE ? new_fill - FM_TILE : new_fill;
+        }
+        __builtin_amdgcn_wave_barrier();
+      }
+    }
+    // chunk done: flush partial tiles (fill rows each)
+    for (uint32_t p = 0; p < k; ++p) {
+      const i64 cur =
+          ((i64)(uint32_t)__builtin_amdgcn_readlane(
+               (int)(uint32_t)((u64)my_cursor >> 32), p)
+           << 32) |
+          (i64)(uint32_t)__builtin_amdgcn_readlane(
+              (int)(uint32_t)((u64)my_cursor & 0xffffffff), p);
+      const int fill = __builtin_amdgcn_readlane(my_fill, p);
+      if (!fill) continue;
+      const i64 flush_base = cur - fill;
+      // tile slots [0, fill) hold this wave's staged rows for positions
+      // [flush_base, cur); flush them
+      for (int i = lane; i < fill; i += BG_WAVE)
+        out_idx[flush_base + i] = t_idx[p * FM_TILE + i];
+      for (int c = 0; c < args.ncols; ++c) {
+        switch (args.c[c].esz) {
+          case 1:
+            for (int i = lane; i < fill; i += BG_WAVE)
+              reinterpret_cast<uint8_t*>(args.c[c].out)[flush_base + i] =
+                  t_cols[c][p * FM_TILE + i];
+            break;
+          case 4:
+            for (int i = lane; i < fill; i += BG_WAVE)
+              reinterpret_cast<uint32_t*>(args.c[c].out)[flush_base + i] =
+                  reinterpret_cast<uint32_t*>(t_cols[c])[p * FM_TILE + i];
+            break;
+          case 8:
+            for (int i = lane; i < fill; i += BG_WAVE)
+              reinterpret_cast<u64*>(args.c[c].out)[flush_base + i] =
+                  reinterpret_cast<u64*>(t_cols[c])[p * FM_TILE + i];
+            break;
+          case 16:
+            for (int i = lane; i < fill; i += BG_WAVE)
+              reinterpret_cast<ulong2*>(args.c[c].out)[flush_base + i] =
+                  reinterpret_cast<ulong2*>(t_cols[c])[p * FM_TILE + i];
+            break;
+        }
+      }
+    }
+  }
+}
+
+extern "C" int bg_hash_repartition_fused(const bg_column* key_cols,
+                                         int32_t nkeys,
+                                         const bg_column* payload_cols,
+                                         int32_t ncols, int64_t n, uint32_t k,
+                                         uint32_t* d_indices,
+                                         int64_t* d_offsets, uint32_t* d_rank,
+                                         void** d_out) {
+  REQUIRE_INIT();
+  if (k == 0 || k > BG_WAVE)
+    return set_err(BG_ERR_INVALID, "fused path needs k in [1,64]");
+  if (ncols > BG_MAX_KEYS)
+    return set_err(BG_ERR_INVALID, "fused path: <= 4 payload cols");
+  KeyArgs keys{};
+  keys.nkeys = nkeys;
+  for (int i = 0; i < nkeys; ++i) {
+    keys.k[i].data = key_cols[i].d_data;
+    keys.k[i].valid = key_cols[i].d_validity;
+    keys.k[i].offsets = key_cols[i].d_offsets;
+    keys.k[i].dtype = key_cols[i].dtype;
+  }
+  FmArgs a{};
+  a.keys = keys;
+  a.ncols = ncols;
+  size_t wave_bytes = (size_t)k * FM_TILE * 4;
+  for (int i = 0; i < ncols; ++i) {
+    a.c[i].data = payload_cols[i].d_data;
+    a.c[i].out = d_out[i];
+    int esz = (int)dtype_size(payload_cols[i].dtype);
+    if (esz != 1 && esz != 4 && esz != 8 && esz != 16)
+      return set_err(BG_ERR_UNSUPPORTED, "fused payload dtype");
+    a.c[i].esz = esz;
+    wave_bytes += (size_t)k * FM_TILE * esz;
+  }
+  const size_t lds = wave_bytes * FM_WAVES;
+  if (lds > 160 * 1024)
+    return set_err(BG_ERR_INVALID, "fused path LDS budget exceeded");
+
+  // hist + scan prework (same stability contract as the split)
+  const int64_t nchunks = (n + PS_ROWS_PER_WAVE - 1) / PS_ROWS_PER_WAVE;
+  const int64_t hist_len = (int64_t)k * (nchunks ? nchunks : 1);
+  uint32_t* d_pids;
+  u64* d_hist;
+  i64* d_start;
+  HIP_TRY(pool_malloc((void**)&d_pids, sizeof(uint32_t) * (n ? n : 1)));
+  HIP_TRY(pool_malloc((void**)&d_hist, sizeof(u64) * hist_len));
+  HIP_TRY(pool_malloc((void**)&d_start, sizeof(i64) * hist_len));
+  uint64_t* d_hashes;
+  HIP_TRY(pool_malloc((void**)&d_hashes, sizeof(u64) * (n ? n : 1)));
+  int rc = bg_hash_columns(key_cols, nkeys, n, d_hashes);
+  if (rc == BG_OK) rc = bg_partition_ids(d_hashes, n, k, d_pids);
+  (void)pool_release(d_hashes);
+  if (rc == BG_OK) {
+    const int waves_per_block = BG_BLOCK / BG_WAVE;
+    const size_t lds_hist = (size_t)waves_per_block * k * sizeof(uint32_t);
+    int blocks = (int)bg_imin64(nchunks, BG_MAX_BLOCKS);
+    if (blocks == 0) blocks = 1;
+    hipLaunchKernelGGL(k_part_hist, dim3(blocks), dim3(BG_BLOCK), lds_hist, 0,
+                       d_pids, n, k, d_hist, nchunks);
+    rc = scan_exclusive_i64(d_hist, hist_len, d_start, nullptr);
+  }
+  if (rc == BG_OK) {
+    hipLaunchKernelGGL(k_extract_offsets, dim3(1), dim3(BG_BLOCK), 0, 0,
+                       d_start, nchunks, k, n, d_offsets);
+    const int64_t fm_threads = FM_WAVES * BG_WAVE;
+    int blocks = (int)bg_imin64(nchunks, BG_MAX_BLOCKS);
+    if (blocks == 0) blocks = 1;
+    hipEvent_t ev0, ev1;
+    (void)hipEventCreate(&ev0);
+    (void)hipEventCreate(&ev1);
+    (void)hipEventRecord(ev0, 0);
+    hipLaunchKernelGGL(k_fused_materialize, dim3(blocks), dim3(fm_threads),
+                       lds, 0, a, n, k, d_start, nchunks, d_indices, d_rank);
+    hipError_t e = hipGetLastError();
+    if (e != hipSuccess) rc = set_hip_err(e, "fused materialize");
+    (void)hipEventRecord(ev1, 0);
+    (void)hipEventSynchronize(ev1);
+    float ms = 0.f;
+    (void)hipEventElapsedTime(&ms, ev0, ev1);
+    g_last_kernel_ms = (double)ms;
+    (void)hipEventDestroy(ev0);
+    (void)hipEventDestroy(ev1);
+  }
+  (void)pool_release(d_pids);
+  (void)pool_release(d_hist);
+  (void)pool_release(d_start);
+  return rc;
+}

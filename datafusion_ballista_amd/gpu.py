@@ -527,3 +527,27 @@ class BgPackJob(ctypes.Structure):
 class BgLz4BlockJob(ctypes.Structure):
     _fields_ = [("d_src", ctypes.c_void_p), ("d_dst_slot", ctypes.c_void_p),
                 ("blen", ctypes.c_int32), ("_pad", ctypes.c_int32)]
+
+
+def _ctx_hash_repartition_fused(self, key_cols, payload_cols, n: int, k: int):
+    """Fused one-pass materialiser (k<=64, <=4 fixed-width payload cols):
+    same outputs/stable order as hash_repartition."""
+    idx = self.alloc(max(4 * n, 4))
+    offs = self.alloc(8 * (k + 1))
+    rank = self.alloc(max(4 * n, 4))
+    outs = []
+    optrs = (ctypes.c_void_p * len(payload_cols))()
+    for i, c in enumerate(payload_cols):
+        esz = _DT_SIZE[c.dtype]
+        b = self.alloc(max(esz * n, esz))
+        outs.append(b)
+        optrs[i] = b.ptr.value
+    karr = (BgColumn * len(key_cols))(*key_cols)
+    parr = (BgColumn * len(payload_cols))(*payload_cols)
+    _check(self.L.bg_hash_repartition_fused(
+        karr, len(key_cols), parr, len(payload_cols), ctypes.c_int64(n), k,
+        idx.ptr, offs.ptr, rank.ptr, optrs), "bg_hash_repartition_fused")
+    return idx, offs, outs
+
+
+GpuStageContext.hash_repartition_fused = _ctx_hash_repartition_fused

@@ -333,6 +333,17 @@ typedef struct {
 } bg_pack_job;
 int bg_pack_blocks(const void* h_jobs, int64_t njobs);
 
+/* Fused repartition materialiser (k <= 64, <= 4 fixed-width payload
+ * columns): one pass re-hashes keys, ranks rows per partition with
+ * ballots, and write-combines k scattered payload streams through
+ * per-partition LDS tiles (64-row flushes).  Same outputs and stable
+ * order as bg_hash_repartition. */
+int bg_hash_repartition_fused(const bg_column* key_cols, int32_t nkeys,
+                              const bg_column* payload_cols, int32_t ncols,
+                              int64_t n, uint32_t k, uint32_t* d_indices,
+                              int64_t* d_offsets, uint32_t* d_rank,
+                              void** d_out);
+
 /* ---- fused filter+aggregate stages ---- */
 
 /* TPC-H q6 stage 1 (scan+filter+aggregate, approved/q6.txt):

@@ -927,3 +927,31 @@ def test_sort_shuffle_stage_gpu_codec(ctx, tmp_path):
         gt = pa.Table.from_batches(got, schema=table.schema) if got else \
             table.schema.empty_table()
         assert gt.equals(wt), f"partition {p} mismatch"
+
+
+@pytest.mark.parametrize("n,k", [(1000, 4), (16384, 16), (16500, 16),
+                                 (500_000, 16), (300_000, 64), (5_000, 1)])
+def test_hash_repartition_fused_parity(ctx, n, k):
+    """The fused LDS-write-combining materialiser must produce bit-exact
+    the unfused path's outputs: indices, offsets, and every partition-major
+    payload buffer (stable order included)."""
+    rng = np.random.default_rng(n * 7 + k)
+    keys = rng.integers(-2**62, 2**62, size=n, dtype=np.int64)
+    vals8 = rng.integers(-10**9, 10**9, size=n, dtype=np.int64)
+    vals4 = rng.integers(-10**6, 10**6, size=n, dtype=np.int32)
+    dec16 = dec_bytes(rng.integers(-10**10, 10**10, size=n, dtype=np.int64))
+    kc, _ = ctx.upload_column(keys, gpu.BG_DT_INT64)
+    c8, _ = ctx.upload_column(vals8, gpu.BG_DT_INT64)
+    c4, _ = ctx.upload_column(vals4, gpu.BG_DT_INT32)
+    cd = ctx.column(gpu.BG_DT_DECIMAL128, ctx.upload(dec16), n)
+    payload = [kc, c8, c4, cd]
+
+    i1, o1, b1 = ctx.hash_repartition([kc], payload, n, k)
+    i2, o2, b2 = ctx.hash_repartition_fused([kc], payload, n, k)
+    ctx.synchronize()
+    assert np.array_equal(o1.download(np.int64, k + 1),
+                          o2.download(np.int64, k + 1))
+    assert np.array_equal(i1.download(np.uint32, n), i2.download(np.uint32, n))
+    for a, b, esz in zip(b1, b2, (8, 8, 4, 16)):
+        assert np.array_equal(a.download(np.uint8, n * esz),
+                              b.download(np.uint8, n * esz))
