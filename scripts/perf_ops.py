@@ -126,6 +126,23 @@ def main():
     report("repartition_pipeline_total", sec, whole_bytes,
            {"rows_per_s": n / sec})
 
+    # ---- fused one-pass repartition materialiser ----
+    fidx = ctx.alloc(4 * n)
+    foffs = ctx.alloc(8 * (k + 1))
+    frank = ctx.alloc(4 * n)
+    fouts = [ctx.alloc(8 * n), ctx.alloc(4 * n), ctx.alloc(4 * n),
+             ctx.alloc(8 * n)]
+    foptrs = (ctypes.c_void_p * 4)(*[b.ptr.value for b in fouts])
+    karr1 = (gpu.BgColumn * 1)(kc)
+    parr4 = (gpu.BgColumn * 4)(*payload)
+    def do_fused():
+        gpu._check(L.bg_hash_repartition_fused(
+            karr1, 1, parr4, 4, ctypes.c_int64(n), k, fidx.ptr, foffs.ptr,
+            frank.ptr, foptrs), "fused")
+    sec = timeit(ctx, do_fused)
+    report("repartition_fused_total", sec, whole_bytes,
+           {"rows_per_s": n / sec, "kernel_ms": L.bg_last_kernel_ms()})
+
     # ---- hash join: build 15M, probe 150M ----
     nb = 15_000_000
     bkeys = torch.randperm(nb, device=dev, dtype=torch.int64) + 1
