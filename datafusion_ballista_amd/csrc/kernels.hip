@@ -3135,3 +3135,175 @@ extern "C" int bg_hash_repartition_fused(const bg_column* key_cols,
   (void)pool_release(d_start);
   return rc;
 }
+
+// ---------------------------------------------------------------------------
+// LZ4-frame DECODE (device shuffle-read ingest — the decompress mirror of
+// bg_lz4_compress, so a GPU-resident next stage can consume shuffle bytes
+// without host decode).  One wave per frame: lane 0 parses the frame
+// header and each block's size word + the LZ4 sequences; 64 lanes execute
+// literal/match copies (pattern replication via modulo, as in Snappy).
+// Handles stored blocks (high-bit size) and linked or independent blocks
+// (the output is contiguous, so back-references past block starts work).
+// ---------------------------------------------------------------------------
+struct Lz4Frame {
+  const uint8_t* src;   // at the frame magic
+  uint8_t* dst;
+  int64_t src_len;
+  int64_t dst_cap;
+};
+
+__global__ void k_lz4_decompress(const Lz4Frame* frames, int64_t nframes,
+                                 int64_t* out_lens) {
+  const int64_t wave_global =
+      ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) / BG_WAVE;
+  const int64_t nwaves = ((int64_t)gridDim.x * blockDim.x) / BG_WAVE;
+  const int lane = lane_id();
+  for (int64_t f = wave_global; f < nframes; f += nwaves) {
+    const uint8_t* s = frames[f].src;
+    const uint8_t* send = s + frames[f].src_len;
+    uint8_t* base_d = frames[f].dst;
+    const int64_t dcap = frames[f].dst_cap;
+    int ok = 1;
+    int64_t si = 0, di = 0;
+    if (lane == 0) {
+      // magic + FLG/BD/HC; flags: content-size bit adds 8 bytes, dict-id 4
+      if (si + 7 > frames[f].src_len ||
+          s[0] != 0x04 || s[1] != 0x22 || s[2] != 0x4d || s[3] != 0x18) {
+        ok = 0;
+      } else {
+        const uint8_t flg = s[4];
+        si = 7;
+        if (flg & 0x08) si += 8;  // content size
+        if (flg & 0x01) si += 4;  // dict id
+        if (flg & 0x10) ok = 0;   // block checksums unsupported
+      }
+    }
+    ok = bcast32(ok);
+    si = bcast64(si);
+    while (ok) {
+      // lane 0 reads the block size word
+      int64_t bsize = 0;
+      int stored = 0;
+      if (lane == 0) {
+        if (si + 4 > frames[f].src_len) { ok = 0; }
+        else {
+          uint32_t w = (uint32_t)s[si] | ((uint32_t)s[si + 1] << 8) |
+                       ((uint32_t)s[si + 2] << 16) |
+                       ((uint32_t)s[si + 3] << 24);
+          si += 4;
+          if (w == 0) { bsize = -1; }  // end mark
+          else {
+            stored = (w >> 31) & 1;
+            bsize = (int64_t)(w & 0x7fffffff);
+          }
+        }
+      }
+      ok = bcast32(ok);
+      if (!ok) break;
+      bsize = bcast64(bsize);
+      stored = bcast32(stored);
+      si = bcast64(si);
+      if (bsize < 0) break;  // end mark
+      if (stored) {
+        if (lane == 0 && (si + bsize > frames[f].src_len ||
+                          di + bsize > dcap))
+          ok = 0;
+        ok = bcast32(ok);
+        if (!ok) break;
+        for (int64_t i = lane; i < bsize; i += BG_WAVE)
+          base_d[di + i] = s[si + i];
+        si += bsize;
+        di += bsize;
+        continue;
+      }
+      // LZ4 block: sequences until block exhausted
+      const int64_t bend = si + bsize;
+      while (ok && si < bend) {
+        // lane 0 parses one full sequence: token, literal length (+ext),
+        // literal start, optional match offset + length (+ext)
+        int64_t lit_len = 0, mlen = 0, moff = 0, lsrc = 0, nsi = si;
+        if (lane == 0) {
+          const uint8_t token = s[nsi++];
+          lit_len = token >> 4;
+          if (lit_len == 15) {
+            while (nsi < bend) {
+              const uint8_t b = s[nsi++];
+              lit_len += b;
+              if (b != 255) break;
+            }
+          }
+          lsrc = nsi;
+          nsi += lit_len;
+          if (nsi > bend || di + lit_len > dcap) {
+            ok = 0;
+          } else if (nsi < bend) {  // a match follows
+            if (nsi + 2 > bend) {
+              ok = 0;
+            } else {
+              moff = (int64_t)s[nsi] | ((int64_t)s[nsi + 1] << 8);
+              nsi += 2;
+              mlen = token & 0x0f;
+              if (mlen == 15) {
+                while (nsi < bend) {
+                  const uint8_t b = s[nsi++];
+                  mlen += b;
+                  if (b != 255) break;
+                }
+              }
+              mlen += 4;
+              if (moff == 0 || moff > di + lit_len ||
+                  di + lit_len + mlen > dcap)
+                ok = 0;
+            }
+          }
+        }
+        ok = bcast32(ok);
+        if (!ok) break;
+        lit_len = bcast64(lit_len);
+        mlen = bcast64(mlen);
+        moff = bcast64(moff);
+        lsrc = bcast64(lsrc);
+        nsi = bcast64(nsi);
+        for (int64_t i = lane; i < lit_len; i += BG_WAVE)
+          base_d[di + i] = s[lsrc + i];
+        di += lit_len;
+        if (mlen) {
+          const uint8_t* win = base_d + di - moff;
+          uint8_t* dst = base_d + di;
+          if (moff >= mlen) {
+            for (int64_t i = lane; i < mlen; i += BG_WAVE) dst[i] = win[i];
+          } else {
+            for (int64_t i = lane; i < mlen; i += BG_WAVE)
+              dst[i] = win[i % moff];
+          }
+          di += mlen;
+        }
+        si = nsi;
+      }
+    }
+    if (lane == 0) out_lens[f] = ok ? di : -1;
+  }
+}
+
+extern "C" int bg_lz4_decompress(const void* h_frames, int64_t nframes,
+                                 int64_t* h_out_lens) {
+  REQUIRE_INIT();
+  Lz4Frame* d_frames;
+  int64_t* d_lens;
+  HIP_TRY(pool_malloc((void**)&d_frames, sizeof(Lz4Frame) * (nframes ? nframes : 1)));
+  HIP_TRY(pool_malloc((void**)&d_lens, sizeof(int64_t) * (nframes ? nframes : 1)));
+  HIP_TRY(hipMemcpy(d_frames, h_frames, sizeof(Lz4Frame) * nframes,
+                    hipMemcpyHostToDevice));
+  const int waves_per_block = BG_BLOCK / BG_WAVE;
+  int blocks = (int)bg_imin64((nframes + waves_per_block - 1) / waves_per_block,
+                              BG_MAX_BLOCKS);
+  if (blocks == 0) blocks = 1;
+  hipLaunchKernelGGL(k_lz4_decompress, dim3(blocks), dim3(BG_BLOCK), 0, 0,
+                     d_frames, nframes, d_lens);
+  HIP_TRY(hipGetLastError());
+  HIP_TRY(hipMemcpy(h_out_lens, d_lens, sizeof(int64_t) * nframes,
+                    hipMemcpyDeviceToHost));
+  (void)pool_release(d_frames);
+  (void)pool_release(d_lens);
+  return BG_OK;
+}

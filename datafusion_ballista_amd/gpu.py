@@ -551,3 +551,28 @@ def _ctx_hash_repartition_fused(self, key_cols, payload_cols, n: int, k: int):
 
 
 GpuStageContext.hash_repartition_fused = _ctx_hash_repartition_fused
+
+
+class BgLz4Frame(ctypes.Structure):
+    _fields_ = [("d_src", ctypes.c_void_p), ("d_dst", ctypes.c_void_p),
+                ("src_len", ctypes.c_int64), ("dst_cap", ctypes.c_int64)]
+
+
+def _ctx_lz4_decompress(self, frames):
+    """frames: list of (src DeviceBuffer, src_len, dst DeviceBuffer,
+    dst_cap) pointing at LZ4 frame magic. -> decompressed lengths."""
+    n = len(frames)
+    arr = (BgLz4Frame * n)()
+    for i, (src, slen, dst, dcap) in enumerate(frames):
+        arr[i] = BgLz4Frame(src.ptr if hasattr(src, "ptr") else src,
+                            dst.ptr if hasattr(dst, "ptr") else dst,
+                            slen, dcap)
+    lens = np.zeros(n, dtype=np.int64)
+    _check(self.L.bg_lz4_decompress(
+        arr, ctypes.c_int64(n),
+        lens.ctypes.data_as(ctypes.POINTER(ctypes.c_int64))),
+        "bg_lz4_decompress")
+    return lens.tolist()
+
+
+GpuStageContext.lz4_decompress = _ctx_lz4_decompress

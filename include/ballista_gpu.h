@@ -323,6 +323,17 @@ typedef struct {
 } bg_lz4_block_job;
 int bg_lz4_compress_flat(const void* h_jobs, int64_t njobs,
                          int64_t* h_block_sizes);
+/* LZ4-frame DECODE on device (shuffle-read ingest; mirror of the
+ * compressor).  h_frames entries point at frame magic; h_out_lens[i] =
+ * decompressed bytes or -1 on malformed input. */
+typedef struct {
+  const void* d_src;
+  void* d_dst;
+  int64_t src_len;
+  int64_t dst_cap;
+} bg_lz4_frame;
+int bg_lz4_decompress(const void* h_frames, int64_t nframes,
+                      int64_t* h_out_lens);
 /* assemble [u32 size][block] sequences on device at precomputed offsets */
 typedef struct {
   const void* d_src;

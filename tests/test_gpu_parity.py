@@ -958,3 +958,39 @@ def test_hash_repartition_fused_parity(ctx, n, k):
     for a, b, esz in zip(b1, b2, esizes):
         assert np.array_equal(a.download(np.uint8, n * esz),
                               b.download(np.uint8, n * esz))
+
+
+@pytest.mark.parametrize("case", ["pyarrow", "own_codec", "stored", "mixed"])
+def test_lz4_decompress_roundtrip(ctx, case):
+    """Device LZ4-frame decode vs frames produced by (a) Arrow's reference
+    encoder, (b) our own device compressor, (c) stored-block frames —
+    the shuffle-read ingest path."""
+    rng = np.random.default_rng(len(case) * 3)
+    data = (b"shuffle bytes " * 20_000 +
+            bytes(rng.integers(0, 256, 50_000, dtype=np.uint8)))
+    if case == "pyarrow":
+        frame = pa.compress(data, codec="lz4", asbytes=True)
+    elif case == "own_codec":
+        src = ctx.upload(np.frombuffer(data, dtype=np.uint8))
+        sizes, slots = ctx.lz4_compress(src, len(data))
+        slot_bytes = slots.download(np.uint8,
+                                    max(len(sizes) * 65544, 1)).tobytes()
+        frame = gpu.lz4_frame_assemble(sizes, slot_bytes, len(data))
+    elif case == "stored":
+        import struct
+        frame = (gpu.LZ4_FRAME_HEADER +
+                 struct.pack("<I", len(data) | 0x80000000) + data +
+                 b"\x00\x00\x00\x00")
+    else:  # mixed: compressible + stored blocks from our codec
+        data = b"\x00" * 70_000 + bytes(rng.integers(0, 256, 70_000,
+                                                     dtype=np.uint8))
+        src = ctx.upload(np.frombuffer(data, dtype=np.uint8))
+        sizes, slots = ctx.lz4_compress(src, len(data))
+        slot_bytes = slots.download(np.uint8,
+                                    max(len(sizes) * 65544, 1)).tobytes()
+        frame = gpu.lz4_frame_assemble(sizes, slot_bytes, len(data))
+    fsrc = ctx.upload(np.frombuffer(frame, dtype=np.uint8))
+    fdst = ctx.alloc(len(data))
+    lens = ctx.lz4_decompress([(fsrc, len(frame), fdst, len(data))])
+    assert lens == [len(data)]
+    assert fdst.download(np.uint8, len(data)).tobytes() == data
