@@ -210,3 +210,71 @@ def stream_from_compressed_batches(schema: pa.Schema, batches) -> bytes:
         out.append(bytes(body))
     out.append(EOS)
     return b"".join(out)
+
+
+# ---------------------------------------------------------------------------
+# minimal reader side: walk an IPC stream's encapsulated messages and parse
+# RecordBatch metadata (the decode mirror of the writer above) — used by the
+# device shuffle-read path to locate compressed buffer extents
+# ---------------------------------------------------------------------------
+
+def _parse_table(meta, tbl):
+    soff = struct.unpack_from("<i", meta, tbl)[0]
+    vt = tbl - soff
+    vt_len, _ = struct.unpack_from("<HH", meta, vt)
+    n = (vt_len - 4) // 2
+    slots = struct.unpack_from(f"<{n}H", meta, vt + 4)
+    return slots
+
+
+def parse_record_batch_meta(meta: bytes):
+    """-> (n_rows, [(buf_off, buf_len)], compressed, body_len) or None if
+    the message is not a RecordBatch."""
+    root = struct.unpack_from("<I", meta, 0)[0]
+    slots = _parse_table(meta, root)
+
+    def fld(i):
+        return slots[i] if i < len(slots) and slots[i] else None
+
+    if not fld(1) or meta[root + slots[1]] != _HDR_RECORD_BATCH:
+        return None
+    body_len = struct.unpack_from("<q", meta, root + slots[3])[0] \
+        if fld(3) else 0
+    hpos = root + slots[2]
+    hoff = struct.unpack_from("<I", meta, hpos)[0]
+    htbl = hpos + hoff
+    hslots = _parse_table(meta, htbl)
+    n_rows = struct.unpack_from("<q", meta, htbl + hslots[0])[0] \
+        if hslots[0] else 0
+    bpos = htbl + hslots[2]
+    boff = struct.unpack_from("<I", meta, bpos)[0]
+    bvec = bpos + boff
+    cnt = struct.unpack_from("<I", meta, bvec)[0]
+    bufs = [struct.unpack_from("<qq", meta, bvec + 4 + 16 * i)
+            for i in range(cnt)]
+    compressed = len(hslots) > 3 and hslots[3] != 0
+    return n_rows, bufs, compressed, body_len
+
+
+def walk_stream(raw: bytes):
+    """Yield (meta_bytes, body_off, body_len) per message in an IPC stream
+    (concatenated streams supported by the caller re-invoking)."""
+    pos = 0
+    while pos + 8 <= len(raw):
+        if raw[pos:pos + 4] != b"\xff\xff\xff\xff":
+            # legacy (no continuation marker) not supported
+            return
+        mlen = struct.unpack_from("<i", raw, pos + 4)[0]
+        if mlen == 0:
+            pos += 8
+            return
+        meta = raw[pos + 8:pos + 8 + mlen]
+        parsed = parse_record_batch_meta(meta)
+        body_off = pos + 8 + mlen
+        if parsed is None:
+            yield None, body_off, 0
+            pos = body_off
+            continue
+        n_rows, bufs, compressed, body_len = parsed
+        yield (n_rows, bufs, compressed), body_off, body_len
+        pos = body_off + body_len

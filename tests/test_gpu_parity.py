@@ -994,3 +994,35 @@ def test_lz4_decompress_roundtrip(ctx, case):
     lens = ctx.lz4_decompress([(fsrc, len(frame), fdst, len(data))])
     assert lens == [len(data)]
     assert fdst.download(np.uint8, len(data)).tobytes() == data
+
+
+def test_device_shuffle_read_roundtrip(ctx, tmp_path):
+    """Stage N writes with the GPU codec; stage N+1 ingests the partition
+    ON DEVICE (read_partition_gpu) — columns must equal the host reader's
+    view (ShuffleReaderExec local-read parity, shuffle_reader.rs:1120-1168)."""
+    from datafusion_ballista_amd import engine, shuffle
+    n, k = 80_000, 8
+    rng = np.random.default_rng(311)
+    table = pa.table({
+        "k": pa.array(rng.integers(0, 7_000, size=n, dtype=np.int64)),
+        "d": pa.array(rng.integers(8000, 11000, size=n, dtype=np.int32),
+                      type=pa.int32()),
+    })
+    ex = engine.GpuQueryStageExecutor(ctx, "job-dr", 2, str(tmp_path),
+                                      key_columns=[0], num_partitions=k,
+                                      gpu_codec=True)
+    summaries = ex.execute_query_stage(0, table)
+    data_path = summaries[0].path
+    index_path = data_path + ".index"
+    for p in range(k):
+        want_batches = shuffle.read_partition(data_path, index_path, p)
+        want = pa.Table.from_batches(want_batches, schema=table.schema) \
+            if want_batches else table.schema.empty_table()
+        m, cols = shuffle.read_partition_gpu(ctx, data_path, index_path, p,
+                                             table.schema)
+        assert m == want.num_rows
+        if m:
+            got_k = cols[0].download(np.int64, m)
+            got_d = cols[1].download(np.int32, m)
+            assert np.array_equal(got_k, want.column("k").to_numpy())
+            assert np.array_equal(got_d, want.column("d").to_numpy())
