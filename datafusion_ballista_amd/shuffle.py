@@ -253,7 +253,7 @@ def read_partition_gpu(ctx, data_path, index_path, partition_id, schema):
             if msg is None:
                 consumed = pos + body_off
                 continue
-            n_rows, bufs, compressed, _bl = msg
+            n_rows, bufs, compressed = msg
             if not compressed:
                 raise RuntimeError("uncompressed batches: host path")
             cols = []
@@ -266,7 +266,7 @@ def read_partition_gpu(ctx, data_path, index_path, partition_id, schema):
                 usize = int.from_bytes(raw[abs_off:abs_off + 8], "little")
                 cols.append((abs_off + 8, blen - 8, usize))
             batches.append((n_rows, cols))
-            consumed = pos + body_off + _bl
+            consumed = pos + body_off + body_len
         if consumed == pos:
             break
         pos = consumed
