@@ -1026,3 +1026,123 @@ def test_device_shuffle_read_roundtrip(ctx, tmp_path):
             got_d = cols[1].download(np.int32, m)
             assert np.array_equal(got_k, want.column("k").to_numpy())
             assert np.array_equal(got_d, want.column("d").to_numpy())
+
+
+# ---------------------------------------------------------------------------
+# edge hardening
+# ---------------------------------------------------------------------------
+def test_empty_inputs(ctx):
+    """n=0 through filter/compact/hash/split/join/sort — no crashes, empty
+    results (the reference's empty-partition paths, writer.rs tests)."""
+    empty64 = np.zeros(0, dtype=np.int64)
+    col, _ = ctx.upload_column(empty64, gpu.BG_DT_INT64)
+    mask = ctx.eval_predicates([col], [(0, gpu.BG_PRED_GT, 0, 0)], 0)
+    _, cnt = ctx.mask_to_indices(mask, 0)
+    assert cnt == 0
+    h = ctx.hash_columns([col], 0)
+    p = ctx.partition_ids(h, 0, 4)
+    idx, offs = ctx.partition_indices(p, 0, 4)
+    assert np.array_equal(offs.download(np.int64, 5), np.zeros(5, np.int64))
+    join = gpu.GpuHashJoin(ctx, col, 0)
+    _, _, m = join.probe(col, 0)
+    assert m == 0
+    join.free()
+    perm = ctx.sort_rows([col], [False], 0)
+    assert perm.download(np.uint32, 0).size == 0
+
+
+def test_join_empty_build_side(ctx):
+    """Probing an empty build side yields zero matches (inner join)."""
+    build = np.zeros(0, dtype=np.int64)
+    probe = np.arange(1000, dtype=np.int64)
+    bcol, _ = ctx.upload_column(build, gpu.BG_DT_INT64)
+    pcol, _ = ctx.upload_column(probe, gpu.BG_DT_INT64)
+    join = gpu.GpuHashJoin(ctx, bcol, 0)
+    _, _, m = join.probe(pcol, 1000)
+    assert m == 0
+    join.free()
+
+
+def test_hashagg_table_full_fails_loudly(ctx):
+    """max_groups too small must error (BG_ERR_INVALID), not corrupt."""
+    n = 10_000
+    keys = np.arange(n, dtype=np.int64)  # n distinct groups
+    vals = np.ones(n, dtype=np.int64)
+    kc, _ = ctx.upload_column(keys, gpu.BG_DT_INT64)
+    vc, _ = ctx.upload_column(vals, gpu.BG_DT_INT64)
+    with pytest.raises(RuntimeError, match="table full"):
+        ctx.hashagg([kc], [vc], [gpu.BG_AGG_OP_SUM_I64], n, max_groups=16)
+
+
+def test_partition_split_k4096(ctx):
+    """Largest supported k (leader-loop path, 4096 partitions)."""
+    n, k = 300_000, 4096
+    rng = np.random.default_rng(4096)
+    keys = rng.integers(-2**62, 2**62, size=n, dtype=np.int64)
+    col, _ = ctx.upload_column(keys, gpu.BG_DT_INT64)
+    hbuf = ctx.hash_columns([col], n)
+    pbuf = ctx.partition_ids(hbuf, n, k)
+    ibuf, obuf = ctx.partition_indices(pbuf, n, k)
+    ctx.synchronize()
+    idx = ibuf.download(np.uint32, n)
+    offs = obuf.download(np.int64, k + 1)
+    want_h = oracle.hash_columns([("i64", keys)], n)
+    want_p = oracle.partition_ids(want_h, k)
+    want_idx, want_offs = oracle.partition_indices(want_p, k)
+    assert np.array_equal(offs, want_offs)
+    assert np.array_equal(idx, want_idx)
+
+
+def test_two_stage_pipeline_device_resident(ctx, tmp_path):
+    """Full two-stage flow, all compute on device: stage 1 hash-repartitions
+    orders by custkey and writes GPU-codec shuffle files; stage 2 device-
+    reads each partition, joins against a filtered customer build side and
+    group-by-sums — exact vs a python restatement (the q3-class two-stage
+    execution model end to end)."""
+    from datafusion_ballista_amd import engine, shuffle
+    rng = np.random.default_rng(555)
+    ncust, nord, k = 5_000, 120_000, 8
+    c_custkey = np.arange(1, ncust + 1, dtype=np.int64)
+    o_custkey = rng.integers(1, ncust + 1, size=nord, dtype=np.int64)
+    o_total = rng.integers(1, 10**6, size=nord, dtype=np.int64)
+    table = pa.table({"o_custkey": pa.array(o_custkey),
+                      "o_total": pa.array(o_total)})
+    ex = engine.GpuQueryStageExecutor(ctx, "job-2s", 1, str(tmp_path),
+                                      key_columns=[0], num_partitions=k,
+                                      gpu_codec=True)
+    summaries = ex.execute_query_stage(0, table)
+    data_path = summaries[0].path
+
+    # stage 2: per partition — device read, semi-join filter on custkey
+    # (custkey <= ncust//3 stands in for the dimension filter), group-by sum
+    keep_hi = ncust // 3
+    got = {}
+    for p in range(k):
+        m, cols = shuffle.read_partition_gpu(ctx, data_path,
+                                             data_path + ".index", p,
+                                             table.schema)
+        if m == 0:
+            continue
+        ck = ctx.column(gpu.BG_DT_INT64, cols[0], m)
+        mask = ctx.eval_predicates([ck], [(0, gpu.BG_PRED_LT, 0, keep_hi + 1)],
+                                   m)
+        tot = ctx.column(gpu.BG_DT_INT64, cols[1], m)
+        first, acc, counts = ctx.hashagg([ck], [tot],
+                                         [gpu.BG_AGG_OP_SUM_I64], m,
+                                         max_groups=max(2 * keep_hi, 64),
+                                         mask=mask)
+        keys_h = cols[0].download(np.int64, m)
+        for g in range(len(first)):
+            key = int(keys_h[first[g]])
+            cnt, s_ = got.get(key, (0, 0))
+            got[key] = (cnt + int(counts[g]),
+                        s_ + gpu.decode_agg_value(gpu.BG_AGG_OP_SUM_I64,
+                                                  bytes(acc[g, 0])))
+    want = {}
+    for i in range(nord):
+        key = int(o_custkey[i])
+        if key > keep_hi:
+            continue
+        cnt, s_ = want.get(key, (0, 0))
+        want[key] = (cnt + 1, s_ + int(o_total[i]))
+    assert got == want
