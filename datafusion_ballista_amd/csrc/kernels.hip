@@ -1160,20 +1160,44 @@ __global__ void k_join_build(const int64_t* keys, int64_t n, int* head,
   }
 }
 
+// 4-way ILP batching: each thread interleaves four independent chain
+// walks so the ~900-cycle random node loads overlap (memory-level
+// parallelism) instead of serialising per probe.
+#define JOIN_ILP 4
 __global__ void k_join_count(const int64_t* probe_keys, int64_t n_probe,
                              const int* head, const ulong2* nodes, u64 mask,
                              u64* counts) {
-  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n_probe;
-       i += (int64_t)gridDim.x * blockDim.x) {
-    const u64 k = (u64)probe_keys[i];
-    const u64 b = bg_hash_u64(k) & mask;
-    u64 c = 0;
-    for (int64_t j = head[b]; j >= 0;) {
-      const ulong2 node = nodes[j];
-      if (node.x == k) c++;
-      j = (int64_t)node.y;
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  const int64_t tid = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  for (int64_t base = tid; base < n_probe; base += stride * JOIN_ILP) {
+    int64_t idx[JOIN_ILP];
+    u64 key[JOIN_ILP];
+    int64_t cur[JOIN_ILP];
+    u64 cnt[JOIN_ILP];
+#pragma unroll
+    for (int j = 0; j < JOIN_ILP; ++j) {
+      idx[j] = base + (int64_t)j * stride;
+      const bool act = idx[j] < n_probe;
+      key[j] = act ? (u64)probe_keys[idx[j]] : 0;
+      cur[j] = act ? (int64_t)head[bg_hash_u64(key[j]) & mask] : -1;
+      cnt[j] = 0;
     }
-    counts[i] = c;
+    bool any = true;
+    while (any) {
+      any = false;
+#pragma unroll
+      for (int j = 0; j < JOIN_ILP; ++j) {
+        if (cur[j] >= 0) {
+          const ulong2 node = nodes[cur[j]];
+          if (node.x == key[j]) cnt[j]++;
+          cur[j] = (int64_t)node.y;
+          any = true;
+        }
+      }
+    }
+#pragma unroll
+    for (int j = 0; j < JOIN_ILP; ++j)
+      if (idx[j] < n_probe) counts[idx[j]] = cnt[j];
   }
 }
 
@@ -1181,19 +1205,37 @@ __global__ void k_join_fill(const int64_t* probe_keys, int64_t n_probe,
                             const int* head, const ulong2* nodes, u64 mask,
                             const i64* offsets, uint32_t* out_probe,
                             uint32_t* out_build) {
-  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n_probe;
-       i += (int64_t)gridDim.x * blockDim.x) {
-    const u64 k = (u64)probe_keys[i];
-    const u64 b = bg_hash_u64(k) & mask;
-    i64 w = offsets[i];
-    for (int64_t j = head[b]; j >= 0;) {
-      const ulong2 node = nodes[j];
-      if (node.x == k) {
-        out_probe[w] = (uint32_t)i;
-        out_build[w] = (uint32_t)j;
-        ++w;
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  const int64_t tid = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  for (int64_t base = tid; base < n_probe; base += stride * JOIN_ILP) {
+    int64_t idx[JOIN_ILP];
+    u64 key[JOIN_ILP];
+    int64_t cur[JOIN_ILP];
+    i64 w[JOIN_ILP];
+#pragma unroll
+    for (int j = 0; j < JOIN_ILP; ++j) {
+      idx[j] = base + (int64_t)j * stride;
+      const bool act = idx[j] < n_probe;
+      key[j] = act ? (u64)probe_keys[idx[j]] : 0;
+      cur[j] = act ? (int64_t)head[bg_hash_u64(key[j]) & mask] : -1;
+      w[j] = act ? offsets[idx[j]] : 0;
+    }
+    bool any = true;
+    while (any) {
+      any = false;
+#pragma unroll
+      for (int j = 0; j < JOIN_ILP; ++j) {
+        if (cur[j] >= 0) {
+          const ulong2 node = nodes[cur[j]];
+          if (node.x == key[j]) {
+            out_probe[w[j]] = (uint32_t)idx[j];
+            out_build[w[j]] = (uint32_t)cur[j];
+            ++w[j];
+          }
+          cur[j] = (int64_t)node.y;
+          any = true;
+        }
       }
-      j = (int64_t)node.y;
     }
   }
 }
