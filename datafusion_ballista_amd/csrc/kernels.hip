@@ -322,9 +322,31 @@ __global__ void k_eval_predicates(PredArgs args, int64_t n, u64* mask_words,
           case BG_DT_DICT8:
             x = (i128) reinterpret_cast<const uint8_t*>(pr.data)[row];
             break;
+          case BG_DT_FLOAT64: {
+            // float predicates compare in f64 (bounds arrive as f64 BITS in
+            // lo_lo/hi_lo); comparisons with NaN are false, matching SQL
+            const double xv = reinterpret_cast<const double*>(pr.data)[row];
+            double flo, fhi;
+            __builtin_memcpy(&flo, &pr.lo_lo, 8);
+            __builtin_memcpy(&fhi, &pr.hi_lo, 8);
+            bool kp;
+            switch (pr.op) {
+              case BG_PRED_GE_LT: kp = xv >= flo && xv < fhi; break;
+              case BG_PRED_BETWEEN: kp = xv >= flo && xv <= fhi; break;
+              case BG_PRED_LT: kp = xv < fhi; break;
+              case BG_PRED_EQ: kp = xv == flo; break;
+              case BG_PRED_GT: kp = xv > flo; break;
+              default: kp = false;
+            }
+            if (!kp) keep = false;
+            x = 0;
+            break;
+          }
           default:
             x = 0;
         }
+        if (!keep) break;      // a failed float predicate short-circuits
+        if (pr.dtype == BG_DT_FLOAT64) continue;  // handled above
         if (!pred_eval(x, pr.op, make_i128(pr.lo_lo, pr.lo_hi),
                        make_i128(pr.hi_lo, pr.hi_hi))) {
           keep = false;
@@ -1360,6 +1382,10 @@ extern "C" int bg_hashjoin_free(void* handle) {
 // (groups always hold >= 1 row, so enc 0 is never read back as a value)
 #define BG_AGG_MIN_I64 2
 #define BG_AGG_MAX_I64 3
+// SUM over Float64: accumulation order is unordered (atomics), so low bits
+// are nondeterministic — covered by the reference comparator's 1e-6
+// relative float tolerance (benchmarks/src/lib.rs:35).
+#define BG_AGG_SUM_F64 4
 
 struct AggArgs {
   int naggs;
@@ -1502,6 +1528,11 @@ __global__ void k_hashagg(KeyArgs keys, AggArgs aggs, const u64* mask_words,
           const u64 v = ~((u64) reinterpret_cast<const int64_t*>(
                               aggs.a[a].data)[i] ^ 0x8000000000000000ull);
           atomicMax(base, v);
+          break;
+        }
+        case BG_AGG_SUM_F64: {
+          const double v = reinterpret_cast<const double*>(aggs.a[a].data)[i];
+          atomicAdd(reinterpret_cast<double*>(base), v);
           break;
         }
         default:

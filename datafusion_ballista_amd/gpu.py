@@ -7,8 +7,14 @@ the GPU is missing — no CPU fallback exists on the product path.
 
 import ctypes
 import os
+import struct as _struct
 
 import numpy as np
+
+
+def struct_unpack_bits(v: float) -> int:
+    """f64 -> its bit pattern as a SIGNED i64 (BgPred bound fields)."""
+    return _struct.unpack("<q", _struct.pack("<d", v))[0]
 
 _DIR = os.path.dirname(os.path.abspath(__file__))
 
@@ -18,6 +24,7 @@ BG_DT_DATE32 = 3
 BG_DT_DECIMAL128 = 4
 BG_DT_DICT8 = 5
 BG_DT_UTF8 = 6
+BG_DT_FLOAT64 = 7
 
 BG_PRED_GE_LT = 0
 BG_PRED_BETWEEN = 1
@@ -26,12 +33,13 @@ BG_PRED_EQ = 3
 BG_PRED_GT = 4
 
 _DT_SIZE = {BG_DT_INT32: 4, BG_DT_DATE32: 4, BG_DT_INT64: 8,
-            BG_DT_DECIMAL128: 16, BG_DT_DICT8: 1}
+            BG_DT_DECIMAL128: 16, BG_DT_DICT8: 1, BG_DT_FLOAT64: 8}
 
 _NP_TO_DT = {
     np.dtype(np.int32): BG_DT_INT32,
     np.dtype(np.int64): BG_DT_INT64,
     np.dtype(np.uint8): BG_DT_DICT8,
+    np.dtype(np.float64): BG_DT_FLOAT64,
 }
 
 
@@ -189,6 +197,11 @@ class GpuStageContext:
         carr = (BgColumn * len(cols))(*cols)
         parr = (BgPred * len(preds))()
         for i, (c, op, lo, hi) in enumerate(preds):
+            if isinstance(lo, float) or isinstance(hi, float):
+                lo_lo = struct_unpack_bits(float(lo))
+                hi_lo = struct_unpack_bits(float(hi))
+                parr[i] = BgPred(c, op, lo_lo, 0, hi_lo, 0)
+                continue
             lo_lo, lo_hi = _split_i128(lo)
             hi_lo, hi_hi = _split_i128(hi)
             parr[i] = BgPred(c, op, lo_lo, lo_hi, hi_lo, hi_hi)
@@ -331,6 +344,7 @@ BG_AGG_OP_SUM_DEC128 = 0
 BG_AGG_OP_SUM_I64 = 1
 BG_AGG_OP_MIN_I64 = 2
 BG_AGG_OP_MAX_I64 = 3
+BG_AGG_OP_SUM_F64 = 4
 
 
 def decode_agg_value(op, raw16: bytes) -> int:

@@ -89,6 +89,7 @@ typedef enum {
   BG_DT_DECIMAL128 = 4,  /* Arrow Decimal128: 16-byte LE two's complement */
   BG_DT_DICT8 = 5,       /* dictionary codes as u8 (small Utf8 dictionaries) */
   BG_DT_UTF8 = 6,        /* Arrow Utf8: i32 offsets (d_offsets) + byte data */
+  BG_DT_FLOAT64 = 7,     /* f64; predicates/aggregates at 1e-6 rel tolerance */
 } bg_dtype;
 
 typedef struct {

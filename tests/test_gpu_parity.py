@@ -1146,3 +1146,36 @@ def test_two_stage_pipeline_device_resident(ctx, tmp_path):
         cnt, s_ = want.get(key, (0, 0))
         want[key] = (cnt + 1, s_ + int(o_total[i]))
     assert got == want
+
+
+def test_f64_filter_and_sum(ctx):
+    """Float64 predicates (exact compares) + grouped SUM(f64): sums within
+    1e-6 relative of numpy (unordered atomic accumulation; the reference
+    comparator's float tolerance, benchmarks/src/lib.rs:35)."""
+    n = 500_000
+    rng = np.random.default_rng(77)
+    vals = rng.standard_normal(n) * 1000.0
+    keys = rng.integers(0, 100, size=n, dtype=np.int64)
+    vcol, _ = ctx.upload_column(vals, gpu.BG_DT_FLOAT64)
+    kcol, _ = ctx.upload_column(keys, gpu.BG_DT_INT64)
+
+    mask = ctx.eval_predicates([vcol], [(0, gpu.BG_PRED_GT, 12.5, 0.0)], n)
+    _, m = ctx.mask_to_indices(mask, n)
+    assert m == int((vals > 12.5).sum())
+
+    first, acc, counts = ctx.hashagg([kcol], [vcol],
+                                     [gpu.BG_AGG_OP_SUM_F64], n,
+                                     max_groups=512, mask=mask)
+    got = {}
+    import struct as st
+    for g in range(len(first)):
+        key = int(keys[first[g]])
+        got[key] = (int(counts[g]),
+                    st.unpack("<d", bytes(acc[g, 0][:8]))[0])
+    sel = vals > 12.5
+    for key in np.unique(keys[sel]):
+        cnt = int((sel & (keys == key)).sum())
+        s_ = float(vals[sel & (keys == key)].sum())
+        gc, gs = got[int(key)]
+        assert gc == cnt
+        assert abs(gs - s_) <= 1e-6 * max(abs(s_), 1e-30)
