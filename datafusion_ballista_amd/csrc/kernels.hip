@@ -242,14 +242,6 @@ __device__ __forceinline__ void atomic_add_i128(u64* lo, u64* hi, i128 v) {
   if (vhi + carry) atomicAdd(hi, vhi + carry);
 }
 
-__device__ __forceinline__ void atomic_add_i128_lds(u64* lo, u64* hi, i128 v) {
-  u64 vlo = (u64)(u128)v;
-  u64 vhi = (u64)((u128)v >> 64);
-  u64 old = atomicAdd(lo, vlo);
-  u64 carry = (old + vlo) < vlo ? 1ull : 0ull;
-  if (vhi + carry) atomicAdd(hi, vhi + carry);
-}
-
 // full-wave i128 sum (all 64 lanes get the total)
 __device__ __forceinline__ i128 wave_reduce_i128(i128 v) {
   u64 lo = (u64)(u128)v;

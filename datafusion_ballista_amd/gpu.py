@@ -181,11 +181,12 @@ class GpuStageContext:
 
     def upload_column(self, arr: np.ndarray, dtype: int = None,
                       validity: np.ndarray = None):
+        """Fixed-width column upload; Decimal128 arrives as raw 16-B rows."""
         if dtype is None:
             dtype = _NP_TO_DT[arr.dtype]
         buf = self.upload(arr)
         vbuf = self.upload(validity) if validity is not None else None
-        n = len(arr) if dtype != BG_DT_DECIMAL128 else arr.nbytes // 16
+        n = arr.nbytes // _DT_SIZE[dtype]
         return self.column(dtype, buf, n, vbuf), buf
 
     # ---- ops ----

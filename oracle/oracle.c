@@ -127,24 +127,15 @@ EXPORT void oracle_partition_indices(const uint32_t* pids, int64_t n,
   for (uint32_t p = 0; p <= k; ++p) out_offsets[p] = 0;
   for (int64_t i = 0; i < n; ++i) out_offsets[pids[i] + 1]++;
   for (uint32_t p = 0; p < k; ++p) out_offsets[p + 1] += out_offsets[p];
-  int64_t* cursor = out_offsets; /* reuse: cursor[p] = next slot for p */
-  /* copy offsets into a scratch cursor on the stack-free path: use a simple
-   * second pass with a local array allocated by caller?  Keep it simple:
-   * recompute cursors in a local VLA-free loop. */
-  /* k is small (<= 4096 in practice); use a fixed heap-less scheme: */
-  {
-    /* temporary cursor array stored at the tail of out_indices is unsafe;
-     * instead do per-partition passes only if k tiny, else a cursor copy. */
+  if (k <= 4096) { /* the device path's k bound; cursors on the stack */
     int64_t cur[4096];
-    if (k <= 4096) {
-      for (uint32_t p = 0; p < k; ++p) cur[p] = out_offsets[p];
-      for (int64_t i = 0; i < n; ++i) out_indices[cur[pids[i]]++] = (uint32_t)i;
-      return;
-    }
+    for (uint32_t p = 0; p < k; ++p) cur[p] = out_offsets[p];
+    for (int64_t i = 0; i < n; ++i) out_indices[cur[pids[i]]++] = (uint32_t)i;
+    return;
   }
-  /* fallback for k > 4096: per-partition scan (O(n*k), oracle sizes only) */
+  /* k > 4096: per-partition scan (O(n*k); oracle sizes are small) */
   for (uint32_t p = 0; p < k; ++p) {
-    int64_t w = cursor[p];
+    int64_t w = out_offsets[p];
     for (int64_t i = 0; i < n; ++i)
       if (pids[i] == p) out_indices[w++] = (uint32_t)i;
   }
