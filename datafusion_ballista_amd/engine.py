@@ -43,6 +43,7 @@ _PA_TO_BG = {
     pa.int64(): gpu.BG_DT_INT64,
     pa.date32(): gpu.BG_DT_DATE32,
     pa.uint8(): gpu.BG_DT_DICT8,
+    pa.float64(): gpu.BG_DT_FLOAT64,
 }
 
 
@@ -61,7 +62,8 @@ def _np_for(t: pa.DataType):
     if pa.types.is_decimal128(t):
         return np.uint8  # 16 B/elem raw
     return {pa.int32(): np.int32, pa.int64(): np.int64,
-            pa.date32(): np.int32, pa.uint8(): np.uint8}[t]
+            pa.date32(): np.int32, pa.uint8(): np.uint8,
+            pa.float64(): np.float64}[t]
 
 
 def _col_raw(arr: pa.Array) -> np.ndarray:

@@ -372,10 +372,26 @@ def _ctx_hashagg(self, key_cols, agg_cols, agg_ops, n, max_groups,
     aarr = (BgColumn * max(naggs, 1))(*(agg_cols or [BgColumn()]))
     oarr = (ctypes.c_int32 * max(naggs, 1))(*(agg_ops or [0]))
     ng = ctypes.c_int64()
-    _check(self.L.bg_hashagg(karr, len(key_cols), aarr, oarr, naggs,
-                             mask.ptr if mask else None, ctypes.c_int64(n),
-                             ctypes.c_int64(max_groups), first.ptr, acc.ptr,
-                             counts.ptr, ctypes.byref(ng)), "bg_hashagg")
+    # auto-grow on table-full (the scheduler's row estimates can be low,
+    # like the reference's AQE-fed group estimates)
+    attempt_groups = max_groups
+    for _ in range(4):
+        rc = self.L.bg_hashagg(karr, len(key_cols), aarr, oarr, naggs,
+                               mask.ptr if mask else None, ctypes.c_int64(n),
+                               ctypes.c_int64(attempt_groups), first.ptr,
+                               acc.ptr, counts.ptr, ctypes.byref(ng))
+        if rc == 0:
+            break
+        err = load_library().bg_last_error().decode()
+        if "table full" not in err or attempt_groups >= n:
+            _check(rc, "bg_hashagg")
+        attempt_groups = min(max(attempt_groups * 4, 64), max(n, 64))
+        # output buffers must grow with the group capacity
+        first = self.alloc(max(4 * attempt_groups, 4))
+        acc = self.alloc(max(16 * attempt_groups * max(naggs, 1), 16))
+        counts = self.alloc(max(8 * attempt_groups, 8))
+    else:
+        _check(rc, "bg_hashagg")
     g = ng.value
     return (first.download(np.uint32, g),
             acc.download(np.uint8, 16 * g * naggs).reshape(g, naggs, 16)

@@ -1179,3 +1179,41 @@ def test_f64_filter_and_sum(ctx):
         gc, gs = got[int(key)]
         assert gc == cnt
         assert abs(gs - s_) <= 1e-6 * max(abs(s_), 1e-30)
+
+
+def test_hashagg_auto_grow_and_dec128_keys(ctx):
+    """Table-full auto-retry (scheduler underestimates) + Decimal128 group
+    keys."""
+    n = 50_000
+    rng = np.random.default_rng(19)
+    kdec = rng.integers(0, 20_000, size=n, dtype=np.int64)  # ~18k groups
+    dec16 = dec_bytes(kdec)
+    vals = rng.integers(0, 1000, size=n, dtype=np.int64)
+    kc = ctx.column(gpu.BG_DT_DECIMAL128, ctx.upload(dec16), n)
+    vc, _ = ctx.upload_column(vals, gpu.BG_DT_INT64)
+    # deliberately low estimate -> auto-grow path
+    first, acc, counts = ctx.hashagg([kc], [vc], [gpu.BG_AGG_OP_SUM_I64], n,
+                                     max_groups=32)
+    got = {}
+    for g in range(len(first)):
+        key = int(kdec[first[g]])
+        got[key] = (int(counts[g]),
+                    [gpu.decode_agg_value(gpu.BG_AGG_OP_SUM_I64,
+                                          bytes(acc[g, 0]))])
+    want = oracle.hashagg([kdec], [("sum", vals)], n)
+    want = {k_[0]: v for k_, v in want.items()}
+    assert got == want
+
+
+def test_filter_with_validity(ctx):
+    """Predicate over a null-carrying column: null rows are excluded
+    (SQL WHERE three-valued logic -> false; FilterExec semantics)."""
+    n = 10_000
+    rng = np.random.default_rng(8)
+    vals = rng.integers(0, 100, size=n, dtype=np.int64)
+    valid = rng.integers(0, 256, size=(n + 7) // 8, dtype=np.uint8)
+    col, _ = ctx.upload_column(vals, gpu.BG_DT_INT64, validity=valid)
+    mask = ctx.eval_predicates([col], [(0, gpu.BG_PRED_GT, 10, 0)], n)
+    _, m = ctx.mask_to_indices(mask, n)
+    want = oracle.filter_mask([("i64", vals, valid, 4, 10, 0)], n)
+    assert m == len(oracle.mask_to_indices(want, n))
