@@ -25,6 +25,7 @@
 #include <hip/hip_runtime.h>
 
 #include <cstdio>
+#include <cstdlib>
 #include <cstring>
 
 #include "../../include/ballista_gpu.h"
@@ -941,6 +942,7 @@ extern "C" int bg_hash_repartition(const bg_column* key_cols, int32_t nkeys,
 // q6: fused scan + filter + SUM(l_extendedprice * l_discount)
 // 52 B/row algorithmic (Date32 4 + 3 x Decimal128 16) — HBM-bound.
 // ---------------------------------------------------------------------------
+template <bool NT>
 __global__ void k_q6_agg(const int32_t* shipdate, const ulong2* discount,
                          const ulong2* quantity, const ulong2* extendedprice,
                          int64_t n, int32_t date_lo, int32_t date_hi,
@@ -950,10 +952,16 @@ __global__ void k_q6_agg(const int32_t* shipdate, const ulong2* discount,
   u64 cnt = 0;
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
        i += (int64_t)gridDim.x * blockDim.x) {
-    const int32_t d = shipdate[i];
-    const ulong2 dv = discount[i];
-    const ulong2 qv = quantity[i];
-    const ulong2 pv = extendedprice[i];
+    // NT: every byte is read exactly once per pass — non-temporal loads
+    // skip L2 retention (guide §nt-weights: streams one CU reads once)
+    const int32_t d = NT ? __builtin_nontemporal_load(&shipdate[i])
+                         : shipdate[i];
+    const ulong2 dv = NT ? __builtin_nontemporal_load(&discount[i])
+                         : discount[i];
+    const ulong2 qv = NT ? __builtin_nontemporal_load(&quantity[i])
+                         : quantity[i];
+    const ulong2 pv = NT ? __builtin_nontemporal_load(&extendedprice[i])
+                         : extendedprice[i];
     const i128 disc = make_i128(dv.x, (i64)dv.y);
     const i128 qty = make_i128(qv.x, (i64)qv.y);
     const bool keep = (d >= date_lo) & (d < date_hi) & (disc >= disc_lo) &
@@ -1008,12 +1016,26 @@ extern "C" int bg_q6_agg(const bg_column* shipdate, const bg_column* discount,
   HIP_TRY(hipEventCreate(&ev0));
   HIP_TRY(hipEventCreate(&ev1));
   HIP_TRY(hipEventRecord(ev0, 0));
-  hipLaunchKernelGGL(k_q6_agg, dim3(blocks), dim3(BG_BLOCK), 0, 0,
-                     (const int32_t*)shipdate->d_data,
-                     (const ulong2*)discount->d_data,
-                     (const ulong2*)quantity->d_data,
-                     (const ulong2*)extendedprice->d_data, n, date_lo, date_hi,
-                     disc_lo, disc_hi, qty_lt, d_acc, d_acc + 1, d_acc + 2);
+  static const bool use_nt = [] {
+    const char* e = getenv("BG_Q6_NT");
+    return e && e[0] == '1';
+  }();
+  if (use_nt)
+    hipLaunchKernelGGL(k_q6_agg<true>, dim3(blocks), dim3(BG_BLOCK), 0, 0,
+                       (const int32_t*)shipdate->d_data,
+                       (const ulong2*)discount->d_data,
+                       (const ulong2*)quantity->d_data,
+                       (const ulong2*)extendedprice->d_data, n, date_lo,
+                       date_hi, disc_lo, disc_hi, qty_lt, d_acc, d_acc + 1,
+                       d_acc + 2);
+  else
+    hipLaunchKernelGGL(k_q6_agg<false>, dim3(blocks), dim3(BG_BLOCK), 0, 0,
+                       (const int32_t*)shipdate->d_data,
+                       (const ulong2*)discount->d_data,
+                       (const ulong2*)quantity->d_data,
+                       (const ulong2*)extendedprice->d_data, n, date_lo,
+                       date_hi, disc_lo, disc_hi, qty_lt, d_acc, d_acc + 1,
+                       d_acc + 2);
   HIP_TRY(hipGetLastError());
   HIP_TRY(hipEventRecord(ev1, 0));
   HIP_TRY(hipEventSynchronize(ev1));
