@@ -953,15 +953,28 @@ __global__ void k_q6_agg(const int32_t* shipdate, const ulong2* discount,
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
        i += (int64_t)gridDim.x * blockDim.x) {
     // NT: every byte is read exactly once per pass — non-temporal loads
-    // skip L2 retention (guide §nt-weights: streams one CU reads once)
-    const int32_t d = NT ? __builtin_nontemporal_load(&shipdate[i])
-                         : shipdate[i];
-    const ulong2 dv = NT ? __builtin_nontemporal_load(&discount[i])
-                         : discount[i];
-    const ulong2 qv = NT ? __builtin_nontemporal_load(&quantity[i])
-                         : quantity[i];
-    const ulong2 pv = NT ? __builtin_nontemporal_load(&extendedprice[i])
-                         : extendedprice[i];
+    // skip L2 retention (guide §nt-weights: streams one CU reads once).
+    // HIP_vector_type isn't a clang vector: load via ext_vector alias.
+    typedef unsigned long long ull2_ev __attribute__((ext_vector_type(2)));
+    int32_t d;
+    ulong2 dv, qv, pv;
+    if (NT) {
+      d = __builtin_nontemporal_load(&shipdate[i]);
+      const ull2_ev dve = __builtin_nontemporal_load(
+          reinterpret_cast<const ull2_ev*>(discount) + i);
+      const ull2_ev qve = __builtin_nontemporal_load(
+          reinterpret_cast<const ull2_ev*>(quantity) + i);
+      const ull2_ev pve = __builtin_nontemporal_load(
+          reinterpret_cast<const ull2_ev*>(extendedprice) + i);
+      dv.x = dve[0]; dv.y = dve[1];
+      qv.x = qve[0]; qv.y = qve[1];
+      pv.x = pve[0]; pv.y = pve[1];
+    } else {
+      d = shipdate[i];
+      dv = discount[i];
+      qv = quantity[i];
+      pv = extendedprice[i];
+    }
     const i128 disc = make_i128(dv.x, (i64)dv.y);
     const i128 qty = make_i128(qv.x, (i64)qv.y);
     const bool keep = (d >= date_lo) & (d < date_hi) & (disc >= disc_lo) &

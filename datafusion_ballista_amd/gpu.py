@@ -375,7 +375,7 @@ def _ctx_hashagg(self, key_cols, agg_cols, agg_ops, n, max_groups,
     # auto-grow on table-full (the scheduler's row estimates can be low,
     # like the reference's AQE-fed group estimates)
     attempt_groups = max_groups
-    for _ in range(4):
+    for _ in range(8):
         rc = self.L.bg_hashagg(karr, len(key_cols), aarr, oarr, naggs,
                                mask.ptr if mask else None, ctypes.c_int64(n),
                                ctypes.c_int64(attempt_groups), first.ptr,
@@ -385,7 +385,7 @@ def _ctx_hashagg(self, key_cols, agg_cols, agg_ops, n, max_groups,
         err = load_library().bg_last_error().decode()
         if "table full" not in err or attempt_groups >= n:
             _check(rc, "bg_hashagg")
-        attempt_groups = min(max(attempt_groups * 4, 64), max(n, 64))
+        attempt_groups = min(max(attempt_groups * 8, 64), max(n, 64))
         # output buffers must grow with the group capacity
         first = self.alloc(max(4 * attempt_groups, 4))
         acc = self.alloc(max(16 * attempt_groups * max(naggs, 1), 16))
