@@ -1681,6 +1681,17 @@ extern "C" int bg_hashagg(const bg_column* key_cols, int32_t nkeys,
   int rc = bg_mask_to_indices(reinterpret_cast<const uint8_t*>(occ_mask),
                               (int64_t)cap, sidx, &ngroups);
   if (rc != BG_OK) return rc;
+  if (ngroups > max_groups) {
+    // the open table (capacity 2*max_groups rounded up) absorbed more
+    // distinct groups than the caller's OUTPUT buffers hold
+    (void)pool_release(slot_data);
+    (void)pool_release(err_flag);
+    (void)pool_release(occ_mask);
+    (void)pool_release(sidx);
+    return set_err(BG_ERR_INVALID,
+                   "bg_hashagg: table full (raise max_groups — more groups "
+                   "than output capacity)");
+  }
 
   int gblocks = (int)bg_imin64((ngroups + BG_BLOCK - 1) / BG_BLOCK,
                                BG_MAX_BLOCKS);

@@ -1064,14 +1064,24 @@ def test_join_empty_build_side(ctx):
 
 
 def test_hashagg_table_full_fails_loudly(ctx):
-    """max_groups too small must error (BG_ERR_INVALID), not corrupt."""
+    """max_groups too small must error (BG_ERR_INVALID) at the RAW ABI, not
+    corrupt (the python wrapper adds auto-grow on top, tested separately)."""
+    import ctypes
     n = 10_000
     keys = np.arange(n, dtype=np.int64)  # n distinct groups
     vals = np.ones(n, dtype=np.int64)
     kc, _ = ctx.upload_column(keys, gpu.BG_DT_INT64)
     vc, _ = ctx.upload_column(vals, gpu.BG_DT_INT64)
-    with pytest.raises(RuntimeError, match="table full"):
-        ctx.hashagg([kc], [vc], [gpu.BG_AGG_OP_SUM_I64], n, max_groups=16)
+    first = ctx.alloc(4 * 16)
+    acc = ctx.alloc(16 * 16)
+    counts = ctx.alloc(8 * 16)
+    ng = ctypes.c_int64()
+    rc = ctx.L.bg_hashagg((gpu.BgColumn * 1)(kc), 1, (gpu.BgColumn * 1)(vc),
+                          (ctypes.c_int32 * 1)(gpu.BG_AGG_OP_SUM_I64), 1,
+                          None, ctypes.c_int64(n), ctypes.c_int64(16),
+                          first.ptr, acc.ptr, counts.ptr, ctypes.byref(ng))
+    assert rc == -3  # BG_ERR_INVALID
+    assert b"table full" in ctx.L.bg_last_error()
 
 
 def test_partition_split_k4096(ctx):
