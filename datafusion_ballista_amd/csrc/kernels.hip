@@ -1029,9 +1029,11 @@ extern "C" int bg_q6_agg(const bg_column* shipdate, const bg_column* discount,
   HIP_TRY(hipEventCreate(&ev0));
   HIP_TRY(hipEventCreate(&ev1));
   HIP_TRY(hipEventRecord(ev0, 0));
+  // non-temporal loads by default: the fused scan reads every byte exactly
+  // once, and NT streams measured +11% over cached loads (6.7 vs 6.05 TB/s)
   static const bool use_nt = [] {
     const char* e = getenv("BG_Q6_NT");
-    return e && e[0] == '1';
+    return !(e && e[0] == '0');
   }();
   if (use_nt)
     hipLaunchKernelGGL(k_q6_agg<true>, dim3(blocks), dim3(BG_BLOCK), 0, 0,
@@ -1095,14 +1097,22 @@ __global__ void k_q1_agg(const uint8_t* rf, const uint8_t* ls,
   for (int i = threadIdx.x; i < Q1_GROUPS; i += blockDim.x) s_counts[i] = 0;
   __syncthreads();
 
+  typedef unsigned long long ull2_ev __attribute__((ext_vector_type(2)));
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
        i += (int64_t)gridDim.x * blockDim.x) {
-    if (shipdate[i] > date_le) continue;
-    const uint32_t g = ((uint32_t)rf[i] << 4) | (uint32_t)ls[i];
-    const i64 qty = (i64)quantity[i].x;       // Decimal(15,2) fits i64
-    const i64 price = (i64)extendedprice[i].x;
-    const i64 disc = (i64)discount[i].x;
-    const i64 tx = (i64)tax[i].x;
+    // non-temporal: every stream byte is read exactly once per pass
+    if (__builtin_nontemporal_load(&shipdate[i]) > date_le) continue;
+    const uint32_t g =
+        ((uint32_t)__builtin_nontemporal_load(&rf[i]) << 4) |
+        (uint32_t)__builtin_nontemporal_load(&ls[i]);
+    const i64 qty = (i64)__builtin_nontemporal_load(
+        reinterpret_cast<const ull2_ev*>(quantity) + i)[0];
+    const i64 price = (i64)__builtin_nontemporal_load(
+        reinterpret_cast<const ull2_ev*>(extendedprice) + i)[0];
+    const i64 disc = (i64)__builtin_nontemporal_load(
+        reinterpret_cast<const ull2_ev*>(discount) + i)[0];
+    const i64 tx = (i64)__builtin_nontemporal_load(
+        reinterpret_cast<const ull2_ev*>(tax) + i)[0];
     const i64 disc_price = price * (100 - disc);
     u64* base = &s_sums[g * Q1_ACCS];
     atomicAdd(&base[0], (u64)qty);
