@@ -304,14 +304,21 @@ __global__ void k_eval_predicates(PredArgs args, int64_t n, u64* mask_words,
         switch (pr.dtype) {
           case BG_DT_INT32:
           case BG_DT_DATE32:
-            x = (i128) reinterpret_cast<const int32_t*>(pr.data)[row];
+            x = (i128)__builtin_nontemporal_load(
+                reinterpret_cast<const int32_t*>(pr.data) + row);
             break;
           case BG_DT_INT64:
-            x = (i128) reinterpret_cast<const int64_t*>(pr.data)[row];
+            x = (i128)__builtin_nontemporal_load(
+                reinterpret_cast<const int64_t*>(pr.data) + row);
             break;
-          case BG_DT_DECIMAL128:
-            x = load_dec128(pr.data, row);
+          case BG_DT_DECIMAL128: {
+            typedef unsigned long long ull2_ev
+                __attribute__((ext_vector_type(2)));
+            const ull2_ev v = __builtin_nontemporal_load(
+                reinterpret_cast<const ull2_ev*>(pr.data) + row);
+            x = make_i128(v[0], (i64)v[1]);
             break;
+          }
           case BG_DT_DICT8:
             x = (i128) reinterpret_cast<const uint8_t*>(pr.data)[row];
             break;
@@ -627,15 +634,20 @@ __global__ void k_hash_columns(KeyArgs keys, int64_t n, u64* hashes) {
       u64 hc;
       switch (keys.k[c].dtype) {
         case BG_DT_INT64:
-          hc = bg_hash_u64((u64) reinterpret_cast<const int64_t*>(keys.k[c].data)[i]);
+          hc = bg_hash_u64((u64)__builtin_nontemporal_load(
+              reinterpret_cast<const int64_t*>(keys.k[c].data) + i));
           break;
         case BG_DT_INT32:
         case BG_DT_DATE32:
-          hc = bg_hash_u32((uint32_t) reinterpret_cast<const int32_t*>(keys.k[c].data)[i]);
+          hc = bg_hash_u32((uint32_t)__builtin_nontemporal_load(
+              reinterpret_cast<const int32_t*>(keys.k[c].data) + i));
           break;
         case BG_DT_DECIMAL128: {
-          const ulong2 v = reinterpret_cast<const ulong2*>(keys.k[c].data)[i];
-          hc = bg_hash_u128(v.x, v.y);
+          typedef unsigned long long ull2_ev
+              __attribute__((ext_vector_type(2)));
+          const ull2_ev v = __builtin_nontemporal_load(
+              reinterpret_cast<const ull2_ev*>(keys.k[c].data) + i);
+          hc = bg_hash_u128(v[0], v[1]);
           break;
         }
         case BG_DT_DICT8:
@@ -859,8 +871,12 @@ template <typename T>
 __global__ void k_scatter_rows(const T* src, const uint32_t* rank, int64_t n,
                                T* dst) {
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
-       i += (int64_t)gridDim.x * blockDim.x)
-    dst[rank[i]] = src[i];
+       i += (int64_t)gridDim.x * blockDim.x) {
+    // sequential inputs are read once: non-temporal; the scattered writes
+    // stay cached so partial lines combine in L2
+    const uint32_t r = __builtin_nontemporal_load(&rank[i]);
+    dst[r] = src[i];
+  }
 }
 
 extern "C" int bg_scatter_rows(const void* d_src, int64_t elem_size,
