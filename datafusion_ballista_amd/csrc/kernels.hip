@@ -795,20 +795,24 @@ __global__ void k_part_scatter(const uint32_t* pids, int64_t n, uint32_t k,
           cur_hi = mine ? (int)(uint32_t)((u64)nxt >> 32) : cur_hi;
         }
       } else {
-        // leader loop: iterations = distinct partitions present (<= 64)
-        u64 remaining = __ballot(active);
-        while (remaining) {
-          const int leader = __ffsll((i64)remaining) - 1;
-          const uint32_t pg = (uint32_t)__shfl((int)pid, leader, BG_WAVE);
-          const u64 m = __ballot(active && pid == pg);
-          const i64 base = my[pg];
-          if (active && pid == pg)
-            my_pos = base + __popcll(m & ((1ull << lane_id()) - 1));
-          __builtin_amdgcn_wave_barrier();
-          if (lane_id() == leader) my[pg] = base + __popcll(m);
-          __builtin_amdgcn_wave_barrier();
-          remaining &= ~m;
+        // ballot-bit multi-split (ceil(log2 k) ballots give each lane the
+        // mask of lanes sharing its partition) — replaces the serialized
+        // per-distinct-partition leader loop
+        const u64 act = __ballot(active);
+        u64 same = act;
+        const int nbits = 32 - __builtin_clz(k - 1);
+        for (int b = 0; b < nbits; ++b) {
+          const u64 bb = __ballot(((pid >> b) & 1u) != 0u);
+          same &= ((pid >> b) & 1u) ? bb : ~bb;
         }
+        if (active) {
+          const u64 lower = same & ((1ull << lane_id()) - 1);
+          const i64 base = my[pid];
+          my_pos = base + __popcll(lower);
+          __builtin_amdgcn_wave_barrier();
+          if (lower == 0) my[pid] = base + __popcll(same);
+        }
+        __builtin_amdgcn_wave_barrier();
       }
       if (active) {
         out[my_pos] = (uint32_t)r;

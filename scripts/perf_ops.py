@@ -196,6 +196,18 @@ def main():
            {"ngroups": ng.value, "rows_per_s": n / sec,
             "kernel_ms": L.bg_last_kernel_ms()})
 
+    # ---- stable split at k=512 (the >64-partition multi-split path) ----
+    ob512 = ctx.alloc(8 * 513)
+    gpu._check(L.bg_partition_ids(hbuf.ptr, ctypes.c_int64(n), 512,
+                                  pbuf.ptr), "pids512")
+    sec = timeit(ctx, lambda: gpu._check(
+        L.bg_partition_indices(pbuf.ptr, ctypes.c_int64(n), 512, ibuf.ptr,
+                               ob512.ptr), "split512"))
+    report("partition_indices_stable_split_k512", sec, 4 * n * 2 + 4 * n,
+           {"rows_per_s": n / sec})
+    gpu._check(L.bg_partition_ids(hbuf.ptr, ctypes.c_int64(n), k,
+                                  pbuf.ptr), "pids-restore")
+
     # ---- sort: 150M i64 keys (8 radix passes) ----
     permbuf = ctx.alloc(4 * n)
     kcarr = (gpu.BgColumn * 1)(kc)
