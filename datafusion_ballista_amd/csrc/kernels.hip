@@ -1225,10 +1225,13 @@ struct BgJoinTable {
   int64_t probe_n = 0;
 };
 
-__global__ void k_join_build(const int64_t* keys, int64_t n, int* head,
-                             ulong2* nodes, u64 mask) {
+__global__ void k_join_build(const int64_t* keys, const uint8_t* valid,
+                             int64_t n, int* head, ulong2* nodes, u64 mask) {
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
        i += (int64_t)gridDim.x * blockDim.x) {
+    // inner-join null semantics (null_equals_null=false in the reference's
+    // HashJoinExec): a NULL build key matches nothing — leave it unlinked
+    if (!bit_valid(valid, i)) continue;
     const u64 k = (u64)keys[i];
     const u64 b = bg_hash_u64(k) & mask;
     const int prev = atomicExch(&head[b], (int)i);
@@ -1243,7 +1246,8 @@ __global__ void k_join_build(const int64_t* keys, int64_t n, int* head,
 // walks so the ~900-cycle random node loads overlap (memory-level
 // parallelism) instead of serialising per probe.
 #define JOIN_ILP 4
-__global__ void k_join_count(const int64_t* probe_keys, int64_t n_probe,
+__global__ void k_join_count(const int64_t* probe_keys,
+                             const uint8_t* probe_valid, int64_t n_probe,
                              const int* head, const ulong2* nodes, u64 mask,
                              u64* counts) {
   const int64_t stride = (int64_t)gridDim.x * blockDim.x;
@@ -1256,7 +1260,7 @@ __global__ void k_join_count(const int64_t* probe_keys, int64_t n_probe,
 #pragma unroll
     for (int j = 0; j < JOIN_ILP; ++j) {
       idx[j] = base + (int64_t)j * stride;
-      const bool act = idx[j] < n_probe;
+      const bool act = idx[j] < n_probe && bit_valid(probe_valid, idx[j]);
       key[j] = act ? (u64)probe_keys[idx[j]] : 0;
       cur[j] = act ? (int64_t)head[bg_hash_u64(key[j]) & mask] : -1;
       cnt[j] = 0;
@@ -1280,7 +1284,8 @@ __global__ void k_join_count(const int64_t* probe_keys, int64_t n_probe,
   }
 }
 
-__global__ void k_join_fill(const int64_t* probe_keys, int64_t n_probe,
+__global__ void k_join_fill(const int64_t* probe_keys,
+                             const uint8_t* probe_valid, int64_t n_probe,
                             const int* head, const ulong2* nodes, u64 mask,
                             const i64* offsets, uint32_t* out_probe,
                             uint32_t* out_build) {
@@ -1294,7 +1299,7 @@ __global__ void k_join_fill(const int64_t* probe_keys, int64_t n_probe,
 #pragma unroll
     for (int j = 0; j < JOIN_ILP; ++j) {
       idx[j] = base + (int64_t)j * stride;
-      const bool act = idx[j] < n_probe;
+      const bool act = idx[j] < n_probe && bit_valid(probe_valid, idx[j]);
       key[j] = act ? (u64)probe_keys[idx[j]] : 0;
       cur[j] = act ? (int64_t)head[bg_hash_u64(key[j]) & mask] : -1;
       w[j] = act ? offsets[idx[j]] : 0;
@@ -1341,8 +1346,8 @@ extern "C" int bg_hashjoin_build(const bg_column* build_keys, int64_t n,
   int blocks = (int)bg_imin64((n + BG_BLOCK - 1) / BG_BLOCK, BG_MAX_BLOCKS);
   if (blocks == 0) blocks = 1;
   hipLaunchKernelGGL(k_join_build, dim3(blocks), dim3(BG_BLOCK), 0, 0,
-                     (const int64_t*)build_keys->d_data, n, t.head, t.nodes,
-                     t.mask);
+                     (const int64_t*)build_keys->d_data,
+                     build_keys->d_validity, n, t.head, t.nodes, t.mask);
   HIP_TRY(hipGetLastError());
   BgJoinTable* h = new BgJoinTable(t);
   *out_handle = h;
@@ -1372,7 +1377,8 @@ extern "C" int bg_hashjoin_probe_count(void* handle,
   int blocks = (int)bg_imin64((n + BG_BLOCK - 1) / BG_BLOCK, BG_MAX_BLOCKS);
   if (blocks == 0) blocks = 1;
   hipLaunchKernelGGL(k_join_count, dim3(blocks), dim3(BG_BLOCK), 0, 0,
-                     (const int64_t*)probe_keys->d_data, n, t->head, t->nodes,
+                     (const int64_t*)probe_keys->d_data,
+                     probe_keys->d_validity, n, t->head, t->nodes,
                      t->mask, d_counts);
   {
     int rc = scan_exclusive_i64(d_counts, n, d_offs, d_total);
@@ -1400,7 +1406,8 @@ extern "C" int bg_hashjoin_probe_fill(void* handle,
   int blocks = (int)bg_imin64((n + BG_BLOCK - 1) / BG_BLOCK, BG_MAX_BLOCKS);
   if (blocks == 0) blocks = 1;
   hipLaunchKernelGGL(k_join_fill, dim3(blocks), dim3(BG_BLOCK), 0, 0,
-                     (const int64_t*)probe_keys->d_data, n, t->head, t->nodes,
+                     (const int64_t*)probe_keys->d_data,
+                     probe_keys->d_validity, n, t->head, t->nodes,
                      t->mask, t->probe_offsets, d_out_probe,
                      d_out_build);
   HIP_TRY(hipGetLastError());
@@ -1446,8 +1453,11 @@ extern "C" int bg_hashjoin_free(void* handle) {
 
 struct AggArgs {
   int naggs;
+  int vmode;  // 1: record carries a per-agg non-null count word (SQL needs
+              // it to report SUM/MIN/MAX of an all-NULL group as NULL)
   struct {
     const void* data;
+    const uint8_t* valid;  // Arrow LSB validity or NULL (all valid)
     int op;
   } a[BG_MAX_AGGS];
 };
@@ -1455,6 +1465,14 @@ struct AggArgs {
 __device__ __forceinline__ bool keys_equal_rows(const KeyArgs& keys,
                                                 int64_t r1, int64_t r2) {
   for (int c = 0; c < keys.nkeys; ++c) {
+    // GROUP BY null semantics (group_values in the reference's DataFusion):
+    // NULL groups with NULL; NULL never equals a value
+    if (keys.k[c].valid) {
+      const bool v1 = bit_valid(keys.k[c].valid, r1);
+      const bool v2 = bit_valid(keys.k[c].valid, r2);
+      if (v1 != v2) return false;
+      if (!v1) continue;
+    }
     switch (keys.k[c].dtype) {
       case BG_DT_INT64:
         if (reinterpret_cast<const int64_t*>(keys.k[c].data)[r1] !=
@@ -1498,6 +1516,7 @@ __device__ __forceinline__ bool keys_equal_rows(const KeyArgs& keys,
 __device__ __forceinline__ u64 hash_keys_row(const KeyArgs& keys, int64_t i) {
   u64 h = 0;
   for (int c = 0; c < keys.nkeys; ++c) {
+    if (!bit_valid(keys.k[c].valid, i)) continue;  // null: leave running hash
     u64 hc;
     switch (keys.k[c].dtype) {
       case BG_DT_INT64:
@@ -1531,7 +1550,8 @@ __device__ __forceinline__ u64 hash_keys_row(const KeyArgs& keys, int64_t i) {
 __global__ void k_hashagg(KeyArgs keys, AggArgs aggs, const u64* mask_words,
                           int64_t n, u64 cap_mask, u64* slot_data,
                           int* err_flag) {
-  const int rec = 2 + 2 * aggs.naggs;
+  const int wpa = 2 + aggs.vmode;  // words per agg in the slot record
+  const int rec = 2 + wpa * aggs.naggs;
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
        i += (int64_t)gridDim.x * blockDim.x) {
     if (mask_words && !((mask_words[i >> 6] >> (i & 63)) & 1)) continue;
@@ -1560,7 +1580,10 @@ __global__ void k_hashagg(KeyArgs keys, AggArgs aggs, const u64* mask_words,
     }
     atomicAdd(&srec[1], 1ull);
     for (int a = 0; a < aggs.naggs; ++a) {
-      u64* base = srec + 2 + 2 * a;
+      u64* base = srec + 2 + wpa * a;
+      // SQL aggregate null semantics: a NULL value contributes nothing
+      if (!bit_valid(aggs.a[a].valid, i)) continue;
+      if (aggs.vmode) atomicAdd(&base[2], 1ull);
       switch (aggs.a[a].op) {
         case BG_AGG_SUM_DEC128: {
           const ulong2 v =
@@ -1617,10 +1640,11 @@ __global__ void k_slot_occupancy(const u64* slot_data, int rec, int64_t cap,
 // gather group outputs: for dense group g = 0..ngroups-1 with slot index
 // sidx[g]: first_row[g], counts_out[g], acc_out[g*naggs*2 ..]
 __global__ void k_hashagg_gather(const uint32_t* sidx, int64_t ngroups,
-                                 const u64* slot_data, int naggs,
+                                 const u64* slot_data, int naggs, int vmode,
                                  uint32_t* first_row, u64* counts_out,
-                                 u64* acc_out) {
-  const int rec = 2 + 2 * naggs;
+                                 u64* acc_out, u64* nncnt_out) {
+  const int wpa = 2 + vmode;
+  const int rec = 2 + wpa * naggs;
   for (int64_t g = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; g < ngroups;
        g += (int64_t)gridDim.x * blockDim.x) {
     const int64_t s = sidx[g];
@@ -1628,19 +1652,22 @@ __global__ void k_hashagg_gather(const uint32_t* sidx, int64_t ngroups,
     first_row[g] = (uint32_t)(srec[0] - 1);
     counts_out[g] = srec[1];
     for (int a = 0; a < naggs; ++a) {
-      acc_out[((u64)g * naggs + a) * 2] = srec[2 + 2 * a];
-      acc_out[((u64)g * naggs + a) * 2 + 1] = srec[3 + 2 * a];
+      acc_out[((u64)g * naggs + a) * 2] = srec[2 + wpa * a];
+      acc_out[((u64)g * naggs + a) * 2 + 1] = srec[2 + wpa * a + 1];
+      if (vmode && nncnt_out)
+        nncnt_out[(u64)g * naggs + a] = srec[2 + wpa * a + 2];
     }
   }
 }
 
-extern "C" int bg_hashagg(const bg_column* key_cols, int32_t nkeys,
-                          const bg_column* agg_cols, const int32_t* agg_ops,
-                          int32_t naggs, const uint8_t* d_mask, int64_t n,
-                          int64_t max_groups, uint32_t* d_first_row,
-                          uint8_t* d_acc_out, int64_t* d_counts_out,
-                          int64_t* out_ngroups) {
+static int hashagg_impl(const bg_column* key_cols, int32_t nkeys,
+                        const bg_column* agg_cols, const int32_t* agg_ops,
+                        int32_t naggs, const uint8_t* d_mask, int64_t n,
+                        int64_t max_groups, uint32_t* d_first_row,
+                        uint8_t* d_acc_out, int64_t* d_counts_out,
+                        int64_t* d_nncnt_out, int64_t* out_ngroups) {
   REQUIRE_INIT();
+  const int vmode = d_nncnt_out ? 1 : 0;
   if (nkeys <= 0 || nkeys > BG_MAX_KEYS)
     return set_err(BG_ERR_INVALID, "nkeys out of range [1,4]");
   if (naggs < 0 || naggs > BG_MAX_AGGS)
@@ -1655,13 +1682,15 @@ extern "C" int bg_hashagg(const bg_column* key_cols, int32_t nkeys,
   }
   AggArgs aggs{};
   aggs.naggs = naggs;
+  aggs.vmode = vmode;
   for (int i = 0; i < naggs; ++i) {
     aggs.a[i].data = agg_cols[i].d_data;
+    aggs.a[i].valid = agg_cols[i].d_validity;
     aggs.a[i].op = agg_ops[i];
   }
   u64 cap = 8;
   while (cap < (u64)(max_groups * 2)) cap <<= 1;
-  const int rec = 2 + 2 * naggs;
+  const int rec = 2 + (2 + vmode) * naggs;
   u64* slot_data;
   int* err_flag;
   HIP_TRY(pool_malloc((void**)&slot_data, sizeof(u64) * cap * rec));
@@ -1727,9 +1756,10 @@ extern "C" int bg_hashagg(const bg_column* key_cols, int32_t nkeys,
                                BG_MAX_BLOCKS);
   if (gblocks == 0) gblocks = 1;
   hipLaunchKernelGGL(k_hashagg_gather, dim3(gblocks), dim3(BG_BLOCK), 0, 0,
-                     sidx, ngroups, slot_data, naggs, d_first_row,
+                     sidx, ngroups, slot_data, naggs, vmode, d_first_row,
                      reinterpret_cast<u64*>(d_counts_out),
-                     reinterpret_cast<u64*>(d_acc_out));
+                     reinterpret_cast<u64*>(d_acc_out),
+                     reinterpret_cast<u64*>(d_nncnt_out));
   HIP_TRY(hipGetLastError());
   HIP_TRY(hipDeviceSynchronize());
   HIP_TRY(pool_release(slot_data));
@@ -1738,6 +1768,33 @@ extern "C" int bg_hashagg(const bg_column* key_cols, int32_t nkeys,
   HIP_TRY(pool_release(sidx));
   *out_ngroups = ngroups;
   return BG_OK;
+}
+
+extern "C" int bg_hashagg(const bg_column* key_cols, int32_t nkeys,
+                          const bg_column* agg_cols, const int32_t* agg_ops,
+                          int32_t naggs, const uint8_t* d_mask, int64_t n,
+                          int64_t max_groups, uint32_t* d_first_row,
+                          uint8_t* d_acc_out, int64_t* d_counts_out,
+                          int64_t* out_ngroups) {
+  return hashagg_impl(key_cols, nkeys, agg_cols, agg_ops, naggs, d_mask, n,
+                      max_groups, d_first_row, d_acc_out, d_counts_out,
+                      nullptr, out_ngroups);
+}
+
+/* As bg_hashagg but also reports the per-(group, aggregate) NON-NULL input
+ * count (d_nncnt_out, ngroups*naggs i64): SQL reports SUM/MIN/MAX of a
+ * group whose inputs were all NULL as NULL, which the accumulator value
+ * alone cannot encode.  Mirrors the reference's accumulators
+ * (datafusion/physical-expr aggregates) returning Option-valued state. */
+extern "C" int bg_hashagg2(const bg_column* key_cols, int32_t nkeys,
+                           const bg_column* agg_cols, const int32_t* agg_ops,
+                           int32_t naggs, const uint8_t* d_mask, int64_t n,
+                           int64_t max_groups, uint32_t* d_first_row,
+                           uint8_t* d_acc_out, int64_t* d_counts_out,
+                           int64_t* d_nncnt_out, int64_t* out_ngroups) {
+  return hashagg_impl(key_cols, nkeys, agg_cols, agg_ops, naggs, d_mask, n,
+                      max_groups, d_first_row, d_acc_out, d_counts_out,
+                      d_nncnt_out, out_ngroups);
 }
 
 // ---------------------------------------------------------------------------

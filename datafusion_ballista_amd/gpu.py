@@ -361,13 +361,16 @@ def decode_agg_value(op, raw16: bytes) -> int:
 
 
 def _ctx_hashagg(self, key_cols, agg_cols, agg_ops, n, max_groups,
-                 mask: "DeviceBuffer" = None):
+                 mask: "DeviceBuffer" = None, want_nncnt: bool = False):
     """General hash group-by. -> (first_row u32[g], acc i128 bytes
-    [g, naggs], counts i64[g]) as numpy arrays (downloaded)."""
+    [g, naggs], counts i64[g]) as numpy arrays (downloaded); with
+    want_nncnt also the per-(group, agg) NON-NULL input counts i64[g, naggs]
+    (SQL: SUM/MIN/MAX of an all-NULL group is NULL — nncnt 0 marks it)."""
     naggs = len(agg_cols)
     first = self.alloc(max(4 * max_groups, 4))
     acc = self.alloc(max(16 * max_groups * max(naggs, 1), 16))
     counts = self.alloc(max(8 * max_groups, 8))
+    nncnt = self.alloc(max(8 * max_groups * max(naggs, 1), 8))         if want_nncnt else None
     karr = (BgColumn * len(key_cols))(*key_cols)
     aarr = (BgColumn * max(naggs, 1))(*(agg_cols or [BgColumn()]))
     oarr = (ctypes.c_int32 * max(naggs, 1))(*(agg_ops or [0]))
@@ -376,10 +379,19 @@ def _ctx_hashagg(self, key_cols, agg_cols, agg_ops, n, max_groups,
     # like the reference's AQE-fed group estimates)
     attempt_groups = max_groups
     for _ in range(8):
-        rc = self.L.bg_hashagg(karr, len(key_cols), aarr, oarr, naggs,
-                               mask.ptr if mask else None, ctypes.c_int64(n),
-                               ctypes.c_int64(attempt_groups), first.ptr,
-                               acc.ptr, counts.ptr, ctypes.byref(ng))
+        if want_nncnt:
+            rc = self.L.bg_hashagg2(karr, len(key_cols), aarr, oarr, naggs,
+                                    mask.ptr if mask else None,
+                                    ctypes.c_int64(n),
+                                    ctypes.c_int64(attempt_groups), first.ptr,
+                                    acc.ptr, counts.ptr, nncnt.ptr,
+                                    ctypes.byref(ng))
+        else:
+            rc = self.L.bg_hashagg(karr, len(key_cols), aarr, oarr, naggs,
+                                   mask.ptr if mask else None,
+                                   ctypes.c_int64(n),
+                                   ctypes.c_int64(attempt_groups), first.ptr,
+                                   acc.ptr, counts.ptr, ctypes.byref(ng))
         if rc == 0:
             break
         err = load_library().bg_last_error().decode()
@@ -390,13 +402,19 @@ def _ctx_hashagg(self, key_cols, agg_cols, agg_ops, n, max_groups,
         first = self.alloc(max(4 * attempt_groups, 4))
         acc = self.alloc(max(16 * attempt_groups * max(naggs, 1), 16))
         counts = self.alloc(max(8 * attempt_groups, 8))
+        if want_nncnt:
+            nncnt = self.alloc(max(8 * attempt_groups * max(naggs, 1), 8))
     else:
         _check(rc, "bg_hashagg")
     g = ng.value
-    return (first.download(np.uint32, g),
-            acc.download(np.uint8, 16 * g * naggs).reshape(g, naggs, 16)
-            if naggs else np.zeros((g, 0, 16), dtype=np.uint8),
-            counts.download(np.int64, g))
+    ret = (first.download(np.uint32, g),
+           acc.download(np.uint8, 16 * g * naggs).reshape(g, naggs, 16)
+           if naggs else np.zeros((g, 0, 16), dtype=np.uint8),
+           counts.download(np.int64, g))
+    if want_nncnt:
+        nn = nncnt.download(np.int64, g * naggs).reshape(g, max(naggs, 1))             if naggs else np.zeros((g, 0), dtype=np.int64)
+        return ret + (nn,)
+    return ret
 
 
 GpuStageContext.hashagg = _ctx_hashagg

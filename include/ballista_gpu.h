@@ -229,6 +229,21 @@ int bg_hashagg(const bg_column* key_cols, int32_t nkeys,
                int64_t* d_counts_out /* max_groups */,
                int64_t* out_ngroups);
 
+/* As bg_hashagg, plus d_nncnt_out (max_groups*naggs i64): the per-
+ * (group, aggregate) count of NON-NULL inputs.  NULL aggregate inputs are
+ * always skipped (SQL semantics; the reference's accumulators in
+ * datafusion/functions-aggregate treat null as no contribution); a group
+ * whose inputs were all NULL has nncnt 0 and its SUM/MIN/MAX is NULL —
+ * the accumulator value alone cannot encode that.  NULL group keys hash
+ * as "no contribution" (hash_utils create_hashes) and group together
+ * (NULL == NULL in GROUP BY). */
+int bg_hashagg2(const bg_column* key_cols, int32_t nkeys,
+                const bg_column* agg_cols, const int32_t* agg_ops,
+                int32_t naggs, const uint8_t* d_mask, int64_t n,
+                int64_t max_groups, uint32_t* d_first_row,
+                uint8_t* d_acc_out, int64_t* d_counts_out,
+                int64_t* d_nncnt_out, int64_t* out_ngroups);
+
 /* ---- ProjectionExec expression subset: Decimal128 arithmetic ----
  * ops: 0 a*b, 1 a+b, 2 a-b, 3 lit-a, 4 a*lit, 5 a+lit (exact i128;
  * lit as (lo,hi) i128 halves).  d_out: n x 16 B. */

@@ -306,6 +306,39 @@ def hashjoin_pairs(build: np.ndarray, probe: np.ndarray):
     return op, ob
 
 
+def hashagg_nulls(keys, aggs, n, mask=None):
+    """Null-aware serial group-by restatement (SQL semantics, exact python
+    ints): keys = [(array, valid_bool_or_None)], aggs = [(op, vals,
+    valid_bool_or_None)] with op in {"sum", "min", "max"}.  NULL group keys
+    group together (None in the key tuple); NULL aggregate inputs contribute
+    nothing.  -> dict key_tuple -> (count, [acc-or-None], [nncnt]).
+    Restates the reference's accumulators (datafusion/functions-aggregate
+    sum.rs/min_max.rs: null input -> no update) and group_values null keys."""
+    out = {}
+    for i in range(n):
+        if mask is not None and not mask[i]:
+            continue
+        kt = tuple(
+            None if (kv is not None and not kv[i]) else
+            (int(k[i]) if hasattr(k[i], "item") else k[i])
+            for k, kv in keys)
+        e = out.setdefault(kt, [0, [None] * len(aggs), [0] * len(aggs)])
+        e[0] += 1
+        for a, (op, vals, vv) in enumerate(aggs):
+            if vv is not None and not vv[i]:
+                continue
+            v = int(vals[i])
+            e[2][a] += 1
+            cur = e[1][a]
+            if op == "sum":
+                e[1][a] = v if cur is None else cur + v
+            elif op == "min":
+                e[1][a] = v if cur is None else min(cur, v)
+            elif op == "max":
+                e[1][a] = v if cur is None else max(cur, v)
+    return {k: (v[0], v[1], v[2]) for k, v in out.items()}
+
+
 def hashagg(keys, aggs, n, mask=None):
     """Serial group-by restatement (exact python-int arithmetic).
 

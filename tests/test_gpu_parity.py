@@ -1227,3 +1227,90 @@ def test_filter_with_validity(ctx):
     _, m = ctx.mask_to_indices(mask, n)
     want = oracle.filter_mask([("i64", vals, valid, 4, 10, 0)], n)
     assert m == len(oracle.mask_to_indices(want, n))
+
+
+# ---------------------------------------------------------------------------
+# NULL semantics through the aggregate and join paths
+# ---------------------------------------------------------------------------
+def _bits(bm, i):
+    return (int(bm[i >> 3]) >> (i & 7)) & 1
+
+
+def test_hashagg_nulls(ctx):
+    """GROUP BY a null-carrying key with SUM/MIN/MAX over null-carrying
+    values: NULL keys group together, NULL inputs contribute nothing, and
+    an all-NULL group's aggregate is NULL (nncnt 0) — the reference's
+    group_values + functions-aggregate null semantics, vs the oracle."""
+    n = 100_000
+    rng = np.random.default_rng(77)
+    k1 = rng.integers(0, 50, size=n, dtype=np.int64)
+    kv = rng.integers(0, 256, size=(n + 7) // 8, dtype=np.uint8)
+    dec = rng.integers(-10**10, 10**10, size=n, dtype=np.int64)
+    dec16 = dec_bytes(dec)
+    decv = rng.integers(0, 256, size=(n + 7) // 8, dtype=np.uint8)
+    v64 = rng.integers(-10**6, 10**6, size=n, dtype=np.int64)
+    v64v = rng.integers(0, 256, size=(n + 7) // 8, dtype=np.uint8)
+
+    c1, _ = ctx.upload_column(k1, gpu.BG_DT_INT64, validity=kv)
+    ca = ctx.column(gpu.BG_DT_DECIMAL128, ctx.upload(dec16), n,
+                    validity=ctx.upload(decv))
+    cb, _ = ctx.upload_column(v64, gpu.BG_DT_INT64, validity=v64v)
+    cc, _ = ctx.upload_column(v64, gpu.BG_DT_INT64, validity=v64v)
+    first, acc, counts, nn = ctx.hashagg(
+        [c1], [ca, cb, cc],
+        [gpu.BG_AGG_OP_SUM_DEC128, gpu.BG_AGG_OP_MIN_I64,
+         gpu.BG_AGG_OP_MAX_I64], n, max_groups=256, want_nncnt=True)
+
+    got = {}
+    for g in range(len(first)):
+        r = int(first[g])
+        key = int(k1[r]) if _bits(kv, r) else None
+        accs = [None if nn[g, a] == 0 else
+                gpu.decode_agg_value(op, bytes(acc[g, a]))
+                for a, op in enumerate([gpu.BG_AGG_OP_SUM_DEC128,
+                                        gpu.BG_AGG_OP_MIN_I64,
+                                        gpu.BG_AGG_OP_MAX_I64])]
+        assert key not in got, "NULL keys must form one group"
+        got[key] = (int(counts[g]), accs, [int(x) for x in nn[g]])
+
+    kvb = np.array([_bits(kv, i) for i in range(n)], dtype=bool)
+    dvb = np.array([_bits(decv, i) for i in range(n)], dtype=bool)
+    vvb = np.array([_bits(v64v, i) for i in range(n)], dtype=bool)
+    want = oracle.hashagg_nulls(
+        [(k1, kvb)], [("sum", dec, dvb), ("min", v64, vvb),
+                      ("max", v64, vvb)], n)
+    want = {k[0]: (c, a, nnw) for k, (c, a, nnw) in want.items()}
+    assert got == want
+
+
+def test_hashjoin_nulls(ctx):
+    """Inner join with NULL keys on both sides: null_equals_null=false —
+    a NULL key matches nothing (the reference HashJoinExec default)."""
+    nb, np_ = 5_000, 20_000
+    rng = np.random.default_rng(88)
+    bk = rng.integers(0, 2_000, size=nb, dtype=np.int64)
+    bv = rng.integers(0, 256, size=(nb + 7) // 8, dtype=np.uint8)
+    pk = rng.integers(0, 2_000, size=np_, dtype=np.int64)
+    pv = rng.integers(0, 256, size=(np_ + 7) // 8, dtype=np.uint8)
+
+    cb, _ = ctx.upload_column(bk, gpu.BG_DT_INT64, validity=bv)
+    cp, _ = ctx.upload_column(pk, gpu.BG_DT_INT64, validity=pv)
+    j = gpu.GpuHashJoin(ctx, cb, nb)
+    ppos, bpos, nm = j.probe(cp, np_)
+    gp = ppos.download(np.uint32, nm)
+    gb = bpos.download(np.uint32, nm)
+    got = set(zip(gp.tolist(), gb.tolist()))
+
+    bvb = np.array([_bits(bv, i) for i in range(nb)], dtype=bool)
+    pvb = np.array([_bits(pv, i) for i in range(np_)], dtype=bool)
+    want = set()
+    from collections import defaultdict
+    idx = defaultdict(list)
+    for i in range(nb):
+        if bvb[i]:
+            idx[int(bk[i])].append(i)
+    for i in range(np_):
+        if pvb[i]:
+            for b in idx.get(int(pk[i]), ()):
+                want.add((i, b))
+    assert got == want

@@ -173,3 +173,22 @@ def test_gather_roundtrip():
     idx = np.array([5, 1, 99, 0, 5], dtype=np.uint32)
     out = oracle.gather(src, 8, idx).view(np.int64)
     assert np.array_equal(out, src[idx])
+
+
+def test_hashagg_nulls_restatement():
+    """Pin the null-aware group-by restatement on a hand-computed case."""
+    import numpy as np
+    import oracle
+    keys = np.array([1, 1, 2, 2, 3], dtype=np.int64)
+    kvalid = np.array([True, True, True, False, False])
+    vals = np.array([10, 20, 30, 40, 50], dtype=np.int64)
+    vvalid = np.array([True, False, False, True, True])
+    got = oracle.hashagg_nulls([(keys, kvalid)],
+                               [("sum", vals, vvalid),
+                                ("min", vals, vvalid)], 5)
+    # rows: (1,10v) (1,20n) (2,30n) (None,40v) (None,50v)
+    assert got == {
+        (1,): (2, [10, 10], [1, 1]),
+        (2,): (1, [None, None], [0, 0]),
+        (None,): (2, [90, 40], [2, 2]),
+    }
