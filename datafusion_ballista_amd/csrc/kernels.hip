@@ -2456,13 +2456,27 @@ extern "C" int bg_snappy_decompress(const void* h_pages, int64_t npages,
 __device__ void k_page_extract_body(const uint8_t* page, int64_t page_len,
                                     uint8_t* out, int64_t nvals,
                                     int64_t src_esz, int has_def,
-                                    int flba_reverse, int* err) {
+                                    int flba_reverse, const uint32_t* vidx,
+                                    const int64_t* n_present, int* err) {
   __shared__ int64_t s_voff;
   __syncthreads();
   if (threadIdx.x == 0) {
     int64_t voff = 0;
     int ok = 1;
-    if (has_def) {
+    if (has_def == 2) {
+      // nullable: levels already decoded by bg_def_levels_batch — just
+      // skip past them and bound the packed value section
+      if (page_len < 4) ok = 0;
+      else {
+        const uint32_t dlen = (uint32_t)page[0] | ((uint32_t)page[1] << 8) |
+                              ((uint32_t)page[2] << 16) |
+                              ((uint32_t)page[3] << 24);
+        voff = 4 + (int64_t)dlen;
+        if (voff + (*n_present) * src_esz > page_len) ok = 0;
+      }
+      if (!ok) atomicExch(err, 2);
+      s_voff = ok ? voff : -1;
+    } else if (has_def) {
       const uint32_t dlen = (uint32_t)page[0] | ((uint32_t)page[1] << 8) |
                             ((uint32_t)page[2] << 16) |
                             ((uint32_t)page[3] << 24);
@@ -2499,14 +2513,35 @@ __device__ void k_page_extract_body(const uint8_t* page, int64_t page_len,
       }
       if (seen < nvals) ok = 0;
       voff = 4 + dlen;
+      if (voff + nvals * src_esz > page_len) ok = 0;  // geometry bound
+      if (!ok) atomicExch(err, 2);  // malformed / unexpected nulls
+      s_voff = ok ? voff : -1;
+    } else {
+      if (voff + nvals * src_esz > page_len) ok = 0;  // geometry bound
+      if (!ok) atomicExch(err, 2);
+      s_voff = ok ? voff : -1;
     }
-    if (voff + nvals * src_esz > page_len) ok = 0;  // geometry bound
-    if (!ok) atomicExch(err, 2);  // nulls / malformed: unsupported this round
-    s_voff = ok ? voff : -1;
   }
   __syncthreads();
   const int64_t voff = s_voff;
   if (voff < 0) return;
+  if (has_def == 2) {
+    // scatter the packed value stream to its slots (NULL slots untouched)
+    for (int64_t v = threadIdx.x; v < nvals; v += blockDim.x) {
+      const uint32_t ix = vidx[v];
+      if (ix == 0xffffffffu) continue;
+      const uint8_t* src = page + voff + (int64_t)ix * src_esz;
+      uint8_t* dst = out + v * (flba_reverse ? 16 : src_esz);
+      if (!flba_reverse) {
+        for (int64_t b = 0; b < src_esz; ++b) dst[b] = src[b];
+      } else {
+        const uint8_t sign = (src[0] & 0x80) ? 0xff : 0x00;
+        for (int64_t b = 0; b < 16; ++b)
+          dst[b] = (b < src_esz) ? src[src_esz - 1 - b] : sign;
+      }
+    }
+    return;
+  }
   if (!flba_reverse) {
     const int64_t nbytes = nvals * src_esz;
     for (int64_t i = threadIdx.x; i < nbytes; i += blockDim.x)
@@ -2530,7 +2565,7 @@ __global__ void k_page_extract(const uint8_t* page, int64_t page_len,
                                int has_def, int flba_reverse, int* err) {
   if (blockIdx.x != 0) return;
   k_page_extract_body(page, page_len, out, nvals, src_esz, has_def,
-                      flba_reverse, err);
+                      flba_reverse, nullptr, nullptr, err);
 }
 
 extern "C" int bg_page_extract(const void* d_page, int64_t page_len,
@@ -2571,15 +2606,35 @@ extern "C" int bg_page_extract(const void* d_page, int64_t page_len,
 // bit-packed groups (one group of 8 per lane) and RLE fills in parallel.
 __device__ void k_dict_indices_body(const uint8_t* page, int64_t page_len,
                                     int64_t nvals, int has_def,
-                                    uint32_t* out_idx, int* err) {
+                                    uint32_t* out_idx, const uint32_t* vidx,
+                                    const int64_t* n_present, uint32_t* dense,
+                                    int* err) {
   const int lane = lane_id();
+  const int64_t nslots = nvals;
+  uint32_t* const slots_out = out_idx;
+  if (has_def == 2) {
+    // nullable: the packed index stream holds n_present entries; decode
+    // them densely into `dense`, then scatter to slots through vidx
+    // (NULL slots get index 0 — a valid dictionary read whose value the
+    // validity bitmap masks, matching the reference's null-padded take)
+    nvals = *n_present;
+    out_idx = dense;
+  }
   int64_t doff = 0;
   int ok = 1;
   int bw = 0;
   if (lane == 0) {
     const uint8_t* d = page;
     const uint8_t* pend = page + page_len;
-    if (has_def) {
+    if (has_def == 2) {
+      if (d + 4 > pend) ok = 0;
+      else {
+        const uint32_t dlen = (uint32_t)d[0] | ((uint32_t)d[1] << 8) |
+                              ((uint32_t)d[2] << 16) | ((uint32_t)d[3] << 24);
+        doff = 4 + (int64_t)dlen;
+        if (doff > page_len) ok = 0;
+      }
+    } else if (has_def) {
       if (d + 4 > pend) ok = 0;
       else {
         const uint32_t dlen = (uint32_t)d[0] | ((uint32_t)d[1] << 8) |
@@ -2634,7 +2689,7 @@ __device__ void k_dict_indices_body(const uint8_t* page, int64_t page_len,
     return;
   }
   if (bw == 0) {
-    for (int64_t t = lane; t < nvals; t += BG_WAVE) out_idx[t] = 0;
+    for (int64_t t = lane; t < nslots; t += BG_WAVE) slots_out[t] = 0;
     return;
   }
   const u64 vmask = (bw == 32) ? 0xffffffffull : ((1ull << bw) - 1);
@@ -2710,13 +2765,21 @@ __device__ void k_dict_indices_body(const uint8_t* page, int64_t page_len,
       si = nsi;
     }
   }
+  if (has_def == 2) {
+    __builtin_amdgcn_wave_barrier();
+    for (int64_t t = lane; t < nslots; t += BG_WAVE) {
+      const uint32_t ix = vidx[t];
+      slots_out[t] = (ix == 0xffffffffu) ? 0u : dense[ix];
+    }
+  }
 }
 
 __global__ void k_dict_indices(const uint8_t* page, int64_t page_len,
                                int64_t nvals, int has_def, uint32_t* out_idx,
                                int* err) {
   if (blockIdx.x != 0 || threadIdx.x >= BG_WAVE) return;
-  k_dict_indices_body(page, page_len, nvals, has_def, out_idx, err);
+  k_dict_indices_body(page, page_len, nvals, has_def, out_idx, nullptr,
+                      nullptr, nullptr, err);
 }
 
 extern "C" int bg_dict_indices(const void* d_page, int64_t page_len,
@@ -2741,6 +2804,123 @@ extern "C" int bg_dict_indices(const void* d_page, int64_t page_len,
 }
 
 // ---------------------------------------------------------------------------
+// Definition-level decode for OPTIONAL (nullable) columns: parquet data
+// pages of a max_def=1 column prefix the payload with [u32 len][RLE/bit-
+// packed bit-width-1 levels]; level 1 = value present, 0 = NULL (format
+// restated from the parquet-format spec Encodings.md, the same hybrid the
+// reference's parquet crate decodes in
+// parquet/src/encodings/rle.rs).  Lane 0 walks one page's levels serially
+// (pages decode concurrently across waves), emitting
+//   - the column's Arrow LSB validity bits (atomicOr into u32 words at an
+//     arbitrary bit offset — page boundaries are not byte-aligned),
+//   - vidx[s] = index of slot s's value in the page's packed value
+//     stream (~0u for NULL slots),
+//   - n_present = how many values the page's value section holds.
+// ---------------------------------------------------------------------------
+struct DefLevelsJob {
+  const uint8_t* page;     // page start ([u32 dlen][levels]...)
+  uint32_t* vidx;          // u32[nvals] (page-local slice)
+  uint32_t* valid_out;     // column validity bitmap as u32 words
+  int64_t page_len;
+  int64_t nvals;
+  int64_t bit_off;         // absolute bit position of this page's slot 0
+  int64_t* n_present;
+};
+
+__device__ void k_def_levels_body(const DefLevelsJob& job, int* err) {
+  if (lane_id() != 0) return;
+  const uint8_t* page = job.page;
+  if (job.page_len < 4) { atomicExch(err, 3); return; }
+  const uint32_t dlen = (uint32_t)page[0] | ((uint32_t)page[1] << 8) |
+                        ((uint32_t)page[2] << 16) | ((uint32_t)page[3] << 24);
+  const uint8_t* d = page + 4;
+  const uint8_t* dend = d + dlen;
+  if (dend > page + job.page_len) { atomicExch(err, 3); return; }
+  int64_t s = 0;        // slot
+  uint32_t cnt = 0;     // values seen
+  int64_t cur_w = -1;   // current validity word index
+  uint32_t cur = 0;
+  auto emit = [&](int present) {
+    const int64_t bit = job.bit_off + s;
+    const int64_t w = bit >> 5;
+    if (w != cur_w) {
+      if (cur_w >= 0 && cur) atomicOr(&job.valid_out[cur_w], cur);
+      cur_w = w;
+      cur = 0;
+    }
+    if (present) {
+      cur |= 1u << (bit & 31);
+      job.vidx[s] = cnt++;
+    } else {
+      job.vidx[s] = 0xffffffffu;
+    }
+    ++s;
+  };
+  while (d < dend && s < job.nvals) {
+    u64 header = 0;
+    int shift = 0;
+    while (d < dend) {
+      const uint8_t b = *d++;
+      header |= (u64)(b & 0x7f) << shift;
+      if (!(b & 0x80)) break;
+      shift += 7;
+    }
+    if (header & 1) {  // bit-packed groups of 8, 1 byte each (bit width 1)
+      const int64_t groups = (int64_t)(header >> 1);
+      for (int64_t g = 0; g < groups && s < job.nvals; ++g) {
+        if (d >= dend) { atomicExch(err, 3); return; }
+        const uint8_t byte = *d++;
+        const int64_t take = job.nvals - s >= 8 ? 8 : job.nvals - s;
+        for (int64_t t = 0; t < take; ++t) emit((byte >> t) & 1);
+      }
+    } else {  // RLE run
+      const int64_t run = (int64_t)(header >> 1);
+      if (d >= dend) { atomicExch(err, 3); return; }
+      const int v = *d++ & 1;
+      for (int64_t t = 0; t < run && s < job.nvals; ++t) emit(v);
+    }
+  }
+  if (cur_w >= 0 && cur) atomicOr(&job.valid_out[cur_w], cur);
+  if (s < job.nvals) { atomicExch(err, 3); return; }
+  *job.n_present = (int64_t)cnt;
+}
+
+__global__ void k_def_levels_batch(const DefLevelsJob* jobs, int64_t njobs,
+                                   int* err) {
+  const int64_t wave_global =
+      ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) / BG_WAVE;
+  const int64_t nwaves = ((int64_t)gridDim.x * blockDim.x) / BG_WAVE;
+  for (int64_t j = wave_global; j < njobs; j += nwaves)
+    k_def_levels_body(jobs[j], err);
+}
+
+extern "C" int bg_def_levels_batch(const void* h_jobs, int64_t njobs) {
+  REQUIRE_INIT();
+  DefLevelsJob* d_jobs;
+  int* d_err;
+  HIP_TRY(pool_malloc((void**)&d_jobs,
+                      sizeof(DefLevelsJob) * (njobs ? njobs : 1)));
+  HIP_TRY(pool_malloc((void**)&d_err, sizeof(int)));
+  HIP_TRY(hipMemset(d_err, 0, sizeof(int)));
+  HIP_TRY(hipMemcpy(d_jobs, h_jobs, sizeof(DefLevelsJob) * njobs,
+                    hipMemcpyHostToDevice));
+  const int waves_per_block = BG_BLOCK / BG_WAVE;
+  int blocks = (int)bg_imin64((njobs + waves_per_block - 1) / waves_per_block,
+                              BG_MAX_BLOCKS);
+  if (blocks == 0) blocks = 1;
+  hipLaunchKernelGGL(k_def_levels_batch, dim3(blocks), dim3(BG_BLOCK), 0, 0,
+                     d_jobs, njobs, d_err);
+  HIP_TRY(hipGetLastError());
+  int err = 0;
+  HIP_TRY(hipMemcpy(&err, d_err, sizeof(int), hipMemcpyDeviceToHost));
+  (void)pool_release(d_jobs);
+  (void)pool_release(d_err);
+  if (err)
+    return set_err(BG_ERR_INVALID, "bg_def_levels_batch: malformed levels");
+  return BG_OK;
+}
+
+// ---------------------------------------------------------------------------
 // Batched page ops: the per-page host loop (launch + sync error check per
 // page) starves the GPU when chunks hold few pages — these take the whole
 // column's page list in one call.
@@ -2751,8 +2931,10 @@ struct PageExtractJob {
   int64_t page_len;
   int64_t nvals;
   int64_t src_esz;
-  int32_t has_def;
+  int32_t has_def;  // 0 none, 1 validate-all-present, 2 nullable (vidx)
   int32_t flba_reverse;
+  const uint32_t* vidx;      // mode 2: per-slot value index, ~0u = null
+  const int64_t* n_present;  // mode 2: value count in this page
 };
 
 __global__ void k_page_extract_batch(const PageExtractJob* jobs, int64_t njobs,
@@ -2760,7 +2942,8 @@ __global__ void k_page_extract_batch(const PageExtractJob* jobs, int64_t njobs,
   for (int64_t j = blockIdx.x; j < njobs; j += gridDim.x) {
     const PageExtractJob job = jobs[j];
     k_page_extract_body(job.page, job.page_len, job.out, job.nvals,
-                        job.src_esz, job.has_def, job.flba_reverse, err);
+                        job.src_esz, job.has_def, job.flba_reverse,
+                        job.vidx, job.n_present, err);
   }
 }
 
@@ -2793,8 +2976,11 @@ struct DictIndicesJob {
   uint32_t* out_idx;
   int64_t page_len;
   int64_t nvals;
-  int32_t has_def;
+  int32_t has_def;  // 0 none, 1 validate-all-present, 2 nullable (vidx)
   int32_t _pad;
+  const uint32_t* vidx;      // mode 2
+  const int64_t* n_present;  // mode 2
+  uint32_t* dense;           // mode 2: scratch for the packed index stream
 };
 
 __global__ void k_dict_indices_batch(const DictIndicesJob* jobs, int64_t njobs,
@@ -2805,7 +2991,8 @@ __global__ void k_dict_indices_batch(const DictIndicesJob* jobs, int64_t njobs,
   for (int64_t j = wave_global; j < njobs; j += nwaves) {
     const DictIndicesJob job = jobs[j];
     k_dict_indices_body(job.page, job.page_len, job.nvals, job.has_def,
-                        job.out_idx, err);
+                        job.out_idx, job.vidx, job.n_present, job.dense,
+                        err);
   }
 }
 

@@ -505,13 +505,26 @@ class BgPageExtractJob(ctypes.Structure):
     _fields_ = [("d_page", ctypes.c_void_p), ("d_out", ctypes.c_void_p),
                 ("page_len", ctypes.c_int64), ("nvals", ctypes.c_int64),
                 ("src_esz", ctypes.c_int64), ("has_def", ctypes.c_int32),
-                ("flba_reverse", ctypes.c_int32)]
+                ("flba_reverse", ctypes.c_int32),
+                ("d_vidx", ctypes.c_void_p),
+                ("d_n_present", ctypes.c_void_p)]
 
 
 class BgDictIndicesJob(ctypes.Structure):
     _fields_ = [("d_page", ctypes.c_void_p), ("d_out_idx", ctypes.c_void_p),
                 ("page_len", ctypes.c_int64), ("nvals", ctypes.c_int64),
-                ("has_def", ctypes.c_int32), ("_pad", ctypes.c_int32)]
+                ("has_def", ctypes.c_int32), ("_pad", ctypes.c_int32),
+                ("d_vidx", ctypes.c_void_p),
+                ("d_n_present", ctypes.c_void_p),
+                ("d_dense", ctypes.c_void_p)]
+
+
+class BgDefLevelsJob(ctypes.Structure):
+    _fields_ = [("d_page", ctypes.c_void_p), ("d_vidx", ctypes.c_void_p),
+                ("d_valid_out", ctypes.c_void_p),
+                ("page_len", ctypes.c_int64), ("nvals", ctypes.c_int64),
+                ("bit_off", ctypes.c_int64),
+                ("d_n_present", ctypes.c_void_p)]
 
 
 def _ctx_gather_varlen(self, src_data: "DeviceBuffer", src_offsets: "DeviceBuffer",

@@ -9,7 +9,10 @@ footer/page metadata (thrift) and drives the copies — the same split as the
 shuffle path (metadata host-side, bytes device-side).
 
 Out of scope this round (fails loudly): V2 data pages, dictionary-encoded
-data pages, pages with nulls, BYTE_ARRAY strings, other codecs.
+data pages, BYTE_ARRAY strings, other codecs.  Nullable (OPTIONAL,
+max_def=1) columns decode on device: definition levels -> Arrow validity
+bitmap + slot->value scatter (bg_def_levels_batch + mode-2 extract/dict
+jobs).
 
 The thrift compact-protocol PageHeader parser below restates the published
 parquet-format spec (PageHeader/DataPageHeader structs); parity is pinned
@@ -158,7 +161,8 @@ class GpuParquetColumnReader:
         """Decode a column across row groups in ONE batched pass (one
         snappy launch over every page of every chunk — chunks alone hold
         too few pages to fill 256 CUs).
-        -> (DeviceBuffer of raw values, num_values, physical_type)."""
+        -> (DeviceBuffer of raw values, num_values, physical_type,
+        validity DeviceBuffer or None (required/non-null column))."""
         ctx = self.ctx
         if rgs is None:
             rgs = range(self.pf.metadata.num_row_groups)
@@ -224,10 +228,11 @@ class GpuParquetColumnReader:
             return ctypes.c_void_p(base.ptr.value + off)
 
         out = ctx.alloc(max(total_values * dst_esz, dst_esz))
-        extracts = []        # PLAIN: (scratch_off, usz, dst_off, nvals)
+        extracts = []        # PLAIN: (scratch_off, usz, dst_off, nvals, pidx)
         dict_runs = {}       # rg -> {"dict": (soff, usz, ndict), "pages": []}
         got_values = 0
         snappy_jobs = []
+        data_pages = []      # (scratch_off, usz, dst_off, nvals) in order
         for (rg, ptype, poff, csz, usz, nvals, enc, ndict, soff) in headers:
             if codec == "SNAPPY":
                 snappy_jobs.append((poff, csz, soff, usz))
@@ -235,13 +240,15 @@ class GpuParquetColumnReader:
                 dict_runs.setdefault(rg, {"dict": None, "pages": []})
                 dict_runs[rg]["dict"] = (soff, usz, ndict)
             elif ptype == 0:
+                pidx = len(data_pages)
+                data_pages.append((soff, usz, got_values, nvals))
                 if enc == 0:
-                    extracts.append((soff, usz, got_values, nvals))
+                    extracts.append((soff, usz, got_values, nvals, pidx))
                 else:
                     if rg not in dict_runs or dict_runs[rg]["dict"] is None:
                         raise RuntimeError("dict-coded page without dict page")
                     dict_runs[rg]["pages"].append(
-                        (soff, usz, got_values, nvals))
+                        (soff, usz, got_values, nvals, pidx))
                 got_values += nvals
         if got_values != total_values:
             raise RuntimeError(f"decoded {got_values} != {total_values} values")
@@ -269,17 +276,40 @@ class GpuParquetColumnReader:
                     ctypes.c_void_p(chunk.ptr.value + poff),
                     ctypes.c_uint64(csz)), "bg_memcpy_dtod")
 
-        # optional column => 4-byte def-level length prefix + RLE run;
-        # required column => values at offset 0  (ONE batched launch)
+        # OPTIONAL column (max_def=1): decode definition levels on device
+        # into the column validity bitmap + per-slot value indices, then
+        # extract with the mode-2 (nullable) jobs; REQUIRED column: values
+        # at offset 0  (ONE batched launch each)
         max_def = self.pf.schema.column(col).max_definition_level
+        valid = vidx = npres = None
+        mode = 0
+        if max_def > 0:
+            mode = 2
+            nwords = (total_values + 31) // 32
+            valid = ctx.upload(np.zeros(nwords, dtype=np.uint32))
+            vidx = ctx.alloc(max(4 * total_values, 4))
+            npres = ctx.alloc(max(8 * len(data_pages), 8))
+            djobs = (gpu.BgDefLevelsJob * len(data_pages))()
+            for i, (soff, usz, dst_off, nvals) in enumerate(data_pages):
+                djobs[i] = gpu.BgDefLevelsJob(
+                    page_ptr(scratch, soff).value,
+                    ctypes.c_void_p(vidx.ptr.value + 4 * dst_off).value,
+                    valid.ptr.value, usz, nvals, dst_off,
+                    ctypes.c_void_p(npres.ptr.value + 8 * i).value)
+            gpu._check(ctx.L.bg_def_levels_batch(
+                djobs, ctypes.c_int64(len(data_pages))),
+                "bg_def_levels_batch")
         if extracts:
             jobs = (gpu.BgPageExtractJob * len(extracts))()
-            for i, (soff, usz, dst_off, nvals) in enumerate(extracts):
+            for i, (soff, usz, dst_off, nvals, pidx) in enumerate(extracts):
                 jobs[i] = gpu.BgPageExtractJob(
                     page_ptr(scratch, soff).value,
                     ctypes.c_void_p(out.ptr.value + dst_off * dst_esz).value,
-                    usz, nvals, src_esz,
-                    1 if max_def > 0 else 0, 1 if flba else 0)
+                    usz, nvals, src_esz, mode, 1 if flba else 0,
+                    ctypes.c_void_p(vidx.ptr.value + 4 * dst_off).value
+                    if mode else None,
+                    ctypes.c_void_p(npres.ptr.value + 8 * pidx).value
+                    if mode else None)
             gpu._check(ctx.L.bg_page_extract_batch(
                 jobs, ctypes.c_int64(len(extracts))), "bg_page_extract_batch")
 
@@ -304,15 +334,22 @@ class GpuParquetColumnReader:
             # ONE batched index expansion over every dict-coded page
             nidx_total = sum(pg[3] for (_, pg) in all_dict_pages)
             idx = ctx.alloc(max(4 * nidx_total, 4))
+            dense = ctx.alloc(max(4 * nidx_total, 4)) if mode else None
             jobs = (gpu.BgDictIndicesJob * len(all_dict_pages))()
             run = 0
             gathers = []  # (rg, idx_off, dst_off, nvals) merged per rg below
-            for i, (rg, (soff, usz, dst_off, nvals)) in \
+            for i, (rg, (soff, usz, dst_off, nvals, pidx)) in \
                     enumerate(all_dict_pages):
                 jobs[i] = gpu.BgDictIndicesJob(
                     page_ptr(scratch, soff).value,
                     ctypes.c_void_p(idx.ptr.value + 4 * run).value,
-                    usz, nvals, 1 if max_def > 0 else 0, 0)
+                    usz, nvals, mode, 0,
+                    ctypes.c_void_p(vidx.ptr.value + 4 * dst_off).value
+                    if mode else None,
+                    ctypes.c_void_p(npres.ptr.value + 8 * pidx).value
+                    if mode else None,
+                    ctypes.c_void_p(dense.ptr.value + 4 * run).value
+                    if mode else None)
                 gathers.append((rg, run, dst_off, nvals))
                 run += nvals
             gpu._check(ctx.L.bg_dict_indices_batch(
@@ -334,4 +371,5 @@ class GpuParquetColumnReader:
                     ctypes.c_void_p(out.ptr.value + dst_off * dst_esz)),
                     "bg_gather(dict)")
         ctx.synchronize()
-        return out, total_values, phys if not flba else "DECIMAL128"
+        return (out, total_values, phys if not flba else "DECIMAL128",
+                valid)

@@ -305,8 +305,10 @@ typedef struct {
   int64_t page_len;
   int64_t nvals;
   int64_t src_esz;
-  int32_t has_def;
+  int32_t has_def;      /* 0 none, 1 validate-all-present, 2 nullable */
   int32_t flba_reverse;
+  const uint32_t* d_vidx;      /* mode 2: slot -> value index, ~0u = NULL */
+  const int64_t* d_n_present;  /* mode 2: values in this page */
 } bg_page_extract_job;
 int bg_page_extract_batch(const void* h_jobs, int64_t njobs);
 
@@ -315,10 +317,31 @@ typedef struct {
   uint32_t* d_out_idx;
   int64_t page_len;
   int64_t nvals;
-  int32_t has_def;
+  int32_t has_def;      /* 0 none, 1 validate-all-present, 2 nullable */
   int32_t _pad;
+  const uint32_t* d_vidx;      /* mode 2 */
+  const int64_t* d_n_present;  /* mode 2 */
+  uint32_t* d_dense;           /* mode 2: scratch, >= n_present u32 */
 } bg_dict_indices_job;
 int bg_dict_indices_batch(const void* h_jobs, int64_t njobs);
+
+/* OPTIONAL-column definition levels (max_def=1; the RLE/bit-packed hybrid
+ * of the parquet spec, as decoded by the reference's parquet crate
+ * rle.rs): per data page, decode [u32 len][levels] into the column's
+ * Arrow validity bitmap (atomicOr into u32 words at bit_off — page
+ * boundaries are not byte-aligned), the slot->value-index map vidx
+ * (~0u = NULL), and n_present.  Feed vidx/n_present to the mode-2
+ * extract/dict jobs above. */
+typedef struct {
+  const void* d_page;    /* page start ([u32 dlen][levels][values...]) */
+  uint32_t* d_vidx;      /* u32[nvals] slice for this page */
+  uint32_t* d_valid_out; /* column validity bitmap (u32 words, zeroed) */
+  int64_t page_len;
+  int64_t nvals;
+  int64_t bit_off;       /* absolute bit position of this page's slot 0 */
+  int64_t* d_n_present;
+} bg_def_levels_job;
+int bg_def_levels_batch(const void* h_jobs, int64_t njobs);
 
 /* Device LZ4 block compression (the GPU shuffle codec's compress half,
  * SURVEY.md §8f row 3): 64 KiB blocks, one wave per block.  d_out_slots

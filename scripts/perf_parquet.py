@@ -44,7 +44,7 @@ def main():
         t0 = time.perf_counter()
         out_bytes = 0
         for col, esz in [(0, 8), (1, 4), (2, 16)]:
-            buf, nv, _ = rd.read_column_all(col)
+            buf, nv, _, _ = rd.read_column_all(col)
             out_bytes += nv * esz
         dt = time.perf_counter() - t0
         rec = {"case": tag, "file_mb": fsize / 1e6, "rows": n,

@@ -52,27 +52,27 @@ def test_parquet_column_decode(ctx, tmp_path, compression, use_dictionary):
     rd = GpuParquetColumnReader(ctx, path)
     n = table.num_rows
 
-    buf, nv, phys = rd.read_column(0, 0)   # k64
+    buf, nv, phys, _ = rd.read_column(0, 0)   # k64
     assert nv == n and phys == "INT64"
     assert np.array_equal(buf.download(np.int64, n),
                           table.column("k64").to_numpy())
 
-    buf, nv, phys = rd.read_column(0, 1)   # d32
+    buf, nv, phys, _ = rd.read_column(0, 1)   # d32
     assert np.array_equal(buf.download(np.int32, n),
                           table.column("d32").to_numpy())
 
-    buf, nv, phys = rd.read_column(0, 2)   # f64
+    buf, nv, phys, _ = rd.read_column(0, 2)   # f64
     assert np.array_equal(buf.download(np.float64, n),
                           table.column("f64").to_numpy())
 
-    buf, nv, phys = rd.read_column(0, 3)   # decimal128 -> Arrow LE bytes
+    buf, nv, phys, _ = rd.read_column(0, 3)   # decimal128 -> Arrow LE bytes
     assert phys == "DECIMAL128"
     got = buf.download(np.uint8, 16 * n)
     want = table.column("dec").combine_chunks().buffers()[1]
     want_np = np.frombuffer(want, dtype=np.uint8, count=16 * n)
     assert np.array_equal(got, want_np)
 
-    buf, nv, phys = rd.read_column(0, 4)   # lowcard: dict-coded when enabled
+    buf, nv, phys, _ = rd.read_column(0, 4)   # lowcard: dict-coded when enabled
     assert np.array_equal(buf.download(np.int64, n),
                           table.column("lowcard").to_numpy())
 
@@ -101,10 +101,10 @@ def test_parquet_decode_feeds_q6_kernel(ctx, tmp_path):
     pq.write_table(table, path, compression="snappy", use_dictionary=False,
                    data_page_size=128 * 1024, write_statistics=False)
     rd = GpuParquetColumnReader(ctx, path)
-    sd_buf, _, _ = rd.read_column(0, 0)
-    disc_buf, _, _ = rd.read_column(0, 1)
-    qty_buf, _, _ = rd.read_column(0, 2)
-    price_buf, _, _ = rd.read_column(0, 3)
+    sd_buf, _, _, _ = rd.read_column(0, 0)
+    disc_buf, _, _, _ = rd.read_column(0, 1)
+    qty_buf, _, _, _ = rd.read_column(0, 2)
+    price_buf, _, _, _ = rd.read_column(0, 3)
     from datafusion_ballista_amd import tpch_synth
     c_sd = ctx.column(gpu.BG_DT_DATE32, sd_buf, n)
     c_d = ctx.column(gpu.BG_DT_DECIMAL128, disc_buf, n)
@@ -124,3 +124,57 @@ def test_parquet_decode_feeds_q6_kernel(ctx, tmp_path):
             int(price[i]).to_bytes(8, "little", signed=True), dtype=np.uint8)
     want_cnt, want_sum = oracle.q6(sd, d16, q16, p16, 8766, 9131, 5, 7, 2400)
     assert cnt == want_cnt and total == want_sum and cnt > 0
+
+
+@pytest.mark.parametrize("compression", ["snappy", "none"])
+@pytest.mark.parametrize("use_dictionary", [False, True])
+def test_parquet_nullable_decode(ctx, tmp_path, compression,
+                                 use_dictionary):
+    """Columns with real NULLs: definition levels decode on device to the
+    Arrow validity bitmap + value scatter; values at valid slots and the
+    bitmap itself must match pyarrow's own reader bit-for-bit."""
+    n = 150_000
+    rng = np.random.default_rng(17)
+    k64 = rng.integers(-2**60, 2**60, size=n, dtype=np.int64)
+    f64 = rng.standard_normal(n)
+    low = rng.integers(0, 40, size=n, dtype=np.int64)
+    dec_vals = [decimal.Decimal(int(v)) / 100 for v in
+                rng.integers(-10**9, 10**9, size=n)]
+    masks = {name: rng.random(n) < p for name, p in
+             [("k64", 0.2), ("f64", 0.5), ("dec", 0.03), ("low", 0.25)]}
+    table = pa.table({
+        "k64": pa.array(k64, mask=masks["k64"]),
+        "f64": pa.array(f64, mask=masks["f64"]),
+        "dec": pa.array([None if m else v
+                         for v, m in zip(dec_vals, masks["dec"])],
+                        type=pa.decimal128(15, 2)),
+        "low": pa.array(low, mask=masks["low"]),
+    })
+    path = str(tmp_path / f"null_{compression}_{use_dictionary}.parquet")
+    pq.write_table(table, path, compression=compression,
+                   use_dictionary=use_dictionary,
+                   data_page_size=32 * 1024, write_statistics=False)
+    rd = GpuParquetColumnReader(ctx, path)
+    ref = pq.read_table(path)
+
+    for ci, (name, npdt, esz) in enumerate(
+            [("k64", np.int64, 8), ("f64", np.float64, 8),
+             ("dec", None, 16), ("low", np.int64, 8)]):
+        buf, nv, phys, valid = rd.read_column_all(ci)
+        assert nv == n and valid is not None
+        vw = valid.download(np.uint32, (n + 31) // 32)
+        got_valid = np.unpackbits(vw.view(np.uint8),
+                                  bitorder="little")[:n].astype(bool)
+        want_valid = ~np.asarray(masks[name])
+        assert np.array_equal(got_valid, want_valid), name
+        col = ref.column(name).combine_chunks()
+        if name == "dec":
+            raw = buf.download(np.uint8, 16 * n).reshape(n, 16)
+            want = [v for v, m in zip(dec_vals, masks["dec"]) if not m]
+            got = [int.from_bytes(bytes(raw[i]), "little", signed=True)
+                   for i in range(n) if got_valid[i]]
+            assert got == [int(w * 100) for w in want]
+        else:
+            vals = buf.download(npdt, n)
+            want = col.drop_null().to_numpy(zero_copy_only=False)
+            assert np.array_equal(vals[got_valid], want.astype(npdt))
