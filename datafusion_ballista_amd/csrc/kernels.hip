@@ -2921,6 +2921,182 @@ extern "C" int bg_def_levels_batch(const void* h_jobs, int64_t njobs) {
 }
 
 // ---------------------------------------------------------------------------
+// BYTE_ARRAY (Utf8/Binary) PLAIN page extraction: the value section is a
+// byte-serial [u32 len][bytes] sequence (parquet spec Encodings.md PLAIN;
+// the reference decodes it in parquet/src/encodings/decoding.rs).  Lane 0
+// walks one page serially recording each slot's length and absolute source
+// address (pages decode concurrently); the column then materialises with
+// one exclusive scan over the lengths + a parallel copy
+// (bg_ba_materialize) into Arrow offsets+data form.
+// ---------------------------------------------------------------------------
+struct BaPageJob {
+  const uint8_t* page;
+  int64_t* lens_out;     // i64[nvals] slice (slot lengths; NULL slot = 0)
+  int64_t* srcaddr_out;  // i64[nvals] slice (device address of the bytes)
+  int64_t page_len;
+  int64_t nvals;
+  int32_t has_def;       // 0 none, 2 nullable
+  int32_t _pad;
+  const uint32_t* vidx;
+  const int64_t* n_present;
+};
+
+__device__ void k_ba_extract_body(const BaPageJob& job, int* err) {
+  const int lane = lane_id();
+  for (int64_t t = lane; t < job.nvals; t += BG_WAVE) {
+    job.lens_out[t] = 0;
+    job.srcaddr_out[t] = 0;
+  }
+  __builtin_amdgcn_wave_barrier();
+  if (lane != 0) return;
+  const uint8_t* p = job.page;
+  int64_t pos = 0;
+  if (job.has_def) {
+    if (job.page_len < 4) { atomicExch(err, 3); return; }
+    const uint32_t dlen = (uint32_t)p[0] | ((uint32_t)p[1] << 8) |
+                          ((uint32_t)p[2] << 16) | ((uint32_t)p[3] << 24);
+    pos = 4 + (int64_t)dlen;
+  }
+  for (int64_t s = 0; s < job.nvals; ++s) {
+    if (job.has_def == 2 && job.vidx[s] == 0xffffffffu) continue;
+    if (pos + 4 > job.page_len) { atomicExch(err, 3); return; }
+    const uint32_t len = (uint32_t)p[pos] | ((uint32_t)p[pos + 1] << 8) |
+                         ((uint32_t)p[pos + 2] << 16) |
+                         ((uint32_t)p[pos + 3] << 24);
+    pos += 4;
+    if (pos + (int64_t)len > job.page_len) { atomicExch(err, 3); return; }
+    job.lens_out[s] = (int64_t)len;
+    job.srcaddr_out[s] = (int64_t)(uintptr_t)(p + pos);
+    pos += (int64_t)len;
+  }
+}
+
+__global__ void k_ba_extract_batch(const BaPageJob* jobs, int64_t njobs,
+                                   int* err) {
+  const int64_t wave_global =
+      ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) / BG_WAVE;
+  const int64_t nwaves = ((int64_t)gridDim.x * blockDim.x) / BG_WAVE;
+  for (int64_t j = wave_global; j < njobs; j += nwaves)
+    k_ba_extract_body(jobs[j], err);
+}
+
+extern "C" int bg_ba_extract_batch(const void* h_jobs, int64_t njobs) {
+  REQUIRE_INIT();
+  BaPageJob* d_jobs;
+  int* d_err;
+  HIP_TRY(pool_malloc((void**)&d_jobs, sizeof(BaPageJob) * (njobs ? njobs : 1)));
+  HIP_TRY(pool_malloc((void**)&d_err, sizeof(int)));
+  HIP_TRY(hipMemset(d_err, 0, sizeof(int)));
+  HIP_TRY(hipMemcpy(d_jobs, h_jobs, sizeof(BaPageJob) * njobs,
+                    hipMemcpyHostToDevice));
+  const int waves_per_block = BG_BLOCK / BG_WAVE;
+  int blocks = (int)bg_imin64((njobs + waves_per_block - 1) / waves_per_block,
+                              BG_MAX_BLOCKS);
+  if (blocks == 0) blocks = 1;
+  hipLaunchKernelGGL(k_ba_extract_batch, dim3(blocks), dim3(BG_BLOCK), 0, 0,
+                     d_jobs, njobs, d_err);
+  HIP_TRY(hipGetLastError());
+  int err = 0;
+  HIP_TRY(hipMemcpy(&err, d_err, sizeof(int), hipMemcpyDeviceToHost));
+  (void)pool_release(d_jobs);
+  (void)pool_release(d_err);
+  if (err)
+    return set_err(BG_ERR_INVALID, "bg_ba_extract_batch: malformed page");
+  return BG_OK;
+}
+
+// dict-coded BYTE_ARRAY pages: slot lengths/addresses come from the
+// PLAIN-decoded dictionary through the expanded indices (NULL slots -> 0)
+__global__ void k_ba_from_dict(const uint32_t* idx, const int32_t* doffs,
+                               const uint8_t* ddata, const uint32_t* vidx,
+                               int64_t n, int64_t* lens_out,
+                               int64_t* srcaddr_out) {
+  for (int64_t s = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; s < n;
+       s += (int64_t)gridDim.x * blockDim.x) {
+    if (vidx && vidx[s] == 0xffffffffu) {
+      lens_out[s] = 0;
+      srcaddr_out[s] = 0;
+      continue;
+    }
+    const uint32_t ix = idx[s];
+    const int32_t o = doffs[ix];
+    lens_out[s] = (int64_t)(doffs[ix + 1] - o);
+    srcaddr_out[s] = (int64_t)(uintptr_t)(ddata + o);
+  }
+}
+
+extern "C" int bg_ba_from_dict(const uint32_t* d_idx, const int32_t* d_doffs,
+                               const void* d_ddata, const uint32_t* d_vidx,
+                               int64_t n, int64_t* d_lens_out,
+                               int64_t* d_srcaddr_out) {
+  REQUIRE_INIT();
+  int blocks = (int)bg_imin64((n + BG_BLOCK - 1) / BG_BLOCK, BG_MAX_BLOCKS);
+  if (blocks == 0) blocks = 1;
+  hipLaunchKernelGGL(k_ba_from_dict, dim3(blocks), dim3(BG_BLOCK), 0, 0,
+                     d_idx, d_doffs, (const uint8_t*)d_ddata, d_vidx, n,
+                     d_lens_out, d_srcaddr_out);
+  HIP_TRY(hipGetLastError());
+  return BG_OK;
+}
+
+// scan lens -> Arrow i32 offsets (+ total) and copy every slot's bytes
+__global__ void k_ba_copy(const int64_t* srcaddr, const int64_t* offs64,
+                          int32_t* offs32, uint8_t* data, int64_t n) {
+  for (int64_t s = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; s < n;
+       s += (int64_t)gridDim.x * blockDim.x) {
+    const int64_t o = offs64[s];
+    offs32[s] = (int32_t)o;
+    const int64_t len = offs64[s + 1] - o;
+    const uint8_t* src = (const uint8_t*)(uintptr_t)srcaddr[s];
+    for (int64_t b = 0; b < len; ++b) data[o + b] = src[b];
+  }
+}
+
+extern "C" int bg_ba_materialize(const int64_t* d_lens,
+                                 const int64_t* d_srcaddr, int64_t n,
+                                 int32_t* d_offs32 /* n+1 */,
+                                 uint8_t* d_data /* total bytes */,
+                                 int64_t data_cap, int64_t* out_total) {
+  REQUIRE_INIT();
+  i64* d_offs64;
+  i64* d_total;
+  HIP_TRY(pool_malloc((void**)&d_offs64, sizeof(i64) * (n + 1)));
+  HIP_TRY(pool_malloc((void**)&d_total, sizeof(i64)));
+  {
+    int rc = scan_exclusive_i64((const u64*)d_lens, n, d_offs64, d_total);
+    if (rc != BG_OK) return rc;
+  }
+  i64 total = 0;
+  HIP_TRY(hipMemcpy(&total, d_total, sizeof(i64), hipMemcpyDeviceToHost));
+  if (d_data == nullptr) {  // dry-run: size query only
+    (void)pool_release(d_offs64);
+    (void)pool_release(d_total);
+    *out_total = total;
+    return BG_OK;
+  }
+  if (total > data_cap || total > 0x7fffffffLL) {
+    (void)pool_release(d_offs64);
+    (void)pool_release(d_total);
+    return set_err(BG_ERR_INVALID, "bg_ba_materialize: data_cap/2GiB bound");
+  }
+  HIP_TRY(hipMemcpy(d_offs64 + n, &total, sizeof(i64),
+                    hipMemcpyHostToDevice));
+  int blocks = (int)bg_imin64((n + BG_BLOCK - 1) / BG_BLOCK, BG_MAX_BLOCKS);
+  if (blocks == 0) blocks = 1;
+  hipLaunchKernelGGL(k_ba_copy, dim3(blocks), dim3(BG_BLOCK), 0, 0,
+                     d_srcaddr, (const int64_t*)d_offs64, d_offs32, d_data,
+                     n);
+  HIP_TRY(hipGetLastError());
+  const int32_t t32 = (int32_t)total;
+  HIP_TRY(hipMemcpy(d_offs32 + n, &t32, sizeof(int32_t),
+                    hipMemcpyHostToDevice));
+  (void)pool_release(d_offs64);
+  (void)pool_release(d_total);
+  *out_total = total;
+  return BG_OK;
+}
+
+// ---------------------------------------------------------------------------
 // Batched page ops: the per-page host loop (launch + sync error check per
 // page) starves the GPU when chunks hold few pages — these take the whole
 // column's page list in one call.

@@ -519,6 +519,16 @@ class BgDictIndicesJob(ctypes.Structure):
                 ("d_dense", ctypes.c_void_p)]
 
 
+class BgBaPageJob(ctypes.Structure):
+    _fields_ = [("d_page", ctypes.c_void_p),
+                ("d_lens_out", ctypes.c_void_p),
+                ("d_srcaddr_out", ctypes.c_void_p),
+                ("page_len", ctypes.c_int64), ("nvals", ctypes.c_int64),
+                ("has_def", ctypes.c_int32), ("_pad", ctypes.c_int32),
+                ("d_vidx", ctypes.c_void_p),
+                ("d_n_present", ctypes.c_void_p)]
+
+
 class BgDefLevelsJob(ctypes.Structure):
     _fields_ = [("d_page", ctypes.c_void_p), ("d_vidx", ctypes.c_void_p),
                 ("d_valid_out", ctypes.c_void_p),

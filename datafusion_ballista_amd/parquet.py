@@ -175,6 +175,9 @@ class GpuParquetColumnReader:
         if phys in _PHYS_NP:
             src_esz = dst_esz = _PHYS_NP[phys][1]
             flba = False
+        elif phys == "BYTE_ARRAY":
+            src_esz = dst_esz = 0  # variable width: lens+addr staging
+            flba = False
         elif phys == "FIXED_LEN_BYTE_ARRAY":
             # parquet stores decimals at the minimal width for the
             # precision (FLBA(7) for Decimal(15,2)), big-endian; the device
@@ -227,7 +230,13 @@ class GpuParquetColumnReader:
         def page_ptr(base, off):
             return ctypes.c_void_p(base.ptr.value + off)
 
-        out = ctx.alloc(max(total_values * dst_esz, dst_esz))
+        ba = phys == "BYTE_ARRAY"
+        if ba:
+            ba_lens = ctx.alloc(max(8 * total_values, 8))
+            ba_srcaddr = ctx.alloc(max(8 * total_values, 8))
+            out = None
+        else:
+            out = ctx.alloc(max(total_values * dst_esz, dst_esz))
         extracts = []        # PLAIN: (scratch_off, usz, dst_off, nvals, pidx)
         dict_runs = {}       # rg -> {"dict": (soff, usz, ndict), "pages": []}
         got_values = 0
@@ -299,7 +308,22 @@ class GpuParquetColumnReader:
             gpu._check(ctx.L.bg_def_levels_batch(
                 djobs, ctypes.c_int64(len(data_pages))),
                 "bg_def_levels_batch")
-        if extracts:
+        if extracts and ba:
+            jobs = (gpu.BgBaPageJob * len(extracts))()
+            for i, (soff, usz, dst_off, nvals, pidx) in enumerate(extracts):
+                jobs[i] = gpu.BgBaPageJob(
+                    page_ptr(scratch, soff).value,
+                    ctypes.c_void_p(ba_lens.ptr.value + 8 * dst_off).value,
+                    ctypes.c_void_p(ba_srcaddr.ptr.value
+                                    + 8 * dst_off).value,
+                    usz, nvals, mode, 0,
+                    ctypes.c_void_p(vidx.ptr.value + 4 * dst_off).value
+                    if mode else None,
+                    ctypes.c_void_p(npres.ptr.value + 8 * pidx).value
+                    if mode else None)
+            gpu._check(ctx.L.bg_ba_extract_batch(
+                jobs, ctypes.c_int64(len(extracts))), "bg_ba_extract_batch")
+        elif extracts:
             jobs = (gpu.BgPageExtractJob * len(extracts))()
             for i, (soff, usz, dst_off, nvals, pidx) in enumerate(extracts):
                 jobs[i] = gpu.BgPageExtractJob(
@@ -318,19 +342,47 @@ class GpuParquetColumnReader:
         if all_dict_pages:
             # PLAIN-decode every row group's dictionary (batched)
             dict_bufs = {}
-            djobs = []
-            for rg, d in dict_runs.items():
-                if not d["pages"]:
-                    continue
-                dsoff, dusz, ndict = d["dict"]
-                dbuf = ctx.alloc(max(ndict * dst_esz, dst_esz))
-                dict_bufs[rg] = dbuf
-                djobs.append(gpu.BgPageExtractJob(
-                    page_ptr(scratch, dsoff).value, dbuf.ptr.value, dusz,
-                    ndict, src_esz, 0, 1 if flba else 0))
-            jarr = (gpu.BgPageExtractJob * len(djobs))(*djobs)
-            gpu._check(ctx.L.bg_page_extract_batch(
-                jarr, ctypes.c_int64(len(djobs))), "bg_page_extract(dicts)")
+            if ba:
+                # string dictionaries: lens+addr walk, then one
+                # materialize per rg into (i32 offsets, bytes)
+                for rg, d in dict_runs.items():
+                    if not d["pages"]:
+                        continue
+                    dsoff, dusz, ndict = d["dict"]
+                    dl = ctx.alloc(max(8 * ndict, 8))
+                    da = ctx.alloc(max(8 * ndict, 8))
+                    j1 = (gpu.BgBaPageJob * 1)(gpu.BgBaPageJob(
+                        page_ptr(scratch, dsoff).value, dl.ptr.value,
+                        da.ptr.value, dusz, ndict, 0, 0, None, None))
+                    gpu._check(ctx.L.bg_ba_extract_batch(
+                        j1, ctypes.c_int64(1)), "bg_ba_extract(dict)")
+                    doffs = ctx.alloc(max(4 * (ndict + 1), 8))
+                    tot = ctypes.c_int64()
+                    gpu._check(ctx.L.bg_ba_materialize(
+                        dl.ptr, da.ptr, ctypes.c_int64(ndict), doffs.ptr,
+                        None, ctypes.c_int64(0), ctypes.byref(tot)),
+                        "bg_ba_materialize(dict size)")
+                    ddata = ctx.alloc(max(tot.value, 1))
+                    gpu._check(ctx.L.bg_ba_materialize(
+                        dl.ptr, da.ptr, ctypes.c_int64(ndict), doffs.ptr,
+                        ddata.ptr, ctypes.c_int64(tot.value),
+                        ctypes.byref(tot)), "bg_ba_materialize(dict)")
+                    dict_bufs[rg] = (doffs, ddata)
+            else:
+                djobs = []
+                for rg, d in dict_runs.items():
+                    if not d["pages"]:
+                        continue
+                    dsoff, dusz, ndict = d["dict"]
+                    dbuf = ctx.alloc(max(ndict * dst_esz, dst_esz))
+                    dict_bufs[rg] = dbuf
+                    djobs.append(gpu.BgPageExtractJob(
+                        page_ptr(scratch, dsoff).value, dbuf.ptr.value, dusz,
+                        ndict, src_esz, 0, 1 if flba else 0))
+                jarr = (gpu.BgPageExtractJob * len(djobs))(*djobs)
+                gpu._check(ctx.L.bg_page_extract_batch(
+                    jarr, ctypes.c_int64(len(djobs))),
+                    "bg_page_extract(dicts)")
             # ONE batched index expansion over every dict-coded page
             nidx_total = sum(pg[3] for (_, pg) in all_dict_pages)
             idx = ctx.alloc(max(4 * nidx_total, 4))
@@ -364,12 +416,41 @@ class GpuParquetColumnReader:
                 else:
                     merged.append([rg, ioff, dst_off, nvals])
             for rg, ioff, dst_off, nvals in merged:
-                gpu._check(ctx.L.bg_gather(
-                    dict_bufs[rg].ptr, ctypes.c_int64(dst_esz),
-                    ctypes.c_void_p(idx.ptr.value + 4 * ioff),
-                    ctypes.c_int64(nvals),
-                    ctypes.c_void_p(out.ptr.value + dst_off * dst_esz)),
-                    "bg_gather(dict)")
+                if ba:
+                    doffs, ddata = dict_bufs[rg]
+                    gpu._check(ctx.L.bg_ba_from_dict(
+                        ctypes.c_void_p(idx.ptr.value + 4 * ioff), doffs.ptr,
+                        ddata.ptr,
+                        ctypes.c_void_p(vidx.ptr.value + 4 * dst_off)
+                        if mode else None,
+                        ctypes.c_int64(nvals),
+                        ctypes.c_void_p(ba_lens.ptr.value + 8 * dst_off),
+                        ctypes.c_void_p(ba_srcaddr.ptr.value
+                                        + 8 * dst_off)),
+                        "bg_ba_from_dict")
+                else:
+                    gpu._check(ctx.L.bg_gather(
+                        dict_bufs[rg].ptr, ctypes.c_int64(dst_esz),
+                        ctypes.c_void_p(idx.ptr.value + 4 * ioff),
+                        ctypes.c_int64(nvals),
+                        ctypes.c_void_p(out.ptr.value + dst_off * dst_esz)),
+                        "bg_gather(dict)")
+        if ba:
+            offs32 = ctx.alloc(max(4 * (total_values + 1), 8))
+            tot = ctypes.c_int64()
+            gpu._check(ctx.L.bg_ba_materialize(
+                ba_lens.ptr, ba_srcaddr.ptr, ctypes.c_int64(total_values),
+                offs32.ptr, None, ctypes.c_int64(0),
+                ctypes.byref(tot)), "bg_ba_materialize(size)")
+            cap = tot.value
+            data = ctx.alloc(max(cap, 1))
+            gpu._check(ctx.L.bg_ba_materialize(
+                ba_lens.ptr, ba_srcaddr.ptr, ctypes.c_int64(total_values),
+                offs32.ptr, data.ptr, ctypes.c_int64(cap),
+                ctypes.byref(tot)), "bg_ba_materialize")
+            ctx.synchronize()
+            return ((offs32, data, tot.value), total_values, "BYTE_ARRAY",
+                    valid)
         ctx.synchronize()
         return (out, total_values, phys if not flba else "DECIMAL128",
                 valid)

@@ -343,6 +343,32 @@ typedef struct {
 } bg_def_levels_job;
 int bg_def_levels_batch(const void* h_jobs, int64_t njobs);
 
+/* BYTE_ARRAY (Utf8/Binary) PLAIN pages (parquet spec PLAIN: [u32 len]
+ * [bytes] per value; the reference's decoding.rs PlainDecoder):
+ * per page, record each slot's length + absolute device source address
+ * (NULL slots length 0); then ONE column-wide bg_ba_materialize scans the
+ * lengths into Arrow i32 offsets and copies every slot's bytes.  Dict-
+ * coded string pages bridge with bg_ba_from_dict (lengths/addresses from
+ * the PLAIN-decoded dictionary through the expanded indices). */
+typedef struct {
+  const void* d_page;
+  int64_t* d_lens_out;     /* i64[nvals] slice */
+  int64_t* d_srcaddr_out;  /* i64[nvals] slice */
+  int64_t page_len;
+  int64_t nvals;
+  int32_t has_def; /* 0 none, 2 nullable */
+  int32_t _pad;
+  const uint32_t* d_vidx;
+  const int64_t* d_n_present;
+} bg_ba_page_job;
+int bg_ba_extract_batch(const void* h_jobs, int64_t njobs);
+int bg_ba_from_dict(const uint32_t* d_idx, const int32_t* d_doffs,
+                    const void* d_ddata, const uint32_t* d_vidx, int64_t n,
+                    int64_t* d_lens_out, int64_t* d_srcaddr_out);
+int bg_ba_materialize(const int64_t* d_lens, const int64_t* d_srcaddr,
+                      int64_t n, int32_t* d_offs32 /* n+1 */,
+                      uint8_t* d_data, int64_t data_cap, int64_t* out_total);
+
 /* Device LZ4 block compression (the GPU shuffle codec's compress half,
  * SURVEY.md §8f row 3): 64 KiB blocks, one wave per block.  d_out_slots
  * holds nblocks slots of 65544 B; h_block_sizes[i] = compressed size, or
