@@ -131,6 +131,8 @@ def parse_page_header(buf, pos):
         3: "i32",   # compressed_page_size
         5: ("struct", {1: "i32", 2: "i32", 3: "i32", 4: "i32"}),  # v1 header
         7: ("struct", {1: "i32", 2: "i32"}),  # dictionary page header
+        8: ("struct", {1: "i32", 2: "i32", 3: "i32", 4: "i32", 5: "i32",
+                       6: "i32", 7: "bool"}),  # DataPageHeaderV2
     })
     return h, r.i
 
@@ -190,7 +192,12 @@ class GpuParquetColumnReader:
 
         import ctypes
         chunk = self._file_buf  # page offsets below are absolute file offsets
-        headers = []  # (rg, ptype, file_off, csz, usz, nvals, enc, ndict, soff)
+        max_def0 = self.pf.schema.column(col).max_definition_level
+        headers = []  # (rg, ptype, file_off, csz, usz, nvals, enc, ndict,
+        #               soff, v2) with v2 = None (V1) or
+        #               (def_len, values_csz, values_usz, is_compressed);
+        #               for V2, usz is the SLOT length (the V1-shaped
+        #               [u32 dlen][levels][values] image synthesized below)
         scratch_total = 0
         total_values = 0
         for rg in rgs:
@@ -208,9 +215,8 @@ class GpuParquetColumnReader:
                 ptype = h.get(1, 0)
                 usz = h[2]
                 csz = h[3]
-                if ptype == 3:
-                    raise RuntimeError("DATA_PAGE_V2: not GPU-decodable yet")
                 nvals = enc = ndict = 0
+                v2 = None
                 if ptype == 2:
                     ndict = h.get(7, {}).get(1, 0)
                 elif ptype == 0:
@@ -220,8 +226,32 @@ class GpuParquetColumnReader:
                     if enc not in (0, 2, 8):
                         raise RuntimeError(
                             f"encoding {enc}: not GPU-decodable yet")
+                elif ptype == 3:
+                    # DataPageV2 (PageHeader field 8): levels sit ahead of
+                    # the (optionally compressed) values, uncompressed
+                    dph = h.get(8, {})
+                    nvals = dph.get(1, 0)
+                    enc = dph.get(4, 0)
+                    def_len = dph.get(5, 0)
+                    rep_len = dph.get(6, 0)
+                    is_comp = dph.get(7, 1)
+                    if rep_len:
+                        raise RuntimeError(
+                            "DATA_PAGE_V2 repetition levels (nested): "
+                            "not GPU-decodable yet")
+                    if enc not in (0, 2, 8):
+                        raise RuntimeError(
+                            f"encoding {enc}: not GPU-decodable yet")
+                    values_csz = csz - def_len
+                    values_usz = usz - def_len
+                    # synthesize the V1 page shape so every downstream
+                    # kernel (def walk, extract, dict) is unchanged
+                    slot_len = (4 + def_len + values_usz) if max_def0 > 0 \
+                        else values_usz
+                    v2 = (def_len, values_csz, values_usz, is_comp)
+                    usz = slot_len
                 headers.append((rg, ptype, data_pos, csz, usz, nvals, enc,
-                                ndict, scratch_total))
+                                ndict, scratch_total, v2))
                 scratch_total += (usz + 255) & ~255
                 pos = data_pos + csz
 
@@ -242,13 +272,26 @@ class GpuParquetColumnReader:
         got_values = 0
         snappy_jobs = []
         data_pages = []      # (scratch_off, usz, dst_off, nvals) in order
-        for (rg, ptype, poff, csz, usz, nvals, enc, ndict, soff) in headers:
-            if codec == "SNAPPY":
+        v2_fixups = []  # (soff, file_def_off, def_len, dst_voff)
+        for (rg, ptype, poff, csz, usz, nvals, enc, ndict, soff,
+             v2) in headers:
+            if v2 is not None:
+                def_len, values_csz, values_usz, is_comp = v2
+                voff = (4 + def_len) if max_def0 > 0 else 0
+                if max_def0 > 0:
+                    v2_fixups.append((soff, poff, def_len))
+                if codec == "SNAPPY" and is_comp:
+                    snappy_jobs.append((poff + def_len, values_csz,
+                                        soff + voff, values_usz))
+                else:
+                    v2_fixups.append((soff + voff, poff + def_len,
+                                      -values_usz))  # raw value copy
+            elif codec == "SNAPPY":
                 snappy_jobs.append((poff, csz, soff, usz))
             if ptype == 2:
                 dict_runs.setdefault(rg, {"dict": None, "pages": []})
                 dict_runs[rg]["dict"] = (soff, usz, ndict)
-            elif ptype == 0:
+            elif ptype in (0, 3):
                 pidx = len(data_pages)
                 data_pages.append((soff, usz, got_values, nvals))
                 if enc == 0:
@@ -261,6 +304,32 @@ class GpuParquetColumnReader:
                 got_values += nvals
         if got_values != total_values:
             raise RuntimeError(f"decoded {got_values} != {total_values} values")
+        if v2_fixups:
+            # synthesize the V1 image pieces: [u32 dlen] prefix + the
+            # uncompressed level bytes (and raw values when the page body
+            # is not compressed), copied device-to-device from the
+            # resident file buffer via a tiny staged prefix table
+            prefixes = np.array([d for (_, _, d) in v2_fixups if d >= 0],
+                                dtype=np.uint32)
+            pbuf = ctx.upload(prefixes) if len(prefixes) else None
+            pi = 0
+            for (soff, foff, dlen) in v2_fixups:
+                if dlen < 0:  # uncompressed values copy
+                    gpu._check(ctx.L.bg_memcpy_dtod(
+                        ctypes.c_void_p(scratch.ptr.value + soff),
+                        ctypes.c_void_p(chunk.ptr.value + foff),
+                        ctypes.c_uint64(-dlen)), "v2 values copy")
+                    continue
+                gpu._check(ctx.L.bg_memcpy_dtod(
+                    ctypes.c_void_p(scratch.ptr.value + soff),
+                    ctypes.c_void_p(pbuf.ptr.value + 4 * pi),
+                    ctypes.c_uint64(4)), "v2 dlen prefix")
+                pi += 1
+                if dlen:
+                    gpu._check(ctx.L.bg_memcpy_dtod(
+                        ctypes.c_void_p(scratch.ptr.value + soff + 4),
+                        ctypes.c_void_p(chunk.ptr.value + foff),
+                        ctypes.c_uint64(dlen)), "v2 def levels copy")
 
         if codec == "SNAPPY":
             arr = (gpu.BgSnappyPage * len(snappy_jobs))()
@@ -279,7 +348,10 @@ class GpuParquetColumnReader:
         else:
             # uncompressed: device-to-device copy page payloads into the
             # aligned scratch slots
-            for (rg, ptype, poff, csz, usz, nvals, enc, ndict, soff) in headers:
+            for (rg, ptype, poff, csz, usz, nvals, enc, ndict, soff,
+                 v2) in headers:
+                if v2 is not None:
+                    continue  # synthesized above
                 gpu._check(ctx.L.bg_memcpy_dtod(
                     ctypes.c_void_p(scratch.ptr.value + soff),
                     ctypes.c_void_p(chunk.ptr.value + poff),
@@ -289,7 +361,7 @@ class GpuParquetColumnReader:
         # into the column validity bitmap + per-slot value indices, then
         # extract with the mode-2 (nullable) jobs; REQUIRED column: values
         # at offset 0  (ONE batched launch each)
-        max_def = self.pf.schema.column(col).max_definition_level
+        max_def = max_def0
         valid = vidx = npres = None
         mode = 0
         if max_def > 0:

@@ -128,8 +128,9 @@ def test_parquet_decode_feeds_q6_kernel(ctx, tmp_path):
 
 @pytest.mark.parametrize("compression", ["snappy", "none"])
 @pytest.mark.parametrize("use_dictionary", [False, True])
+@pytest.mark.parametrize("version", ["1.0", "2.0"])
 def test_parquet_nullable_decode(ctx, tmp_path, compression,
-                                 use_dictionary):
+                                 use_dictionary, version):
     """Columns with real NULLs: definition levels decode on device to the
     Arrow validity bitmap + value scatter; values at valid slots and the
     bitmap itself must match pyarrow's own reader bit-for-bit."""
@@ -150,9 +151,10 @@ def test_parquet_nullable_decode(ctx, tmp_path, compression,
                         type=pa.decimal128(15, 2)),
         "low": pa.array(low, mask=masks["low"]),
     })
-    path = str(tmp_path / f"null_{compression}_{use_dictionary}.parquet")
+    path = str(tmp_path /
+               f"null_{compression}_{use_dictionary}_{version}.parquet")
     pq.write_table(table, path, compression=compression,
-                   use_dictionary=use_dictionary,
+                   use_dictionary=use_dictionary, data_page_version=version,
                    data_page_size=32 * 1024, write_statistics=False)
     rd = GpuParquetColumnReader(ctx, path)
     ref = pq.read_table(path)
@@ -183,8 +185,9 @@ def test_parquet_nullable_decode(ctx, tmp_path, compression,
 @pytest.mark.parametrize("compression", ["snappy", "none"])
 @pytest.mark.parametrize("use_dictionary", [False, True])
 @pytest.mark.parametrize("with_nulls", [False, True])
+@pytest.mark.parametrize("version", ["1.0", "2.0"])
 def test_parquet_byte_array_decode(ctx, tmp_path, compression,
-                                   use_dictionary, with_nulls):
+                                   use_dictionary, with_nulls, version):
     """BYTE_ARRAY (Utf8) columns: PLAIN [u32 len][bytes] pages and dict-
     coded pages decode on device to Arrow offsets+data; strings at valid
     slots must match pyarrow's reader exactly."""
@@ -199,9 +202,9 @@ def test_parquet_byte_array_decode(ctx, tmp_path, compression,
                    type=pa.string())
     table = pa.table({"s": arr})
     path = str(tmp_path / f"ba_{compression}_{use_dictionary}_{with_nulls}"
-               ".parquet")
+               f"_{version}.parquet")
     pq.write_table(table, path, compression=compression,
-                   use_dictionary=use_dictionary,
+                   use_dictionary=use_dictionary, data_page_version=version,
                    data_page_size=16 * 1024, write_statistics=False)
     rd = GpuParquetColumnReader(ctx, path)
     (offs_buf, data_buf, total), nv, phys, valid = rd.read_column_all(0)
