@@ -17,6 +17,7 @@ Device work (hash, partition split, gather) runs through libballista_gpu.so;
 IPC encode + file write stay host-side.
 """
 
+import ctypes as _ct
 import time
 from dataclasses import dataclass
 
@@ -67,9 +68,8 @@ def _np_for(t: pa.DataType):
 
 
 def _col_raw(arr: pa.Array) -> np.ndarray:
-    """Arrow array -> contiguous primitive numpy view (no nulls yet)."""
-    if arr.null_count:
-        raise RuntimeError("null-carrying payload columns: next round")
+    """Arrow array -> contiguous primitive numpy view (values buffer only;
+    validity rides separately — see _repartition_one_input)."""
     t = arr.type
     if pa.types.is_decimal128(t):
         buf = arr.buffers()[1]
@@ -82,8 +82,10 @@ def _col_raw(arr: pa.Array) -> np.ndarray:
                          np.dtype(np_t).itemsize)
 
 
-def _array_from_raw(t: pa.DataType, raw: np.ndarray, n: int) -> pa.Array:
-    return pa.Array.from_buffers(t, n, [None, pa.py_buffer(raw.tobytes())])
+def _array_from_raw(t: pa.DataType, raw: np.ndarray, n: int,
+                    valid_bits: bytes = None) -> pa.Array:
+    vb = pa.py_buffer(valid_bits) if valid_bits is not None else None
+    return pa.Array.from_buffers(t, n, [vb, pa.py_buffer(raw.tobytes())])
 
 
 class GpuQueryStageExecutor:
@@ -169,15 +171,21 @@ class GpuQueryStageExecutor:
 
         cols = []           # BgColumn per table column (fixed OR utf8)
         utf8_src = {}       # col index -> (data buf, offsets buf, total)
+        null_ids = {}       # col index -> host bool validity (null cols)
         for i in range(table.num_columns):
             arr = table.column(i).combine_chunks()
             if isinstance(arr, pa.ChunkedArray):
                 arr = arr.chunk(0) if arr.num_chunks else pa.array(
                     [], type=arr.type)
             t = arr.type
+            vbuf = None
+            if arr.null_count:
+                # Arrow LSB validity (rebased to offset 0); rows with NULL
+                # keys hash as no-contribution (hash_utils create_hashes)
+                vbool = np.asarray(arr.is_valid())
+                vbuf = ctx.upload(np.packbits(vbool, bitorder="little"))
+                null_ids[i] = vbool
             if pa.types.is_string(t):
-                if arr.null_count:
-                    raise RuntimeError("null strings: next round")
                 offs = np.frombuffer(arr.buffers()[1], dtype=np.int32,
                                      count=n + 1 + arr.offset)[arr.offset:]
                 data_buf = arr.buffers()[2]
@@ -189,12 +197,13 @@ class GpuQueryStageExecutor:
                 dbuf = ctx.upload(data_np)
                 obuf = ctx.upload(offs_rb)
                 utf8_src[i] = (dbuf, obuf, nbytes)
-                cols.append(ctx.column(gpu.BG_DT_UTF8, dbuf, n, offsets=obuf))
+                cols.append(ctx.column(gpu.BG_DT_UTF8, dbuf, n,
+                                       offsets=obuf, validity=vbuf))
             else:
                 raw = _col_raw(arr)
                 bgdt = _bg_dtype(t)
                 buf = ctx.upload(raw)
-                cols.append(ctx.column(bgdt, buf, n))
+                cols.append(ctx.column(bgdt, buf, n, validity=vbuf))
 
         key_cols = [cols[i] for i in self.key_columns]
         fixed_ids = [i for i in range(table.num_columns) if i not in utf8_src]
@@ -208,6 +217,13 @@ class GpuQueryStageExecutor:
             oo, od, tot = ctx.gather_varlen(dbuf, obuf, idx_buf, n,
                                             max_bytes=max(nbytes, 1))
             utf8_out[i] = (oo, od, tot)
+        valid_out = {}
+        for i in null_ids:
+            ob = ctx.alloc(max(((n + 63) // 64) * 8, 8))
+            gpu._check(ctx.L.bg_gather_bits(
+                _ct.c_void_p(cols[i].d_validity), idx_buf.ptr,
+                _ct.c_int64(n), ob.ptr), "bg_gather_bits")
+            valid_out[i] = ob
         ctx.synchronize()
         dt_device = time.perf_counter() - t_repart
         offsets = offs_buf.download(np.int64, self.k + 1)
@@ -221,10 +237,18 @@ class GpuQueryStageExecutor:
         for i, (oo, od, tot) in utf8_out.items():
             utf8_raws[i] = (oo.download(np.int32, n + 1),
                             od.download(np.uint8, max(tot, 1))[:tot])
+        valid_perm = {}
+        for i, ob in valid_out.items():
+            words = ob.download(np.uint8, ((n + 63) // 64) * 8)
+            valid_perm[i] = np.unpackbits(
+                words, bitorder="little")[:n].astype(bool)
 
         if self.gpu_codec:
             if utf8_src:
                 raise RuntimeError("gpu_codec: fixed-width columns only (r1)")
+            if null_ids:
+                raise RuntimeError(
+                    "gpu_codec: null columns take the host codec path (r1)")
             return self._encode_partitions_gpu(
                 schema, offsets, cols, out_bufs), dt_device
 
@@ -238,6 +262,12 @@ class GpuQueryStageExecutor:
             arrays = []
             for ci in range(table.num_columns):
                 t = schema.types[ci]
+                vb = None
+                if ci in valid_perm:
+                    part_bool = valid_perm[ci][lo:hi]
+                    if not part_bool.all():
+                        vb = np.packbits(part_bool,
+                                         bitorder="little").tobytes()
                 if ci in utf8_raws:
                     offs_all, data_all = utf8_raws[ci]
                     o_lo, o_hi = int(offs_all[lo]), int(offs_all[hi])
@@ -245,13 +275,14 @@ class GpuQueryStageExecutor:
                     sub_data = data_all[o_lo:o_hi]
                     arrays.append(pa.Array.from_buffers(
                         pa.utf8(), m,
-                        [None, pa.py_buffer(sub_offs.tobytes()),
+                        [pa.py_buffer(vb) if vb is not None else None,
+                         pa.py_buffer(sub_offs.tobytes()),
                          pa.py_buffer(sub_data.tobytes())]))
                     continue
                 esz = 16 if pa.types.is_decimal128(t) else \
                     np.dtype(_np_for(t)).itemsize
                 raw = col_raws[ci][lo * esz: hi * esz]
-                arrays.append(_array_from_raw(t, raw, m))
+                arrays.append(_array_from_raw(t, raw, m, vb))
             part_table = pa.Table.from_arrays(arrays, schema=schema)
             batches = shuffle.rechunk(part_table, self.batch_size)
             streams.append(shuffle.encode_partition_stream(batches, schema))
