@@ -83,13 +83,70 @@ def fuzz_round(ctx, seed):
     want_groups = {kk[0]: v for kk, v in want_groups.items()}
     assert got_groups == want_groups, f"hashagg mismatch seed={seed}"
 
-    # stable sort (asc/desc) vs numpy stable argsort
+    # stable sort (asc/desc) vs numpy stable argsort — both full-range
+    # keys and a narrow range that exercises constant-digit pass skipping
     desc = bool(rng.integers(0, 2))
     perm = ctx.sort_rows([col], [desc], n).download(np.uint32, n)
     u = keys.view(np.uint64) ^ np.uint64(1 << 63)
     want_perm = np.argsort(~u if desc else u, kind="stable")
     assert np.array_equal(perm.astype(np.int64), want_perm), \
         f"sort mismatch seed={seed} desc={desc}"
+    ncol, _ = ctx.upload_column(probe, gpu.BG_DT_INT64)  # narrow range
+    perm2 = ctx.sort_rows([ncol], [desc], n).download(np.uint32, n)
+    u2 = probe.view(np.uint64) ^ np.uint64(1 << 63)
+    want2 = np.argsort(~u2 if desc else u2, kind="stable")
+    assert np.array_equal(perm2.astype(np.int64), want2), \
+        f"narrow sort mismatch seed={seed} desc={desc}"
+
+    # NULL semantics: hash with validity; join with null keys both sides
+    vbits = rng.integers(0, 256, size=(n + 7) // 8, dtype=np.uint8)
+    colv, _ = ctx.upload_column(keys, gpu.BG_DT_INT64, validity=vbits)
+    hv = ctx.hash_columns([colv], n)
+    ctx.synchronize()
+    want_hv = oracle.hash_columns([("i64", keys, vbits)], n)
+    assert np.array_equal(hv.download(np.uint64, n), want_hv), \
+        f"null hash mismatch seed={seed}"
+    bbits = rng.integers(0, 256, size=(nb + 7) // 8, dtype=np.uint8)
+    bcolv, _ = ctx.upload_column(build, gpu.BG_DT_INT64, validity=bbits)
+    pcolv, _ = ctx.upload_column(probe, gpu.BG_DT_INT64, validity=vbits)
+    jn = gpu.GpuHashJoin(ctx, bcolv, nb)
+    jp2, jb2, jm2 = jn.probe(pcolv, n)
+    bmask = np.unpackbits(bbits, bitorder="little")[:nb].astype(bool)
+    pmask = np.unpackbits(vbits, bitorder="little")[:n].astype(bool)
+    bi = np.where(bmask)[0]
+    pi = np.where(pmask)[0]
+    cwp, cwb = oracle.hashjoin_pairs(build[bi], probe[pi])
+    wp2 = pi[cwp.astype(np.int64)].astype(np.uint64)
+    wb2 = bi[cwb.astype(np.int64)].astype(np.uint64)
+    assert jm2 == len(wp2), f"null join count mismatch seed={seed}"
+    if jm2:
+        got2 = np.sort(jp2.download(np.uint32, jm2).astype(np.uint64)
+                       << np.uint64(32) | jb2.download(np.uint32, jm2))
+        want2j = np.sort(wp2 << np.uint64(32) | wb2)
+        assert np.array_equal(got2, want2j), \
+            f"null join pairs mismatch seed={seed}"
+    jn.free()
+
+    # null-aware group-by with non-null counts (bg_hashagg2) on small n
+    if n <= 60_000:
+        gbits = rng.integers(0, 256, size=(n + 7) // 8, dtype=np.uint8)
+        gv, _ = ctx.upload_column(probe, gpu.BG_DT_INT64, validity=gbits)
+        vv, _ = ctx.upload_column(vals, gpu.BG_DT_INT64, validity=vbits)
+        f2, a2, c2, nn2 = ctx.hashagg(
+            [gv], [vv], [gpu.BG_AGG_OP_SUM_I64], n,
+            max_groups=max(2 * kb, 64), want_nncnt=True)
+        gmask = np.unpackbits(gbits, bitorder="little")[:n].astype(bool)
+        got_n = {}
+        for g in range(len(f2)):
+            r = int(f2[g])
+            kk = int(probe[r]) if gmask[r] else None
+            acc_v = None if nn2[g, 0] == 0 else gpu.decode_agg_value(
+                gpu.BG_AGG_OP_SUM_I64, bytes(a2[g, 0]))
+            got_n[kk] = (int(c2[g]), [acc_v], [int(nn2[g, 0])])
+        want_n = oracle.hashagg_nulls([(probe, gmask)],
+                                      [("sum", vals, pmask)], n)
+        want_n = {kk[0]: v for kk, v in want_n.items()}
+        assert got_n == want_n, f"null hashagg mismatch seed={seed}"
     return n, k
 
 
