@@ -1668,6 +1668,11 @@ extern "C" int bg_hashjoin_free(void* handle) {
 // are nondeterministic — covered by the reference comparator's 1e-6
 // relative float tolerance (benchmarks/src/lib.rs:35).
 #define BG_AGG_SUM_F64 4
+// MIN/MAX over Float64: totally-ordered u64 transform (negative values
+// bit-inverted, positives sign-flipped — IEEE total order incl. infs),
+// MIN stores the complement so atomicMax + zero identity works for both.
+#define BG_AGG_MIN_F64 5
+#define BG_AGG_MAX_F64 6
 
 struct AggArgs {
   int naggs;
@@ -1831,6 +1836,27 @@ __global__ void k_hashagg(KeyArgs keys, AggArgs aggs, const u64* mask_words,
         case BG_AGG_SUM_F64: {
           const double v = reinterpret_cast<const double*>(aggs.a[a].data)[i];
           atomicAdd(reinterpret_cast<double*>(base), v);
+          break;
+        }
+        case BG_AGG_MAX_F64: {
+          // IEEE-754 totally-ordered u64 transform (sign-magnitude flip):
+          // monotone for all finite values and infs, as the reference's
+          // min_max.rs float compare is
+          const u64 bits = (u64) reinterpret_cast<const int64_t*>(
+                               aggs.a[a].data)[i];
+          const u64 v = (bits & 0x8000000000000000ull)
+                            ? ~bits
+                            : bits ^ 0x8000000000000000ull;
+          atomicMax(base, v);
+          break;
+        }
+        case BG_AGG_MIN_F64: {
+          const u64 bits = (u64) reinterpret_cast<const int64_t*>(
+                               aggs.a[a].data)[i];
+          const u64 v = (bits & 0x8000000000000000ull)
+                            ? ~bits
+                            : bits ^ 0x8000000000000000ull;
+          atomicMax(base, ~v);
           break;
         }
         default:

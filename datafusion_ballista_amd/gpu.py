@@ -346,16 +346,26 @@ BG_AGG_OP_SUM_I64 = 1
 BG_AGG_OP_MIN_I64 = 2
 BG_AGG_OP_MAX_I64 = 3
 BG_AGG_OP_SUM_F64 = 4
+BG_AGG_OP_MIN_F64 = 5
+BG_AGG_OP_MAX_F64 = 6
 
 
-def decode_agg_value(op, raw16: bytes) -> int:
-    """Accumulator bytes -> python int per aggregate op."""
+def decode_agg_value(op, raw16: bytes):
+    """Accumulator bytes -> python int/float per aggregate op."""
+    import struct as _st
     if op in (BG_AGG_OP_SUM_DEC128, BG_AGG_OP_SUM_I64):
         return int.from_bytes(raw16, "little", signed=True)
+    if op == BG_AGG_OP_SUM_F64:
+        return _st.unpack("<d", raw16[:8])[0]
     enc = int.from_bytes(raw16[:8], "little")
+    if op in (BG_AGG_OP_MIN_F64, BG_AGG_OP_MAX_F64):
+        if op == BG_AGG_OP_MIN_F64:
+            enc = ~enc & 0xFFFFFFFFFFFFFFFF
+        bits = (~enc & 0xFFFFFFFFFFFFFFFF) if not (enc >> 63)             else enc ^ (1 << 63)
+        return _st.unpack("<d", _st.pack("<Q", bits))[0]
     if op == BG_AGG_OP_MAX_I64:
         v = enc ^ (1 << 63)
-    else:  # MIN
+    else:  # MIN_I64
         v = (~enc & 0xFFFFFFFFFFFFFFFF) ^ (1 << 63)
     return v - (1 << 64) if v >= 1 << 63 else v
 

@@ -221,6 +221,9 @@ int bg_hashjoin_free(void* handle);
 #define BG_AGG_OP_SUM_I64 1
 #define BG_AGG_OP_MIN_I64 2 /* acc = order-preserving u64; host decodes */
 #define BG_AGG_OP_MAX_I64 3
+#define BG_AGG_OP_SUM_F64 4
+#define BG_AGG_OP_MIN_F64 5 /* acc = totally-ordered u64 (IEEE sign flip) */
+#define BG_AGG_OP_MAX_F64 6
 int bg_hashagg(const bg_column* key_cols, int32_t nkeys,
                const bg_column* agg_cols, const int32_t* agg_ops,
                int32_t naggs, const uint8_t* d_mask, int64_t n,

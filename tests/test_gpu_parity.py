@@ -7,6 +7,7 @@ Bar (BASELINE.json north_star): bit-exact for integer/byte/index work —
 partition IDs, filter masks, compacted indices, gathered rows, Decimal128
 sums (exact i128).
 """
+import decimal
 import os
 
 import numpy as np
@@ -1356,3 +1357,100 @@ def test_sort_shuffle_stage_with_nulls(ctx, tmp_path):
         rows = idx[offs[p]:offs[p + 1]]
         want = table.take(pa.array(rows, type=pa.uint32()))
         assert got.equals(want), f"partition {p} mismatch"
+
+
+def test_hashagg_f64_min_max_sum(ctx):
+    """Float64 aggregates: MIN/MAX via the totally-ordered u64 transform
+    (exact), SUM within the reference comparator's float tolerance
+    (benchmarks/src/lib.rs:35 — atomics reorder the additions)."""
+    n = 200_000
+    rng = np.random.default_rng(41)
+    keys = rng.integers(0, 300, size=n, dtype=np.int64)
+    vals = rng.standard_normal(n) * 1e6
+    vals[::97] = -vals[::97]  # plenty of negatives
+    kc, _ = ctx.upload_column(keys, gpu.BG_DT_INT64)
+    vc, _ = ctx.upload_column(vals, gpu.BG_DT_FLOAT64)
+    first, acc, counts = ctx.hashagg(
+        [kc], [vc, vc, vc],
+        [gpu.BG_AGG_OP_MIN_F64, gpu.BG_AGG_OP_MAX_F64,
+         gpu.BG_AGG_OP_SUM_F64], n, max_groups=1024)
+    for g in range(len(first)):
+        k = int(keys[first[g]])
+        sel = vals[keys == k]
+        gmin = gpu.decode_agg_value(gpu.BG_AGG_OP_MIN_F64, bytes(acc[g, 0]))
+        gmax = gpu.decode_agg_value(gpu.BG_AGG_OP_MAX_F64, bytes(acc[g, 1]))
+        gsum = gpu.decode_agg_value(gpu.BG_AGG_OP_SUM_F64, bytes(acc[g, 2]))
+        assert gmin == sel.min() and gmax == sel.max(), k
+        assert abs(gsum - sel.sum()) <= 1e-6 * max(1.0, abs(sel.sum())), k
+    assert len(first) == len(np.unique(keys))
+
+
+def test_two_phase_aggregate_via_shuffle(ctx, tmp_path):
+    """The reference's AggregateExec Partial -> hash shuffle -> Final
+    composition: per-split partial SUM(dec)/COUNT, written
+    through the GPU sort-shuffle stage partitioned by group key, then a
+    final merge aggregate per partition; the union of final groups must
+    equal the oracle's one-shot global aggregate exactly (i128 sums)."""
+    from datafusion_ballista_amd import engine, shuffle
+    rng = np.random.default_rng(55)
+    nsplits, k = 3, 4
+    all_keys, all_vals = [], []
+    partial_tables = []
+    for s_i in range(nsplits):
+        n = 40_000 + 1000 * s_i
+        keys = rng.integers(0, 3_000, size=n, dtype=np.int64)
+        vals = rng.integers(-10**9, 10**9, size=n, dtype=np.int64)
+        all_keys.append(keys)
+        all_vals.append(vals)
+        kc, _ = ctx.upload_column(keys, gpu.BG_DT_INT64)
+        dec16 = dec_bytes(vals)
+        vc = ctx.column(gpu.BG_DT_DECIMAL128, ctx.upload(dec16), n)
+        first, acc, counts = ctx.hashagg(
+            [kc], [vc], [gpu.BG_AGG_OP_SUM_DEC128], n, max_groups=8192)
+        g = len(first)
+        pk = keys[first]
+        psum = np.array([int.from_bytes(bytes(acc[i, 0]), "little",
+                                        signed=True) for i in range(g)],
+                        dtype=np.int64)  # fits: |vals| small enough
+        partial_tables.append(pa.table({
+            "k": pa.array(pk.astype(np.int64)),
+            "s": pa.array([decimal.Decimal(int(v)) for v in psum],
+                          type=pa.decimal128(38, 0)),
+            "c": pa.array(counts.astype(np.int64)),
+        }))
+    ex = engine.GpuQueryStageExecutor(ctx, "job-2ph", 9, str(tmp_path),
+                                      key_columns=[0], num_partitions=k)
+    summaries = ex.execute_query_stage(0, partial_tables)
+    data_path = summaries[0].path
+    final = {}
+    for p in range(k):
+        batches = shuffle.read_partition(data_path, data_path + ".index", p)
+        if not batches:
+            continue
+        t = pa.Table.from_batches(batches)
+        pk = t.column("k").to_numpy()
+        ps = t.column("s").combine_chunks()
+        raw = np.frombuffer(ps.buffers()[1], dtype=np.uint8,
+                            count=16 * len(t)).reshape(len(t), 16)
+        pc = t.column("c").to_numpy()
+        n2 = len(t)
+        kc2, _ = ctx.upload_column(pk.astype(np.int64), gpu.BG_DT_INT64)
+        sc2 = ctx.column(gpu.BG_DT_DECIMAL128,
+                         ctx.upload(raw.reshape(-1).copy()), n2)
+        cc2, _ = ctx.upload_column(pc.astype(np.int64), gpu.BG_DT_INT64)
+        f2, a2, c2 = ctx.hashagg(
+            [kc2], [sc2, cc2],
+            [gpu.BG_AGG_OP_SUM_DEC128, gpu.BG_AGG_OP_SUM_I64], n2,
+            max_groups=8192)
+        for g in range(len(f2)):
+            key = int(pk[f2[g]])
+            assert key not in final, "group split across partitions"
+            final[key] = (
+                int.from_bytes(bytes(a2[g, 0]), "little", signed=True),
+                gpu.decode_agg_value(gpu.BG_AGG_OP_SUM_I64,
+                                     bytes(a2[g, 1])))
+    keys_all = np.concatenate(all_keys)
+    vals_all = np.concatenate(all_vals)
+    want = oracle.hashagg([keys_all], [("sum", vals_all)], len(keys_all))
+    want = {kk[0]: (v[1][0], v[0]) for kk, v in want.items()}
+    assert final == want
