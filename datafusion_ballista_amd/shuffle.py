@@ -258,6 +258,14 @@ def read_partition_gpu(ctx, data_path, index_path, partition_id, schema):
                 raise RuntimeError("uncompressed batches: host path")
             cols = []
             for ci in range(len(schema.types)):
+                voff_, vlen_ = bufs[2 * ci]
+                if vlen_:
+                    # null-bearing batches stay on the host read path
+                    # (read_partition), which reassembles validity exactly;
+                    # silent null-dropping here would be a parity hole
+                    raise RuntimeError(
+                        "read_partition_gpu: null-carrying column "
+                        f"{ci}: use read_partition (host) for this stream")
                 boff, blen = bufs[2 * ci + 1]  # data buffer (validity at 2ci)
                 if blen == 0:
                     cols.append(None)
