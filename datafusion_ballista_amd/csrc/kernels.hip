@@ -81,12 +81,14 @@ static int set_hip_err(hipError_t e, const char* what) {
 // ---------------------------------------------------------------------------
 #include <map>
 #include <mutex>
+#include <set>
 #include <unordered_map>
 #include <vector>
 
 static std::mutex g_pool_mu;
 static std::unordered_map<uint64_t, std::vector<void*>> g_pool_free;
 static std::unordered_map<void*, uint64_t> g_pool_sizes;
+static std::set<void*> g_pool_freeset;
 
 static uint64_t pool_round(uint64_t bytes) {
   uint64_t p = 256;
@@ -102,6 +104,7 @@ static hipError_t pool_malloc(void** out, uint64_t bytes) {
     if (it != g_pool_free.end() && !it->second.empty()) {
       *out = it->second.back();
       it->second.pop_back();
+      g_pool_freeset.erase(*out);
       return hipSuccess;
     }
   }
@@ -128,6 +131,13 @@ static hipError_t pool_release(void* p) {
   std::lock_guard<std::mutex> g(g_pool_mu);
   auto it = g_pool_sizes.find(p);
   if (it == g_pool_sizes.end()) return hipFree(p);  // not pooled
+  // double-release guard: handing one buffer to two later allocations
+  // aliases them and corrupts unrelated state — fail loudly instead
+  if (!g_pool_freeset.insert(p).second) {
+    fprintf(stderr, "bg pool: DOUBLE RELEASE of %p (size %llu)\n", p,
+            (unsigned long long)it->second);
+    abort();
+  }
   g_pool_free[it->second].push_back(p);
   return hipSuccess;
 }
@@ -453,6 +463,93 @@ __global__ void k_segment_sums_i64(const u64* in, int64_t m, int64_t seg,
 }
 
 // Host-side hierarchical exclusive scan (two levels; m up to ~10^9).
+// ---------------------------------------------------------------------------
+// Single-pass exclusive scan (decoupled lookback): tiles are acquired in
+// order through a ticket counter (so every predecessor of a spinning tile
+// is resident and progressing — deadlock-free), each tile publishes
+// {flag, value} packed in ONE u64 (flag in bits 63:62: 1 = aggregate,
+// 2 = inclusive prefix; values must stay < 2^62 — counts/byte-lengths do),
+// with agent-scope release/acquire atomics (MI355X XCDs have private L2s;
+// scope-agent makes the handoff visible across them).  16 B/element of
+// traffic vs the 3-pass hierarchical scan's 24, and one launch instead of
+// three.
+// ---------------------------------------------------------------------------
+#define SCAN_TILE 4096
+#define SCAN_ELEMS (SCAN_TILE / BG_BLOCK)
+#define SCAN_FLAG_SHIFT 62
+
+__global__ void k_scan_lookback(const u64* in, int64_t n, i64* out,
+                                i64* total_out, u64* state, int* ticket,
+                                int64_t ntiles) {
+  __shared__ u64 lds[SCAN_TILE];
+  __shared__ u64 tsum[BG_BLOCK];
+  __shared__ u64 texcl[BG_BLOCK];
+  __shared__ u64 s_carry;
+  __shared__ int s_tile;
+  const int tid = threadIdx.x;
+  while (true) {
+    if (tid == 0) s_tile = atomicAdd(ticket, 1);
+    __syncthreads();
+    const int64_t tile = s_tile;
+    if (tile >= ntiles) return;
+    const int64_t base = tile * SCAN_TILE;
+    const int64_t lim = n - base < SCAN_TILE ? n - base : SCAN_TILE;
+    for (int i = tid; i < SCAN_TILE; i += BG_BLOCK)
+      lds[i] = (i < lim) ? __builtin_nontemporal_load(&in[base + i]) : 0;
+    __syncthreads();
+    u64 run = 0;
+    for (int j = 0; j < SCAN_ELEMS; ++j) {
+      const int i = tid * SCAN_ELEMS + j;
+      const u64 v = lds[i];
+      lds[i] = run;  // exclusive within this thread's chunk
+      run += v;
+    }
+    tsum[tid] = run;
+    __syncthreads();
+    for (int off = 1; off < BG_BLOCK; off <<= 1) {
+      const u64 v = (tid >= off) ? tsum[tid - off] : 0;
+      __syncthreads();
+      tsum[tid] += v;
+      __syncthreads();
+    }
+    const u64 agg = tsum[BG_BLOCK - 1];
+    texcl[tid] = (tid == 0) ? 0 : tsum[tid - 1];
+    if (tid == 0) {
+      if (tile == 0) {
+        __hip_atomic_store(&state[0], agg | (2ull << SCAN_FLAG_SHIFT),
+                           __ATOMIC_RELEASE, __HIP_MEMORY_SCOPE_AGENT);
+        s_carry = 0;
+      } else {
+        __hip_atomic_store(&state[tile], agg | (1ull << SCAN_FLAG_SHIFT),
+                           __ATOMIC_RELEASE, __HIP_MEMORY_SCOPE_AGENT);
+        u64 carry = 0;
+        int64_t t = tile - 1;
+        while (t >= 0) {
+          u64 st;
+          do {
+            st = __hip_atomic_load(&state[t], __ATOMIC_ACQUIRE,
+                                   __HIP_MEMORY_SCOPE_AGENT);
+            if (!(st >> SCAN_FLAG_SHIFT)) __builtin_amdgcn_s_sleep(2);
+          } while (!(st >> SCAN_FLAG_SHIFT));
+          carry += st & ((1ull << SCAN_FLAG_SHIFT) - 1);
+          if ((st >> SCAN_FLAG_SHIFT) == 2ull) break;
+          --t;
+        }
+        __hip_atomic_store(&state[tile],
+                           (carry + agg) | (2ull << SCAN_FLAG_SHIFT),
+                           __ATOMIC_RELEASE, __HIP_MEMORY_SCOPE_AGENT);
+        s_carry = carry;
+      }
+      if (tile == ntiles - 1 && total_out) *total_out = (i64)(s_carry + agg);
+    }
+    __syncthreads();
+    const u64 carry = s_carry;
+    for (int i = tid; i < lim; i += BG_BLOCK)
+      out[base + i] = (i64)(carry + texcl[i / SCAN_ELEMS] + lds[i]);
+    __syncthreads();
+  }
+}
+
 static int scan_exclusive_i64(const u64* d_in, int64_t m, i64* d_out,
                               i64* d_total) {
   const int64_t SMALL = 1 << 18;
@@ -461,6 +558,29 @@ static int scan_exclusive_i64(const u64* d_in, int64_t m, i64* d_out,
                        d_in, m, d_out, d_total, m > 0 ? m : 1, nullptr);
     hipError_t e = hipGetLastError();
     if (e != hipSuccess) return set_hip_err(e, "scan small");
+    return BG_OK;
+  }
+  const char* envs = getenv("BG_SCAN_LOOKBACK");
+  if (!(envs && envs[0] == '0')) {
+    const int64_t ntiles = (m + SCAN_TILE - 1) / SCAN_TILE;
+    u64* d_state;
+    int* d_ticket;
+    hipError_t e;
+    e = pool_malloc((void**)&d_state, sizeof(u64) * ntiles);
+    if (e != hipSuccess) return set_hip_err(e, "scan malloc");
+    e = pool_malloc((void**)&d_ticket, sizeof(int));
+    if (e != hipSuccess) return set_hip_err(e, "scan malloc");
+    e = hipMemsetAsync(d_state, 0, sizeof(u64) * ntiles, 0);
+    if (e != hipSuccess) return set_hip_err(e, "scan memset");
+    e = hipMemsetAsync(d_ticket, 0, sizeof(int), 0);
+    if (e != hipSuccess) return set_hip_err(e, "scan memset");
+    int blocks = (int)bg_imin64(ntiles, BG_MAX_BLOCKS);
+    hipLaunchKernelGGL(k_scan_lookback, dim3(blocks), dim3(BG_BLOCK), 0, 0,
+                       d_in, m, d_out, d_total, d_state, d_ticket, ntiles);
+    e = hipGetLastError();
+    (void)pool_release(d_state);
+    (void)pool_release(d_ticket);
+    if (e != hipSuccess) return set_hip_err(e, "scan lookback");
     return BG_OK;
   }
   const int64_t seg = 1 << 16;  // 65536 elems per block segment
