@@ -552,6 +552,15 @@ __global__ void k_scan_lookback(const u64* in, int64_t n, i64* out,
 
 static int scan_exclusive_i64(const u64* d_in, int64_t m, i64* d_out,
                               i64* d_total) {
+  if (m <= 0) {
+    // nothing to scan: the total must still be DEFINED (pool buffers are
+    // recycled — stale bytes here once leaked a phantom match count)
+    if (d_total) {
+      hipError_t e = hipMemsetAsync(d_total, 0, sizeof(i64), 0);
+      if (e != hipSuccess) return set_hip_err(e, "scan total zero");
+    }
+    return BG_OK;
+  }
   const int64_t SMALL = 1 << 18;
   if (m <= SMALL) {
     hipLaunchKernelGGL(k_exclusive_scan_i64, dim3(1), dim3(BG_BLOCK), 0, 0,
@@ -560,8 +569,12 @@ static int scan_exclusive_i64(const u64* d_in, int64_t m, i64* d_out,
     if (e != hipSuccess) return set_hip_err(e, "scan small");
     return BG_OK;
   }
+  // single-pass decoupled-lookback scan: MEASURED NEGATIVE on MI355X
+  // (31.6 ms vs 12.4 ms hierarchical at 322M elements — the ticket +
+  // LDS block-scan + spin overhead at 4096-element tiles outweighs the
+  // 24->16 B/element traffic saving).  Kept behind BG_SCAN_LOOKBACK=1.
   const char* envs = getenv("BG_SCAN_LOOKBACK");
-  if (!(envs && envs[0] == '0')) {
+  if (envs && envs[0] == '1') {
     const int64_t ntiles = (m + SCAN_TILE - 1) / SCAN_TILE;
     u64* d_state;
     int* d_ticket;
@@ -1337,43 +1350,27 @@ extern "C" int bg_q1_agg(const bg_column* rf, const bg_column* ls,
 // offsets and no atomic append nondeterminism in sizes.
 // ---------------------------------------------------------------------------
 struct BgJoinTable {
-  ulong2* nodes;  // packed chain node: {key, next+1 | orig<<32}
+  ulong2* nodes;  // packed chain node: {key, next} — ONE line per hop
   int* head;      // bucket heads (-1 empty)
   int64_t n_build;
   u64 mask;       // nb - 1
   i64* probe_offsets = nullptr;  // per-probe-row output offsets (count phase)
   int64_t probe_n = 0;
-  // L2-locality bucketing (big tables): nodes laid out grouped by the top
-  // log2(JOIN_BUCKETS) bits of the table bucket index, so one probe
-  // bucket touches a contiguous ~(nodes+head)/JOIN_BUCKETS shard that
-  // fits an XCD's L2 instead of striding the whole table
-  int bucketed = 0;
-  int shift = 0;                  // (hash & mask) >> shift = locality bucket
-  int64_t* bkeys = nullptr;       // probe keys in bucketed order (per probe)
-  uint32_t* borig = nullptr;      // original probe index per bucketed slot
 };
 
-#define JOIN_BUCKETS 128
-#define JOIN_BUCKET_MIN_BYTES (24ll << 20)
-
-// orig: original build-row index per node slot (NULL = identity); node.y
-// packs {orig (hi 32) | next+1 (lo 32)} so the probe can return the
-// ORIGINAL build index while nodes sit in locality-bucketed order
 __global__ void k_join_build(const int64_t* keys, const uint8_t* valid,
-                             const uint32_t* orig, int64_t n, int* head,
-                             ulong2* nodes, u64 mask) {
+                             int64_t n, int* head, ulong2* nodes, u64 mask) {
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
        i += (int64_t)gridDim.x * blockDim.x) {
-    const uint32_t og = orig ? orig[i] : (uint32_t)i;
     // inner-join null semantics (null_equals_null=false in the reference's
     // HashJoinExec): a NULL build key matches nothing — leave it unlinked
-    if (!bit_valid(valid, orig ? (int64_t)og : i)) continue;
+    if (!bit_valid(valid, i)) continue;
     const u64 k = (u64)keys[i];
     const u64 b = bg_hash_u64(k) & mask;
     const int prev = atomicExch(&head[b], (int)i);
     ulong2 node;
     node.x = k;
-    node.y = ((u64)og << 32) | (u64)(uint32_t)(prev + 1);  // 0 terminates
+    node.y = (u64)(int64_t)prev;  // sign-extended: -1 terminates
     nodes[i] = node;
   }
 }
@@ -1409,7 +1406,7 @@ __global__ void k_join_count(const int64_t* probe_keys,
         if (cur[j] >= 0) {
           const ulong2 node = nodes[cur[j]];
           if (node.x == key[j]) cnt[j]++;
-          cur[j] = (int64_t)(uint32_t)(node.y & 0xffffffffu) - 1;
+          cur[j] = (int64_t)node.y;
           any = true;
         }
       }
@@ -1449,132 +1446,11 @@ __global__ void k_join_fill(const int64_t* probe_keys,
           const ulong2 node = nodes[cur[j]];
           if (node.x == key[j]) {
             out_probe[w[j]] = (uint32_t)idx[j];
-            out_build[w[j]] = (uint32_t)(node.y >> 32);
+            out_build[w[j]] = (uint32_t)cur[j];
             ++w[j];
           }
-          cur[j] = (int64_t)(uint32_t)(node.y & 0xffffffffu) - 1;
+          cur[j] = (int64_t)node.y;
           any = true;
-        }
-      }
-    }
-  }
-}
-
-// locality-bucket id per key: top bits of the TABLE bucket index, so one
-// probe bucket touches one contiguous head/node shard
-__global__ void k_join_bucket_ids(const int64_t* keys, int64_t n, u64 mask,
-                                  int shift, uint32_t* pids) {
-  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
-       i += (int64_t)gridDim.x * blockDim.x)
-    pids[i] = (uint32_t)((bg_hash_u64((u64)__builtin_nontemporal_load(
-                              &keys[i])) & mask) >> shift);
-}
-
-// Probes in BUCKETED order via a dynamic work queue (front-to-back), so
-// all resident waves hit the same ~table/JOIN_BUCKETS shard at a time and
-// it stays L2-resident; results land at the ORIGINAL probe index, so the
-// output (and the probe-major ordering contract) is unchanged.
-#define JOIN_Q_CHUNK 2048
-
-__device__ __forceinline__ i64 bg_jbcast64(i64 v) {
-  const int lo = __builtin_amdgcn_readfirstlane((int)(v & 0xffffffff));
-  const int hi = __builtin_amdgcn_readfirstlane((int)(((u64)v) >> 32));
-  return (i64)(((u64)(uint32_t)hi << 32) | (u64)(uint32_t)lo);
-}
-
-__global__ void k_join_count_q(const int64_t* bkeys, const uint32_t* borig,
-                               const uint8_t* probe_valid, int64_t n,
-                               const int* head, const ulong2* nodes, u64 mask,
-                               u64* counts, int* ctr) {
-  const int lane = lane_id();
-  while (true) {
-    i64 base = 0;
-    if (lane == 0) base = (i64)atomicAdd(ctr, 1) * JOIN_Q_CHUNK;
-    base = bg_jbcast64(base);
-    if (base >= n) return;
-    const i64 end = base + JOIN_Q_CHUNK < n ? base + JOIN_Q_CHUNK : n;
-    for (i64 ib = base; ib < end; ib += BG_WAVE * JOIN_ILP) {
-      i64 idx[JOIN_ILP];
-      u64 key[JOIN_ILP];
-      i64 cur[JOIN_ILP];
-      u64 cnt[JOIN_ILP];
-      uint32_t og[JOIN_ILP];
-#pragma unroll
-      for (int j = 0; j < JOIN_ILP; ++j) {
-        idx[j] = ib + lane + (i64)j * BG_WAVE;
-        const bool in = idx[j] < end;
-        og[j] = in ? borig[idx[j]] : 0;
-        const bool act = in && bit_valid(probe_valid, (int64_t)og[j]);
-        key[j] = act ? (u64)bkeys[idx[j]] : 0;
-        cur[j] = act ? (i64)head[bg_hash_u64(key[j]) & mask] : -1;
-        cnt[j] = 0;
-      }
-      bool any = true;
-      while (any) {
-        any = false;
-#pragma unroll
-        for (int j = 0; j < JOIN_ILP; ++j) {
-          if (cur[j] >= 0) {
-            const ulong2 node = nodes[cur[j]];
-            if (node.x == key[j]) cnt[j]++;
-            cur[j] = (i64)(uint32_t)(node.y & 0xffffffffu) - 1;
-            any = true;
-          }
-        }
-      }
-#pragma unroll
-      for (int j = 0; j < JOIN_ILP; ++j)
-        if (idx[j] < end && bit_valid(probe_valid, (int64_t)og[j]))
-          counts[og[j]] = cnt[j];
-        else if (idx[j] < end)
-          counts[og[j]] = 0;
-    }
-  }
-}
-
-__global__ void k_join_fill_q(const int64_t* bkeys, const uint32_t* borig,
-                              const uint8_t* probe_valid, int64_t n,
-                              const int* head, const ulong2* nodes, u64 mask,
-                              const i64* offsets, uint32_t* out_probe,
-                              uint32_t* out_build, int* ctr) {
-  const int lane = lane_id();
-  while (true) {
-    i64 base = 0;
-    if (lane == 0) base = (i64)atomicAdd(ctr, 1) * JOIN_Q_CHUNK;
-    base = bg_jbcast64(base);
-    if (base >= n) return;
-    const i64 end = base + JOIN_Q_CHUNK < n ? base + JOIN_Q_CHUNK : n;
-    for (i64 ib = base; ib < end; ib += BG_WAVE * JOIN_ILP) {
-      i64 idx[JOIN_ILP];
-      u64 key[JOIN_ILP];
-      i64 cur[JOIN_ILP];
-      i64 w[JOIN_ILP];
-      uint32_t og[JOIN_ILP];
-#pragma unroll
-      for (int j = 0; j < JOIN_ILP; ++j) {
-        idx[j] = ib + lane + (i64)j * BG_WAVE;
-        const bool in = idx[j] < end;
-        og[j] = in ? borig[idx[j]] : 0;
-        const bool act = in && bit_valid(probe_valid, (int64_t)og[j]);
-        key[j] = act ? (u64)bkeys[idx[j]] : 0;
-        cur[j] = act ? (i64)head[bg_hash_u64(key[j]) & mask] : -1;
-        w[j] = act ? offsets[og[j]] : 0;
-      }
-      bool any = true;
-      while (any) {
-        any = false;
-#pragma unroll
-        for (int j = 0; j < JOIN_ILP; ++j) {
-          if (cur[j] >= 0) {
-            const ulong2 node = nodes[cur[j]];
-            if (node.x == key[j]) {
-              out_probe[w[j]] = og[j];
-              out_build[w[j]] = (uint32_t)(node.y >> 32);
-              ++w[j];
-            }
-            cur[j] = (i64)(uint32_t)(node.y & 0xffffffffu) - 1;
-            any = true;
-          }
         }
       }
     }
@@ -1597,53 +1473,15 @@ extern "C" int bg_hashjoin_build(const bg_column* build_keys, int64_t n,
   t.n_build = n;
   const u64 nb = next_pow2_u64((u64)(n > 4 ? n * 2 : 8));
   t.mask = nb - 1;
-  // L2-locality bucketing — MEASURED NEGATIVE on MI355X (DESIGN.md §8
-  // negative results: 35.8 ms vs 20.3 ms on 150M probes vs 15M build; the
-  // split+gather+queue overhead and 8 XCD-private L2s beat the locality
-  // gain).  Kept behind BG_JOIN_BUCKET=1 as the documented experiment.
-  const char* envb = getenv("BG_JOIN_BUCKET");
-  bool bucketed = envb && envb[0] == '1' && nb >= 8 * JOIN_BUCKETS &&
-      (int64_t)(sizeof(ulong2) * n + sizeof(int) * nb) > JOIN_BUCKET_MIN_BYTES;
   HIP_TRY(pool_malloc((void**)&t.nodes, sizeof(ulong2) * (n ? n : 1)));
   HIP_TRY(pool_malloc((void**)&t.head, sizeof(int) * nb));
   HIP_TRY(hipMemset(t.head, 0xff, sizeof(int) * nb));  // -1
   int blocks = (int)bg_imin64((n + BG_BLOCK - 1) / BG_BLOCK, BG_MAX_BLOCKS);
   if (blocks == 0) blocks = 1;
-  if (bucketed) {
-    int lg = 0;
-    while ((1ull << lg) < nb) ++lg;
-    t.shift = lg - 7;  // JOIN_BUCKETS = 128
-    t.bucketed = 1;
-    uint32_t* d_pids;
-    uint32_t* d_idx;
-    int64_t* d_offs;
-    int64_t* d_bkeys;
-    HIP_TRY(pool_malloc((void**)&d_pids, sizeof(uint32_t) * n));
-    HIP_TRY(pool_malloc((void**)&d_idx, sizeof(uint32_t) * n));
-    HIP_TRY(pool_malloc((void**)&d_offs, sizeof(i64) * (JOIN_BUCKETS + 1)));
-    HIP_TRY(pool_malloc((void**)&d_bkeys, sizeof(int64_t) * n));
-    hipLaunchKernelGGL(k_join_bucket_ids, dim3(blocks), dim3(BG_BLOCK), 0, 0,
-                       (const int64_t*)build_keys->d_data, n, t.mask,
-                       t.shift, d_pids);
-    int rc = bg_partition_indices(d_pids, n, JOIN_BUCKETS, d_idx, d_offs);
-    if (rc != BG_OK) return rc;
-    rc = bg_gather(build_keys->d_data, 8, d_idx, n, d_bkeys);
-    if (rc != BG_OK) return rc;
-    hipLaunchKernelGGL(k_join_build, dim3(blocks), dim3(BG_BLOCK), 0, 0,
-                       (const int64_t*)d_bkeys, build_keys->d_validity,
-                       d_idx, n, t.head, t.nodes, t.mask);
-    HIP_TRY(hipGetLastError());
-    (void)pool_release(d_pids);
-    (void)pool_release(d_offs);
-    (void)pool_release(d_idx);
-    (void)pool_release(d_bkeys);
-  } else {
-    hipLaunchKernelGGL(k_join_build, dim3(blocks), dim3(BG_BLOCK), 0, 0,
-                       (const int64_t*)build_keys->d_data,
-                       build_keys->d_validity, (const uint32_t*)nullptr, n,
-                       t.head, t.nodes, t.mask);
-    HIP_TRY(hipGetLastError());
-  }
+  hipLaunchKernelGGL(k_join_build, dim3(blocks), dim3(BG_BLOCK), 0, 0,
+                     (const int64_t*)build_keys->d_data,
+                     build_keys->d_validity, n, t.head, t.nodes, t.mask);
+  HIP_TRY(hipGetLastError());
   BgJoinTable* h = new BgJoinTable(t);
   *out_handle = h;
   return BG_OK;
@@ -1663,8 +1501,6 @@ extern "C" int bg_hashjoin_probe_count(void* handle,
     (void)pool_release(t->probe_offsets);
     t->probe_offsets = nullptr;
   }
-  if (t->bkeys) { (void)pool_release(t->bkeys); t->bkeys = nullptr; }
-  if (t->borig) { (void)pool_release(t->borig); t->borig = nullptr; }
   u64* d_counts;
   i64* d_offs;
   i64* d_total;
@@ -1673,31 +1509,6 @@ extern "C" int bg_hashjoin_probe_count(void* handle,
   HIP_TRY(pool_malloc((void**)&d_total, sizeof(i64)));
   int blocks = (int)bg_imin64((n + BG_BLOCK - 1) / BG_BLOCK, BG_MAX_BLOCKS);
   if (blocks == 0) blocks = 1;
-  if (t->bucketed && n > 0) {
-    uint32_t* d_pids;
-    int64_t* d_boffs;
-    HIP_TRY(pool_malloc((void**)&d_pids, sizeof(uint32_t) * n));
-    HIP_TRY(pool_malloc((void**)&d_boffs, sizeof(i64) * (JOIN_BUCKETS + 1)));
-    HIP_TRY(pool_malloc((void**)&t->borig, sizeof(uint32_t) * n));
-    HIP_TRY(pool_malloc((void**)&t->bkeys, sizeof(int64_t) * n));
-    hipLaunchKernelGGL(k_join_bucket_ids, dim3(blocks), dim3(BG_BLOCK), 0, 0,
-                       (const int64_t*)probe_keys->d_data, n, t->mask,
-                       t->shift, d_pids);
-    int rc = bg_partition_indices(d_pids, n, JOIN_BUCKETS, t->borig, d_boffs);
-    if (rc != BG_OK) return rc;
-    rc = bg_gather(probe_keys->d_data, 8, t->borig, n, t->bkeys);
-    if (rc != BG_OK) return rc;
-    (void)pool_release(d_pids);
-    (void)pool_release(d_boffs);
-    int* d_ctr;
-    HIP_TRY(pool_malloc((void**)&d_ctr, sizeof(int)));
-    HIP_TRY(hipMemsetAsync(d_ctr, 0, sizeof(int), 0));
-    hipLaunchKernelGGL(k_join_count_q, dim3(blocks), dim3(BG_BLOCK), 0, 0,
-                       t->bkeys, t->borig, probe_keys->d_validity, n,
-                       t->head, t->nodes, t->mask, d_counts, d_ctr);
-    HIP_TRY(hipGetLastError());
-    (void)pool_release(d_ctr);
-  } else
   hipLaunchKernelGGL(k_join_count, dim3(blocks), dim3(BG_BLOCK), 0, 0,
                      (const int64_t*)probe_keys->d_data,
                      probe_keys->d_validity, n, t->head, t->nodes,
@@ -1727,20 +1538,6 @@ extern "C" int bg_hashjoin_probe_fill(void* handle,
     return set_err(BG_ERR_INVALID, "call bg_hashjoin_probe_count first");
   int blocks = (int)bg_imin64((n + BG_BLOCK - 1) / BG_BLOCK, BG_MAX_BLOCKS);
   if (blocks == 0) blocks = 1;
-  if (t->bucketed && n > 0) {
-    if (!t->bkeys || !t->borig)
-      return set_err(BG_ERR_INVALID, "call bg_hashjoin_probe_count first");
-    int* d_ctr;
-    HIP_TRY(pool_malloc((void**)&d_ctr, sizeof(int)));
-    HIP_TRY(hipMemsetAsync(d_ctr, 0, sizeof(int), 0));
-    hipLaunchKernelGGL(k_join_fill_q, dim3(blocks), dim3(BG_BLOCK), 0, 0,
-                       t->bkeys, t->borig, probe_keys->d_validity, n,
-                       t->head, t->nodes, t->mask, t->probe_offsets,
-                       d_out_probe, d_out_build, d_ctr);
-    HIP_TRY(hipGetLastError());
-    (void)pool_release(d_ctr);
-    return BG_OK;
-  }
   hipLaunchKernelGGL(k_join_fill, dim3(blocks), dim3(BG_BLOCK), 0, 0,
                      (const int64_t*)probe_keys->d_data,
                      probe_keys->d_validity, n, t->head, t->nodes,
@@ -1754,8 +1551,6 @@ extern "C" int bg_hashjoin_free(void* handle) {
   REQUIRE_INIT();
   BgJoinTable* t = (BgJoinTable*)handle;
   if (!t) return BG_OK;
-  if (t->bkeys) (void)pool_release(t->bkeys);
-  if (t->borig) (void)pool_release(t->borig);
   (void)pool_release(t->nodes);
   (void)pool_release(t->head);
   if (t->probe_offsets) (void)pool_release(t->probe_offsets);

@@ -107,6 +107,7 @@ class DeviceBuffer:
 
     def __init__(self, ctx, nbytes: int):
         self._ctx = ctx
+        self._owns = True
         self.nbytes = nbytes
         ptr = ctypes.c_void_p()
         _check(ctx.L.bg_malloc(ctypes.c_uint64(nbytes), ctypes.byref(ptr)),
@@ -117,6 +118,20 @@ class DeviceBuffer:
         if self.ptr and self.ptr.value:
             self._ctx.L.bg_free(self.ptr)
             self.ptr = ctypes.c_void_p()
+
+    def __del__(self):
+        # return the buffer to the allocator pool when the last reference
+        # dies — without this every stage leaks its temporaries and later
+        # allocations pay fresh multi-GB hipMallocs (~100 ms/GB) instead
+        # of a free-list hit.  Only buffers WE allocated: wrapper objects
+        # built with __new__ around borrowed pointers (torch tensors) have
+        # no _owns flag and must never be freed here.
+        if not getattr(self, "_owns", False):
+            return
+        try:
+            self.free()
+        except Exception:
+            pass  # interpreter shutdown: the library may already be gone
 
     def upload(self, arr: np.ndarray):
         a = np.ascontiguousarray(arr)
@@ -145,17 +160,18 @@ class GpuStageContext:
 
     # ---- memory ----
     def alloc(self, nbytes: int) -> DeviceBuffer:
-        b = DeviceBuffer(self, nbytes)
-        self._bufs.append(b)
-        return b
+        # ownership lives with the returned DeviceBuffer (__del__ returns
+        # it to the pool when the last reference dies) — pinning every
+        # allocation on the context kept multi-GB stage temporaries alive
+        # for the context's whole life, so long pipelines paid fresh
+        # hipMallocs instead of pool hits
+        return DeviceBuffer(self, nbytes)
 
     def upload(self, arr: np.ndarray) -> DeviceBuffer:
         return self.alloc(arr.nbytes).upload(arr)
 
     def close(self):
-        for b in self._bufs:
-            b.free()
-        self._bufs = []
+        self._bufs = []  # legacy; buffers free themselves via __del__
 
     def synchronize(self):
         _check(self.L.bg_synchronize(), "bg_synchronize")
@@ -164,9 +180,13 @@ class GpuStageContext:
     def column(self, dtype: int, buf: DeviceBuffer, n: int,
                validity: DeviceBuffer = None, precision=0, scale=0,
                offsets: DeviceBuffer = None) -> BgColumn:
-        return BgColumn(dtype, precision, scale, 0, buf.ptr,
-                        validity.ptr if validity else None,
-                        offsets.ptr if offsets else None, n)
+        col = BgColumn(dtype, precision, scale, 0, buf.ptr,
+                       validity.ptr if validity else None,
+                       offsets.ptr if offsets else None, n)
+        # the struct carries raw device pointers: keep the backing
+        # DeviceBuffers alive as long as the column handle is
+        col._keep = (buf, validity, offsets)
+        return col
 
     def upload_utf8_column(self, strings):
         """list[bytes] -> BG_DT_UTF8 column (Arrow i32 offsets + data)."""

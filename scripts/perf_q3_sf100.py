@@ -57,7 +57,12 @@ def main():
     torch.cuda.synchronize()
 
     def run_query():
+        marks = []
+        def mk(name):
+            ctx.synchronize()
+            marks.append((name, time.perf_counter()))
         t0 = time.perf_counter()
+        mk("start")
         # stage A: customer filter -> build
         cc = col_of(c_custkey, gpu.BG_DT_INT64)
         cmask = ctx.eval_predicates([cc], [(0, gpu.BG_PRED_LT, 0,
@@ -66,6 +71,7 @@ def main():
         ckeys = ctx.gather(_wrap(ctx, c_custkey), 8, cidx, nc)
         cjoin = gpu.GpuHashJoin(ctx, ctx.column(gpu.BG_DT_INT64, ckeys, nc),
                                 nc)
+        mk("stageA_cust_build")
         # stage B: orders filter -> probe customers -> build orderkeys
         od = col_of(o_orderdate, gpu.BG_DT_DATE32)
         omask = ctx.eval_predicates([od], [(0, gpu.BG_PRED_LT, 0, CUTOFF)],
@@ -82,29 +88,38 @@ def main():
         ok_matched = ctx.gather(oidx_i64, 8, ppos, nmatch)
         ojoin = gpu.GpuHashJoin(
             ctx, ctx.column(gpu.BG_DT_INT64, ok_matched, nmatch), nmatch)
+        mk("stageB_orders")
         # stage C: lineitem filter -> probe orders -> revenue -> group-by
         ld = col_of(l_shipdate, gpu.BG_DT_DATE32)
         lmask = ctx.eval_predicates([ld], [(0, gpu.BG_PRED_GT, CUTOFF, 0)],
                                     NLI)
+        mk("C.filter")
         lidx, nl = ctx.mask_to_indices(lmask, NLI)
+        mk("C.indices")
         lk = ctx.gather(_wrap(ctx, l_orderkey), 8, lidx, nl)
         lp = ctx.gather(_wrap(ctx, l_price), 16, lidx, nl)
         ldc = ctx.gather(_wrap(ctx, l_disc), 16, lidx, nl)
+        mk("C.gather3")
         lppos, lbpos, nlm = ojoin.probe(
             ctx.column(gpu.BG_DT_INT64, lk, nl), nl)
+        mk("C.probe")
         jk = ctx.gather(lk, 8, lppos, nlm)
         jp = ctx.gather(lp, 16, lppos, nlm)
         jd = ctx.gather(ldc, 16, lppos, nlm)
+        mk("C.gatherj")
         one_minus = ctx.project_dec128(
             gpu.BG_PROJ_RSUB_LIT,
             ctx.column(gpu.BG_DT_DECIMAL128, jd, nlm), None, 100, nlm)
         rev = ctx.project_dec128(
             gpu.BG_PROJ_MUL, ctx.column(gpu.BG_DT_DECIMAL128, jp, nlm),
             ctx.column(gpu.BG_DT_DECIMAL128, one_minus, nlm), 0, nlm)
+        mk("C.proj")
+        mk("stageC_lineitem_join_proj")
         first, acc, counts = ctx.hashagg(
             [ctx.column(gpu.BG_DT_INT64, jk, nlm)],
             [ctx.column(gpu.BG_DT_DECIMAL128, rev, nlm)],
             [gpu.BG_AGG_OP_SUM_DEC128], nlm, max_groups=32_000_000)
+        mk("hashagg_and_downloads")
         ngroups = len(first)
         # top-10 by revenue desc: sort the per-group sums (dec128 keys)
         gsum = ctx.upload(acc.reshape(-1))  # [g,1,16] bytes back to device
@@ -112,8 +127,13 @@ def main():
             [ctx.column(gpu.BG_DT_DECIMAL128, gsum, ngroups)], [True],
             ngroups)
         top = perm.download(np.uint32, min(10, ngroups))
+        mk("upload_sort_topk")
         ctx.synchronize()
         wall = time.perf_counter() - t0
+        prev = t0
+        for nm, tt in marks[1:]:
+            print(f"  [{nm}] {(tt - prev) * 1e3:.1f} ms", flush=True)
+            prev = tt
         lo = acc[:, 0, :8].copy().view(np.uint64).reshape(-1)
         hi = acc[:, 0, 8:].copy().view(np.int64).reshape(-1)
         total_rev = (int(hi.astype(object).sum()) << 64) + \
