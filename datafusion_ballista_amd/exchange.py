@@ -47,6 +47,33 @@ def all_to_all_rows(t: torch.Tensor, offsets, world: int):
     me = dist.get_rank()
     out_splits = [int(all_counts[s * world + me]) for s in range(world)]
     out = torch.empty(sum(out_splits), dtype=t.dtype, device=t.device)
+    if dist.get_backend() == "gloo" and \
+            in_splits[me] != out_splits[me]:
+        # gloo's alltoallv requires equal self send/recv lengths; emulate
+        # with point-to-point sends (CPU test path only — the GPU path is
+        # RCCL, whose all_to_all_single handles ragged self splits)
+        src = t.contiguous()
+        in_off = [0]
+        for c in in_splits:
+            in_off.append(in_off[-1] + c)
+        out_off = [0]
+        for c in out_splits:
+            out_off.append(out_off[-1] + c)
+        reqs = []
+        for peer in range(world):
+            if peer == me:
+                out[out_off[me]:out_off[me + 1]] = \
+                    src[in_off[me]:in_off[me + 1]]
+                continue
+            if in_splits[peer]:
+                reqs.append(dist.isend(
+                    src[in_off[peer]:in_off[peer + 1]], peer))
+            if out_splits[peer]:
+                reqs.append(dist.irecv(
+                    out[out_off[peer]:out_off[peer + 1]], peer))
+        for r in reqs:
+            r.wait()
+        return out, out_splits
     dist.all_to_all_single(out, t.contiguous(), out_splits, in_splits)
     return out, out_splits
 

@@ -163,3 +163,81 @@ def test_broadcast_build_side_world2():
             assert p.exitcode == 0
         want = (1000, sum(i * 7 for i in range(1000)))
         assert results[0] == want and results[1] == want
+
+
+def _twophase_worker(rank, world, port, results):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    os.environ.setdefault("GLOO_SOCKET_IFNAME", "lo")
+    torch.distributed.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        from datafusion_ballista_amd import exchange
+        import oracle
+
+        # each rank holds a shard; partial-aggregate it (oracle restatement
+        # stands in for the GPU hashagg — the exchange LOGIC is under test)
+        rng = np.random.default_rng(100 + rank)
+        n = 5_000
+        keys = rng.integers(0, 200, size=n, dtype=np.int64)
+        vals = rng.integers(-10**6, 10**6, size=n, dtype=np.int64)
+        partial = oracle.hashagg([keys], [("sum", vals)], n)
+        pk = np.array([k[0] for k in partial.keys()], dtype=np.int64)
+        pc = np.array([v[0] for v in partial.values()], dtype=np.int64)
+        ps = np.array([v[1][0] for v in partial.values()], dtype=np.int64)
+
+        # route each partial row to the rank owning hash(key) % world
+        # (the reference's Partial -> RepartitionExec(Hash) -> Final shape)
+        h = oracle.hash_columns([("i64", pk)], len(pk))
+        owner = (h % world).astype(np.int64)
+        order = np.argsort(owner, kind="stable")
+        t = torch.tensor(
+            np.stack([pk[order], pc[order], ps[order]],
+                     axis=1)).reshape(-1)  # 3 units per row (flat exchange)
+        offs = np.zeros(world + 1, dtype=np.int64)
+        for o in owner:
+            offs[o + 1] += 1
+        offs = np.cumsum(offs) * 3
+        out, _ = exchange.all_to_all_rows(t, offs, world)
+
+        # final merge on the owning rank
+        rows = out.numpy().reshape(-1, 3)
+        final = {}
+        for k, c, s_ in rows:
+            e = final.setdefault(int(k), [0, 0])
+            e[0] += int(c)
+            e[1] += int(s_)
+        results[rank] = final
+    finally:
+        torch.distributed.destroy_process_group()
+
+
+def test_two_phase_aggregate_exchange_world2():
+    """Distributed Partial->exchange->Final aggregate (2 ranks, gloo): the
+    union of per-rank final groups must equal the oracle's global
+    aggregate over both shards, with every key on exactly one rank."""
+    ctx = mp.get_context("spawn")
+    with ctx.Manager() as mgr:
+        results = mgr.dict()
+        port = 29637
+        ps = [ctx.Process(target=_twophase_worker, args=(r, 2, port, results))
+              for r in range(2)]
+        for p in ps:
+            p.start()
+        for p in ps:
+            p.join(120)
+            assert p.exitcode == 0
+        f0, f1 = results[0], results[1]
+    assert not (set(f0) & set(f1)), "a group landed on both ranks"
+    merged = {**f0, **f1}
+
+    import oracle
+    all_k, all_v = [], []
+    for r in range(2):
+        rng = np.random.default_rng(100 + r)
+        all_k.append(rng.integers(0, 200, size=5_000, dtype=np.int64))
+        all_v.append(rng.integers(-10**6, 10**6, size=5_000, dtype=np.int64))
+    gk = np.concatenate(all_k)
+    gv = np.concatenate(all_v)
+    want = oracle.hashagg([gk], [("sum", gv)], len(gk))
+    want = {k[0]: [v[0], v[1][0]] for k, v in want.items()}
+    assert merged == want
