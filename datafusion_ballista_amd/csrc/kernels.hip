@@ -406,7 +406,8 @@ extern "C" int bg_eval_predicates(const bg_column* cols, int32_t ncols,
 // Segment-aware exclusive scan: block b owns [b*seg, min((b+1)*seg, m));
 // seg_bases (optional) gives each block's running base.  With gridDim.x == 1
 // and seg >= m this is the whole-array scan (small inputs).
-__global__ void k_exclusive_scan_i64(const u64* in, int64_t m, i64* out,
+template <typename IN>
+__global__ void k_exclusive_scan_i64(const IN* in, int64_t m, i64* out,
                                      i64* total, int64_t seg,
                                      const i64* seg_bases) {
   __shared__ i64 sums[BG_BLOCK];
@@ -443,7 +444,8 @@ __global__ void k_exclusive_scan_i64(const u64* in, int64_t m, i64* out,
 }
 
 // per-segment sums for the hierarchical scan's first phase
-__global__ void k_segment_sums_i64(const u64* in, int64_t m, int64_t seg,
+template <typename IN>
+__global__ void k_segment_sums_i64(const IN* in, int64_t m, int64_t seg,
                                    u64* seg_sums, int64_t nseg) {
   __shared__ i64 sums[BG_BLOCK];
   for (int64_t b = blockIdx.x; b < nseg; b += gridDim.x) {
@@ -550,8 +552,9 @@ __global__ void k_scan_lookback(const u64* in, int64_t n, i64* out,
   }
 }
 
-static int scan_exclusive_i64(const u64* d_in, int64_t m, i64* d_out,
-                              i64* d_total) {
+template <typename IN>
+static int scan_exclusive_t(const IN* d_in, int64_t m, i64* d_out,
+                            i64* d_total) {
   if (m <= 0) {
     // nothing to scan: the total must still be DEFINED (pool buffers are
     // recycled — stale bytes here once leaked a phantom match count)
@@ -574,7 +577,7 @@ static int scan_exclusive_i64(const u64* d_in, int64_t m, i64* d_out,
   // LDS block-scan + spin overhead at 4096-element tiles outweighs the
   // 24->16 B/element traffic saving).  Kept behind BG_SCAN_LOOKBACK=1.
   const char* envs = getenv("BG_SCAN_LOOKBACK");
-  if (envs && envs[0] == '1') {
+  if (sizeof(IN) == 8 && envs && envs[0] == '1') {
     const int64_t ntiles = (m + SCAN_TILE - 1) / SCAN_TILE;
     u64* d_state;
     int* d_ticket;
@@ -589,7 +592,8 @@ static int scan_exclusive_i64(const u64* d_in, int64_t m, i64* d_out,
     if (e != hipSuccess) return set_hip_err(e, "scan memset");
     int blocks = (int)bg_imin64(ntiles, BG_MAX_BLOCKS);
     hipLaunchKernelGGL(k_scan_lookback, dim3(blocks), dim3(BG_BLOCK), 0, 0,
-                       d_in, m, d_out, d_total, d_state, d_ticket, ntiles);
+                       (const u64*)(const void*)d_in, m, d_out, d_total,
+                       d_state, d_ticket, ntiles);
     e = hipGetLastError();
     (void)pool_release(d_state);
     (void)pool_release(d_ticket);
@@ -619,6 +623,17 @@ static int scan_exclusive_i64(const u64* d_in, int64_t m, i64* d_out,
   if (e != hipSuccess) return set_hip_err(e, "scan large");
   return BG_OK;
 }
+
+static int scan_exclusive_i64(const u64* d_in, int64_t m, i64* d_out,
+                              i64* d_total) {
+  return scan_exclusive_t<u64>(d_in, m, d_out, d_total);
+}
+
+static int scan_exclusive_u32(const uint32_t* d_in, int64_t m, i64* d_out,
+                              i64* d_total) {
+  return scan_exclusive_t<uint32_t>(d_in, m, d_out, d_total);
+}
+
 
 // ---------------------------------------------------------------------------
 // stable mask compaction
@@ -1382,7 +1397,7 @@ __global__ void k_join_build(const int64_t* keys, const uint8_t* valid,
 __global__ void k_join_count(const int64_t* probe_keys,
                              const uint8_t* probe_valid, int64_t n_probe,
                              const int* head, const ulong2* nodes, u64 mask,
-                             u64* counts) {
+                             uint32_t* counts) {
   const int64_t stride = (int64_t)gridDim.x * blockDim.x;
   const int64_t tid = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
   for (int64_t base = tid; base < n_probe; base += stride * JOIN_ILP) {
@@ -1413,7 +1428,7 @@ __global__ void k_join_count(const int64_t* probe_keys,
     }
 #pragma unroll
     for (int j = 0; j < JOIN_ILP; ++j)
-      if (idx[j] < n_probe) counts[idx[j]] = cnt[j];
+      if (idx[j] < n_probe) counts[idx[j]] = (uint32_t)cnt[j];
   }
 }
 
@@ -1501,10 +1516,10 @@ extern "C" int bg_hashjoin_probe_count(void* handle,
     (void)pool_release(t->probe_offsets);
     t->probe_offsets = nullptr;
   }
-  u64* d_counts;
+  uint32_t* d_counts;
   i64* d_offs;
   i64* d_total;
-  HIP_TRY(pool_malloc((void**)&d_counts, sizeof(u64) * (n ? n : 1)));
+  HIP_TRY(pool_malloc((void**)&d_counts, sizeof(uint32_t) * (n ? n : 1)));
   HIP_TRY(pool_malloc((void**)&d_offs, sizeof(i64) * (n ? n : 1)));
   HIP_TRY(pool_malloc((void**)&d_total, sizeof(i64)));
   int blocks = (int)bg_imin64((n + BG_BLOCK - 1) / BG_BLOCK, BG_MAX_BLOCKS);
@@ -1514,7 +1529,9 @@ extern "C" int bg_hashjoin_probe_count(void* handle,
                      probe_keys->d_validity, n, t->head, t->nodes,
                      t->mask, d_counts);
   {
-    int rc = scan_exclusive_i64(d_counts, n, d_offs, d_total);
+    // u32 counts: chains are < 2^31 long and per-row match counts fit —
+    // halves the count write and the scan's level-0/3 read traffic
+    int rc = scan_exclusive_u32(d_counts, n, d_offs, d_total);
     if (rc != BG_OK) return rc;
   }
   i64 total = 0;
