@@ -559,6 +559,15 @@ class BgBaPageJob(ctypes.Structure):
                 ("d_n_present", ctypes.c_void_p)]
 
 
+class BgDeltaBpJob(ctypes.Structure):
+    _fields_ = [("d_page", ctypes.c_void_p), ("d_out", ctypes.c_void_p),
+                ("page_len", ctypes.c_int64), ("nvals", ctypes.c_int64),
+                ("esz", ctypes.c_int64),
+                ("has_def", ctypes.c_int32), ("_pad", ctypes.c_int32),
+                ("d_vidx", ctypes.c_void_p),
+                ("d_n_present", ctypes.c_void_p)]
+
+
 class BgDefLevelsJob(ctypes.Structure):
     _fields_ = [("d_page", ctypes.c_void_p), ("d_vidx", ctypes.c_void_p),
                 ("d_valid_out", ctypes.c_void_p),

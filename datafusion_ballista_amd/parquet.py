@@ -223,7 +223,7 @@ class GpuParquetColumnReader:
                     dph = h.get(5, {})
                     nvals = dph.get(1, 0)
                     enc = dph.get(2, 0)
-                    if enc not in (0, 2, 8):
+                    if enc not in (0, 2, 5, 8):
                         raise RuntimeError(
                             f"encoding {enc}: not GPU-decodable yet")
                 elif ptype == 3:
@@ -239,7 +239,7 @@ class GpuParquetColumnReader:
                         raise RuntimeError(
                             "DATA_PAGE_V2 repetition levels (nested): "
                             "not GPU-decodable yet")
-                    if enc not in (0, 2, 8):
+                    if enc not in (0, 2, 5, 8):
                         raise RuntimeError(
                             f"encoding {enc}: not GPU-decodable yet")
                     values_csz = csz - def_len
@@ -268,6 +268,7 @@ class GpuParquetColumnReader:
         else:
             out = ctx.alloc(max(total_values * dst_esz, dst_esz))
         extracts = []        # PLAIN: (scratch_off, usz, dst_off, nvals, pidx)
+        deltas = []          # DELTA_BINARY_PACKED pages (same tuple)
         dict_runs = {}       # rg -> {"dict": (soff, usz, ndict), "pages": []}
         got_values = 0
         snappy_jobs = []
@@ -294,7 +295,12 @@ class GpuParquetColumnReader:
             elif ptype in (0, 3):
                 pidx = len(data_pages)
                 data_pages.append((soff, usz, got_values, nvals))
-                if enc == 0:
+                if enc == 5:
+                    if phys not in ("INT32", "INT64"):
+                        raise RuntimeError(
+                            "DELTA_BINARY_PACKED: INT32/INT64 only")
+                    deltas.append((soff, usz, got_values, nvals, pidx))
+                elif enc == 0:
                     extracts.append((soff, usz, got_values, nvals, pidx))
                 else:
                     if rg not in dict_runs or dict_runs[rg]["dict"] is None:
@@ -408,6 +414,19 @@ class GpuParquetColumnReader:
                     if mode else None)
             gpu._check(ctx.L.bg_page_extract_batch(
                 jobs, ctypes.c_int64(len(extracts))), "bg_page_extract_batch")
+        if deltas:
+            djobs2 = (gpu.BgDeltaBpJob * len(deltas))()
+            for i, (soff, usz, dst_off, nvals, pidx) in enumerate(deltas):
+                djobs2[i] = gpu.BgDeltaBpJob(
+                    page_ptr(scratch, soff).value,
+                    ctypes.c_void_p(out.ptr.value + dst_off * dst_esz).value,
+                    usz, nvals, dst_esz, mode, 0,
+                    ctypes.c_void_p(vidx.ptr.value + 4 * dst_off).value
+                    if mode else None,
+                    ctypes.c_void_p(npres.ptr.value + 8 * pidx).value
+                    if mode else None)
+            gpu._check(ctx.L.bg_delta_bp_batch(
+                djobs2, ctypes.c_int64(len(deltas))), "bg_delta_bp_batch")
 
         all_dict_pages = [(rg, pg) for rg, d in dict_runs.items()
                           for pg in d["pages"]]

@@ -372,6 +372,24 @@ int bg_ba_materialize(const int64_t* d_lens, const int64_t* d_srcaddr,
                       int64_t n, int32_t* d_offs32 /* n+1 */,
                       uint8_t* d_data, int64_t data_cap, int64_t* out_total);
 
+/* DELTA_BINARY_PACKED (encoding 5) INT32/INT64 pages — the reference's
+ * parquet-rs DataPageV2 writer default for integers (spec Encodings.md
+ * "Delta Encoding"; decoding.rs DeltaBitPackDecoder).  Lane-0 serial per
+ * page (the prefix chain is sequential), pages concurrent across waves;
+ * nullable slots scatter through the def-level vidx map. */
+typedef struct {
+  const void* d_page;
+  void* d_out;           /* i32/i64 column slice */
+  int64_t page_len;
+  int64_t nvals;
+  int64_t esz;           /* 4 or 8 */
+  int32_t has_def;       /* 0 none, 2 nullable */
+  int32_t _pad;
+  const uint32_t* d_vidx;
+  const int64_t* d_n_present;
+} bg_delta_bp_job;
+int bg_delta_bp_batch(const void* h_jobs, int64_t njobs);
+
 /* Device LZ4 block compression (the GPU shuffle codec's compress half,
  * SURVEY.md §8f row 3): 64 KiB blocks, one wave per block.  d_out_slots
  * holds nblocks slots of 65544 B; h_block_sizes[i] = compressed size, or
