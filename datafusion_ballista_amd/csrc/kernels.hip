@@ -3107,6 +3107,97 @@ extern "C" int bg_def_levels_batch(const void* h_jobs, int64_t njobs) {
 // sequential); pages decode concurrently across waves.  Nullable slots
 // scatter through vidx like the other extractors.
 // ---------------------------------------------------------------------------
+// Reusable serial reader for one DELTA_BINARY_PACKED integer stream
+// (consumed by encodings 5, 6 and 7).  One lane; returns false on
+// malformed input.  After `count` values, `p` rests at the first byte
+// past the stream (writers pad the final miniblock to full width).
+struct DbpStream {
+  const uint8_t* p;
+  const uint8_t* pend;
+  u64 block_size = 0, mb_per_block = 0, total = 0;
+  int64_t mb_vals = 0;
+  i64 value = 0;     // last emitted value
+  i64 min_delta = 0;
+  const uint8_t* bws = nullptr;  // current block's bit widths
+  u64 mb = 0;        // miniblock index within block (mb_per_block = fresh)
+  int64_t t = 0;     // value index within miniblock
+  int64_t produced = 0;
+  int bw = 0;
+  int64_t mb_bytes = 0;
+
+  __device__ bool varint(u64* out_v) {
+    u64 v = 0;
+    int sh = 0;
+    while (p < pend) {
+      const uint8_t b = *p++;
+      v |= (u64)(b & 0x7f) << sh;
+      if (!(b & 0x80)) { *out_v = v; return true; }
+      sh += 7;
+    }
+    return false;
+  }
+  __device__ bool zigzag(i64* out_v) {
+    u64 v;
+    if (!varint(&v)) return false;
+    *out_v = (i64)(v >> 1) ^ -(i64)(v & 1);
+    return true;
+  }
+  __device__ bool init(const uint8_t* start, const uint8_t* end) {
+    p = start;
+    pend = end;
+    i64 first;
+    if (!varint(&block_size) || !varint(&mb_per_block) || !varint(&total) ||
+        !zigzag(&first))
+      return false;
+    if (!mb_per_block || !block_size || block_size % mb_per_block)
+      return false;
+    mb_vals = (int64_t)(block_size / mb_per_block);
+    value = first;
+    mb = mb_per_block;  // force new block on first next_after_first
+    t = 0;
+    produced = 1;       // `value` holds the first value already
+    return true;
+  }
+  // advance to the next value (call total-1 times after init)
+  __device__ bool next() {
+    if (t >= mb_vals) { t = 0; p += mb_bytes; ++mb; }
+    if (mb >= mb_per_block) {
+      if (!zigzag(&min_delta)) return false;
+      if (p + mb_per_block > pend) return false;
+      bws = p;
+      p += mb_per_block;
+      mb = 0;
+      t = 0;
+    }
+    if (t == 0) {
+      bw = bws[mb];
+      if (bw > 64) return false;
+      mb_bytes = (int64_t)bw * mb_vals / 8;
+      if (p + mb_bytes > pend) return false;
+    }
+    u64 d = 0;
+    if (bw) {
+      const int64_t bit = t * bw;
+      const int64_t byte = bit >> 3;
+      const int sh = (int)(bit & 7);
+      for (int b2 = 0; b2 * 8 < bw + sh; ++b2)
+        if (byte + b2 < mb_bytes) d |= (u64)p[byte + b2] << (8 * b2);
+      d = (d >> sh) & (bw == 64 ? ~0ull : ((1ull << bw) - 1));
+    }
+    ++t;
+    value += min_delta + (i64)d;
+    ++produced;
+    return true;
+  }
+  // position p just past the stream (after consuming exactly `total`)
+  __device__ bool finish() {
+    while (produced < (int64_t)total)
+      if (!next()) return false;
+    if (t > 0 || total == 1) p += mb_bytes;  // skip the last miniblock data
+    return true;
+  }
+};
+
 struct DeltaBpJob {
   const uint8_t* page;   // page start ([u32 dlen][levels] when has_def)
   uint8_t* out;          // i32/i64 column slice (slot-addressed)
@@ -3130,43 +3221,17 @@ __device__ void k_delta_bp_body(const DeltaBpJob& job, int* err) {
     p += 4 + dlen;
     if (p > pend) { atomicExch(err, 3); return; }
   }
-  auto varint = [&](u64* out_v) -> bool {
-    u64 v = 0;
-    int sh = 0;
-    while (p < pend) {
-      const uint8_t b = *p++;
-      v |= (u64)(b & 0x7f) << sh;
-      if (!(b & 0x80)) { *out_v = v; return true; }
-      sh += 7;
-    }
-    return false;
-  };
-  auto zigzag = [&](i64* out_v) -> bool {
-    u64 v;
-    if (!varint(&v)) return false;
-    *out_v = (i64)(v >> 1) ^ -(i64)(v & 1);
-    return true;
-  };
-  u64 block_size, mb_per_block, total;
-  i64 first;
-  if (!varint(&block_size) || !varint(&mb_per_block) || !varint(&total) ||
-      !zigzag(&first)) { atomicExch(err, 3); return; }
-  if (!mb_per_block || !block_size || block_size % mb_per_block) {
-    atomicExch(err, 3);
+  DbpStream st;
+  if (!st.init(p, pend)) { atomicExch(err, 3); return; }
+  const int64_t want = job.has_def == 2 ? *job.n_present : job.nvals;
+  if ((int64_t)st.total < want || want == 0) {
+    if (want != 0) atomicExch(err, 3);
     return;
   }
-  const int64_t mb_vals = (int64_t)(block_size / mb_per_block);
-  const int64_t want = job.has_def == 2 ? *job.n_present : job.nvals;
-  if ((int64_t)total < want) { atomicExch(err, 3); return; }
-
-  int64_t emitted = 0;   // values produced
-  int64_t slot = 0;      // next output slot
-  i64 value = first;
+  int64_t slot = 0;
   auto emit = [&](i64 v) {
-    if (job.has_def == 2) {
+    if (job.has_def == 2)
       while (slot < job.nvals && job.vidx[slot] == 0xffffffffu) ++slot;
-      if (slot >= job.nvals) return;
-    }
     if (slot < job.nvals) {
       if (job.esz == 8)
         *(int64_t*)(job.out + slot * 8) = v;
@@ -3175,37 +3240,171 @@ __device__ void k_delta_bp_body(const DeltaBpJob& job, int* err) {
       ++slot;
     }
   };
-  emit(value);
-  ++emitted;
-  while (emitted < want) {
-    i64 min_delta;
-    if (!zigzag(&min_delta)) { atomicExch(err, 3); return; }
-    if (p + mb_per_block > pend) { atomicExch(err, 3); return; }
-    const uint8_t* bws = p;
-    p += mb_per_block;
-    for (u64 mb = 0; mb < mb_per_block && emitted < want; ++mb) {
-      const int bw = bws[mb];
-      if (bw > 64) { atomicExch(err, 3); return; }
-      const int64_t mb_bytes = (int64_t)bw * mb_vals / 8;
-      if (p + mb_bytes > pend) { atomicExch(err, 3); return; }
-      for (int64_t t = 0; t < mb_vals && emitted < want; ++t) {
-        u64 d = 0;
-        if (bw) {
-          const int64_t bit = t * bw;
-          const int64_t byte = bit >> 3;
-          const int sh = (int)(bit & 7);
-          // assemble up to bw+7 bits little-endian
-          for (int b2 = 0; b2 * 8 < bw + sh; ++b2)
-            if (byte + b2 < mb_bytes) d |= (u64)p[byte + b2] << (8 * b2);
-          d = (d >> sh) & (bw == 64 ? ~0ull : ((1ull << bw) - 1));
-        }
-        value += min_delta + (i64)d;
-        emit(value);
-        ++emitted;
-      }
-      p += mb_bytes;
-    }
+  emit(st.value);
+  for (int64_t k2 = 1; k2 < want; ++k2) {
+    if (!st.next()) { atomicExch(err, 3); return; }
+    emit(st.value);
   }
+}
+
+// DELTA_LENGTH_BYTE_ARRAY (6): one DBP stream of lengths, then the
+// concatenated bytes — lengths + absolute source addresses feed the
+// shared bg_ba_materialize path.  DELTA_BYTE_ARRAY (7): DBP prefix
+// lengths + DBP suffix lengths + suffix bytes; pass 1 records total
+// lengths (srcaddr 0 = "reconstructed later"), pass 2 rebuilds each
+// string from its predecessor's prefix + its suffix at the final
+// offsets (lane-0 serial per page; strings share prefixes only within
+// a page).
+struct DeltaBaJob {
+  const uint8_t* page;
+  int64_t* lens_out;     // slot lengths
+  int64_t* srcaddr_out;  // pass 1: byte addresses (enc 6) or 0 (enc 7)
+  const int32_t* offs32; // pass 2 (enc 7): final column offsets
+  uint8_t* data_out;     // pass 2 (enc 7): column data base
+  int64_t page_len;
+  int64_t nvals;
+  int32_t has_def;
+  int32_t enc;           // 6 or 7
+  const uint32_t* vidx;
+  const int64_t* n_present;
+};
+
+__device__ void k_delta_ba_body(const DeltaBaJob& job, int pass, int* err) {
+  if (lane_id() != 0) return;
+  const uint8_t* p = job.page;
+  const uint8_t* pend = p + job.page_len;
+  if (job.has_def) {
+    if (job.page_len < 4) { atomicExch(err, 3); return; }
+    const uint32_t dlen = (uint32_t)p[0] | ((uint32_t)p[1] << 8) |
+                          ((uint32_t)p[2] << 16) | ((uint32_t)p[3] << 24);
+    p += 4 + dlen;
+    if (p > pend) { atomicExch(err, 3); return; }
+  }
+  const int64_t want = job.has_def == 2 ? *job.n_present : job.nvals;
+  if (pass == 1)
+    for (int64_t t = 0; t < job.nvals; ++t) {
+      job.lens_out[t] = 0;
+      job.srcaddr_out[t] = 0;
+    }
+  if (want == 0) return;
+  if (job.enc == 6) {
+    DbpStream lens;
+    if (!lens.init(p, pend) || (int64_t)lens.total < want) {
+      atomicExch(err, 3);
+      return;
+    }
+    // walk lengths; bytes follow the stream — position after via finish()
+    // (needs a second pass over the stream: re-init and emit)
+    DbpStream probe = lens;
+    if (!probe.finish()) { atomicExch(err, 3); return; }
+    const uint8_t* bytes = probe.p;
+    int64_t slot = 0, cursor = 0;
+    auto emit = [&](i64 ln) {
+      if (job.has_def == 2)
+        while (slot < job.nvals && job.vidx[slot] == 0xffffffffu) ++slot;
+      if (slot < job.nvals) {
+        job.lens_out[slot] = ln;
+        job.srcaddr_out[slot] = (int64_t)(uintptr_t)(bytes + cursor);
+        ++slot;
+      }
+      cursor += ln;
+    };
+    emit(lens.value);
+    for (int64_t k2 = 1; k2 < want; ++k2) {
+      if (!lens.next()) { atomicExch(err, 3); return; }
+      emit(lens.value);
+    }
+    if (bytes + cursor > pend) { atomicExch(err, 3); return; }
+    return;
+  }
+  // enc 7: prefix lens stream, suffix lens stream, suffix bytes
+  DbpStream pre;
+  if (!pre.init(p, pend) || (int64_t)pre.total < want) {
+    atomicExch(err, 3);
+    return;
+  }
+  DbpStream pre_probe = pre;
+  if (!pre_probe.finish()) { atomicExch(err, 3); return; }
+  DbpStream suf;
+  if (!suf.init(pre_probe.p, pend) || (int64_t)suf.total < want) {
+    atomicExch(err, 3);
+    return;
+  }
+  DbpStream suf_probe = suf;
+  if (!suf_probe.finish()) { atomicExch(err, 3); return; }
+  const uint8_t* sbytes = suf_probe.p;
+  int64_t slot = 0, cursor = 0;
+  int64_t prev_off = -1;  // output offset of the previous string (pass 2)
+  int64_t prev_len = 0;
+  bool first = true;
+  auto step = [&](i64 plen, i64 slen) -> bool {
+    if (plen < 0 || slen < 0) return false;
+    if (first && plen != 0) return false;  // nothing to prefix from
+    if (!first && plen > prev_len) return false;
+    if (job.has_def == 2)
+      while (slot < job.nvals && job.vidx[slot] == 0xffffffffu) ++slot;
+    if (slot < job.nvals) {
+      if (pass == 1) {
+        job.lens_out[slot] = plen + slen;
+        // srcaddr stays 0: k_ba_copy skips, pass 2 reconstructs
+      } else {
+        const int64_t off = (int64_t)job.offs32[slot];
+        uint8_t* dst = job.data_out + off;
+        if (!first)
+          for (int64_t b = 0; b < plen; ++b)
+            dst[b] = job.data_out[prev_off + b];
+        for (int64_t b = 0; b < slen; ++b) dst[plen + b] = sbytes[cursor + b];
+        prev_off = off;
+      }
+      prev_len = plen + slen;  // both passes: the prefix bound check
+      ++slot;
+    }
+    cursor += slen;
+    first = false;
+    return true;
+  };
+  if (!step(pre.value, suf.value)) { atomicExch(err, 3); return; }
+  for (int64_t k2 = 1; k2 < want; ++k2) {
+    if (!pre.next() || !suf.next()) { atomicExch(err, 3); return; }
+    if (!step(pre.value, suf.value)) { atomicExch(err, 3); return; }
+  }
+  if (sbytes + cursor > pend) atomicExch(err, 3);
+}
+
+__global__ void k_delta_ba_batch(const DeltaBaJob* jobs, int64_t njobs,
+                                 int pass, int* err) {
+  const int64_t wave_global =
+      ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) / BG_WAVE;
+  const int64_t nwaves = ((int64_t)gridDim.x * blockDim.x) / BG_WAVE;
+  for (int64_t j = wave_global; j < njobs; j += nwaves)
+    k_delta_ba_body(jobs[j], pass, err);
+}
+
+extern "C" int bg_delta_ba_batch(const void* h_jobs, int64_t njobs,
+                                 int32_t pass) {
+  REQUIRE_INIT();
+  DeltaBaJob* d_jobs;
+  int* d_err;
+  HIP_TRY(pool_malloc((void**)&d_jobs,
+                      sizeof(DeltaBaJob) * (njobs ? njobs : 1)));
+  HIP_TRY(pool_malloc((void**)&d_err, sizeof(int)));
+  HIP_TRY(hipMemset(d_err, 0, sizeof(int)));
+  HIP_TRY(hipMemcpy(d_jobs, h_jobs, sizeof(DeltaBaJob) * njobs,
+                    hipMemcpyHostToDevice));
+  const int waves_per_block = BG_BLOCK / BG_WAVE;
+  int blocks = (int)bg_imin64((njobs + waves_per_block - 1) / waves_per_block,
+                              BG_MAX_BLOCKS);
+  if (blocks == 0) blocks = 1;
+  hipLaunchKernelGGL(k_delta_ba_batch, dim3(blocks), dim3(BG_BLOCK), 0, 0,
+                     d_jobs, njobs, (int)pass, d_err);
+  HIP_TRY(hipGetLastError());
+  int err = 0;
+  HIP_TRY(hipMemcpy(&err, d_err, sizeof(int), hipMemcpyDeviceToHost));
+  (void)pool_release(d_jobs);
+  (void)pool_release(d_err);
+  if (err)
+    return set_err(BG_ERR_INVALID, "bg_delta_ba_batch: malformed page");
+  return BG_OK;
 }
 
 __global__ void k_delta_bp_batch(const DeltaBpJob* jobs, int64_t njobs,
@@ -3371,6 +3570,7 @@ __global__ void k_ba_copy(const int64_t* srcaddr, const int64_t* offs64,
     offs32[s] = (int32_t)o;
     const int64_t len = offs64[s + 1] - o;
     const uint8_t* src = (const uint8_t*)(uintptr_t)srcaddr[s];
+    if (!src) continue;  // DELTA_BYTE_ARRAY slots: rebuilt by pass 2
     for (int64_t b = 0; b < len; ++b) data[o + b] = src[b];
   }
 }

@@ -568,6 +568,18 @@ class BgDeltaBpJob(ctypes.Structure):
                 ("d_n_present", ctypes.c_void_p)]
 
 
+class BgDeltaBaJob(ctypes.Structure):
+    _fields_ = [("d_page", ctypes.c_void_p),
+                ("d_lens_out", ctypes.c_void_p),
+                ("d_srcaddr_out", ctypes.c_void_p),
+                ("d_offs32", ctypes.c_void_p),
+                ("d_data_out", ctypes.c_void_p),
+                ("page_len", ctypes.c_int64), ("nvals", ctypes.c_int64),
+                ("has_def", ctypes.c_int32), ("enc", ctypes.c_int32),
+                ("d_vidx", ctypes.c_void_p),
+                ("d_n_present", ctypes.c_void_p)]
+
+
 class BgDefLevelsJob(ctypes.Structure):
     _fields_ = [("d_page", ctypes.c_void_p), ("d_vidx", ctypes.c_void_p),
                 ("d_valid_out", ctypes.c_void_p),

@@ -223,7 +223,7 @@ class GpuParquetColumnReader:
                     dph = h.get(5, {})
                     nvals = dph.get(1, 0)
                     enc = dph.get(2, 0)
-                    if enc not in (0, 2, 5, 8):
+                    if enc not in (0, 2, 5, 6, 7, 8):
                         raise RuntimeError(
                             f"encoding {enc}: not GPU-decodable yet")
                 elif ptype == 3:
@@ -239,7 +239,7 @@ class GpuParquetColumnReader:
                         raise RuntimeError(
                             "DATA_PAGE_V2 repetition levels (nested): "
                             "not GPU-decodable yet")
-                    if enc not in (0, 2, 5, 8):
+                    if enc not in (0, 2, 5, 6, 7, 8):
                         raise RuntimeError(
                             f"encoding {enc}: not GPU-decodable yet")
                     values_csz = csz - def_len
@@ -269,6 +269,7 @@ class GpuParquetColumnReader:
             out = ctx.alloc(max(total_values * dst_esz, dst_esz))
         extracts = []        # PLAIN: (scratch_off, usz, dst_off, nvals, pidx)
         deltas = []          # DELTA_BINARY_PACKED pages (same tuple)
+        delta_ba = []        # DELTA(_LENGTH)_BYTE_ARRAY pages (+enc)
         dict_runs = {}       # rg -> {"dict": (soff, usz, ndict), "pages": []}
         got_values = 0
         snappy_jobs = []
@@ -300,6 +301,12 @@ class GpuParquetColumnReader:
                         raise RuntimeError(
                             "DELTA_BINARY_PACKED: INT32/INT64 only")
                     deltas.append((soff, usz, got_values, nvals, pidx))
+                elif enc in (6, 7):
+                    if phys != "BYTE_ARRAY":
+                        raise RuntimeError(
+                            "DELTA(_LENGTH)_BYTE_ARRAY: BYTE_ARRAY only")
+                    delta_ba.append((soff, usz, got_values, nvals, pidx,
+                                     enc))
                 elif enc == 0:
                     extracts.append((soff, usz, got_values, nvals, pidx))
                 else:
@@ -414,6 +421,23 @@ class GpuParquetColumnReader:
                     if mode else None)
             gpu._check(ctx.L.bg_page_extract_batch(
                 jobs, ctypes.c_int64(len(extracts))), "bg_page_extract_batch")
+        if delta_ba:
+            jobs7 = (gpu.BgDeltaBaJob * len(delta_ba))()
+            for i, (soff, usz, dst_off, nvals, pidx, enc) in \
+                    enumerate(delta_ba):
+                jobs7[i] = gpu.BgDeltaBaJob(
+                    page_ptr(scratch, soff).value,
+                    ctypes.c_void_p(ba_lens.ptr.value + 8 * dst_off).value,
+                    ctypes.c_void_p(ba_srcaddr.ptr.value
+                                    + 8 * dst_off).value,
+                    None, None, usz, nvals, mode, enc,
+                    ctypes.c_void_p(vidx.ptr.value + 4 * dst_off).value
+                    if mode else None,
+                    ctypes.c_void_p(npres.ptr.value + 8 * pidx).value
+                    if mode else None)
+            gpu._check(ctx.L.bg_delta_ba_batch(
+                jobs7, ctypes.c_int64(len(delta_ba)), 1),
+                "bg_delta_ba_batch(1)")
         if deltas:
             djobs2 = (gpu.BgDeltaBpJob * len(deltas))()
             for i, (soff, usz, dst_off, nvals, pidx) in enumerate(deltas):
@@ -539,6 +563,27 @@ class GpuParquetColumnReader:
                 ba_lens.ptr, ba_srcaddr.ptr, ctypes.c_int64(total_values),
                 offs32.ptr, data.ptr, ctypes.c_int64(cap),
                 ctypes.byref(tot)), "bg_ba_materialize")
+            enc7 = [d for d in delta_ba if d[5] == 7]
+            if enc7:
+                jobs8 = (gpu.BgDeltaBaJob * len(enc7))()
+                for i, (soff, usz, dst_off, nvals, pidx, enc) in \
+                        enumerate(enc7):
+                    jobs8[i] = gpu.BgDeltaBaJob(
+                        page_ptr(scratch, soff).value,
+                        ctypes.c_void_p(ba_lens.ptr.value
+                                        + 8 * dst_off).value,
+                        ctypes.c_void_p(ba_srcaddr.ptr.value
+                                        + 8 * dst_off).value,
+                        ctypes.c_void_p(offs32.ptr.value
+                                        + 4 * dst_off).value,
+                        data.ptr.value, usz, nvals, mode, enc,
+                        ctypes.c_void_p(vidx.ptr.value + 4 * dst_off).value
+                        if mode else None,
+                        ctypes.c_void_p(npres.ptr.value + 8 * pidx).value
+                        if mode else None)
+                gpu._check(ctx.L.bg_delta_ba_batch(
+                    jobs8, ctypes.c_int64(len(enc7)), 2),
+                    "bg_delta_ba_batch(2)")
             ctx.synchronize()
             return ((offs32, data, tot.value), total_values, "BYTE_ARRAY",
                     valid)

@@ -390,6 +390,27 @@ typedef struct {
 } bg_delta_bp_job;
 int bg_delta_bp_batch(const void* h_jobs, int64_t njobs);
 
+/* DELTA_LENGTH_BYTE_ARRAY (6) / DELTA_BYTE_ARRAY (7) string pages — the
+ * reference's parquet-rs V2 writer defaults for BYTE_ARRAY (Encodings.md;
+ * decoding.rs DeltaLengthByteArrayDecoder / DeltaByteArrayDecoder).
+ * pass 1 fills slot lengths (+ source addresses for enc 6; 0 for enc 7);
+ * after bg_ba_materialize lays out the column, pass 2 (enc 7 only)
+ * rebuilds each string from its predecessor's prefix + its suffix. */
+typedef struct {
+  const void* d_page;
+  int64_t* d_lens_out;
+  int64_t* d_srcaddr_out;
+  const int32_t* d_offs32;  /* pass 2 */
+  uint8_t* d_data_out;      /* pass 2 */
+  int64_t page_len;
+  int64_t nvals;
+  int32_t has_def;
+  int32_t enc; /* 6 or 7 */
+  const uint32_t* d_vidx;
+  const int64_t* d_n_present;
+} bg_delta_ba_job;
+int bg_delta_ba_batch(const void* h_jobs, int64_t njobs, int32_t pass);
+
 /* Device LZ4 block compression (the GPU shuffle codec's compress half,
  * SURVEY.md §8f row 3): 64 KiB blocks, one wave per block.  d_out_slots
  * holds nblocks slots of 65544 B; h_block_sizes[i] = compressed size, or
