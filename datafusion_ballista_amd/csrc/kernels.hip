@@ -3636,6 +3636,75 @@ struct PageExtractJob {
   const int64_t* n_present;  // mode 2: value count in this page
 };
 
+// ---------------------------------------------------------------------------
+// BYTE_STREAM_SPLIT (encoding 9, fixed-width values): the page stores the
+// k-th byte of every value contiguously (k streams of n_present bytes) —
+// a parallel byte transpose reassembles values; nullable slots scatter
+// through vidx (spec Encodings.md; the reference's parquet crate
+// byte_stream_split decoder).
+// ---------------------------------------------------------------------------
+__global__ void k_bss_batch(const PageExtractJob* jobs, int64_t njobs,
+                            int* err) {
+  for (int64_t j = blockIdx.x; j < njobs; j += gridDim.x) {
+    const PageExtractJob job = jobs[j];
+    const uint8_t* page = job.page;
+    int64_t voff = 0;
+    if (job.has_def) {
+      if (job.page_len < 4) {
+        if (threadIdx.x == 0) atomicExch(err, 3);
+        continue;
+      }
+      const uint32_t dlen = (uint32_t)page[0] | ((uint32_t)page[1] << 8) |
+                            ((uint32_t)page[2] << 16) |
+                            ((uint32_t)page[3] << 24);
+      voff = 4 + (int64_t)dlen;
+    }
+    const int64_t npres =
+        job.has_def == 2 ? *job.n_present : job.nvals;
+    if (voff + npres * job.src_esz > job.page_len) {
+      if (threadIdx.x == 0) atomicExch(err, 3);
+      continue;
+    }
+    const uint8_t* v = page + voff;
+    for (int64_t s2 = threadIdx.x; s2 < job.nvals; s2 += blockDim.x) {
+      int64_t k;
+      if (job.has_def == 2) {
+        const uint32_t ix = job.vidx[s2];
+        if (ix == 0xffffffffu) continue;
+        k = (int64_t)ix;
+      } else {
+        k = s2;
+      }
+      uint8_t* dst = job.out + s2 * job.src_esz;
+      for (int64_t b = 0; b < job.src_esz; ++b)
+        dst[b] = v[b * npres + k];
+    }
+  }
+}
+
+extern "C" int bg_bss_batch(const void* h_jobs, int64_t njobs) {
+  REQUIRE_INIT();
+  PageExtractJob* d_jobs;
+  int* d_err;
+  HIP_TRY(pool_malloc((void**)&d_jobs,
+                      sizeof(PageExtractJob) * (njobs ? njobs : 1)));
+  HIP_TRY(pool_malloc((void**)&d_err, sizeof(int)));
+  HIP_TRY(hipMemset(d_err, 0, sizeof(int)));
+  HIP_TRY(hipMemcpy(d_jobs, h_jobs, sizeof(PageExtractJob) * njobs,
+                    hipMemcpyHostToDevice));
+  int blocks = (int)bg_imin64(njobs, BG_MAX_BLOCKS);
+  if (blocks == 0) blocks = 1;
+  hipLaunchKernelGGL(k_bss_batch, dim3(blocks), dim3(BG_BLOCK), 0, 0, d_jobs,
+                     njobs, d_err);
+  HIP_TRY(hipGetLastError());
+  int err = 0;
+  HIP_TRY(hipMemcpy(&err, d_err, sizeof(int), hipMemcpyDeviceToHost));
+  (void)pool_release(d_jobs);
+  (void)pool_release(d_err);
+  if (err) return set_err(BG_ERR_INVALID, "bg_bss_batch: malformed page");
+  return BG_OK;
+}
+
 __global__ void k_page_extract_batch(const PageExtractJob* jobs, int64_t njobs,
                                      int* err) {
   for (int64_t j = blockIdx.x; j < njobs; j += gridDim.x) {

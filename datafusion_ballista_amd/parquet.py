@@ -223,7 +223,7 @@ class GpuParquetColumnReader:
                     dph = h.get(5, {})
                     nvals = dph.get(1, 0)
                     enc = dph.get(2, 0)
-                    if enc not in (0, 2, 5, 6, 7, 8):
+                    if enc not in (0, 2, 5, 6, 7, 8, 9):
                         raise RuntimeError(
                             f"encoding {enc}: not GPU-decodable yet")
                 elif ptype == 3:
@@ -239,7 +239,7 @@ class GpuParquetColumnReader:
                         raise RuntimeError(
                             "DATA_PAGE_V2 repetition levels (nested): "
                             "not GPU-decodable yet")
-                    if enc not in (0, 2, 5, 6, 7, 8):
+                    if enc not in (0, 2, 5, 6, 7, 8, 9):
                         raise RuntimeError(
                             f"encoding {enc}: not GPU-decodable yet")
                     values_csz = csz - def_len
@@ -270,6 +270,7 @@ class GpuParquetColumnReader:
         extracts = []        # PLAIN: (scratch_off, usz, dst_off, nvals, pidx)
         deltas = []          # DELTA_BINARY_PACKED pages (same tuple)
         delta_ba = []        # DELTA(_LENGTH)_BYTE_ARRAY pages (+enc)
+        bss = []             # BYTE_STREAM_SPLIT pages
         dict_runs = {}       # rg -> {"dict": (soff, usz, ndict), "pages": []}
         got_values = 0
         snappy_jobs = []
@@ -301,6 +302,11 @@ class GpuParquetColumnReader:
                         raise RuntimeError(
                             "DELTA_BINARY_PACKED: INT32/INT64 only")
                     deltas.append((soff, usz, got_values, nvals, pidx))
+                elif enc == 9:
+                    if flba or phys == "BYTE_ARRAY":
+                        raise RuntimeError(
+                            "BYTE_STREAM_SPLIT: fixed-width plain types")
+                    bss.append((soff, usz, got_values, nvals, pidx))
                 elif enc in (6, 7):
                     if phys != "BYTE_ARRAY":
                         raise RuntimeError(
@@ -438,6 +444,19 @@ class GpuParquetColumnReader:
             gpu._check(ctx.L.bg_delta_ba_batch(
                 jobs7, ctypes.c_int64(len(delta_ba)), 1),
                 "bg_delta_ba_batch(1)")
+        if bss:
+            jobs9 = (gpu.BgPageExtractJob * len(bss))()
+            for i, (soff, usz, dst_off, nvals, pidx) in enumerate(bss):
+                jobs9[i] = gpu.BgPageExtractJob(
+                    page_ptr(scratch, soff).value,
+                    ctypes.c_void_p(out.ptr.value + dst_off * dst_esz).value,
+                    usz, nvals, src_esz, mode, 0,
+                    ctypes.c_void_p(vidx.ptr.value + 4 * dst_off).value
+                    if mode else None,
+                    ctypes.c_void_p(npres.ptr.value + 8 * pidx).value
+                    if mode else None)
+            gpu._check(ctx.L.bg_bss_batch(
+                jobs9, ctypes.c_int64(len(bss))), "bg_bss_batch")
         if deltas:
             djobs2 = (gpu.BgDeltaBpJob * len(deltas))()
             for i, (soff, usz, dst_off, nvals, pidx) in enumerate(deltas):

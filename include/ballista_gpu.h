@@ -411,6 +411,11 @@ typedef struct {
 } bg_delta_ba_job;
 int bg_delta_ba_batch(const void* h_jobs, int64_t njobs, int32_t pass);
 
+/* BYTE_STREAM_SPLIT (encoding 9, fixed-width): parallel byte transpose
+ * of the k per-byte streams back into values (bg_page_extract_job fields;
+ * flba_reverse unused). */
+int bg_bss_batch(const void* h_jobs, int64_t njobs);
+
 /* Device LZ4 block compression (the GPU shuffle codec's compress half,
  * SURVEY.md §8f row 3): 64 KiB blocks, one wave per block.  d_out_slots
  * holds nblocks slots of 65544 B; h_block_sizes[i] = compressed size, or
