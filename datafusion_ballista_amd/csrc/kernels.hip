@@ -3180,9 +3180,13 @@ struct DbpStream {
       const int64_t bit = t * bw;
       const int64_t byte = bit >> 3;
       const int sh = (int)(bit & 7);
+      // bw + sh can exceed 64 (full-width deltas off a bit boundary):
+      // assemble through u128 so no high bits are lost
+      u128 acc = 0;
       for (int b2 = 0; b2 * 8 < bw + sh; ++b2)
-        if (byte + b2 < mb_bytes) d |= (u64)p[byte + b2] << (8 * b2);
-      d = (d >> sh) & (bw == 64 ? ~0ull : ((1ull << bw) - 1));
+        if (byte + b2 < mb_bytes) acc |= (u128)p[byte + b2] << (8 * b2);
+      d = (u64)(acc >> sh);
+      if (bw < 64) d &= (1ull << bw) - 1;
     }
     ++t;
     value += min_delta + (i64)d;
