@@ -246,11 +246,8 @@ class GpuQueryStageExecutor:
         if self.gpu_codec:
             if utf8_src:
                 raise RuntimeError("gpu_codec: fixed-width columns only (r1)")
-            if null_ids:
-                raise RuntimeError(
-                    "gpu_codec: null columns take the host codec path (r1)")
             return self._encode_partitions_gpu(
-                schema, offsets, cols, out_bufs), dt_device
+                schema, offsets, cols, out_bufs, valid_perm), dt_device
 
         streams = []
         for p in range(self.k):
@@ -321,14 +318,20 @@ class GpuResidentShuffleStage(GpuQueryStageExecutor):
         return offsets, {i: b for i, b in enumerate(out_bufs)}, table.schema
 
 
-def _encode_partitions_gpu(self, schema, offsets, cols, out_bufs):
+def _encode_partitions_gpu(self, schema, offsets, cols, out_bufs,
+                           valid_perm=None):
     """Device-LZ4 every partition's column slices in ONE flat launch, pack
     the frame bodies on device in ONE launch, then download only the
     compressed bytes (one D2H per (partition, column) buffer); the
-    handwritten IPC writer supplies the metadata."""
+    handwritten IPC writer supplies the metadata.  Null columns ride
+    along: their permuted validity (1/64 of the data bytes, already on
+    host for metrics) is bit-resliced per partition and LZ4-framed with
+    pyarrow's codec — the data plane stays device-compressed."""
     import ctypes
     import struct
     ctx = self.ctx
+    valid_perm = valid_perm or {}
+    lz4 = pa.Codec("lz4") if valid_perm else None
     # phase 1: enumerate every (partition, column, 64KiB block)
     buf_meta = []   # per (p, ci): (length, nblocks, first_job)
     jobs = []
@@ -405,8 +408,20 @@ def _encode_partitions_gpu(self, schema, offsets, cols, out_bufs):
             frame = (gpu.LZ4_FRAME_HEADER +
                      packed_host[poff:poff + body_len] +
                      b"\x00\x00\x00\x00")
-            nodes.append((m, 0))
-            buffer_parts.append(None)
+            if ci in valid_perm:
+                part_bool = valid_perm[ci][lo:hi]
+                nulls = int(m - part_bool.sum())
+                nodes.append((m, nulls))
+                if nulls:
+                    vb = np.packbits(part_bool, bitorder="little").tobytes()
+                    buffer_parts.append(
+                        struct.pack("<q", len(vb)) +
+                        lz4.compress(vb).to_pybytes())
+                else:
+                    buffer_parts.append(None)
+            else:
+                nodes.append((m, 0))
+                buffer_parts.append(None)
             buffer_parts.append(struct.pack("<q", length) + frame)
         streams.append(bgipc.stream_from_compressed_batches(
             schema, [(m, nodes, buffer_parts)]))
