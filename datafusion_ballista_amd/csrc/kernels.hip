@@ -3795,6 +3795,28 @@ extern "C" int bg_dict_indices_batch(const void* h_jobs, int64_t njobs) {
   return BG_OK;
 }
 
+// Arrow offset rebase: partition slices of a shared i32 offsets vector
+// start at arbitrary absolute byte positions — each partition's stream
+// needs offsets rebased to 0 (the writer-side dual of read_partition's
+// slice logic).
+__global__ void k_sub_i32(const int32_t* src, int64_t n, int32_t sub,
+                          int32_t* out) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x)
+    out[i] = src[i] - sub;
+}
+
+extern "C" int bg_sub_i32(const void* d_src, int64_t n, int32_t sub,
+                          void* d_out) {
+  REQUIRE_INIT();
+  int blocks = (int)bg_imin64((n + BG_BLOCK - 1) / BG_BLOCK, BG_MAX_BLOCKS);
+  if (blocks == 0) blocks = 1;
+  hipLaunchKernelGGL(k_sub_i32, dim3(blocks), dim3(BG_BLOCK), 0, 0,
+                     (const int32_t*)d_src, n, sub, (int32_t*)d_out);
+  HIP_TRY(hipGetLastError());
+  return BG_OK;
+}
+
 // ---------------------------------------------------------------------------
 // Variable-length (Utf8/Binary) row gather — the take/interleave the
 // sort-shuffle writer needs for string payload columns

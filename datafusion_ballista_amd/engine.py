@@ -244,10 +244,9 @@ class GpuQueryStageExecutor:
                 words, bitorder="little")[:n].astype(bool)
 
         if self.gpu_codec:
-            if utf8_src:
-                raise RuntimeError("gpu_codec: fixed-width columns only (r1)")
             return self._encode_partitions_gpu(
-                schema, offsets, cols, out_bufs, valid_perm), dt_device
+                schema, offsets, cols, out_bufs, valid_perm, utf8_out,
+                utf8_raws), dt_device
 
         streams = []
         for p in range(self.k):
@@ -319,21 +318,46 @@ class GpuResidentShuffleStage(GpuQueryStageExecutor):
 
 
 def _encode_partitions_gpu(self, schema, offsets, cols, out_bufs,
-                           valid_perm=None):
-    """Device-LZ4 every partition's column slices in ONE flat launch, pack
+                           valid_perm=None, utf8_out=None, utf8_raws=None):
+    """Device-LZ4 every partition's buffer slices in ONE flat launch, pack
     the frame bodies on device in ONE launch, then download only the
-    compressed bytes (one D2H per (partition, column) buffer); the
-    handwritten IPC writer supplies the metadata.  Null columns ride
-    along: their permuted validity (1/64 of the data bytes, already on
-    host for metrics) is bit-resliced per partition and LZ4-framed with
-    pyarrow's codec — the data plane stays device-compressed."""
+    compressed bytes; the handwritten IPC writer supplies the metadata.
+    Fixed-width columns contribute one data buffer; Utf8 columns
+    contribute rebased i32 offsets (bg_sub_i32 on device) + their byte
+    range; null columns ride as host-LZ4'd validity parts (1/64 of the
+    data bytes, already host-resident for metrics)."""
     import ctypes
     import struct
     ctx = self.ctx
     valid_perm = valid_perm or {}
+    utf8_out = utf8_out or {}
+    utf8_raws = utf8_raws or {}
     lz4 = pa.Codec("lz4") if valid_perm else None
-    # phase 1: enumerate every (partition, column, 64KiB block)
-    buf_meta = []   # per (p, ci): (length, nblocks, first_job)
+    fixed_ids = [i for i in range(len(cols)) if i not in utf8_out]
+    fixed_buf = {ci: b for ci, b in zip(fixed_ids, out_bufs)}
+
+    # device-rebase each partition's offset slice for utf8 columns
+    reb = {}
+    for ci, (oo, od, tot) in utf8_out.items():
+        r = ctx.alloc(max(4 * (int(offsets[-1]) + self.k), 4))
+        cur = 0
+        regions = {}
+        offs_all = utf8_raws[ci][0]
+        for p in range(self.k):
+            lo, hi = int(offsets[p]), int(offsets[p + 1])
+            m = hi - lo
+            if m == 0:
+                continue
+            gpu._check(ctx.L.bg_sub_i32(
+                ctypes.c_void_p(oo.ptr.value + 4 * lo),
+                ctypes.c_int64(m + 1), ctypes.c_int32(int(offs_all[lo])),
+                ctypes.c_void_p(r.ptr.value + 4 * cur)), "bg_sub_i32")
+            regions[p] = cur
+            cur += m + 1
+        reb[ci] = (r, regions)
+
+    # phase 1: enumerate every (partition, column, unit, 64KiB block)
+    buf_meta = []   # (p, ci, kind, length, nblocks, first_job)
     jobs = []
     slot_cursor = 0
     for p in range(self.k):
@@ -342,16 +366,25 @@ def _encode_partitions_gpu(self, schema, offsets, cols, out_bufs,
         if m == 0:
             continue
         for ci in range(len(cols)):
-            esz = gpu._DT_SIZE[cols[ci].dtype]
-            length = m * esz
-            nblocks = (length + 65536 - 1) // 65536
-            buf_meta.append((p, ci, m, length, nblocks, len(jobs),
-                             slot_cursor))
-            for i in range(nblocks):
-                blen = min(65536, length - i * 65536)
-                jobs.append((out_bufs[ci].ptr.value + lo * esz + i * 65536,
-                             slot_cursor, blen))
-                slot_cursor += 65544
+            if ci in utf8_out:
+                oo, od, tot = utf8_out[ci]
+                offs_all = utf8_raws[ci][0]
+                r, regions = reb[ci]
+                o_lo, o_hi = int(offs_all[lo]), int(offs_all[hi])
+                units = [("offs", r.ptr.value + 4 * regions[p],
+                          4 * (m + 1)),
+                         ("data", od.ptr.value + o_lo, o_hi - o_lo)]
+            else:
+                esz = gpu._DT_SIZE[cols[ci].dtype]
+                units = [("data", fixed_buf[ci].ptr.value + lo * esz,
+                          m * esz)]
+            for kind, src_ptr, length in units:
+                nblocks = (length + 65536 - 1) // 65536
+                buf_meta.append((p, ci, kind, length, nblocks, len(jobs)))
+                for i in range(nblocks):
+                    blen = min(65536, length - i * 65536)
+                    jobs.append((src_ptr + i * 65536, slot_cursor, blen))
+                    slot_cursor += 65544
     if not jobs:
         return [b""] * self.k
     slots = ctx.alloc(max(slot_cursor, 8))
@@ -366,15 +399,15 @@ def _encode_partitions_gpu(self, schema, offsets, cols, out_bufs,
 
     # phase 2: pack each buffer's [u32 size][block] sequence contiguously
     pack_jobs = []
-    packed_layout = {}  # (p, ci) -> (packed_off, body_len, length, m)
+    packed_layout = {}  # (p, ci, kind) -> (packed_off, body_len, length)
     packed_cursor = 0
-    for (p, ci, m, length, nblocks, job0, _sc) in buf_meta:
+    for (p, ci, kind, length, nblocks, job0) in buf_meta:
         body = 0
         for i in range(nblocks):
             sz = int(sizes[job0 + i])
             payload = -sz if sz < 0 else sz
             body += 4 + payload
-        packed_layout[(p, ci)] = (packed_cursor, body, length, m)
+        packed_layout[(p, ci, kind)] = (packed_cursor, body, length)
         off = packed_cursor
         for i in range(nblocks):
             sz = int(sizes[job0 + i])
@@ -394,6 +427,13 @@ def _encode_partitions_gpu(self, schema, offsets, cols, out_bufs,
 
     # phase 3: one D2H of the whole packed region, then assemble streams
     packed_host = packed.download(np.uint8, packed_cursor).tobytes()
+
+    def part_of(p, ci, kind):
+        poff, body_len, length = packed_layout[(p, ci, kind)]
+        frame = (gpu.LZ4_FRAME_HEADER + packed_host[poff:poff + body_len] +
+                 b"\x00\x00\x00\x00")
+        return struct.pack("<q", length) + frame
+
     streams = []
     for p in range(self.k):
         lo, hi = int(offsets[p]), int(offsets[p + 1])
@@ -404,10 +444,6 @@ def _encode_partitions_gpu(self, schema, offsets, cols, out_bufs,
         nodes = []
         buffer_parts = []
         for ci in range(len(cols)):
-            poff, body_len, length, m2 = packed_layout[(p, ci)]
-            frame = (gpu.LZ4_FRAME_HEADER +
-                     packed_host[poff:poff + body_len] +
-                     b"\x00\x00\x00\x00")
             if ci in valid_perm:
                 part_bool = valid_perm[ci][lo:hi]
                 nulls = int(m - part_bool.sum())
@@ -422,7 +458,11 @@ def _encode_partitions_gpu(self, schema, offsets, cols, out_bufs,
             else:
                 nodes.append((m, 0))
                 buffer_parts.append(None)
-            buffer_parts.append(struct.pack("<q", length) + frame)
+            if ci in utf8_out:
+                buffer_parts.append(part_of(p, ci, "offs"))
+                buffer_parts.append(part_of(p, ci, "data"))
+            else:
+                buffer_parts.append(part_of(p, ci, "data"))
         streams.append(bgipc.stream_from_compressed_batches(
             schema, [(m, nodes, buffer_parts)]))
     slots.free()

@@ -143,6 +143,9 @@ int bg_gather_varlen(const void* d_src_data, const int32_t* d_src_offsets,
                      int32_t* d_out_offsets, void* d_out_data,
                      int64_t out_data_cap, int64_t* out_total_bytes);
 
+/* Arrow i32 offset rebase (writer-side partition slicing). */
+int bg_sub_i32(const void* d_src, int64_t n, int32_t sub, void* d_out);
+
 /* Validity-bitmap gather: out bit i = valid[idx[i]] (Arrow LSB order;
  * d_out_bits holds ceil(m/64)*8 bytes) — take for null-carrying columns. */
 int bg_gather_bits(const uint8_t* d_valid, const uint32_t* d_idx, int64_t m,

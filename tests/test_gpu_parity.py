@@ -1318,20 +1318,24 @@ def test_hashjoin_nulls(ctx):
 
 
 def test_sort_shuffle_stage_with_nulls_gpu_codec(ctx, tmp_path):
-    """gpu_codec stage over null-carrying FIXED-WIDTH columns: the data
-    plane is device-LZ4'd, validity rides as per-partition LZ4-framed
-    bitmap parts; partitions must equal table.take(oracle order)."""
+    """gpu_codec stage over null-carrying columns INCLUDING strings: data
+    and rebased utf8 offsets are device-LZ4'd, validity rides as
+    per-partition LZ4-framed bitmap parts; partitions must equal
+    table.take(oracle order)."""
     from datafusion_ballista_amd import engine, shuffle
     n, k = 50_000, 8
     rng = np.random.default_rng(41)
     kmask = rng.random(n) < 0.15
     vmask = rng.random(n) < 0.4
+    smask = rng.random(n) < 0.25
     kvals = rng.integers(0, 4_000, size=n, dtype=np.int64)
     table = pa.table({
         "k": pa.array(kvals, mask=kmask),
         "v": pa.array(rng.integers(-10**9, 10**9, size=n, dtype=np.int64),
                       mask=vmask),
         "d": pa.array(rng.integers(0, 10**6, size=n, dtype=np.int32)),
+        "s": pa.array([None if m else f"str{i % 1009}-{i % 7}"
+                       for i, m in enumerate(smask)], type=pa.string()),
     })
     ex = engine.GpuQueryStageExecutor(ctx, "job-ngc", 8, str(tmp_path),
                                       key_columns=[0], num_partitions=k,
