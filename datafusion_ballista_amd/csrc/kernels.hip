@@ -2038,30 +2038,40 @@ __global__ void k_sort_digit_seq(const u64* keyu_cur, int64_t n, int shift,
     pids[i] = (uint32_t)((keyu_cur[i] >> shift) & 0xffu);
 }
 
-__global__ void k_key_transform_i64(const int64_t* keys, int64_t n,
+__global__ void k_key_transform_i64(const int64_t* keys,
+                                    const uint8_t* valid, int64_t n,
                                     int descending, u64* keyu) {
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
        i += (int64_t)gridDim.x * blockDim.x) {
+    if (!bit_valid(valid, i)) { keyu[i] = 0; continue; }  // stable nulls
     u64 u = (u64)keys[i] ^ 0x8000000000000000ull;
     keyu[i] = descending ? ~u : u;
   }
 }
 
-__global__ void k_key_transform_i32(const int32_t* keys, int64_t n,
+__global__ void k_key_transform_i32(const int32_t* keys,
+                                    const uint8_t* valid, int64_t n,
                                     int descending, u64* keyu) {
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
        i += (int64_t)gridDim.x * blockDim.x) {
+    if (!bit_valid(valid, i)) { keyu[i] = 0; continue; }
     u64 u = (u64)((uint32_t)keys[i] ^ 0x80000000u);
     keyu[i] = descending ? (~u & 0xffffffffull) : u;
   }
 }
 
 // Decimal128 -> order-preserving (hi, lo) u64 pair (sign-flip the high word)
-__global__ void k_key_transform_dec128(const ulong2* keys, int64_t n,
+__global__ void k_key_transform_dec128(const ulong2* keys,
+                                       const uint8_t* valid, int64_t n,
                                        int descending, u64* key_lo,
                                        u64* key_hi) {
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
        i += (int64_t)gridDim.x * blockDim.x) {
+    if (!bit_valid(valid, i)) {
+      key_lo[i] = 0;
+      key_hi[i] = 0;
+      continue;
+    }
     const ulong2 v = keys[i];
     u64 lo = v.x;
     u64 hi = v.y ^ 0x8000000000000000ull;
@@ -2290,12 +2300,66 @@ static int radix_sort_passes(u64* d_key_words[2], int nwords, int64_t n,
   return rc;
 }
 
+// Utf8 sort keys: LSD radix over 8-byte big-endian chunks with a final
+// length tiebreak chunk sorted FIRST — exact memcmp-then-length order
+// (DataFusion's Utf8 comparator semantics).  Keys are extracted in the
+// CURRENT permutation order each chunk, so stability composes across
+// chunks; constant-digit pass skipping eats the sparse high bytes.
+__global__ void k_utf8_chunk_key(const uint8_t* data, const int32_t* offs,
+                                 const uint8_t* valid, const uint32_t* perm,
+                                 int64_t n,
+                                 int64_t chunk /* -1 = length tiebreak */,
+                                 int descending, u64* keyu) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    const uint32_t r = perm[i];
+    if (!bit_valid(valid, (int64_t)r)) { keyu[i] = 0; continue; }
+    const int32_t lo = offs[r];
+    const int32_t len = offs[r + 1] - lo;
+    u64 key;
+    if (chunk < 0) {
+      key = (u64)(uint32_t)len;
+    } else {
+      key = 0;
+      const int64_t base = chunk * 8;
+      for (int b = 0; b < 8; ++b) {
+        const uint8_t c8 =
+            (base + b < (int64_t)len) ? data[lo + base + b] : 0;
+        key = (key << 8) | (u64)c8;
+      }
+    }
+    keyu[i] = descending ? ~key : key;
+  }
+}
+
+// null-ordering pass: the most significant criterion — valid rows get
+// bit `valid_key`, null rows its complement (nulls_first: nulls -> 0)
+__global__ void k_null_order_key(const uint8_t* valid, const uint32_t* perm,
+                                 int64_t n, int nulls_first, u64* keyu) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    const bool v = bit_valid(valid, (int64_t)perm[i]);
+    keyu[i] = nulls_first ? (v ? 1 : 0) : (v ? 0 : 1);
+  }
+}
+
+__global__ void k_utf8_maxlen(const int32_t* offs, int64_t n, int* out) {
+  int m = 0;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    const int32_t l = offs[i + 1] - offs[i];
+    if (l > m) m = (int)l;
+  }
+  atomicMax(out, m);
+}
+
 /* Multi-column stable sort: keys applied LSD (last column first).  Each
- * column: INT64/INT32/DATE32 (desc via bit-flip).  d_perm out: the row
- * permutation realising the ORDER BY. */
-extern "C" int bg_sort_rows(const bg_column* key_cols,
-                            const int32_t* descending, int32_t nkeys,
-                            int64_t n, uint32_t* d_perm) {
+ * column: INT64/INT32/DATE32/DECIMAL128/UTF8 (desc via bit-flip).
+ * d_perm out: the row permutation realising the ORDER BY. */
+static int sort_rows_impl(const bg_column* key_cols,
+                          const int32_t* descending,
+                          const int32_t* nulls_first, int32_t nkeys,
+                          int64_t n, uint32_t* d_perm) {
   REQUIRE_INIT();
   if (nkeys <= 0 || nkeys > BG_MAX_KEYS)
     return set_err(BG_ERR_INVALID, "nkeys out of range [1,4]");
@@ -2314,27 +2378,69 @@ extern "C" int bg_sort_rows(const bg_column* key_cols,
     switch (key_cols[c].dtype) {
       case BG_DT_INT64:
         hipLaunchKernelGGL(k_key_transform_i64, dim3(blocks), dim3(BG_BLOCK),
-                           0, 0, (const int64_t*)key_cols[c].d_data, n,
+                           0, 0, (const int64_t*)key_cols[c].d_data,
+                           key_cols[c].d_validity, n,
                            descending[c], d_keyu[0]);
         npasses = 8; nwords = 1;
         break;
       case BG_DT_INT32:
       case BG_DT_DATE32:
         hipLaunchKernelGGL(k_key_transform_i32, dim3(blocks), dim3(BG_BLOCK),
-                           0, 0, (const int32_t*)key_cols[c].d_data, n,
+                           0, 0, (const int32_t*)key_cols[c].d_data,
+                           key_cols[c].d_validity, n,
                            descending[c], d_keyu[0]);
         npasses = 4; nwords = 1;
         break;
       case BG_DT_DECIMAL128:
         hipLaunchKernelGGL(k_key_transform_dec128, dim3(blocks),
                            dim3(BG_BLOCK), 0, 0,
-                           (const ulong2*)key_cols[c].d_data, n,
+                           (const ulong2*)key_cols[c].d_data,
+                           key_cols[c].d_validity, n,
                            descending[c], d_keyu[0], d_keyu[1]);
         npasses = 8; nwords = 2;
         break;
+      case BG_DT_UTF8: {
+        // LSD over 8-byte chunks: length tiebreak first, then byte
+        // chunks from the string tail to the head; each chunk's keys are
+        // extracted in the CURRENT perm order (no separate permute)
+        const uint8_t* d_data = (const uint8_t*)key_cols[c].d_data;
+        const int32_t* d_offs = key_cols[c].d_offsets;
+        int* d_maxlen;
+        HIP_TRY(pool_malloc((void**)&d_maxlen, sizeof(int)));
+        HIP_TRY(hipMemsetAsync(d_maxlen, 0, sizeof(int), 0));
+        hipLaunchKernelGGL(k_utf8_maxlen, dim3(blocks), dim3(BG_BLOCK), 0, 0,
+                           d_offs, n, d_maxlen);
+        int maxlen = 0;
+        HIP_TRY(hipMemcpy(&maxlen, d_maxlen, sizeof(int),
+                          hipMemcpyDeviceToHost));
+        (void)pool_release(d_maxlen);
+        if (maxlen > (1 << 20))
+          return set_err(BG_ERR_INVALID, "utf8 sort keys > 1 MiB");
+        const int64_t nch = ((int64_t)maxlen + 7) / 8;
+        for (int64_t j = -1; j < nch && rc == BG_OK; ++j) {
+          const int64_t chunk = (j < 0) ? -1 : (nch - 1 - j);
+          hipLaunchKernelGGL(k_utf8_chunk_key, dim3(blocks), dim3(BG_BLOCK),
+                             0, 0, d_data, d_offs, key_cols[c].d_validity,
+                             d_perm, n, chunk,
+                             descending[c], d_keyperm[0]);
+          u64* words[2] = {d_keyperm[0], d_keyperm[1]};
+          rc = radix_sort_passes(words, 1, n, d_perm, 8);
+          d_keyperm[0] = words[0];
+        }
+        if (rc == BG_OK && key_cols[c].d_validity) {
+          hipLaunchKernelGGL(k_null_order_key, dim3(blocks), dim3(BG_BLOCK),
+                             0, 0, key_cols[c].d_validity, d_perm, n,
+                             nulls_first ? nulls_first[c] : descending[c],
+                             d_keyperm[0]);
+          u64* words[2] = {d_keyperm[0], d_keyperm[1]};
+          rc = radix_sort_passes(words, 1, n, d_perm, 1);
+          d_keyperm[0] = words[0];
+        }
+        continue;
+      }
       default:
         rc = set_err(BG_ERR_UNSUPPORTED,
-                     "sort keys: INT64/INT32/DATE32/DECIMAL128 (round 1)");
+                     "sort keys: INT64/INT32/DATE32/DECIMAL128/UTF8");
         continue;
     }
     // bring the key stream into the CURRENT permutation order once
@@ -2347,6 +2453,18 @@ extern "C" int bg_sort_rows(const bg_column* key_cols,
     // ended up live so the release below frees each buffer exactly once
     d_keyperm[0] = words[0];
     if (nwords > 1) d_keyperm[1] = words[1];
+    if (rc == BG_OK && key_cols[c].d_validity) {
+      // null-ordering pass — the MOST significant criterion for this
+      // key: SQL default NULLS LAST for ASC, NULLS FIRST for DESC
+      // unless the caller says otherwise (bg_sort_rows2)
+      hipLaunchKernelGGL(k_null_order_key, dim3(blocks), dim3(BG_BLOCK),
+                         0, 0, key_cols[c].d_validity, d_perm, n,
+                         nulls_first ? nulls_first[c] : descending[c],
+                         d_keyperm[0]);
+      u64* words2[2] = {d_keyperm[0], d_keyperm[1]};
+      rc = radix_sort_passes(words2, 1, n, d_perm, 1);
+      d_keyperm[0] = words2[0];
+    }
   }
   (void)pool_release(d_keyu[0]);
   (void)pool_release(d_keyu[1]);
@@ -2354,6 +2472,22 @@ extern "C" int bg_sort_rows(const bg_column* key_cols,
   (void)pool_release(d_keyperm[1]);
   if (rc == BG_OK) HIP_TRY(hipGetLastError());
   return rc;
+}
+
+extern "C" int bg_sort_rows(const bg_column* key_cols,
+                            const int32_t* descending, int32_t nkeys,
+                            int64_t n, uint32_t* d_perm) {
+  return sort_rows_impl(key_cols, descending, nullptr, nkeys, n, d_perm);
+}
+
+/* As bg_sort_rows with explicit per-key null ordering (1 = NULLS FIRST);
+ * bg_sort_rows defaults to the SQL convention (ASC -> NULLS LAST,
+ * DESC -> NULLS FIRST).  Null rows sort as one stable group. */
+extern "C" int bg_sort_rows2(const bg_column* key_cols,
+                             const int32_t* descending,
+                             const int32_t* nulls_first, int32_t nkeys,
+                             int64_t n, uint32_t* d_perm) {
+  return sort_rows_impl(key_cols, descending, nulls_first, nkeys, n, d_perm);
 }
 
 

@@ -263,6 +263,14 @@ int bg_project_dec128(int32_t op, const bg_column* a, const bg_column* b,
 int bg_sort_rows(const bg_column* key_cols, const int32_t* descending,
                  int32_t nkeys, int64_t n, uint32_t* d_perm);
 
+/* As bg_sort_rows with explicit per-key null ordering (1 = NULLS FIRST).
+ * bg_sort_rows defaults to the SQL convention (ASC -> NULLS LAST,
+ * DESC -> NULLS FIRST); null keys sort as one stable group and never
+ * scramble by their undefined payload bytes. */
+int bg_sort_rows2(const bg_column* key_cols, const int32_t* descending,
+                  const int32_t* nulls_first, int32_t nkeys, int64_t n,
+                  uint32_t* d_perm);
+
 /* ---- SortMergeJoinExec (INNER, key-sorted Int64 inputs) ----
  * The reference's DEFAULT partitioned join (prefer_hash_join=false,
  * extension.rs:850-858; every approved/q*.txt join stage is SMJ).  Inputs

@@ -1492,3 +1492,71 @@ def test_two_phase_aggregate_via_shuffle(ctx, tmp_path):
     want = oracle.hashagg([keys_all], [("sum", vals_all)], len(keys_all))
     want = {kk[0]: (v[1][0], v[0]) for kk, v in want.items()}
     assert final == want
+
+
+@pytest.mark.parametrize("desc", [False, True])
+def test_sort_utf8_keys(ctx, desc):
+    """ORDER BY a Utf8 column: exact memcmp-then-length order (the
+    reference's Utf8 comparator), stable, incl. embedded NULs, empty
+    strings, shared prefixes and length ties vs python's stable sort."""
+    n = 40_000
+    rng = np.random.default_rng(91)
+    pool = [b"", b"a", b"a\x00", b"a\x00b", b"ab", b"abc", b"abcdefgh",
+            b"abcdefghi", b"abcdefgh\x00", b"zz", b"z" * 23]
+    strs = [pool[i % len(pool)] + (str(rng.integers(0, 50)).encode()
+                                   if i % 3 else b"") for i in range(n)]
+    col, keep = None, []
+    data = b"".join(strs)
+    offs = np.zeros(n + 1, dtype=np.int32)
+    for i, b in enumerate(strs):
+        offs[i + 1] = offs[i] + len(b)
+    dbuf = ctx.upload(np.frombuffer(data, dtype=np.uint8)
+                      if data else np.zeros(1, np.uint8))
+    obuf = ctx.upload(offs)
+    col = ctx.column(gpu.BG_DT_UTF8, dbuf, n, offsets=obuf)
+    perm = ctx.sort_rows([col], [desc], n).download(np.uint32, n)
+    if desc:
+        # stable descending: invert every byte and use inverted length as
+        # the tiebreak (matches the engine's ~key transform exactly)
+        want = sorted(range(n),
+                      key=lambda i: (bytes(255 - b for b in strs[i]) +
+                                     b"\xff" * (64 - len(strs[i])),
+                                     -len(strs[i])))
+    else:
+        want = sorted(range(n), key=lambda i: strs[i])
+    assert list(perm) == want
+
+
+def test_sort_null_ordering(ctx):
+    """ORDER BY with NULL keys: SQL default (ASC -> NULLS LAST, DESC ->
+    NULLS FIRST) via bg_sort_rows, explicit override via bg_sort_rows2;
+    nulls form one stable group (original order preserved)."""
+    import ctypes
+    n = 30_000
+    rng = np.random.default_rng(93)
+    vals = rng.integers(0, 1000, size=n, dtype=np.int64)
+    mask = rng.random(n) < 0.2
+    vbits = np.packbits(~mask, bitorder="little")
+    col, _ = ctx.upload_column(vals, gpu.BG_DT_INT64, validity=vbits)
+    null_rows = [i for i in range(n) if mask[i]]
+    valid_sorted = sorted((i for i in range(n) if not mask[i]),
+                          key=lambda i: (vals[i], i))
+
+    perm = ctx.sort_rows([col], [False], n).download(np.uint32, n)
+    assert list(perm) == valid_sorted + null_rows  # ASC: NULLS LAST
+
+    perm = ctx.sort_rows([col], [True], n).download(np.uint32, n)
+    desc_sorted = sorted((i for i in range(n) if not mask[i]),
+                         key=lambda i: (-vals[i], i))
+    assert list(perm) == null_rows + desc_sorted   # DESC: NULLS FIRST
+
+    # explicit override: ASC + NULLS FIRST
+    pbuf = ctx.alloc(4 * n)
+    karr = (gpu.BgColumn * 1)(col)
+    darr = (ctypes.c_int32 * 1)(0)
+    nfarr = (ctypes.c_int32 * 1)(1)
+    gpu._check(ctx.L.bg_sort_rows2(karr, darr, nfarr, 1,
+                                   ctypes.c_int64(n), pbuf.ptr), "sort2")
+    ctx.synchronize()
+    perm = pbuf.download(np.uint32, n)
+    assert list(perm) == null_rows + valid_sorted
