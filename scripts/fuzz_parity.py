@@ -127,6 +127,37 @@ def fuzz_round(ctx, seed):
             f"null join pairs mismatch seed={seed}"
     jn.free()
 
+    # sort with NULL keys: SQL default null ordering + stable null group
+    perm3 = ctx.sort_rows([colv], [desc], n).download(np.uint32, n)
+    vb = np.unpackbits(vbits, bitorder="little")[:n].astype(bool)
+    nulls = [i for i in range(n) if not vb[i]]
+    u3 = keys.view(np.uint64) ^ np.uint64(1 << 63)
+    if desc:
+        order = sorted((i for i in range(n) if vb[i]),
+                       key=lambda i: (~u3[i] & 0xFFFFFFFFFFFFFFFF, i))
+        want3 = nulls + order
+    else:
+        order = sorted((i for i in range(n) if vb[i]),
+                       key=lambda i: (u3[i], i))
+        want3 = order + nulls
+    assert np.array_equal(perm3.astype(np.int64), np.array(want3)), \
+        f"null sort mismatch seed={seed} desc={desc}"
+
+    # utf8 sort on derived strings (shared prefixes + length ties)
+    if n <= 60_000:
+        strs = [b"p%d/%d" % (int(k2) % 97, int(k2) % 13) for k2 in probe]
+        data = b"".join(strs)
+        offs = np.zeros(n + 1, dtype=np.int32)
+        for i2, b2 in enumerate(strs):
+            offs[i2 + 1] = offs[i2] + len(b2)
+        db = ctx.upload(np.frombuffer(data, dtype=np.uint8)
+                        if data else np.zeros(1, np.uint8))
+        ob = ctx.upload(offs)
+        ucol = ctx.column(gpu.BG_DT_UTF8, db, n, offsets=ob)
+        permu = ctx.sort_rows([ucol], [False], n).download(np.uint32, n)
+        wantu = sorted(range(n), key=lambda i2: strs[i2])
+        assert list(permu) == wantu, f"utf8 sort mismatch seed={seed}"
+
     # null-aware group-by with non-null counts (bg_hashagg2) on small n
     if n <= 60_000:
         gbits = rng.integers(0, 256, size=(n + 7) // 8, dtype=np.uint8)
