@@ -88,6 +88,27 @@ def test_agg_value_codec():
         assert gpu.decode_agg_value(gpu.BG_AGG_OP_MIN_I64, raw) == v
         raw16 = int(v).to_bytes(16, "little", signed=True)
         assert gpu.decode_agg_value(gpu.BG_AGG_OP_SUM_DEC128, raw16) == v
+    # Float64 MIN/MAX: the IEEE totally-ordered u64 transform round-trips
+    for f in (0.0, -0.0, 1.5, -1.5, 1e300, -1e300, 3.14159, -2.5e-308,
+              float("inf"), float("-inf")):
+        bits = struct.unpack("<Q", struct.pack("<d", f))[0]
+        enc = (~bits) & 0xFFFFFFFFFFFFFFFF if bits >> 63 else             bits ^ (1 << 63)
+        raw = struct.pack("<Q", enc) + b"\x00" * 8
+        got = gpu.decode_agg_value(gpu.BG_AGG_OP_MAX_F64, raw)
+        assert struct.pack("<d", got) == struct.pack("<d", f), f
+        raw = struct.pack("<Q", (~enc) & 0xFFFFFFFFFFFFFFFF) + b"\x00" * 8
+        got = gpu.decode_agg_value(gpu.BG_AGG_OP_MIN_F64, raw)
+        assert struct.pack("<d", got) == struct.pack("<d", f), f
+    # the transform is monotone: ordering of encodings == ordering of floats
+    import random
+    rnd = random.Random(5)
+    fs = sorted(rnd.uniform(-1e12, 1e12) for _ in range(200))
+    encs = []
+    for f in fs:
+        bits = struct.unpack("<Q", struct.pack("<d", f))[0]
+        encs.append((~bits) & 0xFFFFFFFFFFFFFFFF if bits >> 63 else
+                    bits ^ (1 << 63))
+    assert encs == sorted(encs)
 
 
 def test_ipc_stream_empty_batchless():
