@@ -32,7 +32,9 @@
 // the pinned DataFusion commit when a Rust toolchain is available:
 //   (1) seeds (0,0,0,0)          — DataFusion's repartition random state
 //   (2) with_seeds xors PI2      — ahash 0.8.x `RandomState::with_seeds`
-//   (3) buffer=k0, pad=k1        — ahash fallback `from_random_state`
+//   (3) buffer=k1, pad=k0        — ahash 0.8.x fallback `from_random_state`
+//       initializes buffer from k1 and pad from k0 (the SWAPPED order,
+//       unlike `new_with_keys`; flagged by round-1 review, changed r2)
 //   (4) fallback (non-AES) path  — default rustc target has no +aes
 // Everything downstream of partition IDs (query results, file layout,
 // row conservation) is invariant to these choices and is pinned by the
@@ -72,8 +74,8 @@
 #define BG_AHASH_K3 BG_AHASH_PI2_3
 
 typedef struct {
-  uint64_t buffer;  // = k0 at init
-  uint64_t pad;     // = k1 at init
+  uint64_t buffer;  // = k1 at init (from_random_state's swapped order)
+  uint64_t pad;     // = k0 at init
   uint64_t extra0;  // = k2
   uint64_t extra1;  // = k3
 } bg_ahasher;
@@ -95,8 +97,10 @@ BG_HD uint64_t bg_folded_multiply(uint64_t s, uint64_t by) {
 
 BG_HD bg_ahasher bg_ahasher_init(void) {
   bg_ahasher h;
-  h.buffer = BG_AHASH_K0;
-  h.pad = BG_AHASH_K1;
+  // ahash 0.8.x fallback AHasher::from_random_state: buffer <- k1,
+  // pad <- k0 (swapped relative to new_with_keys).
+  h.buffer = BG_AHASH_K1;
+  h.pad = BG_AHASH_K0;
   h.extra0 = BG_AHASH_K2;
   h.extra1 = BG_AHASH_K3;
   return h;

@@ -51,6 +51,15 @@ static thread_local double g_last_kernel_ms = 0.0;
 extern "C" const char* bg_last_error(void) { return g_err; }
 extern "C" int bg_version(void) { return 10; }
 
+// Build provenance: the sha256 over (kernels.hip, bg_ahash.h,
+// ballista_gpu.h) injected by __graft_entry__.build(); a CPU test
+// recomputes the hash over the committed sources and fails when the
+// committed .so was not built from them.
+#ifndef BG_SOURCE_HASH
+#define BG_SOURCE_HASH "unverified-local-build"
+#endif
+extern "C" const char* bg_source_hash(void) { return BG_SOURCE_HASH; }
+
 // Duration of the most recent timed hot kernel (bg_q6_agg / bg_q1_agg),
 // measured with hipEvents on the launch stream — feeds bench.py's
 // roofline.achieved (algorithmic bytes / kernel time).
@@ -1689,6 +1698,16 @@ __device__ __forceinline__ u64 hash_keys_row(const KeyArgs& keys, int64_t i) {
       case BG_DT_DICT8:
         hc = bg_hash_u64((u64) reinterpret_cast<const uint8_t*>(keys.k[c].data)[i]);
         break;
+      case BG_DT_UTF8: {
+        // mirror k_hash_columns: without this, every Utf8 key row hashed
+        // to one slot and the group-by degenerated to a single linear
+        // chain
+        const int32_t lo = keys.k[c].offsets[i];
+        const int32_t hi_off = keys.k[c].offsets[i + 1];
+        hc = bg_hash_str(reinterpret_cast<const uint8_t*>(keys.k[c].data) + lo,
+                         (uint64_t)(hi_off - lo));
+        break;
+      }
       default:
         hc = 0;
     }

@@ -275,20 +275,24 @@ class GpuParquetColumnReader:
         got_values = 0
         snappy_jobs = []
         data_pages = []      # (scratch_off, usz, dst_off, nvals) in order
-        v2_fixups = []  # (soff, file_def_off, def_len, dst_voff)
+        v2_fixups = []  # (kind, scratch_off, file_off, nbytes); kind is an
+        # explicit tag ("def" = [u32 dlen] prefix + level bytes, "raw" =
+        # uncompressed values copy) — sign-encoding the length misclassified
+        # zero-length raw copies (all-null uncompressed V2 pages) as prefix
+        # entries and desynchronized the prefix index
         for (rg, ptype, poff, csz, usz, nvals, enc, ndict, soff,
              v2) in headers:
             if v2 is not None:
                 def_len, values_csz, values_usz, is_comp = v2
                 voff = (4 + def_len) if max_def0 > 0 else 0
                 if max_def0 > 0:
-                    v2_fixups.append((soff, poff, def_len))
+                    v2_fixups.append(("def", soff, poff, def_len))
                 if codec == "SNAPPY" and is_comp:
                     snappy_jobs.append((poff + def_len, values_csz,
                                         soff + voff, values_usz))
                 else:
-                    v2_fixups.append((soff + voff, poff + def_len,
-                                      -values_usz))  # raw value copy
+                    v2_fixups.append(("raw", soff + voff, poff + def_len,
+                                      values_usz))
             elif codec == "SNAPPY":
                 snappy_jobs.append((poff, csz, soff, usz))
             if ptype == 2:
@@ -328,16 +332,17 @@ class GpuParquetColumnReader:
             # uncompressed level bytes (and raw values when the page body
             # is not compressed), copied device-to-device from the
             # resident file buffer via a tiny staged prefix table
-            prefixes = np.array([d for (_, _, d) in v2_fixups if d >= 0],
-                                dtype=np.uint32)
+            prefixes = np.array([d for (k, _, _, d) in v2_fixups
+                                 if k == "def"], dtype=np.uint32)
             pbuf = ctx.upload(prefixes) if len(prefixes) else None
             pi = 0
-            for (soff, foff, dlen) in v2_fixups:
-                if dlen < 0:  # uncompressed values copy
-                    gpu._check(ctx.L.bg_memcpy_dtod(
-                        ctypes.c_void_p(scratch.ptr.value + soff),
-                        ctypes.c_void_p(chunk.ptr.value + foff),
-                        ctypes.c_uint64(-dlen)), "v2 values copy")
+            for (kind, soff, foff, dlen) in v2_fixups:
+                if kind == "raw":  # uncompressed values copy (may be empty)
+                    if dlen:
+                        gpu._check(ctx.L.bg_memcpy_dtod(
+                            ctypes.c_void_p(scratch.ptr.value + soff),
+                            ctypes.c_void_p(chunk.ptr.value + foff),
+                            ctypes.c_uint64(dlen)), "v2 values copy")
                     continue
                 gpu._check(ctx.L.bg_memcpy_dtod(
                     ctypes.c_void_p(scratch.ptr.value + soff),

@@ -222,30 +222,15 @@ def read_passthrough_partition(path):
     return list(pa.ipc.open_stream(pa.BufferReader(raw)))
 
 
-def read_partition_gpu(ctx, data_path, index_path, partition_id, schema):
-    """Device shuffle-read: decode a partition's LZ4-compressed batches
-    straight into HBM column buffers (ShuffleReaderExec local-read
-    equivalent for a GPU-resident next stage).  Fixed-width columns, null-
-    free batches.  Returns (n_rows, {col_idx: DeviceBuffer}) with columns
-    concatenated across the partition's batches."""
-    import ctypes
-
-    import numpy as np
-
-    from . import gpu as g
+def walk_partition_batches(raw, schema):
+    """Parse one partition's byte range — a concatenation of COMPLETE IPC
+    sub-streams (one per input, writer.rs:861-884; crossed transparently
+    like multi_stream_reader.rs:17-34) — into
+    [(n_rows, per-col (frame_off_in_raw, frame_len, usize) | None)].
+    Pure host logic (no GPU), so the sub-stream crossing is unit-testable."""
     from . import ipc as bgipc
 
-    offsets = read_index(index_path)
-    lo, hi = offsets[partition_id], offsets[partition_id + 1]
-    with open(data_path, "rb") as f:
-        f.seek(lo)
-        raw = f.read(hi - lo)
-    if not raw:
-        return 0, {}
-
-    esz = [16 if pa.types.is_decimal128(t) else t.bit_width // 8
-           for t in schema.types]
-    batches = []  # (n_rows, per-col (frame_off_in_raw, frame_len, usize))
+    batches = []
     pos = 0
     while pos < len(raw):
         consumed = pos
@@ -277,7 +262,38 @@ def read_partition_gpu(ctx, data_path, index_path, partition_id, schema):
             consumed = pos + body_off + body_len
         if consumed == pos:
             break
+        # walk_stream ends at a sub-stream's EOS marker without consuming
+        # it; skip it so the NEXT concatenated sub-stream is parsed too
+        # (multi-input consolidated partitions, writer.rs:861-884)
+        if raw[consumed:consumed + 8] == bgipc.EOS:
+            consumed += 8
         pos = consumed
+    return batches
+
+
+def read_partition_gpu(ctx, data_path, index_path, partition_id, schema):
+    """Device shuffle-read: decode a partition's LZ4-compressed batches
+    straight into HBM column buffers (ShuffleReaderExec local-read
+    equivalent for a GPU-resident next stage).  Fixed-width columns, null-
+    free batches.  Returns (n_rows, {col_idx: DeviceBuffer}) with columns
+    concatenated across the partition's batches."""
+    import ctypes
+
+    import numpy as np
+
+    from . import gpu as g
+
+    offsets = read_index(index_path)
+    lo, hi = offsets[partition_id], offsets[partition_id + 1]
+    with open(data_path, "rb") as f:
+        f.seek(lo)
+        raw = f.read(hi - lo)
+    if not raw:
+        return 0, {}
+
+    esz = [16 if pa.types.is_decimal128(t) else t.bit_width // 8
+           for t in schema.types]
+    batches = walk_partition_batches(raw, schema)
 
     total = sum(b[0] for b in batches)
     raw_buf = ctx.upload(np.frombuffer(raw, dtype=np.uint8))

@@ -62,6 +62,11 @@ typedef enum {
 
 const char* bg_last_error(void);
 int bg_version(void);
+/* sha256 over the library's own sources, injected at compile time by the
+ * build entry; "unverified-local-build" when compiled without it.  A CPU
+ * test compares it against a fresh hash of the committed sources so a
+ * stale committed binary is detected. */
+const char* bg_source_hash(void);
 /* duration (ms) of the most recent fused-aggregate kernel launch, measured
  * with hipEvents on the launch stream — roofline evidence for bench.py */
 double bg_last_kernel_ms(void);

@@ -34,6 +34,24 @@ def test_version():
     assert L.bg_version() >= 10
 
 
+def test_committed_library_matches_committed_sources():
+    """Build provenance: the library reports the sha256
+    of the sources it was compiled from; recomputing it over the committed
+    sources must match, so a stale committed .so fails here."""
+    import importlib.util
+    spec = importlib.util.spec_from_file_location(
+        "graft_entry", os.path.join(ROOT, "__graft_entry__.py"))
+    mod = importlib.util.module_from_spec(spec)
+    spec.loader.exec_module(mod)
+    from datafusion_ballista_amd import gpu
+    L = gpu.load_library()
+    L.bg_source_hash.restype = ctypes.c_char_p
+    got = L.bg_source_hash().decode()
+    assert got == mod.source_hash(), (
+        "libballista_gpu.so was not built from the committed sources "
+        f"(lib reports {got!r}); run __graft_entry__.build()")
+
+
 def test_init_fails_loudly_without_gpu():
     """On a GPU-less host bg_init must refuse (no silent CPU fallback)."""
     import torch

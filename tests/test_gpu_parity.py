@@ -527,6 +527,33 @@ def test_hashagg_utf8_keys(ctx):
     assert got == want
 
 
+def test_hashagg_utf8_high_cardinality(ctx):
+    """Regression: hash_keys_row must actually hash
+    Utf8 keys — with the missing case every row collided into one slot and
+    a high-cardinality Utf8 group-by degenerated to one linear chain.
+    20k distinct keys over 100k rows both pins correctness and would hang
+    (O(n*groups)) under the old degenerate path."""
+    rng = np.random.default_rng(23)
+    n = 100_000
+    gids = rng.integers(0, 20_000, size=n)
+    strings = [b"cust#%08d" % g for g in gids]
+    col = ctx.upload_utf8_column(strings)
+    vals = rng.integers(-10**9, 10**9, size=n, dtype=np.int64)
+    vcol, _ = ctx.upload_column(vals, gpu.BG_DT_INT64)
+    first, acc, counts = ctx.hashagg([col], [vcol], [gpu.BG_AGG_OP_SUM_I64],
+                                     n, max_groups=40_000)
+    got = {}
+    for g in range(len(first)):
+        key = strings[first[g]]
+        got[key] = (int(counts[g]),
+                    int.from_bytes(bytes(acc[g, 0]), "little", signed=True))
+    want = {}
+    for s_, v in zip(strings, vals):
+        c, t = want.get(s_, (0, 0))
+        want[s_] = (c + 1, t + int(v))
+    assert got == want
+
+
 def test_sort_dec128_keys(ctx):
     """Decimal128 ORDER BY (q3's revenue DESC is Decimal(38,4)): 16-pass
     radix over the (hi, lo) order-preserving pair."""

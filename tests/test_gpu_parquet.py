@@ -182,6 +182,35 @@ def test_parquet_nullable_decode(ctx, tmp_path, compression,
             assert np.array_equal(vals[got_valid], want.astype(npdt))
 
 
+@pytest.mark.parametrize("compression", ["none", "snappy"])
+def test_parquet_v2_all_null_pages(ctx, tmp_path, compression):
+    """Regression: an all-null DataPageV2 page has an
+    EMPTY values section; the old sign-encoded fixup misread the 0-length
+    raw copy as a def-len prefix entry and desynchronized every later V2
+    page.  Small pages force several all-null pages before valued ones."""
+    n = 60_000
+    rng = np.random.default_rng(31)
+    vals = rng.integers(-2**60, 2**60, size=n, dtype=np.int64)
+    # first half all-null, second half mixed -> the column chunk starts
+    # with multiple all-null pages followed by valued pages
+    mask = np.concatenate([np.ones(n // 2, dtype=bool),
+                           rng.random(n - n // 2) < 0.3])
+    table = pa.table({"v": pa.array(vals, mask=mask)})
+    path = str(tmp_path / f"v2null_{compression}.parquet")
+    pq.write_table(table, path, compression=compression,
+                   use_dictionary=False, data_page_version="2.0",
+                   data_page_size=8 * 1024, write_statistics=False)
+    rd = GpuParquetColumnReader(ctx, path)
+    buf, nv, phys, valid = rd.read_column_all(0)
+    assert nv == n and valid is not None
+    vw = valid.download(np.uint32, (n + 31) // 32)
+    got_valid = np.unpackbits(vw.view(np.uint8),
+                              bitorder="little")[:n].astype(bool)
+    assert np.array_equal(got_valid, ~mask)
+    got = buf.download(np.int64, n)
+    assert np.array_equal(got[got_valid], vals[~mask])
+
+
 @pytest.mark.parametrize("compression", ["snappy", "none"])
 @pytest.mark.parametrize("use_dictionary", [False, True])
 @pytest.mark.parametrize("with_nulls", [False, True])

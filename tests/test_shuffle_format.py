@@ -124,6 +124,35 @@ def test_multi_input_concatenated_streams(tmp_work):
         assert got.num_rows == want_rows
 
 
+def test_walk_partition_batches_multi_input():
+    """Regression: the device-read metadata walk must
+    cross the EOS marker between concatenated sub-streams — a partition
+    range holding one complete IPC stream PER INPUT (writer.rs:861-884)
+    must surface every batch, not just the first stream's."""
+    schema = pa.schema([("k", pa.int64()), ("v", pa.int64()),
+                       ("d", pa.int32())])
+    # three inputs -> three complete IPC streams concatenated verbatim
+    parts = []
+    want_rows = []
+    for seed, n in ((1, 300), (2, 257), (3, 129)):
+        t = make_table(n, seed=seed)
+        parts.append(shuffle.encode_partition_stream(
+            shuffle.rechunk(t, 128), t.schema))
+        want_rows += [128] * (n // 128) + ([n % 128] if n % 128 else [])
+    raw = b"".join(parts)
+    batches = shuffle.walk_partition_batches(raw, schema)
+    assert [b[0] for b in batches] == want_rows
+    assert sum(b[0] for b in batches) == 300 + 257 + 129
+    # every parsed frame must carry a plausible (off, len, usize) triple
+    for n_rows, cols in batches:
+        for ci, c in enumerate(cols):
+            assert c is not None
+            foff, flen, usize = c
+            esz = 4 if ci == 2 else 8
+            assert usize == n_rows * esz
+            assert 0 < foff < len(raw) and 0 < flen <= len(raw) - foff
+
+
 def test_empty_partitions(tmp_work):
     """Empty partitions encode as zero bytes; the reader returns no batches
     and the schema survives via the header stream (writer.rs:838-846)."""
