@@ -4789,3 +4789,221 @@ extern "C" int bg_lz4_decompress(const void* h_frames, int64_t nframes,
   (void)pool_release(d_lens);
   return BG_OK;
 }
+
+// ---------------------------------------------------------------------------
+// stage-interpreter support kernels (bg_execute_stage, stage.cpp):
+//  - bg_set_error: lets the separate stage.cpp translation unit report
+//    through the same thread-local bg_last_error channel
+//  - bg_bitcopy: place a batch's decompressed Arrow validity bits into a
+//    column bitmap at an arbitrary bit offset (device shuffle read crosses
+//    batch boundaries that are not byte-aligned)
+//  - bg_agg_materialize / bg_avg_finalize: turn bg_hashagg accumulator
+//    records into Arrow output columns on device (the final/single
+//    AggregateExec materialisation, incl. SQL all-NULL-group semantics and
+//    DataFusion's decimal AVG scale rules)
+// ---------------------------------------------------------------------------
+
+extern "C" int bg_set_error(int code, const char* msg) {
+  return set_err(code, msg);
+}
+
+__global__ void k_bitcopy(const uint8_t* __restrict__ src, int64_t nbits,
+                          uint32_t* __restrict__ dst, int64_t dst_bit_off) {
+  // each thread owns one 32-bit destination word of the target range;
+  // boundary words shared with neighbouring batches go through atomics
+  const int64_t w0 = dst_bit_off >> 5;
+  const int64_t w1 = (dst_bit_off + nbits - 1) >> 5;
+  for (int64_t w = w0 + (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       w <= w1; w += (int64_t)gridDim.x * blockDim.x) {
+    uint32_t value = 0, mask = 0;
+    const int64_t base_bit = (w << 5);
+    for (int b = 0; b < 32; ++b) {
+      const int64_t src_bit = base_bit + b - dst_bit_off;
+      if (src_bit < 0 || src_bit >= nbits) continue;
+      mask |= (1u << b);
+      if ((src[src_bit >> 3] >> (src_bit & 7)) & 1) value |= (1u << b);
+    }
+    if (mask == 0xffffffffu) {
+      dst[w] = value;
+    } else {
+      atomicAnd(&dst[w], ~mask);
+      atomicOr(&dst[w], value);
+    }
+  }
+}
+
+extern "C" int bg_bitcopy(const uint8_t* d_src_bits, int64_t nbits,
+                          uint8_t* d_dst_bits, int64_t dst_bit_off) {
+  REQUIRE_INIT();
+  if (nbits <= 0) return BG_OK;
+  int64_t words = ((dst_bit_off + nbits) >> 5) - (dst_bit_off >> 5) + 1;
+  int blocks = (int)bg_imin64((words + BG_BLOCK - 1) / BG_BLOCK, BG_MAX_BLOCKS);
+  if (blocks == 0) blocks = 1;
+  hipLaunchKernelGGL(k_bitcopy, dim3(blocks), dim3(BG_BLOCK), 0, 0,
+                     d_src_bits, nbits, (uint32_t*)d_dst_bits, dst_bit_off);
+  HIP_TRY(hipGetLastError());
+  return BG_OK;
+}
+
+// accumulator -> natural value transforms (inverse of the kernels' order-
+// preserving encodings; mirrors gpu.py decode_agg_value)
+__global__ void k_agg_materialize(const uint8_t* __restrict__ acc,
+                                  int64_t acc_stride, int32_t op,
+                                  int64_t ngroups, uint8_t* __restrict__ out,
+                                  const int64_t* __restrict__ nncnt,
+                                  int64_t nncnt_stride,
+                                  uint32_t* __restrict__ valid_out) {
+  for (int64_t g = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; g < ngroups;
+       g += (int64_t)gridDim.x * blockDim.x) {
+    const uint8_t* a = acc + g * acc_stride;
+    const uint64_t lo = *reinterpret_cast<const uint64_t*>(a);
+    const uint64_t hi = *reinterpret_cast<const uint64_t*>(a + 8);
+    bool is_null = false;
+    if (nncnt) {
+      const int64_t nn = *reinterpret_cast<const int64_t*>(
+          reinterpret_cast<const uint8_t*>(nncnt) + g * nncnt_stride);
+      is_null = (nn == 0);
+      if (valid_out) {
+        if (is_null) atomicAnd(&valid_out[g >> 5], ~(1u << (g & 31)));
+        else atomicOr(&valid_out[g >> 5], 1u << (g & 31));
+      }
+    }
+    switch (op) {
+      case BG_AGG_SUM_DEC128: {  // i128 LE, 16-B out
+        uint64_t* o = reinterpret_cast<uint64_t*>(out + g * 16);
+        o[0] = is_null ? 0 : lo;
+        o[1] = is_null ? 0 : hi;
+        break;
+      }
+      case BG_AGG_SUM_I64: {  // low limb as i64
+        reinterpret_cast<int64_t*>(out)[g] = is_null ? 0 : (int64_t)lo;
+        break;
+      }
+      case BG_AGG_SUM_F64: {
+        reinterpret_cast<uint64_t*>(out)[g] = is_null ? 0 : lo;
+        break;
+      }
+      case BG_AGG_MAX_I64: {
+        reinterpret_cast<int64_t*>(out)[g] =
+            is_null ? 0 : (int64_t)(lo ^ 0x8000000000000000ull);
+        break;
+      }
+      case BG_AGG_MIN_I64: {
+        reinterpret_cast<int64_t*>(out)[g] =
+            is_null ? 0 : (int64_t)((~lo) ^ 0x8000000000000000ull);
+        break;
+      }
+      case BG_AGG_MIN_F64:
+      case BG_AGG_MAX_F64: {
+        uint64_t enc = lo;
+        if (op == BG_AGG_MIN_F64) enc = ~enc;
+        uint64_t bits = (enc >> 63) ? (enc ^ 0x8000000000000000ull) : ~enc;
+        reinterpret_cast<uint64_t*>(out)[g] = is_null ? 0 : bits;
+        break;
+      }
+      default: break;
+    }
+  }
+}
+
+extern "C" int bg_agg_materialize(const void* d_acc, int64_t acc_stride,
+                                  int32_t op, int64_t ngroups, void* d_out,
+                                  const int64_t* d_nncnt,
+                                  int64_t nncnt_stride,
+                                  uint8_t* d_valid_out) {
+  REQUIRE_INIT();
+  if (ngroups <= 0) return BG_OK;
+  int blocks =
+      (int)bg_imin64((ngroups + BG_BLOCK - 1) / BG_BLOCK, BG_MAX_BLOCKS);
+  hipLaunchKernelGGL(k_agg_materialize, dim3(blocks), dim3(BG_BLOCK), 0, 0,
+                     (const uint8_t*)d_acc, acc_stride, op, ngroups,
+                     (uint8_t*)d_out, d_nncnt, nncnt_stride,
+                     (uint32_t*)d_valid_out);
+  HIP_TRY(hipGetLastError());
+  return BG_OK;
+}
+
+// strided i64 copy (COUNT / non-null-count materialisation)
+__global__ void k_copy_i64_strided(const uint8_t* __restrict__ src,
+                                   int64_t stride, int64_t n,
+                                   int64_t* __restrict__ out) {
+  for (int64_t g = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; g < n;
+       g += (int64_t)gridDim.x * blockDim.x)
+    out[g] = *reinterpret_cast<const int64_t*>(src + g * stride);
+}
+
+extern "C" int bg_copy_i64_strided(const void* d_src, int64_t stride,
+                                   int64_t n, void* d_out) {
+  REQUIRE_INIT();
+  if (n <= 0) return BG_OK;
+  int blocks = (int)bg_imin64((n + BG_BLOCK - 1) / BG_BLOCK, BG_MAX_BLOCKS);
+  hipLaunchKernelGGL(k_copy_i64_strided, dim3(blocks), dim3(BG_BLOCK), 0, 0,
+                     (const uint8_t*)d_src, stride, n, (int64_t*)d_out);
+  HIP_TRY(hipGetLastError());
+  return BG_OK;
+}
+
+// AVG finalisation.  Decimal: DataFusion's AvgAccumulator for Decimal128
+// (datafusion/functions-aggregate avg.rs): target scale = input scale + 4,
+// value = round_half_up(sum * 10^4 / count) computed exactly in i128.
+// Float64: IEEE division.  d_cnt entries of 0 yield NULL (valid bit clear).
+__global__ void k_avg_finalize(const uint8_t* __restrict__ acc,
+                               int64_t acc_stride, int32_t is_f64,
+                               const uint8_t* __restrict__ cnt,
+                               int64_t cnt_stride, int32_t scale_shift,
+                               int64_t ngroups, uint8_t* __restrict__ out,
+                               uint32_t* __restrict__ valid_out) {
+  for (int64_t g = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; g < ngroups;
+       g += (int64_t)gridDim.x * blockDim.x) {
+    const int64_t c = *reinterpret_cast<const int64_t*>(cnt + g * cnt_stride);
+    if (valid_out) {
+      if (c == 0) atomicAnd(&valid_out[g >> 5], ~(1u << (g & 31)));
+      else atomicOr(&valid_out[g >> 5], 1u << (g & 31));
+    }
+    if (is_f64) {
+      const double s =
+          *reinterpret_cast<const double*>(acc + g * acc_stride);
+      reinterpret_cast<double*>(out)[g] = c ? s / (double)c : 0.0;
+      continue;
+    }
+    __int128 s = (__int128)(
+        ((unsigned __int128)*reinterpret_cast<const uint64_t*>(
+             acc + g * acc_stride + 8)
+         << 64) |
+        *reinterpret_cast<const uint64_t*>(acc + g * acc_stride));
+    __int128 v = 0;
+    if (c) {
+      __int128 p10 = 1;
+      for (int i = 0; i < scale_shift; ++i) p10 *= 10;
+      __int128 num = s * p10;
+      const bool neg = num < 0;
+      unsigned __int128 un = neg ? (unsigned __int128)(-num)
+                                 : (unsigned __int128)num;
+      const unsigned __int128 uc = (unsigned __int128)c;
+      unsigned __int128 q = un / uc;
+      const unsigned __int128 r = un - q * uc;
+      if (2 * r >= uc) q += 1;  // round half away from zero
+      v = neg ? -(__int128)q : (__int128)q;
+    }
+    uint64_t* o = reinterpret_cast<uint64_t*>(out + g * 16);
+    o[0] = (uint64_t)v;
+    o[1] = (uint64_t)(v >> 64);
+  }
+}
+
+extern "C" int bg_avg_finalize(const void* d_sum_acc, int64_t acc_stride,
+                               int32_t is_f64, const void* d_cnt,
+                               int64_t cnt_stride, int32_t scale_shift,
+                               int64_t ngroups, void* d_out,
+                               uint8_t* d_valid_out) {
+  REQUIRE_INIT();
+  if (ngroups <= 0) return BG_OK;
+  int blocks =
+      (int)bg_imin64((ngroups + BG_BLOCK - 1) / BG_BLOCK, BG_MAX_BLOCKS);
+  hipLaunchKernelGGL(k_avg_finalize, dim3(blocks), dim3(BG_BLOCK), 0, 0,
+                     (const uint8_t*)d_sum_acc, acc_stride, is_f64,
+                     (const uint8_t*)d_cnt, cnt_stride, scale_shift, ngroups,
+                     (uint8_t*)d_out, (uint32_t*)d_valid_out);
+  HIP_TRY(hipGetLastError());
+  return BG_OK;
+}

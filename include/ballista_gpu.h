@@ -483,6 +483,75 @@ int bg_hash_repartition_fused(const bg_column* key_cols, int32_t nkeys,
                               int64_t* d_offsets, uint32_t* d_rank,
                               void** d_out);
 
+/* ---- stage-interpreter support (bg_execute_stage internals that are
+ * also useful standalone) ---- */
+
+/* Place nbits of an Arrow LSB bitmap at an arbitrary bit offset of a
+ * destination bitmap (device shuffle-read: batch validity bitmaps land at
+ * non-byte-aligned row cursors).  Boundary words use atomics. */
+int bg_bitcopy(const uint8_t* d_src_bits, int64_t nbits, uint8_t* d_dst_bits,
+               int64_t dst_bit_off);
+
+/* Materialize bg_hashagg accumulator records into an Arrow column:
+ * acc_stride in bytes between consecutive groups' records for this
+ * aggregate; op = BG_AGG_OP_*; d_out element size 16 (SUM_DEC128) or 8.
+ * With d_nncnt (stride bytes), groups whose non-null input count is 0 get
+ * their bit cleared in d_valid_out (SQL: SUM/MIN/MAX of all-NULL = NULL). */
+int bg_agg_materialize(const void* d_acc, int64_t acc_stride, int32_t op,
+                       int64_t ngroups, void* d_out, const int64_t* d_nncnt,
+                       int64_t nncnt_stride, uint8_t* d_valid_out);
+
+/* Strided i64 copy (COUNT materialisation from interleaved agg records). */
+int bg_copy_i64_strided(const void* d_src, int64_t stride, int64_t n,
+                        void* d_out);
+
+/* AVG finalisation (AggregateExec Final for avg):
+ * decimal — DataFusion's decimal AvgAccumulator: target scale = input
+ * scale + 4, out = round-half-away-from-zero(sum * 10^scale_shift / count),
+ * exact i128; f64 — IEEE sum/count.  count==0 clears the group's valid
+ * bit. */
+int bg_avg_finalize(const void* d_sum_acc, int64_t acc_stride, int32_t is_f64,
+                    const void* d_cnt, int64_t cnt_stride,
+                    int32_t scale_shift, int64_t ngroups, void* d_out,
+                    uint8_t* d_valid_out);
+
+/* ---- stage interpreter (the product entry of SURVEY.md §8b seam 1) ----
+ *
+ * bg_execute_stage executes ONE whole query stage — the GPU analogue of
+ * QueryStageExecutor::execute_query_stage (executor/src/execution_engine.
+ * rs:78-103, default impl :127-212): operators are sequenced by a plan
+ * tree, key EXPRESSIONS are evaluated before hashing (sort_shuffle/
+ * writer.rs:1265), and the root writer produces the reference's exact
+ * shuffle file bytes + ShuffleWritePartition summaries (proto :779-791).
+ *
+ * plan_json is a faithful JSON restatement of the decoded TaskDefinition
+ * plan (the Rust host decodes task.plan with datafusion-proto exactly as
+ * execution_loop.rs:364-367 does, then serialises the operator tree to
+ * this schema — see INTEGRATION.md "Stage plan JSON").  Operators:
+ * scan / filter / project / hash_join / hash_aggregate / sort /
+ * sort_shuffle_write / passthrough_write / collect.
+ *
+ * On success *out_json receives a malloc'd result document:
+ *   {"partitions": [{"partition_id", "path", "num_batches", "num_rows",
+ *     "num_bytes"}...], "metrics": {...}, "rows": [...] (collect roots)}
+ * Free it with bg_stage_free.  On failure returns a bg_status and
+ * bg_last_error() describes the offending plan node. */
+int bg_execute_stage(const char* plan_json, char** out_json);
+void bg_stage_free(char* p);
+
+/* Parse + type-check a stage plan without touching the GPU (host-side
+ * validation; CPU tests cover the plan grammar through this). */
+int bg_stage_validate(const char* plan_json, char** out_json);
+
+/* Register a device-resident input table for "device" scan sources
+ * (bench/tests feed synthetic in-HBM tables; the Rust host feeds decoded
+ * ShuffleReaderExec batches the same way).  The caller owns the column
+ * buffers; they must outlive the stage. */
+int bg_stage_register_table(const char* name, const bg_column* cols,
+                            const char* const* col_names, int32_t ncols,
+                            int64_t n_rows);
+int bg_stage_unregister_table(const char* name);
+
 /* ---- fused filter+aggregate stages ---- */
 
 /* TPC-H q6 stage 1 (scan+filter+aggregate, approved/q6.txt):

@@ -1,0 +1,374 @@
+"""GPU parity tests of the C++ stage interpreter (bg_execute_stage): whole
+stages — scan -> filter/project/join/aggregate/sort -> shuffle write /
+collect — driven by plan JSON, checked against pyarrow compute (the same
+arrow implementation family the reference executes on) and the oracle.
+
+The interpreter is the product path of SURVEY.md §8b seam 1: one C call
+per task, mirroring QueryStageExecutor::execute_query_stage
+(executor/src/execution_engine.rs:78-103)."""
+import decimal
+import json
+import os
+
+import numpy as np
+import pyarrow as pa
+import pyarrow.compute as pc
+import pytest
+
+import oracle
+from datafusion_ballista_amd import gpu, shuffle, stage
+
+pytestmark = pytest.mark.gpu
+
+
+@pytest.fixture(scope="module")
+def ctx():
+    c = gpu.GpuStageContext(0)
+    yield c
+    c.close()
+
+
+@pytest.fixture()
+def reg(ctx):
+    """Register tables for a test; unregister afterwards."""
+    keep = []
+    names = []
+
+    def _reg(name, table):
+        keep.append(stage.register_table(ctx, name, table))
+        names.append(name)
+        return table
+
+    yield _reg
+    for n in names:
+        stage.unregister_table(n)
+
+
+def _doc(plan, **kw):
+    d = {"job_id": "job-s", "stage_id": 1, "task_id": 0,
+         "work_dir": "/tmp/stage-tests", "plan": plan}
+    d.update(kw)
+    return d
+
+
+def scan_of(table, name):
+    return {"op": "scan", "schema": stage.schema_json(table.schema),
+            "source": {"kind": "device", "table": name}}
+
+
+def make_lineitem(n, seed=0):
+    rng = np.random.default_rng(seed)
+    qty = rng.integers(100, 5100, size=n)          # scale 2
+    price = rng.integers(90000, 10500000, size=n)  # scale 2
+    disc = rng.integers(0, 11, size=n) * 100       # scale 4-ish (use 2)
+    ship = rng.integers(8400, 9800, size=n).astype(np.int32)
+    okey = rng.integers(0, n // 4 + 1, size=n, dtype=np.int64)
+    return pa.table({
+        "l_orderkey": pa.array(okey),
+        "l_quantity": pa.array([decimal.Decimal(int(v)) / 100 for v in qty],
+                               type=pa.decimal128(15, 2)),
+        "l_extendedprice": pa.array(
+            [decimal.Decimal(int(v)) / 100 for v in price],
+            type=pa.decimal128(15, 2)),
+        "l_discount": pa.array([decimal.Decimal(int(v)) / 100 for v in disc],
+                               type=pa.decimal128(15, 2)),
+        "l_shipdate": pa.array(ship, type=pa.date32()),
+    })
+
+
+def test_collect_filter_project(ctx, reg):
+    t = make_lineitem(50_000, seed=1)
+    reg("li", t)
+    plan = {"op": "collect", "input": {
+        "op": "project", "exprs": [
+            {"as": "okey", "expr": {"col": "l_orderkey"}},
+            {"as": "rev", "expr": {"mul": [{"col": "l_extendedprice"},
+                                           {"col": "l_discount"}]}},
+        ],
+        "input": {"op": "filter", "predicates": [
+            {"col": "l_shipdate", "cmp": "ge_lt", "lo": 8500, "hi": 8600}],
+            "input": scan_of(t, "li")}}}
+    res = stage.execute(_doc(plan))
+    mask = pc.and_(pc.greater_equal(t["l_shipdate"], pa.scalar(8500,
+                                                               pa.date32())),
+                   pc.less(t["l_shipdate"], pa.scalar(8600, pa.date32())))
+    want = t.filter(mask)
+    assert len(res["rows"]) == want.num_rows
+    got_keys = [r[0] for r in res["rows"]]
+    assert got_keys == want["l_orderkey"].to_pylist()
+    # revenue = price * discount exactly (scale 4)
+    wp = want["l_extendedprice"].to_pylist()
+    wd = want["l_discount"].to_pylist()
+    got_rev = [int(r[1]) for r in res["rows"]]
+    want_rev = [int(p.scaleb(2)) * int(d.scaleb(2)) for p, d in zip(wp, wd)]
+    assert got_rev == want_rev
+
+
+def test_q6_stage_fused_matches_oracle(ctx, reg):
+    from datafusion_ballista_amd import tpch_synth
+    n = 200_000
+    li = tpch_synth.lineitem_numpy(n, seed=7)
+    d16 = tpch_synth.dec128_pairs_np(li["l_discount"]).view(np.uint8).reshape(-1)
+    q16 = tpch_synth.dec128_pairs_np(li["l_quantity"]).view(np.uint8).reshape(-1)
+    p16 = tpch_synth.dec128_pairs_np(li["l_extendedprice"]).view(np.uint8).reshape(-1)
+
+    def dec_col(raw16):
+        vals = [int.from_bytes(bytes(raw16[16*i:16*(i+1)]), "little",
+                               signed=True) for i in range(n)]
+        return pa.array([decimal.Decimal(v).scaleb(-2) for v in vals],
+                        type=pa.decimal128(15, 2))
+
+    t = pa.table({
+        "l_shipdate": pa.array(li["l_shipdate"], type=pa.date32()),
+        "l_discount": dec_col(d16),
+        "l_quantity": dec_col(q16),
+        "l_extendedprice": dec_col(p16),
+    })
+    reg("li6", t)
+    plan = {"op": "collect", "input": {
+        "op": "hash_aggregate", "mode": "single", "group_by": [],
+        "aggs": [
+            {"fn": "sum", "as": "revenue",
+             "expr": {"mul": [{"col": "l_extendedprice"},
+                              {"col": "l_discount"}]}},
+            {"fn": "count", "as": "cnt"},
+        ],
+        "input": {"op": "filter", "predicates": [
+            {"col": "l_shipdate", "cmp": "ge_lt",
+             "lo": tpch_synth.Q6_DATE_LO, "hi": tpch_synth.Q6_DATE_HI},
+            {"col": "l_discount", "cmp": "between",
+             "lo": tpch_synth.Q6_DISC_LO, "hi": tpch_synth.Q6_DISC_HI},
+            {"col": "l_quantity", "cmp": "lt", "hi": tpch_synth.Q6_QTY_LT}],
+            "input": scan_of(t, "li6")}}}
+    res = stage.execute(_doc(plan))
+    want_cnt, want_sum = oracle.q6(li["l_shipdate"], d16, q16, p16,
+                                   tpch_synth.Q6_DATE_LO,
+                                   tpch_synth.Q6_DATE_HI,
+                                   tpch_synth.Q6_DISC_LO,
+                                   tpch_synth.Q6_DISC_HI,
+                                   tpch_synth.Q6_QTY_LT)
+    assert len(res["rows"]) == 1
+    assert int(res["rows"][0][0]) == want_sum
+    assert res["rows"][0][1] == want_cnt
+    # the fused kernel must actually have run (kernel_ms recorded)
+    assert res["metrics"]["gpu_kernel_ms"] > 0
+
+
+def test_grouped_aggregate_with_nulls(ctx, reg):
+    rng = np.random.default_rng(11)
+    n = 80_000
+    keys = rng.integers(0, 500, size=n, dtype=np.int64)
+    vals = rng.integers(-10**6, 10**6, size=n, dtype=np.int64)
+    vmask = rng.random(n) < 0.3  # 30% NULL values
+    f64 = rng.standard_normal(n)
+    t = pa.table({
+        "k": pa.array(keys),
+        "v": pa.array(vals, mask=vmask),
+        "x": pa.array(f64),
+    })
+    reg("g", t)
+    plan = {"op": "collect", "input": {
+        "op": "hash_aggregate", "mode": "single", "group_by": ["k"],
+        "aggs": [
+            {"fn": "sum", "as": "sv", "expr": {"col": "v"}},
+            {"fn": "min", "as": "mn", "expr": {"col": "v"}},
+            {"fn": "max", "as": "mx", "expr": {"col": "v"}},
+            {"fn": "count", "as": "c"},
+            {"fn": "avg", "as": "ax", "expr": {"col": "x"}},
+        ], "input": scan_of(t, "g")}}
+    res = stage.execute(_doc(plan))
+    want = t.group_by("k").aggregate([
+        ("v", "sum"), ("v", "min"), ("v", "max"), ("k", "count"),
+        ("x", "mean")]).sort_by("k")
+    got = sorted(res["rows"], key=lambda r: r[0])
+    assert len(got) == want.num_rows
+    for i, r in enumerate(got):
+        assert r[0] == want["k"][i].as_py()
+        assert r[1] == want["v_sum"][i].as_py()
+        assert r[2] == want["v_min"][i].as_py()
+        assert r[3] == want["v_max"][i].as_py()
+        assert r[4] == want["k_count"][i].as_py()
+        assert abs(r[5] - want["x_mean"][i].as_py()) < 1e-9
+
+
+def test_join_stage(ctx, reg):
+    rng = np.random.default_rng(13)
+    nb, np_ = 3_000, 40_000
+    ck = np.arange(nb, dtype=np.int64)
+    seg = rng.integers(0, 5, size=nb).astype(np.uint8)
+    ok = rng.integers(0, nb * 2, size=np_, dtype=np.int64)  # half miss
+    od = rng.integers(9000, 9400, size=np_).astype(np.int32)
+    cust = pa.table({"c_custkey": pa.array(ck), "c_seg": pa.array(seg)})
+    orders = pa.table({"o_custkey": pa.array(ok),
+                       "o_orderdate": pa.array(od, type=pa.date32()),
+                       "o_orderkey": pa.array(
+                           np.arange(np_, dtype=np.int64))})
+    reg("cust", cust)
+    reg("ord", orders)
+    plan = {"op": "collect", "input": {
+        "op": "hash_join",
+        "build": {"op": "filter",
+                  "predicates": [{"col": "c_seg", "cmp": "eq", "lo": 2}],
+                  "input": scan_of(cust, "cust")},
+        "probe": scan_of(orders, "ord"),
+        "build_keys": ["c_custkey"], "probe_keys": ["o_custkey"],
+        "join_type": "inner",
+        "output": [{"side": "probe", "col": "o_orderkey"},
+                   {"side": "probe", "col": "o_orderdate"},
+                   {"side": "build", "col": "c_custkey"}]}}
+    res = stage.execute(_doc(plan))
+    want_keys = set(ck[seg == 2].tolist())
+    want_pairs = [(int(k), int(d), int(c)) for k, d, c in
+                  zip(np.arange(np_), od, ok) if c in want_keys]
+    got_pairs = [(r[0], r[1], r[2]) for r in res["rows"]]
+    assert sorted(got_pairs) == sorted(want_pairs)
+
+
+def test_sort_topk_stage(ctx, reg):
+    rng = np.random.default_rng(17)
+    n = 30_000
+    t = pa.table({"a": pa.array(rng.integers(0, 10**9, size=n,
+                                             dtype=np.int64)),
+                  "b": pa.array(rng.integers(0, 100, size=n,
+                                             dtype=np.int64))})
+    reg("s", t)
+    plan = {"op": "collect", "input": {
+        "op": "sort", "keys": [{"col": "a", "desc": True}], "limit": 25,
+        "input": scan_of(t, "s")}}
+    res = stage.execute(_doc(plan))
+    want = t.sort_by([("a", "descending")]).slice(0, 25)
+    assert [r[0] for r in res["rows"]] == want["a"].to_pylist()
+
+
+def test_sort_shuffle_write_roundtrip(ctx, reg, tmp_path):
+    """The consolidated file the C++ writer emits must be byte-compatible:
+    pyarrow's own IPC reader (via shuffle.read_partition) reconstructs
+    every partition exactly, and the index/partition assignment matches
+    the oracle's compute_partition_indices restatement."""
+    rng = np.random.default_rng(19)
+    n, k = 57_000, 8
+    keys = rng.integers(-2**62, 2**62, size=n, dtype=np.int64)
+    vals = rng.integers(-10**9, 10**9, size=n, dtype=np.int64)
+    vmask = rng.random(n) < 0.15
+    strs = [f"s{int(v) % 997}" if v % 7 else "" for v in vals]
+    t = pa.table({
+        "k": pa.array(keys),
+        "v": pa.array(vals, mask=vmask),
+        "s": pa.array(strs, type=pa.string()),
+    })
+    reg("sw", t)
+    doc = _doc({"op": "sort_shuffle_write", "k": k,
+                "keys": [{"col": "k"}],
+                "input": scan_of(t, "sw")},
+               work_dir=str(tmp_path), stage_id=3, task_id=5,
+               schema_msg_hex=stage.schema_msg_hex(t.schema),
+               batch_size=4096)
+    res = stage.execute(doc)
+    parts = res["partitions"]
+    assert len(parts) == k
+    assert sum(p["num_rows"] for p in parts) == n
+    data_path = parts[0]["path"]
+    assert data_path.endswith(
+        os.path.join(str(tmp_path), "job-s", "3", "5", "data.arrow"))
+
+    # oracle partition assignment
+    h = oracle.hash_columns([("i64", keys)], n)
+    pids = oracle.partition_ids(h, k)
+    idx, offs = oracle.partition_indices(pids, k)
+
+    schema = shuffle.read_schema(data_path)
+    assert schema.equals(t.schema)
+    for p in range(k):
+        batches = shuffle.read_partition(data_path, data_path + ".index", p)
+        got = (pa.Table.from_batches(batches, schema=schema)
+               if batches else t.schema.empty_table())
+        rows = idx[offs[p]:offs[p + 1]]
+        want = t.take(pa.array(rows, type=pa.uint32()))
+        assert got.equals(want), f"partition {p}"
+        assert got.num_rows == parts[p]["num_rows"]
+        nb_expect = (got.num_rows + 4095) // 4096
+        assert parts[p]["num_batches"] == nb_expect
+
+
+def test_two_phase_aggregate_over_shuffle(ctx, reg, tmp_path):
+    """Partial aggregate -> sort_shuffle_write -> (scan shuffle) -> final
+    aggregate == single-shot aggregate: the whole two-stage exchange runs
+    through the interpreter, files in between (the reference's execution
+    shape for FinalPartitioned)."""
+    rng = np.random.default_rng(23)
+    n, k = 120_000, 4
+    keys = rng.integers(0, 3_000, size=n, dtype=np.int64)
+    vals = rng.integers(-10**8, 10**8, size=n, dtype=np.int64)
+    vmask = rng.random(n) < 0.2
+    t = pa.table({"k": pa.array(keys), "v": pa.array(vals, mask=vmask)})
+    reg("tp", t)
+
+    partial = {"op": "hash_aggregate", "mode": "partial", "group_by": ["k"],
+               "aggs": [{"fn": "sum", "as": "sv", "expr": {"col": "v"}},
+                        {"fn": "count", "as": "c"},
+                        {"fn": "avg", "as": "av", "expr": {"col": "v"}}],
+               "input": scan_of(t, "tp")}
+    part_schema = pa.schema([
+        ("k", pa.int64()), ("sv", pa.decimal128(38, 0)), ("sv$n", pa.int64()),
+        ("c", pa.int64()), ("av$s", pa.decimal128(38, 0)),
+        ("av$n", pa.int64())])
+    # NOTE: SUM(Int64) partial is int64 in Arrow; here we declare the
+    # partial sum columns as the interpreter emits them (int64)
+    part_schema = pa.schema([
+        ("k", pa.int64()), ("sv", pa.int64()), ("sv$n", pa.int64()),
+        ("c", pa.int64()), ("av$s", pa.int64()), ("av$n", pa.int64())])
+    doc1 = _doc({"op": "sort_shuffle_write", "k": k,
+                 "keys": [{"col": "k"}], "input": partial},
+                work_dir=str(tmp_path), stage_id=1, task_id=0,
+                schema_msg_hex=stage.schema_msg_hex(part_schema))
+    res1 = stage.execute(doc1)
+    data_path = res1["partitions"][0]["path"]
+
+    # stage 2: scan every partition of the shuffle file, final-aggregate
+    sc2 = {"op": "scan",
+           "schema": stage.schema_json(part_schema),
+           "source": {"kind": "shuffle", "data": data_path,
+                      "index": data_path + ".index",
+                      "partitions": list(range(k))}}
+    final = {"op": "hash_aggregate", "mode": "final", "group_by": ["k"],
+             "aggs": [{"fn": "sum", "as": "sv"}, {"fn": "count", "as": "c"},
+                      {"fn": "avg", "as": "av"}], "input": sc2}
+    res2 = stage.execute(_doc({"op": "collect", "input": final}))
+
+    df = t.group_by("k").aggregate(
+        [("v", "sum"), ("k", "count"), ("v", "mean")]).sort_by("k")
+    got = sorted(res2["rows"], key=lambda r: r[0])
+    assert len(got) == df.num_rows
+    for i, r in enumerate(got):
+        assert r[0] == df["k"][i].as_py()
+        assert r[1] == df["v_sum"][i].as_py()
+        assert r[2] == df["k_count"][i].as_py()
+        want_avg = df["v_mean"][i].as_py()
+        # decimal avg at scale +4 vs float mean
+        assert abs(float(r[3]) / 10**4 - want_avg) < 0.51 / 10**4
+
+
+def test_passthrough_write_stage(ctx, reg, tmp_path):
+    rng = np.random.default_rng(29)
+    n = 23_000
+    t = pa.table({"a": pa.array(rng.integers(0, 10**9, size=n,
+                                             dtype=np.int64)),
+                  "s": pa.array([f"v{i%311}" for i in range(n)])})
+    reg("pt", t)
+    doc = _doc({"op": "passthrough_write", "global_partition": 7,
+                "input": scan_of(t, "pt")},
+               work_dir=str(tmp_path), stage_id=2, task_id=9,
+               schema_msg_hex=stage.schema_msg_hex(t.schema))
+    res = stage.execute(doc)
+    path = res["partitions"][0]["path"]
+    assert path.endswith(os.path.join("2", "7", "data-9.arrow"))
+    got = pa.Table.from_batches(shuffle.read_passthrough_partition(path),
+                                schema=t.schema)
+    assert got.equals(t)
+
+
+def test_stage_errors_fail_loudly(ctx):
+    with pytest.raises(RuntimeError, match="unregistered device table"):
+        stage.execute(_doc({"op": "collect", "input": {
+            "op": "scan", "schema": [{"name": "x", "dtype": "int64"}],
+            "source": {"kind": "device", "table": "missing-table"}}}))
