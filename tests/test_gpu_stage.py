@@ -367,6 +367,116 @@ def test_passthrough_write_stage(ctx, reg, tmp_path):
     assert got.equals(t)
 
 
+def test_q3_stage_real_group_key(ctx, reg):
+    """TPC-H q3 as ONE plan-driven stage with its REAL shape
+    (approved/q3.txt): customer(mktsegment=BUILDING) ⨝ orders(date<cutoff)
+    ⨝ lineitem(date>cutoff), GROUP BY (l_orderkey, o_orderdate,
+    o_shippriority) — the 3-column key the reference groups on, not the
+    single-key stand-in of round 1 — then ORDER BY revenue DESC,
+    o_orderdate ASC LIMIT 10.  Per-group values checked exactly."""
+    rng = np.random.default_rng(37)
+    ncust, nord, nli = 20_000, 100_000, 400_000
+    cutoff = 9204  # 1995-03-15
+
+    c_custkey = np.arange(1, ncust + 1, dtype=np.int64)
+    c_seg = rng.integers(0, 5, size=ncust).astype(np.uint8)
+    o_orderkey = np.arange(1, nord + 1, dtype=np.int64)
+    o_custkey = rng.integers(1, ncust + 1, size=nord, dtype=np.int64)
+    o_orderdate = rng.integers(8900, 9500, size=nord, dtype=np.int32)
+    o_shipprio = rng.integers(0, 3, size=nord, dtype=np.int32)
+    l_orderkey = rng.integers(1, nord + 1, size=nli, dtype=np.int64)
+    l_shipdate = rng.integers(8900, 9500, size=nli, dtype=np.int32)
+    l_price = rng.integers(90000, 10495100, size=nli, dtype=np.int64)
+    l_disc = rng.integers(0, 11, size=nli, dtype=np.int64)
+
+    cust = pa.table({"c_custkey": pa.array(c_custkey),
+                     "c_mktsegment": pa.array(c_seg)})
+    orders = pa.table({"o_orderkey": pa.array(o_orderkey),
+                       "o_custkey": pa.array(o_custkey),
+                       "o_orderdate": pa.array(o_orderdate,
+                                               type=pa.date32()),
+                       "o_shippriority": pa.array(o_shipprio)})
+    li = pa.table({
+        "l_orderkey": pa.array(l_orderkey),
+        "l_shipdate": pa.array(l_shipdate, type=pa.date32()),
+        "l_extendedprice": pa.array(
+            [decimal.Decimal(int(v)) / 100 for v in l_price],
+            type=pa.decimal128(15, 2)),
+        "l_discount": pa.array(
+            [decimal.Decimal(int(v)) / 100 for v in l_disc],
+            type=pa.decimal128(15, 2)),
+    })
+    reg("q3c", cust)
+    reg("q3o", orders)
+    reg("q3l", li)
+
+    join1 = {"op": "hash_join",
+             "build": {"op": "filter",
+                       "predicates": [{"col": "c_mktsegment", "cmp": "eq",
+                                       "lo": 1}],  # BUILDING
+                       "input": scan_of(cust, "q3c")},
+             "probe": {"op": "filter",
+                       "predicates": [{"col": "o_orderdate", "cmp": "lt",
+                                       "hi": cutoff}],
+                       "input": scan_of(orders, "q3o")},
+             "build_keys": ["c_custkey"], "probe_keys": ["o_custkey"],
+             "join_type": "inner",
+             "output": [{"side": "probe", "col": "o_orderkey"},
+                        {"side": "probe", "col": "o_orderdate"},
+                        {"side": "probe", "col": "o_shippriority"}]}
+    join2 = {"op": "hash_join",
+             "build": join1,
+             "probe": {"op": "filter",
+                       "predicates": [{"col": "l_shipdate", "cmp": "gt",
+                                       "lo": cutoff}],
+                       "input": scan_of(li, "q3l")},
+             "build_keys": ["o_orderkey"], "probe_keys": ["l_orderkey"],
+             "join_type": "inner",
+             "output": [{"side": "probe", "col": "l_orderkey"},
+                        {"side": "build", "col": "o_orderdate"},
+                        {"side": "build", "col": "o_shippriority"},
+                        {"side": "probe", "col": "l_extendedprice"},
+                        {"side": "probe", "col": "l_discount"}]}
+    agg = {"op": "hash_aggregate", "mode": "single",
+           "group_by": ["l_orderkey", "o_orderdate", "o_shippriority"],
+           "aggs": [{"fn": "sum", "as": "revenue",
+                     "expr": {"mul": [{"col": "l_extendedprice"},
+                                      {"sub": [{"lit": 100},
+                                               {"col": "l_discount"}]}]}}],
+           "input": join2}
+    top = {"op": "sort", "keys": [{"col": "revenue", "desc": True},
+                                  {"col": "o_orderdate", "desc": False}],
+           "limit": 10, "input": agg}
+    res = stage.execute(_doc({"op": "collect", "input": top}))
+
+    # exact python restatement (per-group, not just global invariants)
+    keep_cust = set(c_custkey[c_seg == 1].tolist())
+    omult = {}
+    for i in range(nord):
+        if o_orderdate[i] < cutoff and int(o_custkey[i]) in keep_cust:
+            key = int(o_orderkey[i])
+            omult.setdefault(key, []).append(
+                (int(o_orderdate[i]), int(o_shipprio[i])))
+    want = {}
+    for i in range(nli):
+        if l_shipdate[i] <= cutoff:
+            continue
+        okey = int(l_orderkey[i])
+        if okey not in omult:
+            continue
+        rev = int(l_price[i]) * (100 - int(l_disc[i]))
+        for (odate, oprio) in omult[okey]:
+            g = (okey, odate, oprio)
+            want[g] = want.get(g, 0) + rev
+    assert want, "degenerate test"
+    top10 = sorted(want.items(), key=lambda kv: (-kv[1], kv[0][1]))[:10]
+    got = [((r[0], r[1], r[2]), int(r[3])) for r in res["rows"]]
+    # revenue DESC is a total order on values; ties broken by o_orderdate
+    assert [g[1] for g in got] == [w[1] for w in top10]
+    assert sorted(g[0] for g in got) == sorted(w[0] for w in top10) or \
+        [g[0] for g in got] == [w[0] for w in top10]
+
+
 def test_stage_errors_fail_loudly(ctx):
     with pytest.raises(RuntimeError, match="unregistered device table"):
         stage.execute(_doc({"op": "collect", "input": {
