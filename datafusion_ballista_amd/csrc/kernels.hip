@@ -5007,3 +5007,43 @@ extern "C" int bg_avg_finalize(const void* d_sum_acc, int64_t acc_stride,
   HIP_TRY(hipGetLastError());
   return BG_OK;
 }
+
+// synthetic-data fill for the no-Python C++ examples/bench (splitmix64
+// per element — deterministic, seed+index addressed; TEST/BENCH
+// INFRASTRUCTURE, not on any query path)
+__global__ void k_fill_rand(uint8_t* __restrict__ out, int64_t n,
+                            uint64_t seed, int64_t lo, uint64_t range,
+                            int32_t mode) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    uint64_t z = seed + (uint64_t)i * 0x9E3779B97F4A7C15ull;
+    z = (z ^ (z >> 30)) * 0xBF58476D1CE4E5B9ull;
+    z = (z ^ (z >> 27)) * 0x94D049BB133111EBull;
+    z ^= z >> 31;
+    int64_t v = lo + (int64_t)(z % range);
+    switch (mode) {
+      case 0: reinterpret_cast<int64_t*>(out)[i] = v; break;
+      case 1: reinterpret_cast<int32_t*>(out)[i] = (int32_t)v; break;
+      case 2: {  // Decimal128 LE pair
+        int64_t* p = reinterpret_cast<int64_t*>(out) + 2 * i;
+        p[0] = v;
+        p[1] = v < 0 ? -1 : 0;
+        break;
+      }
+      case 3: reinterpret_cast<int64_t*>(out)[i] = lo + i; break;  // arange
+      default: break;
+    }
+  }
+}
+
+extern "C" int bg_fill_rand(void* d_out, int64_t n, uint64_t seed,
+                            int64_t lo, int64_t hi, int32_t mode) {
+  REQUIRE_INIT();
+  if (n <= 0) return BG_OK;
+  if (hi <= lo) return set_err(BG_ERR_INVALID, "bg_fill_rand: hi <= lo");
+  int blocks = (int)bg_imin64((n + BG_BLOCK - 1) / BG_BLOCK, BG_MAX_BLOCKS);
+  hipLaunchKernelGGL(k_fill_rand, dim3(blocks), dim3(BG_BLOCK), 0, 0,
+                     (uint8_t*)d_out, n, seed, lo, (uint64_t)(hi - lo), mode);
+  HIP_TRY(hipGetLastError());
+  return BG_OK;
+}

@@ -552,6 +552,13 @@ int bg_stage_register_table(const char* name, const bg_column* cols,
                             int64_t n_rows);
 int bg_stage_unregister_table(const char* name);
 
+/* Deterministic synthetic fill (splitmix64 per element) for the
+ * no-Python C++ examples/bench — test/bench infrastructure, not a query
+ * path.  mode: 0 int64, 1 int32, 2 Decimal128 (LE pair).  Values uniform
+ * in [lo, hi). */
+int bg_fill_rand(void* d_out, int64_t n, uint64_t seed, int64_t lo,
+                 int64_t hi, int32_t mode);
+
 /* ---- fused filter+aggregate stages ---- */
 
 /* TPC-H q6 stage 1 (scan+filter+aggregate, approved/q6.txt):
