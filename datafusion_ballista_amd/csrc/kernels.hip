@@ -2747,30 +2747,215 @@ __global__ void k_snappy_decompress(const SnappyPage* pages, int64_t npages,
   }
 }
 
+// ---------------------------------------------------------------------------
+// 3-pass parallel Snappy (round 2): the wave-serial decoder above peaked at
+// 5.4 GB/s because ONE lane parsed tags and every element paid 4 wave
+// broadcasts before 64 lanes copied ~30 bytes.  The tag stream is the only
+// truly serial part, so:
+//   pass 1  k_snap_parse    — one THREAD per page walks the tag stream and
+//            materialises element descriptors {dst, src|off, len, kind}
+//            (no copying, no broadcasts; thousands of pages in flight);
+//   pass 2  k_snap_literals — every LITERAL copy is independent of all
+//            other elements (source = compressed stream), so the whole
+//            grid executes them wave-per-element, bandwidth-bound;
+//   pass 3  k_snap_matches  — LZ77 matches replay in order per page (one
+//            wave, 64-lane copies): a match window only references
+//            EARLIER output bytes, and those are complete (literals from
+//            pass 2, earlier matches from pass-3 order).
+// Descriptor bound: every element consumes >= 2 source bytes, so
+// nelems <= src_len/2 + 2.
+// ---------------------------------------------------------------------------
+
+struct SnapDesc {
+  uint32_t dst;   // output byte offset
+  uint32_t aux;   // literal: source offset in page; match: back offset
+  uint32_t len;
+  uint32_t kind;  // 0 literal, 1 match
+};
+
+__global__ void k_snap_parse(const SnappyPage* __restrict__ pages,
+                             int64_t npages, SnapDesc* __restrict__ descs,
+                             const int64_t* __restrict__ desc_base,
+                             int64_t* __restrict__ counts,
+                             int64_t* __restrict__ out_lens) {
+  for (int64_t p = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; p < npages;
+       p += (int64_t)gridDim.x * blockDim.x) {
+    const uint8_t* s = pages[p].src;
+    const int64_t src_len = pages[p].src_len;
+    SnapDesc* d = descs + desc_base[p];
+    int64_t si = 0, di = 0, ulen = 0, nd = 0;
+    int ok = 1, shift = 0;
+    while (si < src_len) {
+      const uint8_t b = s[si++];
+      ulen |= (int64_t)(b & 0x7f) << shift;
+      if (!(b & 0x80)) break;
+      shift += 7;
+      if (shift > 32) { ok = 0; break; }
+    }
+    if (ulen > pages[p].dst_cap) ok = 0;
+    while (ok && si < src_len && di < ulen) {
+      const uint8_t tag = s[si++];
+      if ((tag & 3) == 0) {
+        int64_t len = (tag >> 2) + 1;
+        if (len > 60) {
+          const int nb = (int)len - 60;
+          if (si + nb > src_len) { ok = 0; break; }
+          len = 0;
+          for (int i = 0; i < nb; ++i) len |= (int64_t)s[si + i] << (8 * i);
+          len += 1;
+          si += nb;
+        }
+        if (si + len > src_len || di + len > ulen) { ok = 0; break; }
+        d[nd].dst = (uint32_t)di;
+        d[nd].aux = (uint32_t)si;
+        d[nd].len = (uint32_t)len;
+        d[nd].kind = 0;
+        ++nd;
+        si += len;
+        di += len;
+      } else {
+        int64_t len = 0, off = 0;
+        if ((tag & 3) == 1) {
+          if (si >= src_len) { ok = 0; break; }
+          len = ((tag >> 2) & 7) + 4;
+          off = ((int64_t)(tag >> 5) << 8) | s[si];
+          si += 1;
+        } else if ((tag & 3) == 2) {
+          if (si + 2 > src_len) { ok = 0; break; }
+          len = (tag >> 2) + 1;
+          off = (int64_t)s[si] | ((int64_t)s[si + 1] << 8);
+          si += 2;
+        } else {
+          if (si + 4 > src_len) { ok = 0; break; }
+          len = (tag >> 2) + 1;
+          off = (int64_t)s[si] | ((int64_t)s[si + 1] << 8) |
+                ((int64_t)s[si + 2] << 16) | ((int64_t)s[si + 3] << 24);
+          si += 4;
+        }
+        if (off == 0 || off > di || di + len > ulen) { ok = 0; break; }
+        d[nd].dst = (uint32_t)di;
+        d[nd].aux = (uint32_t)off;
+        d[nd].len = (uint32_t)len;
+        d[nd].kind = 1;
+        ++nd;
+        di += len;
+      }
+    }
+    counts[p] = ok ? nd : 0;
+    out_lens[p] = (ok && di == ulen) ? ulen : -1;
+  }
+}
+
+// grid.y = page (chunked to <=65535), blocks-x * waves stride elements
+__global__ void k_snap_literals(const SnappyPage* __restrict__ pages,
+                                int64_t page0,
+                                const SnapDesc* __restrict__ descs,
+                                const int64_t* __restrict__ desc_base,
+                                const int64_t* __restrict__ counts) {
+  const int64_t p = page0 + blockIdx.y;
+  const uint8_t* src = pages[p].src;
+  uint8_t* dst = pages[p].dst;
+  const SnapDesc* d = descs + desc_base[p];
+  const int64_t nd = counts[p];
+  const int waves_per_block = blockDim.x / BG_WAVE;
+  const int64_t wave = (int64_t)blockIdx.x * waves_per_block +
+                       (int64_t)(threadIdx.x / BG_WAVE);
+  const int64_t wstride = (int64_t)gridDim.x * waves_per_block;
+  const int lane = threadIdx.x & (BG_WAVE - 1);
+  for (int64_t e = wave; e < nd; e += wstride) {
+    if (d[e].kind != 0) continue;
+    const uint8_t* s = src + d[e].aux;
+    uint8_t* o = dst + d[e].dst;
+    const int64_t len = d[e].len;
+    for (int64_t i = lane; i < len; i += BG_WAVE) o[i] = s[i];
+  }
+}
+
+__global__ void k_snap_matches(const SnappyPage* __restrict__ pages,
+                               int64_t npages,
+                               const SnapDesc* __restrict__ descs,
+                               const int64_t* __restrict__ desc_base,
+                               const int64_t* __restrict__ counts) {
+  const int64_t wave_global =
+      ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) / BG_WAVE;
+  const int64_t nwaves = ((int64_t)gridDim.x * blockDim.x) / BG_WAVE;
+  const int lane = lane_id();
+  for (int64_t p = wave_global; p < npages; p += nwaves) {
+    uint8_t* dst = pages[p].dst;
+    const SnapDesc* d = descs + desc_base[p];
+    const int64_t nd = counts[p];
+    for (int64_t e = 0; e < nd; ++e) {
+      if (d[e].kind != 1) continue;
+      const int64_t off = d[e].aux;
+      const int64_t len = d[e].len;
+      uint8_t* o = dst + d[e].dst;
+      const uint8_t* win = o - off;
+      if (off >= len) {
+        for (int64_t i = lane; i < len; i += BG_WAVE) o[i] = win[i];
+      } else {
+        for (int64_t i = lane; i < len; i += BG_WAVE) o[i] = win[i % off];
+      }
+      __builtin_amdgcn_wave_barrier();
+    }
+  }
+}
+
 /* Decompress npages independent Snappy blocks.  pages: HOST array copied
  * internally; each entry's src/dst are DEVICE pointers.  out_lens (host,
  * npages): decompressed length or -1 on malformed input. */
 extern "C" int bg_snappy_decompress(const void* h_pages, int64_t npages,
                                     int64_t* h_out_lens) {
   REQUIRE_INIT();
+  if (npages <= 0) return BG_OK;
+  const SnappyPage* hp = (const SnappyPage*)h_pages;
+  std::vector<int64_t> base(npages + 1);
+  base[0] = 0;
+  for (int64_t p = 0; p < npages; ++p)
+    base[p + 1] = base[p] + hp[p].src_len / 2 + 2;
   SnappyPage* d_pages;
   int64_t* d_lens;
-  HIP_TRY(pool_malloc((void**)&d_pages, sizeof(SnappyPage) * (npages ? npages : 1)));
-  HIP_TRY(pool_malloc((void**)&d_lens, sizeof(int64_t) * (npages ? npages : 1)));
+  int64_t* d_base;
+  int64_t* d_counts;
+  SnapDesc* d_descs;
+  HIP_TRY(pool_malloc((void**)&d_pages, sizeof(SnappyPage) * npages));
+  HIP_TRY(pool_malloc((void**)&d_lens, sizeof(int64_t) * npages));
+  HIP_TRY(pool_malloc((void**)&d_base, sizeof(int64_t) * (npages + 1)));
+  HIP_TRY(pool_malloc((void**)&d_counts, sizeof(int64_t) * npages));
+  HIP_TRY(pool_malloc((void**)&d_descs, sizeof(SnapDesc) * base[npages]));
   HIP_TRY(hipMemcpy(d_pages, h_pages, sizeof(SnappyPage) * npages,
                     hipMemcpyHostToDevice));
-  // one WAVE per page
+  HIP_TRY(hipMemcpy(d_base, base.data(), sizeof(int64_t) * (npages + 1),
+                    hipMemcpyHostToDevice));
+  int blocks1 = (int)bg_imin64((npages + BG_BLOCK - 1) / BG_BLOCK, BG_MAX_BLOCKS);
+  if (blocks1 == 0) blocks1 = 1;
+  hipLaunchKernelGGL(k_snap_parse, dim3(blocks1), dim3(BG_BLOCK), 0, 0,
+                     d_pages, npages, d_descs, d_base, d_counts, d_lens);
+  HIP_TRY(hipGetLastError());
+  // pass 2: chunk grid.y at 65535 pages
   const int waves_per_block = BG_BLOCK / BG_WAVE;
-  int blocks = (int)bg_imin64((npages + waves_per_block - 1) / waves_per_block,
-                              BG_MAX_BLOCKS);
-  if (blocks == 0) blocks = 1;
-  hipLaunchKernelGGL(k_snappy_decompress, dim3(blocks), dim3(BG_BLOCK), 0, 0,
-                     d_pages, npages, d_lens);
+  // enough x-blocks to spread big pages' literal lists over the chip
+  int xb = (int)bg_imin64(
+      (16384 + npages - 1) / (npages > 0 ? npages : 1) + 1, 1024);
+  for (int64_t p0 = 0; p0 < npages; p0 += 65535) {
+    uint32_t ny = (uint32_t)bg_imin64(npages - p0, 65535);
+    hipLaunchKernelGGL(k_snap_literals, dim3((uint32_t)xb, ny),
+                       dim3(BG_BLOCK), 0, 0, d_pages, p0, d_descs, d_base,
+                       d_counts);
+    HIP_TRY(hipGetLastError());
+  }
+  int blocks3 = (int)bg_imin64(
+      (npages + waves_per_block - 1) / waves_per_block, BG_MAX_BLOCKS);
+  if (blocks3 == 0) blocks3 = 1;
+  hipLaunchKernelGGL(k_snap_matches, dim3(blocks3), dim3(BG_BLOCK), 0, 0,
+                     d_pages, npages, d_descs, d_base, d_counts);
   HIP_TRY(hipGetLastError());
   HIP_TRY(hipMemcpy(h_out_lens, d_lens, sizeof(int64_t) * npages,
                     hipMemcpyDeviceToHost));
   (void)pool_release(d_pages);
   (void)pool_release(d_lens);
+  (void)pool_release(d_base);
+  (void)pool_release(d_counts);
+  (void)pool_release(d_descs);
   return BG_OK;
 }
 

@@ -148,7 +148,7 @@ static int run_shuffle(int64_t n, int k) {
   void* kcol = dmalloc((uint64_t)n * 8);
   void* vcol = dmalloc((uint64_t)n * 16);
   void* dcol = dmalloc((uint64_t)n * 4);
-  CHK(bg_fill_rand(kcol, n, 11, -(int64_t)1 << 60, (int64_t)1 << 60, 0));
+  CHK(bg_fill_rand(kcol, n, 11, -((int64_t)1 << 60), (int64_t)1 << 60, 0));
   CHK(bg_fill_rand(vcol, n, 12, -1000000000, 1000000000, 2));
   CHK(bg_fill_rand(dcol, n, 13, 8000, 11000, 1));
   bg_column cols[3] = {};
