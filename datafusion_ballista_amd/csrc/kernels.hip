@@ -2871,6 +2871,13 @@ __global__ void k_snap_literals(const SnappyPage* __restrict__ pages,
   }
 }
 
+// Match replay with 64-wide dependency batching: a snappy match is <= 64
+// bytes, so one LANE can own one match.  Load 64 consecutive descriptors
+// (coalesced); every match whose window ends BEFORE the batch's first
+// output byte only reads bytes from already-completed elements, so the
+// longest such prefix executes concurrently, one match per lane.  Only
+// genuinely chained matches (window into the current batch — tight RLE
+// runs) fall back to the one-element wave-parallel path.
 __global__ void k_snap_matches(const SnappyPage* __restrict__ pages,
                                int64_t npages,
                                const SnapDesc* __restrict__ descs,
@@ -2884,16 +2891,46 @@ __global__ void k_snap_matches(const SnappyPage* __restrict__ pages,
     uint8_t* dst = pages[p].dst;
     const SnapDesc* d = descs + desc_base[p];
     const int64_t nd = counts[p];
-    for (int64_t e = 0; e < nd; ++e) {
-      if (d[e].kind != 1) continue;
-      const int64_t off = d[e].aux;
-      const int64_t len = d[e].len;
-      uint8_t* o = dst + d[e].dst;
-      const uint8_t* win = o - off;
-      if (off >= len) {
-        for (int64_t i = lane; i < len; i += BG_WAVE) o[i] = win[i];
+    int64_t e = 0;
+    while (e < nd) {
+      const int64_t batch = nd - e < BG_WAVE ? nd - e : BG_WAVE;
+      SnapDesc my{};
+      if (lane < batch) my = d[e + lane];
+      const uint32_t batch_dst0 =
+          (uint32_t)__builtin_amdgcn_readfirstlane((int)my.dst);
+      // literals are pass-2 no-ops; matches eligible when their window is
+      // entirely before this batch's writes
+      bool elig = true;
+      if (lane < batch && my.kind == 1)
+        elig = ((int64_t)my.dst - (int64_t)my.aux + (int64_t)my.len <=
+                (int64_t)batch_dst0);
+      const uint64_t ball = __ballot(elig);
+      int prefix = (~ball == 0) ? BG_WAVE : __builtin_ctzll(~ball);
+      if (prefix > batch) prefix = (int)batch;
+      if (prefix == 0) {
+        // chained match at the head: whole wave executes it
+        const uint32_t h_dst =
+            (uint32_t)__builtin_amdgcn_readfirstlane((int)my.dst);
+        const uint32_t h_off =
+            (uint32_t)__builtin_amdgcn_readfirstlane((int)my.aux);
+        const uint32_t h_len =
+            (uint32_t)__builtin_amdgcn_readfirstlane((int)my.len);
+        uint8_t* o = dst + h_dst;
+        const uint8_t* win = o - h_off;
+        if (h_off >= h_len) {
+          for (uint32_t i = lane; i < h_len; i += BG_WAVE) o[i] = win[i];
+        } else {
+          for (uint32_t i = lane; i < h_len; i += BG_WAVE)
+            o[i] = win[i % h_off];
+        }
+        e += 1;
       } else {
-        for (int64_t i = lane; i < len; i += BG_WAVE) o[i] = win[i % off];
+        if (lane < prefix && my.kind == 1) {
+          uint8_t* o = dst + my.dst;
+          const uint8_t* win = o - my.aux;
+          for (uint32_t i = 0; i < my.len; ++i) o[i] = win[i];
+        }
+        e += prefix;
       }
       __builtin_amdgcn_wave_barrier();
     }
