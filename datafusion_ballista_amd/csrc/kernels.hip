@@ -24,6 +24,7 @@
 
 #include <hip/hip_runtime.h>
 
+#include <chrono>
 #include <cstdio>
 #include <cstdlib>
 #include <cstring>
@@ -2773,20 +2774,74 @@ struct SnapDesc {
   uint32_t kind;  // 0 literal, 1 match
 };
 
+// Register-buffered tag-stream cursor: the parse is strictly serial, so
+// its latency is the critical path — fetching tag bytes through a 16-B
+// register window (refilled with two aligned u64 loads) replaces ~5
+// dependent byte loads per element with ~1 refill per ~8 elements.
+struct SnapCursor {
+  const uint8_t* s;
+  int64_t len;
+  uint64_t w0, w1;
+  int64_t wbase;
+
+  __device__ void refill(int64_t pos) {
+    // align on the ABSOLUTE address (pages sit at arbitrary offsets of the
+    // uploaded chunk; the chunk base itself is allocator-aligned, so
+    // aligning down never leaves the allocation)
+    const uintptr_t abs0 = (uintptr_t)s + (uintptr_t)pos;
+    const uint8_t* ap = (const uint8_t*)(abs0 & ~(uintptr_t)7);
+    wbase = (int64_t)(ap - s);
+    if (wbase >= 0 && wbase + 16 <= len) {
+      const uint64_t* p = reinterpret_cast<const uint64_t*>(ap);
+      w0 = p[0];
+      w1 = p[1];
+    } else {  // tail/head: byte-safe fill (never read past the page)
+      w0 = w1 = 0;
+      for (int i = 0; i < 16 && wbase + i < len; ++i) {
+        if (wbase + i < 0) continue;
+        const uint64_t b = s[wbase + i];
+        if (i < 8) w0 |= b << (8 * i);
+        else w1 |= b << (8 * (i - 8));
+      }
+    }
+  }
+  __device__ uint8_t at(int64_t pos) {
+    int64_t off = pos - wbase;
+    if (off < 0 || off > 15) {
+      refill(pos);
+      off = pos - wbase;
+    }
+    const uint64_t w = off < 8 ? w0 : w1;
+    return (uint8_t)(w >> (8 * (off & 7)));
+  }
+};
+
+// One WAVE per page, executing the (serial) parse UNIFORMLY on all 64
+// lanes: same control flow and addresses everywhere, so tag-window loads
+// merge into one transaction, while the otherwise-idle lanes double as a
+// software prefetcher — each lane touches one cache line of the tag
+// stream ~4 KB ahead, so the serial chain's refills hit L2 instead of
+// paying a cold HBM round trip each 16 bytes.  Descriptor stores are
+// lane-0-gated.
 __global__ void k_snap_parse(const SnappyPage* __restrict__ pages,
                              int64_t npages, SnapDesc* __restrict__ descs,
                              const int64_t* __restrict__ desc_base,
                              int64_t* __restrict__ counts,
                              int64_t* __restrict__ out_lens) {
-  for (int64_t p = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; p < npages;
-       p += (int64_t)gridDim.x * blockDim.x) {
-    const uint8_t* s = pages[p].src;
+  const int lane = lane_id();
+  const int64_t wave_global =
+      ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) / BG_WAVE;
+  const int64_t nwaves = ((int64_t)gridDim.x * blockDim.x) / BG_WAVE;
+  for (int64_t p = wave_global; p < npages; p += nwaves) {
     const int64_t src_len = pages[p].src_len;
+    const uint8_t* __restrict__ s_pref = pages[p].src;
+    int64_t pref = 0;
+    SnapCursor cur{pages[p].src, src_len, 0, 0, -32};
     SnapDesc* d = descs + desc_base[p];
     int64_t si = 0, di = 0, ulen = 0, nd = 0;
     int ok = 1, shift = 0;
     while (si < src_len) {
-      const uint8_t b = s[si++];
+      const uint8_t b = cur.at(si++);
       ulen |= (int64_t)(b & 0x7f) << shift;
       if (!(b & 0x80)) break;
       shift += 7;
@@ -2794,22 +2849,25 @@ __global__ void k_snap_parse(const SnappyPage* __restrict__ pages,
     }
     if (ulen > pages[p].dst_cap) ok = 0;
     while (ok && si < src_len && di < ulen) {
-      const uint8_t tag = s[si++];
+      const uint8_t tag = cur.at(si++);
       if ((tag & 3) == 0) {
         int64_t len = (tag >> 2) + 1;
         if (len > 60) {
           const int nb = (int)len - 60;
           if (si + nb > src_len) { ok = 0; break; }
           len = 0;
-          for (int i = 0; i < nb; ++i) len |= (int64_t)s[si + i] << (8 * i);
+          for (int i = 0; i < nb; ++i)
+            len |= (int64_t)cur.at(si + i) << (8 * i);
           len += 1;
           si += nb;
         }
         if (si + len > src_len || di + len > ulen) { ok = 0; break; }
-        d[nd].dst = (uint32_t)di;
-        d[nd].aux = (uint32_t)si;
-        d[nd].len = (uint32_t)len;
-        d[nd].kind = 0;
+        if (lane == 0) {
+          d[nd].dst = (uint32_t)di;
+          d[nd].aux = (uint32_t)si;
+          d[nd].len = (uint32_t)len;
+          d[nd].kind = 0;
+        }
         ++nd;
         si += len;
         di += len;
@@ -2818,31 +2876,45 @@ __global__ void k_snap_parse(const SnappyPage* __restrict__ pages,
         if ((tag & 3) == 1) {
           if (si >= src_len) { ok = 0; break; }
           len = ((tag >> 2) & 7) + 4;
-          off = ((int64_t)(tag >> 5) << 8) | s[si];
+          off = ((int64_t)(tag >> 5) << 8) | cur.at(si);
           si += 1;
         } else if ((tag & 3) == 2) {
           if (si + 2 > src_len) { ok = 0; break; }
           len = (tag >> 2) + 1;
-          off = (int64_t)s[si] | ((int64_t)s[si + 1] << 8);
+          off = (int64_t)cur.at(si) | ((int64_t)cur.at(si + 1) << 8);
           si += 2;
         } else {
           if (si + 4 > src_len) { ok = 0; break; }
           len = (tag >> 2) + 1;
-          off = (int64_t)s[si] | ((int64_t)s[si + 1] << 8) |
-                ((int64_t)s[si + 2] << 16) | ((int64_t)s[si + 3] << 24);
+          off = (int64_t)cur.at(si) | ((int64_t)cur.at(si + 1) << 8) |
+                ((int64_t)cur.at(si + 2) << 16) |
+                ((int64_t)cur.at(si + 3) << 24);
           si += 4;
         }
         if (off == 0 || off > di || di + len > ulen) { ok = 0; break; }
-        d[nd].dst = (uint32_t)di;
-        d[nd].aux = (uint32_t)off;
-        d[nd].len = (uint32_t)len;
-        d[nd].kind = 1;
+        if (lane == 0) {
+          d[nd].dst = (uint32_t)di;
+          d[nd].aux = (uint32_t)off;
+          d[nd].len = (uint32_t)len;
+          d[nd].kind = 1;
+        }
         ++nd;
         di += len;
       }
+      // distributed prefetch: keep the tag stream ~4 KB ahead in L2
+      if (si + 2048 > pref && pref < src_len) {
+        const int64_t a = pref + (int64_t)lane * 64;
+        if (a < src_len) {
+          volatile uint8_t t = s_pref[a];
+          (void)t;
+        }
+        pref += 4096;
+      }
     }
-    counts[p] = ok ? nd : 0;
-    out_lens[p] = (ok && di == ulen) ? ulen : -1;
+    if (lane == 0) {
+      counts[p] = ok ? nd : 0;
+      out_lens[p] = (ok && di == ulen) ? ulen : -1;
+    }
   }
 }
 
@@ -2896,42 +2968,43 @@ __global__ void k_snap_matches(const SnappyPage* __restrict__ pages,
       const int64_t batch = nd - e < BG_WAVE ? nd - e : BG_WAVE;
       SnapDesc my{};
       if (lane < batch) my = d[e + lane];
-      const uint32_t batch_dst0 =
-          (uint32_t)__builtin_amdgcn_readfirstlane((int)my.dst);
-      // literals are pass-2 no-ops; matches eligible when their window is
-      // entirely before this batch's writes
-      bool elig = true;
-      if (lane < batch && my.kind == 1)
-        elig = ((int64_t)my.dst - (int64_t)my.aux + (int64_t)my.len <=
-                (int64_t)batch_dst0);
-      const uint64_t ball = __ballot(elig);
-      int prefix = (~ball == 0) ? BG_WAVE : __builtin_ctzll(~ball);
-      if (prefix > batch) prefix = (int)batch;
-      if (prefix == 0) {
-        // chained match at the head: whole wave executes it
-        const uint32_t h_dst =
-            (uint32_t)__builtin_amdgcn_readfirstlane((int)my.dst);
-        const uint32_t h_off =
-            (uint32_t)__builtin_amdgcn_readfirstlane((int)my.aux);
-        const uint32_t h_len =
-            (uint32_t)__builtin_amdgcn_readfirstlane((int)my.len);
-        uint8_t* o = dst + h_dst;
-        const uint8_t* win = o - h_off;
-        if (h_off >= h_len) {
-          for (uint32_t i = lane; i < h_len; i += BG_WAVE) o[i] = win[i];
+      // consume the WHOLE batch from registers: at step j, matches whose
+      // windows end before element j's output start can run one-per-lane
+      // concurrently; a chained head executes wave-parallel.  No
+      // descriptor reloads inside the batch.
+      int j = 0;
+      while (j < batch) {
+        const uint32_t dst_j = (uint32_t)__shfl((int)my.dst, j);
+        bool elig = lane >= j && lane < batch;
+        if (elig && my.kind == 1)
+          elig = ((int64_t)my.dst - (int64_t)my.aux + (int64_t)my.len <=
+                  (int64_t)dst_j);
+        uint64_t ball = __ballot(elig) >> j;
+        int prefix = (~ball == 0) ? BG_WAVE : __builtin_ctzll(~ball);
+        if (j + prefix > batch) prefix = (int)batch - j;
+        if (prefix == 0) {
+          const uint32_t h_off = (uint32_t)__shfl((int)my.aux, j);
+          const uint32_t h_len = (uint32_t)__shfl((int)my.len, j);
+          uint8_t* o = dst + dst_j;
+          const uint8_t* win = o - h_off;
+          if (h_off >= h_len) {
+            for (uint32_t i = lane; i < h_len; i += BG_WAVE) o[i] = win[i];
+          } else {
+            for (uint32_t i = lane; i < h_len; i += BG_WAVE)
+              o[i] = win[i % h_off];
+          }
+          j += 1;
         } else {
-          for (uint32_t i = lane; i < h_len; i += BG_WAVE)
-            o[i] = win[i % h_off];
+          if (lane >= j && lane < j + prefix && my.kind == 1) {
+            uint8_t* o = dst + my.dst;
+            const uint8_t* win = o - my.aux;
+            for (uint32_t i = 0; i < my.len; ++i) o[i] = win[i];
+          }
+          j += prefix;
         }
-        e += 1;
-      } else {
-        if (lane < prefix && my.kind == 1) {
-          uint8_t* o = dst + my.dst;
-          const uint8_t* win = o - my.aux;
-          for (uint32_t i = 0; i < my.len; ++i) o[i] = win[i];
-        }
-        e += prefix;
+        __builtin_amdgcn_wave_barrier();
       }
+      e += batch;
       __builtin_amdgcn_wave_barrier();
     }
   }
@@ -2963,11 +3036,24 @@ extern "C" int bg_snappy_decompress(const void* h_pages, int64_t npages,
                     hipMemcpyHostToDevice));
   HIP_TRY(hipMemcpy(d_base, base.data(), sizeof(int64_t) * (npages + 1),
                     hipMemcpyHostToDevice));
-  int blocks1 = (int)bg_imin64((npages + BG_BLOCK - 1) / BG_BLOCK, BG_MAX_BLOCKS);
+  const bool dbg = getenv("BG_SNAPPY_DEBUG") != nullptr;
+  double t_parse = 0, t_lit = 0, t_match = 0;
+  auto tick = [&]() -> double {
+    if (!dbg) return 0.0;
+    (void)hipDeviceSynchronize();
+    return (double)std::chrono::duration_cast<std::chrono::nanoseconds>(
+               std::chrono::steady_clock::now().time_since_epoch())
+               .count() / 1e6;
+  };
+  double t0 = tick();
+  const int wpb1 = BG_BLOCK / BG_WAVE;  // parse waves per block
+  int blocks1 = (int)bg_imin64((npages + wpb1 - 1) / wpb1, BG_MAX_BLOCKS);
   if (blocks1 == 0) blocks1 = 1;
   hipLaunchKernelGGL(k_snap_parse, dim3(blocks1), dim3(BG_BLOCK), 0, 0,
                      d_pages, npages, d_descs, d_base, d_counts, d_lens);
   HIP_TRY(hipGetLastError());
+  double t1 = tick();
+  t_parse = t1 - t0;
   // pass 2: chunk grid.y at 65535 pages
   const int waves_per_block = BG_BLOCK / BG_WAVE;
   // enough x-blocks to spread big pages' literal lists over the chip
@@ -2980,12 +3066,21 @@ extern "C" int bg_snappy_decompress(const void* h_pages, int64_t npages,
                        d_counts);
     HIP_TRY(hipGetLastError());
   }
+  double t2 = tick();
+  t_lit = t2 - t1;
   int blocks3 = (int)bg_imin64(
       (npages + waves_per_block - 1) / waves_per_block, BG_MAX_BLOCKS);
   if (blocks3 == 0) blocks3 = 1;
   hipLaunchKernelGGL(k_snap_matches, dim3(blocks3), dim3(BG_BLOCK), 0, 0,
                      d_pages, npages, d_descs, d_base, d_counts);
   HIP_TRY(hipGetLastError());
+  if (dbg) {
+    t_match = tick() - t2;
+    fprintf(stderr,
+            "[bg_snappy] npages=%lld parse=%.3fms literals=%.3fms "
+            "matches=%.3fms\n",
+            (long long)npages, t_parse, t_lit, t_match);
+  }
   HIP_TRY(hipMemcpy(h_out_lens, d_lens, sizeof(int64_t) * npages,
                     hipMemcpyDeviceToHost));
   (void)pool_release(d_pages);

@@ -74,8 +74,15 @@ def main():
     ctx = gpu.GpuStageContext(0)
     rng = np.random.default_rng(3)
     out = []
-    for kind in ("random", "lowcard", "runs"):
-        for page_kb, npages in ((1024, 160), (1024, 2000), (64, 4000)):
+    only = sys.argv[1] if len(sys.argv) > 1 else None
+    cases = [("random", 1024, 160), ("random", 1024, 2000),
+             ("random", 64, 4000), ("lowcard", 1024, 160),
+             ("lowcard", 1024, 2000), ("lowcard", 64, 4000),
+             ("runs", 1024, 160), ("runs", 1024, 2000), ("runs", 64, 4000)]
+    if only:
+        kind, page_kb, npages = only.split(",")
+        cases = [(kind, int(page_kb), int(npages))]
+    for kind, page_kb, npages in cases:
             pages = make_pages(kind, page_kb * 1024, npages, rng)
             comp = sum(len(p[0]) for p in pages)
             total_u, dt = bench(ctx, pages)
