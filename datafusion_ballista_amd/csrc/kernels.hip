@@ -2774,55 +2774,15 @@ struct SnapDesc {
   uint32_t kind;  // 0 literal, 1 match
 };
 
-// Register-buffered tag-stream cursor: the parse is strictly serial, so
-// its latency is the critical path — fetching tag bytes through a 16-B
-// register window (refilled with two aligned u64 loads) replaces ~5
-// dependent byte loads per element with ~1 refill per ~8 elements.
-struct SnapCursor {
-  const uint8_t* s;
-  int64_t len;
-  uint64_t w0, w1;
-  int64_t wbase;
-
-  __device__ void refill(int64_t pos) {
-    // align on the ABSOLUTE address (pages sit at arbitrary offsets of the
-    // uploaded chunk; the chunk base itself is allocator-aligned, so
-    // aligning down never leaves the allocation)
-    const uintptr_t abs0 = (uintptr_t)s + (uintptr_t)pos;
-    const uint8_t* ap = (const uint8_t*)(abs0 & ~(uintptr_t)7);
-    wbase = (int64_t)(ap - s);
-    if (wbase >= 0 && wbase + 16 <= len) {
-      const uint64_t* p = reinterpret_cast<const uint64_t*>(ap);
-      w0 = p[0];
-      w1 = p[1];
-    } else {  // tail/head: byte-safe fill (never read past the page)
-      w0 = w1 = 0;
-      for (int i = 0; i < 16 && wbase + i < len; ++i) {
-        if (wbase + i < 0) continue;
-        const uint64_t b = s[wbase + i];
-        if (i < 8) w0 |= b << (8 * i);
-        else w1 |= b << (8 * (i - 8));
-      }
-    }
-  }
-  __device__ uint8_t at(int64_t pos) {
-    int64_t off = pos - wbase;
-    if (off < 0 || off > 15) {
-      refill(pos);
-      off = pos - wbase;
-    }
-    const uint64_t w = off < 8 ? w0 : w1;
-    return (uint8_t)(w >> (8 * (off & 7)));
-  }
-};
-
-// One WAVE per page, executing the (serial) parse UNIFORMLY on all 64
-// lanes: same control flow and addresses everywhere, so tag-window loads
-// merge into one transaction, while the otherwise-idle lanes double as a
-// software prefetcher — each lane touches one cache line of the tag
-// stream ~4 KB ahead, so the serial chain's refills hit L2 instead of
-// paying a cold HBM round trip each 16 bytes.  Descriptor stores are
-// lane-0-gated.
+// Speculative wave-parallel tag parse: every lane decodes a CANDIDATE
+// element at byte si+lane (tag + extra bytes -> consumed/out_len/kind/aux,
+// mostly garbage), then the wave chases the TRUE element chain through
+// registers: each chase step reads the lane at the current offset via two
+// u64 shuffles (no memory traffic), emits its descriptor (predicated store
+// by the owning lane), and jumps to its successor.  ~24 real elements per
+// 64-B window on the worst FLBA-decimal pages cost ~24 shuffle steps
+// instead of 24 dependent memory round trips — the tag stream is the only
+// serial part of snappy and this walks it at register speed.
 __global__ void k_snap_parse(const SnappyPage* __restrict__ pages,
                              int64_t npages, SnapDesc* __restrict__ descs,
                              const int64_t* __restrict__ desc_base,
@@ -2833,15 +2793,13 @@ __global__ void k_snap_parse(const SnappyPage* __restrict__ pages,
       ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) / BG_WAVE;
   const int64_t nwaves = ((int64_t)gridDim.x * blockDim.x) / BG_WAVE;
   for (int64_t p = wave_global; p < npages; p += nwaves) {
+    const uint8_t* __restrict__ s = pages[p].src;
     const int64_t src_len = pages[p].src_len;
-    const uint8_t* __restrict__ s_pref = pages[p].src;
-    int64_t pref = 0;
-    SnapCursor cur{pages[p].src, src_len, 0, 0, -32};
     SnapDesc* d = descs + desc_base[p];
     int64_t si = 0, di = 0, ulen = 0, nd = 0;
     int ok = 1, shift = 0;
-    while (si < src_len) {
-      const uint8_t b = cur.at(si++);
+    while (si < src_len) {  // uncompressed-length varint (uniform)
+      const uint8_t b = s[si++];
       ulen |= (int64_t)(b & 0x7f) << shift;
       if (!(b & 0x80)) break;
       shift += 7;
@@ -2849,67 +2807,88 @@ __global__ void k_snap_parse(const SnappyPage* __restrict__ pages,
     }
     if (ulen > pages[p].dst_cap) ok = 0;
     while (ok && si < src_len && di < ulen) {
-      const uint8_t tag = cur.at(si++);
-      if ((tag & 3) == 0) {
-        int64_t len = (tag >> 2) + 1;
-        if (len > 60) {
-          const int nb = (int)len - 60;
-          if (si + nb > src_len) { ok = 0; break; }
-          len = 0;
-          for (int i = 0; i < nb; ++i)
-            len |= (int64_t)cur.at(si + i) << (8 * i);
-          len += 1;
-          si += nb;
-        }
-        if (si + len > src_len || di + len > ulen) { ok = 0; break; }
-        if (lane == 0) {
-          d[nd].dst = (uint32_t)di;
-          d[nd].aux = (uint32_t)si;
-          d[nd].len = (uint32_t)len;
-          d[nd].kind = 0;
-        }
-        ++nd;
-        si += len;
-        di += len;
-      } else {
-        int64_t len = 0, off = 0;
-        if ((tag & 3) == 1) {
-          if (si >= src_len) { ok = 0; break; }
-          len = ((tag >> 2) & 7) + 4;
-          off = ((int64_t)(tag >> 5) << 8) | cur.at(si);
-          si += 1;
-        } else if ((tag & 3) == 2) {
-          if (si + 2 > src_len) { ok = 0; break; }
-          len = (tag >> 2) + 1;
-          off = (int64_t)cur.at(si) | ((int64_t)cur.at(si + 1) << 8);
-          si += 2;
+      // ---- speculative decode at si + lane ----
+      const int64_t o = si + lane;
+      uint64_t q0 = 0;  // [next_rel(32) | out_len(32)]; next_rel 0 = invalid
+      uint64_t q1 = 0;  // [aux(32) | kind(1)]
+      if (o < src_len) {
+        const uint8_t tag = s[o];
+        const int t = tag & 3;
+        int64_t next_rel = 0, out = 0;
+        uint32_t aux = 0, kind = 0;
+        if (t == 0) {
+          int64_t len = (tag >> 2) + 1;
+          int nb = 0;
+          if (len > 60) {
+            nb = (int)len - 60;
+            if (o + 1 + nb <= src_len && nb <= 4) {
+              len = 0;
+              for (int i = 0; i < nb; ++i)
+                len |= (int64_t)s[o + 1 + i] << (8 * i);
+              len += 1;
+            } else {
+              len = -1;  // malformed
+            }
+          }
+          if (len >= 0) {
+            out = len;
+            aux = (uint32_t)(o + 1 + nb);  // ABS literal body offset
+            next_rel = (int64_t)lane + 1 + nb + len;
+            kind = 0;
+          }
+        } else if (t == 1) {
+          if (o + 1 < src_len) {
+            out = ((tag >> 2) & 7) + 4;
+            aux = ((uint32_t)(tag >> 5) << 8) | s[o + 1];
+            next_rel = lane + 2;
+            kind = 1;
+          }
+        } else if (t == 2) {
+          if (o + 2 < src_len) {
+            out = (tag >> 2) + 1;
+            aux = (uint32_t)s[o + 1] | ((uint32_t)s[o + 2] << 8);
+            next_rel = lane + 3;
+            kind = 1;
+          }
         } else {
-          if (si + 4 > src_len) { ok = 0; break; }
-          len = (tag >> 2) + 1;
-          off = (int64_t)cur.at(si) | ((int64_t)cur.at(si + 1) << 8) |
-                ((int64_t)cur.at(si + 2) << 16) |
-                ((int64_t)cur.at(si + 3) << 24);
-          si += 4;
+          if (o + 4 < src_len) {
+            out = (tag >> 2) + 1;
+            aux = (uint32_t)s[o + 1] | ((uint32_t)s[o + 2] << 8) |
+                  ((uint32_t)s[o + 3] << 16) | ((uint32_t)s[o + 4] << 24);
+            next_rel = lane + 5;
+            kind = 1;
+          }
         }
-        if (off == 0 || off > di || di + len > ulen) { ok = 0; break; }
-        if (lane == 0) {
+        if (next_rel > 0 && out >= 0 && out < ((int64_t)1 << 31) &&
+            next_rel < ((int64_t)1 << 31)) {
+          q0 = ((uint64_t)(uint32_t)next_rel << 32) | (uint64_t)(uint32_t)out;
+          q1 = ((uint64_t)aux << 32) | kind;
+        }
+      }
+      // ---- chase the true chain through registers ----
+      int64_t cur = 0;
+      while (ok && cur < BG_WAVE && si + cur < src_len && di < ulen) {
+        const uint64_t cq0 = (uint64_t)__shfl((long long)q0, (int)cur);
+        const uint64_t cq1 = (uint64_t)__shfl((long long)q1, (int)cur);
+        const uint32_t next_rel = (uint32_t)(cq0 >> 32);
+        const uint32_t out = (uint32_t)cq0;
+        const uint32_t aux = (uint32_t)(cq1 >> 32);
+        const uint32_t kind = (uint32_t)cq1 & 1;
+        if (next_rel == 0 || si + next_rel > src_len) { ok = 0; break; }
+        if (di + out > ulen) { ok = 0; break; }
+        if (kind == 1 && (aux == 0 || (int64_t)aux > di)) { ok = 0; break; }
+        if (lane == cur) {
           d[nd].dst = (uint32_t)di;
-          d[nd].aux = (uint32_t)off;
-          d[nd].len = (uint32_t)len;
-          d[nd].kind = 1;
+          d[nd].aux = aux;
+          d[nd].len = out;
+          d[nd].kind = kind;
         }
         ++nd;
-        di += len;
+        di += out;
+        cur = (int64_t)next_rel;
       }
-      // distributed prefetch: keep the tag stream ~4 KB ahead in L2
-      if (si + 2048 > pref && pref < src_len) {
-        const int64_t a = pref + (int64_t)lane * 64;
-        if (a < src_len) {
-          volatile uint8_t t = s_pref[a];
-          (void)t;
-        }
-        pref += 4096;
-      }
+      if (!ok) break;
+      si += cur;
     }
     if (lane == 0) {
       counts[p] = ok ? nd : 0;
