@@ -2787,7 +2787,8 @@ __global__ void k_snap_parse(const SnappyPage* __restrict__ pages,
                              int64_t npages, SnapDesc* __restrict__ descs,
                              const int64_t* __restrict__ desc_base,
                              int64_t* __restrict__ counts,
-                             int64_t* __restrict__ out_lens) {
+                             int64_t* __restrict__ out_lens,
+                             int store_lane, int ablate_shfl) {
   const int lane = lane_id();
   const int64_t wave_global =
       ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) / BG_WAVE;
@@ -2806,70 +2807,92 @@ __global__ void k_snap_parse(const SnappyPage* __restrict__ pages,
       if (shift > 32) { ok = 0; break; }
     }
     if (ulen > pages[p].dst_cap) ok = 0;
-    while (ok && si < src_len && di < ulen) {
-      // ---- speculative decode at si + lane ----
-      const int64_t o = si + lane;
-      uint64_t q0 = 0;  // [next_rel(32) | out_len(32)]; next_rel 0 = invalid
-      uint64_t q1 = 0;  // [aux(32) | kind(1)]
-      if (o < src_len) {
-        const uint8_t tag = s[o];
-        const int t = tag & 3;
-        int64_t next_rel = 0, out = 0;
-        uint32_t aux = 0, kind = 0;
-        if (t == 0) {
-          int64_t len = (tag >> 2) + 1;
-          int nb = 0;
-          if (len > 60) {
-            nb = (int)len - 60;
-            if (o + 1 + nb <= src_len && nb <= 4) {
-              len = 0;
-              for (int i = 0; i < nb; ++i)
-                len |= (int64_t)s[o + 1 + i] << (8 * i);
-              len += 1;
-            } else {
-              len = -1;  // malformed
-            }
-          }
-          if (len >= 0) {
-            out = len;
-            aux = (uint32_t)(o + 1 + nb);  // ABS literal body offset
-            next_rel = (int64_t)lane + 1 + nb + len;
-            kind = 0;
-          }
-        } else if (t == 1) {
-          if (o + 1 < src_len) {
-            out = ((tag >> 2) & 7) + 4;
-            aux = ((uint32_t)(tag >> 5) << 8) | s[o + 1];
-            next_rel = lane + 2;
-            kind = 1;
-          }
-        } else if (t == 2) {
-          if (o + 2 < src_len) {
-            out = (tag >> 2) + 1;
-            aux = (uint32_t)s[o + 1] | ((uint32_t)s[o + 2] << 8);
-            next_rel = lane + 3;
-            kind = 1;
-          }
-        } else {
-          if (o + 4 < src_len) {
-            out = (tag >> 2) + 1;
-            aux = (uint32_t)s[o + 1] | ((uint32_t)s[o + 2] << 8) |
-                  ((uint32_t)s[o + 3] << 16) | ((uint32_t)s[o + 4] << 24);
-            next_rel = lane + 5;
-            kind = 1;
+    // speculative decode of the candidate element at absolute offset o
+    auto spec_decode = [&](int64_t o, uint64_t* q0, uint64_t* q1,
+                           int64_t rel_base) {
+      *q0 = 0;
+      *q1 = 0;
+      if (o >= src_len) return;
+      const uint8_t tag = s[o];
+      const int t = tag & 3;
+      int64_t next_rel = 0, out = 0;
+      uint32_t aux = 0, kind = 0;
+      if (t == 0) {
+        int64_t len = (tag >> 2) + 1;
+        int nb = 0;
+        if (len > 60) {
+          nb = (int)len - 60;
+          if (o + 1 + nb <= src_len && nb <= 4) {
+            len = 0;
+            for (int i = 0; i < nb; ++i)
+              len |= (int64_t)s[o + 1 + i] << (8 * i);
+            len += 1;
+          } else {
+            len = -1;  // malformed
           }
         }
-        if (next_rel > 0 && out >= 0 && out < ((int64_t)1 << 31) &&
-            next_rel < ((int64_t)1 << 31)) {
-          q0 = ((uint64_t)(uint32_t)next_rel << 32) | (uint64_t)(uint32_t)out;
-          q1 = ((uint64_t)aux << 32) | kind;
+        if (len >= 0) {
+          out = len;
+          aux = (uint32_t)(o + 1 + nb);  // ABS literal body offset
+          next_rel = rel_base + 1 + nb + len;
+          kind = 0;
+        }
+      } else if (t == 1) {
+        if (o + 1 < src_len) {
+          out = ((tag >> 2) & 7) + 4;
+          aux = ((uint32_t)(tag >> 5) << 8) | s[o + 1];
+          next_rel = rel_base + 2;
+          kind = 1;
+        }
+      } else if (t == 2) {
+        if (o + 2 < src_len) {
+          out = (tag >> 2) + 1;
+          aux = (uint32_t)s[o + 1] | ((uint32_t)s[o + 2] << 8);
+          next_rel = rel_base + 3;
+          kind = 1;
+        }
+      } else {
+        if (o + 4 < src_len) {
+          out = (tag >> 2) + 1;
+          aux = (uint32_t)s[o + 1] | ((uint32_t)s[o + 2] << 8) |
+                ((uint32_t)s[o + 3] << 16) | ((uint32_t)s[o + 4] << 24);
+          next_rel = rel_base + 5;
+          kind = 1;
+        }
+      }
+      if (next_rel > 0 && out >= 0 && out < ((int64_t)1 << 31) &&
+          next_rel < ((int64_t)1 << 31)) {
+        *q0 = ((uint64_t)(uint32_t)next_rel << 32) | (uint64_t)(uint32_t)out;
+        *q1 = ((uint64_t)aux << 32) | kind;
+      }
+    };
+    while (ok && si < src_len && di < ulen) {
+      // ---- 128-B window: two candidate elements per lane ----
+      uint64_t qa0, qa1, qb0, qb1;
+      spec_decode(si + lane, &qa0, &qa1, lane);
+      spec_decode(si + BG_WAVE + lane, &qb0, &qb1, BG_WAVE + lane);
+      // prefetch the LIKELY next window while the chase runs (the chase
+      // itself touches no memory beyond descriptor stores)
+      {
+        const int64_t pf = si + 2 * BG_WAVE + (int64_t)lane * BG_WAVE;
+        if (lane < 8 && pf < src_len) {
+          volatile uint8_t t_ = s[pf];
+          (void)t_;
         }
       }
       // ---- chase the true chain through registers ----
       int64_t cur = 0;
-      while (ok && cur < BG_WAVE && si + cur < src_len && di < ulen) {
-        const uint64_t cq0 = (uint64_t)__shfl((long long)q0, (int)cur);
-        const uint64_t cq1 = (uint64_t)__shfl((long long)q1, (int)cur);
+      while (ok && cur < 2 * BG_WAVE && si + cur < src_len && di < ulen) {
+        const bool hi_half = cur >= BG_WAVE;
+        const int sl = (int)(cur & (BG_WAVE - 1));
+        uint64_t cq0, cq1;
+        if (ablate_shfl) {  // debug-only: own-lane regs (wrong results)
+          cq0 = hi_half ? qb0 : qa0;
+          cq1 = hi_half ? qb1 : qa1;
+        } else {
+          cq0 = (uint64_t)__shfl((long long)(hi_half ? qb0 : qa0), sl);
+          cq1 = (uint64_t)__shfl((long long)(hi_half ? qb1 : qa1), sl);
+        }
         const uint32_t next_rel = (uint32_t)(cq0 >> 32);
         const uint32_t out = (uint32_t)cq0;
         const uint32_t aux = (uint32_t)(cq1 >> 32);
@@ -2877,7 +2900,7 @@ __global__ void k_snap_parse(const SnappyPage* __restrict__ pages,
         if (next_rel == 0 || si + next_rel > src_len) { ok = 0; break; }
         if (di + out > ulen) { ok = 0; break; }
         if (kind == 1 && (aux == 0 || (int64_t)aux > di)) { ok = 0; break; }
-        if (lane == cur) {
+        if (lane == store_lane) {  // values are uniform after the shuffle
           d[nd].dst = (uint32_t)di;
           d[nd].aux = aux;
           d[nd].len = out;
@@ -3028,8 +3051,12 @@ extern "C" int bg_snappy_decompress(const void* h_pages, int64_t npages,
   const int wpb1 = BG_BLOCK / BG_WAVE;  // parse waves per block
   int blocks1 = (int)bg_imin64((npages + wpb1 - 1) / wpb1, BG_MAX_BLOCKS);
   if (blocks1 == 0) blocks1 = 1;
+  const char* abl = getenv("BG_SNAP_ABLATE");
+  const int store_lane = (abl && abl[0] == '1') ? BG_WAVE : 0;
+  const int ablate_shfl = (abl && abl[0] == '2') ? 1 : 0;
   hipLaunchKernelGGL(k_snap_parse, dim3(blocks1), dim3(BG_BLOCK), 0, 0,
-                     d_pages, npages, d_descs, d_base, d_counts, d_lens);
+                     d_pages, npages, d_descs, d_base, d_counts, d_lens,
+                     store_lane, ablate_shfl);
   HIP_TRY(hipGetLastError());
   double t1 = tick();
   t_parse = t1 - t0;

@@ -29,6 +29,9 @@ def make_pages(kind, page_bytes, npages, rng):
         elif kind == "lowcard":
             raw = rng.integers(0, 100, size=page_bytes // 8,
                                dtype=np.int64).tobytes()
+        elif kind == "flba7":  # parquet FLBA(7) decimals (price column)
+            vals = rng.integers(90000, 10495100, size=page_bytes // 7)
+            raw = b"".join(int(v).to_bytes(7, "big") for v in vals)
         else:  # runs
             raw = (rng.integers(0, 256, size=64, dtype=np.uint8).tobytes()
                    * (page_bytes // 64))
@@ -55,10 +58,11 @@ def bench(ctx, pages, iters=5):
     # warmup + verify
     gpu._check(L.bg_snappy_decompress(arr, ctypes.c_int64(len(pages)), lp),
                "snappy")
-    assert all(lens[i] == len(pages[i][1]) for i in range(len(pages)))
-    got = dst.download(np.uint8, min(total_u, 1 << 20))
-    want = b"".join(p[1] for p in pages)[:1 << 20]
-    assert got.tobytes() == want, "decode mismatch"
+    if not os.environ.get("BG_SNAP_ABLATE"):
+        assert all(lens[i] == len(pages[i][1]) for i in range(len(pages)))
+        got = dst.download(np.uint8, min(total_u, 1 << 20))
+        want = b"".join(p[1] for p in pages)[:1 << 20]
+        assert got.tobytes() == want, "decode mismatch"
     ctx.synchronize()
     best = 1e30
     for _ in range(iters):
