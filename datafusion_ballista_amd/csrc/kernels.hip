@@ -2787,8 +2787,12 @@ __global__ void k_snap_parse(const SnappyPage* __restrict__ pages,
                              int64_t npages, SnapDesc* __restrict__ descs,
                              const int64_t* __restrict__ desc_base,
                              int64_t* __restrict__ counts,
-                             int64_t* __restrict__ out_lens,
-                             int store_lane, int ablate_shfl) {
+                             int64_t* __restrict__ out_lens) {
+  // chase staging: candidates deposited once per window, then the serial
+  // chain walks LDS (~50-cy broadcast reads) instead of paired dependent
+  // ds_bpermute shuffles (~600 cy/step measured)
+  __shared__ uint64_t ldsq[BG_BLOCK / BG_WAVE][2 * BG_WAVE][2];
+  uint64_t(*myq)[2] = ldsq[threadIdx.x / BG_WAVE];
   const int lane = lane_id();
   const int64_t wave_global =
       ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) / BG_WAVE;
@@ -2871,6 +2875,11 @@ __global__ void k_snap_parse(const SnappyPage* __restrict__ pages,
       uint64_t qa0, qa1, qb0, qb1;
       spec_decode(si + lane, &qa0, &qa1, lane);
       spec_decode(si + BG_WAVE + lane, &qb0, &qb1, BG_WAVE + lane);
+      myq[lane][0] = qa0;
+      myq[lane][1] = qa1;
+      myq[BG_WAVE + lane][0] = qb0;
+      myq[BG_WAVE + lane][1] = qb1;
+      __builtin_amdgcn_wave_barrier();
       // prefetch the LIKELY next window while the chase runs (the chase
       // itself touches no memory beyond descriptor stores)
       {
@@ -2881,26 +2890,25 @@ __global__ void k_snap_parse(const SnappyPage* __restrict__ pages,
         }
       }
       // ---- chase the true chain through registers ----
+      // branchless chase: errors ACCUMULATE into err and force loop exit
+      // via a poisoned cur; on any error the whole page is discarded, so
+      // garbage descriptors stored after the fault never surface.
       int64_t cur = 0;
-      while (ok && cur < 2 * BG_WAVE && si + cur < src_len && di < ulen) {
-        const bool hi_half = cur >= BG_WAVE;
-        const int sl = (int)(cur & (BG_WAVE - 1));
-        uint64_t cq0, cq1;
-        if (ablate_shfl) {  // debug-only: own-lane regs (wrong results)
-          cq0 = hi_half ? qb0 : qa0;
-          cq1 = hi_half ? qb1 : qa1;
-        } else {
-          cq0 = (uint64_t)__shfl((long long)(hi_half ? qb0 : qa0), sl);
-          cq1 = (uint64_t)__shfl((long long)(hi_half ? qb1 : qa1), sl);
-        }
+      uint32_t err = 0;
+      while (cur < 2 * BG_WAVE && si + cur < src_len && di < ulen) {
+        const uint64_t cq0 = myq[cur][0];
+        const uint64_t cq1 = myq[cur][1];
         const uint32_t next_rel = (uint32_t)(cq0 >> 32);
         const uint32_t out = (uint32_t)cq0;
         const uint32_t aux = (uint32_t)(cq1 >> 32);
         const uint32_t kind = (uint32_t)cq1 & 1;
-        if (next_rel == 0 || si + next_rel > src_len) { ok = 0; break; }
-        if (di + out > ulen) { ok = 0; break; }
-        if (kind == 1 && (aux == 0 || (int64_t)aux > di)) { ok = 0; break; }
-        if (lane == store_lane) {  // values are uniform after the shuffle
+        const uint32_t bad =
+            (uint32_t)(next_rel == 0) |
+            (uint32_t)(si + next_rel > src_len) |
+            (uint32_t)(di + out > ulen) |
+            (kind & ((uint32_t)(aux == 0) | (uint32_t)((int64_t)aux > di)));
+        err |= bad;
+        if (lane == 0) {
           d[nd].dst = (uint32_t)di;
           d[nd].aux = aux;
           d[nd].len = out;
@@ -2908,9 +2916,9 @@ __global__ void k_snap_parse(const SnappyPage* __restrict__ pages,
         }
         ++nd;
         di += out;
-        cur = (int64_t)next_rel;
+        cur = bad ? (int64_t)4 * BG_WAVE : (int64_t)next_rel;
       }
-      if (!ok) break;
+      if (err) { ok = 0; break; }
       si += cur;
     }
     if (lane == 0) {
@@ -3051,12 +3059,8 @@ extern "C" int bg_snappy_decompress(const void* h_pages, int64_t npages,
   const int wpb1 = BG_BLOCK / BG_WAVE;  // parse waves per block
   int blocks1 = (int)bg_imin64((npages + wpb1 - 1) / wpb1, BG_MAX_BLOCKS);
   if (blocks1 == 0) blocks1 = 1;
-  const char* abl = getenv("BG_SNAP_ABLATE");
-  const int store_lane = (abl && abl[0] == '1') ? BG_WAVE : 0;
-  const int ablate_shfl = (abl && abl[0] == '2') ? 1 : 0;
   hipLaunchKernelGGL(k_snap_parse, dim3(blocks1), dim3(BG_BLOCK), 0, 0,
-                     d_pages, npages, d_descs, d_base, d_counts, d_lens,
-                     store_lane, ablate_shfl);
+                     d_pages, npages, d_descs, d_base, d_counts, d_lens);
   HIP_TRY(hipGetLastError());
   double t1 = tick();
   t_parse = t1 - t0;
