@@ -45,6 +45,11 @@ def parse_args():
     p.add_argument("--warmup", type=int, default=3)
     p.add_argument("--rows", type=int, default=ROWS_SF100,
                    help="rows per GPU (default SF100 lineitem)")
+    p.add_argument("--mode", choices=["q6", "q5"], default="q6",
+                   help="q6: flagship resident scan+filter+agg (driver "
+                        "contract); q5: q5-class hash-repartition resident "
+                        "shuffle (hash->split->all-to-all->final agg, "
+                        "SURVEY.md §8e)")
     p.add_argument("--cpu-sample-rows", type=int, default=100_000_000)
     p.add_argument("--skip-cpu-baseline", action="store_true")
     return p.parse_args()
@@ -120,6 +125,212 @@ def cpu_baseline_leg(sample_rows: int):
     }
 
 
+def q5_mode(args, dist, rank, world, local_rank):
+    """q5-class resident shuffle: each rank hash-
+    repartitions its shard into K = 16*world global partitions on device
+    (compute_partition_indices + split + partition-major materialise,
+    writer.rs:1259-1279/564-753), the partition-major buffers cross ranks
+    as an RCCL all-to-all over xGMI (exchange.py; replaces the remote-read
+    byte movement of shuffle_reader.rs:704-716), and each rank runs the
+    FINAL grouped aggregate over the raw rows it received.  Byte
+    conservation and an exact global-sum invariant assert every step;
+    partition ownership of received keys is device-verified on step 1."""
+    import ctypes
+
+    import numpy as np
+    import torch
+
+    from datafusion_ballista_amd import exchange, gpu
+
+    device = torch.device(f"cuda:{local_rank}")
+    ctx = gpu.GpuStageContext(local_rank)
+    n = args.rows if args.rows != ROWS_SF100 else 150_000_000
+    k_local = 16
+    K = k_local * world
+    g = torch.Generator(device=device)
+    g.manual_seed(500 + rank)
+    nkeys = 15_000_000 * world
+    keys = torch.randint(0, nkeys, (n,), generator=g, device=device,
+                         dtype=torch.int64)
+    price = torch.zeros((n, 2), dtype=torch.int64, device=device)
+    price[:, 0] = torch.randint(90000, 10495100, (n,), generator=g,
+                                device=device)
+    qty = torch.randint(1, 51, (n,), generator=g, device=device,
+                        dtype=torch.int64)
+    torch.cuda.synchronize()
+
+    # exact global invariant, computed once
+    local_sum = int(price[:, 0].sum().item())
+    local_rows = n
+    if dist is not None:
+        t = torch.tensor(i128_to_limbs(local_sum) + [local_rows],
+                         dtype=torch.int64, device=device)
+        dist.all_reduce(t)
+        m = t.cpu().tolist()
+        global_sum, global_rows = limbs_to_i128(m[:4]), m[4]
+    else:
+        global_sum, global_rows = local_sum, local_rows
+
+    def col_of(t, dtype):
+        return gpu.BgColumn(dtype, 15, 2, 0, t.data_ptr(), None, None,
+                            t.shape[0])
+
+    idx = torch.empty(n, dtype=torch.int32, device=device)
+    offs = torch.empty(K + 1, dtype=torch.int64, device=device)
+    out_keys = torch.empty(n, dtype=torch.int64, device=device)
+    out_price = torch.empty((n, 2), dtype=torch.int64, device=device)
+    out_qty = torch.empty(n, dtype=torch.int64, device=device)
+
+    agg_cap = max(2 * nkeys // max(world, 1), 1 << 20)
+    agg_first = ctx.alloc(4 * agg_cap)
+    agg_acc = ctx.alloc(16 * 2 * agg_cap)
+    agg_counts = ctx.alloc(8 * agg_cap)
+
+    ev0 = torch.cuda.Event(enable_timing=True)
+    ev1 = torch.cuda.Event(enable_timing=True)
+    repart_ms = []
+    checked = {"ownership": False}
+
+    def step():
+        kc = col_of(keys, gpu.BG_DT_INT64)
+        payload = (gpu.BgColumn * 3)(col_of(keys, gpu.BG_DT_INT64),
+                                     col_of(price, gpu.BG_DT_DECIMAL128),
+                                     col_of(qty, gpu.BG_DT_INT64))
+        outp = (ctypes.c_void_p * 3)(out_keys.data_ptr(),
+                                     out_price.data_ptr(),
+                                     out_qty.data_ptr())
+        ev0.record()
+        gpu._check(ctx.L.bg_hash_repartition(
+            ctypes.byref(kc), 1, payload, 3, ctypes.c_int64(n),
+            ctypes.c_uint32(K), ctypes.c_void_p(idx.data_ptr()),
+            ctypes.c_void_p(offs.data_ptr()), outp), "bg_hash_repartition")
+        ev1.record()
+        offsets = offs.cpu().numpy()
+        # exchange: keys (1 i64/row), price (2 i64/row), qty (1 i64/row)
+        if dist is not None:
+            rk, splits = exchange.all_to_all_rows(out_keys, offsets, world)
+            rp, _ = exchange.all_to_all_rows(out_price.view(-1), offsets * 2,
+                                             world)
+            rq, _ = exchange.all_to_all_rows(out_qty, offsets, world)
+        else:
+            rk, rp, rq = out_keys, out_price.view(-1), out_qty
+            splits = [n]
+        m = rk.shape[0]
+        # final grouped aggregate over the received raw rows (device
+        # outputs; parity of the aggregate itself is pinned by the test
+        # suite — the bench asserts the EXCHANGE invariants below)
+        rkc = gpu.BgColumn(gpu.BG_DT_INT64, 0, 0, 0, rk.data_ptr(), None,
+                           None, m)
+        rpc = gpu.BgColumn(gpu.BG_DT_DECIMAL128, 15, 2, 0, rp.data_ptr(),
+                           None, None, m)
+        rqc = gpu.BgColumn(gpu.BG_DT_INT64, 0, 0, 0, rq.data_ptr(), None,
+                           None, m)
+        karr = (gpu.BgColumn * 1)(rkc)
+        aarr = (gpu.BgColumn * 2)(rpc, rqc)
+        oarr = (ctypes.c_int32 * 2)(gpu.BG_AGG_OP_SUM_DEC128,
+                                    gpu.BG_AGG_OP_SUM_I64)
+        ng = ctypes.c_int64()
+        gpu._check(ctx.L.bg_hashagg(
+            karr, 1, aarr, oarr, 2, None, ctypes.c_int64(m),
+            ctypes.c_int64(agg_cap), agg_first.ptr, agg_acc.ptr,
+            agg_counts.ptr, ctypes.byref(ng)), "bg_hashagg")
+        torch.cuda.synchronize()
+        repart_ms.append(ev0.elapsed_time(ev1))
+        # conservation + exact global-sum invariants (every step): the
+        # received price ints sum exactly to the generated global sum
+        part_sum = int(rp.view(-1, 2)[:, 0].sum().item())
+        rows_recv = m
+        if dist is not None:
+            t = torch.tensor(i128_to_limbs(part_sum) + [rows_recv],
+                             dtype=torch.int64, device=device)
+            dist.all_reduce(t)
+            mm = t.cpu().tolist()
+            tot_sum, tot_rows = limbs_to_i128(mm[:4]), mm[4]
+        else:
+            tot_sum, tot_rows = part_sum, rows_recv
+        assert tot_rows == global_rows, (tot_rows, global_rows)
+        assert tot_sum == global_sum, (tot_sum, global_sum)
+        assert int(ng.value) > 0
+        if not checked["ownership"] and dist is not None:
+            # received keys must hash to MY partitions (device check)
+            hb = ctx.alloc(max(8 * m, 8))
+            pb = ctx.alloc(max(4 * m, 4))
+            gpu._check(ctx.L.bg_hash_columns(ctypes.byref(rkc), 1,
+                                             ctypes.c_int64(m), hb.ptr),
+                       "hash")
+            gpu._check(ctx.L.bg_partition_ids(hb.ptr, ctypes.c_int64(m),
+                                              ctypes.c_uint32(K), pb.ptr),
+                       "pids")
+            pids = pb.download(np.uint32, m)
+            owners = pids // k_local
+            assert (owners == rank).all(), "foreign rows received"
+            checked["ownership"] = True
+        return m
+
+    def barrier_sync():
+        if dist is not None:
+            dist.barrier()
+        import torch
+        torch.cuda.synchronize()
+
+    for _ in range(args.warmup):
+        step()
+    barrier_sync()
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        step()
+    barrier_sync()
+    elapsed = time.perf_counter() - t0
+    if dist is not None:
+        t = torch.tensor([elapsed], dtype=torch.float64, device=device)
+        dist.all_reduce(t, op=torch.distributed.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    if rank == 0:
+        total_rows = n * world * args.steps
+        value = total_rows / elapsed
+        kms = sorted(repart_ms)[len(repart_ms) // 2]
+        # repartition algorithmic bytes/row (DESIGN.md §3): read key+payload
+        # once + write once (2 x 32 B) + hash/pids/idx/rank bookkeeping 16 B
+        bytes_per_row = 80.0
+        achieved_gbs = n * bytes_per_row / (kms * 1e-3) / 1e9
+        out = {
+            "metric": "tpch_q5class_resident_shuffle_rows_per_s",
+            "value": value,
+            "unit": "rows/s",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": elapsed / args.steps * 1000.0,
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "int128",
+            "data": "synthetic",
+            "config": {
+                "workload": "tpch_q5class_resident_shuffle",
+                "rows_per_gpu": n,
+                "k_partitions": K,
+                "exchanged_bytes_per_row": 32,
+                "parallelism": f"shuffle{world}",
+            },
+            "roofline": {
+                "bound": "hbm",
+                "achieved": achieved_gbs,
+                "peak": HBM_PEAK_GBS,
+                "unit": "GB/s",
+                "frac": achieved_gbs / HBM_PEAK_GBS,
+                "traffic": None,
+                "kernel": "bg_hash_repartition pipeline",
+                "kernel_ms_median": kms,
+            },
+            "cpu_baseline": None,
+        }
+        print(json.dumps(out), flush=True)
+    if dist is not None:
+        dist.destroy_process_group()
+
+
 def main():
     args = parse_args()
     import torch
@@ -127,6 +338,9 @@ def main():
         print(json.dumps({"error": "no GPU visible; bench requires MI355X"}))
         sys.exit(1)
     dist, rank, world, local_rank = setup_dist(args)
+    if args.mode == "q5":
+        q5_mode(args, dist, rank, world, local_rank)
+        return
     device = torch.device(f"cuda:{local_rank}")
 
     from datafusion_ballista_amd import gpu, tpch_synth

@@ -5374,3 +5374,332 @@ extern "C" int bg_fill_rand(void* d_out, int64_t n, uint64_t seed,
   HIP_TRY(hipGetLastError());
   return BG_OK;
 }
+
+// ---------------------------------------------------------------------------
+// Generalized HashJoinExec build/probe (round 2): multi-column keys over
+// Int64/Int32/Date32/Decimal128/Utf8/dict8 and the non-inner join types
+// DataFusion's HashJoinExec supports on the probe side (LeftSemi/LeftAnti/
+// Left with the probe side as the preserved side — SURVEY.md §8a row 3;
+// q2/q9-class shapes).  Nodes store {hash, next} (the general analogue of
+// the Int64 path's {key, next}: full-hash reject before the cross-table
+// key compare).  Join null semantics (null_equals_null=false): a row with
+// ANY null key matches nothing — excluded from INNER/SEMI, EMITTED by
+// ANTI and OUTER (with a BG_JOIN_NULL_IDX build id).
+// ---------------------------------------------------------------------------
+
+__device__ __forceinline__ bool row_has_null_key(const KeyArgs& keys,
+                                                 int64_t i) {
+  for (int c = 0; c < keys.nkeys; ++c)
+    if (keys.k[c].valid && !bit_valid(keys.k[c].valid, i)) return true;
+  return false;
+}
+
+// cross-table key equality (build row vs probe row); nulls already
+// excluded by row_has_null_key
+__device__ __forceinline__ bool keys_equal_cross(const KeyArgs& a, int64_t ra,
+                                                 const KeyArgs& b,
+                                                 int64_t rb) {
+  for (int c = 0; c < a.nkeys; ++c) {
+    switch (a.k[c].dtype) {
+      case BG_DT_INT64:
+        if (reinterpret_cast<const int64_t*>(a.k[c].data)[ra] !=
+            reinterpret_cast<const int64_t*>(b.k[c].data)[rb])
+          return false;
+        break;
+      case BG_DT_INT32:
+      case BG_DT_DATE32:
+        if (reinterpret_cast<const int32_t*>(a.k[c].data)[ra] !=
+            reinterpret_cast<const int32_t*>(b.k[c].data)[rb])
+          return false;
+        break;
+      case BG_DT_DECIMAL128: {
+        const ulong2 x = reinterpret_cast<const ulong2*>(a.k[c].data)[ra];
+        const ulong2 y = reinterpret_cast<const ulong2*>(b.k[c].data)[rb];
+        if (x.x != y.x || x.y != y.y) return false;
+        break;
+      }
+      case BG_DT_DICT8:
+        if (reinterpret_cast<const uint8_t*>(a.k[c].data)[ra] !=
+            reinterpret_cast<const uint8_t*>(b.k[c].data)[rb])
+          return false;
+        break;
+      case BG_DT_UTF8: {
+        const int32_t la = a.k[c].offsets[ra], ha = a.k[c].offsets[ra + 1];
+        const int32_t lb = b.k[c].offsets[rb], hb = b.k[c].offsets[rb + 1];
+        if (ha - la != hb - lb) return false;
+        const uint8_t* da = reinterpret_cast<const uint8_t*>(a.k[c].data);
+        const uint8_t* db = reinterpret_cast<const uint8_t*>(b.k[c].data);
+        for (int32_t j = 0; j < ha - la; ++j)
+          if (da[la + j] != db[lb + j]) return false;
+        break;
+      }
+      default:
+        return false;
+    }
+  }
+  return true;
+}
+
+__global__ void k_join_build2(KeyArgs keys, int64_t n, int* head,
+                              ulong2* nodes, u64 mask) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    if (row_has_null_key(keys, i)) continue;
+    const u64 h = hash_keys_row(keys, i);
+    const int prev = atomicExch(&head[h & mask], (int)i);
+    ulong2 node;
+    node.x = h;
+    node.y = (u64)(int64_t)prev;
+    nodes[i] = node;
+  }
+}
+
+__global__ void k_join_count2(KeyArgs bkeys, KeyArgs pkeys, int64_t n_probe,
+                              const int* head, const ulong2* nodes, u64 mask,
+                              int32_t join_type, uint32_t* counts) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       i < n_probe; i += (int64_t)gridDim.x * blockDim.x) {
+    u64 cnt = 0;
+    if (!row_has_null_key(pkeys, i)) {
+      const u64 h = hash_keys_row(pkeys, i);
+      int64_t cur = head[h & mask];
+      while (cur >= 0) {
+        const ulong2 node = nodes[cur];
+        if (node.x == h && keys_equal_cross(bkeys, cur, pkeys, i)) {
+          ++cnt;
+          if (join_type == BG_JOIN_SEMI || join_type == BG_JOIN_ANTI) break;
+        }
+        cur = (int64_t)node.y;
+      }
+    }
+    switch (join_type) {
+      case BG_JOIN_INNER: break;
+      case BG_JOIN_SEMI: cnt = cnt ? 1 : 0; break;
+      case BG_JOIN_ANTI: cnt = cnt ? 0 : 1; break;
+      case BG_JOIN_OUTER_PROBE: cnt = cnt ? cnt : 1; break;
+      default: break;
+    }
+    counts[i] = (uint32_t)cnt;
+  }
+}
+
+__global__ void k_join_fill2(KeyArgs bkeys, KeyArgs pkeys, int64_t n_probe,
+                             const int* head, const ulong2* nodes, u64 mask,
+                             int32_t join_type, const i64* offsets,
+                             uint32_t* out_probe, uint32_t* out_build) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       i < n_probe; i += (int64_t)gridDim.x * blockDim.x) {
+    i64 w = offsets[i];
+    u64 matched = 0;
+    if (!row_has_null_key(pkeys, i)) {
+      const u64 h = hash_keys_row(pkeys, i);
+      int64_t cur = head[h & mask];
+      while (cur >= 0) {
+        const ulong2 node = nodes[cur];
+        if (node.x == h && keys_equal_cross(bkeys, cur, pkeys, i)) {
+          ++matched;
+          if (join_type == BG_JOIN_SEMI) {
+            out_probe[w] = (uint32_t)i;
+            out_build[w] = (uint32_t)cur;
+            ++w;
+            break;
+          }
+          if (join_type == BG_JOIN_ANTI) break;
+          if (join_type != BG_JOIN_ANTI) {
+            out_probe[w] = (uint32_t)i;
+            out_build[w] = (uint32_t)cur;
+            ++w;
+          }
+        }
+        cur = (int64_t)node.y;
+      }
+    }
+    if (!matched &&
+        (join_type == BG_JOIN_ANTI || join_type == BG_JOIN_OUTER_PROBE)) {
+      out_probe[w] = (uint32_t)i;
+      out_build[w] = BG_JOIN_NULL_IDX;
+    }
+  }
+}
+
+static int keyargs_from_cols(const bg_column* cols, int32_t nkeys,
+                             KeyArgs* out) {
+  if (nkeys <= 0 || nkeys > BG_MAX_KEYS) return BG_ERR_INVALID;
+  out->nkeys = nkeys;
+  for (int c = 0; c < nkeys; ++c) {
+    out->k[c].data = cols[c].d_data;
+    out->k[c].valid = cols[c].d_validity;
+    out->k[c].offsets = cols[c].d_offsets;
+    out->k[c].dtype = cols[c].dtype;
+    switch (cols[c].dtype) {
+      case BG_DT_INT64:
+      case BG_DT_INT32:
+      case BG_DT_DATE32:
+      case BG_DT_DECIMAL128:
+      case BG_DT_DICT8:
+      case BG_DT_UTF8:
+        break;
+      default:
+        return BG_ERR_UNSUPPORTED;
+    }
+  }
+  return BG_OK;
+}
+
+struct BgJoinTable2 {
+  ulong2* nodes;
+  int* head;
+  int64_t n_build;
+  u64 mask;
+  KeyArgs bkeys;  // caller's device buffers must outlive the handle
+  int32_t nkeys;
+  i64* probe_offsets = nullptr;
+  int64_t probe_n = 0;
+  int32_t probe_jt = -1;
+};
+
+extern "C" int bg_hashjoin_build2(const bg_column* keys, int32_t nkeys,
+                                  int64_t n, void** out_handle) {
+  REQUIRE_INIT();
+  if (n > 0x7fffffffLL)
+    return set_err(BG_ERR_INVALID, "build side > 2^31 rows");
+  BgJoinTable2 t{};
+  if (keyargs_from_cols(keys, nkeys, &t.bkeys) != BG_OK)
+    return set_err(BG_ERR_UNSUPPORTED, "join key dtype/count unsupported");
+  t.nkeys = nkeys;
+  t.n_build = n;
+  const u64 nb = next_pow2_u64((u64)(n > 4 ? n * 2 : 8));
+  t.mask = nb - 1;
+  HIP_TRY(pool_malloc((void**)&t.nodes, sizeof(ulong2) * (n ? n : 1)));
+  HIP_TRY(pool_malloc((void**)&t.head, sizeof(int) * nb));
+  HIP_TRY(hipMemset(t.head, 0xff, sizeof(int) * nb));
+  int blocks = (int)bg_imin64((n + BG_BLOCK - 1) / BG_BLOCK, BG_MAX_BLOCKS);
+  if (blocks == 0) blocks = 1;
+  hipLaunchKernelGGL(k_join_build2, dim3(blocks), dim3(BG_BLOCK), 0, 0,
+                     t.bkeys, n, t.head, t.nodes, t.mask);
+  HIP_TRY(hipGetLastError());
+  *out_handle = new BgJoinTable2(t);
+  return BG_OK;
+}
+
+extern "C" int bg_hashjoin_probe_count2(void* handle, const bg_column* keys,
+                                        int32_t nkeys, int64_t n,
+                                        int32_t join_type,
+                                        int64_t* out_matches) {
+  REQUIRE_INIT();
+  BgJoinTable2* t = (BgJoinTable2*)handle;
+  KeyArgs pk;
+  if (keyargs_from_cols(keys, nkeys, &pk) != BG_OK || nkeys != t->nkeys)
+    return set_err(BG_ERR_UNSUPPORTED, "probe key dtype/count mismatch");
+  for (int c = 0; c < nkeys; ++c)
+    if (pk.k[c].dtype != t->bkeys.k[c].dtype)
+      return set_err(BG_ERR_INVALID, "probe/build key dtypes differ");
+  if (t->probe_offsets) {
+    (void)pool_release(t->probe_offsets);
+    t->probe_offsets = nullptr;
+  }
+  uint32_t* d_counts;
+  i64* d_offs;
+  i64* d_total;
+  HIP_TRY(pool_malloc((void**)&d_counts, sizeof(uint32_t) * (n ? n : 1)));
+  HIP_TRY(pool_malloc((void**)&d_offs, sizeof(i64) * (n ? n : 1)));
+  HIP_TRY(pool_malloc((void**)&d_total, sizeof(i64)));
+  int blocks = (int)bg_imin64((n + BG_BLOCK - 1) / BG_BLOCK, BG_MAX_BLOCKS);
+  if (blocks == 0) blocks = 1;
+  hipLaunchKernelGGL(k_join_count2, dim3(blocks), dim3(BG_BLOCK), 0, 0,
+                     t->bkeys, pk, n, t->head, t->nodes, t->mask, join_type,
+                     d_counts);
+  HIP_TRY(hipGetLastError());
+  int rc = scan_exclusive_u32(d_counts, n, d_offs, d_total);
+  if (rc != BG_OK) return rc;
+  i64 total = 0;
+  HIP_TRY(hipMemcpy(&total, d_total, sizeof(i64), hipMemcpyDeviceToHost));
+  (void)pool_release(d_counts);
+  (void)pool_release(d_total);
+  t->probe_offsets = d_offs;
+  t->probe_n = n;
+  t->probe_jt = join_type;
+  *out_matches = total;
+  return BG_OK;
+}
+
+extern "C" int bg_hashjoin_probe_fill2(void* handle, const bg_column* keys,
+                                       int32_t nkeys, int64_t n,
+                                       int32_t join_type,
+                                       uint32_t* d_out_probe,
+                                       uint32_t* d_out_build) {
+  REQUIRE_INIT();
+  BgJoinTable2* t = (BgJoinTable2*)handle;
+  KeyArgs pk;
+  if (keyargs_from_cols(keys, nkeys, &pk) != BG_OK)
+    return set_err(BG_ERR_UNSUPPORTED, "probe key dtype/count mismatch");
+  if (!t->probe_offsets || t->probe_n != n || t->probe_jt != join_type)
+    return set_err(BG_ERR_INVALID,
+                   "call bg_hashjoin_probe_count2 first (same join_type)");
+  int blocks = (int)bg_imin64((n + BG_BLOCK - 1) / BG_BLOCK, BG_MAX_BLOCKS);
+  if (blocks == 0) blocks = 1;
+  hipLaunchKernelGGL(k_join_fill2, dim3(blocks), dim3(BG_BLOCK), 0, 0,
+                     t->bkeys, pk, n, t->head, t->nodes, t->mask, join_type,
+                     t->probe_offsets, d_out_probe, d_out_build);
+  HIP_TRY(hipGetLastError());
+  return BG_OK;
+}
+
+extern "C" int bg_hashjoin_free2(void* handle) {
+  REQUIRE_INIT();
+  BgJoinTable2* t = (BgJoinTable2*)handle;
+  if (!t) return BG_OK;
+  (void)pool_release(t->nodes);
+  (void)pool_release(t->head);
+  if (t->probe_offsets) (void)pool_release(t->probe_offsets);
+  delete t;
+  return BG_OK;
+}
+
+// sentinel handling for probe-outer joins: split an index vector carrying
+// BG_JOIN_NULL_IDX markers into a clamped (safe-to-gather) vector + an
+// Arrow validity bitmap of the non-sentinel slots
+__global__ void k_idx_sentinel(const uint32_t* __restrict__ idx, int64_t m,
+                               uint32_t* __restrict__ clamped,
+                               uint32_t* __restrict__ valid_words) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < m;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    const uint32_t v = idx[i];
+    const bool ok = v != 0xFFFFFFFFu;
+    clamped[i] = ok ? v : 0;
+    if (ok) atomicOr(&valid_words[i >> 5], 1u << (i & 31));
+    else atomicAnd(&valid_words[i >> 5], ~(1u << (i & 31)));
+  }
+}
+
+extern "C" int bg_idx_sentinel(const uint32_t* d_idx, int64_t m,
+                               uint32_t* d_clamped, uint8_t* d_valid_bits) {
+  REQUIRE_INIT();
+  if (m <= 0) return BG_OK;
+  int blocks = (int)bg_imin64((m + BG_BLOCK - 1) / BG_BLOCK, BG_MAX_BLOCKS);
+  hipLaunchKernelGGL(k_idx_sentinel, dim3(blocks), dim3(BG_BLOCK), 0, 0,
+                     d_idx, m, d_clamped, (uint32_t*)d_valid_bits);
+  HIP_TRY(hipGetLastError());
+  return BG_OK;
+}
+
+__global__ void k_bitmap_and(const uint8_t* __restrict__ a,
+                             const uint8_t* __restrict__ b, int64_t nbytes,
+                             uint8_t* __restrict__ out) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < nbytes;
+       i += (int64_t)gridDim.x * blockDim.x)
+    out[i] = a[i] & b[i];
+}
+
+extern "C" int bg_bitmap_and(const uint8_t* d_a, const uint8_t* d_b,
+                             int64_t nbits, uint8_t* d_out) {
+  REQUIRE_INIT();
+  int64_t nbytes = (nbits + 7) / 8;
+  if (nbytes <= 0) return BG_OK;
+  int blocks =
+      (int)bg_imin64((nbytes + BG_BLOCK - 1) / BG_BLOCK, BG_MAX_BLOCKS);
+  hipLaunchKernelGGL(k_bitmap_and, dim3(blocks), dim3(BG_BLOCK), 0, 0, d_a,
+                     d_b, nbytes, d_out);
+  HIP_TRY(hipGetLastError());
+  return BG_OK;
+}

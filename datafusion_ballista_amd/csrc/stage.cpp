@@ -808,29 +808,70 @@ static Table exec_join(const Value& node, Metrics& m) {
   Table probe = exec_plan(node.at("probe"), m);
   auto& bkeys = node.get_arr("build_keys");
   auto& pkeys = node.get_arr("probe_keys");
-  if (bkeys.size() != 1 || pkeys.size() != 1)
-    throw StageError(BG_ERR_UNSUPPORTED,
-                     "hash_join: single Int64 key only (round 2 start)");
+  if (bkeys.size() != pkeys.size() || bkeys.empty() || bkeys.size() > 4)
+    throw StageError(BG_ERR_INVALID, "hash_join: 1-4 matching key columns");
   const std::string jt = node.get_str_or("join_type", "inner");
-  if (jt != "inner")
-    throw StageError(BG_ERR_UNSUPPORTED, "hash_join: inner only for now");
+  int32_t join_type;
+  if (jt == "inner") join_type = BG_JOIN_INNER;
+  else if (jt == "semi") join_type = BG_JOIN_SEMI;
+  else if (jt == "anti") join_type = BG_JOIN_ANTI;
+  else if (jt == "left") join_type = BG_JOIN_OUTER_PROBE;  // probe preserved
+  else
+    throw StageError(BG_ERR_UNSUPPORTED, "hash_join: join_type " + jt);
 
-  bg_column bk = to_bg(build.cols[(size_t)build.idx(bkeys[0]->s)], build.n);
-  bg_column pk = to_bg(probe.cols[(size_t)probe.idx(pkeys[0]->s)], probe.n);
+  std::vector<bg_column> bk, pk;
+  for (size_t i = 0; i < bkeys.size(); ++i) {
+    bk.push_back(to_bg(build.cols[(size_t)build.idx(bkeys[i]->s)], build.n));
+    pk.push_back(to_bg(probe.cols[(size_t)probe.idx(pkeys[i]->s)], probe.n));
+    if (bk[i].dtype != pk[i].dtype)
+      throw StageError(BG_ERR_INVALID, "hash_join: key dtype mismatch");
+  }
+  const bool fast_int64 =
+      bk.size() == 1 && bk[0].dtype == BG_DT_INT64 && join_type == BG_JOIN_INNER;
+
   void* handle = nullptr;
-  chk(bg_hashjoin_build(&bk, build.n, &handle), "bg_hashjoin_build");
   int64_t matches = 0;
-  int rc = bg_hashjoin_probe_count(handle, &pk, probe.n, &matches);
+  int rc;
+  if (fast_int64) {
+    chk(bg_hashjoin_build(&bk[0], build.n, &handle), "bg_hashjoin_build");
+    rc = bg_hashjoin_probe_count(handle, &pk[0], probe.n, &matches);
+  } else {
+    chk(bg_hashjoin_build2(bk.data(), (int32_t)bk.size(), build.n, &handle),
+        "bg_hashjoin_build2");
+    rc = bg_hashjoin_probe_count2(handle, pk.data(), (int32_t)pk.size(),
+                                  probe.n, join_type, &matches);
+  }
   if (rc != BG_OK) {
-    (void)bg_hashjoin_free(handle);
-    chk(rc, "bg_hashjoin_probe_count");
+    if (fast_int64) (void)bg_hashjoin_free(handle);
+    else (void)bg_hashjoin_free2(handle);
+    chk(rc, "hash_join probe count");
   }
   DBufPtr pidx = dalloc((uint64_t)(matches > 0 ? matches : 1) * 4);
   DBufPtr bidx = dalloc((uint64_t)(matches > 0 ? matches : 1) * 4);
-  rc = bg_hashjoin_probe_fill(handle, &pk, probe.n, (uint32_t*)pidx->p,
-                              (uint32_t*)bidx->p);
-  (void)bg_hashjoin_free(handle);
-  chk(rc, "bg_hashjoin_probe_fill");
+  if (fast_int64) {
+    rc = bg_hashjoin_probe_fill(handle, &pk[0], probe.n, (uint32_t*)pidx->p,
+                                (uint32_t*)bidx->p);
+    (void)bg_hashjoin_free(handle);
+  } else {
+    rc = bg_hashjoin_probe_fill2(handle, pk.data(), (int32_t)pk.size(),
+                                 probe.n, join_type, (uint32_t*)pidx->p,
+                                 (uint32_t*)bidx->p);
+    (void)bg_hashjoin_free2(handle);
+  }
+  chk(rc, "hash_join probe fill");
+
+  // probe-outer: the build index vector carries BG_JOIN_NULL_IDX for
+  // unmatched probe rows — clamp it and remember the null mask so build-
+  // side gathers produce NULL columns
+  DBufPtr bidx_clamped, outer_valid;
+  if (join_type == BG_JOIN_OUTER_PROBE) {
+    bidx_clamped = dalloc((uint64_t)(matches > 0 ? matches : 1) * 4);
+    outer_valid = dalloc((uint64_t)((matches + 63) / 64) * 8 + 8);
+    chk(bg_memset(outer_valid->p, 0xFF, outer_valid->bytes), "bg_memset");
+    chk(bg_idx_sentinel((const uint32_t*)bidx->p, matches,
+                        (uint32_t*)bidx_clamped->p, outer_valid->u8()),
+        "bg_idx_sentinel");
+  }
 
   Table out;
   out.n = matches;
@@ -838,11 +879,26 @@ static Table exec_join(const Value& node, Metrics& m) {
     const std::string side = o->get_str("side");
     const std::string col = o->get_str("col");
     const Table& src = (side == "build") ? build : probe;
+    const bool outer_build =
+        side == "build" && join_type == BG_JOIN_OUTER_PROBE;
     const uint32_t* idx =
-        (side == "build") ? (const uint32_t*)bidx->p : (const uint32_t*)pidx->p;
+        (side == "build")
+            ? (const uint32_t*)(outer_build ? bidx_clamped->p : bidx->p)
+            : (const uint32_t*)pidx->p;
     out.names.push_back(o->get_str_or("as", col));
-    out.cols.push_back(
-        gather_col(src.cols[(size_t)src.idx(col)], src.n, idx, matches));
+    Col c = gather_col(src.cols[(size_t)src.idx(col)], src.n, idx, matches);
+    if (outer_build) {
+      if (c.validity) {  // combine source validity with the outer mask
+        DBufPtr comb = dalloc((uint64_t)((matches + 63) / 64) * 8 + 8);
+        chk(bg_bitmap_and(c.validity->u8(), outer_valid->u8(), matches,
+                          comb->u8()),
+            "bg_bitmap_and");
+        c.validity = comb;
+      } else {
+        c.validity = outer_valid;
+      }
+    }
+    out.cols.push_back(std::move(c));
   }
   return out;
 }
@@ -2000,6 +2056,9 @@ static VSchema validate_plan(const Value& node) {
     VSchema p = validate_plan(node.at("probe"));
     for (auto& k : node.get_arr("build_keys")) b.idx(k->s);
     for (auto& k : node.get_arr("probe_keys")) p.idx(k->s);
+    const std::string jt = node.get_str_or("join_type", "inner");
+    if (jt != "inner" && jt != "semi" && jt != "anti" && jt != "left")
+      throw StageError(BG_ERR_UNSUPPORTED, "hash_join: join_type " + jt);
     VSchema out;
     for (auto& o : node.get_arr("output")) {
       const VSchema& src = o->get_str("side") == "build" ? b : p;

@@ -216,6 +216,38 @@ int bg_hashjoin_probe_fill(void* handle, const bg_column* probe_keys,
                            uint32_t* d_out_build);
 int bg_hashjoin_free(void* handle);
 
+/* ---- generalized hash join (round 2): multi-column keys over
+ * Int64/Int32/Date32/Decimal128/Utf8/dict8 and the probe-side join types
+ * of DataFusion's HashJoinExec (q2/q9-class shapes).  Nodes store
+ * {hash, next}; equality is the cross-table key compare.  Join null
+ * semantics (null_equals_null=false): a row with ANY null key matches
+ * nothing — excluded by INNER/SEMI, emitted by ANTI/OUTER_PROBE with
+ * build id BG_JOIN_NULL_IDX.  The build-key device buffers must outlive
+ * the handle (nodes reference them for compares). */
+#define BG_JOIN_INNER 0
+#define BG_JOIN_SEMI 1        /* probe rows with >=1 match, emitted once */
+#define BG_JOIN_ANTI 2        /* probe rows with no match */
+#define BG_JOIN_OUTER_PROBE 3 /* matches + unmatched probe rows */
+#define BG_JOIN_NULL_IDX 0xFFFFFFFFu
+int bg_hashjoin_build2(const bg_column* keys, int32_t nkeys, int64_t n,
+                       void** out_handle);
+int bg_hashjoin_probe_count2(void* handle, const bg_column* keys,
+                             int32_t nkeys, int64_t n, int32_t join_type,
+                             int64_t* out_matches);
+int bg_hashjoin_probe_fill2(void* handle, const bg_column* keys,
+                            int32_t nkeys, int64_t n, int32_t join_type,
+                            uint32_t* d_out_probe, uint32_t* d_out_build);
+int bg_hashjoin_free2(void* handle);
+
+/* Split an index vector carrying BG_JOIN_NULL_IDX sentinels (probe-outer
+ * build side) into a clamped gather-safe vector + the validity bitmap of
+ * non-sentinel slots. */
+int bg_idx_sentinel(const uint32_t* d_idx, int64_t m, uint32_t* d_clamped,
+                    uint8_t* d_valid_bits);
+/* out = a AND b over ceil(nbits/8) bytes (combine validity bitmaps). */
+int bg_bitmap_and(const uint8_t* d_a, const uint8_t* d_b, int64_t nbits,
+                  uint8_t* d_out);
+
 /* ---- general hash group-by (AggregateExec Partial/Single) ----
  * Arbitrary group cardinality (q3-class: millions of groups).  Aggregate
  * ops: 0 = SUM over Decimal128 (exact i128), 1 = SUM over Int64 (exact,

@@ -477,6 +477,112 @@ def test_q3_stage_real_group_key(ctx, reg):
         [g[0] for g in got] == [w[0] for w in top10]
 
 
+def _join_plan(build_scan, probe_scan, bk, pk, jt, output):
+    return {"op": "hash_join", "build": build_scan, "probe": probe_scan,
+            "build_keys": bk, "probe_keys": pk, "join_type": jt,
+            "output": output}
+
+
+def test_join_types_semi_anti_left(ctx, reg):
+    """Probe-side join types (q2/q9/q16-class shapes): semi, anti, and
+    left (probe side preserved, build side NULL for unmatched), with NULL
+    probe keys — null_equals_null=false: nulls never match, so they are
+    excluded by semi/inner and EMITTED by anti/left."""
+    rng = np.random.default_rng(41)
+    nb, np_ = 500, 20_000
+    bkeys = np.arange(0, 2 * nb, 2, dtype=np.int64)  # evens only
+    bval = rng.integers(0, 100, size=nb, dtype=np.int64)
+    pkeys = rng.integers(0, 2 * nb, size=np_, dtype=np.int64)
+    pmask = rng.random(np_) < 0.1  # 10% NULL probe keys
+    build = pa.table({"bk": pa.array(bkeys), "bv": pa.array(bval)})
+    probe = pa.table({"pk": pa.array(pkeys, mask=pmask),
+                      "pid": pa.array(np.arange(np_, dtype=np.int64))})
+    reg("jb", build)
+    reg("jp", probe)
+    bs, ps = scan_of(build, "jb"), scan_of(probe, "jp")
+
+    matched = {int(k) for k in bkeys}
+    pk_list = [None if m else int(k) for k, m in zip(pkeys, pmask)]
+
+    # SEMI: probe rows with a match, once
+    res = stage.execute(_doc({"op": "collect", "input": _join_plan(
+        bs, ps, ["bk"], ["pk"], "semi",
+        [{"side": "probe", "col": "pid"}])}))
+    want = [i for i, k in enumerate(pk_list)
+            if k is not None and k in matched]
+    assert sorted(r[0] for r in res["rows"]) == want
+
+    # ANTI: probe rows with no match, including NULL-key rows
+    res = stage.execute(_doc({"op": "collect", "input": _join_plan(
+        bs, ps, ["bk"], ["pk"], "anti",
+        [{"side": "probe", "col": "pid"}])}))
+    want = [i for i, k in enumerate(pk_list) if k is None or k not in matched]
+    assert sorted(r[0] for r in res["rows"]) == want
+
+    # LEFT (probe preserved): every probe row; unmatched build side -> NULL
+    res = stage.execute(_doc({"op": "collect", "input": _join_plan(
+        bs, ps, ["bk"], ["pk"], "left",
+        [{"side": "probe", "col": "pid"},
+         {"side": "build", "col": "bv"}])}))
+    assert len(res["rows"]) == np_
+    bmap = {int(k): int(v) for k, v in zip(bkeys, bval)}
+    for pid, bv in res["rows"]:
+        k = pk_list[pid]
+        if k is None or k not in bmap:
+            assert bv is None
+        else:
+            assert bv == bmap[k]
+
+
+def test_join_utf8_and_composite_keys(ctx, reg):
+    """Utf8 and composite (Int64, Utf8) join keys through the generalized
+    build/probe — the q2/q9-class key shapes round 1 lacked."""
+    rng = np.random.default_rng(43)
+    nb, np_ = 800, 30_000
+    names = [f"part#{i:05d}" for i in range(nb)]
+    bval = rng.integers(0, 10**6, size=nb, dtype=np.int64)
+    build = pa.table({"name": pa.array(names), "bv": pa.array(bval)})
+    p_name = [names[i] if i < nb else f"miss#{i}"
+              for i in rng.integers(0, nb + 200, size=np_)]
+    probe = pa.table({"pname": pa.array(p_name),
+                      "pid": pa.array(np.arange(np_, dtype=np.int64))})
+    reg("ub", build)
+    reg("up", probe)
+    res = stage.execute(_doc({"op": "collect", "input": _join_plan(
+        scan_of(build, "ub"), scan_of(probe, "up"),
+        ["name"], ["pname"], "inner",
+        [{"side": "probe", "col": "pid"},
+         {"side": "build", "col": "bv"}])}))
+    bmap = dict(zip(names, (int(v) for v in bval)))
+    want = sorted((i, bmap[nm]) for i, nm in enumerate(p_name)
+                  if nm in bmap)
+    got = sorted((r[0], r[1]) for r in res["rows"])
+    assert got == want
+
+    # composite (int64, utf8) key
+    b2 = pa.table({"k1": pa.array(np.arange(nb, dtype=np.int64) % 50),
+                   "k2": pa.array(names),
+                   "bv": pa.array(bval)})
+    p2 = pa.table({"q1": pa.array(
+        rng.integers(0, 50, size=np_, dtype=np.int64)),
+        "q2": pa.array(p_name),
+        "pid": pa.array(np.arange(np_, dtype=np.int64))})
+    reg("cb", b2)
+    reg("cp", p2)
+    res = stage.execute(_doc({"op": "collect", "input": _join_plan(
+        scan_of(b2, "cb"), scan_of(p2, "cp"),
+        ["k1", "k2"], ["q1", "q2"], "inner",
+        [{"side": "probe", "col": "pid"},
+         {"side": "build", "col": "bv"}])}))
+    bmap2 = {(int(i % 50), names[i]): int(bval[i]) for i in range(nb)}
+    want2 = sorted((i, bmap2[(int(q1), q2)])
+                   for i, (q1, q2) in enumerate(zip(p2["q1"].to_numpy(),
+                                                    p_name))
+                   if (int(q1), q2) in bmap2)
+    got2 = sorted((r[0], r[1]) for r in res["rows"])
+    assert got2 == want2
+
+
 def test_stage_errors_fail_loudly(ctx):
     with pytest.raises(RuntimeError, match="unregistered device table"):
         stage.execute(_doc({"op": "collect", "input": {
