@@ -74,16 +74,17 @@ def orders_lineitem(sf: int, device, seed: int = 19920101,
     # lines per order: uniform 1..7 (spec); expand with repeat_interleave
     lines = _gen((n_ord,), 1, 8, g, device)
     if lines_cap is not None:
-        # trim the tail so the lineitem table hits an exact target size
-        csum = torch.cumsum(lines, 0)
-        keep = int(torch.searchsorted(csum, torch.tensor(
-            lines_cap, device=device)).item())
-        lines = lines[:keep + 1]
-        if csum[keep] > lines_cap:
-            lines[keep] -= int((csum[keep] - lines_cap).item())
-        o_idx_src = torch.arange(lines.shape[0], device=device)
-    else:
-        o_idx_src = torch.arange(n_ord, device=device)
+        # nudge to the exact pinned row count (fixtures.rs): add or remove
+        # single lines from the first orders with slack — a vanishing
+        # perturbation of the 1..7 distribution (~1e-4 of orders)
+        total = int(lines.sum().item())
+        if total < lines_cap:
+            slack = (lines < 7).nonzero(as_tuple=True)[0][:lines_cap - total]
+            lines[slack] += 1
+        elif total > lines_cap:
+            slack = (lines > 1).nonzero(as_tuple=True)[0][:total - lines_cap]
+            lines[slack] -= 1
+    o_idx_src = torch.arange(n_ord, device=device)
     l_order_idx = torch.repeat_interleave(o_idx_src, lines)
     n_li = l_order_idx.shape[0]
 
