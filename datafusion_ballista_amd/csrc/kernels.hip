@@ -1818,7 +1818,177 @@ __global__ void k_hashagg(KeyArgs keys, AggArgs aggs, const u64* mask_words,
   }
 }
 
+// Low-cardinality fast path (q1-class: a handful of groups over hundreds
+// of millions of rows): the global open table serialises every row on a
+// few device-scope atomic lines (measured 21 s for SF100 q1 through the
+// generic path).  Each BLOCK accumulates into its own LDS table with the
+// same record layout, then flushes occupied slots into the global table
+// with op-aware atomic merges.  If a block overflows its LDS capacity it
+// aborts the whole launch with err=2 (no flush) and the host falls back
+// to the global-table kernel on a zeroed table.
+__global__ void k_hashagg_lds(KeyArgs keys, AggArgs aggs,
+                              const u64* mask_words, int64_t n,
+                              u64 lds_cap_mask, u64 gcap_mask,
+                              u64* slot_data, int* err_flag) {
+  extern __shared__ u64 lt[];
+  const int wpa = 2 + aggs.vmode;
+  const int rec = 2 + wpa * aggs.naggs;
+  const int64_t lds_cap = (int64_t)lds_cap_mask + 1;
+  for (int64_t w = threadIdx.x; w < lds_cap * rec; w += blockDim.x) lt[w] = 0;
+  __shared__ int overflow;
+  if (threadIdx.x == 0) overflow = 0;
+  __syncthreads();
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    if (__hip_atomic_load(&overflow, __ATOMIC_RELAXED,
+                          __HIP_MEMORY_SCOPE_WORKGROUP))
+      break;
+    if (mask_words && !((mask_words[i >> 6] >> (i & 63)) & 1)) continue;
+    const u64 h = hash_keys_row(keys, i);
+    u64 slot = h & lds_cap_mask;
+    int64_t probes = 0;
+    u64* srec = nullptr;
+    while (true) {
+      u64* cand = lt + slot * rec;
+      u64 cur = __hip_atomic_load(cand, __ATOMIC_RELAXED,
+                                  __HIP_MEMORY_SCOPE_WORKGROUP);
+      if (cur == 0) {
+        u64 old = atomicCAS((unsigned long long*)cand, 0ull, (u64)i + 1);
+        if (old == 0) { srec = cand; break; }
+        cur = old;
+      }
+      if (cur != 0 && keys_equal_rows(keys, (int64_t)cur - 1, i)) {
+        srec = cand;
+        break;
+      }
+      slot = (slot + 1) & lds_cap_mask;
+      if (++probes > (int64_t)lds_cap_mask) {
+        atomicExch(&overflow, 1);
+        atomicExch(err_flag, 2);
+        srec = nullptr;
+        break;
+      }
+    }
+    if (!srec) break;
+    atomicAdd((unsigned long long*)&srec[1], 1ull);
+    for (int a = 0; a < aggs.naggs; ++a) {
+      u64* base = srec + 2 + wpa * a;
+      if (!bit_valid(aggs.a[a].valid, i)) continue;
+      if (aggs.vmode) atomicAdd((unsigned long long*)&base[2], 1ull);
+      switch (aggs.a[a].op) {
+        case BG_AGG_SUM_DEC128: {
+          const ulong2 v =
+              reinterpret_cast<const ulong2*>(aggs.a[a].data)[i];
+          atomic_add_i128(base, base + 1, make_i128(v.x, (i64)v.y));
+          break;
+        }
+        case BG_AGG_SUM_I64: {
+          const i64 v = reinterpret_cast<const int64_t*>(aggs.a[a].data)[i];
+          atomic_add_i128(base, base + 1, (i128)v);
+          break;
+        }
+        case BG_AGG_MAX_I64: {
+          const u64 v = (u64) reinterpret_cast<const int64_t*>(
+                            aggs.a[a].data)[i] ^ 0x8000000000000000ull;
+          atomicMax((unsigned long long*)base, v);
+          break;
+        }
+        case BG_AGG_MIN_I64: {
+          const u64 v = ~((u64) reinterpret_cast<const int64_t*>(
+                              aggs.a[a].data)[i] ^ 0x8000000000000000ull);
+          atomicMax((unsigned long long*)base, v);
+          break;
+        }
+        case BG_AGG_SUM_F64: {
+          const double v = reinterpret_cast<const double*>(aggs.a[a].data)[i];
+          atomicAdd(reinterpret_cast<double*>(base), v);
+          break;
+        }
+        case BG_AGG_MAX_F64: {
+          const u64 bits = (u64) reinterpret_cast<const int64_t*>(
+                               aggs.a[a].data)[i];
+          const u64 v = (bits & 0x8000000000000000ull)
+                            ? ~bits
+                            : bits ^ 0x8000000000000000ull;
+          atomicMax((unsigned long long*)base, v);
+          break;
+        }
+        case BG_AGG_MIN_F64: {
+          const u64 bits = (u64) reinterpret_cast<const int64_t*>(
+                               aggs.a[a].data)[i];
+          const u64 v = (bits & 0x8000000000000000ull)
+                            ? ~bits
+                            : bits ^ 0x8000000000000000ull;
+          atomicMax((unsigned long long*)base, ~v);
+          break;
+        }
+        default:
+          break;
+      }
+    }
+  }
+  __syncthreads();
+  if (overflow) return;
+  // flush: merge each occupied LDS record into the global table
+  for (int64_t s = threadIdx.x; s < lds_cap; s += blockDim.x) {
+    u64* lrec = lt + s * rec;
+    const u64 claim = lrec[0];
+    if (!claim) continue;
+    const int64_t row = (int64_t)claim - 1;
+    const u64 h = hash_keys_row(keys, row);
+    u64 slot = h & gcap_mask;
+    int64_t probes = 0;
+    u64* grec = nullptr;
+    while (true) {
+      u64* cand = slot_data + slot * rec;
+      u64 cur = __hip_atomic_load(cand, __ATOMIC_RELAXED,
+                                  __HIP_MEMORY_SCOPE_AGENT);
+      if (cur == 0) {
+        u64 old = atomicCAS(cand, 0ull, claim);
+        if (old == 0) { grec = cand; break; }
+        cur = old;
+      }
+      if (cur != 0 && keys_equal_rows(keys, (int64_t)cur - 1, row)) {
+        grec = cand;
+        break;
+      }
+      slot = (slot + 1) & gcap_mask;
+      if (++probes > (int64_t)gcap_mask) {
+        atomicExch(err_flag, 1);
+        return;
+      }
+    }
+    atomicAdd(&grec[1], lrec[1]);
+    for (int a = 0; a < aggs.naggs; ++a) {
+      u64* g = grec + 2 + wpa * a;
+      const u64* l = lrec + 2 + wpa * a;
+      if (aggs.vmode) atomicAdd(&g[2], l[2]);
+      switch (aggs.a[a].op) {
+        case BG_AGG_SUM_DEC128:
+        case BG_AGG_SUM_I64:
+          atomic_add_i128(g, g + 1, make_i128(l[0], (i64)l[1]));
+          break;
+        case BG_AGG_MAX_I64:
+        case BG_AGG_MIN_I64:
+        case BG_AGG_MAX_F64:
+        case BG_AGG_MIN_F64:
+          atomicMax(g, l[0]);
+          break;
+        case BG_AGG_SUM_F64: {
+          double v;
+          memcpy(&v, &l[0], 8);
+          atomicAdd(reinterpret_cast<double*>(g), v);
+          break;
+        }
+        default:
+          break;
+      }
+    }
+  }
+}
+
 // occupancy bitmask over slots (feeds the stable compaction)
+
 __global__ void k_slot_occupancy(const u64* slot_data, int rec, int64_t cap,
                                  u64* mask_words) {
   const int64_t wave_global =
@@ -1896,13 +2066,37 @@ static int hashagg_impl(const bg_column* key_cols, int32_t nkeys,
 
   int blocks = (int)bg_imin64((n + BG_BLOCK - 1) / BG_BLOCK, BG_MAX_BLOCKS);
   if (blocks == 0) blocks = 1;
+  // low-cardinality first attempt: per-block LDS tables (48 KB budget).
+  // Overflow (more groups than the LDS capacity) aborts with err=2 and
+  // we fall back to the global-table kernel on a zeroed table.
+  u64 lds_cap = 8;
+  while (lds_cap * 2 * (u64)rec * 8 <= 48 * 1024 && lds_cap < 4096)
+    lds_cap <<= 1;
+  const bool try_lds = lds_cap >= 64 && n >= (1 << 16);
   hipEvent_t ev0, ev1;
   HIP_TRY(hipEventCreate(&ev0));
   HIP_TRY(hipEventCreate(&ev1));
   HIP_TRY(hipEventRecord(ev0, 0));
-  hipLaunchKernelGGL(k_hashagg, dim3(blocks), dim3(BG_BLOCK), 0, 0, keys, aggs,
-                     reinterpret_cast<const u64*>(d_mask), n, cap - 1,
-                     slot_data, err_flag);
+  bool lds_done = false;
+  if (try_lds) {
+    hipLaunchKernelGGL(k_hashagg_lds, dim3(blocks), dim3(BG_BLOCK),
+                       lds_cap * (u64)rec * 8, 0, keys, aggs,
+                       reinterpret_cast<const u64*>(d_mask), n, lds_cap - 1,
+                       cap - 1, slot_data, err_flag);
+    HIP_TRY(hipGetLastError());
+    int lerr = 0;
+    HIP_TRY(hipMemcpy(&lerr, err_flag, sizeof(int), hipMemcpyDeviceToHost));
+    if (lerr == 2) {  // LDS overflow: clean retry on the global path
+      HIP_TRY(hipMemset(slot_data, 0, sizeof(u64) * cap * rec));
+      HIP_TRY(hipMemset(err_flag, 0, sizeof(int)));
+    } else {
+      lds_done = true;
+    }
+  }
+  if (!lds_done)
+    hipLaunchKernelGGL(k_hashagg, dim3(blocks), dim3(BG_BLOCK), 0, 0, keys,
+                       aggs, reinterpret_cast<const u64*>(d_mask), n, cap - 1,
+                       slot_data, err_flag);
   HIP_TRY(hipGetLastError());
   HIP_TRY(hipEventRecord(ev1, 0));
   HIP_TRY(hipEventSynchronize(ev1));
