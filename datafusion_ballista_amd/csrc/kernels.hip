@@ -1374,72 +1374,53 @@ extern "C" int bg_q1_agg(const bg_column* rf, const bg_column* ls,
 // (count -> exclusive scan -> fill) so the output is probe-major with exact
 // offsets and no atomic append nondeterminism in sizes.
 // ---------------------------------------------------------------------------
-static inline u64 next_pow2_u64(u64 v) {
-  u64 p = 8;
-  while (p < v) p <<= 1;
-  return p;
-}
-
 struct BgJoinTable {
-  // open-addressing table (round 2): one u64 slot per build row at load
-  // factor <= 0.5 — [fingerprint(32) | row+1(32)], fingerprint = high hash
-  // bits (never 0).  8 slots share a cache line, so a probe's
-  // scan-to-empty usually touches ONE random line (the chained layout's
-  // head+node walk touched ~2.5); the build-key verify load only happens
-  // on fingerprint hits.  The single-word CAS publish removes the
-  // two-word store/publish race the chained node layout had to tiptoe
-  // around (the og-encoding anomaly).
-  u64* slots;
+  ulong2* nodes;  // packed chain node: {key, next} — ONE line per hop
+  int* head;      // bucket heads (-1 empty)
   int64_t n_build;
-  u64 mask;  // nslots - 1
-  const int64_t* keys;        // caller's build keys (must outlive handle)
-  const uint8_t* valid;
-  i64* probe_offsets = nullptr;
+  u64 mask;       // nb - 1
+  i64* probe_offsets = nullptr;  // per-probe-row output offsets (count phase)
   int64_t probe_n = 0;
 };
 
-__device__ __forceinline__ u64 join_pack(u64 h, int64_t row) {
-  const u64 fp = (u64)((uint32_t)(h >> 32) | 1u);
-  return (fp << 32) | (u64)(uint32_t)(row + 1);
-}
-
 __global__ void k_join_build(const int64_t* keys, const uint8_t* valid,
-                             int64_t n, u64* slots, u64 mask) {
+                             int64_t n, int* head, ulong2* nodes, u64 mask) {
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
        i += (int64_t)gridDim.x * blockDim.x) {
-    // inner-join null semantics (null_equals_null=false): NULL matches
-    // nothing — the row is never inserted
+    // inner-join null semantics (null_equals_null=false in the reference's
+    // HashJoinExec): a NULL build key matches nothing — leave it unlinked
     if (!bit_valid(valid, i)) continue;
-    const u64 h = bg_hash_u64((u64)keys[i]);
-    const u64 packed = join_pack(h, i);
-    u64 slot = h & mask;
-    while (atomicCAS(&slots[slot], 0ull, packed) != 0ull)
-      slot = (slot + 1) & mask;
+    const u64 k = (u64)keys[i];
+    const u64 b = bg_hash_u64(k) & mask;
+    const int prev = atomicExch(&head[b], (int)i);
+    ulong2 node;
+    node.x = k;
+    node.y = (u64)(int64_t)prev;  // sign-extended: -1 terminates
+    nodes[i] = node;
   }
 }
 
-// 4-way ILP batching: interleave four independent slot scans so the
-// random-line loads overlap (memory-level parallelism)
+// 4-way ILP batching: each thread interleaves four independent chain
+// walks so the ~900-cycle random node loads overlap (memory-level
+// parallelism) instead of serialising per probe.
 #define JOIN_ILP 4
 __global__ void k_join_count(const int64_t* probe_keys,
                              const uint8_t* probe_valid, int64_t n_probe,
-                             const int64_t* build_keys, const u64* slots,
-                             u64 mask, uint32_t* counts) {
+                             const int* head, const ulong2* nodes, u64 mask,
+                             uint32_t* counts) {
   const int64_t stride = (int64_t)gridDim.x * blockDim.x;
   const int64_t tid = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
   for (int64_t base = tid; base < n_probe; base += stride * JOIN_ILP) {
     int64_t idx[JOIN_ILP];
-    u64 key[JOIN_ILP], fp[JOIN_ILP], slot[JOIN_ILP];
+    u64 key[JOIN_ILP];
+    int64_t cur[JOIN_ILP];
     u64 cnt[JOIN_ILP];
-    bool act[JOIN_ILP];
 #pragma unroll
     for (int j = 0; j < JOIN_ILP; ++j) {
       idx[j] = base + (int64_t)j * stride;
-      act[j] = idx[j] < n_probe && bit_valid(probe_valid, idx[j]);
-      key[j] = act[j] ? (u64)probe_keys[idx[j]] : 0;
-      const u64 h = bg_hash_u64(key[j]);
-      fp[j] = (u64)((uint32_t)(h >> 32) | 1u);
-      slot[j] = h & mask;
+      const bool act = idx[j] < n_probe && bit_valid(probe_valid, idx[j]);
+      key[j] = act ? (u64)probe_keys[idx[j]] : 0;
+      cur[j] = act ? (int64_t)head[bg_hash_u64(key[j]) & mask] : -1;
       cnt[j] = 0;
     }
     bool any = true;
@@ -1447,15 +1428,12 @@ __global__ void k_join_count(const int64_t* probe_keys,
       any = false;
 #pragma unroll
       for (int j = 0; j < JOIN_ILP; ++j) {
-        if (!act[j]) continue;
-        const u64 s = slots[slot[j]];
-        if (s == 0) { act[j] = false; continue; }
-        if ((s >> 32) == fp[j]) {
-          const int64_t row = (int64_t)(uint32_t)s - 1;
-          if ((u64)build_keys[row] == key[j]) ++cnt[j];
+        if (cur[j] >= 0) {
+          const ulong2 node = nodes[cur[j]];
+          if (node.x == key[j]) cnt[j]++;
+          cur[j] = (int64_t)node.y;
+          any = true;
         }
-        slot[j] = (slot[j] + 1) & mask;
-        any = true;
       }
     }
 #pragma unroll
@@ -1465,71 +1443,72 @@ __global__ void k_join_count(const int64_t* probe_keys,
 }
 
 __global__ void k_join_fill(const int64_t* probe_keys,
-                            const uint8_t* probe_valid, int64_t n_probe,
-                            const int64_t* build_keys, const u64* slots,
-                            u64 mask, const i64* offsets, uint32_t* out_probe,
+                             const uint8_t* probe_valid, int64_t n_probe,
+                            const int* head, const ulong2* nodes, u64 mask,
+                            const i64* offsets, uint32_t* out_probe,
                             uint32_t* out_build) {
   const int64_t stride = (int64_t)gridDim.x * blockDim.x;
   const int64_t tid = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
   for (int64_t base = tid; base < n_probe; base += stride * JOIN_ILP) {
     int64_t idx[JOIN_ILP];
-    u64 key[JOIN_ILP], fp[JOIN_ILP], slot[JOIN_ILP];
+    u64 key[JOIN_ILP];
+    int64_t cur[JOIN_ILP];
     i64 w[JOIN_ILP];
-    bool act[JOIN_ILP];
 #pragma unroll
     for (int j = 0; j < JOIN_ILP; ++j) {
       idx[j] = base + (int64_t)j * stride;
-      act[j] = idx[j] < n_probe && bit_valid(probe_valid, idx[j]);
-      key[j] = act[j] ? (u64)probe_keys[idx[j]] : 0;
-      const u64 h = bg_hash_u64(key[j]);
-      fp[j] = (u64)((uint32_t)(h >> 32) | 1u);
-      slot[j] = h & mask;
-      w[j] = act[j] ? offsets[idx[j]] : 0;
+      const bool act = idx[j] < n_probe && bit_valid(probe_valid, idx[j]);
+      key[j] = act ? (u64)probe_keys[idx[j]] : 0;
+      cur[j] = act ? (int64_t)head[bg_hash_u64(key[j]) & mask] : -1;
+      w[j] = act ? offsets[idx[j]] : 0;
     }
     bool any = true;
     while (any) {
       any = false;
 #pragma unroll
       for (int j = 0; j < JOIN_ILP; ++j) {
-        if (!act[j]) continue;
-        const u64 s = slots[slot[j]];
-        if (s == 0) { act[j] = false; continue; }
-        if ((s >> 32) == fp[j]) {
-          const int64_t row = (int64_t)(uint32_t)s - 1;
-          if ((u64)build_keys[row] == key[j]) {
+        if (cur[j] >= 0) {
+          const ulong2 node = nodes[cur[j]];
+          if (node.x == key[j]) {
             out_probe[w[j]] = (uint32_t)idx[j];
-            out_build[w[j]] = (uint32_t)row;
+            out_build[w[j]] = (uint32_t)cur[j];
             ++w[j];
           }
+          cur[j] = (int64_t)node.y;
+          any = true;
         }
-        slot[j] = (slot[j] + 1) & mask;
-        any = true;
       }
     }
   }
+}
+
+static u64 next_pow2_u64(u64 x) {
+  u64 p = 1;
+  while (p < x) p <<= 1;
+  return p;
 }
 
 extern "C" int bg_hashjoin_build(const bg_column* build_keys, int64_t n,
                                  void** out_handle) {
   REQUIRE_INIT();
   if (build_keys->dtype != BG_DT_INT64)
-    return set_err(BG_ERR_UNSUPPORTED,
-                   "bg_hashjoin_build: INT64 fast path (use build2)");
+    return set_err(BG_ERR_UNSUPPORTED, "join keys must be INT64 (round 1)");
   if (n > 0x7fffffffLL) return set_err(BG_ERR_INVALID, "build side > 2^31 rows");
   BgJoinTable t{};
   t.n_build = n;
-  t.keys = (const int64_t*)build_keys->d_data;
-  t.valid = build_keys->d_validity;
   const u64 nb = next_pow2_u64((u64)(n > 4 ? n * 2 : 8));
   t.mask = nb - 1;
-  HIP_TRY(pool_malloc((void**)&t.slots, sizeof(u64) * nb));
-  HIP_TRY(hipMemset(t.slots, 0, sizeof(u64) * nb));
+  HIP_TRY(pool_malloc((void**)&t.nodes, sizeof(ulong2) * (n ? n : 1)));
+  HIP_TRY(pool_malloc((void**)&t.head, sizeof(int) * nb));
+  HIP_TRY(hipMemset(t.head, 0xff, sizeof(int) * nb));  // -1
   int blocks = (int)bg_imin64((n + BG_BLOCK - 1) / BG_BLOCK, BG_MAX_BLOCKS);
   if (blocks == 0) blocks = 1;
   hipLaunchKernelGGL(k_join_build, dim3(blocks), dim3(BG_BLOCK), 0, 0,
-                     t.keys, t.valid, n, t.slots, t.mask);
+                     (const int64_t*)build_keys->d_data,
+                     build_keys->d_validity, n, t.head, t.nodes, t.mask);
   HIP_TRY(hipGetLastError());
-  *out_handle = new BgJoinTable(t);
+  BgJoinTable* h = new BgJoinTable(t);
+  *out_handle = h;
   return BG_OK;
 }
 
@@ -1539,8 +1518,10 @@ extern "C" int bg_hashjoin_probe_count(void* handle,
   REQUIRE_INIT();
   BgJoinTable* t = (BgJoinTable*)handle;
   if (probe_keys->dtype != BG_DT_INT64)
-    return set_err(BG_ERR_UNSUPPORTED,
-                   "bg_hashjoin_probe: INT64 fast path (use probe2)");
+    return set_err(BG_ERR_UNSUPPORTED, "join keys must be INT64 (round 1)");
+  // release the previous probe's offsets FIRST so the pool can recycle the
+  // buffer for this call (allocating before releasing forces a fresh
+  // multi-GB hipMalloc inside the hot path)
   if (t->probe_offsets) {
     (void)pool_release(t->probe_offsets);
     t->probe_offsets = nullptr;
@@ -1555,15 +1536,18 @@ extern "C" int bg_hashjoin_probe_count(void* handle,
   if (blocks == 0) blocks = 1;
   hipLaunchKernelGGL(k_join_count, dim3(blocks), dim3(BG_BLOCK), 0, 0,
                      (const int64_t*)probe_keys->d_data,
-                     probe_keys->d_validity, n, t->keys, t->slots, t->mask,
-                     d_counts);
+                     probe_keys->d_validity, n, t->head, t->nodes,
+                     t->mask, d_counts);
   {
+    // u32 counts: chains are < 2^31 long and per-row match counts fit —
+    // halves the count write and the scan's level-0/3 read traffic
     int rc = scan_exclusive_u32(d_counts, n, d_offs, d_total);
     if (rc != BG_OK) return rc;
   }
   i64 total = 0;
   HIP_TRY(hipMemcpy(&total, d_total, sizeof(i64), hipMemcpyDeviceToHost));
   HIP_TRY(pool_release(d_counts));
+  // stash offsets on the handle for the fill call
   t->probe_offsets = d_offs;
   t->probe_n = n;
   HIP_TRY(pool_release(d_total));
@@ -1583,8 +1567,9 @@ extern "C" int bg_hashjoin_probe_fill(void* handle,
   if (blocks == 0) blocks = 1;
   hipLaunchKernelGGL(k_join_fill, dim3(blocks), dim3(BG_BLOCK), 0, 0,
                      (const int64_t*)probe_keys->d_data,
-                     probe_keys->d_validity, n, t->keys, t->slots, t->mask,
-                     t->probe_offsets, d_out_probe, d_out_build);
+                     probe_keys->d_validity, n, t->head, t->nodes,
+                     t->mask, t->probe_offsets, d_out_probe,
+                     d_out_build);
   HIP_TRY(hipGetLastError());
   return BG_OK;
 }
@@ -1593,7 +1578,8 @@ extern "C" int bg_hashjoin_free(void* handle) {
   REQUIRE_INIT();
   BgJoinTable* t = (BgJoinTable*)handle;
   if (!t) return BG_OK;
-  (void)pool_release(t->slots);
+  (void)pool_release(t->nodes);
+  (void)pool_release(t->head);
   if (t->probe_offsets) (void)pool_release(t->probe_offsets);
   delete t;
   return BG_OK;
