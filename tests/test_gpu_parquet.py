@@ -41,7 +41,7 @@ def make_file(tmp_path, compression, n=200_000, use_dictionary=False):
     return path, table
 
 
-@pytest.mark.parametrize("compression", ["snappy", "none"])
+@pytest.mark.parametrize("compression", ["snappy", "none", "zstd", "gzip"])
 @pytest.mark.parametrize("use_dictionary", [False, True])
 def test_parquet_column_decode(ctx, tmp_path, compression, use_dictionary):
     """PLAIN and dictionary-encoded chunks (pyarrow's default is
