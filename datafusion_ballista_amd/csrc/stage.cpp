@@ -255,6 +255,89 @@ static Table gather_table(const Table& t, const uint32_t* d_idx, int64_t m) {
   return out;
 }
 
+// Zero-copy row slice of a column: [start, start+len).  start must be a
+// multiple of 64 so the validity bitmap slices at a byte boundary (the
+// chunked-probe spill path below slices at 64-row multiples).
+static Col slice_col(const Col& c, int64_t start, int64_t len) {
+  Col s = c;  // shares ownership of the underlying buffers
+  if (c.dtype == BG_DT_UTF8) {
+    // absolute offsets stay valid against the shared data buffer
+    s.ext_offs = c.optr() + start;
+    s.offsets = nullptr;
+  } else {
+    s.ext_data = (const uint8_t*)c.dptr() + start * dt_size(c.dtype);
+    s.data = nullptr;
+  }
+  if (c.nullable()) {
+    s.ext_valid = c.vptr() + start / 8;
+    s.validity = nullptr;
+  }
+  (void)len;
+  return s;
+}
+
+// Concatenate row-aligned column pieces (the chunked-probe merge).
+static Col concat_cols(const std::vector<Col>& pieces,
+                       const std::vector<int64_t>& lens) {
+  Col out;
+  const Col& c0 = pieces[0];
+  out.dtype = c0.dtype;
+  out.precision = c0.precision;
+  out.scale = c0.scale;
+  int64_t total = 0;
+  bool any_null = false;
+  for (size_t i = 0; i < pieces.size(); ++i) {
+    total += lens[i];
+    any_null |= pieces[i].nullable();
+  }
+  if (c0.dtype == BG_DT_UTF8) {
+    int64_t bytes = 0;
+    for (auto& p : pieces) bytes += p.data_bytes;
+    out.offsets = dalloc((uint64_t)(total + 1) * 4);
+    out.data = dalloc((uint64_t)(bytes > 0 ? bytes : 1));
+    out.data_bytes = bytes;
+    int64_t row = 0, db = 0;
+    for (size_t i = 0; i < pieces.size(); ++i) {
+      const Col& p = pieces[i];
+      if (lens[i] == 0) continue;
+      // piece offsets are 0-based (gather outputs); shift by the running
+      // data cursor: out = in - (-db)
+      chk(bg_sub_i32(p.optr(), lens[i] + 1, (int32_t)(-db),
+                     out.offsets->u8() + 4 * row),
+          "bg_sub_i32");
+      if (p.data_bytes)
+        chk(bg_memcpy_dtod(out.data->u8() + db, p.dptr(),
+                           (uint64_t)p.data_bytes),
+            "bg_memcpy_dtod");
+      row += lens[i];
+      db += p.data_bytes;
+    }
+  } else {
+    const int64_t esz = dt_size(c0.dtype);
+    out.data = dalloc((uint64_t)(total > 0 ? total : 1) * (uint64_t)esz);
+    int64_t row = 0;
+    for (size_t i = 0; i < pieces.size(); ++i) {
+      if (lens[i])
+        chk(bg_memcpy_dtod(out.data->u8() + row * esz, pieces[i].dptr(),
+                           (uint64_t)lens[i] * esz),
+            "bg_memcpy_dtod");
+      row += lens[i];
+    }
+  }
+  if (any_null) {
+    out.validity = dalloc((uint64_t)((total + 63) / 64) * 8 + 8);
+    chk(bg_memset(out.validity->p, 0xFF, out.validity->bytes), "bg_memset");
+    int64_t row = 0;
+    for (size_t i = 0; i < pieces.size(); ++i) {
+      if (lens[i] && pieces[i].nullable())
+        chk(bg_bitcopy(pieces[i].vptr(), lens[i], out.validity->u8(), row),
+            "bg_bitcopy");
+      row += lens[i];
+    }
+  }
+  return out;
+}
+
 // ---------------------------------------------------------------------------
 // scan sources
 // ---------------------------------------------------------------------------
@@ -819,86 +902,133 @@ static Table exec_join(const Value& node, Metrics& m) {
   else
     throw StageError(BG_ERR_UNSUPPORTED, "hash_join: join_type " + jt);
 
-  std::vector<bg_column> bk, pk;
+  std::vector<bg_column> bk;
+  std::vector<int> pk_ids;
   for (size_t i = 0; i < bkeys.size(); ++i) {
     bk.push_back(to_bg(build.cols[(size_t)build.idx(bkeys[i]->s)], build.n));
-    pk.push_back(to_bg(probe.cols[(size_t)probe.idx(pkeys[i]->s)], probe.n));
-    if (bk[i].dtype != pk[i].dtype)
+    pk_ids.push_back(probe.idx(pkeys[i]->s));
+    if (bk[i].dtype != probe.cols[(size_t)pk_ids[i]].dtype)
       throw StageError(BG_ERR_INVALID, "hash_join: key dtype mismatch");
   }
-  const bool fast_int64 =
-      bk.size() == 1 && bk[0].dtype == BG_DT_INT64 && join_type == BG_JOIN_INNER;
+  const bool fast_int64 = bk.size() == 1 && bk[0].dtype == BG_DT_INT64 &&
+                          join_type == BG_JOIN_INNER;
 
-  void* handle = nullptr;
-  int64_t matches = 0;
-  int rc;
-  if (fast_int64) {
-    chk(bg_hashjoin_build(&bk[0], build.n, &handle), "bg_hashjoin_build");
-    rc = bg_hashjoin_probe_count(handle, &pk[0], probe.n, &matches);
-  } else {
-    chk(bg_hashjoin_build2(bk.data(), (int32_t)bk.size(), build.n, &handle),
+  struct HandleGuard {
+    void* h = nullptr;
+    bool fast = false;
+    ~HandleGuard() {
+      if (h) {
+        if (fast) (void)bg_hashjoin_free(h);
+        else (void)bg_hashjoin_free2(h);
+      }
+    }
+  } guard;
+  guard.fast = fast_int64;
+  if (fast_int64)
+    chk(bg_hashjoin_build(&bk[0], build.n, &guard.h), "bg_hashjoin_build");
+  else
+    chk(bg_hashjoin_build2(bk.data(), (int32_t)bk.size(), build.n, &guard.h),
         "bg_hashjoin_build2");
-    rc = bg_hashjoin_probe_count2(handle, pk.data(), (int32_t)pk.size(),
-                                  probe.n, join_type, &matches);
-  }
-  if (rc != BG_OK) {
-    if (fast_int64) (void)bg_hashjoin_free(handle);
-    else (void)bg_hashjoin_free2(handle);
-    chk(rc, "hash_join probe count");
-  }
-  DBufPtr pidx = dalloc((uint64_t)(matches > 0 ? matches : 1) * 4);
-  DBufPtr bidx = dalloc((uint64_t)(matches > 0 ? matches : 1) * 4);
-  if (fast_int64) {
-    rc = bg_hashjoin_probe_fill(handle, &pk[0], probe.n, (uint32_t*)pidx->p,
-                                (uint32_t*)bidx->p);
-    (void)bg_hashjoin_free(handle);
-  } else {
-    rc = bg_hashjoin_probe_fill2(handle, pk.data(), (int32_t)pk.size(),
-                                 probe.n, join_type, (uint32_t*)pidx->p,
-                                 (uint32_t*)bidx->p);
-    (void)bg_hashjoin_free2(handle);
-  }
-  chk(rc, "hash_join probe fill");
 
-  // probe-outer: the build index vector carries BG_JOIN_NULL_IDX for
-  // unmatched probe rows — clamp it and remember the null mask so build-
-  // side gathers produce NULL columns
-  DBufPtr bidx_clamped, outer_valid;
-  if (join_type == BG_JOIN_OUTER_PROBE) {
-    bidx_clamped = dalloc((uint64_t)(matches > 0 ? matches : 1) * 4);
-    outer_valid = dalloc((uint64_t)((matches + 63) / 64) * 8 + 8);
-    chk(bg_memset(outer_valid->p, 0xFF, outer_valid->bytes), "bg_memset");
-    chk(bg_idx_sentinel((const uint32_t*)bidx->p, matches,
-                        (uint32_t*)bidx_clamped->p, outer_valid->u8()),
-        "bg_idx_sentinel");
+  // Memory-pressure bound (the spill analogue of sort_shuffle/writer.rs:
+  // 650-686 for join temporaries): probe in slices of probe_chunk_rows
+  // so the pair buffers + gathered outputs stay bounded regardless of the
+  // probe side's size; chunk outputs are concatenated afterwards.  0 =
+  // unchunked.  Slices start at 64-row multiples (validity byte-aligned).
+  int64_t chunk = node.get_int_or("probe_chunk_rows", 0);
+  if (chunk <= 0 || chunk >= probe.n) chunk = probe.n > 0 ? probe.n : 1;
+  chunk = (chunk + 63) & ~63LL;
+
+  auto& outputs = node.get_arr("output");
+  std::vector<std::vector<Col>> pieces(outputs.size());
+  std::vector<int64_t> piece_lens;
+  int64_t total_matches = 0;
+
+  for (int64_t start = 0; start < probe.n || (probe.n == 0 && start == 0);
+       start += chunk) {
+    const int64_t len =
+        probe.n - start < chunk ? probe.n - start : chunk;
+    // sliced probe key columns
+    std::vector<bg_column> pk;
+    std::vector<Col> pk_sliced;
+    for (int id : pk_ids) {
+      pk_sliced.push_back(slice_col(probe.cols[(size_t)id], start, len));
+      pk.push_back(to_bg(pk_sliced.back(), len));
+    }
+    int64_t matches = 0;
+    if (fast_int64)
+      chk(bg_hashjoin_probe_count(guard.h, &pk[0], len, &matches),
+          "bg_hashjoin_probe_count");
+    else
+      chk(bg_hashjoin_probe_count2(guard.h, pk.data(), (int32_t)pk.size(),
+                                   len, join_type, &matches),
+          "bg_hashjoin_probe_count2");
+    DBufPtr pidx = dalloc((uint64_t)(matches > 0 ? matches : 1) * 4);
+    DBufPtr bidx = dalloc((uint64_t)(matches > 0 ? matches : 1) * 4);
+    if (fast_int64)
+      chk(bg_hashjoin_probe_fill(guard.h, &pk[0], len, (uint32_t*)pidx->p,
+                                 (uint32_t*)bidx->p),
+          "bg_hashjoin_probe_fill");
+    else
+      chk(bg_hashjoin_probe_fill2(guard.h, pk.data(), (int32_t)pk.size(),
+                                  len, join_type, (uint32_t*)pidx->p,
+                                  (uint32_t*)bidx->p),
+          "bg_hashjoin_probe_fill2");
+
+    DBufPtr bidx_clamped, outer_valid;
+    if (join_type == BG_JOIN_OUTER_PROBE) {
+      bidx_clamped = dalloc((uint64_t)(matches > 0 ? matches : 1) * 4);
+      outer_valid = dalloc((uint64_t)((matches + 63) / 64) * 8 + 8);
+      chk(bg_memset(outer_valid->p, 0xFF, outer_valid->bytes), "bg_memset");
+      chk(bg_idx_sentinel((const uint32_t*)bidx->p, matches,
+                          (uint32_t*)bidx_clamped->p, outer_valid->u8()),
+          "bg_idx_sentinel");
+    }
+
+    for (size_t oi = 0; oi < outputs.size(); ++oi) {
+      const Value& o = *outputs[oi];
+      const std::string side = o.get_str("side");
+      const std::string col = o.get_str("col");
+      const bool outer_build =
+          side == "build" && join_type == BG_JOIN_OUTER_PROBE;
+      Col c;
+      if (side == "build") {
+        const uint32_t* idx =
+            (const uint32_t*)(outer_build ? bidx_clamped->p : bidx->p);
+        c = gather_col(build.cols[(size_t)build.idx(col)], build.n, idx,
+                       matches);
+        if (outer_build) {
+          if (c.validity) {
+            DBufPtr comb = dalloc((uint64_t)((matches + 63) / 64) * 8 + 8);
+            chk(bg_bitmap_and(c.validity->u8(), outer_valid->u8(), matches,
+                              comb->u8()),
+                "bg_bitmap_and");
+            c.validity = comb;
+          } else {
+            c.validity = outer_valid;
+          }
+        }
+      } else {
+        // probe indices are chunk-relative: gather from the SLICE
+        Col ps = slice_col(probe.cols[(size_t)probe.idx(col)], start, len);
+        c = gather_col(ps, len, (const uint32_t*)pidx->p, matches);
+      }
+      pieces[oi].push_back(std::move(c));
+    }
+    piece_lens.push_back(matches);
+    total_matches += matches;
+    if (probe.n == 0) break;
   }
 
   Table out;
-  out.n = matches;
-  for (auto& o : node.get_arr("output")) {
-    const std::string side = o->get_str("side");
-    const std::string col = o->get_str("col");
-    const Table& src = (side == "build") ? build : probe;
-    const bool outer_build =
-        side == "build" && join_type == BG_JOIN_OUTER_PROBE;
-    const uint32_t* idx =
-        (side == "build")
-            ? (const uint32_t*)(outer_build ? bidx_clamped->p : bidx->p)
-            : (const uint32_t*)pidx->p;
-    out.names.push_back(o->get_str_or("as", col));
-    Col c = gather_col(src.cols[(size_t)src.idx(col)], src.n, idx, matches);
-    if (outer_build) {
-      if (c.validity) {  // combine source validity with the outer mask
-        DBufPtr comb = dalloc((uint64_t)((matches + 63) / 64) * 8 + 8);
-        chk(bg_bitmap_and(c.validity->u8(), outer_valid->u8(), matches,
-                          comb->u8()),
-            "bg_bitmap_and");
-        c.validity = comb;
-      } else {
-        c.validity = outer_valid;
-      }
-    }
-    out.cols.push_back(std::move(c));
+  out.n = total_matches;
+  for (size_t oi = 0; oi < outputs.size(); ++oi) {
+    const Value& o = *outputs[oi];
+    out.names.push_back(o.get_str_or("as", o.get_str("col")));
+    if (pieces[oi].size() == 1)
+      out.cols.push_back(std::move(pieces[oi][0]));
+    else
+      out.cols.push_back(concat_cols(pieces[oi], piece_lens));
   }
   return out;
 }

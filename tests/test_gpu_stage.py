@@ -583,6 +583,40 @@ def test_join_utf8_and_composite_keys(ctx, reg):
     assert got2 == want2
 
 
+def test_join_chunked_probe_matches_unchunked(ctx, reg):
+    """probe_chunk_rows (the join-temporary spill bound, writer.rs:650-686
+    analogue): slicing the probe side must reproduce the unchunked result
+    exactly — incl. Utf8 payloads, NULL probe keys, and a left join whose
+    unmatched rows carry NULL build columns across chunk boundaries."""
+    rng = np.random.default_rng(47)
+    nb, np_ = 700, 25_000
+    bkeys = np.arange(0, 2 * nb, 2, dtype=np.int64)
+    bstr = [f"b{int(k) % 313}" for k in bkeys]
+    pkeys = rng.integers(0, 2 * nb, size=np_, dtype=np.int64)
+    pmask = rng.random(np_) < 0.08
+    pstr = [f"p{i % 511}" for i in range(np_)]
+    build = pa.table({"bk": pa.array(bkeys), "bs": pa.array(bstr)})
+    probe = pa.table({"pk": pa.array(pkeys, mask=pmask),
+                      "ps": pa.array(pstr),
+                      "pid": pa.array(np.arange(np_, dtype=np.int64))})
+    reg("ckb", build)
+    reg("ckp", probe)
+    for jt in ("inner", "left"):
+        plans = []
+        for chunk in (0, 4096):
+            jp = _join_plan(scan_of(build, "ckb"), scan_of(probe, "ckp"),
+                            ["bk"], ["pk"], jt,
+                            [{"side": "probe", "col": "pid"},
+                             {"side": "probe", "col": "ps"},
+                             {"side": "build", "col": "bs"}])
+            if chunk:
+                jp = dict(jp, probe_chunk_rows=chunk)
+            plans.append(stage.execute(_doc({"op": "collect", "input": jp})))
+        r0 = sorted(map(tuple, plans[0]["rows"]))
+        r1 = sorted(map(tuple, plans[1]["rows"]))
+        assert r0 == r1, jt
+
+
 def test_stage_errors_fail_loudly(ctx):
     with pytest.raises(RuntimeError, match="unregistered device table"):
         stage.execute(_doc({"op": "collect", "input": {
