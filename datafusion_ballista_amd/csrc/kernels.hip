@@ -4554,6 +4554,13 @@ extern "C" int bg_gather_varlen(const void* d_src_data,
   if (rc != BG_OK) return rc;
   i64 total = 0;
   HIP_TRY(hipMemcpy(&total, d_total, sizeof(i64), hipMemcpyDeviceToHost));
+  if (d_out_data == nullptr) {  // sizing call: report the needed bytes
+    (void)pool_release(d_lens);
+    (void)pool_release(d_offs);
+    (void)pool_release(d_total);
+    *out_total_bytes = total;
+    return BG_OK;
+  }
   if (total > out_data_cap)
     return set_err(BG_ERR_INVALID, "bg_gather_varlen: out_data too small");
   if (total > 0x7fffffffLL)

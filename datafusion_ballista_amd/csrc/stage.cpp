@@ -226,11 +226,17 @@ static Col gather_col(const Col& c, int64_t n_src, const uint32_t* d_idx,
   out.scale = c.scale;
   if (c.dtype == BG_DT_UTF8) {
     out.offsets = dalloc((uint64_t)(m + 1) * 4);
-    int64_t cap = c.data_bytes > 0 ? c.data_bytes : 1;
-    out.data = dalloc((uint64_t)cap);
+    // one-to-many joins can duplicate rows, so the gathered payload may
+    // EXCEED the source bytes: size exactly first (d_out_data NULL =
+    // sizing call), then copy
     int64_t total = 0;
     chk(bg_gather_varlen(c.dptr(), c.optr(), d_idx, m,
-                         (int32_t*)out.offsets->p, out.data->p, cap, &total),
+                         (int32_t*)out.offsets->p, nullptr, 0, &total),
+        "bg_gather_varlen(size)");
+    out.data = dalloc((uint64_t)(total > 0 ? total : 1));
+    chk(bg_gather_varlen(c.dptr(), c.optr(), d_idx, m,
+                         (int32_t*)out.offsets->p, out.data->p,
+                         total > 0 ? total : 1, &total),
         "bg_gather_varlen");
     out.data_bytes = total;
   } else {
