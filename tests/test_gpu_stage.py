@@ -191,6 +191,46 @@ def test_grouped_aggregate_with_nulls(ctx, reg):
         assert abs(r[5] - want["x_mean"][i].as_py()) < 1e-9
 
 
+def test_grouped_aggregate_low_cardinality_lds(ctx, reg):
+    """~50 groups over 200k rows: hits the per-block LDS aggregate path
+    (k_hashagg_lds; the q1-class shape whose global-atomic version was
+    346x slower at SF100) — values incl. NULLs and min/max/avg must match
+    pyarrow exactly."""
+    rng = np.random.default_rng(53)
+    n = 200_000
+    keys = rng.integers(0, 50, size=n, dtype=np.int64)
+    vals = rng.integers(-10**7, 10**7, size=n, dtype=np.int64)
+    vmask = rng.random(n) < 0.25
+    f64 = rng.standard_normal(n)
+    t = pa.table({"k": pa.array(keys), "v": pa.array(vals, mask=vmask),
+                  "x": pa.array(f64)})
+    reg("lds", t)
+    plan = {"op": "collect", "input": {
+        "op": "hash_aggregate", "mode": "single", "group_by": ["k"],
+        "aggs": [
+            {"fn": "sum", "as": "sv", "expr": {"col": "v"}},
+            {"fn": "min", "as": "mn", "expr": {"col": "v"}},
+            {"fn": "max", "as": "mx", "expr": {"col": "v"}},
+            {"fn": "min", "as": "mnx", "expr": {"col": "x"}},
+            {"fn": "max", "as": "mxx", "expr": {"col": "x"}},
+            {"fn": "count", "as": "c"},
+        ], "input": scan_of(t, "lds")}}
+    res = stage.execute(_doc(plan))
+    want = t.group_by("k").aggregate([
+        ("v", "sum"), ("v", "min"), ("v", "max"), ("x", "min"),
+        ("x", "max"), ("k", "count")]).sort_by("k")
+    got = sorted(res["rows"], key=lambda r: r[0])
+    assert len(got) == want.num_rows == 50
+    for i, r in enumerate(got):
+        assert r[0] == want["k"][i].as_py()
+        assert r[1] == want["v_sum"][i].as_py()
+        assert r[2] == want["v_min"][i].as_py()
+        assert r[3] == want["v_max"][i].as_py()
+        assert r[4] == want["x_min"][i].as_py()
+        assert r[5] == want["x_max"][i].as_py()
+        assert r[6] == want["k_count"][i].as_py()
+
+
 def test_join_stage(ctx, reg):
     rng = np.random.default_rng(13)
     nb, np_ = 3_000, 40_000
