@@ -657,6 +657,205 @@ def test_join_chunked_probe_matches_unchunked(ctx, reg):
         assert r0 == r1, jt
 
 
+def test_q3_staged_pipeline_with_shuffle_files(ctx, reg, tmp_path):
+    """q3 as the reference actually RUNS it: four tasks exchanging real
+    consolidated shuffle files (the staged execution of
+    execution_graph.rs:62-105 — each stage's output is a durable
+    addressable artifact, shuffle.md:66-68), every stage one
+    bg_execute_stage call:
+      stage 1: customer filter -> sort_shuffle_write (hash c_custkey, k=4)
+      stage 2: orders filter -> sort_shuffle_write (hash o_custkey, k=4)
+      stage 3 (x4 tasks, one per partition p): scan both shuffles at p ->
+               join -> partial agg -> sort_shuffle_write (hash group key)
+      stage 4: scan stage-3 shuffles -> final agg -> top-10
+    Cross-checked against the single-stage q3 plan on the same data."""
+    rng = np.random.default_rng(59)
+    ncust, nord, nli = 8_000, 60_000, 240_000
+    cutoff = 9204
+    cust = pa.table({
+        "c_custkey": pa.array(np.arange(1, ncust + 1, dtype=np.int64)),
+        "c_mktsegment": pa.array(rng.integers(0, 5, size=ncust)
+                                 .astype(np.uint8))})
+    orders = pa.table({
+        "o_orderkey": pa.array(np.arange(1, nord + 1, dtype=np.int64)),
+        "o_custkey": pa.array(rng.integers(1, ncust + 1, size=nord,
+                                           dtype=np.int64)),
+        "o_orderdate": pa.array(rng.integers(8900, 9500, size=nord,
+                                             dtype=np.int32),
+                                type=pa.date32()),
+        "o_shippriority": pa.array(np.zeros(nord, dtype=np.int32))})
+    li = pa.table({
+        "l_orderkey": pa.array(rng.integers(1, nord + 1, size=nli,
+                                            dtype=np.int64)),
+        "l_shipdate": pa.array(rng.integers(8900, 9500, size=nli,
+                                            dtype=np.int32),
+                               type=pa.date32()),
+        "l_extendedprice": pa.array(
+            [decimal.Decimal(int(v)) / 100 for v in
+             rng.integers(90000, 10495100, size=nli)],
+            type=pa.decimal128(15, 2)),
+        "l_discount": pa.array(
+            [decimal.Decimal(int(v)) / 100 for v in
+             rng.integers(0, 11, size=nli)],
+            type=pa.decimal128(15, 2))})
+    reg("sgc", cust)
+    reg("sgo", orders)
+    reg("sgl", li)
+    K = 4
+    wd = str(tmp_path)
+
+    def doc(plan, stage_id, task_id=0, schema=None, **kw):
+        d = {"job_id": "q3stg", "stage_id": stage_id, "task_id": task_id,
+             "work_dir": wd, "plan": plan}
+        if schema is not None:
+            d["schema_msg_hex"] = stage.schema_msg_hex(schema)
+        d.update(kw)
+        return d
+
+    # stage 1: filtered customer, hash-partitioned by c_custkey
+    s1_schema = pa.schema([("c_custkey", pa.int64())])
+    r1 = stage.execute(doc({
+        "op": "sort_shuffle_write", "k": K, "keys": [{"col": "c_custkey"}],
+        "input": {"op": "project",
+                  "exprs": [{"as": "c_custkey", "expr": {"col": "c_custkey"}}],
+                  "input": {"op": "filter",
+                            "predicates": [{"col": "c_mktsegment",
+                                            "cmp": "eq", "lo": 1}],
+                            "input": scan_of(cust, "sgc")}}},
+        1, schema=s1_schema))
+    s1_path = r1["partitions"][0]["path"]
+
+    # stage 2: filtered orders, hash-partitioned by o_custkey
+    s2_schema = pa.schema([("o_orderkey", pa.int64()),
+                           ("o_custkey", pa.int64()),
+                           ("o_orderdate", pa.date32()),
+                           ("o_shippriority", pa.int32())])
+    r2 = stage.execute(doc({
+        "op": "sort_shuffle_write", "k": K, "keys": [{"col": "o_custkey"}],
+        "input": {"op": "filter",
+                  "predicates": [{"col": "o_orderdate", "cmp": "lt",
+                                  "hi": cutoff}],
+                  "input": scan_of(orders, "sgo")}},
+        2, schema=s2_schema))
+    s2_path = r2["partitions"][0]["path"]
+
+    # stage 3 (one task per partition): co-partitioned join + partial agg,
+    # output hash-partitioned by l_orderkey for the final stage
+    s3_schema = pa.schema([
+        ("l_orderkey", pa.int64()), ("o_orderdate", pa.date32()),
+        ("o_shippriority", pa.int32()),
+        ("revenue", pa.decimal128(38, 4)), ("revenue$n", pa.int64())])
+    s3_paths = []
+    for p in range(K):
+        join1 = {"op": "hash_join",
+                 "build": {"op": "scan",
+                           "schema": stage.schema_json(s1_schema),
+                           "source": {"kind": "shuffle", "data": s1_path,
+                                      "index": s1_path + ".index",
+                                      "partitions": [p]}},
+                 "probe": {"op": "scan",
+                           "schema": stage.schema_json(s2_schema),
+                           "source": {"kind": "shuffle", "data": s2_path,
+                                      "index": s2_path + ".index",
+                                      "partitions": [p]}},
+                 "build_keys": ["c_custkey"], "probe_keys": ["o_custkey"],
+                 "join_type": "inner",
+                 "output": [{"side": "probe", "col": "o_orderkey"},
+                            {"side": "probe", "col": "o_orderdate"},
+                            {"side": "probe", "col": "o_shippriority"}]}
+        join2 = {"op": "hash_join", "build": join1,
+                 "probe": {"op": "filter",
+                           "predicates": [{"col": "l_shipdate", "cmp": "gt",
+                                           "lo": cutoff}],
+                           "input": scan_of(li, "sgl")},
+                 "build_keys": ["o_orderkey"], "probe_keys": ["l_orderkey"],
+                 "join_type": "inner",
+                 "output": [{"side": "probe", "col": "l_orderkey"},
+                            {"side": "build", "col": "o_orderdate"},
+                            {"side": "build", "col": "o_shippriority"},
+                            {"side": "probe", "col": "l_extendedprice"},
+                            {"side": "probe", "col": "l_discount"}]}
+        partial = {"op": "hash_aggregate", "mode": "partial",
+                   "group_by": ["l_orderkey", "o_orderdate",
+                                "o_shippriority"],
+                   "aggs": [{"fn": "sum", "as": "revenue",
+                             "expr": {"mul": [
+                                 {"col": "l_extendedprice"},
+                                 {"sub": [{"lit": 100},
+                                          {"col": "l_discount"}]}]}}],
+                   "input": join2}
+        rp = stage.execute(doc({
+            "op": "sort_shuffle_write", "k": K,
+            "keys": [{"col": "l_orderkey"}], "input": partial},
+            3, task_id=p, schema=s3_schema))
+        s3_paths.append(rp["partitions"][0]["path"])
+
+    # stage 4: final aggregate over every stage-3 output, top-10
+    final = {"op": "hash_aggregate", "mode": "final",
+             "group_by": ["l_orderkey", "o_orderdate", "o_shippriority"],
+             "aggs": [{"fn": "sum", "as": "revenue"}],
+             "input": {"op": "scan",
+                       "schema": stage.schema_json(s3_schema),
+                       "source": {"kind": "shuffle", "locations": [
+                           {"data": pth, "index": pth + ".index",
+                            "partitions": list(range(K))}
+                           for pth in s3_paths]}}}
+    r4 = stage.execute(doc({
+        "op": "collect", "limit": 10, "input": {
+            "op": "sort", "keys": [{"col": "revenue", "desc": True},
+                                   {"col": "o_orderdate", "desc": False}],
+            "limit": 10, "input": final}}, 4))
+
+    # reference: the single-stage q3 plan over the same registered tables
+    single = stage.execute(_doc({"op": "collect", "limit": 10, "input": {
+        "op": "sort", "keys": [{"col": "revenue", "desc": True},
+                               {"col": "o_orderdate", "desc": False}],
+        "limit": 10, "input": {
+            "op": "hash_aggregate", "mode": "single",
+            "group_by": ["l_orderkey", "o_orderdate", "o_shippriority"],
+            "aggs": [{"fn": "sum", "as": "revenue",
+                      "expr": {"mul": [{"col": "l_extendedprice"},
+                                       {"sub": [{"lit": 100},
+                                                {"col": "l_discount"}]}]}}],
+            "input": {"op": "hash_join",
+                      "build": {"op": "hash_join",
+                                "build": {"op": "filter",
+                                          "predicates": [
+                                              {"col": "c_mktsegment",
+                                               "cmp": "eq", "lo": 1}],
+                                          "input": scan_of(cust, "sgc")},
+                                "probe": {"op": "filter",
+                                          "predicates": [
+                                              {"col": "o_orderdate",
+                                               "cmp": "lt", "hi": cutoff}],
+                                          "input": scan_of(orders, "sgo")},
+                                "build_keys": ["c_custkey"],
+                                "probe_keys": ["o_custkey"],
+                                "join_type": "inner",
+                                "output": [
+                                    {"side": "probe", "col": "o_orderkey"},
+                                    {"side": "probe", "col": "o_orderdate"},
+                                    {"side": "probe",
+                                     "col": "o_shippriority"}]},
+                      "probe": {"op": "filter",
+                                "predicates": [{"col": "l_shipdate",
+                                                "cmp": "gt", "lo": cutoff}],
+                                "input": scan_of(li, "sgl")},
+                      "build_keys": ["o_orderkey"],
+                      "probe_keys": ["l_orderkey"],
+                      "join_type": "inner",
+                      "output": [{"side": "probe", "col": "l_orderkey"},
+                                 {"side": "build", "col": "o_orderdate"},
+                                 {"side": "build",
+                                  "col": "o_shippriority"},
+                                 {"side": "probe",
+                                  "col": "l_extendedprice"},
+                                 {"side": "probe",
+                                  "col": "l_discount"}]}}}}))
+    assert r4["rows"] == single["rows"]
+    assert len(r4["rows"]) == 10
+
+
 def test_stage_errors_fail_loudly(ctx):
     with pytest.raises(RuntimeError, match="unregistered device table"):
         stage.execute(_doc({"op": "collect", "input": {
