@@ -146,6 +146,36 @@ def unregister_table(name: str):
     gpu.load_library().bg_stage_unregister_table(name.encode())
 
 
+def parquet_source(path: str) -> dict:
+    """Build a scan{kind:"parquet"} source dict from the file's footer
+    (pyarrow parses footers natively; the Rust host uses parquet-rs the
+    same way).  The C++ scan walks page headers and decodes on device;
+    columns outside its subset fail loudly at execute time."""
+    import pyarrow.parquet as pq
+    pf = pq.ParquetFile(path)
+    md = pf.metadata
+    cols = []
+    for c in range(md.num_columns):
+        sc = pf.schema.column(c)
+        phys = md.row_group(0).column(c).physical_type
+        spec = {"phys": "FLBA" if phys == "FIXED_LEN_BYTE_ARRAY" else phys,
+                "max_def": sc.max_definition_level,
+                "codec": md.row_group(0).column(c).compression,
+                "chunks": []}
+        if spec["phys"] == "FLBA":
+            spec["flba_len"] = sc.length
+        for rg in range(md.num_row_groups):
+            m = md.row_group(rg).column(c)
+            start = m.data_page_offset
+            if m.has_dictionary_page and m.dictionary_page_offset is not None:
+                start = min(start, m.dictionary_page_offset)
+            spec["chunks"].append({"start": start,
+                                   "size": m.total_compressed_size,
+                                   "num_values": m.num_values})
+        cols.append(spec)
+    return {"kind": "parquet", "path": path, "columns": cols}
+
+
 def debug_rb_message(n_rows, nodes, bufs, body_len, compressed) -> bytes:
     L = _lib()
     nd = (ctypes.c_int64 * (2 * len(nodes)))(

@@ -115,11 +115,53 @@ def main():
                                    tpch_synth.Q6_QTY_LT)
     assert cnt == want_cnt and total == want_sum, \
         (cnt, total, want_cnt, want_sum)
+    # interpreter leg: the SAME q6 over scan{kind:"parquet"} — C++ page
+    # walk + device decode + fused kernel in one bg_execute_stage call
+    from datafusion_ballista_amd import stage as bgstage
+    scan_node = {"op": "scan", "schema": [
+        {"name": "l_shipdate", "dtype": "date32"},
+        {"name": "l_quantity", "dtype": "decimal128", "precision": 15,
+         "scale": 2},
+        {"name": "l_extendedprice", "dtype": "decimal128", "precision": 15,
+         "scale": 2},
+        {"name": "l_discount", "dtype": "decimal128", "precision": 15,
+         "scale": 2}],
+        "source": bgstage.parquet_source(path)}
+    q6_doc = {"job_id": "pq", "stage_id": 6, "task_id": 0,
+              "work_dir": "/tmp/x", "plan": {"op": "collect", "input": {
+                  "op": "hash_aggregate", "mode": "single", "group_by": [],
+                  "aggs": [{"fn": "sum", "as": "revenue",
+                            "expr": {"mul": [{"col": "l_extendedprice"},
+                                             {"col": "l_discount"}]}},
+                           {"fn": "count", "as": "cnt"}],
+                  "input": {"op": "filter", "predicates": [
+                      {"col": "l_shipdate", "cmp": "ge_lt",
+                       "lo": tpch_synth.Q6_DATE_LO,
+                       "hi": tpch_synth.Q6_DATE_HI},
+                      {"col": "l_discount", "cmp": "between",
+                       "lo": tpch_synth.Q6_DISC_LO,
+                       "hi": tpch_synth.Q6_DISC_HI},
+                      {"col": "l_quantity", "cmp": "lt",
+                       "hi": tpch_synth.Q6_QTY_LT}],
+                      "input": scan_node}}}}
+    ri = bgstage.execute(q6_doc)  # warmup (file-read + pool warm)
+    assert ri["rows"][0][1] == want_cnt
+    assert int(ri["rows"][0][0]) == want_sum
+    t_i = 1e30
+    for _ in range(5):
+        t0 = time.perf_counter()
+        ri = bgstage.execute(q6_doc)
+        t_i = min(t_i, time.perf_counter() - t0)
+    # NOTE: the interpreter leg re-reads the FILE from page cache and
+    # re-uploads every call (stateless stages) — PCIe-inclusive
+
     rec = {"rows": n, "file_mb": os.path.getsize(path) / 1e6,
            "upload_s": t_upload, "decode_s": t_dec,
            "decode_gbps": dbytes / t_dec / 1e9,
            "q6_from_parquet_s": t_all,
            "rows_per_s": n / t_all, "count": cnt,
+           "interp_q6_s_incl_file_read": t_i,
+           "interp_rows_per_s": n / t_i,
            "crosscheck": "exact vs oracle"}
     print(json.dumps(rec), flush=True)
     os.makedirs(os.path.join(ROOT, "gpurun_out"), exist_ok=True)

@@ -856,6 +856,70 @@ def test_q3_staged_pipeline_with_shuffle_files(ctx, reg, tmp_path):
     assert len(r4["rows"]) == 10
 
 
+def test_parquet_scan_in_stage(ctx, tmp_path):
+    """scan{kind:"parquet"}: the C++ page walk + device decode inside
+    bg_execute_stage — q6 over a pyarrow-written snappy lineitem file
+    (dictionary pages for low-cardinality columns, PLAIN fallback for
+    price, FLBA decimals, OPTIONAL columns), vs the oracle exactly.
+    Also: a whole-file collect must equal pyarrow's own reader."""
+    import pyarrow.parquet as pq
+
+    import oracle
+    from datafusion_ballista_amd import tpch_synth
+    n = 150_000
+    li = tpch_synth.lineitem_numpy(n, seed=29)
+
+    def dec_col(vals):
+        return pa.array([decimal.Decimal(int(v)) / 100 for v in vals],
+                        type=pa.decimal128(15, 2))
+
+    t = pa.table({
+        "l_shipdate": pa.array(li["l_shipdate"], type=pa.date32()),
+        "l_quantity": dec_col(li["l_quantity"]),
+        "l_extendedprice": dec_col(li["l_extendedprice"]),
+        "l_discount": dec_col(li["l_discount"]),
+    })
+    path = str(tmp_path / "li_stage.parquet")
+    pq.write_table(t, path, compression="snappy",
+                   data_page_size=32 * 1024, write_statistics=False)
+
+    scan_node = {"op": "scan", "schema": stage.schema_json(t.schema),
+                 "source": stage.parquet_source(path)}
+    plan = {"op": "collect", "input": {
+        "op": "hash_aggregate", "mode": "single", "group_by": [],
+        "aggs": [{"fn": "sum", "as": "revenue",
+                  "expr": {"mul": [{"col": "l_extendedprice"},
+                                   {"col": "l_discount"}]}},
+                 {"fn": "count", "as": "cnt"}],
+        "input": {"op": "filter", "predicates": [
+            {"col": "l_shipdate", "cmp": "ge_lt",
+             "lo": tpch_synth.Q6_DATE_LO, "hi": tpch_synth.Q6_DATE_HI},
+            {"col": "l_discount", "cmp": "between",
+             "lo": tpch_synth.Q6_DISC_LO, "hi": tpch_synth.Q6_DISC_HI},
+            {"col": "l_quantity", "cmp": "lt",
+             "hi": tpch_synth.Q6_QTY_LT}],
+            "input": scan_node}}}
+    res = stage.execute(_doc(plan))
+    d16 = tpch_synth.dec128_pairs_np(li["l_discount"]).view(np.uint8).reshape(-1)
+    q16 = tpch_synth.dec128_pairs_np(li["l_quantity"]).view(np.uint8).reshape(-1)
+    p16 = tpch_synth.dec128_pairs_np(li["l_extendedprice"]).view(np.uint8).reshape(-1)
+    want_cnt, want_sum = oracle.q6(
+        li["l_shipdate"], d16, q16, p16, tpch_synth.Q6_DATE_LO,
+        tpch_synth.Q6_DATE_HI, tpch_synth.Q6_DISC_LO, tpch_synth.Q6_DISC_HI,
+        tpch_synth.Q6_QTY_LT)
+    assert res["rows"][0][1] == want_cnt
+    assert int(res["rows"][0][0]) == want_sum
+
+    # whole-file scan collect vs pyarrow reader (first/last 100 rows)
+    res2 = stage.execute(_doc({"op": "collect", "limit": 100,
+                               "input": scan_node}))
+    ref = pq.read_table(path)
+    for i, row in enumerate(res2["rows"]):
+        assert row[0] == ref["l_shipdate"][i].as_py().toordinal() - 719163
+        assert int(row[2]) == int(ref["l_extendedprice"][i].as_py()
+                                  .scaleb(2))
+
+
 def test_stage_errors_fail_loudly(ctx):
     with pytest.raises(RuntimeError, match="unregistered device table"):
         stage.execute(_doc({"op": "collect", "input": {
