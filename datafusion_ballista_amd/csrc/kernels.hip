@@ -3214,6 +3214,281 @@ __global__ void k_snap_matches(const SnappyPage* __restrict__ pages,
   }
 }
 
+// ---------------------------------------------------------------------------
+// Giant-page snappy parse (round 2): the speculative chase above is serial
+// PER PAGE, so one 1-MB dict-fallback page (~300k elements) bounds the
+// whole decode (~70 ms) no matter how many other pages run in parallel.
+// Snappy parses SELF-SYNCHRONIZE: chains started at different offsets
+// merge within a few elements.  So, for pages above a size threshold:
+//   A  k_snapbig_spec    — one THREAD per 8-KB segment chases
+//      speculatively from the segment start (segment 0 starts at the real
+//      header end, so its chain is TRUE), recording per visited offset the
+//      consumed length (membership) and prefix (count, out-bytes), plus
+//      the segment's totals and exit offset.  77k segments across the
+//      file run concurrently.
+//   B  k_snapbig_resolve — one thread per page walks SEGMENTS (128/MB,
+//      not 300k elements): if the true entry offset lands on the
+//      segment's speculative chain (the common, self-synced case) the
+//      whole suffix is absorbed with two subtractions; otherwise it
+//      catches up element by element until it merges.  Emits per-segment
+//      true entry + descriptor/output prefixes.
+//   C  k_snapbig_emit    — one thread per segment re-decodes from its
+//      TRUE entry and writes descriptors at the precomputed positions,
+//      with the same validation as the serial parse.
+// Cost: ~10 B of scratch per source byte of big pages; the per-page
+// serial work drops from ~300k elements to ~128 segment steps.
+// ---------------------------------------------------------------------------
+
+#define SNAPBIG_SEG 8192
+#define SNAPBIG_THRESHOLD 98304  // pages with csz above this use A/B/C
+
+// scalar speculative decode of the element at page-relative offset o.
+// Returns false when no element can start at o (malformed/truncated).
+__device__ __forceinline__ bool snap_decode_at(
+    const uint8_t* __restrict__ s, int64_t src_len, int64_t o,
+    int64_t* consumed, int64_t* out, uint32_t* aux, uint32_t* kind) {
+  if (o >= src_len) return false;
+  const uint8_t tag = s[o];
+  const int t = tag & 3;
+  if (t == 0) {
+    int64_t len = (tag >> 2) + 1;
+    int nb = 0;
+    if (len > 60) {
+      nb = (int)len - 60;
+      if (o + 1 + nb > src_len || nb > 4) return false;
+      len = 0;
+      for (int i = 0; i < nb; ++i) len |= (int64_t)s[o + 1 + i] << (8 * i);
+      len += 1;
+    }
+    *consumed = 1 + nb + len;
+    *out = len;
+    *aux = (uint32_t)(o + 1 + nb);
+    *kind = 0;
+  } else if (t == 1) {
+    if (o + 1 >= src_len) return false;
+    *out = ((tag >> 2) & 7) + 4;
+    *aux = ((uint32_t)(tag >> 5) << 8) | s[o + 1];
+    *consumed = 2;
+    *kind = 1;
+  } else if (t == 2) {
+    if (o + 2 >= src_len) return false;
+    *out = (tag >> 2) + 1;
+    *aux = (uint32_t)s[o + 1] | ((uint32_t)s[o + 2] << 8);
+    *consumed = 3;
+    *kind = 1;
+  } else {
+    if (o + 4 >= src_len) return false;
+    *out = (tag >> 2) + 1;
+    *aux = (uint32_t)s[o + 1] | ((uint32_t)s[o + 2] << 8) |
+           ((uint32_t)s[o + 3] << 16) | ((uint32_t)s[o + 4] << 24);
+    *consumed = 5;
+    *kind = 1;
+  }
+  if (*consumed <= 0 || *out < 0 || o + *consumed > src_len) return false;
+  return true;
+}
+
+struct SnapBigPage {
+  int64_t page_idx;  // into the SnappyPage array
+  int64_t arr_base;  // per-byte array base (bytes)
+  int64_t seg_base;  // per-seg array base
+  int64_t nsegs;
+};
+
+// phase A: speculative per-segment chase
+__global__ void k_snapbig_spec(const SnappyPage* __restrict__ pages,
+                               const SnapBigPage* __restrict__ bigs,
+                               const int32_t* __restrict__ seg_page,
+                               int64_t total_segs,
+                               uint16_t* __restrict__ next16,
+                               uint32_t* __restrict__ cnt_pre,
+                               uint32_t* __restrict__ out_pre,
+                               uint32_t* __restrict__ seg_cnt_tot,
+                               uint32_t* __restrict__ seg_out_tot,
+                               int64_t* __restrict__ seg_exit,
+                               int64_t* __restrict__ page_hdr,
+                               int64_t* __restrict__ page_ulen) {
+  for (int64_t gs = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       gs < total_segs; gs += (int64_t)gridDim.x * blockDim.x) {
+    const SnapBigPage& bp = bigs[seg_page[gs]];
+    const int64_t k = gs - bp.seg_base;
+    const SnappyPage& pg = pages[bp.page_idx];
+    const uint8_t* s = pg.src;
+    const int64_t src_len = pg.src_len;
+    int64_t o = k * SNAPBIG_SEG;
+    if (k == 0) {
+      // parse the uncompressed-length varint; segment 0 starts at the
+      // TRUE element boundary
+      int64_t ulen = 0, si = 0;
+      int shift = 0, ok = 1;
+      while (si < src_len) {
+        const uint8_t b = s[si++];
+        ulen |= (int64_t)(b & 0x7f) << shift;
+        if (!(b & 0x80)) break;
+        shift += 7;
+        if (shift > 32) { ok = 0; break; }
+      }
+      if (ulen > pg.dst_cap) ok = 0;
+      page_hdr[seg_page[gs]] = ok ? si : -1;
+      page_ulen[seg_page[gs]] = ulen;
+      o = si;
+      if (!ok) {
+        seg_exit[gs] = -1;
+        seg_cnt_tot[gs] = 0;
+        seg_out_tot[gs] = 0;
+        continue;
+      }
+    }
+    const int64_t seg_end =
+        (k + 1) * SNAPBIG_SEG < src_len ? (k + 1) * SNAPBIG_SEG : src_len;
+    uint32_t cnt = 0, outsum = 0;
+    int64_t exitv = -1;
+    while (true) {
+      if (o >= seg_end) { exitv = o; break; }
+      int64_t consumed, out;
+      uint32_t aux, kind;
+      if (!snap_decode_at(s, src_len, o, &consumed, &out, &aux, &kind))
+        break;  // exitv stays -1 (chain dies inside the segment)
+      next16[bp.arr_base + o] =
+          consumed > 0xFFFE ? (uint16_t)0xFFFF : (uint16_t)consumed;
+      cnt_pre[bp.arr_base + o] = cnt;
+      out_pre[bp.arr_base + o] = outsum;
+      cnt += 1;
+      outsum += (uint32_t)out;
+      o += consumed;
+    }
+    seg_cnt_tot[gs] = cnt;
+    seg_out_tot[gs] = outsum;
+    seg_exit[gs] = exitv;
+  }
+}
+
+// phase B: one thread per page resolves segment entries/prefixes
+__global__ void k_snapbig_resolve(const SnappyPage* __restrict__ pages,
+                                  const SnapBigPage* __restrict__ bigs,
+                                  int64_t nbig,
+                                  const uint16_t* __restrict__ next16,
+                                  const uint32_t* __restrict__ cnt_pre,
+                                  const uint32_t* __restrict__ out_pre,
+                                  const uint32_t* __restrict__ seg_cnt_tot,
+                                  const uint32_t* __restrict__ seg_out_tot,
+                                  const int64_t* __restrict__ seg_exit,
+                                  const int64_t* __restrict__ page_hdr,
+                                  const int64_t* __restrict__ page_ulen,
+                                  int64_t* __restrict__ seg_entry,
+                                  int64_t* __restrict__ seg_nd,
+                                  int64_t* __restrict__ seg_di,
+                                  int64_t* __restrict__ counts,
+                                  int64_t* __restrict__ out_lens) {
+  for (int64_t b = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; b < nbig;
+       b += (int64_t)gridDim.x * blockDim.x) {
+    const SnapBigPage& bp = bigs[b];
+    const SnappyPage& pg = pages[bp.page_idx];
+    const uint8_t* s = pg.src;
+    const int64_t src_len = pg.src_len;
+    const int64_t hdr = page_hdr[b];
+    const int64_t ulen = page_ulen[b];
+    int err = hdr < 0;
+    int64_t cur = hdr, nd = 0, di = 0;
+    for (int64_t k = 0; k < bp.nsegs && !err; ++k) {
+      const int64_t gs = bp.seg_base + k;
+      const int64_t seg_end =
+          (k + 1) * SNAPBIG_SEG < src_len ? (k + 1) * SNAPBIG_SEG : src_len;
+      if (cur >= seg_end || di >= ulen) {
+        seg_entry[gs] = -1;
+        continue;
+      }
+      seg_entry[gs] = cur;
+      seg_nd[gs] = nd;
+      seg_di[gs] = di;
+      // catch up element by element until we merge with the segment's
+      // speculative chain (usually immediately) or leave the segment
+      bool advanced = false;
+      while (cur < seg_end) {
+        if (next16[bp.arr_base + cur] != 0) {
+          nd += (int64_t)seg_cnt_tot[gs] - cnt_pre[bp.arr_base + cur];
+          di += (int64_t)seg_out_tot[gs] - out_pre[bp.arr_base + cur];
+          const int64_t ex = seg_exit[gs];
+          if (ex < 0) { err = 1; }
+          cur = ex;
+          advanced = true;
+          break;
+        }
+        int64_t consumed, out;
+        uint32_t aux, kind;
+        if (!snap_decode_at(s, src_len, cur, &consumed, &out, &aux,
+                            &kind)) {
+          err = 1;
+          break;
+        }
+        nd += 1;
+        di += out;
+        cur += consumed;
+        advanced = true;
+      }
+      (void)advanced;
+      if (di > ulen) err = 1;
+    }
+    // a valid stream's last element ends exactly at src_len with
+    // di == ulen (snappy has no trailing bytes)
+    if (!err && di != ulen) err = 1;
+    counts[bp.page_idx] = err ? 0 : nd;
+    out_lens[bp.page_idx] = err ? -1 : ulen;
+  }
+}
+
+// phase C: one thread per live segment re-decodes its true chain and
+// emits descriptors at the precomputed positions (validation included)
+__global__ void k_snapbig_emit(const SnappyPage* __restrict__ pages,
+                               const SnapBigPage* __restrict__ bigs,
+                               const int32_t* __restrict__ seg_page,
+                               int64_t total_segs,
+                               const int64_t* __restrict__ seg_entry,
+                               const int64_t* __restrict__ seg_nd,
+                               const int64_t* __restrict__ seg_di,
+                               const int64_t* __restrict__ page_ulen,
+                               SnapDesc* __restrict__ descs,
+                               const int64_t* __restrict__ desc_base,
+                               int64_t* __restrict__ out_lens) {
+  for (int64_t gs = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       gs < total_segs; gs += (int64_t)gridDim.x * blockDim.x) {
+    if (seg_entry[gs] < 0) continue;
+    const SnapBigPage& bp = bigs[seg_page[gs]];
+    const int64_t k = gs - bp.seg_base;
+    const SnappyPage& pg = pages[bp.page_idx];
+    const uint8_t* s = pg.src;
+    const int64_t src_len = pg.src_len;
+    const int64_t ulen = page_ulen[seg_page[gs]];
+    const int64_t seg_end =
+        (k + 1) * SNAPBIG_SEG < src_len ? (k + 1) * SNAPBIG_SEG : src_len;
+    SnapDesc* d = descs + desc_base[bp.page_idx];
+    int64_t cur = seg_entry[gs];
+    int64_t nd = seg_nd[gs], di = seg_di[gs];
+    while (cur < seg_end && di < ulen) {
+      int64_t consumed, out;
+      uint32_t aux, kind;
+      if (!snap_decode_at(s, src_len, cur, &consumed, &out, &aux, &kind)) {
+        atomicExch((unsigned long long*)&out_lens[bp.page_idx],
+                   (unsigned long long)(-1LL));
+        return;
+      }
+      if (di + out > ulen ||
+          (kind == 1 && (aux == 0 || (int64_t)aux > di))) {
+        atomicExch((unsigned long long*)&out_lens[bp.page_idx],
+                   (unsigned long long)(-1LL));
+        return;
+      }
+      d[nd].dst = (uint32_t)di;
+      d[nd].aux = aux;
+      d[nd].len = (uint32_t)out;
+      d[nd].kind = kind;
+      ++nd;
+      di += out;
+      cur += consumed;
+    }
+  }
+}
+
 /* Decompress npages independent Snappy blocks.  pages: HOST array copied
  * internally; each entry's src/dst are DEVICE pointers.  out_lens (host,
  * npages): decompressed length or -1 on malformed input. */
@@ -3253,9 +3528,158 @@ extern "C" int bg_snappy_decompress(const void* h_pages, int64_t npages,
   const int wpb1 = BG_BLOCK / BG_WAVE;  // parse waves per block
   int blocks1 = (int)bg_imin64((npages + wpb1 - 1) / wpb1, BG_MAX_BLOCKS);
   if (blocks1 == 0) blocks1 = 1;
-  hipLaunchKernelGGL(k_snap_parse, dim3(blocks1), dim3(BG_BLOCK), 0, 0,
-                     d_pages, npages, d_descs, d_base, d_counts, d_lens);
-  HIP_TRY(hipGetLastError());
+  // giant pages (dict-fallback ~1 MB pages) go through the segmented
+  // self-synchronizing parse; the wave chase handles the rest.  The
+  // chase kernel still runs over ALL pages but exits immediately for big
+  // ones (counts/lens are overwritten by phase B/C afterwards) — to keep
+  // it simple we simply run the chase on the small pages only by
+  // swapping big pages out of its view via a filtered page list.
+  std::vector<SnapBigPage> bigs;
+  std::vector<int32_t> seg_page;
+  int64_t arr_total = 0, seg_total = 0;
+  for (int64_t p = 0; p < npages; ++p) {
+    if (hp[p].src_len <= SNAPBIG_THRESHOLD) continue;
+    SnapBigPage bp;
+    bp.page_idx = p;
+    bp.arr_base = arr_total;
+    bp.seg_base = seg_total;
+    bp.nsegs = (hp[p].src_len + SNAPBIG_SEG - 1) / SNAPBIG_SEG;
+    arr_total += hp[p].src_len;
+    for (int64_t k = 0; k < bp.nsegs; ++k)
+      seg_page.push_back((int32_t)bigs.size());
+    seg_total += bp.nsegs;
+    bigs.push_back(bp);
+  }
+  if (bigs.empty()) {
+    hipLaunchKernelGGL(k_snap_parse, dim3(blocks1), dim3(BG_BLOCK), 0, 0,
+                       d_pages, npages, d_descs, d_base, d_counts, d_lens);
+    HIP_TRY(hipGetLastError());
+  } else {
+    // small pages via the chase (on a compacted page list mapped back)
+    std::vector<SnappyPage> smalls;
+    std::vector<int64_t> small_map, small_base;
+    for (int64_t p = 0; p < npages; ++p) {
+      if (hp[p].src_len > SNAPBIG_THRESHOLD) continue;
+      smalls.push_back(hp[p]);
+      small_map.push_back(p);
+      small_base.push_back(base[p]);
+    }
+    if (!smalls.empty()) {
+      SnappyPage* d_small;
+      int64_t* d_small_base;
+      int64_t* d_small_counts;
+      int64_t* d_small_lens;
+      HIP_TRY(pool_malloc((void**)&d_small,
+                          sizeof(SnappyPage) * smalls.size()));
+      HIP_TRY(pool_malloc((void**)&d_small_base,
+                          sizeof(int64_t) * smalls.size()));
+      HIP_TRY(pool_malloc((void**)&d_small_counts,
+                          sizeof(int64_t) * smalls.size()));
+      HIP_TRY(pool_malloc((void**)&d_small_lens,
+                          sizeof(int64_t) * smalls.size()));
+      HIP_TRY(hipMemcpy(d_small, smalls.data(),
+                        sizeof(SnappyPage) * smalls.size(),
+                        hipMemcpyHostToDevice));
+      HIP_TRY(hipMemcpy(d_small_base, small_base.data(),
+                        sizeof(int64_t) * smalls.size(),
+                        hipMemcpyHostToDevice));
+      int sb = (int)bg_imin64(((int64_t)smalls.size() + wpb1 - 1) / wpb1,
+                              BG_MAX_BLOCKS);
+      if (sb == 0) sb = 1;
+      hipLaunchKernelGGL(k_snap_parse, dim3(sb), dim3(BG_BLOCK), 0, 0,
+                         d_small, (int64_t)smalls.size(), d_descs,
+                         d_small_base, d_small_counts, d_small_lens);
+      HIP_TRY(hipGetLastError());
+      // scatter counts/lens back to the full arrays
+      std::vector<int64_t> tmpc(smalls.size()), tmpl(smalls.size());
+      HIP_TRY(hipMemcpy(tmpc.data(), d_small_counts,
+                        sizeof(int64_t) * smalls.size(),
+                        hipMemcpyDeviceToHost));
+      HIP_TRY(hipMemcpy(tmpl.data(), d_small_lens,
+                        sizeof(int64_t) * smalls.size(),
+                        hipMemcpyDeviceToHost));
+      std::vector<int64_t> full_c(npages, 0), full_l(npages, 0);
+      for (size_t i = 0; i < smalls.size(); ++i) {
+        full_c[small_map[i]] = tmpc[i];
+        full_l[small_map[i]] = tmpl[i];
+      }
+      HIP_TRY(hipMemcpy(d_counts, full_c.data(), sizeof(int64_t) * npages,
+                        hipMemcpyHostToDevice));
+      HIP_TRY(hipMemcpy(d_lens, full_l.data(), sizeof(int64_t) * npages,
+                        hipMemcpyHostToDevice));
+      (void)pool_release(d_small);
+      (void)pool_release(d_small_base);
+      (void)pool_release(d_small_counts);
+      (void)pool_release(d_small_lens);
+    }
+    // big pages: A (spec) -> B (resolve) -> C (emit)
+    SnapBigPage* d_bigs;
+    int32_t* d_seg_page;
+    uint16_t* d_next16;
+    uint32_t* d_cnt_pre;
+    uint32_t* d_out_pre;
+    uint32_t* d_seg_ct;
+    uint32_t* d_seg_ot;
+    int64_t* d_seg_exit;
+    int64_t* d_seg_entry;
+    int64_t* d_seg_nd;
+    int64_t* d_seg_di;
+    int64_t* d_page_hdr;
+    int64_t* d_page_ulen;
+    HIP_TRY(pool_malloc((void**)&d_bigs, sizeof(SnapBigPage) * bigs.size()));
+    HIP_TRY(pool_malloc((void**)&d_seg_page, sizeof(int32_t) * seg_total));
+    HIP_TRY(pool_malloc((void**)&d_next16, sizeof(uint16_t) * arr_total));
+    HIP_TRY(pool_malloc((void**)&d_cnt_pre, sizeof(uint32_t) * arr_total));
+    HIP_TRY(pool_malloc((void**)&d_out_pre, sizeof(uint32_t) * arr_total));
+    HIP_TRY(pool_malloc((void**)&d_seg_ct, sizeof(uint32_t) * seg_total));
+    HIP_TRY(pool_malloc((void**)&d_seg_ot, sizeof(uint32_t) * seg_total));
+    HIP_TRY(pool_malloc((void**)&d_seg_exit, sizeof(int64_t) * seg_total));
+    HIP_TRY(pool_malloc((void**)&d_seg_entry, sizeof(int64_t) * seg_total));
+    HIP_TRY(pool_malloc((void**)&d_seg_nd, sizeof(int64_t) * seg_total));
+    HIP_TRY(pool_malloc((void**)&d_seg_di, sizeof(int64_t) * seg_total));
+    HIP_TRY(pool_malloc((void**)&d_page_hdr, sizeof(int64_t) * bigs.size()));
+    HIP_TRY(pool_malloc((void**)&d_page_ulen, sizeof(int64_t) * bigs.size()));
+    HIP_TRY(hipMemcpy(d_bigs, bigs.data(), sizeof(SnapBigPage) * bigs.size(),
+                      hipMemcpyHostToDevice));
+    HIP_TRY(hipMemcpy(d_seg_page, seg_page.data(),
+                      sizeof(int32_t) * seg_total, hipMemcpyHostToDevice));
+    HIP_TRY(hipMemset(d_next16, 0, sizeof(uint16_t) * arr_total));
+    int ab = (int)bg_imin64((seg_total + BG_BLOCK - 1) / BG_BLOCK,
+                            BG_MAX_BLOCKS);
+    if (ab == 0) ab = 1;
+    hipLaunchKernelGGL(k_snapbig_spec, dim3(ab), dim3(BG_BLOCK), 0, 0,
+                       d_pages, d_bigs, d_seg_page, seg_total, d_next16,
+                       d_cnt_pre, d_out_pre, d_seg_ct, d_seg_ot, d_seg_exit,
+                       d_page_hdr, d_page_ulen);
+    HIP_TRY(hipGetLastError());
+    int bb = (int)bg_imin64(((int64_t)bigs.size() + BG_BLOCK - 1) / BG_BLOCK,
+                            BG_MAX_BLOCKS);
+    if (bb == 0) bb = 1;
+    hipLaunchKernelGGL(k_snapbig_resolve, dim3(bb), dim3(BG_BLOCK), 0, 0,
+                       d_pages, d_bigs, (int64_t)bigs.size(), d_next16,
+                       d_cnt_pre, d_out_pre, d_seg_ct, d_seg_ot, d_seg_exit,
+                       d_page_hdr, d_page_ulen, d_seg_entry, d_seg_nd,
+                       d_seg_di, d_counts, d_lens);
+    HIP_TRY(hipGetLastError());
+    hipLaunchKernelGGL(k_snapbig_emit, dim3(ab), dim3(BG_BLOCK), 0, 0,
+                       d_pages, d_bigs, d_seg_page, seg_total, d_seg_entry,
+                       d_seg_nd, d_seg_di, d_page_ulen, d_descs, d_base,
+                       d_lens);
+    HIP_TRY(hipGetLastError());
+    (void)pool_release(d_bigs);
+    (void)pool_release(d_seg_page);
+    (void)pool_release(d_next16);
+    (void)pool_release(d_cnt_pre);
+    (void)pool_release(d_out_pre);
+    (void)pool_release(d_seg_ct);
+    (void)pool_release(d_seg_ot);
+    (void)pool_release(d_seg_exit);
+    (void)pool_release(d_seg_entry);
+    (void)pool_release(d_seg_nd);
+    (void)pool_release(d_seg_di);
+    (void)pool_release(d_page_hdr);
+    (void)pool_release(d_page_ulen);
+  }
   double t1 = tick();
   t_parse = t1 - t0;
   // pass 2: chunk grid.y at 65535 pages
