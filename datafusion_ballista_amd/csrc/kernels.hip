@@ -3300,6 +3300,7 @@ __global__ void k_snapbig_spec(const SnappyPage* __restrict__ pages,
                                const SnapBigPage* __restrict__ bigs,
                                const int32_t* __restrict__ seg_page,
                                int64_t total_segs,
+                               unsigned long long* dbg_ctr,
                                uint16_t* __restrict__ next16,
                                uint32_t* __restrict__ cnt_pre,
                                uint32_t* __restrict__ out_pre,
@@ -3347,8 +3348,15 @@ __global__ void k_snapbig_spec(const SnappyPage* __restrict__ pages,
       if (o >= seg_end) { exitv = o; break; }
       int64_t consumed, out;
       uint32_t aux, kind;
-      if (!snap_decode_at(s, src_len, o, &consumed, &out, &aux, &kind))
-        break;  // exitv stays -1 (chain dies inside the segment)
+      if (!snap_decode_at(s, src_len, o, &consumed, &out, &aux, &kind)) {
+        // the chase is speculative GARBAGE until it merges with the true
+        // chain (which never dies on a valid page): restart one byte on.
+        // Cumulative cnt/out prefixes stay consistent for the surviving
+        // final fragment, which is the only one the true entry can land
+        // on, so the absorb arithmetic is unaffected.
+        o += 1;
+        continue;
+      }
       next16[bp.arr_base + o] =
           consumed > 0xFFFE ? (uint16_t)0xFFFF : (uint16_t)consumed;
       cnt_pre[bp.arr_base + o] = cnt;
@@ -3360,13 +3368,17 @@ __global__ void k_snapbig_spec(const SnappyPage* __restrict__ pages,
     seg_cnt_tot[gs] = cnt;
     seg_out_tot[gs] = outsum;
     seg_exit[gs] = exitv;
+    if (dbg_ctr) {
+      atomicAdd(&dbg_ctr[2], (unsigned long long)cnt);
+      if (exitv < 0) atomicAdd(&dbg_ctr[3], 1ull);
+    }
   }
 }
 
 // phase B: one thread per page resolves segment entries/prefixes
 __global__ void k_snapbig_resolve(const SnappyPage* __restrict__ pages,
                                   const SnapBigPage* __restrict__ bigs,
-                                  int64_t nbig,
+                                  int64_t nbig, unsigned long long* dbg_ctr,
                                   const uint16_t* __restrict__ next16,
                                   const uint32_t* __restrict__ cnt_pre,
                                   const uint32_t* __restrict__ out_pre,
@@ -3425,8 +3437,10 @@ __global__ void k_snapbig_resolve(const SnappyPage* __restrict__ pages,
         di += out;
         cur += consumed;
         advanced = true;
+        if (dbg_ctr) atomicAdd(&dbg_ctr[1], 1ull);
       }
       (void)advanced;
+      if (dbg_ctr) atomicAdd(&dbg_ctr[0], 1ull);
       if (di > ulen) err = 1;
     }
     // a valid stream's last element ends exactly at src_len with
@@ -3647,25 +3661,47 @@ extern "C" int bg_snappy_decompress(const void* h_pages, int64_t npages,
     int ab = (int)bg_imin64((seg_total + BG_BLOCK - 1) / BG_BLOCK,
                             BG_MAX_BLOCKS);
     if (ab == 0) ab = 1;
+    unsigned long long* d_dbg = nullptr;
+    if (dbg) {
+      HIP_TRY(pool_malloc((void**)&d_dbg, 32));
+      HIP_TRY(hipMemset(d_dbg, 0, 32));
+    }
+
+    double tb0 = tick();
     hipLaunchKernelGGL(k_snapbig_spec, dim3(ab), dim3(BG_BLOCK), 0, 0,
-                       d_pages, d_bigs, d_seg_page, seg_total, d_next16,
+                       d_pages, d_bigs, d_seg_page, seg_total, d_dbg,
+                       d_next16,
                        d_cnt_pre, d_out_pre, d_seg_ct, d_seg_ot, d_seg_exit,
                        d_page_hdr, d_page_ulen);
     HIP_TRY(hipGetLastError());
+    double tb1 = tick();
     int bb = (int)bg_imin64(((int64_t)bigs.size() + BG_BLOCK - 1) / BG_BLOCK,
                             BG_MAX_BLOCKS);
     if (bb == 0) bb = 1;
     hipLaunchKernelGGL(k_snapbig_resolve, dim3(bb), dim3(BG_BLOCK), 0, 0,
-                       d_pages, d_bigs, (int64_t)bigs.size(), d_next16,
+                       d_pages, d_bigs, (int64_t)bigs.size(), d_dbg, d_next16,
                        d_cnt_pre, d_out_pre, d_seg_ct, d_seg_ot, d_seg_exit,
                        d_page_hdr, d_page_ulen, d_seg_entry, d_seg_nd,
                        d_seg_di, d_counts, d_lens);
     HIP_TRY(hipGetLastError());
+    double tb2 = tick();
     hipLaunchKernelGGL(k_snapbig_emit, dim3(ab), dim3(BG_BLOCK), 0, 0,
                        d_pages, d_bigs, d_seg_page, seg_total, d_seg_entry,
                        d_seg_nd, d_seg_di, d_page_ulen, d_descs, d_base,
                        d_lens);
     HIP_TRY(hipGetLastError());
+    if (dbg) {
+      double tb3 = tick();
+      unsigned long long hc[4] = {0, 0, 0, 0};
+      (void)hipMemcpy(hc, d_dbg, 32, hipMemcpyDeviceToHost);
+      (void)pool_release(d_dbg);
+      fprintf(stderr,
+              "[bg_snappy]   big: nseg=%lld spec=%.3fms resolve=%.3fms "
+              "emit=%.3fms seg_steps=%llu catchup=%llu spec_elems=%llu "
+              "spec_died=%llu\n",
+              (long long)seg_total, tb1 - tb0, tb2 - tb1, tb3 - tb2,
+              hc[0], hc[1], hc[2], hc[3]);
+    }
     (void)pool_release(d_bigs);
     (void)pool_release(d_seg_page);
     (void)pool_release(d_next16);
