@@ -3553,6 +3553,12 @@ extern "C" int bg_snappy_decompress(const void* h_pages, int64_t npages,
   int64_t arr_total = 0, seg_total = 0;
   for (int64_t p = 0; p < npages; ++p) {
     if (hp[p].src_len <= SNAPBIG_THRESHOLD) continue;
+    // near-incompressible pages are literal-dominated: few elements, the
+    // wave chase handles them at full rate — the segmented machinery
+    // only pays off on element-dense (match-heavy) pages
+    if (hp[p].dst_cap > 0 &&
+        (double)hp[p].src_len > 0.85 * (double)hp[p].dst_cap)
+      continue;
     SnapBigPage bp;
     bp.page_idx = p;
     bp.arr_base = arr_total;
