@@ -6370,3 +6370,117 @@ extern "C" int bg_bitmap_and(const uint8_t* d_a, const uint8_t* d_b,
   HIP_TRY(hipGetLastError());
   return BG_OK;
 }
+
+// ---------------------------------------------------------------------------
+// LIKE predicates over Utf8 columns (FilterExec breadth for the
+// q9/q13/q14/q16 shapes: '%green%', 'PROMO%', '%BRASS',
+// '%special%requests%').  Pattern restated as up to 4 literal TERMS that
+// must appear in order, with anchors: prefix (pattern does not start with
+// %) pins term 0 at position 0; suffix (does not end with %) pins the
+// last term at the end.  This covers every LIKE in the reference's
+// approved/q*.txt plans (no '_' wildcards there).  Output mask ANDs into
+// an Arrow bitmap like bg_eval_predicates; NULL input => false.
+// ---------------------------------------------------------------------------
+#define BG_LIKE_MAX_TERMS 4
+#define BG_LIKE_MAX_PAT 64
+
+struct BgLikeArgs {
+  const uint8_t* d_data;
+  const int32_t* d_offsets;
+  const uint8_t* d_validity;
+  uint8_t terms[BG_LIKE_MAX_TERMS][BG_LIKE_MAX_PAT];
+  int32_t term_len[BG_LIKE_MAX_TERMS];
+  int32_t nterms;
+  int32_t anchor_prefix, anchor_suffix;
+  int32_t negate;  // NOT LIKE
+};
+
+__device__ __forceinline__ bool bg_like_row(const BgLikeArgs& a,
+                                            const uint8_t* s, int32_t len) {
+  int32_t pos = 0;
+  for (int t = 0; t < a.nterms; ++t) {
+    const int32_t tl = a.term_len[t];
+    const bool first = t == 0, last = t == a.nterms - 1;
+    if (first && a.anchor_prefix) {
+      if (len < tl) return false;
+      for (int32_t j = 0; j < tl; ++j)
+        if (s[j] != a.terms[t][j]) return false;
+      pos = tl;
+      if (last && a.anchor_suffix) return len == tl;
+      continue;
+    }
+    if (last && a.anchor_suffix) {
+      if (len - pos < tl) return false;
+      for (int32_t j = 0; j < tl; ++j)
+        if (s[len - tl + j] != a.terms[t][j]) return false;
+      return true;
+    }
+    // find the term anywhere at/after pos
+    bool found = false;
+    for (int32_t i = pos; i + tl <= len; ++i) {
+      bool m = true;
+      for (int32_t j = 0; j < tl; ++j)
+        if (s[i + j] != a.terms[t][j]) { m = false; break; }
+      if (m) { pos = i + tl; found = true; break; }
+    }
+    if (!found) return false;
+  }
+  return true;
+}
+
+__global__ void k_eval_like(BgLikeArgs a, int64_t n, u64* mask_words) {
+  const int64_t wave_global =
+      ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) / BG_WAVE;
+  const int64_t nwaves = ((int64_t)gridDim.x * blockDim.x) / BG_WAVE;
+  const int lane = lane_id();
+  const int64_t nwords = (n + 63) / 64;
+  for (int64_t w = wave_global; w < nwords; w += nwaves) {
+    const int64_t i = w * BG_WAVE + lane;
+    bool hit = false;
+    if (i < n && bit_valid(a.d_validity, i)) {
+      const int32_t lo = a.d_offsets[i], hi = a.d_offsets[i + 1];
+      hit = bg_like_row(a, a.d_data + lo, hi - lo);
+      if (a.negate) hit = !hit;
+    }
+    const u64 m = __ballot(hit);
+    if (lane == 0) mask_words[w] &= m;  // AND-fold like bg_eval_predicates
+  }
+}
+
+/* AND a LIKE predicate over a Utf8 column into an EXISTING Arrow LSB
+ * bitmask (initialise the mask to all-ones or with bg_eval_predicates
+ * first).  terms/lens describe the in-order literal fragments of the
+ * pattern; anchors pin the first fragment at the start (no leading %)
+ * and/or the last at the end (no trailing %).  negate = NOT LIKE. */
+extern "C" int bg_eval_like(const bg_column* col, const char* const* terms,
+                            const int32_t* term_lens, int32_t nterms,
+                            int32_t anchor_prefix, int32_t anchor_suffix,
+                            int32_t negate, int64_t n, uint8_t* d_mask) {
+  REQUIRE_INIT();
+  if (col->dtype != BG_DT_UTF8 || !col->d_offsets)
+    return set_err(BG_ERR_INVALID, "bg_eval_like: Utf8 column required");
+  if (nterms < 1 || nterms > BG_LIKE_MAX_TERMS)
+    return set_err(BG_ERR_INVALID, "bg_eval_like: 1..4 pattern terms");
+  BgLikeArgs a{};
+  a.d_data = (const uint8_t*)col->d_data;
+  a.d_offsets = col->d_offsets;
+  a.d_validity = col->d_validity;
+  a.nterms = nterms;
+  a.anchor_prefix = anchor_prefix;
+  a.anchor_suffix = anchor_suffix;
+  a.negate = negate;
+  for (int t = 0; t < nterms; ++t) {
+    if (term_lens[t] <= 0 || term_lens[t] > BG_LIKE_MAX_PAT)
+      return set_err(BG_ERR_INVALID, "bg_eval_like: term length 1..64");
+    a.term_len[t] = term_lens[t];
+    memcpy(a.terms[t], terms[t], (size_t)term_lens[t]);
+  }
+  const int64_t nwords = (n + 63) / 64;
+  int blocks = (int)bg_imin64((nwords * BG_WAVE + BG_BLOCK - 1) / BG_BLOCK,
+                              BG_MAX_BLOCKS);
+  if (blocks == 0) blocks = 1;
+  hipLaunchKernelGGL(k_eval_like, dim3(blocks), dim3(BG_BLOCK), 0, 0, a, n,
+                     (u64*)d_mask);
+  HIP_TRY(hipGetLastError());
+  return BG_OK;
+}

@@ -1088,7 +1088,12 @@ static DBufPtr eval_filter_mask(const Table& t, const Value& node) {
   std::vector<bg_column> cols;
   for (auto& c : t.cols) cols.push_back(to_bg(c, t.n));
   std::vector<bg_pred> bp;
+  std::vector<const Value*> likes;
   for (auto& p : preds) {
+    if (p->has("like")) {
+      likes.push_back(p.get());
+      continue;
+    }
     bg_pred q{};
     q.column = t.idx(p->get_str("col"));
     q.op = cmp_code(p->get_str("cmp"));
@@ -1097,9 +1102,47 @@ static DBufPtr eval_filter_mask(const Table& t, const Value& node) {
     bp.push_back(q);
   }
   DBufPtr mask = dalloc((uint64_t)((t.n + 63) / 64) * 8 + 8);
-  chk(bg_eval_predicates(cols.data(), (int32_t)cols.size(), bp.data(),
-                         (int32_t)bp.size(), t.n, mask->u8()),
-      "bg_eval_predicates");
+  if (!bp.empty()) {
+    chk(bg_eval_predicates(cols.data(), (int32_t)cols.size(), bp.data(),
+                           (int32_t)bp.size(), t.n, mask->u8()),
+        "bg_eval_predicates");
+  } else {
+    chk(bg_memset(mask->p, 0xFF, mask->bytes), "bg_memset");
+  }
+  for (auto* lp : likes) {
+    // split the LIKE pattern into in-order literal fragments + anchors
+    const std::string& pat = lp->get_str("like");
+    const bool anchor_prefix = !pat.empty() && pat.front() != '%';
+    const bool anchor_suffix = !pat.empty() && pat.back() != '%';
+    std::vector<std::string> frags;
+    std::string cur;
+    for (char ch : pat) {
+      if (ch == '%') {
+        if (!cur.empty()) frags.push_back(cur);
+        cur.clear();
+      } else if (ch == '_') {
+        throw StageError(BG_ERR_UNSUPPORTED,
+                         "LIKE '_' wildcard not supported");
+      } else {
+        cur += ch;
+      }
+    }
+    if (!cur.empty()) frags.push_back(cur);
+    if (frags.empty())
+      throw StageError(BG_ERR_INVALID, "LIKE pattern has no literal text");
+    std::vector<const char*> terms;
+    std::vector<int32_t> lens;
+    for (auto& f : frags) {
+      terms.push_back(f.c_str());
+      lens.push_back((int32_t)f.size());
+    }
+    bg_column c = to_bg(t.cols[(size_t)t.idx(lp->get_str("col"))], t.n);
+    chk(bg_eval_like(&c, terms.data(), lens.data(), (int32_t)terms.size(),
+                     anchor_prefix ? 1 : 0, anchor_suffix ? 1 : 0,
+                     lp->get_bool_or("negate", false) ? 1 : 0, t.n,
+                     mask->u8()),
+        "bg_eval_like");
+  }
   return mask;
 }
 
@@ -2422,6 +2465,7 @@ static VSchema validate_plan(const Value& node) {
     VSchema s = validate_plan(node.at("input"));
     for (auto& p : node.get_arr("predicates")) {
       s.idx(p->get_str("col"));
+      if (p->has("like")) continue;
       cmp_code(p->get_str("cmp"));
     }
     return s;

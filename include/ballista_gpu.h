@@ -156,6 +156,15 @@ int bg_sub_i32(const void* d_src, int64_t n, int32_t sub, void* d_out);
 int bg_gather_bits(const uint8_t* d_valid, const uint32_t* d_idx, int64_t m,
                    uint8_t* d_out_bits);
 
+/* LIKE / NOT LIKE over a Utf8 column, ANDed into an existing Arrow LSB
+ * bitmask (q9/q13/q14/q16-class predicates: in-order literal fragments
+ * with optional start/end anchors — the forms the reference's
+ * approved plans use; no '_' wildcards there). */
+int bg_eval_like(const bg_column* col, const char* const* terms,
+                 const int32_t* term_lens, int32_t nterms,
+                 int32_t anchor_prefix, int32_t anchor_suffix,
+                 int32_t negate, int64_t n, uint8_t* d_mask);
+
 /* ---- hash repartition (SortShuffleWriterExec device half) ---- */
 
 /* create_hashes restatement over the key columns (bg_ahash.h; parity

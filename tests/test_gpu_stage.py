@@ -920,6 +920,50 @@ def test_parquet_scan_in_stage(ctx, tmp_path):
                                   .scaleb(2))
 
 
+def test_like_predicates(ctx, reg):
+    """LIKE / NOT LIKE filter predicates (the q9/q13/q14/q16 pattern
+    shapes): prefix, suffix, contains, multi-fragment, and negation —
+    each checked against a Python regex restatement, incl. NULLs
+    (NULL LIKE ... => excluded; NULL NOT LIKE ... => excluded too)."""
+    import re
+    rng = np.random.default_rng(61)
+    n = 40_000
+    words = ["PROMO BRUSHED BRASS", "STANDARD green shiny", "promo brass",
+             "special packages requests", "special requests",
+             "ECONOMY green BRASS", "MEDIUM POLISHED TIN", "",
+             "the specials make requests later", "PROMOTION green"]
+    strs = [words[i % len(words)] + (f" #{i%97}" if i % 3 else "")
+            for i in range(n)]
+    mask = rng.random(n) < 0.07
+    t = pa.table({"s": pa.array([None if m else v
+                                 for v, m in zip(strs, mask)]),
+                  "pid": pa.array(np.arange(n, dtype=np.int64))})
+    reg("lk", t)
+
+    cases = [("PROMO%", False, r"\APROMO.*\Z"),
+             ("%BRASS", False, r"\A.*BRASS\Z"),
+             ("%green%", False, r"\A.*green.*\Z"),
+             ("%special%requests%", False, r"\A.*special.*requests.*\Z"),
+             ("%green%", True, None)]
+    for pat, neg, rx in cases:
+        pred = {"col": "s", "like": pat}
+        if neg:
+            pred["negate"] = True
+        res = stage.execute(_doc({"op": "collect", "input": {
+            "op": "filter", "predicates": [pred],
+            "input": scan_of(t, "lk")}}))
+        if rx is None:
+            rx_pos = re.compile(r"\A.*green.*\Z", re.S)
+            want = [i for i, (v, m) in enumerate(zip(strs, mask))
+                    if not m and not rx_pos.match(v)]
+        else:
+            creg = re.compile(rx, re.S)
+            want = [i for i, (v, m) in enumerate(zip(strs, mask))
+                    if not m and creg.match(v)]
+        got = sorted(r[1] for r in res["rows"])
+        assert got == want, (pat, neg, len(got), len(want))
+
+
 def test_stage_errors_fail_loudly(ctx):
     with pytest.raises(RuntimeError, match="unregistered device table"):
         stage.execute(_doc({"op": "collect", "input": {
