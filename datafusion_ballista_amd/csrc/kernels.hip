@@ -6484,3 +6484,61 @@ extern "C" int bg_eval_like(const bg_column* col, const char* const* terms,
   HIP_TRY(hipGetLastError());
   return BG_OK;
 }
+
+// CASE WHEN <pred> THEN a ELSE b (ProjectionExec breadth: q8/q12/q14-class
+// conditional expressions): elementwise select through an Arrow bitmask.
+__global__ void k_select(const u64* __restrict__ mask,
+                         const uint8_t* __restrict__ a,
+                         const uint8_t* __restrict__ b, int64_t esz,
+                         int64_t n, uint8_t* __restrict__ out) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    const bool sel = (mask[i >> 6] >> (i & 63)) & 1;
+    const uint8_t* src = (sel ? a : b) + i * esz;
+    uint8_t* dst = out + i * esz;
+    for (int64_t j = 0; j < esz; ++j) dst[j] = src[j];
+  }
+}
+
+extern "C" int bg_select(const uint8_t* d_mask, const void* d_a,
+                         const void* d_b, int64_t esz, int64_t n,
+                         void* d_out) {
+  REQUIRE_INIT();
+  if (n <= 0) return BG_OK;
+  int blocks = (int)bg_imin64((n + BG_BLOCK - 1) / BG_BLOCK, BG_MAX_BLOCKS);
+  hipLaunchKernelGGL(k_select, dim3(blocks), dim3(BG_BLOCK), 0, 0,
+                     (const u64*)d_mask, (const uint8_t*)d_a,
+                     (const uint8_t*)d_b, esz, n, (uint8_t*)d_out);
+  HIP_TRY(hipGetLastError());
+  return BG_OK;
+}
+
+// constant fill (literal branches of CASE, constant columns)
+__global__ void k_fill_const(uint8_t* __restrict__ out, int64_t n,
+                             int64_t esz, int64_t lo, int64_t hi) {
+  const uint8_t v[16] = {
+      (uint8_t)(lo), (uint8_t)(lo >> 8), (uint8_t)(lo >> 16),
+      (uint8_t)(lo >> 24), (uint8_t)(lo >> 32), (uint8_t)(lo >> 40),
+      (uint8_t)(lo >> 48), (uint8_t)(lo >> 56), (uint8_t)(hi),
+      (uint8_t)(hi >> 8), (uint8_t)(hi >> 16), (uint8_t)(hi >> 24),
+      (uint8_t)(hi >> 32), (uint8_t)(hi >> 40), (uint8_t)(hi >> 48),
+      (uint8_t)(hi >> 56)};
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    uint8_t* dst = out + i * esz;
+    for (int64_t j = 0; j < esz; ++j) dst[j] = v[j];
+  }
+}
+
+extern "C" int bg_fill_const(void* d_out, int64_t n, int64_t esz,
+                             int64_t lo, int64_t hi) {
+  REQUIRE_INIT();
+  if (n <= 0) return BG_OK;
+  if (esz <= 0 || esz > 16)
+    return set_err(BG_ERR_INVALID, "bg_fill_const: esz 1..16");
+  int blocks = (int)bg_imin64((n + BG_BLOCK - 1) / BG_BLOCK, BG_MAX_BLOCKS);
+  hipLaunchKernelGGL(k_fill_const, dim3(blocks), dim3(BG_BLOCK), 0, 0,
+                     (uint8_t*)d_out, n, esz, lo, hi);
+  HIP_TRY(hipGetLastError());
+  return BG_OK;
+}

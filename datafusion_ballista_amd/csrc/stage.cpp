@@ -951,6 +951,8 @@ static Table scan(const Value& node) {
 // expressions
 // ---------------------------------------------------------------------------
 
+static DBufPtr eval_filter_mask(const Table& t, const Value& node);
+
 struct ExprVal {
   bool is_lit = false;
   Col col;          // when !is_lit (shares buffers)
@@ -1063,6 +1065,46 @@ static ExprVal eval_expr(const Table& t, const Value& e) {
     }
     v.col = eval_dec_binary(t, 2, -1, 3, *ops[0], *ops[1]);
     return v;
+  }
+  // CASE WHEN <predicates> THEN <expr> ELSE <expr>  (q8/q12/q14-class):
+  // {"case": {"when": [pred...], "then": expr, "else": expr}}
+  if (e.has("case")) {
+    const Value& c = e.at("case");
+    // when-predicates evaluate over the input table (same grammar as
+    // filter, incl. LIKE)
+    bgjson::Value fake;
+    fake.kind = Value::OBJ;
+    fake.obj.emplace_back("predicates", *c.find("when"));
+    DBufPtr mask = eval_filter_mask(t, fake);
+    ExprVal a = eval_expr(t, c.at("then"));
+    ExprVal b = eval_expr(t, c.at("else"));
+    // materialise literal branches as constant columns
+    auto materialise = [&](ExprVal& ev, const ExprVal& other) -> Col {
+      if (!ev.is_lit) return ev.col;
+      Col out;
+      out.dtype = other.is_lit ? BG_DT_DECIMAL128 : other.col.dtype;
+      out.precision = other.is_lit ? 38 : other.col.precision;
+      out.scale = other.is_lit ? 0 : other.col.scale;
+      int64_t esz = dt_size(out.dtype);
+      out.data = dalloc((uint64_t)(t.n > 0 ? t.n : 1) * (uint64_t)esz);
+      chk(bg_fill_const(out.data->p, t.n, esz, ev.lit_lo, ev.lit_hi),
+          "bg_fill_const");
+      return out;
+    };
+    Col ca = materialise(a, b);
+    Col cb = materialise(b, a);
+    if (ca.dtype != cb.dtype)
+      throw StageError(BG_ERR_INVALID, "case: branch dtype mismatch");
+    ExprVal v2;
+    v2.col.dtype = ca.dtype;
+    v2.col.precision = ca.precision ? ca.precision : cb.precision;
+    v2.col.scale = ca.scale ? ca.scale : cb.scale;
+    int64_t esz = dt_size(ca.dtype);
+    v2.col.data = dalloc((uint64_t)(t.n > 0 ? t.n : 1) * (uint64_t)esz);
+    chk(bg_select(mask->u8(), ca.dptr(), cb.dptr(), esz, t.n,
+                  v2.col.data->p),
+        "bg_select");
+    return v2;
   }
   throw StageError(BG_ERR_INVALID, "plan: unknown expression node");
 }
@@ -2573,6 +2615,16 @@ static VSchema validate_plan(const Value& node) {
 static DtSpec validate_expr(const VSchema& s, const Value& e) {
   if (e.has("col")) return s.dts[(size_t)s.idx(e.get_str("col"))];
   if (e.has("lit")) return {BG_DT_DECIMAL128, 38, 0};
+  if (e.has("case")) {
+    const Value& c = e.at("case");
+    for (auto& p : c.get_arr("when")) {
+      s.idx(p->get_str("col"));
+      if (!p->has("like")) cmp_code(p->get_str("cmp"));
+    }
+    DtSpec a = validate_expr(s, c.at("then"));
+    DtSpec b = validate_expr(s, c.at("else"));
+    return a.dt == BG_DT_DECIMAL128 || b.dt != BG_DT_DECIMAL128 ? a : b;
+  }
   for (const char* k : {"mul", "add", "sub"}) {
     if (e.has(k)) {
       auto& ops = e.get_arr(k);

@@ -165,6 +165,13 @@ int bg_eval_like(const bg_column* col, const char* const* terms,
                  int32_t anchor_prefix, int32_t anchor_suffix,
                  int32_t negate, int64_t n, uint8_t* d_mask);
 
+/* CASE WHEN mask THEN a ELSE b: elementwise select (esz 1..16). */
+int bg_select(const uint8_t* d_mask, const void* d_a, const void* d_b,
+              int64_t esz, int64_t n, void* d_out);
+/* constant fill (literal CASE branches): value as i128 (lo, hi). */
+int bg_fill_const(void* d_out, int64_t n, int64_t esz, int64_t lo,
+                  int64_t hi);
+
 /* ---- hash repartition (SortShuffleWriterExec device half) ---- */
 
 /* create_hashes restatement over the key columns (bg_ahash.h; parity

@@ -964,6 +964,51 @@ def test_like_predicates(ctx, reg):
         assert got == want, (pat, neg, len(got), len(want))
 
 
+def test_case_expression_q14_shape(ctx, reg):
+    """q14's CASE WHEN p_type LIKE 'PROMO%' THEN revenue ELSE 0 shape:
+    conditional projection feeding SUMs, exact vs a python restatement."""
+    rng = np.random.default_rng(67)
+    n = 50_000
+    types = ["PROMO BRUSHED", "STANDARD TIN", "PROMO POLISHED", "ECONOMY",
+             "MEDIUM PLATED"]
+    tcol = [types[i % len(types)] for i in range(n)]
+    price = rng.integers(90000, 10495100, size=n)
+    disc = rng.integers(0, 11, size=n)
+    t = pa.table({
+        "p_type": pa.array(tcol),
+        "l_extendedprice": pa.array(
+            [decimal.Decimal(int(v)) / 100 for v in price],
+            type=pa.decimal128(15, 2)),
+        "l_discount": pa.array(
+            [decimal.Decimal(int(v)) / 100 for v in disc],
+            type=pa.decimal128(15, 2)),
+    })
+    reg("q14", t)
+    rev = {"mul": [{"col": "l_extendedprice"},
+                   {"sub": [{"lit": 100}, {"col": "l_discount"}]}]}
+    plan = {"op": "collect", "input": {
+        "op": "hash_aggregate", "mode": "single", "group_by": [],
+        "aggs": [
+            {"fn": "sum", "as": "promo",
+             "expr": {"case": {"when": [{"col": "p_type",
+                                         "like": "PROMO%"}],
+                               "then": rev, "else": {"lit": 0}}}},
+            {"fn": "sum", "as": "total", "expr": rev}],
+        "input": {"op": "project", "exprs": [
+            {"as": "p_type", "expr": {"col": "p_type"}},
+            {"as": "l_extendedprice", "expr": {"col": "l_extendedprice"}},
+            {"as": "l_discount", "expr": {"col": "l_discount"}}],
+            "input": scan_of(t, "q14")}}}
+    res = stage.execute(_doc(plan))
+    want_promo = sum(int(p) * (100 - int(d))
+                     for ty, p, d in zip(tcol, price, disc)
+                     if ty.startswith("PROMO"))
+    want_total = sum(int(p) * (100 - int(d))
+                     for p, d in zip(price, disc))
+    assert int(res["rows"][0][0]) == want_promo
+    assert int(res["rows"][0][1]) == want_total
+
+
 def test_stage_errors_fail_loudly(ctx):
     with pytest.raises(RuntimeError, match="unregistered device table"):
         stage.execute(_doc({"op": "collect", "input": {
