@@ -3548,17 +3548,21 @@ extern "C" int bg_snappy_decompress(const void* h_pages, int64_t npages,
   // ones (counts/lens are overwritten by phase B/C afterwards) — to keep
   // it simple we simply run the chase on the small pages only by
   // swapping big pages out of its view via a filtered page list.
+  // route to the segmented path: big AND element-dense.  Near-
+  // incompressible pages are literal-dominated (few elements) and the
+  // wave chase handles them at full rate.
+  auto is_big = [&](int64_t p) {
+    if (hp[p].src_len <= SNAPBIG_THRESHOLD) return false;
+    if (hp[p].dst_cap > 0 &&
+        (double)hp[p].src_len > 0.85 * (double)hp[p].dst_cap)
+      return false;
+    return true;
+  };
   std::vector<SnapBigPage> bigs;
   std::vector<int32_t> seg_page;
   int64_t arr_total = 0, seg_total = 0;
   for (int64_t p = 0; p < npages; ++p) {
-    if (hp[p].src_len <= SNAPBIG_THRESHOLD) continue;
-    // near-incompressible pages are literal-dominated: few elements, the
-    // wave chase handles them at full rate — the segmented machinery
-    // only pays off on element-dense (match-heavy) pages
-    if (hp[p].dst_cap > 0 &&
-        (double)hp[p].src_len > 0.85 * (double)hp[p].dst_cap)
-      continue;
+    if (!is_big(p)) continue;
     SnapBigPage bp;
     bp.page_idx = p;
     bp.arr_base = arr_total;
@@ -3579,7 +3583,7 @@ extern "C" int bg_snappy_decompress(const void* h_pages, int64_t npages,
     std::vector<SnappyPage> smalls;
     std::vector<int64_t> small_map, small_base;
     for (int64_t p = 0; p < npages; ++p) {
-      if (hp[p].src_len > SNAPBIG_THRESHOLD) continue;
+      if (is_big(p)) continue;
       smalls.push_back(hp[p]);
       small_map.push_back(p);
       small_base.push_back(base[p]);
