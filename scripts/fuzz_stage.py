@@ -155,7 +155,7 @@ def run_round(ctx, rng, trial, tmpdir):
     bt = pa.table({"bk": pa.array(np.unique(bkeys)),
                    "bv": pa.array(np.arange(len(np.unique(bkeys)),
                                             dtype=np.int64))})
-    stage.register_table(ctx, name + "b", bt)
+    keep2 = stage.register_table(ctx, name + "b", bt)
     jp = {"op": "hash_join",
           "build": {"op": "scan", "schema": stage.schema_json(bt.schema),
                     "source": {"kind": "device", "table": name + "b"}},
@@ -175,7 +175,7 @@ def run_round(ctx, rng, trial, tmpdir):
                                          want_n)
     stage.unregister_table(name)
     stage.unregister_table(name + "b")
-    del keep
+    del keep, keep2
 
 
 def main():
