@@ -58,7 +58,7 @@ def run_round(ctx, rng, trial, tmpdir):
     hi = lo + int(rng.integers(50, 2000))
     which = rng.integers(0, 4)
     mask = np.ones(n, dtype=bool)
-    d_np = t["d"].to_numpy()
+    d_np = t["d"].to_numpy().astype("datetime64[D]").astype(np.int64)
     if which == 0:
         preds = [{"col": "d", "cmp": "ge_lt", "lo": lo, "hi": hi}]
         mask = (d_np >= lo) & (d_np < hi)
