@@ -6546,3 +6546,81 @@ extern "C" int bg_fill_const(void* d_out, int64_t n, int64_t esz,
   HIP_TRY(hipGetLastError());
   return BG_OK;
 }
+
+// IN-list predicate (q12 l_shipmode IN (...), q19 p_container IN (...)):
+// AND-fold `x IN {v0..vk}` over an integer/dict column into an existing
+// Arrow bitmask.  k <= 16 values, compared as i64 (narrower dtypes
+// sign-extend; dict8 zero-extends).
+__global__ void k_eval_in(const uint8_t* __restrict__ data,
+                          const uint8_t* __restrict__ valid, int32_t dtype,
+                          const int64_t* __restrict__ vals, int32_t nvals,
+                          int64_t n, u64* __restrict__ mask_words) {
+  const int64_t wave_global =
+      ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) / BG_WAVE;
+  const int64_t nwaves = ((int64_t)gridDim.x * blockDim.x) / BG_WAVE;
+  const int lane = lane_id();
+  const int64_t nwords = (n + 63) / 64;
+  for (int64_t w = wave_global; w < nwords; w += nwaves) {
+    const int64_t i = w * BG_WAVE + lane;
+    bool hit = false;
+    if (i < n && bit_valid(valid, i)) {
+      int64_t x;
+      switch (dtype) {
+        case BG_DT_INT64: x = reinterpret_cast<const int64_t*>(data)[i]; break;
+        case BG_DT_INT32:
+        case BG_DT_DATE32: x = reinterpret_cast<const int32_t*>(data)[i]; break;
+        case BG_DT_DICT8: x = data[i]; break;
+        default: x = 0;
+      }
+      for (int v = 0; v < nvals; ++v)
+        if (vals[v] == x) { hit = true; break; }
+    }
+    const u64 m = __ballot(hit);
+    if (lane == 0) mask_words[w] &= m;
+  }
+}
+
+extern "C" int bg_eval_in(const bg_column* col, const int64_t* values,
+                          int32_t nvalues, int64_t n, uint8_t* d_mask) {
+  REQUIRE_INIT();
+  if (nvalues < 1 || nvalues > 16)
+    return set_err(BG_ERR_INVALID, "bg_eval_in: 1..16 values");
+  if (col->dtype != BG_DT_INT64 && col->dtype != BG_DT_INT32 &&
+      col->dtype != BG_DT_DATE32 && col->dtype != BG_DT_DICT8)
+    return set_err(BG_ERR_UNSUPPORTED, "bg_eval_in: integer/dict column");
+  int64_t* d_vals;
+  HIP_TRY(pool_malloc((void**)&d_vals, sizeof(int64_t) * 16));
+  HIP_TRY(hipMemcpy(d_vals, values, sizeof(int64_t) * nvalues,
+                    hipMemcpyHostToDevice));
+  const int64_t nwords = (n + 63) / 64;
+  int blocks = (int)bg_imin64((nwords * BG_WAVE + BG_BLOCK - 1) / BG_BLOCK,
+                              BG_MAX_BLOCKS);
+  if (blocks == 0) blocks = 1;
+  hipLaunchKernelGGL(k_eval_in, dim3(blocks), dim3(BG_BLOCK), 0, 0,
+                     (const uint8_t*)col->d_data, col->d_validity,
+                     col->dtype, d_vals, nvalues, n, (u64*)d_mask);
+  HIP_TRY(hipGetLastError());
+  (void)pool_release(d_vals);
+  return BG_OK;
+}
+
+__global__ void k_bitmap_or(const uint8_t* __restrict__ a,
+                            const uint8_t* __restrict__ b, int64_t nbytes,
+                            uint8_t* __restrict__ out) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < nbytes;
+       i += (int64_t)gridDim.x * blockDim.x)
+    out[i] = a[i] | b[i];
+}
+
+extern "C" int bg_bitmap_or(const uint8_t* d_a, const uint8_t* d_b,
+                            int64_t nbits, uint8_t* d_out) {
+  REQUIRE_INIT();
+  int64_t nbytes = (nbits + 7) / 8;
+  if (nbytes <= 0) return BG_OK;
+  int blocks =
+      (int)bg_imin64((nbytes + BG_BLOCK - 1) / BG_BLOCK, BG_MAX_BLOCKS);
+  hipLaunchKernelGGL(k_bitmap_or, dim3(blocks), dim3(BG_BLOCK), 0, 0, d_a,
+                     d_b, nbytes, d_out);
+  HIP_TRY(hipGetLastError());
+  return BG_OK;
+}

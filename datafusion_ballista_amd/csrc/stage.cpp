@@ -1131,9 +1131,14 @@ static DBufPtr eval_filter_mask(const Table& t, const Value& node) {
   for (auto& c : t.cols) cols.push_back(to_bg(c, t.n));
   std::vector<bg_pred> bp;
   std::vector<const Value*> likes;
+  std::vector<const Value*> ins;
   for (auto& p : preds) {
     if (p->has("like")) {
       likes.push_back(p.get());
+      continue;
+    }
+    if (p->has("in")) {
+      ins.push_back(p.get());
       continue;
     }
     bg_pred q{};
@@ -1150,6 +1155,17 @@ static DBufPtr eval_filter_mask(const Table& t, const Value& node) {
         "bg_eval_predicates");
   } else {
     chk(bg_memset(mask->p, 0xFF, mask->bytes), "bg_memset");
+  }
+  for (auto* ip : ins) {
+    std::vector<int64_t> vals;
+    for (auto& v : ip->get_arr("in")) {
+      int64_t lo, hi;
+      parse_i128(*v, &lo, &hi);
+      vals.push_back(lo);
+    }
+    bg_column c = to_bg(t.cols[(size_t)t.idx(ip->get_str("col"))], t.n);
+    chk(bg_eval_in(&c, vals.data(), (int32_t)vals.size(), t.n, mask->u8()),
+        "bg_eval_in");
   }
   for (auto* lp : likes) {
     // split the LIKE pattern into in-order literal fragments + anchors
@@ -1186,6 +1202,28 @@ static DBufPtr eval_filter_mask(const Table& t, const Value& node) {
         "bg_eval_like");
   }
   return mask;
+}
+
+// filter node: {"predicates": [AND...]} or {"any": [[AND...], ...]}
+// (disjunctive groups, the reference's q19-class OR-of-ANDs)
+static DBufPtr eval_filter_node(const Table& t, const Value& node) {
+  if (!node.has("any")) return eval_filter_mask(t, node);
+  auto& groups = node.get_arr("any");
+  DBufPtr acc;
+  for (auto& grp : groups) {
+    bgjson::Value fake;
+    fake.kind = Value::OBJ;
+    fake.obj.emplace_back("predicates", grp);
+    DBufPtr m2 = eval_filter_mask(t, fake);
+    if (!acc) {
+      acc = m2;
+    } else {
+      chk(bg_bitmap_or(acc->u8(), m2->u8(), t.n, acc->u8()),
+          "bg_bitmap_or");
+    }
+  }
+  if (!acc) throw StageError(BG_ERR_INVALID, "filter: empty 'any'");
+  return acc;
 }
 
 static Table filter_materialize(const Table& t, DBufPtr mask) {
@@ -1488,7 +1526,7 @@ static Table exec_aggregate(const Value& node, Metrics& m) {
   const Value& child = node.at("input");
   if (child.get_str("op") == "filter") {
     in = exec_plan(child.at("input"), m);
-    mask = eval_filter_mask(in, child);
+    mask = eval_filter_node(in, child);
   } else {
     in = exec_plan(child, m);
   }
@@ -1813,7 +1851,7 @@ static Table exec_plan(const Value& node, Metrics& m) {
   }
   if (op == "filter") {
     Table in = exec_plan(node.at("input"), m);
-    return filter_materialize(in, eval_filter_mask(in, node));
+    return filter_materialize(in, eval_filter_node(in, node));
   }
   if (op == "project") return exec_project(node, m);
   if (op == "hash_join") return exec_join(node, m);
@@ -2505,11 +2543,17 @@ static VSchema validate_plan(const Value& node) {
   }
   if (op == "filter") {
     VSchema s = validate_plan(node.at("input"));
-    for (auto& p : node.get_arr("predicates")) {
-      s.idx(p->get_str("col"));
-      if (p->has("like")) continue;
-      cmp_code(p->get_str("cmp"));
-    }
+    auto check_group = [&](const std::vector<ValuePtr>& preds) {
+      for (auto& p : preds) {
+        s.idx(p->get_str("col"));
+        if (p->has("like") || p->has("in")) continue;
+        cmp_code(p->get_str("cmp"));
+      }
+    };
+    if (node.has("any"))
+      for (auto& grp : node.get_arr("any")) check_group(grp->arr);
+    else
+      check_group(node.get_arr("predicates"));
     return s;
   }
   if (op == "project") {

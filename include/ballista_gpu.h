@@ -172,6 +172,14 @@ int bg_select(const uint8_t* d_mask, const void* d_a, const void* d_b,
 int bg_fill_const(void* d_out, int64_t n, int64_t esz, int64_t lo,
                   int64_t hi);
 
+/* x IN {v0..v15} over integer/dict columns, ANDed into an existing
+ * bitmask (q12/q19-class IN-lists). */
+int bg_eval_in(const bg_column* col, const int64_t* values, int32_t nvalues,
+               int64_t n, uint8_t* d_mask);
+/* out = a OR b (disjunctive predicate groups, q19's OR-of-ANDs). */
+int bg_bitmap_or(const uint8_t* d_a, const uint8_t* d_b, int64_t nbits,
+                 uint8_t* d_out);
+
 /* ---- hash repartition (SortShuffleWriterExec device half) ---- */
 
 /* create_hashes restatement over the key columns (bg_ahash.h; parity

@@ -1009,6 +1009,55 @@ def test_case_expression_q14_shape(ctx, reg):
     assert int(res["rows"][0][1]) == want_total
 
 
+def test_in_list_and_or_groups(ctx, reg):
+    """IN-list predicates (q12 l_shipmode IN (...)) and OR-of-AND groups
+    (q19's (p=a AND qty in r1) OR (p=b AND qty in r2) shape), exact vs
+    numpy."""
+    rng = np.random.default_rng(71)
+    n = 30_000
+    mode = rng.integers(0, 7, size=n).astype(np.uint8)
+    qty = rng.integers(1, 51, size=n, dtype=np.int64)
+    d = rng.integers(8000, 11000, size=n, dtype=np.int32)
+    t = pa.table({"mode": pa.array(mode), "qty": pa.array(qty),
+                  "d": pa.array(d, type=pa.date32()),
+                  "pid": pa.array(np.arange(n, dtype=np.int64))})
+    reg("inor", t)
+
+    # IN over dict8
+    res = stage.execute(_doc({"op": "collect", "input": {
+        "op": "filter", "predicates": [{"col": "mode", "in": [1, 4, 6]}],
+        "input": scan_of(t, "inor")}}))
+    want = sorted(np.nonzero(np.isin(mode, [1, 4, 6]))[0].tolist())
+    assert sorted(r[3] for r in res["rows"]) == want
+
+    # OR of AND-groups (q19 shape)
+    res = stage.execute(_doc({"op": "collect", "input": {
+        "op": "filter", "any": [
+            [{"col": "mode", "in": [1]},
+             {"col": "qty", "cmp": "between", "lo": 1, "hi": 11}],
+            [{"col": "mode", "in": [2]},
+             {"col": "qty", "cmp": "between", "lo": 10, "hi": 20}],
+            [{"col": "mode", "in": [3]},
+             {"col": "qty", "cmp": "between", "lo": 20, "hi": 30}]],
+        "input": scan_of(t, "inor")}}))
+    want_mask = (((mode == 1) & (qty >= 1) & (qty <= 11)) |
+                 ((mode == 2) & (qty >= 10) & (qty <= 20)) |
+                 ((mode == 3) & (qty >= 20) & (qty <= 30)))
+    assert sorted(r[3] for r in res["rows"]) == \
+        sorted(np.nonzero(want_mask)[0].tolist())
+
+    # OR groups fused into an aggregate (mask path)
+    res = stage.execute(_doc({"op": "collect", "input": {
+        "op": "hash_aggregate", "mode": "single", "group_by": [],
+        "aggs": [{"fn": "count", "as": "c"}],
+        "input": {"op": "filter", "any": [
+            [{"col": "d", "cmp": "lt", "hi": 9000}],
+            [{"col": "qty", "cmp": "gt", "lo": 45}]],
+            "input": scan_of(t, "inor")}}}))
+    want_c = int(((d < 9000) | (qty > 45)).sum())
+    assert res["rows"][0][0] == want_c
+
+
 def test_stage_errors_fail_loudly(ctx):
     with pytest.raises(RuntimeError, match="unregistered device table"):
         stage.execute(_doc({"op": "collect", "input": {
