@@ -270,6 +270,110 @@ def main():
                  "top10": r3["rows"]}
     print("q3:", json.dumps(out["q3"]), flush=True)
 
+    # ---------------- q14 (LIKE + CASE + part join at SF100) ----------
+    # part: 20M rows, p_type from the spec's 150-combo vocabulary; the
+    # first 25 combos are 'PROMO ...' so the cross-check knows membership
+    # by type code.  The utf8 column is built ON DEVICE by gathering the
+    # 150-string dictionary through the per-row codes.
+    import ctypes as ct
+    npart = SF * 200_000
+    t1s = ["PROMO", "STANDARD", "SMALL", "MEDIUM", "LARGE", "ECONOMY"]
+    t2s = ["ANODIZED", "BURNISHED", "PLATED", "POLISHED", "BRUSHED"]
+    t3s = ["TIN", "NICKEL", "BRASS", "STEEL", "COPPER"]
+    vocab = [f"{a} {b} {c}" for a in t1s for b in t2s for c in t3s]
+    npromo = len(t2s) * len(t3s)  # 'PROMO *' codes are 0..24
+    g2 = torch.Generator(device=dev)
+    g2.manual_seed(77)
+    pcodes = torch.randint(0, len(vocab), (npart,), generator=g2,
+                           device=dev, dtype=torch.int32)
+    dict_data = "".join(vocab).encode()
+    dict_offs = np.zeros(len(vocab) + 1, dtype=np.int32)
+    for i, w in enumerate(vocab):
+        dict_offs[i + 1] = dict_offs[i] + len(w)
+    ctx2 = gpu.GpuStageContext(0)
+    ddata = ctx2.upload(np.frombuffer(dict_data, dtype=np.uint8))
+    doffs = ctx2.upload(dict_offs)
+    ptype_offs = ctx2.alloc(4 * (npart + 1))
+    cap = npart * 32
+    ptype_data = ctx2.alloc(cap)
+    tot = ct.c_int64()
+    gpu._check(gpu.load_library().bg_gather_varlen(
+        ddata.ptr, doffs.ptr, ct.c_void_p(pcodes.data_ptr()),
+        ct.c_int64(npart), ptype_offs.ptr, ptype_data.ptr,
+        ct.c_int64(cap), ct.byref(tot)), "gather_varlen(p_type)")
+    p_partkey = torch.arange(1, npart + 1, device=dev, dtype=torch.int64)
+    L = gpu.load_library()
+    arr = (gpu.BgColumn * 2)(
+        gpu.BgColumn(gpu.BG_DT_INT64, 0, 0, 0,
+                     ct.c_void_p(p_partkey.data_ptr()), None, None, npart),
+        gpu.BgColumn(gpu.BG_DT_UTF8, 0, 0, 0, ptype_data.ptr, None,
+                     ptype_offs.ptr, npart))
+    names = (ct.c_char_p * 2)(b"p_partkey", b"p_type")
+    gpu._check(L.bg_stage_register_table(b"part", arr, names, 2,
+                                         ct.c_int64(npart)), "register")
+
+    rev = {"mul": [{"col": "l_extendedprice"},
+                   {"sub": [{"lit": 100}, {"col": "l_discount"}]}]}
+    q14 = {"op": "collect", "input": {
+        "op": "hash_aggregate", "mode": "single", "group_by": [],
+        "aggs": [
+            {"fn": "sum", "as": "promo",
+             "expr": {"case": {"when": [{"col": "p_type",
+                                         "like": "PROMO%"}],
+                               "then": rev, "else": {"lit": 0}}}},
+            {"fn": "sum", "as": "total", "expr": rev},
+            {"fn": "count", "as": "cnt"}],
+        "input": {"op": "hash_join",
+                  "build": {"op": "scan", "schema": [
+                      {"name": "p_partkey", "dtype": "int64"},
+                      {"name": "p_type", "dtype": "utf8"}],
+                      "source": {"kind": "device", "table": "part"}},
+                  "probe": {"op": "filter",
+                            "predicates": [{"col": "l_shipdate",
+                                            "cmp": "ge_lt", "lo": 9374,
+                                            "hi": 9404}],
+                            "input": {"op": "scan", "schema": li_schema + [
+                                {"name": "l_partkey", "dtype": "int64"}],
+                                "source": {"kind": "device",
+                                           "table": "lineitem2"},
+                                "projection": ["l_partkey", "l_shipdate",
+                                               "l_extendedprice",
+                                               "l_discount"]}},
+                  "build_keys": ["p_partkey"], "probe_keys": ["l_partkey"],
+                  "join_type": "inner",
+                  "output": [{"side": "build", "col": "p_type"},
+                             {"side": "probe", "col": "l_extendedprice"},
+                             {"side": "probe", "col": "l_discount"}]}}}
+    register("lineitem2", [
+        ("l_orderkey", gpu.BG_DT_INT64, li["l_orderkey"], 0, 0),
+        ("l_shipdate", gpu.BG_DT_DATE32, li["l_shipdate"], 0, 0),
+        ("l_quantity", gpu.BG_DT_DECIMAL128, li["l_quantity"], 15, 2),
+        ("l_extendedprice", gpu.BG_DT_DECIMAL128, li["l_extendedprice"],
+         15, 2),
+        ("l_discount", gpu.BG_DT_DECIMAL128, li["l_discount"], 15, 2),
+        ("l_tax", gpu.BG_DT_DECIMAL128, li["l_tax"], 15, 2),
+        ("l_returnflag", gpu.BG_DT_DICT8, li["l_returnflag"], 0, 0),
+        ("l_linestatus", gpu.BG_DT_DICT8, li["l_linestatus"], 0, 0),
+        ("l_partkey", gpu.BG_DT_INT64, li["l_partkey"], 0, 0),
+    ])
+    r14, cold, warm = run_doc(q14, 14)
+    # torch cross-check
+    lm = (li["l_shipdate"] >= 9374) & (li["l_shipdate"] < 9404)
+    pk = li["l_partkey"][lm]
+    rev_t = (li["l_extendedprice"][:, 0][lm] *
+             (100 - li["l_discount"][:, 0][lm]))
+    promo_m = pcodes.to(torch.int64)[pk - 1] < npromo
+    want_promo = int(rev_t[promo_m].sum().item())
+    want_total = int(rev_t.sum().item())
+    want_cnt = int(lm.sum().item())
+    assert int(r14["rows"][0][0]) == want_promo
+    assert int(r14["rows"][0][1]) == want_total
+    assert r14["rows"][0][2] == want_cnt
+    out["q14"] = {"wall_s_warm": warm, "wall_s_cold": cold,
+                  "joined_rows": want_cnt,
+                  "promo_frac": want_promo / max(want_total, 1)}
+    print("q14:", json.dumps(out["q14"]), flush=True)
+
     os.makedirs(os.path.join(ROOT, "gpurun_out"), exist_ok=True)
     with open(os.path.join(ROOT, "gpurun_out", "perf_dbgen_sf100.json"),
               "w") as f:
