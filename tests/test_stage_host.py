@@ -144,6 +144,48 @@ def test_validate_join_schema():
     assert [f["name"] for f in res["schema"]] == ["o_orderkey", "c_seg"]
 
 
+def test_validate_new_grammar():
+    """Round-2 grammar: LIKE / IN / OR-groups / CASE / parquet scan
+    validate host-side (no GPU)."""
+    sch = [{"name": "s", "dtype": "utf8"}, {"name": "m", "dtype": "dict8"},
+           {"name": "v", "dtype": "decimal128", "precision": 15,
+            "scale": 2}]
+    sc = {"op": "scan", "schema": sch,
+          "source": {"kind": "device", "table": "t"}}
+    res = stage.validate(_doc({"op": "collect", "input": {
+        "op": "filter", "predicates": [
+            {"col": "s", "like": "%green%"},
+            {"col": "m", "in": [1, 2, 3]}], "input": sc}}))
+    assert res["ok"] is True
+
+    res = stage.validate(_doc({"op": "collect", "input": {
+        "op": "filter", "any": [
+            [{"col": "m", "in": [1]}],
+            [{"col": "s", "like": "PROMO%"}]], "input": sc}}))
+    assert res["ok"] is True
+
+    res = stage.validate(_doc({"op": "collect", "input": {
+        "op": "project", "exprs": [{"as": "c", "expr": {"case": {
+            "when": [{"col": "s", "like": "PROMO%"}],
+            "then": {"col": "v"}, "else": {"lit": 0}}}}],
+        "input": sc}}))
+    assert res["ok"] is True
+    assert res["schema"][0]["dtype"] == "decimal128"
+
+    # parquet scan source kind accepted
+    res = stage.validate(_doc({"op": "collect", "input": {
+        "op": "scan", "schema": [{"name": "a", "dtype": "int64"}],
+        "source": {"kind": "parquet", "path": "/nonexistent",
+                   "columns": []}}}))
+    assert res["ok"] is True
+
+    # unknown column inside an OR group still errors
+    msg = stage.validate_error(_doc({"op": "collect", "input": {
+        "op": "filter", "any": [[{"col": "nope", "in": [1]}]],
+        "input": sc}}))
+    assert "nope" in msg
+
+
 def test_validate_bad_json():
     import ctypes
     import json
