@@ -2263,6 +2263,23 @@ __global__ void k_key_transform_i64(const int64_t* keys,
   }
 }
 
+// Float64 ORDER BY: IEEE-754 total-order transform (sign-magnitude
+// flip — negatives bit-inverted, positives sign-flipped; same monotone
+// map the MIN/MAX aggregates use)
+__global__ void k_key_transform_f64(const int64_t* keys,
+                                    const uint8_t* valid, int64_t n,
+                                    int descending, u64* keyu) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    if (!bit_valid(valid, i)) { keyu[i] = 0; continue; }
+    const u64 bits = (u64)keys[i];
+    const u64 u = (bits & 0x8000000000000000ull)
+                      ? ~bits
+                      : bits ^ 0x8000000000000000ull;
+    keyu[i] = descending ? ~u : u;
+  }
+}
+
 __global__ void k_key_transform_i32(const int32_t* keys,
                                     const uint8_t* valid, int64_t n,
                                     int descending, u64* keyu) {
@@ -2597,6 +2614,13 @@ static int sort_rows_impl(const bg_column* key_cols,
                            descending[c], d_keyu[0]);
         npasses = 8; nwords = 1;
         break;
+      case BG_DT_FLOAT64:
+        hipLaunchKernelGGL(k_key_transform_f64, dim3(blocks), dim3(BG_BLOCK),
+                           0, 0, (const int64_t*)key_cols[c].d_data,
+                           key_cols[c].d_validity, n,
+                           descending[c], d_keyu[0]);
+        npasses = 8; nwords = 1;
+        break;
       case BG_DT_INT32:
       case BG_DT_DATE32:
         hipLaunchKernelGGL(k_key_transform_i32, dim3(blocks), dim3(BG_BLOCK),
@@ -2654,7 +2678,7 @@ static int sort_rows_impl(const bg_column* key_cols,
       }
       default:
         rc = set_err(BG_ERR_UNSUPPORTED,
-                     "sort keys: INT64/INT32/DATE32/DECIMAL128/UTF8");
+                     "sort keys: INT64/INT32/DATE32/DECIMAL128/UTF8/FLOAT64");
         continue;
     }
     // bring the key stream into the CURRENT permutation order once

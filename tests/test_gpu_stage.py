@@ -1058,6 +1058,30 @@ def test_in_list_and_or_groups(ctx, reg):
     assert res["rows"][0][0] == want_c
 
 
+def test_sort_float64_keys(ctx, reg):
+    """ORDER BY over Float64 (IEEE total order incl. negatives/infs), asc
+    and desc, with NULLs defaulting to the SQL convention."""
+    rng = np.random.default_rng(73)
+    n = 20_000
+    x = rng.standard_normal(n) * 1e6
+    x[0] = float("inf")
+    x[1] = float("-inf")
+    mask = rng.random(n) < 0.05
+    t = pa.table({"x": pa.array(x, mask=mask),
+                  "pid": pa.array(np.arange(n, dtype=np.int64))})
+    reg("f64s", t)
+    for desc in (False, True):
+        res = stage.execute(_doc({"op": "collect", "input": {
+            "op": "sort", "keys": [{"col": "x", "desc": desc}],
+            "limit": 200, "input": scan_of(t, "f64s")}}))
+        order = "descending" if desc else "ascending"
+        want = t.sort_by([("x", order)]).slice(0, 200)
+        got = [r[0] for r in res["rows"]]
+        assert got == want["x"].to_pylist()
+    # nulls: ASC -> last; with limit < non-null count none appear
+    assert all(g is not None for g in got)
+
+
 def test_stage_errors_fail_loudly(ctx):
     with pytest.raises(RuntimeError, match="unregistered device table"):
         stage.execute(_doc({"op": "collect", "input": {
