@@ -286,6 +286,11 @@ struct Writer {
   void str(const std::string& s) { escape_to(out, s); }
   void num(int64_t v) { out += std::to_string(v); }
   void dbl(double v) {
+    // JSON has no literals for non-finite values; emit the Python-json
+    // extensions (Infinity/-Infinity/NaN) which the host parser accepts
+    if (v != v) { out += "NaN"; return; }
+    if (v > 1.7976931348623157e308) { out += "Infinity"; return; }
+    if (v < -1.7976931348623157e308) { out += "-Infinity"; return; }
     char buf[40];
     snprintf(buf, sizeof(buf), "%.17g", v);
     out += buf;
