@@ -1071,8 +1071,12 @@ def test_sort_float64_keys(ctx, reg):
                   "pid": pa.array(np.arange(n, dtype=np.int64))})
     reg("f64s", t)
     for desc in (False, True):
+        # pyarrow sort_by places nulls at_end for BOTH directions; pin the
+        # plan's nulls_first accordingly (the kernel's default is the SQL
+        # convention: DESC -> NULLS FIRST)
         res = stage.execute(_doc({"op": "collect", "input": {
-            "op": "sort", "keys": [{"col": "x", "desc": desc}],
+            "op": "sort", "keys": [{"col": "x", "desc": desc,
+                                    "nulls_first": False}],
             "limit": 200, "input": scan_of(t, "f64s")}}))
         order = "descending" if desc else "ascending"
         want = t.sort_by([("x", order)]).slice(0, 200)
