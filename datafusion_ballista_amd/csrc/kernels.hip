@@ -6648,3 +6648,131 @@ extern "C" int bg_bitmap_or(const uint8_t* d_a, const uint8_t* d_b,
   HIP_TRY(hipGetLastError());
   return BG_OK;
 }
+
+// ---------------------------------------------------------------------------
+// Multi-expression Decimal128 projection in ONE pass (round 2): q1-class
+// stages evaluate several arithmetic expressions over the same columns;
+// chaining bg_project_dec128 launches materialises every intermediate
+// (q1: ~77 GB of extra HBM traffic at SF100).  This kernel interprets a
+// tiny stack bytecode per row — each input column is read once, each
+// output written once, intermediates live in registers.
+// ---------------------------------------------------------------------------
+#define BG_EXPR_PUSH_COL 0 /* arg = column index */
+#define BG_EXPR_PUSH_LIT 1 /* arg = literal index */
+#define BG_EXPR_MUL 2
+#define BG_EXPR_ADD 3
+#define BG_EXPR_SUB 4      /* top = below - top */
+#define BG_EXPR_MAX_OPS 24
+#define BG_EXPR_MAX_OUT 6
+#define BG_EXPR_MAX_LITS 8
+
+struct BgExprProg {
+  int32_t ops[BG_EXPR_MAX_OPS];
+  int32_t args[BG_EXPR_MAX_OPS];
+  int32_t nops;
+  int32_t expr_end[BG_EXPR_MAX_OUT];  // bytecode index AFTER each expr
+  int32_t nexprs;
+  i64 lit_lo[BG_EXPR_MAX_LITS];
+  i64 lit_hi[BG_EXPR_MAX_LITS];
+  const void* col_data[BG_MAX_AGGS];
+  int32_t col_dtype[BG_MAX_AGGS];
+  int32_t ncols;
+  void* out[BG_EXPR_MAX_OUT];
+};
+
+__global__ void k_project_multi(BgExprProg p, int64_t n) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    i128 col[BG_MAX_AGGS];
+    for (int c = 0; c < p.ncols; ++c) {
+      switch (p.col_dtype[c]) {
+        case BG_DT_DECIMAL128: {
+          const ulong2 v =
+              reinterpret_cast<const ulong2*>(p.col_data[c])[i];
+          col[c] = make_i128(v.x, (i64)v.y);
+          break;
+        }
+        case BG_DT_INT64:
+          col[c] = (i128) reinterpret_cast<const int64_t*>(p.col_data[c])[i];
+          break;
+        default:
+          col[c] = 0;
+      }
+    }
+    i128 stack[5];
+    int sp = 0;
+    int e = 0;
+    for (int k = 0; k < p.nops; ++k) {
+      switch (p.ops[k]) {
+        case BG_EXPR_PUSH_COL: stack[sp++] = col[p.args[k]]; break;
+        case BG_EXPR_PUSH_LIT:
+          stack[sp++] =
+              make_i128((u64)p.lit_lo[p.args[k]], p.lit_hi[p.args[k]]);
+          break;
+        case BG_EXPR_MUL:
+          stack[sp - 2] = stack[sp - 2] * stack[sp - 1];
+          --sp;
+          break;
+        case BG_EXPR_ADD:
+          stack[sp - 2] = stack[sp - 2] + stack[sp - 1];
+          --sp;
+          break;
+        case BG_EXPR_SUB:
+          stack[sp - 2] = stack[sp - 2] - stack[sp - 1];
+          --sp;
+          break;
+        default: break;
+      }
+      if (k + 1 == p.expr_end[e]) {
+        const i128 v = stack[--sp];
+        ulong2* o = reinterpret_cast<ulong2*>(p.out[e]);
+        ulong2 w;
+        w.x = (u64)(u128)v;
+        w.y = (u64)((u128)v >> 64);
+        o[i] = w;
+        ++e;
+      }
+    }
+  }
+}
+
+extern "C" int bg_project_dec128_multi(
+    const bg_column* cols, int32_t ncols, const int32_t* ops,
+    const int32_t* args, int32_t nops, const int32_t* expr_end,
+    int32_t nexprs, const int64_t* lit_lo, const int64_t* lit_hi,
+    int32_t nlits, int64_t n, void* const* d_outs) {
+  REQUIRE_INIT();
+  if (ncols < 0 || ncols > BG_MAX_AGGS || nops < 1 ||
+      nops > BG_EXPR_MAX_OPS || nexprs < 1 || nexprs > BG_EXPR_MAX_OUT ||
+      nlits < 0 || nlits > BG_EXPR_MAX_LITS)
+    return set_err(BG_ERR_INVALID, "bg_project_dec128_multi: limits");
+  BgExprProg p{};
+  p.nops = nops;
+  p.nexprs = nexprs;
+  p.ncols = ncols;
+  for (int k = 0; k < nops; ++k) {
+    p.ops[k] = ops[k];
+    p.args[k] = args[k];
+  }
+  for (int e = 0; e < nexprs; ++e) {
+    p.expr_end[e] = expr_end[e];
+    p.out[e] = d_outs[e];
+  }
+  for (int l = 0; l < nlits; ++l) {
+    p.lit_lo[l] = lit_lo[l];
+    p.lit_hi[l] = lit_hi[l];
+  }
+  for (int c = 0; c < ncols; ++c) {
+    p.col_data[c] = cols[c].d_data;
+    p.col_dtype[c] = cols[c].dtype;
+    if (cols[c].dtype != BG_DT_DECIMAL128 && cols[c].dtype != BG_DT_INT64)
+      return set_err(BG_ERR_UNSUPPORTED,
+                     "bg_project_dec128_multi: dec128/int64 inputs");
+  }
+  int blocks = (int)bg_imin64((n + BG_BLOCK - 1) / BG_BLOCK, BG_MAX_BLOCKS);
+  if (blocks == 0) blocks = 1;
+  hipLaunchKernelGGL(k_project_multi, dim3(blocks), dim3(BG_BLOCK), 0, 0, p,
+                     n);
+  HIP_TRY(hipGetLastError());
+  return BG_OK;
+}

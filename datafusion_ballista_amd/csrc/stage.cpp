@@ -1109,6 +1109,133 @@ static ExprVal eval_expr(const Table& t, const Value& e) {
   throw StageError(BG_ERR_INVALID, "plan: unknown expression node");
 }
 
+
+// ---- fused multi-expression Decimal128 projection ----
+// Compile a dec128 arithmetic tree to the stack bytecode of
+// bg_project_dec128_multi; returns false when the tree contains nodes
+// outside {col(dec128), lit, mul, add, sub} (caller falls back to the
+// per-op path).
+struct ExprProgBuild {
+  std::vector<int32_t> ops, args;
+  std::vector<int32_t> expr_end;
+  std::vector<int64_t> lit_lo, lit_hi;
+  std::vector<int> col_ids;  // table column indices
+};
+
+static bool compile_expr_rec(const Table& t, const Value& e,
+                             ExprProgBuild* p, int* scale, int* depth) {
+  if (*depth > 8) return false;
+  ++*depth;
+  if (e.has("col")) {
+    int ci;
+    try {
+      ci = t.idx(e.get_str("col"));
+    } catch (...) {
+      return false;
+    }
+    const Col& c = t.cols[(size_t)ci];
+    if (c.dtype != BG_DT_DECIMAL128 || c.nullable()) return false;
+    int slot = -1;
+    for (size_t s = 0; s < p->col_ids.size(); ++s)
+      if (p->col_ids[s] == ci) slot = (int)s;
+    if (slot < 0) {
+      if (p->col_ids.size() >= 8) return false;
+      slot = (int)p->col_ids.size();
+      p->col_ids.push_back(ci);
+    }
+    p->ops.push_back(0);  // PUSH_COL
+    p->args.push_back(slot);
+    *scale = c.scale;
+    return true;
+  }
+  if (e.has("lit")) {
+    if (p->lit_lo.size() >= 8) return false;
+    int64_t lo, hi;
+    parse_i128(e.at("lit"), &lo, &hi);
+    p->ops.push_back(1);  // PUSH_LIT
+    p->args.push_back((int32_t)p->lit_lo.size());
+    p->lit_lo.push_back(lo);
+    p->lit_hi.push_back(hi);
+    *scale = -1;  // literal adopts the sibling's scale
+    return true;
+  }
+  const char* kinds[3] = {"mul", "add", "sub"};
+  const int32_t opcodes[3] = {2, 3, 4};
+  for (int k = 0; k < 3; ++k) {
+    if (!e.has(kinds[k])) continue;
+    auto& opnds = e.get_arr(kinds[k]);
+    if (opnds.size() != 2) return false;
+    int sa = -1, sb = -1;
+    if (!compile_expr_rec(t, *opnds[0], p, &sa, depth)) return false;
+    if (!compile_expr_rec(t, *opnds[1], p, &sb, depth)) return false;
+    p->ops.push_back(opcodes[k]);
+    p->args.push_back(0);
+    if (k == 0)  // mul: scales add (literals contribute 0)
+      *scale = (sa < 0 ? 0 : sa) + (sb < 0 ? 0 : sb);
+    else
+      *scale = sa >= 0 ? sa : sb;
+    return true;
+  }
+  return false;
+}
+
+// Evaluate a set of expressions over t: arithmetic trees fuse into ONE
+// kernel pass (each input column read once); anything else falls back to
+// eval_expr.  outs[i] receives the i-th expression's column.
+static void eval_exprs_fused(const Table& t,
+                             const std::vector<const Value*>& exprs,
+                             std::vector<Col>* outs) {
+  outs->assign(exprs.size(), Col{});
+  ExprProgBuild p;
+  std::vector<size_t> fused;
+  std::vector<int> scales;
+  for (size_t i = 0; i < exprs.size(); ++i) {
+    if (exprs[i]->has("col")) continue;  // plain refs share buffers
+    ExprProgBuild trial = p;
+    int scale = 0, depth = 0;
+    if (fused.size() < 6 && trial.ops.size() + 12 < 24 &&
+        compile_expr_rec(t, *exprs[i], &trial, &scale, &depth) &&
+        trial.ops.size() <= 24) {
+      trial.expr_end.push_back((int32_t)trial.ops.size());
+      p = trial;
+      fused.push_back(i);
+      scales.push_back(scale);
+    }
+  }
+  if (fused.size() >= 2) {  // fusion pays only with shared passes
+    std::vector<bg_column> cols;
+    for (int ci : p.col_ids) cols.push_back(to_bg(t.cols[(size_t)ci], t.n));
+    std::vector<void*> douts;
+    for (size_t fi = 0; fi < fused.size(); ++fi) {
+      Col c;
+      c.dtype = BG_DT_DECIMAL128;
+      c.precision = 38;
+      c.scale = scales[fi];
+      c.data = dalloc((uint64_t)(t.n > 0 ? t.n : 1) * 16);
+      douts.push_back(c.data->p);
+      (*outs)[fused[fi]] = std::move(c);
+    }
+    chk(bg_project_dec128_multi(
+            cols.data(), (int32_t)cols.size(), p.ops.data(), p.args.data(),
+            (int32_t)p.ops.size(), p.expr_end.data(),
+            (int32_t)p.expr_end.size(), p.lit_lo.data(), p.lit_hi.data(),
+            (int32_t)p.lit_lo.size(), t.n, douts.data()),
+        "bg_project_dec128_multi");
+  } else {
+    fused.clear();
+  }
+  for (size_t i = 0; i < exprs.size(); ++i) {
+    bool done = false;
+    for (size_t fi : fused)
+      if (fi == i) done = true;
+    if (done && (*outs)[i].data) continue;
+    ExprVal v = eval_expr(t, *exprs[i]);
+    if (v.is_lit)
+      throw StageError(BG_ERR_INVALID, "bare literal expression column");
+    (*outs)[i] = v.col;
+  }
+}
+
 // rsub: lit - col  (op 3) — expressed as {"sub":[{"lit":...}, {"col":...}]}
 // handled by eval_dec_binary's op_lc above.
 
@@ -1253,12 +1380,14 @@ static Table exec_project(const Value& node, Metrics& m) {
   Table in = exec_plan(node.at("input"), m);
   Table out;
   out.n = in.n;
+  std::vector<const Value*> exprs;
+  for (auto& ex : node.get_arr("exprs")) exprs.push_back(&ex->at("expr"));
+  std::vector<Col> cols;
+  eval_exprs_fused(in, exprs, &cols);
+  size_t i = 0;
   for (auto& ex : node.get_arr("exprs")) {
-    ExprVal v = eval_expr(in, ex->at("expr"));
-    if (v.is_lit)
-      throw StageError(BG_ERR_INVALID, "project: bare literal column");
     out.names.push_back(ex->get_str("as"));
-    out.cols.push_back(v.col);
+    out.cols.push_back(std::move(cols[i++]));
   }
   return out;
 }
@@ -1544,7 +1673,30 @@ static Table exec_aggregate(const Value& node, Metrics& m) {
         "bg_memset");
   }
 
-  // aggregate inputs
+  // aggregate inputs — arithmetic expressions fuse into one projection
+  // pass (q1-class chains cost ~77 GB of intermediates otherwise)
+  std::vector<const Value*> pre_exprs;
+  std::vector<size_t> pre_idx;
+  {
+    size_t ai = 0;
+    for (auto& a : node.get_arr("aggs")) {
+      const std::string fn = a->get_str("fn");
+      if (!is_final && a->has("expr") && !(fn == "count" && !a->has("expr"))) {
+        pre_exprs.push_back(&a->at("expr"));
+        pre_idx.push_back(ai);
+      }
+      ++ai;
+    }
+  }
+  std::vector<Col> pre_cols;
+  if (!pre_exprs.empty()) eval_exprs_fused(in, pre_exprs, &pre_cols);
+  auto pre_col_of = [&](size_t agg_index) -> Col* {
+    for (size_t i = 0; i < pre_idx.size(); ++i)
+      if (pre_idx[i] == agg_index) return &pre_cols[i];
+    return nullptr;
+  };
+  size_t agg_index = 0;
+
   std::vector<AggSpec> specs;
   std::vector<Col> agg_in;       // evaluated agg input columns
   std::vector<int32_t> agg_ops;
@@ -1602,9 +1754,14 @@ static Table exec_aggregate(const Value& node, Metrics& m) {
       if (s.fn == "count" && !a->has("expr")) {
         s.agg_slot = -1;  // COUNT(*): the kernel's group counts
       } else {
-        ExprVal v = eval_expr(in, a->at("expr"));
-        if (v.is_lit)
-          throw StageError(BG_ERR_INVALID, "aggregate of a literal");
+        ExprVal v;
+        Col* pc = pre_col_of(agg_index);
+        if (pc) v.col = *pc;
+        else {
+          v = eval_expr(in, a->at("expr"));
+          if (v.is_lit)
+            throw StageError(BG_ERR_INVALID, "aggregate of a literal");
+        }
         s.in_dt = v.col.dtype; s.in_prec = v.col.precision;
         s.in_scale = v.col.scale;
         if (s.fn == "sum" || s.fn == "avg" || s.fn == "count")
@@ -1624,6 +1781,7 @@ static Table exec_aggregate(const Value& node, Metrics& m) {
       }
     }
     specs.push_back(s);
+    ++agg_index;
   }
 
   // run the kernel (auto-grow on table-full, like gpu.py)

@@ -317,6 +317,18 @@ int bg_hashagg2(const bg_column* key_cols, int32_t nkeys,
 int bg_project_dec128(int32_t op, const bg_column* a, const bg_column* b,
                       int64_t lit_lo, int64_t lit_hi, int64_t n, void* d_out);
 
+/* Multi-expression Decimal128 projection in one pass: stack bytecode
+ * (PUSH_COL/PUSH_LIT/MUL/ADD/SUB) over <=8 input columns, <=6 output
+ * dec128 columns; each input read once, intermediates in registers (the
+ * q1-class expression chains cost ~77 GB of materialised intermediates
+ * at SF100 without this). */
+int bg_project_dec128_multi(const bg_column* cols, int32_t ncols,
+                            const int32_t* ops, const int32_t* args,
+                            int32_t nops, const int32_t* expr_end,
+                            int32_t nexprs, const int64_t* lit_lo,
+                            const int64_t* lit_hi, int32_t nlits, int64_t n,
+                            void* const* d_outs);
+
 /* ---- SortExec (stable multi-column ORDER BY) ----
  * d_perm (u32[n]) receives the stable row permutation realising ORDER BY
  * key_cols[0] [DESC], key_cols[1] [DESC], ...; Top-K = first K entries.
