@@ -193,6 +193,14 @@ def main():
                  "groups": len(r1_["rows"]), "rows_per_s": n / warm}
     print("q1:", json.dumps(out["q1"]), flush=True)
 
+    if "--stop-after-q1" in sys.argv:
+        path = os.path.join(ROOT, "gpurun_out", "perf_dbgen_sf100.json")
+        os.makedirs(os.path.dirname(path), exist_ok=True)
+        with open(path, "w") as f:
+            json.dump(out, f, indent=1)
+        print("stopped after q1 (flag)", flush=True)
+        return
+
     # ---------------- q3 (real group key, dbgen correlations) ----------
     join1 = {"op": "hash_join",
              "build": {"op": "filter",
