@@ -3182,12 +3182,14 @@ __global__ void k_snap_matches(const SnappyPage* __restrict__ pages,
                                int64_t npages,
                                const SnapDesc* __restrict__ descs,
                                const int64_t* __restrict__ desc_base,
-                               const int64_t* __restrict__ counts) {
+                               const int64_t* __restrict__ counts,
+                               const uint8_t* __restrict__ skip) {
   const int64_t wave_global =
       ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) / BG_WAVE;
   const int64_t nwaves = ((int64_t)gridDim.x * blockDim.x) / BG_WAVE;
   const int lane = lane_id();
   for (int64_t p = wave_global; p < npages; p += nwaves) {
+    if (skip && skip[p]) continue;  // big pages replay via list ranking
     uint8_t* dst = pages[p].dst;
     const SnapDesc* d = descs + desc_base[p];
     const int64_t nd = counts[p];
@@ -3317,6 +3319,7 @@ struct SnapBigPage {
   int64_t arr_base;  // per-byte array base (bytes)
   int64_t seg_base;  // per-seg array base
   int64_t nsegs;
+  int64_t par_base;  // parent-array base (output bytes; replay phase)
 };
 
 // phase A: speculative per-segment chase
@@ -3527,6 +3530,121 @@ __global__ void k_snapbig_emit(const SnappyPage* __restrict__ pages,
   }
 }
 
+// ---------------------------------------------------------------------------
+// Giant-page match REPLAY by list ranking (round 2, final): the in-order
+// wave replay (k_snap_matches) is serial per page, so one 1-MB page with
+// ~150k matches bounds the decode no matter how many pages run.  But the
+// copy structure is a FOREST: every output byte of a match has exactly ONE
+// parent byte — the window byte it copies, (dst-aux) + ((b-dst) mod aux)
+// (the mod makes overlapping/RLE matches periodic fills of the pre-match
+// window, exactly the serial semantics) — and parents always point
+// strictly backward, with literal bytes as roots.  So:
+//   par_init     P[b] = b                     (literal bytes stay roots)
+//   par_scatter  match descs write their bytes' parents     (parallel)
+//   par_double   P[b] = P^4[b] until fixpoint (jump-4 pointer doubling:
+//                ceil(log4 depth) rounds; measured chain depths 200-1000
+//                => 5-6 rounds; races only jump FURTHER toward the root,
+//                so the passes are correct without any synchronisation)
+//   par_fill     dst[b] = dst[P[b]]           (after literals land; every
+//                root is a literal byte, so one unordered gather)
+// The doubling/fill loops are XCD-swizzled: page -> XCD by index so each
+// page's parent array (4 B/byte; 4 MB for a 1-MB page) lives in ONE
+// XCD's L2 instead of thrashing all eight.
+// ---------------------------------------------------------------------------
+
+__global__ void k_snap_par_init(const SnapBigPage* __restrict__ bigs,
+                                int64_t nbig,
+                                const int64_t* __restrict__ lens,
+                                uint32_t* __restrict__ P) {
+  const int xcd = blockIdx.x & 7;
+  const int slot = blockIdx.x >> 3;
+  const int64_t nslot = gridDim.x >> 3;
+  for (int64_t bi = xcd; bi < nbig; bi += 8) {
+    const SnapBigPage& bp = bigs[bi];
+    const int64_t ulen = lens[bp.page_idx];
+    if (ulen <= 0) continue;
+    uint32_t* p = P + bp.par_base;
+    for (int64_t b = slot * blockDim.x + threadIdx.x; b < ulen;
+         b += nslot * blockDim.x)
+      p[b] = (uint32_t)b;
+  }
+}
+
+__global__ void k_snap_par_scatter(const SnapDesc* __restrict__ descs,
+                                   const int64_t* __restrict__ desc_base,
+                                   const int64_t* __restrict__ counts,
+                                   const SnapBigPage* __restrict__ bigs,
+                                   int64_t big0,
+                                   const int64_t* __restrict__ lens,
+                                   uint32_t* __restrict__ P) {
+  const SnapBigPage& bp = bigs[big0 + blockIdx.y];
+  if (lens[bp.page_idx] <= 0) return;
+  const SnapDesc* d = descs + desc_base[bp.page_idx];
+  const int64_t nd = counts[bp.page_idx];
+  uint32_t* p = P + bp.par_base;
+  for (int64_t e = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; e < nd;
+       e += (int64_t)gridDim.x * blockDim.x) {
+    if (d[e].kind != 1) continue;
+    const uint32_t dst = d[e].dst, aux = d[e].aux, len = d[e].len;
+    const uint32_t base = dst - aux;  // aux <= dst validated at emit
+    if (aux >= len) {
+      for (uint32_t i = 0; i < len; ++i) p[dst + i] = base + i;
+    } else {
+      for (uint32_t i = 0; i < len; ++i) p[dst + i] = base + (i % aux);
+    }
+  }
+}
+
+__global__ void k_snap_par_double(const SnapBigPage* __restrict__ bigs,
+                                  int64_t nbig,
+                                  const int64_t* __restrict__ lens,
+                                  uint32_t* __restrict__ P,
+                                  int* __restrict__ changed) {
+  const int xcd = blockIdx.x & 7;
+  const int slot = blockIdx.x >> 3;
+  const int64_t nslot = gridDim.x >> 3;
+  int any = 0;
+  for (int64_t bi = xcd; bi < nbig; bi += 8) {
+    const SnapBigPage& bp = bigs[bi];
+    const int64_t ulen = lens[bp.page_idx];
+    if (ulen <= 0) continue;
+    uint32_t* p = P + bp.par_base;
+    for (int64_t b = slot * blockDim.x + threadIdx.x; b < ulen;
+         b += nslot * blockDim.x) {
+      const uint32_t r0 = p[b];
+      const uint32_t r1 = p[r0];
+      if (r1 == r0) continue;  // already at root
+      const uint32_t r2 = p[r1];
+      const uint32_t r3 = p[r2];
+      p[b] = p[r3];
+      any = 1;
+    }
+  }
+  if (any) *changed = 1;
+}
+
+__global__ void k_snap_par_fill(const SnappyPage* __restrict__ pages,
+                                const SnapBigPage* __restrict__ bigs,
+                                int64_t nbig,
+                                const int64_t* __restrict__ lens,
+                                const uint32_t* __restrict__ P) {
+  const int xcd = blockIdx.x & 7;
+  const int slot = blockIdx.x >> 3;
+  const int64_t nslot = gridDim.x >> 3;
+  for (int64_t bi = xcd; bi < nbig; bi += 8) {
+    const SnapBigPage& bp = bigs[bi];
+    const int64_t ulen = lens[bp.page_idx];
+    if (ulen <= 0) continue;
+    const uint32_t* p = P + bp.par_base;
+    uint8_t* out = pages[bp.page_idx].dst;
+    for (int64_t b = slot * blockDim.x + threadIdx.x; b < ulen;
+         b += nslot * blockDim.x) {
+      const uint32_t r = p[b];
+      if ((int64_t)r != b) out[b] = out[r];
+    }
+  }
+}
+
 /* Decompress npages independent Snappy blocks.  pages: HOST array copied
  * internally; each entry's src/dst are DEVICE pointers.  out_lens (host,
  * npages): decompressed length or -1 on malformed input. */
@@ -3592,12 +3710,32 @@ extern "C" int bg_snappy_decompress(const void* h_pages, int64_t npages,
     bp.arr_base = arr_total;
     bp.seg_base = seg_total;
     bp.nsegs = (hp[p].src_len + SNAPBIG_SEG - 1) / SNAPBIG_SEG;
+    bp.par_base = 0;
     arr_total += hp[p].src_len;
     for (int64_t k = 0; k < bp.nsegs; ++k)
       seg_page.push_back((int32_t)bigs.size());
     seg_total += bp.nsegs;
     bigs.push_back(bp);
   }
+  // list-ranking replay eligibility: parent arrays are sized from dst_cap
+  // (4 B per output byte), so every big page must declare its output size
+  // and the total must stay bounded; otherwise big pages fall back to the
+  // serial in-order replay (correct either way)
+  int64_t par_total = 0;
+  bool use_parents = !bigs.empty();
+  for (auto& bp : bigs) {
+    const int64_t cap = hp[bp.page_idx].dst_cap;
+    if (cap <= 0 || cap > (int64_t)UINT32_MAX) {
+      use_parents = false;
+      break;
+    }
+    bp.par_base = par_total;
+    par_total += cap;
+  }
+  if (par_total * (int64_t)sizeof(uint32_t) > ((int64_t)16 << 30))
+    use_parents = false;
+  SnapBigPage* d_bigs_keep = nullptr;
+  uint8_t* d_isbig = nullptr;
   if (bigs.empty()) {
     hipLaunchKernelGGL(k_snap_parse, dim3(blocks1), dim3(BG_BLOCK), 0, 0,
                        d_pages, npages, d_descs, d_base, d_counts, d_lens);
@@ -3736,7 +3874,10 @@ extern "C" int bg_snappy_decompress(const void* h_pages, int64_t npages,
               (long long)seg_total, tb1 - tb0, tb2 - tb1, tb3 - tb2,
               hc[0], hc[1], hc[2], hc[3]);
     }
-    (void)pool_release(d_bigs);
+    if (use_parents)
+      d_bigs_keep = d_bigs;  // replay phases below still need it
+    else
+      (void)pool_release(d_bigs);
     (void)pool_release(d_seg_page);
     (void)pool_release(d_next16);
     (void)pool_release(d_cnt_pre);
@@ -3766,19 +3907,69 @@ extern "C" int bg_snappy_decompress(const void* h_pages, int64_t npages,
   }
   double t2 = tick();
   t_lit = t2 - t1;
+  // big pages: list-ranking replay (parents were NOT built yet — all
+  // phases run here, after the literal roots have landed)
+  double t_par = 0;
+  int par_rounds = 0;
+  if (use_parents) {
+    uint32_t* d_par;
+    int* d_changed;
+    HIP_TRY(pool_malloc((void**)&d_par, sizeof(uint32_t) * par_total));
+    HIP_TRY(pool_malloc((void**)&d_changed, sizeof(int)));
+    std::vector<uint8_t> isbig(npages, 0);
+    for (auto& bp : bigs) isbig[bp.page_idx] = 1;
+    HIP_TRY(pool_malloc((void**)&d_isbig, npages));
+    HIP_TRY(hipMemcpy(d_isbig, isbig.data(), npages,
+                      hipMemcpyHostToDevice));
+    const int pb = 1024;  // multiple of 8: page->XCD swizzle
+    hipLaunchKernelGGL(k_snap_par_init, dim3(pb), dim3(BG_BLOCK), 0, 0,
+                       d_bigs_keep, (int64_t)bigs.size(), d_lens, d_par);
+    HIP_TRY(hipGetLastError());
+    for (int64_t b0 = 0; b0 < (int64_t)bigs.size(); b0 += 65535) {
+      uint32_t ny = (uint32_t)bg_imin64((int64_t)bigs.size() - b0, 65535);
+      hipLaunchKernelGGL(k_snap_par_scatter, dim3(64, ny), dim3(BG_BLOCK),
+                         0, 0, d_descs, d_base, d_counts, d_bigs_keep, b0,
+                         d_lens, d_par);
+      HIP_TRY(hipGetLastError());
+    }
+    // jump-4 doubling: parents strictly decrease (validated at emit), so
+    // depth <= 2^32 converges within 16 rounds of x4 path compression
+    for (par_rounds = 0; par_rounds < 20; ++par_rounds) {
+      HIP_TRY(hipMemset(d_changed, 0, sizeof(int)));
+      hipLaunchKernelGGL(k_snap_par_double, dim3(pb), dim3(BG_BLOCK), 0, 0,
+                         d_bigs_keep, (int64_t)bigs.size(), d_lens, d_par,
+                         d_changed);
+      HIP_TRY(hipGetLastError());
+      int h_changed = 0;
+      HIP_TRY(hipMemcpy(&h_changed, d_changed, sizeof(int),
+                        hipMemcpyDeviceToHost));
+      if (!h_changed) break;
+    }
+    hipLaunchKernelGGL(k_snap_par_fill, dim3(pb), dim3(BG_BLOCK), 0, 0,
+                       d_pages, d_bigs_keep, (int64_t)bigs.size(), d_lens,
+                       d_par);
+    HIP_TRY(hipGetLastError());
+    (void)pool_release(d_par);
+    (void)pool_release(d_changed);
+    if (dbg) t_par = tick() - t2;
+  }
+  double t2b = tick();
   int blocks3 = (int)bg_imin64(
       (npages + waves_per_block - 1) / waves_per_block, BG_MAX_BLOCKS);
   if (blocks3 == 0) blocks3 = 1;
   hipLaunchKernelGGL(k_snap_matches, dim3(blocks3), dim3(BG_BLOCK), 0, 0,
-                     d_pages, npages, d_descs, d_base, d_counts);
+                     d_pages, npages, d_descs, d_base, d_counts, d_isbig);
   HIP_TRY(hipGetLastError());
   if (dbg) {
-    t_match = tick() - t2;
+    t_match = tick() - t2b;
     fprintf(stderr,
             "[bg_snappy] npages=%lld parse=%.3fms literals=%.3fms "
-            "matches=%.3fms\n",
-            (long long)npages, t_parse, t_lit, t_match);
+            "par=%.3fms(rounds=%d) matches=%.3fms\n",
+            (long long)npages, t_parse, t_lit, t_par, par_rounds + 1,
+            t_match);
   }
+  if (d_bigs_keep) (void)pool_release(d_bigs_keep);
+  if (d_isbig) (void)pool_release(d_isbig);
   HIP_TRY(hipMemcpy(h_out_lens, d_lens, sizeof(int64_t) * npages,
                     hipMemcpyDeviceToHost));
   (void)pool_release(d_pages);

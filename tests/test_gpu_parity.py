@@ -734,6 +734,43 @@ def test_snappy_decompress_parity(ctx):
         assert got == p, f"page {i} content mismatch"
 
 
+def test_snappy_big_page_list_ranking_parity(ctx):
+    """Pages with compressed size above the big-page threshold (98304)
+    replay matches by list ranking (parent pointers + jump-4 doubling +
+    root gather) instead of the serial in-order wave replay.  Exact-byte
+    parity on the page shapes that route there: FLBA(7) decimal pages
+    (the q6-feed bound: ~150k matches, chain depth ~450), low-cardinality
+    int64 pages (depth ~200), and a just-above-threshold page."""
+    rng = np.random.default_rng(11)
+    payloads = []
+    vals = rng.integers(90000, 10495100, size=1_048_576 // 7)
+    payloads.append(b"".join(int(v).to_bytes(7, "big") for v in vals))
+    payloads.append(rng.integers(0, 100, size=1_048_576 // 8,
+                                 dtype=np.int64).tobytes())
+    # binary-search a raw size whose flba7 compressed size lands just
+    # above the threshold
+    for raw_bytes in (131_072, 160_000, 200_000):
+        vv = rng.integers(90000, 10495100, size=raw_bytes // 7)
+        p = b"".join(int(v).to_bytes(7, "big") for v in vv)
+        if len(pa.compress(p, codec="snappy", asbytes=True)) > 98304:
+            payloads.append(p)
+            break
+    comp = [pa.compress(p, codec="snappy", asbytes=True) for p in payloads]
+    assert any(len(c) > 98304 for c in comp)
+    pages = []
+    dsts = []
+    for c, p in zip(comp, payloads):
+        src = ctx.upload(np.frombuffer(c, dtype=np.uint8))
+        dst = ctx.alloc(len(p))
+        pages.append((src, len(c), dst, len(p)))
+        dsts.append(dst)
+    lens = ctx.snappy_decompress(pages)
+    for i, p in enumerate(payloads):
+        assert lens[i] == len(p), f"big page {i} length mismatch"
+        got = dsts[i].download(np.uint8, len(p)).tobytes()
+        assert got == p, f"big page {i} content mismatch"
+
+
 def test_snappy_malformed_is_rejected(ctx):
     """Truncated/garbage pages must fail loudly (-1), not write junk."""
     good = pa.compress(b"A" * 5000, codec="snappy", asbytes=True)
