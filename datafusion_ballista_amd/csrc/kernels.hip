@@ -3922,9 +3922,11 @@ extern "C" int bg_snappy_decompress(const void* h_pages, int64_t npages,
     HIP_TRY(hipMemcpy(d_isbig, isbig.data(), npages,
                       hipMemcpyHostToDevice));
     const int pb = 1024;  // multiple of 8: page->XCD swizzle
+    double tp0 = tick();
     hipLaunchKernelGGL(k_snap_par_init, dim3(pb), dim3(BG_BLOCK), 0, 0,
                        d_bigs_keep, (int64_t)bigs.size(), d_lens, d_par);
     HIP_TRY(hipGetLastError());
+    double tp1 = tick();
     for (int64_t b0 = 0; b0 < (int64_t)bigs.size(); b0 += 65535) {
       uint32_t ny = (uint32_t)bg_imin64((int64_t)bigs.size() - b0, 65535);
       hipLaunchKernelGGL(k_snap_par_scatter, dim3(64, ny), dim3(BG_BLOCK),
@@ -3932,6 +3934,7 @@ extern "C" int bg_snappy_decompress(const void* h_pages, int64_t npages,
                          d_lens, d_par);
       HIP_TRY(hipGetLastError());
     }
+    double tp2 = tick();
     // jump-4 doubling: parents strictly decrease (validated at emit), so
     // depth <= 2^32 converges within 16 rounds of x4 path compression
     for (par_rounds = 0; par_rounds < 20; ++par_rounds) {
@@ -3945,13 +3948,21 @@ extern "C" int bg_snappy_decompress(const void* h_pages, int64_t npages,
                         hipMemcpyDeviceToHost));
       if (!h_changed) break;
     }
+    double tp3 = tick();
     hipLaunchKernelGGL(k_snap_par_fill, dim3(pb), dim3(BG_BLOCK), 0, 0,
                        d_pages, d_bigs_keep, (int64_t)bigs.size(), d_lens,
                        d_par);
     HIP_TRY(hipGetLastError());
     (void)pool_release(d_par);
     (void)pool_release(d_changed);
-    if (dbg) t_par = tick() - t2;
+    if (dbg) {
+      t_par = tick() - t2;
+      fprintf(stderr,
+              "[bg_snappy]   par: nbig=%lld init=%.3fms scatter=%.3fms "
+              "double=%.3fms fill=%.3fms\n",
+              (long long)bigs.size(), tp1 - tp0, tp2 - tp1, tp3 - tp2,
+              t_par - (tp3 - t2));
+    }
   }
   double t2b = tick();
   int blocks3 = (int)bg_imin64(
