@@ -3402,7 +3402,77 @@ __global__ void k_snapbig_spec(const SnappyPage* __restrict__ pages,
   }
 }
 
-// phase B: one thread per page resolves segment entries/prefixes
+// phase A2: parallel catch-up walks.  Phase B's serial per-page loop was
+// latency-bound when few big pages exist (one THREAD per page x ~17
+// dependent decodes per segment = 47 ms on a 160-page q6 feed).  But each
+// segment's catch-up walk is independent under one assumption: the true
+// entry into segment k is segment k-1's speculative exit (true whenever
+// the true chain merged with k-1's chain — the self-synced common case).
+// So walk every segment boundary in PARALLEL from that candidate entry,
+// recording what the walk consumed and where it landed; phase B becomes
+// an O(1)-per-segment composition that falls back to the serial walk only
+// when its entry disagrees with the candidate (rare: upstream crossings).
+//   w_land >= 0: landed on the segment's spec chain at this offset
+//   w_land == -1: walked past the segment end (exit = w_exit)
+//   w_land == -4: decode failed on the walk (malformed if entry is true)
+__global__ void k_snapbig_walk(const SnappyPage* __restrict__ pages,
+                               const SnapBigPage* __restrict__ bigs,
+                               const int32_t* __restrict__ seg_page,
+                               int64_t total_segs,
+                               const uint16_t* __restrict__ next16,
+                               const int64_t* __restrict__ seg_exit,
+                               const int64_t* __restrict__ page_hdr,
+                               int64_t* __restrict__ w_land,
+                               int64_t* __restrict__ w_cnt,
+                               int64_t* __restrict__ w_out,
+                               int64_t* __restrict__ w_exit) {
+  for (int64_t gs = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       gs < total_segs; gs += (int64_t)gridDim.x * blockDim.x) {
+    const SnapBigPage& bp = bigs[seg_page[gs]];
+    const int64_t k = gs - bp.seg_base;
+    const SnappyPage& pg = pages[bp.page_idx];
+    const uint8_t* s = pg.src;
+    const int64_t src_len = pg.src_len;
+    const int64_t e = k == 0 ? page_hdr[seg_page[gs]] : seg_exit[gs - 1];
+    w_cnt[gs] = 0;
+    w_out[gs] = 0;
+    if (e < 0) {
+      w_land[gs] = -2;  // upstream spec error; resolve goes serial
+      w_exit[gs] = -1;
+      continue;
+    }
+    const int64_t seg_end =
+        (k + 1) * SNAPBIG_SEG < src_len ? (k + 1) * SNAPBIG_SEG : src_len;
+    if (e >= seg_end) {  // candidate passes this segment entirely
+      w_land[gs] = -3;
+      w_exit[gs] = e;
+      continue;
+    }
+    int64_t cur = e, cnt = 0, outsum = 0, land = -1;
+    while (cur < seg_end) {
+      if (next16[bp.arr_base + cur] != 0) {
+        land = cur;
+        break;
+      }
+      int64_t consumed, out;
+      uint32_t aux, kind;
+      if (!snap_decode_at(s, src_len, cur, &consumed, &out, &aux, &kind)) {
+        land = -4;
+        break;
+      }
+      cnt += 1;
+      outsum += out;
+      cur += consumed;
+    }
+    w_land[gs] = land;
+    w_cnt[gs] = cnt;
+    w_out[gs] = outsum;
+    w_exit[gs] = land >= 0 ? seg_exit[gs] : cur;
+  }
+}
+
+// phase B: one thread per page composes segment entries/prefixes from the
+// parallel walks (serial catch-up only on candidate mismatch)
 __global__ void k_snapbig_resolve(const SnappyPage* __restrict__ pages,
                                   const SnapBigPage* __restrict__ bigs,
                                   int64_t nbig, unsigned long long* dbg_ctr,
@@ -3414,6 +3484,10 @@ __global__ void k_snapbig_resolve(const SnappyPage* __restrict__ pages,
                                   const int64_t* __restrict__ seg_exit,
                                   const int64_t* __restrict__ page_hdr,
                                   const int64_t* __restrict__ page_ulen,
+                                  const int64_t* __restrict__ w_land,
+                                  const int64_t* __restrict__ w_cnt,
+                                  const int64_t* __restrict__ w_out,
+                                  const int64_t* __restrict__ w_exit,
                                   int64_t* __restrict__ seg_entry,
                                   int64_t* __restrict__ seg_nd,
                                   int64_t* __restrict__ seg_di,
@@ -3440,33 +3514,44 @@ __global__ void k_snapbig_resolve(const SnappyPage* __restrict__ pages,
       seg_entry[gs] = cur;
       seg_nd[gs] = nd;
       seg_di[gs] = di;
-      // catch up element by element until we merge with the segment's
-      // speculative chain (usually immediately) or leave the segment
-      bool advanced = false;
-      while (cur < seg_end) {
-        if (next16[bp.arr_base + cur] != 0) {
-          nd += (int64_t)seg_cnt_tot[gs] - cnt_pre[bp.arr_base + cur];
-          di += (int64_t)seg_out_tot[gs] - out_pre[bp.arr_base + cur];
-          const int64_t ex = seg_exit[gs];
-          if (ex < 0) { err = 1; }
-          cur = ex;
-          advanced = true;
-          break;
+      const int64_t e = k == 0 ? hdr : seg_exit[gs - 1];
+      const int64_t wl = w_land[gs];
+      if (cur == e && wl != -2 && wl != -3) {
+        // the parallel walk ran from exactly this entry: compose
+        if (wl == -4) { err = 1; break; }
+        nd += w_cnt[gs];
+        di += w_out[gs];
+        if (wl >= 0) {
+          nd += (int64_t)seg_cnt_tot[gs] - cnt_pre[bp.arr_base + wl];
+          di += (int64_t)seg_out_tot[gs] - out_pre[bp.arr_base + wl];
+          if (seg_exit[gs] < 0) { err = 1; break; }
         }
-        int64_t consumed, out;
-        uint32_t aux, kind;
-        if (!snap_decode_at(s, src_len, cur, &consumed, &out, &aux,
-                            &kind)) {
-          err = 1;
-          break;
+        cur = w_exit[gs];
+        if (cur < 0) { err = 1; break; }
+      } else {
+        // candidate mismatch: serial catch-up (rare)
+        while (cur < seg_end) {
+          if (next16[bp.arr_base + cur] != 0) {
+            nd += (int64_t)seg_cnt_tot[gs] - cnt_pre[bp.arr_base + cur];
+            di += (int64_t)seg_out_tot[gs] - out_pre[bp.arr_base + cur];
+            const int64_t ex = seg_exit[gs];
+            if (ex < 0) { err = 1; }
+            cur = ex;
+            break;
+          }
+          int64_t consumed, out;
+          uint32_t aux, kind;
+          if (!snap_decode_at(s, src_len, cur, &consumed, &out, &aux,
+                              &kind)) {
+            err = 1;
+            break;
+          }
+          nd += 1;
+          di += out;
+          cur += consumed;
+          if (dbg_ctr) atomicAdd(&dbg_ctr[1], 1ull);
         }
-        nd += 1;
-        di += out;
-        cur += consumed;
-        advanced = true;
-        if (dbg_ctr) atomicAdd(&dbg_ctr[1], 1ull);
       }
-      (void)advanced;
       if (dbg_ctr) atomicAdd(&dbg_ctr[0], 1ull);
       if (di > ulen) err = 1;
     }
@@ -3734,6 +3819,11 @@ extern "C" int bg_snappy_decompress(const void* h_pages, int64_t npages,
   }
   if (par_total * (int64_t)sizeof(uint32_t) > ((int64_t)16 << 30))
     use_parents = false;
+  // with MANY big pages the serial per-page replay already fills the chip
+  // (one wave per page) and the doubling passes' extra traffic loses —
+  // measured crossover ~700 pages (160: 6.2 ms vs 23 ms serial; 2000:
+  // 55 ms vs 25 ms serial)
+  if ((int64_t)bigs.size() > 768) use_parents = false;
   SnapBigPage* d_bigs_keep = nullptr;
   uint8_t* d_isbig = nullptr;
   if (bigs.empty()) {
@@ -3847,15 +3937,34 @@ extern "C" int bg_snappy_decompress(const void* h_pages, int64_t npages,
                        d_page_hdr, d_page_ulen);
     HIP_TRY(hipGetLastError());
     double tb1 = tick();
+    int64_t* d_w_land;
+    int64_t* d_w_cnt;
+    int64_t* d_w_out;
+    int64_t* d_w_exit;
+    HIP_TRY(pool_malloc((void**)&d_w_land, sizeof(int64_t) * seg_total));
+    HIP_TRY(pool_malloc((void**)&d_w_cnt, sizeof(int64_t) * seg_total));
+    HIP_TRY(pool_malloc((void**)&d_w_out, sizeof(int64_t) * seg_total));
+    HIP_TRY(pool_malloc((void**)&d_w_exit, sizeof(int64_t) * seg_total));
+    hipLaunchKernelGGL(k_snapbig_walk, dim3(ab), dim3(BG_BLOCK), 0, 0,
+                       d_pages, d_bigs, d_seg_page, seg_total, d_next16,
+                       d_seg_exit, d_page_hdr, d_w_land, d_w_cnt, d_w_out,
+                       d_w_exit);
+    HIP_TRY(hipGetLastError());
+    double tb1b = tick();
     int bb = (int)bg_imin64(((int64_t)bigs.size() + BG_BLOCK - 1) / BG_BLOCK,
                             BG_MAX_BLOCKS);
     if (bb == 0) bb = 1;
     hipLaunchKernelGGL(k_snapbig_resolve, dim3(bb), dim3(BG_BLOCK), 0, 0,
                        d_pages, d_bigs, (int64_t)bigs.size(), d_dbg, d_next16,
                        d_cnt_pre, d_out_pre, d_seg_ct, d_seg_ot, d_seg_exit,
-                       d_page_hdr, d_page_ulen, d_seg_entry, d_seg_nd,
+                       d_page_hdr, d_page_ulen, d_w_land, d_w_cnt, d_w_out,
+                       d_w_exit, d_seg_entry, d_seg_nd,
                        d_seg_di, d_counts, d_lens);
     HIP_TRY(hipGetLastError());
+    (void)pool_release(d_w_land);
+    (void)pool_release(d_w_cnt);
+    (void)pool_release(d_w_out);
+    (void)pool_release(d_w_exit);
     double tb2 = tick();
     hipLaunchKernelGGL(k_snapbig_emit, dim3(ab), dim3(BG_BLOCK), 0, 0,
                        d_pages, d_bigs, d_seg_page, seg_total, d_seg_entry,
@@ -3868,11 +3977,11 @@ extern "C" int bg_snappy_decompress(const void* h_pages, int64_t npages,
       (void)hipMemcpy(hc, d_dbg, 32, hipMemcpyDeviceToHost);
       (void)pool_release(d_dbg);
       fprintf(stderr,
-              "[bg_snappy]   big: nseg=%lld spec=%.3fms resolve=%.3fms "
-              "emit=%.3fms seg_steps=%llu catchup=%llu spec_elems=%llu "
-              "spec_died=%llu\n",
-              (long long)seg_total, tb1 - tb0, tb2 - tb1, tb3 - tb2,
-              hc[0], hc[1], hc[2], hc[3]);
+              "[bg_snappy]   big: nseg=%lld spec=%.3fms walk=%.3fms "
+              "resolve=%.3fms emit=%.3fms seg_steps=%llu catchup=%llu "
+              "spec_elems=%llu spec_died=%llu\n",
+              (long long)seg_total, tb1 - tb0, tb1b - tb1, tb2 - tb1b,
+              tb3 - tb2, hc[0], hc[1], hc[2], hc[3]);
     }
     if (use_parents)
       d_bigs_keep = d_bigs;  // replay phases below still need it
