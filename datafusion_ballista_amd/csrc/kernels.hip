@@ -3169,6 +3169,19 @@ __global__ void k_snap_literals(const SnappyPage* __restrict__ pages,
                        (int64_t)(threadIdx.x / BG_WAVE);
   const int64_t wstride = (int64_t)gridDim.x * waves_per_block;
   const int lane = threadIdx.x & (BG_WAVE - 1);
+  if (nd <= wstride) {
+    // fewer descriptors than waves (incompressible pages: a handful of
+    // giant literals) — wave-per-descriptor uses every wave; batching
+    // would serialise them all onto one wave (measured 295 -> 31 GB/s)
+    for (int64_t e = wave; e < nd; e += wstride) {
+      if (d[e].kind != 0) continue;
+      const uint8_t* s = src + d[e].aux;
+      uint8_t* o = dst + d[e].dst;
+      const int64_t len = d[e].len;
+      for (int64_t i = lane; i < len; i += BG_WAVE) o[i] = s[i];
+    }
+    return;
+  }
   for (int64_t e0 = wave * BG_WAVE; e0 < nd; e0 += wstride * BG_WAVE) {
     const int batch = nd - e0 < BG_WAVE ? (int)(nd - e0) : BG_WAVE;
     SnapDesc my{};
