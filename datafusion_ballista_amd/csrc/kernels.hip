@@ -3838,29 +3838,38 @@ extern "C" int bg_snappy_decompress(const void* h_pages, int64_t npages,
     seg_total += bp.nsegs;
     bigs.push_back(bp);
   }
-  // list-ranking replay eligibility: parent arrays are sized from dst_cap
-  // (4 B per output byte), so every big page must declare its output size
-  // and the total must stay bounded; otherwise big pages fall back to the
-  // serial in-order replay (correct either way)
+  // List-ranking replay routing — INDEPENDENT of the segmented-parse
+  // routing above: replay cost is set by matches x chain depth, not by
+  // compressed size, so it also covers highly-expanding pages whose
+  // compressed bytes are small (1-MB RLE "runs" pages: 50 KB compressed,
+  // 16k matches chained 1000 deep — 1.35 GB/s serial).  Eligible: output
+  // large enough that the serial in-order replay is the bound, not
+  // near-incompressible (replay there is trivial), output size declared
+  // (parent arrays are 4 B per output byte).  With MANY such pages the
+  // serial per-page replay already fills the chip (one wave per page)
+  // and the doubling passes' extra traffic loses — measured crossover
+  // ~700 1-MB pages (160: 6.2 ms vs 23 ms serial; 2000: 55 vs 25 ms).
+  std::vector<SnapBigPage> par_pages;
   int64_t par_total = 0;
-  bool use_parents = !bigs.empty();
-  for (auto& bp : bigs) {
-    const int64_t cap = hp[bp.page_idx].dst_cap;
-    if (cap <= 0 || cap > (int64_t)UINT32_MAX) {
+  bool use_parents = true;
+  for (int64_t p = 0; p < npages; ++p) {
+    const int64_t cap = hp[p].dst_cap;
+    if (cap <= (int64_t)131072) continue;
+    if ((double)hp[p].src_len > 0.85 * (double)cap) continue;
+    if (cap > (int64_t)UINT32_MAX) {
       use_parents = false;
       break;
     }
-    bp.par_base = par_total;
+    SnapBigPage pp{};
+    pp.page_idx = p;
+    pp.par_base = par_total;
     par_total += cap;
+    par_pages.push_back(pp);
   }
-  if (par_total * (int64_t)sizeof(uint32_t) > ((int64_t)16 << 30))
+  if (par_pages.empty() ||
+      par_total * (int64_t)sizeof(uint32_t) > ((int64_t)16 << 30) ||
+      (int64_t)par_pages.size() > 768)
     use_parents = false;
-  // with MANY big pages the serial per-page replay already fills the chip
-  // (one wave per page) and the doubling passes' extra traffic loses —
-  // measured crossover ~700 pages (160: 6.2 ms vs 23 ms serial; 2000:
-  // 55 ms vs 25 ms serial)
-  if ((int64_t)bigs.size() > 768) use_parents = false;
-  SnapBigPage* d_bigs_keep = nullptr;
   uint8_t* d_isbig = nullptr;
   if (bigs.empty()) {
     hipLaunchKernelGGL(k_snap_parse, dim3(blocks1), dim3(BG_BLOCK), 0, 0,
@@ -4019,10 +4028,7 @@ extern "C" int bg_snappy_decompress(const void* h_pages, int64_t npages,
               (long long)seg_total, tb1 - tb0, tb1b - tb1, tb2 - tb1b,
               tb3 - tb2, hc[0], hc[1], hc[2], hc[3]);
     }
-    if (use_parents)
-      d_bigs_keep = d_bigs;  // replay phases below still need it
-    else
-      (void)pool_release(d_bigs);
+    (void)pool_release(d_bigs);
     (void)pool_release(d_seg_page);
     (void)pool_release(d_next16);
     (void)pool_release(d_cnt_pre);
@@ -4057,25 +4063,30 @@ extern "C" int bg_snappy_decompress(const void* h_pages, int64_t npages,
   double t_par = 0;
   int par_rounds = 0;
   if (use_parents) {
+    const int64_t npar = (int64_t)par_pages.size();
+    SnapBigPage* d_parp;
     uint32_t* d_par;
     int* d_changed;
+    HIP_TRY(pool_malloc((void**)&d_parp, sizeof(SnapBigPage) * npar));
     HIP_TRY(pool_malloc((void**)&d_par, sizeof(uint32_t) * par_total));
     HIP_TRY(pool_malloc((void**)&d_changed, sizeof(int)));
+    HIP_TRY(hipMemcpy(d_parp, par_pages.data(),
+                      sizeof(SnapBigPage) * npar, hipMemcpyHostToDevice));
     std::vector<uint8_t> isbig(npages, 0);
-    for (auto& bp : bigs) isbig[bp.page_idx] = 1;
+    for (auto& pp : par_pages) isbig[pp.page_idx] = 1;
     HIP_TRY(pool_malloc((void**)&d_isbig, npages));
     HIP_TRY(hipMemcpy(d_isbig, isbig.data(), npages,
                       hipMemcpyHostToDevice));
     const int pb = 1024;  // multiple of 8: page->XCD swizzle
     double tp0 = tick();
     hipLaunchKernelGGL(k_snap_par_init, dim3(pb), dim3(BG_BLOCK), 0, 0,
-                       d_bigs_keep, (int64_t)bigs.size(), d_lens, d_par);
+                       d_parp, npar, d_lens, d_par);
     HIP_TRY(hipGetLastError());
     double tp1 = tick();
-    for (int64_t b0 = 0; b0 < (int64_t)bigs.size(); b0 += 65535) {
-      uint32_t ny = (uint32_t)bg_imin64((int64_t)bigs.size() - b0, 65535);
+    for (int64_t b0 = 0; b0 < npar; b0 += 65535) {
+      uint32_t ny = (uint32_t)bg_imin64(npar - b0, 65535);
       hipLaunchKernelGGL(k_snap_par_scatter, dim3(64, ny), dim3(BG_BLOCK),
-                         0, 0, d_descs, d_base, d_counts, d_bigs_keep, b0,
+                         0, 0, d_descs, d_base, d_counts, d_parp, b0,
                          d_lens, d_par);
       HIP_TRY(hipGetLastError());
     }
@@ -4085,8 +4096,7 @@ extern "C" int bg_snappy_decompress(const void* h_pages, int64_t npages,
     for (par_rounds = 0; par_rounds < 20; ++par_rounds) {
       HIP_TRY(hipMemset(d_changed, 0, sizeof(int)));
       hipLaunchKernelGGL(k_snap_par_double, dim3(pb), dim3(BG_BLOCK), 0, 0,
-                         d_bigs_keep, (int64_t)bigs.size(), d_lens, d_par,
-                         d_changed);
+                         d_parp, npar, d_lens, d_par, d_changed);
       HIP_TRY(hipGetLastError());
       int h_changed = 0;
       HIP_TRY(hipMemcpy(&h_changed, d_changed, sizeof(int),
@@ -4095,17 +4105,17 @@ extern "C" int bg_snappy_decompress(const void* h_pages, int64_t npages,
     }
     double tp3 = tick();
     hipLaunchKernelGGL(k_snap_par_fill, dim3(pb), dim3(BG_BLOCK), 0, 0,
-                       d_pages, d_bigs_keep, (int64_t)bigs.size(), d_lens,
-                       d_par);
+                       d_pages, d_parp, npar, d_lens, d_par);
     HIP_TRY(hipGetLastError());
+    (void)pool_release(d_parp);
     (void)pool_release(d_par);
     (void)pool_release(d_changed);
     if (dbg) {
       t_par = tick() - t2;
       fprintf(stderr,
-              "[bg_snappy]   par: nbig=%lld init=%.3fms scatter=%.3fms "
+              "[bg_snappy]   par: npar=%lld init=%.3fms scatter=%.3fms "
               "double=%.3fms fill=%.3fms\n",
-              (long long)bigs.size(), tp1 - tp0, tp2 - tp1, tp3 - tp2,
+              (long long)npar, tp1 - tp0, tp2 - tp1, tp3 - tp2,
               t_par - (tp3 - t2));
     }
   }
@@ -4124,7 +4134,6 @@ extern "C" int bg_snappy_decompress(const void* h_pages, int64_t npages,
             (long long)npages, t_parse, t_lit, t_par, par_rounds + 1,
             t_match);
   }
-  if (d_bigs_keep) (void)pool_release(d_bigs_keep);
   if (d_isbig) (void)pool_release(d_isbig);
   HIP_TRY(hipMemcpy(h_out_lens, d_lens, sizeof(int64_t) * npages,
                     hipMemcpyDeviceToHost));
