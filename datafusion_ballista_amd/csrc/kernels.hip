@@ -3849,27 +3849,45 @@ extern "C" int bg_snappy_decompress(const void* h_pages, int64_t npages,
   // serial per-page replay already fills the chip (one wave per page)
   // and the doubling passes' extra traffic loses — measured crossover
   // ~700 1-MB pages (160: 6.2 ms vs 23 ms serial; 2000: 55 vs 25 ms).
+  // Two eligibility classes (both need a declared output size <= 4 GB):
+  //   A — heavy-RLE pages (csz < 0.15 x usz): their match chains are deep
+  //       (offset-1 runs), so the serial replay is a per-page ~100 ms
+  //       latency wall no matter how many pages run concurrently (1-MB
+  //       runs pages: 7.3 us/descriptor).  ALWAYS list-rank, any count.
+  //   B — moderate-expansion pages: serial replay is throughput-bound and
+  //       page-parallelism covers it once enough pages exist — measured
+  //       crossover ~700 1-MB pages (160: 6.2 ms parents vs 23 ms serial;
+  //       2000: 55 vs 25 ms).  List-rank only when the count is low.
+  // The combined set is processed in chunks of <= 4 GB of parent arrays
+  // to bound replay memory.
   std::vector<SnapBigPage> par_pages;
-  int64_t par_total = 0;
-  bool use_parents = true;
+  bool caps_ok = true;
+  int64_t class_b = 0;
   for (int64_t p = 0; p < npages; ++p) {
     const int64_t cap = hp[p].dst_cap;
     if (cap <= (int64_t)131072) continue;
-    if ((double)hp[p].src_len > 0.85 * (double)cap) continue;
+    const double ratio = (double)hp[p].src_len / (double)cap;
+    if (ratio > 0.85) continue;
     if (cap > (int64_t)UINT32_MAX) {
-      use_parents = false;
+      caps_ok = false;
       break;
     }
     SnapBigPage pp{};
     pp.page_idx = p;
-    pp.par_base = par_total;
-    par_total += cap;
+    pp.par_base = ratio < 0.15 ? -1 : 0;  // -1 marks class A for the cut
     par_pages.push_back(pp);
+    if (pp.par_base == 0) ++class_b;
   }
-  if (par_pages.empty() ||
-      par_total * (int64_t)sizeof(uint32_t) > ((int64_t)16 << 30) ||
-      (int64_t)par_pages.size() > 768)
-    use_parents = false;
+  if (!caps_ok) {
+    par_pages.clear();
+  } else if (class_b > 768) {
+    // keep only class A
+    std::vector<SnapBigPage> keep;
+    for (auto& pp : par_pages)
+      if (pp.par_base < 0) keep.push_back(pp);
+    par_pages.swap(keep);
+  }
+  const bool use_parents = !par_pages.empty();
   uint8_t* d_isbig = nullptr;
   if (bigs.empty()) {
     hipLaunchKernelGGL(k_snap_parse, dim3(blocks1), dim3(BG_BLOCK), 0, 0,
@@ -4063,60 +4081,84 @@ extern "C" int bg_snappy_decompress(const void* h_pages, int64_t npages,
   double t_par = 0;
   int par_rounds = 0;
   if (use_parents) {
-    const int64_t npar = (int64_t)par_pages.size();
-    SnapBigPage* d_parp;
-    uint32_t* d_par;
-    int* d_changed;
-    HIP_TRY(pool_malloc((void**)&d_parp, sizeof(SnapBigPage) * npar));
-    HIP_TRY(pool_malloc((void**)&d_par, sizeof(uint32_t) * par_total));
-    HIP_TRY(pool_malloc((void**)&d_changed, sizeof(int)));
-    HIP_TRY(hipMemcpy(d_parp, par_pages.data(),
-                      sizeof(SnapBigPage) * npar, hipMemcpyHostToDevice));
     std::vector<uint8_t> isbig(npages, 0);
     for (auto& pp : par_pages) isbig[pp.page_idx] = 1;
     HIP_TRY(pool_malloc((void**)&d_isbig, npages));
     HIP_TRY(hipMemcpy(d_isbig, isbig.data(), npages,
                       hipMemcpyHostToDevice));
+    int* d_changed;
+    HIP_TRY(pool_malloc((void**)&d_changed, sizeof(int)));
+    const int64_t CHUNK_ELEMS = ((int64_t)4 << 30) / sizeof(uint32_t);
     const int pb = 1024;  // multiple of 8: page->XCD swizzle
-    double tp0 = tick();
-    hipLaunchKernelGGL(k_snap_par_init, dim3(pb), dim3(BG_BLOCK), 0, 0,
-                       d_parp, npar, d_lens, d_par);
-    HIP_TRY(hipGetLastError());
-    double tp1 = tick();
-    for (int64_t b0 = 0; b0 < npar; b0 += 65535) {
-      uint32_t ny = (uint32_t)bg_imin64(npar - b0, 65535);
-      hipLaunchKernelGGL(k_snap_par_scatter, dim3(64, ny), dim3(BG_BLOCK),
-                         0, 0, d_descs, d_base, d_counts, d_parp, b0,
-                         d_lens, d_par);
+    double tp_init = 0, tp_scat = 0, tp_dbl = 0, tp_fill = 0;
+    size_t ci = 0;
+    while (ci < par_pages.size()) {
+      std::vector<SnapBigPage> chunk;
+      int64_t elems = 0;
+      while (ci < par_pages.size()) {
+        const int64_t cap = hp[par_pages[ci].page_idx].dst_cap;
+        if (!chunk.empty() && elems + cap > CHUNK_ELEMS) break;
+        SnapBigPage pp = par_pages[ci];
+        pp.par_base = elems;
+        elems += cap;
+        chunk.push_back(pp);
+        ++ci;
+      }
+      const int64_t npar = (int64_t)chunk.size();
+      SnapBigPage* d_parp;
+      uint32_t* d_par;
+      HIP_TRY(pool_malloc((void**)&d_parp, sizeof(SnapBigPage) * npar));
+      HIP_TRY(pool_malloc((void**)&d_par, sizeof(uint32_t) * elems));
+      HIP_TRY(hipMemcpy(d_parp, chunk.data(), sizeof(SnapBigPage) * npar,
+                        hipMemcpyHostToDevice));
+      double tp0 = tick();
+      hipLaunchKernelGGL(k_snap_par_init, dim3(pb), dim3(BG_BLOCK), 0, 0,
+                         d_parp, npar, d_lens, d_par);
       HIP_TRY(hipGetLastError());
-    }
-    double tp2 = tick();
-    // jump-4 doubling: parents strictly decrease (validated at emit), so
-    // depth <= 2^32 converges within 16 rounds of x4 path compression
-    for (par_rounds = 0; par_rounds < 20; ++par_rounds) {
-      HIP_TRY(hipMemset(d_changed, 0, sizeof(int)));
-      hipLaunchKernelGGL(k_snap_par_double, dim3(pb), dim3(BG_BLOCK), 0, 0,
-                         d_parp, npar, d_lens, d_par, d_changed);
+      double tp1 = tick();
+      for (int64_t b0 = 0; b0 < npar; b0 += 65535) {
+        uint32_t ny = (uint32_t)bg_imin64(npar - b0, 65535);
+        hipLaunchKernelGGL(k_snap_par_scatter, dim3(64, ny),
+                           dim3(BG_BLOCK), 0, 0, d_descs, d_base, d_counts,
+                           d_parp, b0, d_lens, d_par);
+        HIP_TRY(hipGetLastError());
+      }
+      double tp2 = tick();
+      // jump-4 doubling: parents strictly decrease (validated at parse),
+      // so convergence is guaranteed; x4 path compression per round
+      for (int round = 0; round < 20; ++round) {
+        HIP_TRY(hipMemset(d_changed, 0, sizeof(int)));
+        hipLaunchKernelGGL(k_snap_par_double, dim3(pb), dim3(BG_BLOCK), 0,
+                           0, d_parp, npar, d_lens, d_par, d_changed);
+        HIP_TRY(hipGetLastError());
+        int h_changed = 0;
+        HIP_TRY(hipMemcpy(&h_changed, d_changed, sizeof(int),
+                          hipMemcpyDeviceToHost));
+        ++par_rounds;
+        if (!h_changed) break;
+      }
+      double tp3 = tick();
+      hipLaunchKernelGGL(k_snap_par_fill, dim3(pb), dim3(BG_BLOCK), 0, 0,
+                         d_pages, d_parp, npar, d_lens, d_par);
       HIP_TRY(hipGetLastError());
-      int h_changed = 0;
-      HIP_TRY(hipMemcpy(&h_changed, d_changed, sizeof(int),
-                        hipMemcpyDeviceToHost));
-      if (!h_changed) break;
+      (void)pool_release(d_parp);
+      (void)pool_release(d_par);
+      if (dbg) {
+        double tp4 = tick();
+        tp_init += tp1 - tp0;
+        tp_scat += tp2 - tp1;
+        tp_dbl += tp3 - tp2;
+        tp_fill += tp4 - tp3;
+      }
     }
-    double tp3 = tick();
-    hipLaunchKernelGGL(k_snap_par_fill, dim3(pb), dim3(BG_BLOCK), 0, 0,
-                       d_pages, d_parp, npar, d_lens, d_par);
-    HIP_TRY(hipGetLastError());
-    (void)pool_release(d_parp);
-    (void)pool_release(d_par);
     (void)pool_release(d_changed);
     if (dbg) {
       t_par = tick() - t2;
       fprintf(stderr,
               "[bg_snappy]   par: npar=%lld init=%.3fms scatter=%.3fms "
               "double=%.3fms fill=%.3fms\n",
-              (long long)npar, tp1 - tp0, tp2 - tp1, tp3 - tp2,
-              t_par - (tp3 - t2));
+              (long long)par_pages.size(), tp_init, tp_scat, tp_dbl,
+              tp_fill);
     }
   }
   double t2b = tick();
