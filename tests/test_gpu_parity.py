@@ -747,6 +747,13 @@ def test_snappy_big_page_list_ranking_parity(ctx):
     payloads.append(b"".join(int(v).to_bytes(7, "big") for v in vals))
     payloads.append(rng.integers(0, 100, size=1_048_576 // 8,
                                  dtype=np.int64).tobytes())
+    # heavy-RLE page (class A: always list-ranked regardless of count):
+    # a repeated 64-B pattern compresses ~20x with deep offset-64 chains
+    payloads.append(rng.integers(0, 256, size=64,
+                                 dtype=np.uint8).tobytes() * (1_048_576
+                                                              // 64))
+    # and a pure zero page (offset-1 chains, the deepest possible)
+    payloads.append(b"\x00" * 1_048_576)
     # binary-search a raw size whose flba7 compressed size lands just
     # above the threshold
     for raw_bytes in (131_072, 160_000, 200_000):
