@@ -3735,6 +3735,9 @@ __global__ void k_snap_par_double(const SnapBigPage* __restrict__ bigs,
       const uint32_t r0 = p[b];
       const uint32_t r1 = p[r0];
       if (r1 == r0) continue;  // already at root
+      // jump-4; jump-8 measured WORSE (flba7 double 5.0 -> 5.8 ms, runs
+      // 17.4 -> 15.7 GB/s): the longer dependent-load chain costs more
+      // than the saved write passes
       const uint32_t r2 = p[r1];
       const uint32_t r3 = p[r2];
       p[b] = p[r3];
