@@ -4666,6 +4666,230 @@ extern "C" int bg_def_levels_batch(const void* h_jobs, int64_t njobs) {
 }
 
 // ---------------------------------------------------------------------------
+// LIST columns (parquet repetition levels, max_rep == 1): a V1 page body
+// is [u32 rlen][rep levels][u32 dlen][def levels][values].  rep == 0
+// starts a row; def tells how far down the schema the slot is defined
+// (0 = NULL list, 1 = empty list, >= def_entry = an element slot exists,
+// == max_def = the element is non-null and consumes one packed value).
+// Rows never span pages (checked), so one wave-lane walks both level
+// streams per page: pass 1 counts {rows, entries, present} and reports
+// rlen (the host needs it to re-point the UNCHANGED extract/dict kernels
+// at the [u32 dlen][def][values] suffix); pass 2 emits per-row entry
+// counts, the two validity bitmaps and the page-local value index per
+// entry (the same vidx contract bg_def_levels_batch feeds mode-2
+// extraction with).  Levels are the RLE/bit-packed hybrid at bit width
+// <= 8 (max_def <= 255); deeper nesting (max_rep > 1) is rejected by the
+// host reader.
+// ---------------------------------------------------------------------------
+struct ListLevelsJob {
+  const uint8_t* page;  // [u32 rlen][rep][u32 dlen][def][values]
+  int64_t page_len;
+  int64_t nslots;      // level entries (page header num_values)
+  int32_t max_def;
+  int32_t def_entry;   // min def meaning an element slot exists
+  int32_t def_valid;   // min def meaning the LIST itself is non-null
+                       // (1 for an optional list, 0 for a required one)
+  int32_t _pad;
+  int64_t row_base;    // pass 2: column-global bases
+  int64_t entry_base;
+  int64_t* counts;     // pass 1: [rows, entries, present, rlen]
+  int32_t* row_sizes;  // pass 2: entries per row (column-global index)
+  uint32_t* list_valid;  // pass 2: row-space bitmap words
+  uint32_t* elem_valid;  // pass 2: entry-space bitmap words
+  uint32_t* vidx;        // pass 2: PAGE-LOCAL slice (host offsets it)
+};
+
+struct LvlRd {
+  const uint8_t* d;
+  const uint8_t* end;
+  int w;
+  int64_t rle_left = 0;
+  int rle_val = 0;
+  int64_t bp_left = 0;
+  u64 bitbuf = 0;
+  int bits = 0;
+  bool ok = true;
+  __device__ int next() {
+    while (true) {
+      if (rle_left > 0) {
+        --rle_left;
+        return rle_val;
+      }
+      if (bp_left > 0) {
+        if (bits < w) {
+          while (bits <= 56 && d < end) {
+            bitbuf |= (u64)(*d++) << bits;
+            bits += 8;
+          }
+          if (bits < w) {
+            ok = false;
+            return 0;
+          }
+        }
+        const int v = (int)(bitbuf & ((1u << w) - 1));
+        bitbuf >>= w;
+        bits -= w;
+        --bp_left;
+        return v;
+      }
+      if (d >= end) {
+        ok = false;
+        return 0;
+      }
+      u64 h = 0;
+      int sh = 0;
+      while (d < end) {
+        const uint8_t b = *d++;
+        h |= (u64)(b & 0x7f) << sh;
+        if (!(b & 0x80)) break;
+        sh += 7;
+      }
+      if (h & 1) {
+        bp_left = (int64_t)(h >> 1) * 8;  // groups of 8 w-bit values
+        bitbuf = 0;
+        bits = 0;
+      } else {
+        rle_left = (int64_t)(h >> 1);
+        if (d >= end) {
+          ok = false;
+          return 0;
+        }
+        rle_val = *d++;  // bit width <= 8: value fits one byte
+      }
+    }
+  }
+};
+
+__device__ __forceinline__ int bg_bitwidth(int v) {
+  int w = 0;
+  while ((1 << w) <= v) ++w;
+  return w < 1 ? 1 : w;
+}
+
+__device__ void k_list_levels_body(const ListLevelsJob& job, int pass,
+                                   int* err) {
+  if (lane_id() != 0) return;
+  const uint8_t* page = job.page;
+  const int64_t plen = job.page_len;
+  if (plen < 8) { atomicExch(err, 4); return; }
+  const uint32_t rlen = (uint32_t)page[0] | ((uint32_t)page[1] << 8) |
+                        ((uint32_t)page[2] << 16) |
+                        ((uint32_t)page[3] << 24);
+  if (4 + (int64_t)rlen + 4 > plen) { atomicExch(err, 4); return; }
+  const uint8_t* dpos = page + 4 + rlen;
+  const uint32_t dlen = (uint32_t)dpos[0] | ((uint32_t)dpos[1] << 8) |
+                        ((uint32_t)dpos[2] << 16) |
+                        ((uint32_t)dpos[3] << 24);
+  if (4 + (int64_t)rlen + 4 + (int64_t)dlen > plen) {
+    atomicExch(err, 4);
+    return;
+  }
+  LvlRd rr{page + 4, page + 4 + rlen, 1};
+  LvlRd dr{dpos + 4, dpos + 4 + dlen, bg_bitwidth(job.max_def)};
+  int64_t rows = 0, entries = 0, present = 0;
+  int64_t cur_row = -1;
+  // word-batched bitmap emission (same pattern as the def-level walk)
+  int64_t lv_w = -1, ev_w = -1;
+  uint32_t lv_cur = 0, ev_cur = 0;
+  for (int64_t s = 0; s < job.nslots; ++s) {
+    const int rep = rr.next();
+    const int def = dr.next();
+    if (!rr.ok || !dr.ok || rep > 1 || def > job.max_def) {
+      atomicExch(err, 4);
+      return;
+    }
+    if (s == 0 && rep != 0) {  // a row spanning pages: unsupported
+      atomicExch(err, 5);
+      return;
+    }
+    if (rep == 0) {
+      cur_row = job.row_base + rows;
+      ++rows;
+      if (pass == 2) {
+        job.row_sizes[cur_row] = 0;
+        if (def >= job.def_valid) {
+          const int64_t w = cur_row >> 5;
+          if (w != lv_w) {
+            if (lv_w >= 0 && lv_cur) atomicOr(&job.list_valid[lv_w], lv_cur);
+            lv_w = w;
+            lv_cur = 0;
+          }
+          lv_cur |= 1u << (cur_row & 31);
+        }
+      }
+    }
+    if (def >= job.def_entry) {
+      if (pass == 2) {
+        const int64_t ent = job.entry_base + entries;
+        job.row_sizes[cur_row] += 1;
+        if (def == job.max_def) {
+          const int64_t w = ent >> 5;
+          if (w != ev_w) {
+            if (ev_w >= 0 && ev_cur) atomicOr(&job.elem_valid[ev_w], ev_cur);
+            ev_w = w;
+            ev_cur = 0;
+          }
+          ev_cur |= 1u << (ent & 31);
+          job.vidx[entries] = (uint32_t)present;
+        } else {
+          job.vidx[entries] = 0xffffffffu;
+        }
+      }
+      ++entries;
+    }
+    if (def == job.max_def) ++present;
+  }
+  if (pass == 2) {
+    if (lv_w >= 0 && lv_cur) atomicOr(&job.list_valid[lv_w], lv_cur);
+    if (ev_w >= 0 && ev_cur) atomicOr(&job.elem_valid[ev_w], ev_cur);
+  } else {
+    job.counts[0] = rows;
+    job.counts[1] = entries;
+    job.counts[2] = present;
+    job.counts[3] = (int64_t)rlen;
+  }
+}
+
+__global__ void k_list_levels_batch(const ListLevelsJob* jobs, int64_t njobs,
+                                    int pass, int* err) {
+  const int64_t wave_global =
+      ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) / BG_WAVE;
+  const int64_t nwaves = ((int64_t)gridDim.x * blockDim.x) / BG_WAVE;
+  for (int64_t j = wave_global; j < njobs; j += nwaves)
+    k_list_levels_body(jobs[j], pass, err);
+}
+
+extern "C" int bg_list_levels_batch(const void* h_jobs, int64_t njobs,
+                                    int32_t pass) {
+  REQUIRE_INIT();
+  if (njobs <= 0) return BG_OK;
+  ListLevelsJob* d_jobs;
+  int* d_err;
+  HIP_TRY(pool_malloc((void**)&d_jobs, sizeof(ListLevelsJob) * njobs));
+  HIP_TRY(pool_malloc((void**)&d_err, sizeof(int)));
+  HIP_TRY(hipMemset(d_err, 0, sizeof(int)));
+  HIP_TRY(hipMemcpy(d_jobs, h_jobs, sizeof(ListLevelsJob) * njobs,
+                    hipMemcpyHostToDevice));
+  const int waves_per_block = BG_BLOCK / BG_WAVE;
+  int blocks = (int)bg_imin64((njobs + waves_per_block - 1) / waves_per_block,
+                              BG_MAX_BLOCKS);
+  if (blocks == 0) blocks = 1;
+  hipLaunchKernelGGL(k_list_levels_batch, dim3(blocks), dim3(BG_BLOCK), 0, 0,
+                     d_jobs, njobs, (int)pass, d_err);
+  HIP_TRY(hipGetLastError());
+  int err = 0;
+  HIP_TRY(hipMemcpy(&err, d_err, sizeof(int), hipMemcpyDeviceToHost));
+  (void)pool_release(d_jobs);
+  (void)pool_release(d_err);
+  if (err == 5)
+    return set_err(BG_ERR_UNSUPPORTED,
+                   "bg_list_levels_batch: row spans pages");
+  if (err)
+    return set_err(BG_ERR_INVALID, "bg_list_levels_batch: malformed levels");
+  return BG_OK;
+}
+
+// ---------------------------------------------------------------------------
 // DELTA_BINARY_PACKED (encoding 5) integer pages — the reference's
 // parquet-rs V2 writer default for INT32/INT64 (restated from the parquet
 // spec Encodings.md "Delta Encoding"): header = <block_size varint>

@@ -588,6 +588,20 @@ class BgDefLevelsJob(ctypes.Structure):
                 ("d_n_present", ctypes.c_void_p)]
 
 
+class BgListLevelsJob(ctypes.Structure):
+    _fields_ = [("d_page", ctypes.c_void_p), ("page_len", ctypes.c_int64),
+                ("nslots", ctypes.c_int64), ("max_def", ctypes.c_int32),
+                ("def_entry", ctypes.c_int32),
+                ("def_valid", ctypes.c_int32), ("_pad", ctypes.c_int32),
+                ("row_base", ctypes.c_int64),
+                ("entry_base", ctypes.c_int64),
+                ("d_counts", ctypes.c_void_p),
+                ("d_row_sizes", ctypes.c_void_p),
+                ("d_list_valid", ctypes.c_void_p),
+                ("d_elem_valid", ctypes.c_void_p),
+                ("d_vidx", ctypes.c_void_p)]
+
+
 def _ctx_gather_varlen(self, src_data: "DeviceBuffer", src_offsets: "DeviceBuffer",
                        idx: "DeviceBuffer", m: int, max_bytes: int):
     """Variable-length take -> (offsets buf i32[m+1], data buf, total)."""

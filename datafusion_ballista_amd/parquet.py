@@ -198,6 +198,12 @@ class GpuParquetColumnReader:
 
         import ctypes
         chunk = self._file_buf  # page offsets below are absolute file offsets
+        if self.pf.schema.column(col).max_repetition_level > 0:
+            # nested column through the FLAT path would misread the page
+            # body ([u32 rlen][rep] sits ahead of the def section): route
+            # LIST columns to read_list_column_all, reject deeper nesting
+            raise RuntimeError(
+                "nested column: use read_list_column_all (max_rep == 1)")
         max_def0 = self.pf.schema.column(col).max_definition_level
         headers = []  # (rg, ptype, file_off, csz, usz, nvals, enc, ndict,
         #               soff, v2) with v2 = None (V1) or
@@ -634,3 +640,230 @@ class GpuParquetColumnReader:
         ctx.synchronize()
         return (out, total_values, phys if not flba else "DECIMAL128",
                 valid)
+
+    def read_list_column_all(self, col: int, rgs=None):
+        """Decode a LIST<primitive> column (max_rep == 1) across row
+        groups: repetition + definition levels walk on device
+        (bg_list_levels_batch: parquet-format.md "Nested Encoding"; the
+        reference reads these via arrow-rs' list reader), element values
+        via the UNCHANGED mode-2 extract/dict kernels re-pointed past the
+        [u32 rlen][rep] section.  List offsets are the prefix sum of the
+        per-row entry counts — computed host-side from one row_sizes
+        download (i32/row; an explicit bridge, like the ZSTD host codec).
+
+        -> dict(offsets=np.int32[nrows+1], n_rows, n_entries,
+                values=DeviceBuffer (entry space), phys,
+                elem_valid=DeviceBuffer u32 words,
+                list_valid=DeviceBuffer u32 words or None (required list),
+                row_sizes=DeviceBuffer i32[nrows])"""
+        import ctypes
+        ctx = self.ctx
+        if rgs is None:
+            rgs = range(self.pf.metadata.num_row_groups)
+        rgs = list(rgs)
+        sc = self.pf.schema.column(col)
+        max_rep = sc.max_repetition_level
+        max_def = sc.max_definition_level
+        if max_rep != 1:
+            raise RuntimeError(
+                f"max_rep={max_rep}: only 1-level LIST is GPU-decodable")
+        # level thresholds from the schema's nullability shape:
+        #   optional list adds 1, the repeated level adds 1, optional
+        #   element adds 1 (parquet-format.md Nested Encoding)
+        arrow_field = self.pf.schema_arrow.field(
+            sc.path.split(".")[0])
+        list_nullable = arrow_field.nullable
+        elem_nullable = max_def == (2 if not list_nullable else 3)
+        def_entry = max_def - (1 if elem_nullable else 0)
+        def_valid = 1 if list_nullable else 0
+        meta = self.pf.metadata.row_group(rgs[0]).column(col)
+        codec = meta.compression
+        if codec not in ("SNAPPY", "UNCOMPRESSED"):
+            raise RuntimeError(f"LIST + codec {codec}: not GPU-decodable")
+        phys = meta.physical_type
+        if phys not in _PHYS_NP:
+            raise RuntimeError(f"LIST of {phys}: not GPU-decodable yet")
+        esz = _PHYS_NP[phys][1]
+
+        chunk = self._file_buf
+        headers = []  # (soff, usz, nslots, enc, ndict_or_0, rg)
+        scratch_total = 0
+        total_slots = 0
+        for rg in rgs:
+            m = self.pf.metadata.row_group(rg).column(col)
+            start = m.data_page_offset
+            if m.has_dictionary_page and \
+                    m.dictionary_page_offset is not None and \
+                    m.dictionary_page_offset < start:
+                start = m.dictionary_page_offset
+            end = start + m.total_compressed_size
+            pos = start
+            total_slots += m.num_values
+            while pos < end:
+                h, data_pos = parse_page_header(self.raw, pos)
+                ptype = h.get(1, 0)
+                usz, csz = h[2], h[3]
+                if ptype == 3:
+                    raise RuntimeError(
+                        "LIST + DataPageV2: not GPU-decodable yet")
+                if ptype == 2:
+                    ndict = h.get(7, {}).get(1, 0)
+                    headers.append((data_pos, csz, usz, 0, 0, ndict, rg))
+                elif ptype == 0:
+                    dph = h.get(5, {})
+                    nvals, enc = dph.get(1, 0), dph.get(2, 0)
+                    if enc not in (0, 2, 8):
+                        raise RuntimeError(
+                            f"LIST + encoding {enc}: not GPU-decodable")
+                    headers.append((data_pos, csz, usz, nvals, enc, 0, rg))
+                pos = data_pos + csz
+        scratch_offs = []
+        for (_, _, usz, *_rest) in headers:
+            scratch_offs.append(scratch_total)
+            scratch_total += (usz + 255) & ~255
+        scratch = ctx.alloc(max(scratch_total, 256))
+        snappy_jobs = []
+        for (poff, csz, usz, *_rest), soff in zip(headers, scratch_offs):
+            if codec == "SNAPPY":
+                snappy_jobs.append((poff, csz, soff, usz))
+            else:
+                gpu._check(ctx.L.bg_memcpy_dtod(
+                    ctypes.c_void_p(scratch.ptr.value + soff),
+                    ctypes.c_void_p(chunk.ptr.value + poff),
+                    ctypes.c_uint64(csz)), "bg_memcpy_dtod")
+        if snappy_jobs:
+            arr = (gpu.BgSnappyPage * len(snappy_jobs))()
+            for i, (poff, csz, soff, usz) in enumerate(snappy_jobs):
+                arr[i] = gpu.BgSnappyPage(
+                    ctypes.c_void_p(chunk.ptr.value + poff),
+                    ctypes.c_void_p(scratch.ptr.value + soff), csz, usz)
+            lens = np.zeros(len(snappy_jobs), dtype=np.int64)
+            gpu._check(ctx.L.bg_snappy_decompress(
+                arr, ctypes.c_int64(len(snappy_jobs)),
+                lens.ctypes.data_as(ctypes.POINTER(ctypes.c_int64))),
+                "bg_snappy_decompress")
+            for i, (_, _, _, usz) in enumerate(snappy_jobs):
+                if lens[i] != usz:
+                    raise RuntimeError(f"snappy page {i} failed")
+
+        data_idx = [i for i, h in enumerate(headers) if h[3] > 0]
+        npages = len(data_idx)
+        # pass 1: per-page {rows, entries, present, rlen}
+        counts = ctx.alloc(max(32 * npages, 32))
+        jobs = (gpu.BgListLevelsJob * npages)()
+        for j, i in enumerate(data_idx):
+            (_, _, usz, nslots, enc, _, rg) = headers[i]
+            jobs[j] = gpu.BgListLevelsJob(
+                ctypes.c_void_p(scratch.ptr.value + scratch_offs[i]).value,
+                usz, nslots, max_def, def_entry, def_valid, 0, 0, 0,
+                ctypes.c_void_p(counts.ptr.value + 32 * j).value,
+                None, None, None, None)
+        gpu._check(ctx.L.bg_list_levels_batch(
+            jobs, ctypes.c_int64(npages), ctypes.c_int32(1)),
+            "bg_list_levels_batch(1)")
+        ctx.synchronize()
+        cts = counts.download(np.int64, 4 * npages).reshape(npages, 4)
+        rows_pp, ents_pp, pres_pp, rlen_pp = (cts[:, 0], cts[:, 1],
+                                              cts[:, 2], cts[:, 3])
+        n_rows = int(rows_pp.sum())
+        n_entries = int(ents_pp.sum())
+        row_bases = np.concatenate(([0], np.cumsum(rows_pp)))
+        ent_bases = np.concatenate(([0], np.cumsum(ents_pp)))
+
+        row_sizes = ctx.upload(np.zeros(max(n_rows, 1), dtype=np.int32))
+        lv_words = (n_rows + 31) // 32
+        ev_words = (n_entries + 31) // 32
+        list_valid = ctx.upload(np.zeros(max(lv_words, 1),
+                                         dtype=np.uint32)) \
+            if list_nullable else None
+        elem_valid = ctx.upload(np.zeros(max(ev_words, 1), dtype=np.uint32))
+        vidx = ctx.alloc(max(4 * n_entries, 4))
+        all_valid = ctx.upload(np.full(max(lv_words, ev_words, 1),
+                                       0xffffffff, dtype=np.uint32))
+        # pass 2
+        for j, i in enumerate(data_idx):
+            (_, _, usz, nslots, enc, _, rg) = headers[i]
+            jobs[j] = gpu.BgListLevelsJob(
+                ctypes.c_void_p(scratch.ptr.value + scratch_offs[i]).value,
+                usz, nslots, max_def, def_entry, def_valid, 0,
+                int(row_bases[j]), int(ent_bases[j]), None,
+                row_sizes.ptr.value,
+                (list_valid if list_valid is not None else all_valid).ptr.value,
+                elem_valid.ptr.value,
+                ctypes.c_void_p(vidx.ptr.value + 4 * int(ent_bases[j]))
+                .value)
+        gpu._check(ctx.L.bg_list_levels_batch(
+            jobs, ctypes.c_int64(npages), ctypes.c_int32(2)),
+            "bg_list_levels_batch(2)")
+
+        # element values (entry space) — mode-2 extract/dict re-pointed
+        # past [u32 rlen][rep]
+        out = ctx.alloc(max(n_entries * esz, esz))
+        npres = ctx.upload(pres_pp.astype(np.int64)) if npages else None
+        plain, dict_pages = [], []
+        for j, i in enumerate(data_idx):
+            (_, _, usz, nslots, enc, _, rg) = headers[i]
+            shift = 4 + int(rlen_pp[j])
+            tup = (scratch_offs[i] + shift, usz - shift,
+                   int(ent_bases[j]), int(ents_pp[j]), j, rg)
+            (plain if enc == 0 else dict_pages).append(tup)
+        if plain:
+            ejobs = (gpu.BgPageExtractJob * len(plain))()
+            for i, (soff, plen, ebase, ents, j, rg) in enumerate(plain):
+                ejobs[i] = gpu.BgPageExtractJob(
+                    ctypes.c_void_p(scratch.ptr.value + soff).value,
+                    ctypes.c_void_p(out.ptr.value + ebase * esz).value,
+                    plen, ents, esz, 2, 0,
+                    ctypes.c_void_p(vidx.ptr.value + 4 * ebase).value,
+                    ctypes.c_void_p(npres.ptr.value + 8 * j).value)
+            gpu._check(ctx.L.bg_page_extract_batch(
+                ejobs, ctypes.c_int64(len(plain))),
+                "bg_page_extract_batch(list)")
+        if dict_pages:
+            dict_bufs = {}
+            for i, h in enumerate(headers):
+                if h[5] > 0:  # dictionary page of rg h[6]
+                    dbuf = ctx.alloc(max(h[5] * esz, esz))
+                    j1 = (gpu.BgPageExtractJob * 1)(gpu.BgPageExtractJob(
+                        ctypes.c_void_p(scratch.ptr.value
+                                        + scratch_offs[i]).value,
+                        dbuf.ptr.value, h[2], h[5], esz, 0, 0))
+                    gpu._check(ctx.L.bg_page_extract_batch(
+                        j1, ctypes.c_int64(1)), "bg_page_extract(ldict)")
+                    dict_bufs[h[6]] = dbuf
+            nidx = sum(p[3] for p in dict_pages)
+            idx = ctx.alloc(max(4 * nidx, 4))
+            dense = ctx.alloc(max(4 * nidx, 4))
+            djobs = (gpu.BgDictIndicesJob * len(dict_pages))()
+            run = 0
+            gathers = []
+            for i, (soff, plen, ebase, ents, j, rg) in \
+                    enumerate(dict_pages):
+                djobs[i] = gpu.BgDictIndicesJob(
+                    ctypes.c_void_p(scratch.ptr.value + soff).value,
+                    ctypes.c_void_p(idx.ptr.value + 4 * run).value,
+                    plen, ents, 2, 0,
+                    ctypes.c_void_p(vidx.ptr.value + 4 * ebase).value,
+                    ctypes.c_void_p(npres.ptr.value + 8 * j).value,
+                    ctypes.c_void_p(dense.ptr.value + 4 * run).value)
+                gathers.append((rg, run, ebase, ents))
+                run += ents
+            gpu._check(ctx.L.bg_dict_indices_batch(
+                djobs, ctypes.c_int64(len(dict_pages))),
+                "bg_dict_indices_batch(list)")
+            for rg, ioff, ebase, ents in gathers:
+                gpu._check(ctx.L.bg_gather(
+                    dict_bufs[rg].ptr, ctypes.c_int64(esz),
+                    ctypes.c_void_p(idx.ptr.value + 4 * ioff),
+                    ctypes.c_int64(ents),
+                    ctypes.c_void_p(out.ptr.value + ebase * esz)),
+                    "bg_gather(ldict)")
+        ctx.synchronize()
+        sizes = row_sizes.download(np.int32, max(n_rows, 1))[:n_rows]
+        offsets = np.zeros(n_rows + 1, dtype=np.int32)
+        np.cumsum(sizes, out=offsets[1:])
+        assert offsets[-1] == n_entries, (offsets[-1], n_entries)
+        return {"offsets": offsets, "n_rows": n_rows,
+                "n_entries": n_entries, "values": out, "phys": phys,
+                "elem_valid": elem_valid, "list_valid": list_valid,
+                "row_sizes": row_sizes}

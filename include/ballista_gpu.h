@@ -430,6 +430,32 @@ typedef struct {
 } bg_def_levels_job;
 int bg_def_levels_batch(const void* h_jobs, int64_t njobs);
 
+/* LIST columns (repetition levels, max_rep == 1; parquet-format.md
+ * "Nested Encoding" / the reference's arrow-rs list reader).  A V1 page
+ * body is [u32 rlen][rep][u32 dlen][def][values].  pass 1 fills
+ * d_counts[4] = {rows, entries, present, rlen} per page; the host
+ * prefix-sums bases and re-points the UNCHANGED mode-2 extract/dict
+ * jobs at page + 4 + rlen.  pass 2 emits per-row entry counts (list
+ * offsets = their prefix sum), the list/element validity bitmaps and
+ * the page-local value index per entry.  Rows must not span pages. */
+typedef struct {
+  const void* d_page;   /* [u32 rlen][rep][u32 dlen][def][values] */
+  int64_t page_len;
+  int64_t nslots;       /* level entries (page header num_values) */
+  int32_t max_def;
+  int32_t def_entry;    /* min def meaning an element slot exists */
+  int32_t def_valid;    /* min def meaning the LIST is non-null */
+  int32_t _pad;
+  int64_t row_base;     /* pass 2: column-global bases */
+  int64_t entry_base;
+  int64_t* d_counts;    /* pass 1 */
+  int32_t* d_row_sizes; /* pass 2 (column-global) */
+  uint32_t* d_list_valid;
+  uint32_t* d_elem_valid;
+  uint32_t* d_vidx;     /* pass 2: PAGE-LOCAL slice */
+} bg_list_levels_job;
+int bg_list_levels_batch(const void* h_jobs, int64_t njobs, int32_t pass);
+
 /* BYTE_ARRAY (Utf8/Binary) PLAIN pages (parquet spec PLAIN: [u32 len]
  * [bytes] per value; the reference's decoding.rs PlainDecoder):
  * per page, record each slot's length + absolute device source address
