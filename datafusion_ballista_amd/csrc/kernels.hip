@@ -4706,6 +4706,7 @@ struct LvlRd {
   int64_t rle_left = 0;
   int rle_val = 0;
   int64_t bp_left = 0;
+  int64_t bp_bytes = 0;  // bytes remaining in the current bit-packed run
   u64 bitbuf = 0;
   int bits = 0;
   bool ok = true;
@@ -4717,9 +4718,12 @@ struct LvlRd {
       }
       if (bp_left > 0) {
         if (bits < w) {
-          while (bits <= 56 && d < end) {
+          // refill ONLY from this run's own bytes — reading ahead would
+          // swallow the next run's header
+          while (bits <= 56 && bp_bytes > 0 && d < end) {
             bitbuf |= (u64)(*d++) << bits;
             bits += 8;
+            --bp_bytes;
           }
           if (bits < w) {
             ok = false;
@@ -4746,6 +4750,7 @@ struct LvlRd {
       }
       if (h & 1) {
         bp_left = (int64_t)(h >> 1) * 8;  // groups of 8 w-bit values
+        bp_bytes = (int64_t)(h >> 1) * w;
         bitbuf = 0;
         bits = 0;
       } else {
