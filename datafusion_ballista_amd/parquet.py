@@ -8,11 +8,14 @@ ON DEVICE (bg_snappy_decompress + bg_page_extract); the host only parses
 footer/page metadata (thrift) and drives the copies — the same split as the
 shuffle path (metadata host-side, bytes device-side).
 
-Out of scope this round (fails loudly): V2 data pages, dictionary-encoded
-data pages, BYTE_ARRAY strings, other codecs.  Nullable (OPTIONAL,
-max_def=1) columns decode on device: definition levels -> Arrow validity
-bitmap + slot->value scatter (bg_def_levels_batch + mode-2 extract/dict
-jobs).
+Coverage: V1 + V2 data pages, PLAIN + dictionary encodings, the three
+DELTA encodings + BYTE_STREAM_SPLIT, BYTE_ARRAY strings, FLBA decimals,
+nullable columns (definition levels -> Arrow validity bitmap + slot->value
+scatter on device), and LIST<primitive> columns (repetition + definition
+levels walked on device by bg_list_levels_batch; read_list_column_all).
+Fails loudly: deeper nesting (max_rep > 1), LIST + V2/strings, codecs
+beyond SNAPPY/UNCOMPRESSED/ZSTD/GZIP (the latter two via an explicit
+host-codec bridge).
 
 The thrift compact-protocol PageHeader parser below restates the published
 parquet-format spec (PageHeader/DataPageHeader structs); parity is pinned
