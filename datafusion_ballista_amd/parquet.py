@@ -684,9 +684,16 @@ class GpuParquetColumnReader:
         if codec not in ("SNAPPY", "UNCOMPRESSED"):
             raise RuntimeError(f"LIST + codec {codec}: not GPU-decodable")
         phys = meta.physical_type
-        if phys not in _PHYS_NP:
+        ba = phys == "BYTE_ARRAY"
+        flba = phys == "FIXED_LEN_BYTE_ARRAY"
+        if ba:
+            src_esz, esz = 0, 0
+        elif flba:
+            src_esz, esz = self.pf.schema.column(col).length, 16
+        elif phys in _PHYS_NP:
+            src_esz = esz = _PHYS_NP[phys][1]
+        else:
             raise RuntimeError(f"LIST of {phys}: not GPU-decodable yet")
-        esz = _PHYS_NP[phys][1]
 
         chunk = self._file_buf
         headers = []  # (soff, usz, nslots, enc, ndict_or_0, rg)
@@ -800,8 +807,14 @@ class GpuParquetColumnReader:
             "bg_list_levels_batch(2)")
 
         # element values (entry space) — mode-2 extract/dict re-pointed
-        # past [u32 rlen][rep]
-        out = ctx.alloc(max(n_entries * esz, esz))
+        # past [u32 rlen][rep]; BYTE_ARRAY elements via the same
+        # lens/addr staging + materialize the flat path uses
+        if ba:
+            out = None
+            ba_lens = ctx.alloc(max(8 * n_entries, 8))
+            ba_srcaddr = ctx.alloc(max(8 * n_entries, 8))
+        else:
+            out = ctx.alloc(max(n_entries * esz, esz))
         npres = ctx.upload(pres_pp.astype(np.int64)) if npages else None
         plain, dict_pages = [], []
         for j, i in enumerate(data_idx):
@@ -810,13 +823,27 @@ class GpuParquetColumnReader:
             tup = (scratch_offs[i] + shift, usz - shift,
                    int(ent_bases[j]), int(ents_pp[j]), j, rg)
             (plain if enc == 0 else dict_pages).append(tup)
-        if plain:
+        if plain and ba:
+            bjobs = (gpu.BgBaPageJob * len(plain))()
+            for i, (soff, plen, ebase, ents, j, rg) in enumerate(plain):
+                bjobs[i] = gpu.BgBaPageJob(
+                    ctypes.c_void_p(scratch.ptr.value + soff).value,
+                    ctypes.c_void_p(ba_lens.ptr.value + 8 * ebase).value,
+                    ctypes.c_void_p(ba_srcaddr.ptr.value
+                                    + 8 * ebase).value,
+                    plen, ents, 2, 0,
+                    ctypes.c_void_p(vidx.ptr.value + 4 * ebase).value,
+                    ctypes.c_void_p(npres.ptr.value + 8 * j).value)
+            gpu._check(ctx.L.bg_ba_extract_batch(
+                bjobs, ctypes.c_int64(len(plain))),
+                "bg_ba_extract_batch(list)")
+        elif plain:
             ejobs = (gpu.BgPageExtractJob * len(plain))()
             for i, (soff, plen, ebase, ents, j, rg) in enumerate(plain):
                 ejobs[i] = gpu.BgPageExtractJob(
                     ctypes.c_void_p(scratch.ptr.value + soff).value,
                     ctypes.c_void_p(out.ptr.value + ebase * esz).value,
-                    plen, ents, esz, 2, 0,
+                    plen, ents, src_esz, 2, 1 if flba else 0,
                     ctypes.c_void_p(vidx.ptr.value + 4 * ebase).value,
                     ctypes.c_void_p(npres.ptr.value + 8 * j).value)
             gpu._check(ctx.L.bg_page_extract_batch(
@@ -826,14 +853,42 @@ class GpuParquetColumnReader:
             dict_bufs = {}
             for i, h in enumerate(headers):
                 if h[5] > 0:  # dictionary page of rg h[6]
-                    dbuf = ctx.alloc(max(h[5] * esz, esz))
-                    j1 = (gpu.BgPageExtractJob * 1)(gpu.BgPageExtractJob(
-                        ctypes.c_void_p(scratch.ptr.value
-                                        + scratch_offs[i]).value,
-                        dbuf.ptr.value, h[2], h[5], esz, 0, 0))
-                    gpu._check(ctx.L.bg_page_extract_batch(
-                        j1, ctypes.c_int64(1)), "bg_page_extract(ldict)")
-                    dict_bufs[h[6]] = dbuf
+                    if ba:
+                        dsoff, dusz, ndict = scratch_offs[i], h[2], h[5]
+                        dl = ctx.alloc(max(8 * ndict, 8))
+                        da = ctx.alloc(max(8 * ndict, 8))
+                        j1 = (gpu.BgBaPageJob * 1)(gpu.BgBaPageJob(
+                            ctypes.c_void_p(scratch.ptr.value
+                                            + dsoff).value,
+                            dl.ptr.value, da.ptr.value, dusz, ndict,
+                            0, 0, None, None))
+                        gpu._check(ctx.L.bg_ba_extract_batch(
+                            j1, ctypes.c_int64(1)), "bg_ba_extract(ldict)")
+                        doffs = ctx.alloc(max(4 * (ndict + 1), 8))
+                        tot = ctypes.c_int64()
+                        gpu._check(ctx.L.bg_ba_materialize(
+                            dl.ptr, da.ptr, ctypes.c_int64(ndict),
+                            doffs.ptr, None, ctypes.c_int64(0),
+                            ctypes.byref(tot)), "bg_ba_materialize(lsize)")
+                        ddata = ctx.alloc(max(tot.value, 1))
+                        gpu._check(ctx.L.bg_ba_materialize(
+                            dl.ptr, da.ptr, ctypes.c_int64(ndict),
+                            doffs.ptr, ddata.ptr,
+                            ctypes.c_int64(tot.value),
+                            ctypes.byref(tot)), "bg_ba_materialize(ldict)")
+                        dict_bufs[h[6]] = (doffs, ddata)
+                    else:
+                        dbuf = ctx.alloc(max(h[5] * esz, esz))
+                        j1 = (gpu.BgPageExtractJob * 1)(
+                            gpu.BgPageExtractJob(
+                                ctypes.c_void_p(scratch.ptr.value
+                                                + scratch_offs[i]).value,
+                                dbuf.ptr.value, h[2], h[5], src_esz, 0,
+                                1 if flba else 0))
+                        gpu._check(ctx.L.bg_page_extract_batch(
+                            j1, ctypes.c_int64(1)),
+                            "bg_page_extract(ldict)")
+                        dict_bufs[h[6]] = dbuf
             nidx = sum(p[3] for p in dict_pages)
             idx = ctx.alloc(max(4 * nidx, 4))
             dense = ctx.alloc(max(4 * nidx, 4))
@@ -855,12 +910,37 @@ class GpuParquetColumnReader:
                 djobs, ctypes.c_int64(len(dict_pages))),
                 "bg_dict_indices_batch(list)")
             for rg, ioff, ebase, ents in gathers:
-                gpu._check(ctx.L.bg_gather(
-                    dict_bufs[rg].ptr, ctypes.c_int64(esz),
-                    ctypes.c_void_p(idx.ptr.value + 4 * ioff),
-                    ctypes.c_int64(ents),
-                    ctypes.c_void_p(out.ptr.value + ebase * esz)),
-                    "bg_gather(ldict)")
+                if ba:
+                    doffs, ddata = dict_bufs[rg]
+                    gpu._check(ctx.L.bg_ba_from_dict(
+                        ctypes.c_void_p(idx.ptr.value + 4 * ioff),
+                        doffs.ptr, ddata.ptr,
+                        ctypes.c_void_p(vidx.ptr.value + 4 * ebase),
+                        ctypes.c_int64(ents),
+                        ctypes.c_void_p(ba_lens.ptr.value + 8 * ebase),
+                        ctypes.c_void_p(ba_srcaddr.ptr.value
+                                        + 8 * ebase)),
+                        "bg_ba_from_dict(list)")
+                else:
+                    gpu._check(ctx.L.bg_gather(
+                        dict_bufs[rg].ptr, ctypes.c_int64(esz),
+                        ctypes.c_void_p(idx.ptr.value + 4 * ioff),
+                        ctypes.c_int64(ents),
+                        ctypes.c_void_p(out.ptr.value + ebase * esz)),
+                        "bg_gather(ldict)")
+        if ba:
+            offs32 = ctx.alloc(max(4 * (n_entries + 1), 8))
+            tot = ctypes.c_int64()
+            gpu._check(ctx.L.bg_ba_materialize(
+                ba_lens.ptr, ba_srcaddr.ptr, ctypes.c_int64(n_entries),
+                offs32.ptr, None, ctypes.c_int64(0),
+                ctypes.byref(tot)), "bg_ba_materialize(lsz)")
+            data = ctx.alloc(max(tot.value, 1))
+            gpu._check(ctx.L.bg_ba_materialize(
+                ba_lens.ptr, ba_srcaddr.ptr, ctypes.c_int64(n_entries),
+                offs32.ptr, data.ptr, ctypes.c_int64(tot.value),
+                ctypes.byref(tot)), "bg_ba_materialize(list)")
+            out = (offs32, data, tot.value)
         ctx.synchronize()
         sizes = row_sizes.download(np.int32, max(n_rows, 1))[:n_rows]
         offsets = np.zeros(n_rows + 1, dtype=np.int32)

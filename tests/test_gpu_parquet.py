@@ -465,3 +465,73 @@ def test_parquet_list_no_null_elements(ctx, tmp_path):
                           want.values.to_numpy(zero_copy_only=False))
     with pytest.raises(RuntimeError, match="nested"):
         rd.read_column_all(0)
+
+
+@pytest.mark.parametrize("use_dictionary", [False, True])
+def test_parquet_list_strings_and_decimals(ctx, tmp_path, use_dictionary):
+    """LIST<string> (BYTE_ARRAY elements via the lens/addr staging +
+    materialize path) and LIST<decimal(15,2)> (FLBA big-endian
+    sign-extension) vs pyarrow."""
+    rng = np.random.default_rng(33)
+    n = 20_000
+    words = ["green", "lemon", "", "navajo peru", "x" * 40, "midnight"]
+
+    def mk(valf):
+        out = []
+        for _ in range(n):
+            r = rng.random()
+            if r < 0.10:
+                out.append(None)
+            elif r < 0.20:
+                out.append([])
+            else:
+                out.append([None if rng.random() < 0.15 else valf()
+                            for _ in range(int(rng.integers(1, 6)))])
+        return out
+
+    ls = mk(lambda: words[int(rng.integers(0, len(words)))])
+    dv = mk(lambda: decimal.Decimal(int(rng.integers(-10**9, 10**9))) / 100)
+    table = pa.table({
+        "ls": pa.array(ls, type=pa.list_(pa.string())),
+        "ldec": pa.array(dv, type=pa.list_(pa.decimal128(15, 2))),
+    })
+    path = str(tmp_path / f"lists_sd_{use_dictionary}.parquet")
+    pq.write_table(table, path, compression="snappy",
+                   use_dictionary=use_dictionary,
+                   data_page_size=32 * 1024, row_group_size=n // 2,
+                   write_statistics=False)
+    rd = GpuParquetColumnReader(ctx, path)
+    want_tbl = pq.read_table(path)
+
+    res = rd.read_list_column_all(0)
+    want = want_tbl.column("ls").combine_chunks()
+    w_offs = want.offsets.to_numpy(zero_copy_only=False)
+    assert np.array_equal(res["offsets"], w_offs.astype(np.int32))
+    offs32, data, tot = res["values"]
+    got_offs = offs32.download(np.int32, res["n_entries"] + 1)
+    got_data = data.download(np.uint8, max(tot, 1)).tobytes()
+    w_elems = want.values
+    w_ev = ~np.array(w_elems.is_null())
+    got_ev = _bitmap_bools(
+        res["elem_valid"].download(np.uint32,
+                                   (res["n_entries"] + 31) // 32),
+        res["n_entries"])
+    assert np.array_equal(got_ev, w_ev)
+    for i in range(res["n_entries"]):
+        if w_ev[i]:
+            s = got_data[got_offs[i]:got_offs[i + 1]].decode()
+            assert s == w_elems[i].as_py(), i
+
+    res = rd.read_list_column_all(1)
+    want = want_tbl.column("ldec").combine_chunks()
+    w_offs = want.offsets.to_numpy(zero_copy_only=False)
+    assert np.array_equal(res["offsets"], w_offs.astype(np.int32))
+    got = res["values"].download(np.int64,
+                                 2 * max(res["n_entries"], 1))
+    w_elems = want.values
+    w_ev = ~np.array(w_elems.is_null())
+    for i in range(res["n_entries"]):
+        if w_ev[i]:
+            lo, hi = int(got[2 * i]), int(got[2 * i + 1])
+            v = (hi << 64) | (lo & (2**64 - 1))
+            assert v == int(w_elems[i].as_py() * 100), i
