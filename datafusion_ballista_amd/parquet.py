@@ -11,11 +11,12 @@ shuffle path (metadata host-side, bytes device-side).
 Coverage: V1 + V2 data pages, PLAIN + dictionary encodings, the three
 DELTA encodings + BYTE_STREAM_SPLIT, BYTE_ARRAY strings, FLBA decimals,
 nullable columns (definition levels -> Arrow validity bitmap + slot->value
-scatter on device), and LIST<primitive> columns (repetition + definition
-levels walked on device by bg_list_levels_batch; read_list_column_all).
-Fails loudly: deeper nesting (max_rep > 1), LIST + V2/strings, codecs
+scatter on device), and LIST columns of primitives,
+strings and decimals (repetition + definition levels walked on device by
+bg_list_levels_batch; read_list_column_all).
+Fails loudly: deeper nesting (max_rep > 1), LIST + V2 pages, codecs
 beyond SNAPPY/UNCOMPRESSED/ZSTD/GZIP (the latter two via an explicit
-host-codec bridge).
+host-codec bridge; LIST itself is SNAPPY/UNCOMPRESSED only).
 
 The thrift compact-protocol PageHeader parser below restates the published
 parquet-format spec (PageHeader/DataPageHeader structs); parity is pinned
