@@ -3301,7 +3301,11 @@ __global__ void k_snap_matches(const SnappyPage* __restrict__ pages,
 // serial work drops from ~300k elements to ~128 segment steps.
 // ---------------------------------------------------------------------------
 
-#define SNAPBIG_SEG 8192
+#define SNAPBIG_SEG 8192   /* default; 4096 when few big pages (the
+                               per-segment chase is latency-bound there
+                               and shorter chases halve it; with many
+                               pages the chip is throughput-bound and
+                               larger segments have less overhead) */
 #define SNAPBIG_THRESHOLD 98304  // pages with csz above this use A/B/C
 
 // scalar speculative decode of the element at page-relative offset o.
@@ -3356,6 +3360,7 @@ struct SnapBigPage {
   int64_t seg_base;  // per-seg array base
   int64_t nsegs;
   int64_t par_base;  // parent-array base (output bytes; replay phase)
+  int64_t seg_sz;    // segment size this call (uniform; see SNAPBIG_SEG)
 };
 
 // phase A: speculative per-segment chase
@@ -3379,7 +3384,7 @@ __global__ void k_snapbig_spec(const SnappyPage* __restrict__ pages,
     const SnappyPage& pg = pages[bp.page_idx];
     const uint8_t* s = pg.src;
     const int64_t src_len = pg.src_len;
-    int64_t o = k * SNAPBIG_SEG;
+    int64_t o = k * bp.seg_sz;
     if (k == 0) {
       // parse the uncompressed-length varint; segment 0 starts at the
       // TRUE element boundary
@@ -3404,7 +3409,7 @@ __global__ void k_snapbig_spec(const SnappyPage* __restrict__ pages,
       }
     }
     const int64_t seg_end =
-        (k + 1) * SNAPBIG_SEG < src_len ? (k + 1) * SNAPBIG_SEG : src_len;
+        (k + 1) * bp.seg_sz < src_len ? (k + 1) * bp.seg_sz : src_len;
     uint32_t cnt = 0, outsum = 0;
     int64_t exitv = -1;
     while (true) {
@@ -3478,7 +3483,7 @@ __global__ void k_snapbig_walk(const SnappyPage* __restrict__ pages,
       continue;
     }
     const int64_t seg_end =
-        (k + 1) * SNAPBIG_SEG < src_len ? (k + 1) * SNAPBIG_SEG : src_len;
+        (k + 1) * bp.seg_sz < src_len ? (k + 1) * bp.seg_sz : src_len;
     if (e >= seg_end) {  // candidate passes this segment entirely
       w_land[gs] = -3;
       w_exit[gs] = e;
@@ -3542,7 +3547,7 @@ __global__ void k_snapbig_resolve(const SnappyPage* __restrict__ pages,
     for (int64_t k = 0; k < bp.nsegs && !err; ++k) {
       const int64_t gs = bp.seg_base + k;
       const int64_t seg_end =
-          (k + 1) * SNAPBIG_SEG < src_len ? (k + 1) * SNAPBIG_SEG : src_len;
+          (k + 1) * bp.seg_sz < src_len ? (k + 1) * bp.seg_sz : src_len;
       if (cur >= seg_end || di >= ulen) {
         seg_entry[gs] = -1;
         continue;
@@ -3622,7 +3627,7 @@ __global__ void k_snapbig_emit(const SnappyPage* __restrict__ pages,
     const int64_t src_len = pg.src_len;
     const int64_t ulen = page_ulen[seg_page[gs]];
     const int64_t seg_end =
-        (k + 1) * SNAPBIG_SEG < src_len ? (k + 1) * SNAPBIG_SEG : src_len;
+        (k + 1) * bp.seg_sz < src_len ? (k + 1) * bp.seg_sz : src_len;
     SnapDesc* d = descs + desc_base[bp.page_idx];
     int64_t cur = seg_entry[gs];
     int64_t nd = seg_nd[gs], di = seg_di[gs];
@@ -3827,13 +3832,18 @@ extern "C" int bg_snappy_decompress(const void* h_pages, int64_t npages,
   std::vector<SnapBigPage> bigs;
   std::vector<int32_t> seg_page;
   int64_t arr_total = 0, seg_total = 0;
+  int64_t nbig_est = 0;
+  for (int64_t p = 0; p < npages; ++p)
+    if (is_big(p)) ++nbig_est;
+  const int64_t segsz = nbig_est <= 768 ? SNAPBIG_SEG / 2 : SNAPBIG_SEG;
   for (int64_t p = 0; p < npages; ++p) {
     if (!is_big(p)) continue;
     SnapBigPage bp;
     bp.page_idx = p;
     bp.arr_base = arr_total;
     bp.seg_base = seg_total;
-    bp.nsegs = (hp[p].src_len + SNAPBIG_SEG - 1) / SNAPBIG_SEG;
+    bp.nsegs = (hp[p].src_len + segsz - 1) / segsz;
+    bp.seg_sz = segsz;
     bp.par_base = 0;
     arr_total += hp[p].src_len;
     for (int64_t k = 0; k < bp.nsegs; ++k)
