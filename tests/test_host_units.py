@@ -119,3 +119,87 @@ def test_ipc_stream_empty_batchless():
     stream = bgipc.schema_message_bytes(schema) + bgipc.EOS
     got = pa.ipc.open_stream(pa.BufferReader(stream)).read_all()
     assert got.num_rows == 0 and got.schema.equals(schema)
+
+
+def test_snappy_list_ranking_replay_restatement():
+    """CPU restatement of the giant-page match-replay algorithm
+    (kernels.hip k_snap_par_*): every output byte of a match has exactly
+    ONE parent byte, (dst-off) + ((b-dst) mod off) — periodic/RLE matches
+    included — literal bytes are roots, jump-4 pointer doubling converges
+    in ceil(log4 depth) rounds, and the final gather from literal roots
+    reproduces the stream byte-exactly.  Pinned against pyarrow's own
+    snappy codec on the page shapes that route to this path."""
+    import numpy as np
+    import pyarrow as pa
+
+    rng = np.random.default_rng(13)
+    payloads = [
+        # structured FLBA-like (deep-ish chains)
+        b"".join(int(v).to_bytes(7, "big")
+                 for v in rng.integers(90000, 10495100, size=40_000)),
+        # heavy RLE: repeated 64-B pattern and pure zeros (deepest chains)
+        rng.integers(0, 256, size=64, dtype=np.uint8).tobytes() * 2048,
+        b"\x00" * 100_000,
+        # low-cardinality ints
+        rng.integers(0, 50, size=20_000, dtype=np.int64).tobytes(),
+    ]
+    for raw in payloads:
+        comp = pa.compress(raw, codec="snappy", asbytes=True)
+        # parse the element stream (tags per snappy format)
+        s, i = comp, 0
+        ulen, sh = 0, 0
+        while True:
+            b = s[i]
+            i += 1
+            ulen |= (b & 0x7F) << sh
+            if not (b & 0x80):
+                break
+            sh += 7
+        assert ulen == len(raw)
+        out = bytearray(ulen)
+        parents = np.arange(ulen, dtype=np.int64)  # identity = roots
+        dst = 0
+        while i < len(s):
+            tag = s[i]
+            t = tag & 3
+            if t == 0:
+                ln = (tag >> 2) + 1
+                nb = 0
+                if ln > 60:
+                    nb = ln - 60
+                    ln = int.from_bytes(s[i + 1:i + 1 + nb], "little") + 1
+                out[dst:dst + ln] = s[i + 1 + nb:i + 1 + nb + ln]  # literal
+                dst += ln
+                i += 1 + nb + ln
+                continue
+            if t == 1:
+                ln = ((tag >> 2) & 7) + 4
+                off = ((tag >> 5) << 8) | s[i + 1]
+                i += 2
+            elif t == 2:
+                ln = (tag >> 2) + 1
+                off = s[i + 1] | (s[i + 2] << 8)
+                i += 3
+            else:
+                ln = (tag >> 2) + 1
+                off = int.from_bytes(s[i + 1:i + 5], "little")
+                i += 5
+            assert 0 < off <= dst
+            base = dst - off
+            idx = np.arange(ln, dtype=np.int64)
+            parents[dst:dst + ln] = base + (idx % off)
+            dst += ln
+        assert dst == ulen
+        # jump-4 doubling until fixpoint; rounds must stay logarithmic
+        rounds = 0
+        while True:
+            nxt = parents[parents[parents[parents[parents]]]]
+            rounds += 1
+            if np.array_equal(nxt, parents):
+                break
+            parents = nxt
+            assert rounds < 20
+        # all roots are literal bytes (already written); one gather
+        ob = np.frombuffer(bytes(out), dtype=np.uint8).copy()
+        ob = ob[parents]
+        assert ob.tobytes() == raw
